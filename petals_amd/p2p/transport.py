@@ -1,0 +1,347 @@
+"""Framed-msgpack TCP transport with unary and bidirectional-stream RPCs.
+
+This replaces the reference's go-libp2p daemon (`hivemind.P2P`, spawned per
+process — see SURVEY §2.3) with an in-process asyncio transport. The wire unit
+is a *frame*:
+
+    [u32 big-endian length][msgpack map]
+
+and the msgpack map is an *envelope*:
+
+    {"rpc": str, "rid": int, "kind": "req"|"item"|"end"|"err",
+     "meta": {...},                      # msgpack-able metadata side-channel
+     "tdescs": [tensor descriptors],     # see utils.serialization
+     "tbufs": [bytes, ...]}              # raw tensor payloads
+
+Every RPC is a message stream keyed by `rid`: a unary call is a stream with one
+"req" inbound and one "end" outbound. Handlers receive an `RpcStream` with an
+async-iterator of inbound messages and `send()`/`close()` for outbound ones.
+
+Identity: every node owns a random 16-byte peer id (hex). There is no
+encryption layer (the reference relied on libp2p's); swarm deployments should
+front this with their own network fabric.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import logging
+import os
+import socket
+import struct
+from typing import Any, Awaitable, Callable, Dict, List, Optional, Tuple
+
+import msgpack
+import torch
+
+from petals_amd.utils import serialization
+
+logger = logging.getLogger(__name__)
+
+MAX_FRAME_SIZE = 1 << 30  # 1 GiB hard cap per frame
+# payloads larger than this are split into multiple stream items by callers
+MAX_UNARY_PAYLOAD_SIZE = 64 << 20
+
+
+class RpcError(Exception):
+    pass
+
+
+class RpcMessage:
+    """One envelope: metadata + a list of tensors."""
+
+    __slots__ = ("meta", "tensors", "kind")
+
+    def __init__(self, meta: Optional[Dict[str, Any]] = None, tensors: Optional[List[torch.Tensor]] = None, kind: str = "item"):
+        self.meta = meta or {}
+        self.tensors = tensors or []
+        self.kind = kind
+
+    def __repr__(self):
+        return f"RpcMessage(kind={self.kind}, meta_keys={list(self.meta)}, n_tensors={len(self.tensors)})"
+
+
+def _encode_envelope(rpc: str, rid: int, kind: str, msg: RpcMessage, compressions=None) -> bytes:
+    tdescs, tbufs = serialization.serialize_tensors(msg.tensors, compressions)
+    body = msgpack.packb(
+        {"rpc": rpc, "rid": rid, "kind": kind, "meta": msg.meta, "tdescs": tdescs, "tbufs": tbufs},
+        use_bin_type=True,
+    )
+    return struct.pack(">I", len(body)) + body
+
+
+def _decode_envelope(body: bytes) -> Tuple[str, int, str, RpcMessage]:
+    obj = msgpack.unpackb(body, raw=False, strict_map_key=False)
+    tensors = serialization.deserialize_tensors(obj.get("tdescs", []), obj.get("tbufs", []))
+    msg = RpcMessage(meta=obj.get("meta", {}), tensors=tensors, kind=obj["kind"])
+    return obj["rpc"], obj["rid"], obj["kind"], msg
+
+
+class RpcStream:
+    """Bidirectional message stream for one RPC invocation."""
+
+    def __init__(self, conn: "Connection", rpc: str, rid: int):
+        self._conn = conn
+        self.rpc = rpc
+        self.rid = rid
+        self._inbox: asyncio.Queue = asyncio.Queue()
+        self._closed_outbound = False
+        self._finished_inbound = False
+
+    async def send(self, msg: RpcMessage, kind: str = "item", compressions=None) -> None:
+        if self._closed_outbound:
+            raise RpcError(f"stream {self.rpc}#{self.rid} already closed for sending")
+        if kind == "end":
+            self._closed_outbound = True
+        await self._conn.send_frame(_encode_envelope(self.rpc, self.rid, kind, msg, compressions))
+
+    async def close(self, msg: Optional[RpcMessage] = None) -> None:
+        """Send the final message (or an empty end marker)."""
+        if not self._closed_outbound:
+            await self.send(msg or RpcMessage(), kind="end")
+
+    async def error(self, text: str) -> None:
+        if not self._closed_outbound:
+            self._closed_outbound = True
+            await self._conn.send_frame(
+                _encode_envelope(self.rpc, self.rid, "err", RpcMessage(meta={"error": text}))
+            )
+
+    def _feed(self, msg: RpcMessage) -> None:
+        self._inbox.put_nowait(msg)
+
+    def _feed_eof(self, exc_text: Optional[str] = None) -> None:
+        if not self._finished_inbound:
+            self._finished_inbound = True
+            self._inbox.put_nowait(exc_text if exc_text is not None else StopAsyncIteration)
+
+    async def receive(self, timeout: Optional[float] = None) -> RpcMessage:
+        """Next inbound message; raises RpcError on remote error / EOF."""
+        item = await asyncio.wait_for(self._inbox.get(), timeout)
+        if item is StopAsyncIteration:
+            self._inbox.put_nowait(StopAsyncIteration)  # keep EOF sticky
+            raise RpcError(f"stream {self.rpc}#{self.rid}: closed")
+        if isinstance(item, str):
+            self._inbox.put_nowait(item)
+            raise RpcError(f"stream {self.rpc}#{self.rid}: remote error: {item}")
+        return item
+
+    def __aiter__(self):
+        return self
+
+    async def __anext__(self) -> RpcMessage:
+        item = await self._inbox.get()
+        if item is StopAsyncIteration:
+            self._inbox.put_nowait(StopAsyncIteration)
+            raise StopAsyncIteration
+        if isinstance(item, str):
+            self._inbox.put_nowait(item)
+            raise RpcError(f"stream {self.rpc}#{self.rid}: remote error: {item}")
+        if item.kind == "end":
+            self._finished_inbound = True
+            # deliver the final message, next iteration stops
+            self._inbox.put_nowait(StopAsyncIteration)
+        return item
+
+
+RpcHandler = Callable[[RpcMessage, RpcStream], Awaitable[None]]
+
+
+class Connection:
+    """One TCP connection; multiplexes many RPC streams."""
+
+    def __init__(self, node: "P2PNode", reader: asyncio.StreamReader, writer: asyncio.StreamWriter):
+        self.node = node
+        self.reader = reader
+        self.writer = writer
+        self.streams: Dict[int, RpcStream] = {}
+        self._rid_counter = 0
+        self._send_lock = asyncio.Lock()
+        self._closed = False
+        self._reader_task: Optional[asyncio.Task] = None
+
+    def start(self):
+        self._reader_task = asyncio.get_event_loop().create_task(self._read_loop())
+
+    @property
+    def is_closed(self) -> bool:
+        return self._closed
+
+    def next_rid(self) -> int:
+        self._rid_counter += 2  # client uses even rids, 0 reserved
+        return self._rid_counter
+
+    async def send_frame(self, frame: bytes) -> None:
+        if self._closed:
+            raise RpcError("connection closed")
+        async with self._send_lock:
+            self.writer.write(frame)
+            await self.writer.drain()
+
+    async def _read_loop(self):
+        exc_text = None
+        try:
+            while True:
+                header = await self.reader.readexactly(4)
+                (length,) = struct.unpack(">I", header)
+                if length > MAX_FRAME_SIZE:
+                    raise RpcError(f"frame too large: {length}")
+                body = await self.reader.readexactly(length)
+                rpc, rid, kind, msg = _decode_envelope(body)
+                if kind in ("req", "req_end"):
+                    msg.kind = "end" if kind == "req_end" else "item"
+                    stream = RpcStream(self, rpc, rid)
+                    self.streams[rid] = stream
+                    handler = self.node.handlers.get(rpc)
+                    if handler is None:
+                        await stream.error(f"unknown rpc {rpc!r}")
+                        continue
+                    asyncio.get_event_loop().create_task(self._run_handler(handler, msg, stream))
+                else:
+                    stream = self.streams.get(rid)
+                    if stream is None:
+                        continue  # late message for a finished stream
+                    if kind == "err":
+                        stream._feed_eof(msg.meta.get("error", "remote error"))
+                        self.streams.pop(rid, None)
+                    else:
+                        stream._feed(msg)
+                        if kind == "end":
+                            self.streams.pop(rid, None)
+        except (asyncio.IncompleteReadError, ConnectionResetError, BrokenPipeError, OSError) as e:
+            exc_text = f"connection lost: {type(e).__name__}"
+        except Exception as e:  # noqa: BLE001
+            exc_text = f"transport error: {e!r}"
+            logger.exception("transport read loop failed")
+        finally:
+            await self._shutdown(exc_text or "connection closed")
+
+    async def _run_handler(self, handler: RpcHandler, first: RpcMessage, stream: RpcStream):
+        try:
+            stream._feed(first)
+            if first.kind == "end":
+                stream._feed_eof()
+            await handler(first, stream)
+            await stream.close()
+        except asyncio.CancelledError:
+            raise
+        except Exception as e:  # noqa: BLE001
+            logger.debug("rpc handler %s failed: %r", stream.rpc, e, exc_info=True)
+            try:
+                await stream.error(f"{type(e).__name__}: {e}")
+            except Exception:  # noqa: BLE001
+                pass
+        finally:
+            self.streams.pop(stream.rid, None)
+
+    async def _shutdown(self, reason: str):
+        if self._closed:
+            return
+        self._closed = True
+        for stream in list(self.streams.values()):
+            stream._feed_eof(reason)
+        self.streams.clear()
+        try:
+            self.writer.close()
+        except Exception:  # noqa: BLE001
+            pass
+        self.node._forget_connection(self)
+
+    async def close(self):
+        if self._reader_task is not None:
+            self._reader_task.cancel()
+        await self._shutdown("closed locally")
+
+
+class P2PNode:
+    """A peer: can listen for inbound RPCs and open outbound RPC streams.
+
+    Replaces the reference's per-process `p2pd` daemon + hivemind.P2P wrapper.
+    """
+
+    def __init__(self, peer_id: Optional[str] = None):
+        self.peer_id = peer_id or os.urandom(16).hex()
+        self.handlers: Dict[str, RpcHandler] = {}
+        self._server: Optional[asyncio.base_events.Server] = None
+        self.listen_addr: Optional[Tuple[str, int]] = None
+        self._conns: Dict[Tuple[str, int], Connection] = {}
+        self._inbound: List[Connection] = []
+        self._conn_locks: Dict[Tuple[str, int], asyncio.Lock] = {}
+
+    def add_handler(self, rpc: str, handler: RpcHandler) -> None:
+        self.handlers[rpc] = handler
+
+    async def listen(self, host: str = "127.0.0.1", port: int = 0) -> Tuple[str, int]:
+        self._server = await asyncio.start_server(self._on_inbound, host=host, port=port)
+        sockets = self._server.sockets
+        addr = sockets[0].getsockname()
+        self.listen_addr = (host, addr[1])
+        return self.listen_addr
+
+    async def _on_inbound(self, reader: asyncio.StreamReader, writer: asyncio.StreamWriter):
+        conn = Connection(self, reader, writer)
+        self._inbound.append(conn)
+        conn.start()
+
+    def _forget_connection(self, conn: Connection):
+        for key, val in list(self._conns.items()):
+            if val is conn:
+                self._conns.pop(key, None)
+        if conn in self._inbound:
+            self._inbound.remove(conn)
+
+    async def connect(self, addr: Tuple[str, int], timeout: float = 10.0) -> Connection:
+        addr = (addr[0], int(addr[1]))
+        lock = self._conn_locks.setdefault(addr, asyncio.Lock())
+        async with lock:
+            conn = self._conns.get(addr)
+            if conn is not None and not conn.is_closed:
+                return conn
+            reader, writer = await asyncio.wait_for(asyncio.open_connection(*addr), timeout)
+            sock = writer.get_extra_info("socket")
+            if sock is not None:
+                sock.setsockopt(socket.IPPROTO_TCP, socket.TCP_NODELAY, 1)
+            conn = Connection(self, reader, writer)
+            conn.start()
+            self._conns[addr] = conn
+            return conn
+
+    async def open_stream(
+        self,
+        addr: Tuple[str, int],
+        rpc: str,
+        request: RpcMessage,
+        timeout: float = 10.0,
+        compressions=None,
+        end: bool = False,
+    ) -> RpcStream:
+        """Send the initial request message and return the live stream.
+
+        With ``end=True`` the request also closes our outbound side (unary
+        request)."""
+        conn = await self.connect(addr, timeout)
+        rid = conn.next_rid()
+        stream = RpcStream(conn, rpc, rid)
+        conn.streams[rid] = stream
+        if end:
+            stream._closed_outbound = True
+        await conn.send_frame(_encode_envelope(rpc, rid, "req_end" if end else "req", request, compressions))
+        return stream
+
+    async def call_unary(
+        self, addr: Tuple[str, int], rpc: str, request: RpcMessage, timeout: float = 30.0, compressions=None
+    ) -> RpcMessage:
+        stream = await self.open_stream(
+            addr, rpc, request, timeout=timeout, compressions=compressions, end=True
+        )
+        return await stream.receive(timeout=timeout)
+
+    async def shutdown(self):
+        for conn in list(self._conns.values()) + list(self._inbound):
+            await conn.close()
+        if self._server is not None:
+            self._server.close()
+            try:
+                await self._server.wait_closed()
+            except Exception:  # noqa: BLE001
+                pass

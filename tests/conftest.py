@@ -1,0 +1,17 @@
+import os
+import sys
+
+import pytest
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
+def pytest_configure(config):
+    config.addinivalue_line("markers", "gpu: test requires an MI355X GPU (run via gpurun)")
+
+
+@pytest.fixture(scope="session")
+def event_loop_policy():
+    import asyncio
+
+    return asyncio.DefaultEventLoopPolicy()
