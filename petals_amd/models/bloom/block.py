@@ -1,0 +1,88 @@
+"""Native BLOOM decoder block: LayerNorm -> fused-QKV attention with ALiBi ->
+LayerNorm -> GELU MLP. Weight names match HF BLOOM per-layer state dicts
+(contrast: reference models/bloom/block.py wraps HF's BloomBlock).
+
+HF fused-QKV layout: query_key_value output reshapes to
+[batch, seq, n_heads, 3, head_dim] (per-head interleave of q,k,v).
+"""
+
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import torch
+from torch import nn
+
+from petals_amd import ops
+from petals_amd.models.bloom.config import BloomConfig
+
+
+class BloomAttention(nn.Module):
+    def __init__(self, config: BloomConfig):
+        super().__init__()
+        self.num_heads = config.num_attention_heads
+        self.head_dim = config.head_dim
+        self.hidden_size = config.hidden_size
+        self.query_key_value = nn.Linear(self.hidden_size, 3 * self.hidden_size, bias=True)
+        self.dense = nn.Linear(self.hidden_size, self.hidden_size, bias=True)
+        self.register_buffer("alibi_slopes", ops.build_alibi_slopes(self.num_heads), persistent=False)
+
+    def forward(self, hidden_states, kv_cache=None, prefix_length: int = 0):
+        b, q_len, _ = hidden_states.shape
+        fused = self.query_key_value(hidden_states).view(b, q_len, self.num_heads, 3, self.head_dim)
+        q = fused[..., 0, :].transpose(1, 2)  # [b, heads, q_len, hd]
+        k = fused[..., 1, :].transpose(1, 2)
+        v = fused[..., 2, :].transpose(1, 2)
+
+        end = prefix_length + q_len
+        slopes = self.alibi_slopes.to(hidden_states.device)
+        if kv_cache is not None:
+            k_cache, v_cache = kv_cache
+            k_cache[:b, :, prefix_length:end].copy_(k)
+            v_cache[:b, :, prefix_length:end].copy_(v)
+            k_pos = torch.arange(end, device=hidden_states.device, dtype=torch.float32)
+            bias = (slopes[:, None, None] * k_pos[None, None, :]).unsqueeze(0)
+            attn = ops.attention_decode(q, k_cache[:b], v_cache[:b], end, attn_bias=bias)
+        else:
+            assert prefix_length == 0
+            k_pos = torch.arange(q_len, device=hidden_states.device, dtype=torch.float32)
+            bias = (slopes[:, None, None] * k_pos[None, None, :]).unsqueeze(0)
+            attn = ops.attention(q, k, v, causal=True, attn_bias=bias)
+        attn = attn.transpose(1, 2).reshape(b, q_len, self.hidden_size)
+        return self.dense(attn)
+
+
+class BloomMLP(nn.Module):
+    def __init__(self, config: BloomConfig):
+        super().__init__()
+        self.dense_h_to_4h = nn.Linear(config.hidden_size, config.intermediate_size, bias=True)
+        self.dense_4h_to_h = nn.Linear(config.intermediate_size, config.hidden_size, bias=True)
+
+    def forward(self, x):
+        return self.dense_4h_to_h(ops.gelu(self.dense_h_to_4h(x)))
+
+
+class BloomBlock(nn.Module):
+    def __init__(self, config: BloomConfig, layer_idx: int = 0):
+        super().__init__()
+        self.config = config
+        self.layer_idx = layer_idx
+        self.input_layernorm = nn.LayerNorm(config.hidden_size, eps=config.layer_norm_eps)
+        self.self_attention = BloomAttention(config)
+        self.post_attention_layernorm = nn.LayerNorm(config.hidden_size, eps=config.layer_norm_eps)
+        self.mlp = BloomMLP(config)
+        self.apply_residual_post_ln = config.apply_residual_connection_post_layernorm
+
+    def forward(self, hidden_states, kv_cache=None, prefix_length: int = 0):
+        ln_out = self.input_layernorm(hidden_states)
+        residual = ln_out if self.apply_residual_post_ln else hidden_states
+        attn = self.self_attention(ln_out, kv_cache=kv_cache, prefix_length=prefix_length)
+        hidden_states = residual + attn
+
+        ln_out = self.post_attention_layernorm(hidden_states)
+        residual = ln_out if self.apply_residual_post_ln else hidden_states
+        return residual + self.mlp(ln_out)
+
+    def kv_cache_shape(self, batch_size: int, max_length: int) -> Tuple[Tuple[int, ...], Tuple[int, ...]]:
+        shape = (batch_size, self.config.num_attention_heads, max_length, self.config.head_dim)
+        return shape, shape

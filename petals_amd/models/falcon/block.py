@@ -1,0 +1,128 @@
+"""Native Falcon decoder block (7B MQA and 40B/180B GQA new-decoder variants).
+
+Weight names match HF Falcon per-layer state dicts. Fused-QKV layouts:
+  * new_decoder_architecture: view [b, s, n_kv, n_heads//n_kv + 2, hd] — per
+    kv-group: (n_heads//n_kv) query heads, then 1 key and 1 value head
+  * multi_query (falcon-7b): view [b, s, n_heads + 2, hd]
+(contrast: reference models/falcon/block.py wraps HF FalconDecoderLayer and
+CUDA-graphs the QKV split; here the split is plain views and the decode fast
+path is the fused HIP kernel.)
+"""
+
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import torch
+from torch import nn
+
+from petals_amd import ops
+from petals_amd.models.falcon.config import FalconConfig
+
+
+class FalconAttention(nn.Module):
+    def __init__(self, config: FalconConfig):
+        super().__init__()
+        self.config = config
+        self.num_heads = config.num_attention_heads
+        self.num_kv = config.n_kv_heads
+        self.head_dim = config.head_dim
+        h = config.hidden_size
+        if config.new_decoder_architecture:
+            qkv_out = (self.num_heads + 2 * self.num_kv) * self.head_dim
+        elif config.multi_query:
+            qkv_out = h + 2 * self.head_dim
+        else:
+            qkv_out = 3 * h
+        self.query_key_value = nn.Linear(h, qkv_out, bias=config.bias)
+        self.dense = nn.Linear(self.num_heads * self.head_dim, h, bias=config.bias)
+        cos, sin = ops.build_rope_cache(self.head_dim, config.max_position_embeddings, theta=config.rope_theta)
+        self.register_buffer("rope_cos", cos, persistent=False)
+        self.register_buffer("rope_sin", sin, persistent=False)
+
+    def _split_qkv(self, fused: torch.Tensor, b: int, q_len: int):
+        if self.config.new_decoder_architecture:
+            fused = fused.view(b, q_len, self.num_kv, self.num_heads // self.num_kv + 2, self.head_dim)
+            q = fused[..., :-2, :].reshape(b, q_len, self.num_heads, self.head_dim)
+            k = fused[..., -2, :]
+            v = fused[..., -1, :]
+        elif self.config.multi_query:
+            fused = fused.view(b, q_len, self.num_heads + 2, self.head_dim)
+            q = fused[..., :-2, :]
+            k = fused[..., -2:-1, :]
+            v = fused[..., -1:, :]
+        else:
+            fused = fused.view(b, q_len, self.num_heads, 3, self.head_dim)
+            q, k, v = fused[..., 0, :], fused[..., 1, :], fused[..., 2, :]
+        return (x.transpose(1, 2) for x in (q, k, v))  # [b, heads, len, hd]
+
+    def _ensure_rope(self, needed: int, device):
+        if self.rope_cos.shape[0] < needed:
+            cos, sin = ops.build_rope_cache(self.head_dim, max(needed, 2 * self.rope_cos.shape[0]), theta=self.config.rope_theta)
+            self.rope_cos, self.rope_sin = cos.to(device), sin.to(device)
+        elif self.rope_cos.device != device:
+            self.rope_cos, self.rope_sin = self.rope_cos.to(device), self.rope_sin.to(device)
+
+    def forward(self, hidden_states, kv_cache=None, prefix_length: int = 0):
+        b, q_len, _ = hidden_states.shape
+        q, k, v = self._split_qkv(self.query_key_value(hidden_states), b, q_len)
+        end = prefix_length + q_len
+        self._ensure_rope(end, hidden_states.device)
+        pos = torch.arange(prefix_length, end, device=hidden_states.device)
+        q, k = ops.apply_rope(q.contiguous(), k.contiguous(), self.rope_cos, self.rope_sin, pos)
+        if kv_cache is not None:
+            k_cache, v_cache = kv_cache
+            k_cache[:b, :, prefix_length:end].copy_(k)
+            v_cache[:b, :, prefix_length:end].copy_(v)
+            attn = ops.attention_decode(q, k_cache[:b], v_cache[:b], end)
+        else:
+            assert prefix_length == 0
+            attn = ops.attention(q, k, v, causal=True)
+        attn = attn.transpose(1, 2).reshape(b, q_len, self.num_heads * self.head_dim)
+        return self.dense(attn)
+
+
+class FalconMLP(nn.Module):
+    def __init__(self, config: FalconConfig):
+        super().__init__()
+        self.dense_h_to_4h = nn.Linear(config.hidden_size, config.intermediate_size, bias=config.bias)
+        self.dense_4h_to_h = nn.Linear(config.intermediate_size, config.hidden_size, bias=config.bias)
+
+    def forward(self, x):
+        return self.dense_4h_to_h(ops.gelu(self.dense_h_to_4h(x)))
+
+
+class FalconBlock(nn.Module):
+    def __init__(self, config: FalconConfig, layer_idx: int = 0):
+        super().__init__()
+        self.config = config
+        self.layer_idx = layer_idx
+        self.self_attention = FalconAttention(config)
+        self.mlp = FalconMLP(config)
+        eps = config.layer_norm_eps
+        if config.new_decoder_architecture:
+            self.ln_attn = nn.LayerNorm(config.hidden_size, eps=eps)
+            self.ln_mlp = nn.LayerNorm(config.hidden_size, eps=eps)
+        else:
+            self.input_layernorm = nn.LayerNorm(config.hidden_size, eps=eps)
+            if not config.parallel_attn:
+                self.post_attention_layernorm = nn.LayerNorm(config.hidden_size, eps=eps)
+
+    def forward(self, hidden_states, kv_cache=None, prefix_length: int = 0):
+        residual = hidden_states
+        if self.config.new_decoder_architecture:
+            attn_in = self.ln_attn(hidden_states)
+            mlp_in = self.ln_mlp(hidden_states)
+            attn = self.self_attention(attn_in, kv_cache=kv_cache, prefix_length=prefix_length)
+            return residual + attn + self.mlp(mlp_in)
+        attn_in = self.input_layernorm(hidden_states)
+        attn = self.self_attention(attn_in, kv_cache=kv_cache, prefix_length=prefix_length)
+        if self.config.parallel_attn:
+            return residual + attn + self.mlp(attn_in)
+        hidden_states = residual + attn
+        mlp_in = self.post_attention_layernorm(hidden_states)
+        return hidden_states + self.mlp(mlp_in)
+
+    def kv_cache_shape(self, batch_size: int, max_length: int) -> Tuple[Tuple[int, ...], Tuple[int, ...]]:
+        shape = (batch_size, self.config.n_kv_heads, max_length, self.config.head_dim)
+        return shape, shape

@@ -1,0 +1,153 @@
+"""Native Llama decoder block.
+
+Our own implementation (NOT wrapping transformers — contrast with reference
+`models/llama/block.py:226` which wraps HF's LlamaDecoderLayer): RMSNorm ->
+GQA attention with RoPE and a preallocated KV cache -> RMSNorm -> SwiGLU MLP.
+All hot math goes through `petals_amd.ops` (HIP kernels on MI355X, torch
+reference on CPU).
+
+KV cache layout: K and V both [batch, n_kv_heads, max_length, head_dim]
+(natural layout; the reference's transposed bloom-layout K is a hivemind wire
+convention we do not carry over — conversion happens at the session boundary
+if ever needed).
+
+Block forward contract (used by the server backend and tests):
+    forward(hidden_states, kv_cache=None, prefix_length=0)
+      * kv_cache=None: stateless causal forward over positions [0, seq_len)
+      * kv_cache=(k_cache, v_cache): write new K/V at
+        [prefix_length : prefix_length + q_len], attend over the full prefix.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Optional, Tuple
+
+import torch
+from torch import nn
+
+from petals_amd import ops
+from petals_amd.models.llama.config import LlamaConfig
+
+
+class LlamaAttention(nn.Module):
+    def __init__(self, config: LlamaConfig):
+        super().__init__()
+        self.config = config
+        self.hidden_size = config.hidden_size
+        self.num_heads = config.num_attention_heads
+        self.num_kv_heads = config.n_kv_heads
+        self.head_dim = config.head_dim
+        bias = config.attention_bias
+        self.q_proj = nn.Linear(self.hidden_size, self.num_heads * self.head_dim, bias=bias)
+        self.k_proj = nn.Linear(self.hidden_size, self.num_kv_heads * self.head_dim, bias=bias)
+        self.v_proj = nn.Linear(self.hidden_size, self.num_kv_heads * self.head_dim, bias=bias)
+        self.o_proj = nn.Linear(self.num_heads * self.head_dim, self.hidden_size, bias=bias)
+
+        cos, sin = ops.build_rope_cache(
+            self.head_dim,
+            config.max_position_embeddings,
+            theta=config.rope_theta,
+            rope_scaling=config.rope_scaling,
+        )
+        self.register_buffer("rope_cos", cos, persistent=False)
+        self.register_buffer("rope_sin", sin, persistent=False)
+
+    def _ensure_rope(self, needed_len: int, device, ref: torch.Tensor):
+        if self.rope_cos.shape[0] < needed_len:
+            cos, sin = ops.build_rope_cache(
+                self.head_dim,
+                max(needed_len, 2 * self.rope_cos.shape[0]),
+                theta=self.config.rope_theta,
+                rope_scaling=self.config.rope_scaling,
+            )
+            self.rope_cos = cos.to(device)
+            self.rope_sin = sin.to(device)
+        elif self.rope_cos.device != device:
+            self.rope_cos = self.rope_cos.to(device)
+            self.rope_sin = self.rope_sin.to(device)
+
+    def forward(
+        self,
+        hidden_states: torch.Tensor,
+        kv_cache: Optional[Tuple[torch.Tensor, torch.Tensor]] = None,
+        prefix_length: int = 0,
+    ) -> torch.Tensor:
+        b, q_len, _ = hidden_states.shape
+        q = self.q_proj(hidden_states).view(b, q_len, self.num_heads, self.head_dim).transpose(1, 2)
+        k = self.k_proj(hidden_states).view(b, q_len, self.num_kv_heads, self.head_dim).transpose(1, 2)
+        v = self.v_proj(hidden_states).view(b, q_len, self.num_kv_heads, self.head_dim).transpose(1, 2)
+
+        end = prefix_length + q_len
+        self._ensure_rope(end, hidden_states.device, hidden_states)
+        position_ids = torch.arange(prefix_length, end, device=hidden_states.device)
+        q, k = ops.apply_rope(q, k, self.rope_cos, self.rope_sin, position_ids)
+
+        if kv_cache is not None:
+            k_cache, v_cache = kv_cache
+            k_cache[:b, :, prefix_length:end].copy_(k)
+            v_cache[:b, :, prefix_length:end].copy_(v)
+            attn = ops.attention_decode(q, k_cache[:b], v_cache[:b], end)
+        else:
+            assert prefix_length == 0, "stateless forward starts at position 0"
+            attn = ops.attention(q, k, v, causal=True)
+
+        attn = attn.transpose(1, 2).reshape(b, q_len, self.num_heads * self.head_dim)
+        return self.o_proj(attn)
+
+
+class LlamaMLP(nn.Module):
+    def __init__(self, config: LlamaConfig):
+        super().__init__()
+        bias = config.mlp_bias
+        self.gate_proj = nn.Linear(config.hidden_size, config.intermediate_size, bias=bias)
+        self.up_proj = nn.Linear(config.hidden_size, config.intermediate_size, bias=bias)
+        self.down_proj = nn.Linear(config.intermediate_size, config.hidden_size, bias=bias)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return self.down_proj(ops.swiglu(self.gate_proj(x), self.up_proj(x)))
+
+
+class RMSNorm(nn.Module):
+    def __init__(self, hidden_size: int, eps: float = 1e-5):
+        super().__init__()
+        self.weight = nn.Parameter(torch.ones(hidden_size))
+        self.variance_epsilon = eps
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return ops.rms_norm(x, self.weight, self.variance_epsilon)
+
+
+class LlamaBlock(nn.Module):
+    """One decoder layer; the unit served by a swarm server."""
+
+    def __init__(self, config: LlamaConfig, layer_idx: int = 0):
+        super().__init__()
+        self.config = config
+        self.layer_idx = layer_idx
+        self.self_attn = LlamaAttention(config)
+        self.mlp = LlamaMLP(config)
+        self.input_layernorm = RMSNorm(config.hidden_size, eps=config.layer_norm_eps)
+        self.post_attention_layernorm = RMSNorm(config.hidden_size, eps=config.layer_norm_eps)
+
+    def forward(
+        self,
+        hidden_states: torch.Tensor,
+        kv_cache: Optional[Tuple[torch.Tensor, torch.Tensor]] = None,
+        prefix_length: int = 0,
+    ) -> torch.Tensor:
+        residual = hidden_states
+        hidden_states = self.input_layernorm(hidden_states)
+        hidden_states = self.self_attn(hidden_states, kv_cache=kv_cache, prefix_length=prefix_length)
+        hidden_states = residual + hidden_states
+
+        residual = hidden_states
+        hidden_states = self.post_attention_layernorm(hidden_states)
+        hidden_states = self.mlp(hidden_states)
+        return residual + hidden_states
+
+    # --- cache geometry used by the server's MemoryCache ---
+
+    def kv_cache_shape(self, batch_size: int, max_length: int) -> Tuple[Tuple[int, ...], Tuple[int, ...]]:
+        shape = (batch_size, self.config.n_kv_heads, max_length, self.config.head_dim)
+        return shape, shape

@@ -1,0 +1,114 @@
+"""Op dispatch: HIP/CDNA4 kernels on MI355X, torch reference on CPU.
+
+Policy: on a GPU box the HIP extension is REQUIRED — ops raise if a CUDA
+tensor arrives and `petals_amd._hip_ops` is missing (no silent eager
+fallback; set PETALS_AMD_ALLOW_TORCH_FALLBACK=1 to override for debugging).
+"""
+
+from __future__ import annotations
+
+import os
+from typing import Optional, Tuple
+
+import torch
+
+from petals_amd.ops import reference
+
+_hip_ops = None
+_hip_import_error: Optional[BaseException] = None
+
+
+def _load_hip_ops():
+    global _hip_ops, _hip_import_error
+    if _hip_ops is not None:
+        return _hip_ops
+    try:
+        from petals_amd.ops import _hip  # thin loader for the in-tree .so
+
+        _hip_ops = _hip.load()
+    except BaseException as e:  # noqa: BLE001
+        _hip_import_error = e
+        _hip_ops = None
+    return _hip_ops
+
+
+def hip_available() -> bool:
+    return _load_hip_ops() is not None
+
+
+def _require_hip(op_name: str):
+    ops = _load_hip_ops()
+    if ops is None:
+        if os.environ.get("PETALS_AMD_ALLOW_TORCH_FALLBACK") == "1":
+            return None
+        raise RuntimeError(
+            f"petals_amd HIP extension is required for {op_name} on GPU but could not be "
+            f"loaded (build it with `python -m petals_amd.ops.build` or __graft_entry__.build()). "
+            f"Import error: {_hip_import_error!r}"
+        )
+    return ops
+
+
+def rms_norm(x: torch.Tensor, weight: torch.Tensor, eps: float = 1e-5) -> torch.Tensor:
+    if x.is_cuda:
+        ops = _require_hip("rms_norm")
+        if ops is not None:
+            return ops.rms_norm(x.contiguous(), weight.contiguous(), eps)
+    return reference.rms_norm(x, weight, eps)
+
+
+def apply_rope(q, k, cos, sin, position_ids):
+    if q.is_cuda:
+        ops = _require_hip("apply_rope")
+        if ops is not None:
+            if position_ids.dim() == 1:
+                position_ids = position_ids.unsqueeze(0).expand(q.shape[0], -1)
+            return ops.apply_rope(q, k, cos, sin, position_ids.contiguous())
+    return reference.apply_rope(q, k, cos, sin, position_ids)
+
+
+def attention(q, k, v, *, causal: bool, kv_offset: int = 0, attn_bias=None, scale=None):
+    if q.is_cuda:
+        ops = _require_hip("attention")
+        if ops is not None and attn_bias is None and q.dtype in (torch.bfloat16, torch.float16):
+            return ops.attention_prefill(
+                q.contiguous(), k.contiguous(), v.contiguous(), bool(causal), int(kv_offset),
+                float(scale) if scale is not None else -1.0,
+            )
+    return reference.attention(q, k, v, causal=causal, kv_offset=kv_offset, attn_bias=attn_bias, scale=scale)
+
+
+def attention_decode(q, k_cache, v_cache, kv_len: int, *, attn_bias=None, scale=None):
+    """Decode attention over a preallocated cache.
+
+    q: [b, n_heads, q_len, hd]; k_cache/v_cache: [b, n_kv, max_len, hd] with
+    valid prefix of length kv_len (incl. the current step already written)."""
+    if q.is_cuda:
+        ops = _require_hip("attention_decode")
+        if ops is not None and attn_bias is None and q.dtype in (torch.bfloat16, torch.float16):
+            return ops.attention_decode(
+                q.contiguous(), k_cache, v_cache, int(kv_len),
+                float(scale) if scale is not None else -1.0,
+            )
+    k = k_cache[:, :, :kv_len]
+    v = v_cache[:, :, :kv_len]
+    q_len = q.shape[2]
+    return reference.attention(
+        q, k, v, causal=q_len > 1, kv_offset=kv_len - q_len, attn_bias=attn_bias, scale=scale
+    )
+
+
+def swiglu(gate, up):
+    if gate.is_cuda:
+        ops = _require_hip("swiglu")
+        if ops is not None:
+            return ops.swiglu(gate.contiguous(), up.contiguous())
+    return reference.swiglu(gate, up)
+
+
+# re-exports used by blocks
+build_rope_cache = reference.build_rope_cache
+build_alibi_bias = reference.build_alibi_bias
+build_alibi_slopes = reference.build_alibi_slopes
+gelu = reference.gelu
+repeat_kv = reference.repeat_kv
