@@ -217,8 +217,9 @@ class Connection:
             await self._shutdown(exc_text or "connection closed")
 
     async def _run_handler(self, handler: RpcHandler, first: RpcMessage, stream: RpcStream):
+        # NB: the initial request is passed to the handler as an argument and is
+        # NOT fed into the stream inbox — the iterator yields only later messages
         try:
-            stream._feed(first)
             if first.kind == "end":
                 stream._feed_eof()
             await handler(first, stream)

@@ -111,7 +111,7 @@ class TransformerBackend:
         *before* this block on the forward pass; its gradient is the slice of
         grad_inputs over those positions (parity: block_functions.py:84-141)."""
         with torch.enable_grad():
-            inputs = inputs.detach().requires_grad_(True)
+            inputs = inputs.detach().clone().requires_grad_(True)
             if prompt is not None and not is_dummy(prompt):
                 pre = prompt.shape[1]
                 hidden = inputs.clone()

@@ -1,0 +1,378 @@
+"""The server's RPC surface — 7 RPCs matching the reference's wire API
+(`server/handler.py:132-592`): rpc_inference (bidi stream), rpc_forward,
+rpc_backward, rpc_forward_stream, rpc_backward_stream, rpc_push, rpc_info.
+
+Single-process MI355X design: handlers are asyncio coroutines on the server
+loop; all GPU work is submitted to the PriorityRuntime thread (inference
+priority beats training). Server-to-server activation push (`rpc_push`)
+delivers a step's inputs straight into the *next* server's open inference
+session, skipping the client round-trip.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import contextlib
+import logging
+import time
+from typing import Any, Dict, List, Optional, Sequence, Tuple
+
+import torch
+
+from petals_amd.data_structures import CHAIN_DELIMITER, InferenceMetadata, ModuleUID
+from petals_amd.p2p.streaming import receive_tensors_streamed, send_tensors_streamed
+from petals_amd.p2p.transport import P2PNode, RpcError, RpcMessage, RpcStream
+from petals_amd.server.backend import TransformerBackend
+from petals_amd.server.memory_cache import MemoryCache, TensorDescriptor
+from petals_amd.server.scheduler import PriorityRuntime, TaskPrioritizer
+from petals_amd.utils.misc import DUMMY, is_dummy
+
+logger = logging.getLogger(__name__)
+
+
+class _Session:
+    """Server-side state of one rpc_inference session."""
+
+    def __init__(self, session_id: str, uids: List[ModuleUID], max_length: int):
+        self.session_id = session_id
+        self.uids = uids
+        self.max_length = max_length
+        self.prefix_length = 0
+        self.pushed_inputs: asyncio.Queue = asyncio.Queue()
+
+
+class TransformerConnectionHandler:
+    def __init__(
+        self,
+        *,
+        backends: Dict[ModuleUID, TransformerBackend],
+        memory_cache: MemoryCache,
+        runtime: PriorityRuntime,
+        prioritizer: Optional[TaskPrioritizer] = None,
+        inference_max_length: int = 8192,
+        request_timeout: float = 180.0,
+        session_timeout: float = 30 * 60,
+        step_timeout: float = 5 * 60,
+        server_info_extra=None,
+        p2p: Optional[P2PNode] = None,
+        adapters: Sequence[str] = (),
+    ):
+        self.backends = backends
+        self.memory_cache = memory_cache
+        self.runtime = runtime
+        self.prioritizer = prioritizer or TaskPrioritizer()
+        self.inference_max_length = inference_max_length
+        self.request_timeout = request_timeout
+        self.session_timeout = session_timeout
+        self.step_timeout = step_timeout
+        self.server_info_extra = server_info_extra or (lambda: {})
+        self.adapters = tuple(adapters)
+        self.p2p = p2p
+        self._sessions: Dict[str, _Session] = {}
+
+    def register(self, p2p: P2PNode) -> None:
+        self.p2p = p2p
+        for name, fn in (
+            ("petals.rpc_info", self.rpc_info),
+            ("petals.rpc_forward", self.rpc_forward),
+            ("petals.rpc_forward_stream", self.rpc_forward_stream),
+            ("petals.rpc_backward", self.rpc_backward),
+            ("petals.rpc_backward_stream", self.rpc_backward_stream),
+            ("petals.rpc_inference", self.rpc_inference),
+            ("petals.rpc_push", self.rpc_push),
+        ):
+            p2p.add_handler(name, fn)
+
+    # ------------------------------------------------------------- helpers
+
+    def _parse_uids(self, meta: Dict[str, Any]) -> List[ModuleUID]:
+        uids_field = meta.get("uids")
+        if isinstance(uids_field, str):
+            uids = uids_field.split(CHAIN_DELIMITER)
+        else:
+            uids = list(uids_field or ())
+        if not uids:
+            raise RpcError("request must name at least one module uid")
+        for uid in uids:
+            if uid not in self.backends:
+                raise RpcError(f"uid {uid!r} is not served here")
+        return uids
+
+    def _split_prompts(self, prompts: Optional[torch.Tensor], n: int, dtype, device) -> List[Optional[torch.Tensor]]:
+        if prompts is None or is_dummy(prompts):
+            return [None] * n
+        assert prompts.ndim == 4 and prompts.shape[0] == n, "prompts must be [num_blocks, batch, pre_seq, hidden]"
+        return [p.to(device=device, dtype=dtype) for p in prompts]
+
+    # ----------------------------------------------------------- rpc_info
+
+    async def rpc_info(self, request: RpcMessage, stream: RpcStream) -> None:
+        first = next(iter(self.backends.values()), None)
+        cache_bytes_left = self.memory_cache.bytes_left
+        bytes_per_token = first.cache_bytes_per_token() if first is not None else 1
+        info = {
+            "version": "petals_amd-0.1",
+            "dht_client_mode": False,
+            "cache_tokens_left": int(cache_bytes_left // max(1, bytes_per_token)),
+            "adapters": list(self.adapters),
+            **self.server_info_extra(),
+        }
+        await stream.close(RpcMessage(meta=info))
+
+    # -------------------------------------------------------- rpc_forward
+
+    def _forward_chain(
+        self, uids: List[ModuleUID], hidden_states: torch.Tensor, prompts: Optional[torch.Tensor]
+    ) -> torch.Tensor:
+        """Runs IN the runtime thread: chain of stateless block forwards."""
+        backend0 = self.backends[uids[0]]
+        device, dtype = backend0.device, backend0.dtype
+        hidden_states = hidden_states.to(device=device, dtype=dtype)
+        prompt_list = self._split_prompts(prompts, len(uids), dtype, device)
+        for uid, prompt in zip(uids, prompt_list):
+            if prompt is not None:
+                hidden_states = hidden_states.clone()
+                hidden_states[:, : prompt.shape[1]] += prompt
+            hidden_states = self.backends[uid].forward(hidden_states)
+        return hidden_states.cpu()
+
+    async def _handle_forward(self, meta: Dict[str, Any], tensors: List[torch.Tensor]) -> List[torch.Tensor]:
+        uids = self._parse_uids(meta)
+        hidden_states = tensors[0]
+        prompts = tensors[1] if len(tensors) > 1 else None
+        assert hidden_states.ndim == 3
+        priority = self.prioritizer.prioritize(hidden_states, points=meta.get("points", 0), type="forward")
+        out = await self.runtime.submit(priority, self._forward_chain, uids, hidden_states, prompts)
+        return [out]
+
+    async def rpc_forward(self, request: RpcMessage, stream: RpcStream) -> None:
+        outs = await asyncio.wait_for(self._handle_forward(request.meta, request.tensors), self.request_timeout)
+        await stream.close(RpcMessage(tensors=outs))
+
+    async def rpc_forward_stream(self, request: RpcMessage, stream: RpcStream) -> None:
+        meta, tensors = await receive_tensors_streamed(stream, timeout=self.request_timeout)
+        meta = {**request.meta, **meta}
+        outs = await asyncio.wait_for(self._handle_forward(meta, tensors), self.request_timeout)
+        await send_tensors_streamed(stream, outs, close=True)
+
+    # ------------------------------------------------------- rpc_backward
+
+    def _backward_chain(
+        self, uids: List[ModuleUID], inputs: torch.Tensor, grad_outputs: torch.Tensor, prompts: Optional[torch.Tensor]
+    ) -> Tuple[torch.Tensor, torch.Tensor]:
+        """Runs IN the runtime thread. Re-runs forward to recover intermediate
+        activations, then backward in reverse (parity: block_functions.py:84-141)."""
+        backend0 = self.backends[uids[0]]
+        device, dtype = backend0.device, backend0.dtype
+        inputs = inputs.to(device=device, dtype=dtype)
+        grad_outputs = grad_outputs.to(device=device, dtype=dtype)
+        prompt_list = self._split_prompts(prompts, len(uids), dtype, device)
+
+        # re-run forward (plain no_grad, NOT inference_mode: these activations
+        # feed autograd below) to recover intermediate inputs
+        inter_inputs: List[torch.Tensor] = []
+        hidden = inputs
+        with torch.no_grad():
+            for uid, prompt in zip(uids[:-1], prompt_list[:-1]):
+                if prompt is not None:
+                    hidden = hidden.clone()
+                    hidden[:, : prompt.shape[1]] += prompt
+                inter_inputs.append(hidden)
+                hidden = self.backends[uid].block(hidden)
+            if prompt_list[-1] is not None:
+                hidden = hidden.clone()
+                hidden[:, : prompt_list[-1].shape[1]] += prompt_list[-1]
+            inter_inputs.append(hidden)
+
+        grad_prompts: List[Optional[torch.Tensor]] = []
+        grad = grad_outputs
+        for uid, hidden, prompt in zip(reversed(uids), reversed(inter_inputs), reversed(prompt_list)):
+            # prompt was already added into `hidden`; grad wrt prompt is the
+            # slice of grad wrt hidden over the prompt positions
+            grad, _ = self.backends[uid].backward(hidden, grad, prompt=None)
+            if prompt is not None:
+                grad_prompts.append(grad[:, : prompt.shape[1]].clone())
+            else:
+                grad_prompts.append(None)
+        grad_prompts.reverse()
+        if any(gp is not None for gp in grad_prompts):
+            ref = next(gp for gp in grad_prompts if gp is not None)
+            grad_prompts_t = torch.stack([gp if gp is not None else torch.zeros_like(ref) for gp in grad_prompts])
+        else:
+            grad_prompts_t = DUMMY
+        return grad.cpu(), grad_prompts_t.cpu() if not is_dummy(grad_prompts_t) else DUMMY
+
+    async def _handle_backward(self, meta: Dict[str, Any], tensors: List[torch.Tensor]) -> List[torch.Tensor]:
+        uids = self._parse_uids(meta)
+        inputs, grad_outputs = tensors[0], tensors[1]
+        prompts = tensors[2] if len(tensors) > 2 else None
+        priority = self.prioritizer.prioritize(inputs, points=meta.get("points", 0), type="backward")
+        grad_inputs, grad_prompts = await self.runtime.submit(
+            priority, self._backward_chain, uids, inputs, grad_outputs, prompts
+        )
+        return [grad_inputs, grad_prompts]
+
+    async def rpc_backward(self, request: RpcMessage, stream: RpcStream) -> None:
+        outs = await asyncio.wait_for(self._handle_backward(request.meta, request.tensors), self.request_timeout)
+        await stream.close(RpcMessage(tensors=outs))
+
+    async def rpc_backward_stream(self, request: RpcMessage, stream: RpcStream) -> None:
+        meta, tensors = await receive_tensors_streamed(stream, timeout=self.request_timeout)
+        meta = {**request.meta, **meta}
+        outs = await asyncio.wait_for(self._handle_backward(meta, tensors), self.request_timeout)
+        await send_tensors_streamed(stream, outs, close=True)
+
+    # ------------------------------------------------------ rpc_inference
+
+    def _inference_step_chain(
+        self,
+        uids: List[ModuleUID],
+        hidden_states: torch.Tensor,
+        hypo_ids: torch.Tensor,
+        prompts: Optional[torch.Tensor],
+        handles: List[Tuple[int, int]],
+        prefix_length: int,
+        active_adapter: Optional[str],
+    ) -> torch.Tensor:
+        """Runs IN the runtime thread: one inference step through the whole span
+        (the single-process analog of reference _MergedInferenceStep)."""
+        backend0 = self.backends[uids[0]]
+        device, dtype = backend0.device, backend0.dtype
+        hidden_states = hidden_states.to(device=device, dtype=dtype)
+        if hypo_ids is not None and not is_dummy(hypo_ids):
+            hypo_ids = hypo_ids.to(device)
+        prompt_list = self._split_prompts(prompts, len(uids), dtype, device)
+        for uid, prompt, handle_pair in zip(uids, prompt_list, handles):
+            if prompt is not None:
+                hidden_states = hidden_states.clone()
+                hidden_states[:, : prompt.shape[1]] += prompt
+            info = InferenceMetadata(uid, prefix_length, tuple(handle_pair), active_adapter)
+            (hidden_states,) = self.backends[uid].inference_step(hidden_states, hypo_ids, info)
+        return hidden_states.cpu()
+
+    async def rpc_inference(self, request: RpcMessage, stream: RpcStream) -> None:
+        meta = request.meta
+        uids = self._parse_uids(meta)
+        max_length = int(meta.get("max_length", 0))
+        if not 0 < max_length <= self.inference_max_length:
+            raise RpcError(f"max_length must be in (0, {self.inference_max_length}], got {max_length}")
+        batch_size = int(meta.get("batch_size", 1))
+        active_adapter = meta.get("active_adapter") or None
+        session_id = meta.get("session_id") or f"anon-{id(stream)}"
+        alloc_timeout = float(meta.get("alloc_timeout", self.memory_cache.alloc_timeout))
+
+        descriptors: List[TensorDescriptor] = []
+        for uid in uids:
+            descriptors.extend(self.backends[uid].get_inference_cache_descriptors(batch_size, max_length))
+
+        session = _Session(session_id, uids, max_length)
+        self._sessions[session_id] = session
+        t_start = time.monotonic()
+        try:
+            async with self.memory_cache.allocate_cache(*descriptors, timeout=alloc_timeout) as flat_handles:
+                handles = [tuple(flat_handles[2 * i : 2 * i + 2]) for i in range(len(uids))]
+                # confirm session is open (client waits for this before step 1)
+                await stream.send(RpcMessage(meta={"session_open": True, "session_id": session_id}))
+                async for step_meta, tensors in self._iterate_inference_steps(stream, session):
+                    if time.monotonic() - t_start > self.session_timeout:
+                        raise RpcError("session timed out")
+                    if "start_from_position" in step_meta:
+                        pos = int(step_meta["start_from_position"])
+                        if pos > session.prefix_length:
+                            raise RpcError("start_from_position is ahead of the cache")
+                        session.prefix_length = pos
+                    hidden_states = tensors[0]
+                    prompts = tensors[1] if len(tensors) > 1 else None
+                    hypo_ids = tensors[2] if len(tensors) > 2 else None
+                    length_increment = hidden_states.shape[1] if hidden_states.numel() > 0 else 0
+                    if session.prefix_length + length_increment > max_length:
+                        raise RpcError(
+                            f"max_length exceeded: prefix {session.prefix_length} + {length_increment} > {max_length}"
+                        )
+                    if length_increment > 0:
+                        priority = self.prioritizer.prioritize(hidden_states, type="inference")
+                        output = await self.runtime.submit(
+                            priority,
+                            self._inference_step_chain,
+                            uids,
+                            hidden_states,
+                            hypo_ids,
+                            prompts,
+                            handles,
+                            session.prefix_length,
+                            active_adapter,
+                        )
+                    else:
+                        output = hidden_states
+                    session.prefix_length += length_increment
+
+                    has_prompts = prompts is not None and not is_dummy(prompts)
+                    next_servers = step_meta.get("next_servers")
+                    pushed = False
+                    if next_servers and not has_prompts and length_increment > 0:
+                        pushed = await self._push_outputs(output, step_meta, next_servers)
+                    if not pushed:
+                        # the last server of a push chain (or any server when push
+                        # is off/failed) returns outputs on its client stream
+                        await stream.send(
+                            RpcMessage(meta={"step_id": step_meta.get("step_id")}, tensors=[output])
+                        )
+        finally:
+            self._sessions.pop(session_id, None)
+
+    async def _iterate_inference_steps(self, stream: RpcStream, session: _Session):
+        """Multiplex client-stream steps with server-pushed steps (parity:
+        handler.py:247-308)."""
+        while True:
+            client_task = asyncio.ensure_future(stream.receive(timeout=self.step_timeout))
+            push_task = asyncio.ensure_future(session.pushed_inputs.get())
+            done, pending = await asyncio.wait(
+                {client_task, push_task}, return_when=asyncio.FIRST_COMPLETED, timeout=self.step_timeout
+            )
+            for task in pending:
+                task.cancel()
+                with contextlib.suppress(asyncio.CancelledError):
+                    await task
+            if not done:
+                raise RpcError("inference step timed out")
+            task = done.pop()
+            try:
+                item = task.result()
+            except RpcError as e:
+                if "closed" in str(e):
+                    return  # client closed the stream: session over
+                raise
+            if isinstance(item, RpcMessage):
+                if item.kind == "end" and not item.tensors and not item.meta.get("step_id"):
+                    return  # graceful close
+                yield item.meta, item.tensors
+            else:  # pushed step: (meta, tensors)
+                yield item
+
+    async def rpc_push(self, request: RpcMessage, stream: RpcStream) -> None:
+        session_id = request.meta.get("session_id")
+        session = self._sessions.get(session_id)
+        if session is None:
+            raise RpcError(f"no active inference session {session_id!r}")
+        session.pushed_inputs.put_nowait((dict(request.meta), list(request.tensors)))
+        await stream.close(RpcMessage(meta={"ok": True}))
+
+    async def _push_outputs(self, output: torch.Tensor, step_meta: Dict[str, Any], next_servers) -> bool:
+        """Push this step's output into the next server's session. next_servers:
+        [[host, port, session_id, start_block, end_block], ...]; we contact the
+        first entry."""
+        try:
+            host, port, next_session_id = next_servers[0][0], next_servers[0][1], next_servers[0][2]
+            meta = {
+                "session_id": next_session_id,
+                "step_id": step_meta.get("step_id"),
+                "next_servers": next_servers[1:],
+            }
+            await asyncio.wait_for(
+                self.p2p.call_unary((host, port), "petals.rpc_push", RpcMessage(meta=meta, tensors=[output])),
+                timeout=10.0,
+            )
+            return True
+        except Exception as e:  # noqa: BLE001
+            logger.warning("rpc_push to next server failed: %r", e)
+            return False

@@ -1,0 +1,357 @@
+"""Server process: owns the DHT handle, decides which blocks to serve, loads
+them, serves the 7 RPCs, announces itself, and rebalances when the swarm is
+uneven.
+
+Parity with reference ``server/server.py`` (Server :46-429, ModuleContainer
+:431-672, ModuleAnnouncerThread :674-768) in a single-process MI355X design:
+one asyncio loop thread runs DHT + RPC handlers; one PriorityRuntime thread
+owns the GPU; no handler subprocesses (the reference needed them because of
+Python-level serialization costs per libp2p stream; asyncio + zero-copy
+msgpack framing covers swarm-scale traffic here).
+"""
+
+from __future__ import annotations
+
+import asyncio
+import logging
+import random
+import threading
+import time
+from typing import Dict, List, Optional, Sequence, Tuple
+
+import torch
+
+from petals_amd.data_structures import ServerInfo, ServerState, get_dht_time, make_uid
+from petals_amd.dht.node import DHTNode
+from petals_amd.models.config_base import ModelConfig, load_model_config
+from petals_amd.p2p.transport import P2PNode
+from petals_amd.server import block_selection
+from petals_amd.server.backend import TransformerBackend
+from petals_amd.server.from_pretrained import load_pretrained_block
+from petals_amd.server.handler import TransformerConnectionHandler
+from petals_amd.server.memory_cache import MemoryCache
+from petals_amd.server.scheduler import PriorityRuntime
+from petals_amd.server.throughput import get_server_throughput
+from petals_amd.utils.dht import declare_active_modules, declare_model, get_remote_module_infos
+from petals_amd.utils.misc import get_size_in_bytes
+
+logger = logging.getLogger(__name__)
+
+DTYPE_MAP = {"bfloat16": torch.bfloat16, "float16": torch.float16, "float32": torch.float32, "auto": None}
+
+
+class Server:
+    """One swarm server hosting a contiguous span of transformer blocks."""
+
+    def __init__(
+        self,
+        model_name_or_dir: str,
+        *,
+        initial_peers: Sequence[Tuple[str, int]] = (),
+        host: str = "127.0.0.1",
+        port: int = 0,
+        device: Optional[str] = None,
+        torch_dtype: str = "auto",
+        num_blocks: Optional[int] = None,
+        block_indices: Optional[str] = None,
+        dht_prefix: Optional[str] = None,
+        attn_cache_tokens: int = 16384,
+        max_batch_size: int = 8,
+        inference_max_length: Optional[int] = None,
+        throughput: str | float = "auto",
+        update_period: float = 60.0,
+        expiration: Optional[float] = None,
+        balance_quality: float = 0.75,
+        mean_balance_check_period: float = 120.0,
+        quant_type: str = "none",
+        public_name: Optional[str] = None,
+    ):
+        self.config = load_model_config(model_name_or_dir)
+        self.model_name_or_dir = model_name_or_dir
+        if dht_prefix:
+            self.config.dht_prefix = dht_prefix
+        self.initial_peers = [tuple(p) for p in initial_peers]
+        self.host, self.port = host, port
+
+        if device is None:
+            device = "cuda" if torch.cuda.is_available() else "cpu"
+        self.device = torch.device(device)
+        dtype = DTYPE_MAP[torch_dtype]
+        if dtype is None:
+            dtype = torch.bfloat16 if self.device.type == "cuda" else torch.float32
+        self.torch_dtype = dtype
+        self.quant_type = quant_type
+
+        self.block_indices: Optional[List[int]] = None
+        if block_indices is not None:
+            start, end = block_indices.split(":")
+            self.block_indices = list(range(int(start), int(end)))
+            num_blocks = len(self.block_indices)
+        self.num_blocks = num_blocks or self._choose_num_blocks(attn_cache_tokens)
+
+        self.attn_cache_tokens = attn_cache_tokens
+        self.max_batch_size = max_batch_size
+        self.inference_max_length = inference_max_length or attn_cache_tokens
+        self.update_period = update_period
+        self.expiration = expiration or max(3 * update_period, 60.0)
+        self.balance_quality = balance_quality
+        self.mean_balance_check_period = mean_balance_check_period
+        self.public_name = public_name
+        self._throughput_setting = throughput
+
+        self.module_uids = [make_uid(self.config.dht_prefix, i) for i in range(self.config.num_blocks)]
+
+        # runtime state
+        self.p2p: Optional[P2PNode] = None
+        self.dht_node: Optional[DHTNode] = None
+        self.runtime: Optional[PriorityRuntime] = None
+        self.memory_cache: Optional[MemoryCache] = None
+        self.backends: Dict[str, TransformerBackend] = {}
+        self.handler: Optional[TransformerConnectionHandler] = None
+        self.loop: Optional[asyncio.AbstractEventLoop] = None
+        self._thread: Optional[threading.Thread] = None
+        self._ready = threading.Event()
+        self._stop = threading.Event()
+        self._restart_requested = False
+        self.server_info: Optional[ServerInfo] = None
+        self.listen_addr: Optional[Tuple[str, int]] = None
+
+    # -------------------------------------------------------------- sizing
+
+    def _block_param_bytes(self) -> int:
+        cfg = self.config
+        dtype_bytes = get_size_in_bytes(self.torch_dtype)
+        if self.quant_type == "nf4":
+            dtype_bytes = 0.53125  # 4.25 bits/param (parity: block_utils.py:46)
+        h, inter = cfg.hidden_size, cfg.intermediate_size
+        kv = cfg.n_kv_heads * cfg.head_dim
+        attn = h * h + 2 * h * kv + h * h
+        n_experts = getattr(cfg, "num_local_experts", 1)
+        mlp = 3 * h * inter * n_experts
+        return int((attn + mlp) * dtype_bytes)
+
+    def _choose_num_blocks(self, attn_cache_tokens: int = 16384) -> int:
+        """Fit blocks to device memory: block bytes + attn-cache + autograd
+        reserve (parity: server.py:275-326)."""
+        if self.device.type == "cuda":
+            total = torch.cuda.get_device_properties(self.device).total_memory
+        else:
+            total = 16 << 30
+        block_bytes = self._block_param_bytes()
+        cache_bytes_per_block = (
+            2 * self.config.n_kv_heads * self.config.head_dim * attn_cache_tokens * get_size_in_bytes(self.torch_dtype)
+        )
+        autograd_reserve = 2 << 30
+        usable = total * 0.92 - autograd_reserve
+        n = max(1, int(usable // (block_bytes + cache_bytes_per_block)))
+        return min(n, self.config.num_blocks)
+
+    # ------------------------------------------------------------ lifecycle
+
+    def start(self):
+        self._thread = threading.Thread(target=self._run_thread, name="PetalsServer", daemon=True)
+        self._thread.start()
+        self._ready.wait(timeout=600)
+        if not self._ready.is_set() or self.listen_addr is None:
+            raise RuntimeError("server failed to start")
+        return self
+
+    def _run_thread(self):
+        loop = asyncio.new_event_loop()
+        asyncio.set_event_loop(loop)
+        self.loop = loop
+        try:
+            loop.run_until_complete(self._amain())
+        except Exception:  # noqa: BLE001
+            logger.exception("server main loop crashed")
+        finally:
+            self._ready.set()  # unblock start() even on failure
+            try:
+                loop.run_until_complete(self._ashutdown())
+            except Exception:  # noqa: BLE001
+                pass
+            loop.close()
+
+    async def _amain(self):
+        self.p2p = P2PNode()
+        await self.p2p.listen(host=self.host, port=self.port)
+        self.listen_addr = self.p2p.listen_addr
+        self.dht_node = await DHTNode.create(initial_peers=self.initial_peers, p2p=self.p2p)
+
+        self.runtime = PriorityRuntime(self.device).start()
+        while not self._stop.is_set():
+            await self._serve_once()
+            if not self._restart_requested:
+                break
+            self._restart_requested = False
+
+    async def _serve_once(self):
+        """Load a span of blocks and serve until shutdown or rebalance."""
+        # --- choose blocks
+        if self.block_indices is not None:
+            block_indices = self.block_indices
+        else:
+            infos, _ = await self._get_infos()
+            block_indices = block_selection.choose_best_blocks(self.num_blocks, infos)
+        logger.info("serving blocks %s..%s", block_indices[0], block_indices[-1])
+
+        # --- throughput estimate (cheap on CPU; measured once per device/model)
+        if self._throughput_setting == "auto":
+            tp = get_server_throughput(
+                self.config,
+                device=self.device,
+                dtype=self.torch_dtype,
+                num_blocks=len(block_indices),
+                quant_type=self.quant_type,
+            )
+            throughput = tp["throughput"]
+            inference_rps, forward_rps, network_rps = tp["inference_rps"], tp["forward_rps"], tp["network_rps"]
+        else:
+            throughput = float(self._throughput_setting)
+            inference_rps = forward_rps = network_rps = throughput
+
+        cache_bytes = int(
+            2
+            * self.config.n_kv_heads
+            * self.config.head_dim
+            * self.attn_cache_tokens
+            * get_size_in_bytes(self.torch_dtype)
+            * len(block_indices)
+        )
+        self.memory_cache = MemoryCache(cache_bytes, self.device)
+
+        self.server_info = ServerInfo(
+            state=ServerState.JOINING,
+            throughput=throughput,
+            start_block=block_indices[0],
+            end_block=block_indices[-1] + 1,
+            public_name=self.public_name,
+            version="0.1",
+            network_rps=network_rps,
+            forward_rps=forward_rps,
+            inference_rps=inference_rps,
+            torch_dtype=str(self.torch_dtype).replace("torch.", ""),
+            quant_type=self.quant_type,
+        )
+        await self._announce()
+
+        # --- load blocks (in a worker thread: file IO + H2D copies)
+        served_uids = [make_uid(self.config.dht_prefix, i) for i in block_indices]
+        self.backends = {}
+        for i, uid in zip(block_indices, served_uids):
+            block = await asyncio.get_event_loop().run_in_executor(
+                None,
+                lambda idx=i: load_pretrained_block(
+                    self.model_name_or_dir, self.config, idx, torch_dtype=self.torch_dtype, device=self.device
+                ),
+            )
+            self.backends[uid] = TransformerBackend(
+                uid, block, config=self.config, memory_cache=self.memory_cache, dtype=self.torch_dtype
+            )
+
+        self.handler = TransformerConnectionHandler(
+            backends=self.backends,
+            memory_cache=self.memory_cache,
+            runtime=self.runtime,
+            inference_max_length=self.inference_max_length,
+            p2p=self.p2p,
+        )
+        self.handler.register(self.p2p)
+
+        self.server_info.state = ServerState.ONLINE
+        await self._announce()
+        declare_model_info = {
+            "model_type": self.config.model_type,
+            "num_blocks": self.config.num_blocks,
+            "dht_prefix": self.config.dht_prefix,
+            "config": self.config.to_dict(),
+        }
+        await self.dht_node.store_many(
+            [("_petals_amd.models", self.config.dht_prefix, declare_model_info, get_dht_time() + self.expiration)]
+        )
+        self._ready.set()
+
+        # --- announce + health + rebalance loop
+        next_balance_check = time.monotonic() + random.random() * 2 * self.mean_balance_check_period
+        last_announce = time.monotonic()
+        try:
+            while not self._stop.is_set():
+                if time.monotonic() - last_announce >= self.update_period:
+                    await self._announce()
+                    last_announce = time.monotonic()
+                if not self.runtime.alive:
+                    raise RuntimeError("runtime thread died")
+                await asyncio.sleep(min(self.update_period, 0.5))
+                if self.block_indices is None and time.monotonic() > next_balance_check:
+                    next_balance_check = time.monotonic() + random.random() * 2 * self.mean_balance_check_period
+                    infos, _ = await self._get_infos()
+                    if block_selection.should_choose_other_blocks(self.p2p.peer_id, infos, self.balance_quality):
+                        logger.info("swarm is imbalanced: restarting with a new span")
+                        self._restart_requested = True
+                        break
+        finally:
+            self.server_info.state = ServerState.OFFLINE
+            await self._announce()
+            for backend in self.backends.values():
+                backend.shutdown()
+            self.backends = {}
+
+    async def _get_infos(self):
+        from petals_amd.data_structures import RemoteModuleInfo, ServerInfo as SI
+
+        found = await self.dht_node.get_many(self.module_uids)
+        infos, addrs = [], {}
+        for uid in self.module_uids:
+            servers = {}
+            for peer_id, (value, _exp) in (found.get(uid) or {}).items():
+                try:
+                    info = SI.from_dict(value["info"])
+                    addrs[peer_id] = tuple(value["addr"])
+                except (KeyError, TypeError):
+                    continue
+                servers[peer_id] = info
+            infos.append(RemoteModuleInfo(uid=uid, servers=servers) if servers else None)
+        return infos, addrs
+
+    async def _announce(self):
+        if self.server_info is None:
+            return
+        first = next(iter(self.backends.values()), None)
+        if first is not None and self.memory_cache is not None:
+            bytes_per_token = first.cache_bytes_per_token() * max(1, len(self.backends))
+            self.server_info.cache_tokens_left = int(self.memory_cache.bytes_left // max(1, bytes_per_token))
+        uids = [
+            make_uid(self.config.dht_prefix, i)
+            for i in range(self.server_info.start_block or 0, self.server_info.end_block or 0)
+        ]
+        if not uids:
+            return
+        value = {"info": self.server_info.to_dict(), "addr": list(self.listen_addr)}
+        entries = [(uid, self.p2p.peer_id, value, get_dht_time() + self.expiration) for uid in uids]
+        try:
+            await self.dht_node.store_many(entries)
+        except Exception as e:  # noqa: BLE001
+            logger.warning("announce failed: %r", e)
+
+    async def _ashutdown(self):
+        if self.runtime is not None:
+            self.runtime.shutdown()
+        if self.dht_node is not None:
+            await self.dht_node.shutdown()
+
+    @property
+    def peer_id(self) -> str:
+        return self.p2p.peer_id if self.p2p else ""
+
+    def shutdown(self, timeout: float = 15.0):
+        self._stop.set()
+        if self._thread is not None:
+            self._thread.join(timeout=timeout)
+
+    def is_healthy(self) -> bool:
+        return (
+            self._thread is not None
+            and self._thread.is_alive()
+            and self.runtime is not None
+            and self.runtime.alive
+        )

@@ -1,0 +1,155 @@
+"""End-to-end mini-swarm on CPU (BASELINE config #1 pattern): two fp32 servers
+hosting disjoint spans + a thin client; generate() and training fwd/bwd are
+exact-matched against the local HF model (golden-reference pattern, reference
+tests/test_full_model.py)."""
+
+import os
+import tempfile
+
+import pytest
+import torch
+
+HF_CFG = dict(
+    hidden_size=64,
+    num_hidden_layers=4,
+    num_attention_heads=4,
+    num_key_value_heads=2,
+    intermediate_size=128,
+    vocab_size=128,
+    max_position_embeddings=256,
+    tie_word_embeddings=False,
+)
+
+
+@pytest.fixture(scope="module")
+def hf_checkpoint(tmp_path_factory):
+    transformers = pytest.importorskip("transformers")
+    torch.manual_seed(0)
+    cfg = transformers.LlamaConfig(**HF_CFG)
+    model = transformers.LlamaForCausalLM(cfg).eval()
+    path = tmp_path_factory.mktemp("llama_ckpt")
+    model.save_pretrained(path, safe_serialization=True)
+    return str(path), model
+
+
+@pytest.fixture(scope="module")
+def swarm(hf_checkpoint):
+    from petals_amd.dht.node import DHT
+    from petals_amd.server.server import Server
+
+    path, _ = hf_checkpoint
+    bootstrap = DHT(host="127.0.0.1")
+    servers = []
+    # server1: blocks 0..2, server2: blocks 2..4, server3 overlaps 1..4
+    for spec in ("0:2", "2:4", "1:4"):
+        s = Server(
+            path,
+            initial_peers=[bootstrap.listen_addr],
+            host="127.0.0.1",
+            device="cpu",
+            torch_dtype="float32",
+            block_indices=spec,
+            dht_prefix="test-llama-e2e",
+            throughput=1.0,
+            update_period=2.0,
+        )
+        s.start()
+        servers.append(s)
+    yield bootstrap, servers, path
+    for s in servers:
+        s.shutdown()
+    bootstrap.shutdown()
+
+
+@pytest.fixture(scope="module")
+def client_model(swarm, hf_checkpoint):
+    from petals_amd.utils.auto_config import AutoDistributedModelForCausalLM
+
+    bootstrap, _, path = swarm
+    model = AutoDistributedModelForCausalLM.from_pretrained(
+        path,
+        initial_peers=[bootstrap.listen_addr],
+        dht_prefix="test-llama-e2e",
+        show_route=False,
+        max_retries=2,
+        min_backoff=0.2,
+        request_timeout=30.0,
+    )
+    yield model
+    model.transformer.h.sequence_manager.shutdown()
+
+
+def test_forward_exact_match(client_model, hf_checkpoint):
+    _, hf_model = hf_checkpoint
+    torch.manual_seed(1)
+    ids = torch.randint(0, 128, (2, 10))
+    with torch.no_grad():
+        ref = hf_model(ids).logits
+        out = client_model(input_ids=ids).logits
+    assert out.shape == ref.shape
+    assert torch.allclose(out, ref, atol=1e-4, rtol=1e-3), (out - ref).abs().max()
+
+
+def test_greedy_generate_matches_hf(client_model, hf_checkpoint):
+    _, hf_model = hf_checkpoint
+    torch.manual_seed(2)
+    ids = torch.randint(0, 128, (1, 5))
+    ref = hf_model.generate(ids, max_new_tokens=6, do_sample=False)
+    out = client_model.generate(ids, max_new_tokens=6, do_sample=False)
+    assert torch.equal(out, ref), (out, ref)
+
+
+def test_batched_generate(client_model):
+    torch.manual_seed(3)
+    ids = torch.randint(0, 128, (3, 4))
+    out = client_model.generate(ids, max_new_tokens=5, do_sample=False)
+    assert out.shape == (3, 9)
+
+
+def test_sampling_generate_deterministic(client_model):
+    ids = torch.randint(0, 128, (1, 4))
+    torch.manual_seed(0)
+    out1 = client_model.generate(ids, max_new_tokens=5, do_sample=True, top_k=10)
+    torch.manual_seed(0)
+    out2 = client_model.generate(ids, max_new_tokens=5, do_sample=True, top_k=10)
+    assert torch.equal(out1, out2)
+
+
+def test_inference_matches_forward(client_model):
+    """Token-by-token session logits == one-shot forward logits."""
+    torch.manual_seed(4)
+    ids = torch.randint(0, 128, (1, 8))
+    with torch.no_grad():
+        ref = client_model(input_ids=ids).logits
+    outs = []
+    with client_model.transformer.h.inference_session(max_length=16) as sess:
+        with client_model.transformer.h.use_session(sess):
+            with torch.no_grad():
+                outs.append(client_model(input_ids=ids[:, :3]).logits)
+                for t in range(3, 8):
+                    outs.append(client_model(input_ids=ids[:, t : t + 1]).logits)
+    step_logits = torch.cat(outs, dim=1)
+    assert torch.allclose(step_logits, ref, atol=1e-4, rtol=1e-3), (step_logits - ref).abs().max()
+
+
+def test_training_forward_backward(client_model, hf_checkpoint):
+    """Remote fwd+bwd grads match local HF grads (prompt-tuning style: grads
+    wrt inputs_embeds)."""
+    _, hf_model = hf_checkpoint
+    torch.manual_seed(5)
+    ids = torch.randint(0, 128, (2, 6))
+
+    embeds_ref = hf_model.get_input_embeddings()(ids).detach().requires_grad_(True)
+    ref_out = hf_model(inputs_embeds=embeds_ref).logits
+    ref_loss = ref_out.square().mean()
+    ref_loss.backward()
+
+    embeds = hf_model.get_input_embeddings()(ids).detach().requires_grad_(True)
+    out = client_model(inputs_embeds=embeds).logits
+    loss = out.square().mean()
+    loss.backward()
+
+    assert torch.allclose(out, ref_out, atol=1e-4, rtol=1e-3)
+    assert torch.allclose(embeds.grad, embeds_ref.grad, atol=1e-4, rtol=1e-3), (
+        (embeds.grad - embeds_ref.grad).abs().max()
+    )
