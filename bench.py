@@ -1,0 +1,216 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: single-batch autoregressive generate, Llama-2-70B,
+blocks pipelined across N MI355X GPUs over RCCL/xGMI (BASELINE.json metric).
+
+  python bench.py --gpus N --steps K --warmup W [--model llama-2-70b]
+
+For N>1 the driver launches this under torch.distributed.run with one rank per
+GPU; rank r hosts a contiguous span of the 80 blocks (the petals_amd span
+architecture), activations hop rank->rank via RCCL send/recv (the framework's
+intra-node hand-off path, petals_amd/parallel/pipeline.py), rank 0 holds
+embeddings + final norm + LM head (the thin petals client role). Weights are
+random-init bf16 (no network for checkpoints), data is a synthetic prompt.
+
+A "step" = one generated token. Weak scaling: each GPU keeps the whole model's
+per-token read spread over N, i.e. per-GPU work shrinks but single-batch
+decode is sequential across spans (petals semantics: more servers host bigger
+models; they do not parallelize one batch).
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+
+def log(msg):
+    if int(os.environ.get("RANK", "0")) == 0:
+        print(msg, flush=True)
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=64)
+    p.add_argument("--warmup", type=int, default=8)
+    p.add_argument("--model", default="llama-2-70b")
+    p.add_argument("--prompt-len", type=int, default=128)
+    p.add_argument("--batch", type=int, default=1)
+    p.add_argument("--device", default="cuda")
+    args = p.parse_args()
+
+    import torch.distributed as dist
+
+    from petals_amd.models import get_model_block
+    from petals_amd.models.config_base import load_model_config
+    from petals_amd.parallel.pipeline import PipelineStage, init_process_group_from_env, split_blocks
+    from petals_amd.server.from_pretrained import init_random_block_
+
+    rank, world = init_process_group_from_env()
+    assert world == args.gpus or args.gpus == 1, f"WORLD_SIZE={world} but --gpus={args.gpus}"
+    world = max(world, 1)
+
+    use_cuda = args.device == "cuda" and torch.cuda.is_available()
+    if use_cuda:
+        device = torch.device("cuda", int(os.environ.get("LOCAL_RANK", 0)))
+        torch.cuda.set_device(device)
+        dtype = torch.bfloat16
+    else:
+        device = torch.device("cpu")
+        dtype = torch.float32
+
+    config = load_model_config(args.model)
+    H = config.hidden_size
+    B = args.batch
+    max_len = args.prompt_len + args.warmup + args.steps + 8
+
+    spans = split_blocks(config.num_blocks, world)
+    my_span = spans[rank]
+    log(f"[bench] model={args.model} blocks={config.num_blocks} world={world} "
+        f"span[0]={list(spans[0])[:1]}..{list(spans[0])[-1:]} dtype={dtype}")
+
+    # --- build this rank's span (deterministic random init, bf16, fused path)
+    t0 = time.time()
+    blocks = []
+    for i in my_span:
+        blk = get_model_block(config, i)
+        init_random_block_(blk, config, i)
+        blk = blk.to(device=device, dtype=dtype).eval()
+        if use_cuda and hasattr(blk, "optimize_for_inference"):
+            blk.optimize_for_inference()
+        blocks.append(blk)
+    kv_caches = []
+    for blk in blocks:
+        ks, vs = blk.kv_cache_shape(B, max_len)
+        kv_caches.append((
+            torch.zeros(ks, device=device, dtype=dtype),
+            torch.zeros(vs, device=device, dtype=dtype),
+        ))
+    stage = PipelineStage(blocks, rank, world, device, H, dtype)
+    log(f"[bench] rank {rank}: {len(blocks)} blocks built in {time.time()-t0:.1f}s")
+
+    # --- rank 0 client-side parts
+    if rank == 0:
+        gen = torch.Generator().manual_seed(1234)
+        embed = (torch.randn(config.vocab_size, H, generator=gen) * 0.02).to(device=device, dtype=dtype)
+        norm_w = torch.ones(H, device=device, dtype=dtype)
+        head_t = embed.t().contiguous()  # tied head, [H, vocab] for gemv/matmul
+        prompt = torch.randint(0, config.vocab_size, (B, args.prompt_len), generator=gen).to(device)
+
+    from petals_amd import ops
+
+    hip = ops._load_hip_ops() if use_cuda else None
+    if use_cuda and hip is None:
+        raise RuntimeError(f"HIP extension required on GPU: {ops._hip_import_error!r}")
+
+    def head_logits(h_last: torch.Tensor) -> torch.Tensor:
+        """h_last [B, H] -> logits [B, vocab] (final norm + tied head)."""
+        if hip is not None:
+            xn = hip.rms_norm_f32out(h_last.to(dtype), norm_w, config.layer_norm_eps)
+            ws = torch.empty(0, device=device)
+            return hip.gemv_bf16(head_t, xn, ws, None, 0)
+        xn = ops.reference.rms_norm(h_last, norm_w, config.layer_norm_eps)
+        return xn @ head_t.float() if head_t.dtype != xn.dtype else xn @ head_t
+
+    def run_pipeline(h: torch.Tensor, prefix: int, seq: int) -> torch.Tensor:
+        """Push [B, seq, H] through all spans; returns last hidden on rank 0."""
+        if world == 1:
+            return stage.forward_span(h, kv_caches, prefix)
+        if rank == 0:
+            h = stage.forward_span(h, kv_caches, prefix)
+            stage.send(h)
+            return stage.recv(B, seq)  # final output comes back from last rank
+        else:
+            h = stage.recv(B, seq)
+            h = stage.forward_span(h, kv_caches, prefix)
+            stage.send(h)
+            return h  # unused on non-zero ranks
+
+    def one_token(token_ids, prefix: int):
+        """Generate the next token given current ids [B,1]."""
+        if rank == 0:
+            h = embed[token_ids.view(-1)].view(B, 1, H)
+        else:
+            h = torch.empty(B, 1, H, device=device, dtype=dtype)
+        out = run_pipeline(h, prefix, 1)
+        if rank == 0:
+            logits = head_logits(out[:, -1, :])
+            return logits.argmax(dim=-1, keepdim=True)
+        return token_ids
+
+    # --- prefill
+    prefix = 0
+    if rank == 0:
+        h = embed[prompt.view(-1)].view(B, args.prompt_len, H)
+    else:
+        h = torch.empty(B, args.prompt_len, H, device=device, dtype=dtype)
+    out = run_pipeline(h, 0, args.prompt_len)
+    prefix = args.prompt_len
+    if rank == 0:
+        cur = head_logits(out[:, -1, :]).argmax(dim=-1, keepdim=True)
+    else:
+        cur = torch.zeros(B, 1, dtype=torch.long, device=device)
+
+    # --- warmup decode
+    for _ in range(args.warmup):
+        cur = one_token(cur, prefix)
+        prefix += 1
+
+    # --- timed region
+    if world > 1:
+        dist.barrier()
+    if use_cuda:
+        torch.cuda.synchronize(device)
+    t_start = time.perf_counter()
+    for _ in range(args.steps):
+        cur = one_token(cur, prefix)
+        prefix += 1
+    if use_cuda:
+        torch.cuda.synchronize(device)
+    if world > 1:
+        dist.barrier()
+    elapsed = time.perf_counter() - t_start
+
+    # max over ranks
+    if world > 1:
+        e = torch.tensor([elapsed], device=device)
+        dist.all_reduce(e, op=dist.ReduceOp.MAX)
+        elapsed = float(e.item())
+
+    if rank == 0:
+        ms_per_step = elapsed / args.steps * 1000
+        tokens_per_s = args.steps * B / elapsed
+        baseline = 6.0  # reference headline: Llama-2-70B, 6 tok/s on the public swarm
+        result = {
+            "metric": "single-batch generate tokens/sec, Llama-2-70B across 1/2/4/8 MI355X servers",
+            "value": tokens_per_s,
+            "unit": "tokens/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": tokens_per_s / baseline if args.model == "llama-2-70b" else None,
+            "dtype": "bf16" if use_cuda else "fp32",
+            "data": "synthetic prompt, random-init weights (no network)",
+            "config": {
+                "model": args.model,
+                "global_batch": B,
+                "seq_len": args.prompt_len + args.warmup + args.steps,
+                "prompt_len": args.prompt_len,
+                "parallelism": f"pp{world}" if world > 1 else "single",
+            },
+        }
+        print(json.dumps(result), flush=True)
+
+    if world > 1:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

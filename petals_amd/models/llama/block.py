@@ -130,12 +130,38 @@ class LlamaBlock(nn.Module):
         self.input_layernorm = RMSNorm(config.hidden_size, eps=config.layer_norm_eps)
         self.post_attention_layernorm = RMSNorm(config.hidden_size, eps=config.layer_norm_eps)
 
+    _fast = None  # LlamaFastPath after optimize_for_inference()
+
+    def optimize_for_inference(self) -> "LlamaBlock":
+        """Repack weights into the MI355X kernel layout and enable the fused
+        decode path. Requires the HIP extension; frees nn.Linear weights."""
+        from petals_amd import ops as _ops
+        from petals_amd.ops.fused_decode import LlamaFastPath
+
+        hip = _ops._load_hip_ops()
+        if hip is None:
+            raise RuntimeError(
+                f"cannot optimize block for MI355X: HIP extension missing ({_ops._hip_import_error!r})"
+            )
+        assert next(self.parameters()).device.type == "cuda", "optimize_for_inference needs a GPU block"
+        assert not self.config.attention_bias and not self.config.mlp_bias, "fast path assumes no biases"
+        self._fast = LlamaFastPath(self, hip)
+        return self
+
     def forward(
         self,
         hidden_states: torch.Tensor,
         kv_cache: Optional[Tuple[torch.Tensor, torch.Tensor]] = None,
         prefix_length: int = 0,
     ) -> torch.Tensor:
+        if self._fast is not None:
+            if torch.is_grad_enabled() and hidden_states.requires_grad:
+                assert kv_cache is None, "training forward does not use the KV cache"
+                return self._fast.forward_autograd(hidden_states, prefix_length)
+            if kv_cache is not None and hidden_states.shape[1] == 1 and hidden_states.shape[0] <= 8:
+                return self._fast.decode_step(hidden_states, kv_cache[0], kv_cache[1], prefix_length)
+            return self._fast.forward(hidden_states, kv_cache, prefix_length)
+
         residual = hidden_states
         hidden_states = self.input_layernorm(hidden_states)
         hidden_states = self.self_attn(hidden_states, kv_cache=kv_cache, prefix_length=prefix_length)

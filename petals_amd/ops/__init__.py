@@ -49,32 +49,37 @@ def _require_hip(op_name: str):
     return ops
 
 
+def _grad_mode(*tensors) -> bool:
+    """True if this op must stay on differentiable torch ops (training path):
+    the HIP kernels are inference-only and would silently break autograd."""
+    return torch.is_grad_enabled() and any(t is not None and t.requires_grad for t in tensors)
+
+
 def rms_norm(x: torch.Tensor, weight: torch.Tensor, eps: float = 1e-5) -> torch.Tensor:
-    if x.is_cuda:
+    if x.is_cuda and not _grad_mode(x, weight):
         ops = _require_hip("rms_norm")
-        if ops is not None:
-            return ops.rms_norm(x.contiguous(), weight.contiguous(), eps)
+        if ops is not None and x.dtype == torch.bfloat16:
+            return ops.rms_norm(x.contiguous(), weight.to(torch.bfloat16).contiguous(), eps)
     return reference.rms_norm(x, weight, eps)
 
 
 def apply_rope(q, k, cos, sin, position_ids):
-    if q.is_cuda:
+    if q.is_cuda and not _grad_mode(q, k) and q.dtype == torch.bfloat16:
         ops = _require_hip("apply_rope")
         if ops is not None:
             if position_ids.dim() == 1:
                 position_ids = position_ids.unsqueeze(0).expand(q.shape[0], -1)
-            return ops.apply_rope(q, k, cos, sin, position_ids.contiguous())
+            out = ops.apply_rope(
+                q.contiguous(), k.contiguous(), cos.contiguous(), sin.contiguous(), position_ids.contiguous()
+            )
+            return out[0], out[1]
     return reference.apply_rope(q, k, cos, sin, position_ids)
 
 
 def attention(q, k, v, *, causal: bool, kv_offset: int = 0, attn_bias=None, scale=None):
-    if q.is_cuda:
-        ops = _require_hip("attention")
-        if ops is not None and attn_bias is None and q.dtype in (torch.bfloat16, torch.float16):
-            return ops.attention_prefill(
-                q.contiguous(), k.contiguous(), v.contiguous(), bool(causal), int(kv_offset),
-                float(scale) if scale is not None else -1.0,
-            )
+    # prefill attention currently runs the fp32-softmax torch composition
+    # (chunked by the backend); a flash-style HIP prefill kernel is the
+    # planned replacement. Decode attention uses attn_decode_fused.
     return reference.attention(q, k, v, causal=causal, kv_offset=kv_offset, attn_bias=attn_bias, scale=scale)
 
 
@@ -83,13 +88,30 @@ def attention_decode(q, k_cache, v_cache, kv_len: int, *, attn_bias=None, scale=
 
     q: [b, n_heads, q_len, hd]; k_cache/v_cache: [b, n_kv, max_len, hd] with
     valid prefix of length kv_len (incl. the current step already written)."""
-    if q.is_cuda:
+    if (
+        q.is_cuda
+        and not _grad_mode(q)
+        and attn_bias is None
+        and q.shape[2] == 1
+        and q.dtype in (torch.bfloat16, torch.float16)
+        and k_cache.dtype == torch.bfloat16
+    ):
         ops = _require_hip("attention_decode")
-        if ops is not None and attn_bias is None and q.dtype in (torch.bfloat16, torch.float16):
-            return ops.attention_decode(
-                q.contiguous(), k_cache, v_cache, int(kv_len),
-                float(scale) if scale is not None else -1.0,
-            )
+        if ops is not None:
+            import math
+
+            b, n_heads, _, hd = q.shape
+            n_kv = k_cache.shape[1]
+            gq = n_heads // n_kv
+            if hd in (64, 128) and gq in (1, 2, 4, 6, 8, 16):
+                kv_len_t = torch.tensor([kv_len], dtype=torch.int32, device=q.device)
+                empty = torch.empty(0, dtype=torch.float32, device=q.device)
+                qf = q.permute(0, 2, 1, 3).reshape(b, n_heads * hd).float().contiguous()
+                out = ops.attn_decode_fused(
+                    qf, k_cache.contiguous(), v_cache.contiguous(), kv_len_t, gq, 0, empty, empty,
+                    float(scale) if scale is not None else 1.0 / math.sqrt(hd),
+                )
+                return out.view(b, 1, n_heads, hd).permute(0, 2, 1, 3).to(q.dtype)
     k = k_cache[:, :, :kv_len]
     v = v_cache[:, :, :kv_len]
     q_len = q.shape[2]
@@ -99,7 +121,7 @@ def attention_decode(q, k_cache, v_cache, kv_len: int, *, attn_bias=None, scale=
 
 
 def swiglu(gate, up):
-    if gate.is_cuda:
+    if gate.is_cuda and not _grad_mode(gate, up) and gate.dtype == torch.bfloat16:
         ops = _require_hip("swiglu")
         if ops is not None:
             return ops.swiglu(gate.contiguous(), up.contiguous())

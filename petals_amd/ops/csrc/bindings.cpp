@@ -1,0 +1,28 @@
+// Python bindings for the petals_amd CDNA4 kernel suite.
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+
+torch::Tensor rms_norm(torch::Tensor x, torch::Tensor w, double eps);
+torch::Tensor rms_norm_f32out(torch::Tensor x, torch::Tensor w, double eps);
+torch::Tensor swiglu(torch::Tensor gate, torch::Tensor up);
+std::vector<torch::Tensor> apply_rope(
+    torch::Tensor q, torch::Tensor k, torch::Tensor cos_t, torch::Tensor sin_t, torch::Tensor pos);
+void rope_cache_write(
+    torch::Tensor qkv, torch::Tensor cos_t, torch::Tensor sin_t, torch::Tensor pos,
+    torch::Tensor k_cache, torch::Tensor v_cache, int64_t qh, int64_t kh);
+torch::Tensor gemv_bf16(
+    torch::Tensor wt, torch::Tensor x, torch::Tensor workspace,
+    c10::optional<torch::Tensor> residual, int64_t epilogue);
+torch::Tensor attn_decode_fused(
+    torch::Tensor q, torch::Tensor k_cache, torch::Tensor v_cache, torch::Tensor kv_len,
+    int64_t gq, int64_t n_splits, torch::Tensor part_o, torch::Tensor part_ml, double scale);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rms_norm", &rms_norm, "RMSNorm (bf16 -> bf16)");
+  m.def("rms_norm_f32out", &rms_norm_f32out, "RMSNorm (bf16 -> f32)");
+  m.def("swiglu", &swiglu, "silu(gate) * up (bf16)");
+  m.def("apply_rope", &apply_rope, "rotate q,k by positions (bf16)");
+  m.def("rope_cache_write", &rope_cache_write, "fused decode rope + kv cache write");
+  m.def("gemv_bf16", &gemv_bf16, "split-K bf16 gemv with fused epilogue");
+  m.def("attn_decode_fused", &attn_decode_fused, "GQA decode attention (flash-decoding)");
+}

@@ -1,0 +1,281 @@
+// Elementwise / normalization kernels for CDNA4 (gfx950): RMSNorm, LayerNorm,
+// RoPE + KV-cache write, SwiGLU. All bf16 IO vectorized as short8 (G13 of the
+// CDNA4 guide: hipcc does not auto-vectorize bf16 loads).
+
+#include "common.h"
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+
+// --------------------------------------------------------------- rms_norm
+
+// one workgroup (256 threads) per row; row held in registers (short8 chunks)
+template <bool OUT_F32>
+__global__ void rms_norm_kernel(
+    const unsigned short* __restrict__ x,  // [rows, dim] bf16
+    const unsigned short* __restrict__ w,  // [dim] bf16
+    void* __restrict__ y,                  // [rows, dim] bf16 or f32
+    int dim,
+    float eps) {
+  constexpr int T = 256;
+  const int row = blockIdx.x;
+  const int tid = threadIdx.x;
+  const unsigned short* xr = x + (size_t)row * dim;
+
+  float sumsq = 0.f;
+  const int chunk = T * 8;
+  for (int base = 0; base < dim; base += chunk) {
+    const int idx = base + tid * 8;
+    if (idx + 8 <= dim) {
+      const short8 v = *reinterpret_cast<const short8*>(xr + idx);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const float f = bf16_to_f32((unsigned short)v[j]);
+        sumsq = fmaf(f, f, sumsq);
+      }
+    } else {
+      for (int j = idx; j < dim; ++j) {
+        const float f = bf16_to_f32(xr[j]);
+        sumsq = fmaf(f, f, sumsq);
+      }
+    }
+  }
+  __shared__ float red[T / WAVE];
+  sumsq = wave_reduce_sum(sumsq);
+  if ((tid & (WAVE - 1)) == 0) red[tid / WAVE] = sumsq;
+  __syncthreads();
+  float total = 0.f;
+#pragma unroll
+  for (int i = 0; i < T / WAVE; ++i) total += red[i];
+  const float inv = rsqrtf(total / dim + eps);
+
+  for (int base = 0; base < dim; base += chunk) {
+    const int idx = base + tid * 8;
+    if (idx + 8 <= dim) {
+      const short8 v = *reinterpret_cast<const short8*>(xr + idx);
+      const short8 wv = *reinterpret_cast<const short8*>(w + idx);
+      if (OUT_F32) {
+        float* yr = reinterpret_cast<float*>(y) + (size_t)row * dim + idx;
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          yr[j] = bf16_to_f32((unsigned short)v[j]) * inv * bf16_to_f32((unsigned short)wv[j]);
+      } else {
+        short8 out;
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          out[j] = (short)f32_to_bf16(bf16_to_f32((unsigned short)v[j]) * inv * bf16_to_f32((unsigned short)wv[j]));
+        *reinterpret_cast<short8*>(reinterpret_cast<unsigned short*>(y) + (size_t)row * dim + idx) = out;
+      }
+    } else {
+      for (int j = idx; j < dim; ++j) {
+        const float f = bf16_to_f32(xr[j]) * inv * bf16_to_f32(w[j]);
+        if (OUT_F32)
+          reinterpret_cast<float*>(y)[(size_t)row * dim + j] = f;
+        else
+          reinterpret_cast<unsigned short*>(y)[(size_t)row * dim + j] = f32_to_bf16(f);
+      }
+    }
+  }
+}
+
+torch::Tensor rms_norm(torch::Tensor x, torch::Tensor w, double eps) {
+  TORCH_CHECK(x.is_cuda() && x.dtype() == torch::kBFloat16, "rms_norm expects bf16 CUDA tensor");
+  auto shape = x.sizes().vec();
+  const int dim = shape.back();
+  auto x2 = x.contiguous().view({-1, dim});
+  const int rows = x2.size(0);
+  auto y = torch::empty_like(x2);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  rms_norm_kernel<false><<<rows, 256, 0, stream>>>(
+      reinterpret_cast<const unsigned short*>(x2.data_ptr()),
+      reinterpret_cast<const unsigned short*>(w.contiguous().data_ptr()),
+      y.data_ptr(), dim, (float)eps);
+  HIP_CHECK_LAST();
+  return y.view(shape);
+}
+
+torch::Tensor rms_norm_f32out(torch::Tensor x, torch::Tensor w, double eps) {
+  TORCH_CHECK(x.is_cuda() && x.dtype() == torch::kBFloat16);
+  auto shape = x.sizes().vec();
+  const int dim = shape.back();
+  auto x2 = x.contiguous().view({-1, dim});
+  const int rows = x2.size(0);
+  auto y = torch::empty({rows, dim}, x.options().dtype(torch::kFloat32));
+  auto stream = at::cuda::getCurrentCUDAStream();
+  rms_norm_kernel<true><<<rows, 256, 0, stream>>>(
+      reinterpret_cast<const unsigned short*>(x2.data_ptr()),
+      reinterpret_cast<const unsigned short*>(w.contiguous().data_ptr()),
+      y.data_ptr(), dim, (float)eps);
+  HIP_CHECK_LAST();
+  return y.view(shape);
+}
+
+// ------------------------------------------------------------------ swiglu
+
+__global__ void swiglu_kernel(
+    const unsigned short* __restrict__ gate,
+    const unsigned short* __restrict__ up,
+    unsigned short* __restrict__ y,
+    long n) {
+  const long i = (long)(blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  if (i + 8 <= n) {
+    const short8 g8 = *reinterpret_cast<const short8*>(gate + i);
+    const short8 u8 = *reinterpret_cast<const short8*>(up + i);
+    short8 o8;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const float g = bf16_to_f32((unsigned short)g8[j]);
+      const float u = bf16_to_f32((unsigned short)u8[j]);
+      o8[j] = (short)f32_to_bf16(g / (1.f + __expf(-g)) * u);
+    }
+    *reinterpret_cast<short8*>(y + i) = o8;
+  } else {
+    for (long j = i; j < n; ++j) {
+      const float g = bf16_to_f32(gate[j]);
+      const float u = bf16_to_f32(up[j]);
+      y[j] = f32_to_bf16(g / (1.f + __expf(-g)) * u);
+    }
+  }
+}
+
+torch::Tensor swiglu(torch::Tensor gate, torch::Tensor up) {
+  TORCH_CHECK(gate.is_cuda() && gate.dtype() == torch::kBFloat16);
+  auto g = gate.contiguous(), u = up.contiguous();
+  auto y = torch::empty_like(g);
+  const long n = g.numel();
+  const long blocks = std::min((n / 8 + 255) / 256 + 1, (long)4096);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  swiglu_kernel<<<blocks, 256, 0, stream>>>(
+      reinterpret_cast<const unsigned short*>(g.data_ptr()),
+      reinterpret_cast<const unsigned short*>(u.data_ptr()),
+      reinterpret_cast<unsigned short*>(y.data_ptr()), n);
+  HIP_CHECK_LAST();
+  return y;
+}
+
+// ----------------------------------------------- rope (batched, bf16 q/k)
+
+// q,k: [b, heads, len, hd] bf16; cos/sin: [max_pos, hd] f32; positions [b, len]
+__global__ void rope_kernel(
+    unsigned short* __restrict__ q,
+    unsigned short* __restrict__ k,
+    const float* __restrict__ cos_t,
+    const float* __restrict__ sin_t,
+    const long* __restrict__ pos,  // [b, len]
+    int b, int qh, int kh, int len, int hd) {
+  const int half = hd / 2;
+  const long total = (long)b * (qh + kh) * len * half;
+  for (long idx = (long)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * blockDim.x) {
+    long t = idx;
+    const int d = t % half; t /= half;
+    const int l = t % len; t /= len;
+    const int h = t % (qh + kh); t /= (qh + kh);
+    const int bi = t;
+    unsigned short* base = (h < qh)
+        ? q + (((size_t)bi * qh + h) * len + l) * hd
+        : k + (((size_t)bi * kh + (h - qh)) * len + l) * hd;
+    const long p = pos[(size_t)bi * len + l];
+    const float c = cos_t[(size_t)p * hd + d];
+    const float s = sin_t[(size_t)p * hd + d];
+    const float x1 = bf16_to_f32(base[d]);
+    const float x2 = bf16_to_f32(base[d + half]);
+    base[d] = f32_to_bf16(x1 * c - x2 * s);
+    base[d + half] = f32_to_bf16(x2 * c + x1 * s);
+  }
+}
+
+// returns (q, k) rotated copies
+std::vector<torch::Tensor> apply_rope(
+    torch::Tensor q, torch::Tensor k, torch::Tensor cos_t, torch::Tensor sin_t, torch::Tensor pos) {
+  TORCH_CHECK(q.is_cuda() && q.dim() == 4 && k.dim() == 4);
+  TORCH_CHECK(q.dtype() == torch::kBFloat16 && k.dtype() == torch::kBFloat16);
+  auto qc = q.contiguous().clone();
+  auto kc = k.contiguous().clone();
+  auto cf = cos_t.to(torch::kFloat32).contiguous();
+  auto sf = sin_t.to(torch::kFloat32).contiguous();
+  auto pc = pos.to(torch::kInt64).contiguous();
+  const int b = qc.size(0), qh = qc.size(1), len = qc.size(2), hd = qc.size(3);
+  const int kh = kc.size(1);
+  TORCH_CHECK(pc.dim() == 2 && pc.size(0) == b && pc.size(1) == len, "positions must be [b, len]");
+  const long total = (long)b * (qh + kh) * len * (hd / 2);
+  const long blocks = std::min((total + 255) / 256 + 1, (long)4096);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  rope_kernel<<<blocks, 256, 0, stream>>>(
+      reinterpret_cast<unsigned short*>(qc.data_ptr()),
+      reinterpret_cast<unsigned short*>(kc.data_ptr()),
+      cf.data_ptr<float>(), sf.data_ptr<float>(), pc.data_ptr<long>(),
+      b, qh, kh, len, hd);
+  HIP_CHECK_LAST();
+  return {qc, kc};
+}
+
+// -------------------------- fused decode rope + kv-cache write (fast path)
+
+// qkv: [b, qh*hd + 2*kh*hd] f32 (one token per row). Rotates q in place,
+// rotates k and writes k/v into the caches (bf16) at row *pos_ptr.
+__global__ void rope_cache_write_kernel(
+    float* __restrict__ qkv,
+    const float* __restrict__ cos_t,
+    const float* __restrict__ sin_t,
+    const int* __restrict__ pos_ptr,
+    unsigned short* __restrict__ k_cache,  // [bcap, kh, lmax, hd]
+    unsigned short* __restrict__ v_cache,
+    int b, int qh, int kh, int lmax, int hd) {
+  const int half = hd / 2;
+  const int p = *pos_ptr;
+  const float* crow = cos_t + (size_t)p * hd;
+  const float* srow = sin_t + (size_t)p * hd;
+  const int row_elems = qh * hd + 2 * kh * hd;
+  // rotate q+k pairs: b * (qh+kh) * half work items; copy v: b * kh * hd
+  const long rot_total = (long)b * (qh + kh) * half;
+  const long v_total = (long)b * kh * hd;
+  for (long idx = (long)blockIdx.x * blockDim.x + threadIdx.x; idx < rot_total + v_total;
+       idx += (long)gridDim.x * blockDim.x) {
+    if (idx < rot_total) {
+      long t = idx;
+      const int d = t % half; t /= half;
+      const int h = t % (qh + kh); t /= (qh + kh);
+      const int bi = t;
+      const float c = crow[d], s = srow[d];
+      if (h < qh) {
+        float* base = qkv + (size_t)bi * row_elems + h * hd;
+        const float x1 = base[d], x2 = base[d + half];
+        base[d] = x1 * c - x2 * s;
+        base[d + half] = x2 * c + x1 * s;
+      } else {
+        const int kh_i = h - qh;
+        float* base = qkv + (size_t)bi * row_elems + qh * hd + kh_i * hd;
+        const float x1 = base[d], x2 = base[d + half];
+        unsigned short* krow = k_cache + (((size_t)bi * kh + kh_i) * lmax + p) * hd;
+        krow[d] = f32_to_bf16(x1 * c - x2 * s);
+        krow[d + half] = f32_to_bf16(x2 * c + x1 * s);
+      }
+    } else {
+      long t = idx - rot_total;
+      const int d = t % hd; t /= hd;
+      const int kh_i = t % kh; t /= kh;
+      const int bi = t;
+      const float v = qkv[(size_t)bi * row_elems + qh * hd + kh * hd + kh_i * hd + d];
+      v_cache[(((size_t)bi * kh + kh_i) * lmax + p) * hd + d] = f32_to_bf16(v);
+    }
+  }
+}
+
+void rope_cache_write(
+    torch::Tensor qkv, torch::Tensor cos_t, torch::Tensor sin_t, torch::Tensor pos,
+    torch::Tensor k_cache, torch::Tensor v_cache, int64_t qh, int64_t kh) {
+  TORCH_CHECK(qkv.is_cuda() && qkv.dtype() == torch::kFloat32 && qkv.dim() == 2);
+  TORCH_CHECK(k_cache.dtype() == torch::kBFloat16 && k_cache.dim() == 4);
+  const int b = qkv.size(0);
+  const int hd = k_cache.size(3), lmax = k_cache.size(2);
+  const long total = (long)b * (qh + kh) * (hd / 2) + (long)b * kh * hd;
+  const long blocks = std::min((total + 255) / 256 + 1, (long)1024);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  rope_cache_write_kernel<<<blocks, 256, 0, stream>>>(
+      qkv.data_ptr<float>(), cos_t.data_ptr<float>(), sin_t.data_ptr<float>(),
+      pos.data_ptr<int>(),
+      reinterpret_cast<unsigned short*>(k_cache.data_ptr()),
+      reinterpret_cast<unsigned short*>(v_cache.data_ptr()),
+      b, (int)qh, (int)kh, lmax, hd);
+  HIP_CHECK_LAST();
+}

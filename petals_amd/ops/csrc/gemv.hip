@@ -1,0 +1,217 @@
+// Decode-path GEMV suite for CDNA4 (gfx950).
+//
+// Weights are stored TRANSPOSED: Wt[in, out] bf16 row-major (= W^T). A wave
+// owns a strip of 512 output columns (8 per lane, one short8 = 16 B/lane per
+// input row) and iterates over a chunk of the input dimension; x[i] is a
+// wave-uniform scalar load (s_load, L2-resident) broadcast into 8 FMAs.
+// The input dimension is split across gridDim.y workgroups so 256 CUs stay
+// busy even for a single-token GEMV; each split writes a partial fp32 strip,
+// and gemv_reduce sums splits deterministically while fusing the epilogue
+// (plain / +residual / SwiGLU-pair) and dtype conversion.
+//
+// Replaces (capability-wise) the per-token GEMMs the reference runs through
+// torch/CUDA inside HF blocks (reference models/llama/block.py:44-127); the
+// design follows /opt/skills/guides/cdna_hip_programming.md Appendix B
+// ("Element-wise"/"GEMM" vectorization rules; G13).
+
+#include "common.h"
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+
+#ifndef GEMV_OUT_PER_WAVE
+#define GEMV_OUT_PER_WAVE 512  // 64 lanes x 8 bf16
+#endif
+
+// ---------------------------------------------------------------- gemv core
+
+template <int BATCH>
+__global__ void gemv_bf16_kernel(
+    const unsigned short* __restrict__ wt,  // [in, out]
+    const float* __restrict__ x,            // [BATCH, in]
+    float* __restrict__ partials,           // [n_splits, BATCH, out]
+    int in_dim,
+    int out_dim,
+    int i_per_split) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int out0 = blockIdx.x * GEMV_OUT_PER_WAVE + lane * 8;
+  if (out0 >= out_dim) return;
+  const int split = blockIdx.y;
+  const int i_begin = split * i_per_split;
+  const int i_end = min(i_begin + i_per_split, in_dim);
+
+  float acc[BATCH][8];
+#pragma unroll
+  for (int b = 0; b < BATCH; ++b)
+#pragma unroll
+    for (int v = 0; v < 8; ++v) acc[b][v] = 0.f;
+
+  const bool full = (out0 + 8) <= out_dim;
+  if (full) {
+    for (int i = i_begin; i < i_end; ++i) {
+      const short8 w8 = *reinterpret_cast<const short8*>(wt + (size_t)i * out_dim + out0);
+      float wf[8];
+#pragma unroll
+      for (int v = 0; v < 8; ++v) wf[v] = bf16_to_f32((unsigned short)w8[v]);
+#pragma unroll
+      for (int b = 0; b < BATCH; ++b) {
+        const float xv = x[(size_t)b * in_dim + i];  // wave-uniform -> s_load
+#pragma unroll
+        for (int v = 0; v < 8; ++v) acc[b][v] = fmaf(wf[v], xv, acc[b][v]);
+      }
+    }
+  } else {
+    const int nv = out_dim - out0;
+    for (int i = i_begin; i < i_end; ++i) {
+      for (int b = 0; b < BATCH; ++b) {
+        const float xv = x[(size_t)b * in_dim + i];
+        for (int v = 0; v < nv; ++v)
+          acc[b][v] = fmaf(bf16_to_f32(wt[(size_t)i * out_dim + out0 + v]), xv, acc[b][v]);
+      }
+    }
+  }
+
+#pragma unroll
+  for (int b = 0; b < BATCH; ++b) {
+    float* dst = partials + ((size_t)split * BATCH + b) * out_dim + out0;
+    if (full) {
+      float4v* d4 = reinterpret_cast<float4v*>(dst);
+      d4[0] = float4v{acc[b][0], acc[b][1], acc[b][2], acc[b][3]};
+      d4[1] = float4v{acc[b][4], acc[b][5], acc[b][6], acc[b][7]};
+    } else {
+      for (int v = 0; v < out_dim - out0; ++v) dst[v] = acc[b][v];
+    }
+  }
+}
+
+// ------------------------------------------------------------- reduce + epi
+
+enum GemvEpilogue : int {
+  EPI_PLAIN_F32 = 0,     // y_f32[b, out] = sum
+  EPI_PLAIN_BF16 = 1,    // y_bf16[b, out] = sum
+  EPI_RESIDUAL_BF16 = 2, // y_bf16 = residual_bf16 + sum
+  EPI_SWIGLU_F32 = 3,    // out = silu(sum[:half]) * sum[half:], y_f32[b, half]
+};
+
+__global__ void gemv_reduce_kernel(
+    const float* __restrict__ partials,  // [n_splits, batch, out]
+    const unsigned short* __restrict__ residual,  // [batch, out] or null
+    void* __restrict__ y,
+    int n_splits,
+    int batch,
+    int out_dim,
+    int epilogue) {
+  const int half = out_dim >> 1;
+  const int n_out = (epilogue == EPI_SWIGLU_F32) ? half : out_dim;
+  const int total = batch * n_out;
+  for (int idx = blockIdx.x * blockDim.x + threadIdx.x; idx < total; idx += gridDim.x * blockDim.x) {
+    const int b = idx / n_out;
+    const int o = idx - b * n_out;
+    if (epilogue == EPI_SWIGLU_F32) {
+      float g = 0.f, u = 0.f;
+      for (int s = 0; s < n_splits; ++s) {
+        const float* base = partials + ((size_t)s * batch + b) * out_dim;
+        g += base[o];
+        u += base[o + half];
+      }
+      const float act = g / (1.f + __expf(-g)) * u;
+      reinterpret_cast<float*>(y)[(size_t)b * half + o] = act;
+    } else {
+      float sum = 0.f;
+      for (int s = 0; s < n_splits; ++s) sum += partials[((size_t)s * batch + b) * out_dim + o];
+      if (epilogue == EPI_PLAIN_F32) {
+        reinterpret_cast<float*>(y)[(size_t)b * out_dim + o] = sum;
+      } else if (epilogue == EPI_PLAIN_BF16) {
+        reinterpret_cast<unsigned short*>(y)[(size_t)b * out_dim + o] = f32_to_bf16(sum);
+      } else {  // EPI_RESIDUAL_BF16
+        const float r = bf16_to_f32(residual[(size_t)b * out_dim + o]);
+        reinterpret_cast<unsigned short*>(y)[(size_t)b * out_dim + o] = f32_to_bf16(r + sum);
+      }
+    }
+  }
+}
+
+// ------------------------------------------------------------------- host
+
+static int pick_splits(int in_dim, int out_dim) {
+  const long out_waves = (out_dim + GEMV_OUT_PER_WAVE - 1) / GEMV_OUT_PER_WAVE;
+  // target ~1024 workgroups so 256 CUs have >=4 each; round splits to a
+  // divisor-ish value of in_dim chunks of >=128 rows
+  long splits = (1024 + out_waves - 1) / out_waves;
+  long max_splits = (in_dim + 127) / 128;
+  if (splits > max_splits) splits = max_splits;
+  if (splits < 1) splits = 1;
+  return (int)splits;
+}
+
+torch::Tensor gemv_bf16(
+    torch::Tensor wt,        // [in, out] bf16
+    torch::Tensor x,         // [batch, in] f32
+    torch::Tensor workspace, // [n_splits_max, batch, out] f32 (preallocated, may be empty)
+    c10::optional<torch::Tensor> residual,  // [batch, out] bf16
+    int64_t epilogue) {
+  TORCH_CHECK(wt.is_cuda() && wt.dtype() == torch::kBFloat16 && wt.dim() == 2);
+  TORCH_CHECK(x.is_cuda() && x.dtype() == torch::kFloat32 && x.dim() == 2);
+  const int in_dim = wt.size(0), out_dim = wt.size(1);
+  TORCH_CHECK(x.size(1) == in_dim, "x/in mismatch");
+  const int batch = x.size(0);
+  TORCH_CHECK(batch <= 8, "decode gemv supports batch <= 8");
+
+  const int splits = pick_splits(in_dim, out_dim);
+  const int i_per_split = (in_dim + splits - 1) / splits;
+
+  torch::Tensor partials;
+  if (workspace.numel() >= (int64_t)splits * batch * out_dim) {
+    partials = workspace;
+  } else {
+    partials = torch::empty({(int64_t)splits, batch, out_dim}, x.options());
+  }
+
+  const int out_waves = (out_dim + GEMV_OUT_PER_WAVE - 1) / GEMV_OUT_PER_WAVE;
+  dim3 grid(out_waves, splits);
+  dim3 block(WAVE);
+  auto stream = at::cuda::getCurrentCUDAStream();
+
+  const unsigned short* wt_p = reinterpret_cast<const unsigned short*>(wt.data_ptr());
+  const float* x_p = x.data_ptr<float>();
+  float* part_p = partials.data_ptr<float>();
+
+#define LAUNCH_GEMV(B)                                                        \
+  gemv_bf16_kernel<B><<<grid, block, 0, stream>>>(wt_p, x_p, part_p, in_dim, out_dim, i_per_split)
+  switch (batch) {
+    case 1: LAUNCH_GEMV(1); break;
+    case 2: LAUNCH_GEMV(2); break;
+    case 3: LAUNCH_GEMV(3); break;
+    case 4: LAUNCH_GEMV(4); break;
+    case 5: LAUNCH_GEMV(5); break;
+    case 6: LAUNCH_GEMV(6); break;
+    case 7: LAUNCH_GEMV(7); break;
+    case 8: LAUNCH_GEMV(8); break;
+    default: TORCH_CHECK(false, "decode gemv supports batch <= 8");
+  }
+#undef LAUNCH_GEMV
+  HIP_CHECK_LAST();
+
+  // epilogue
+  const int half = out_dim / 2;
+  const int n_out = (epilogue == EPI_SWIGLU_F32) ? half : out_dim;
+  torch::Tensor y;
+  if (epilogue == EPI_PLAIN_F32) {
+    y = torch::empty({batch, out_dim}, x.options());
+  } else if (epilogue == EPI_SWIGLU_F32) {
+    y = torch::empty({batch, half}, x.options());
+  } else {
+    y = torch::empty({batch, out_dim}, wt.options());
+  }
+  const unsigned short* res_p = nullptr;
+  if (epilogue == EPI_RESIDUAL_BF16) {
+    TORCH_CHECK(residual.has_value(), "residual required");
+    TORCH_CHECK(residual->dtype() == torch::kBFloat16 && residual->is_contiguous());
+    res_p = reinterpret_cast<const unsigned short*>(residual->data_ptr());
+  }
+  const int total = batch * n_out;
+  int rblocks = std::min((total + 255) / 256, 2048);
+  gemv_reduce_kernel<<<rblocks, 256, 0, stream>>>(
+      part_p, res_p, y.data_ptr(), splits, batch, out_dim, (int)epilogue);
+  HIP_CHECK_LAST();
+  return y;
+}

@@ -1,0 +1,197 @@
+"""MI355X fast path for Llama-family blocks.
+
+`LlamaFastPath` repacks a loaded LlamaBlock's weights into kernel-optimal
+layout (transposed [in, out] bf16, QKV and gate/up concatenated) and then runs:
+
+  decode (q_len == 1, any batch <= 8):
+      rms_norm_f32out -> gemv(QKV) -> rope_cache_write -> attn_decode_fused
+      -> gemv(O, +residual) -> rms_norm_f32out -> gemv(gate|up, SwiGLU epi)
+      -> gemv(down, +residual)
+    ... 8 kernel launches per block, every byte of weights read exactly once
+    (HBM-bound by design; see profiles/).
+
+  prefill (q_len > 1): bf16 GEMMs via torch.matmul on the SAME transposed
+    weights (rocBLAS), HIP rms_norm / rope / swiglu kernels, fp32-softmax
+    attention (chunked by the backend; flash prefill kernel is the planned
+    upgrade).
+
+The original nn.Linear weights are FREED after repacking (a 70B span must not
+hold two copies).
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Dict, Optional, Tuple
+
+import torch
+
+from petals_amd.ops import reference
+
+_EPI_PLAIN_F32 = 0
+_EPI_PLAIN_BF16 = 1
+_EPI_RESIDUAL_BF16 = 2
+_EPI_SWIGLU_F32 = 3
+
+_workspaces: Dict[Tuple[torch.device, str], torch.Tensor] = {}
+
+
+def _get_ws(device: torch.device, key: str, numel: int) -> torch.Tensor:
+    k = (device, key)
+    ws = _workspaces.get(k)
+    if ws is None or ws.numel() < numel:
+        ws = torch.empty(numel, dtype=torch.float32, device=device)
+        _workspaces[k] = ws
+    return ws
+
+
+class LlamaFastPath:
+    def __init__(self, block, hip_ops):
+        cfg = block.config
+        self.hip = hip_ops
+        self.cfg = cfg
+        self.hd = cfg.head_dim
+        self.qh = cfg.num_attention_heads
+        self.kh = cfg.n_kv_heads
+        self.gq = self.qh // self.kh
+        self.scale = 1.0 / math.sqrt(self.hd)
+        self.eps = cfg.layer_norm_eps
+
+        attn = block.self_attn
+        device = attn.q_proj.weight.device
+
+        def t(w):
+            return w.detach().to(torch.bfloat16).t().contiguous()
+
+        self.wqkv_t = torch.cat(
+            [t(attn.q_proj.weight), t(attn.k_proj.weight), t(attn.v_proj.weight)], dim=1
+        )  # [H, qh*hd + 2*kh*hd]
+        self.wo_t = t(attn.o_proj.weight)
+        mlp = block.mlp
+        self.wgateup_t = torch.cat([t(mlp.gate_proj.weight), t(mlp.up_proj.weight)], dim=1)
+        self.wdown_t = t(mlp.down_proj.weight)
+        self.ln1_w = block.input_layernorm.weight.detach().to(torch.bfloat16).contiguous()
+        self.ln2_w = block.post_attention_layernorm.weight.detach().to(torch.bfloat16).contiguous()
+
+        # free the originals (they'd double the span's memory)
+        empty = torch.empty(0, device=device, dtype=torch.bfloat16)
+        for lin in (attn.q_proj, attn.k_proj, attn.v_proj, attn.o_proj, mlp.gate_proj, mlp.up_proj, mlp.down_proj):
+            lin.weight.data = empty
+        self.has_bias = attn.q_proj.bias is not None
+
+        self.rope_cos: Optional[torch.Tensor] = None
+        self.rope_sin: Optional[torch.Tensor] = None
+        self._pos = torch.zeros(1, dtype=torch.int32, device=device)
+        self._kv_len = torch.zeros(1, dtype=torch.int32, device=device)
+        self.device = device
+        self._empty_f32 = torch.empty(0, dtype=torch.float32, device=device)
+
+    def _ensure_rope(self, needed: int):
+        if self.rope_cos is None or self.rope_cos.shape[0] < needed:
+            cos, sin = reference.build_rope_cache(
+                self.hd,
+                max(needed, self.cfg.max_position_embeddings),
+                theta=self.cfg.rope_theta,
+                rope_scaling=self.cfg.rope_scaling,
+            )
+            self.rope_cos = cos.to(self.device, torch.float32).contiguous()
+            self.rope_sin = sin.to(self.device, torch.float32).contiguous()
+
+    # ------------------------------------------------------------- decode
+
+    @torch.inference_mode()
+    def decode_step(
+        self,
+        hidden: torch.Tensor,  # [B, 1, H] bf16
+        k_cache: torch.Tensor,  # [Bc, KV, Lmax, hd] bf16
+        v_cache: torch.Tensor,
+        prefix_length: int,
+    ) -> torch.Tensor:
+        B = hidden.shape[0]
+        H = hidden.shape[-1]
+        h = hidden.view(B, H)
+        if h.dtype != torch.bfloat16:
+            h = h.to(torch.bfloat16)
+        h = h.contiguous()
+        self._ensure_rope(prefix_length + 1)
+        self._pos.fill_(prefix_length)
+        self._kv_len.fill_(prefix_length + 1)
+        ws = _get_ws(self.device, "gemv", 64 * B * max(self.wgateup_t.shape[1], self.wqkv_t.shape[1], H))
+
+        xn = self.hip.rms_norm_f32out(h, self.ln1_w, self.eps)  # [B, H] f32
+        qkv = self.hip.gemv_bf16(self.wqkv_t, xn, ws, None, _EPI_PLAIN_F32)  # [B, qkv] f32
+        self.hip.rope_cache_write(
+            qkv, self.rope_cos, self.rope_sin, self._pos, k_cache[:B], v_cache[:B], self.qh, self.kh
+        )
+        q = qkv[:, : self.qh * self.hd]
+        attn = self.hip.attn_decode_fused(
+            q.contiguous(), k_cache[:B], v_cache[:B], self._kv_len, self.gq, 0,
+            self._empty_f32, self._empty_f32, self.scale,
+        )  # [B, H] f32
+        h2 = self.hip.gemv_bf16(self.wo_t, attn, ws, h, _EPI_RESIDUAL_BF16)  # [B, H] bf16
+        xn2 = self.hip.rms_norm_f32out(h2, self.ln2_w, self.eps)
+        act = self.hip.gemv_bf16(self.wgateup_t, xn2, ws, None, _EPI_SWIGLU_F32)  # [B, I] f32
+        h3 = self.hip.gemv_bf16(self.wdown_t, act, ws, h2, _EPI_RESIDUAL_BF16)
+        return h3.view(B, 1, H)
+
+    # --------------------------------------------------- training (autograd)
+
+    def forward_autograd(self, hidden: torch.Tensor, prefix_length: int = 0) -> torch.Tensor:
+        """Differentiable forward on the transposed weights (torch primitives
+        only — the HIP kernels are inference-only). Used by rpc_backward on GPU
+        servers; weights are frozen, grads flow to inputs/prompts."""
+        B, S, H = hidden.shape
+        end = prefix_length + S
+        self._ensure_rope(end)
+        xn = reference.rms_norm(hidden, self.ln1_w, self.eps)
+        qkv = torch.matmul(xn, self.wqkv_t)
+        q = qkv[..., : self.qh * self.hd].view(B, S, self.qh, self.hd).transpose(1, 2)
+        k = qkv[..., self.qh * self.hd : (self.qh + self.kh) * self.hd].view(B, S, self.kh, self.hd).transpose(1, 2)
+        v = qkv[..., (self.qh + self.kh) * self.hd :].view(B, S, self.kh, self.hd).transpose(1, 2)
+        pos = torch.arange(prefix_length, end, device=hidden.device)
+        q, k = reference.apply_rope(q, k, self.rope_cos, self.rope_sin, pos)
+        attn = reference.attention(q, k, v, causal=True)
+        attn = attn.transpose(1, 2).reshape(B, S, self.qh * self.hd).to(hidden.dtype)
+        h2 = hidden + torch.matmul(attn, self.wo_t)
+        xn2 = reference.rms_norm(h2, self.ln2_w, self.eps)
+        gateup = torch.matmul(xn2, self.wgateup_t)
+        inter = self.wgateup_t.shape[1] // 2
+        act = reference.swiglu(gateup[..., :inter], gateup[..., inter:]).to(hidden.dtype)
+        return h2 + torch.matmul(act, self.wdown_t)
+
+    # ------------------------------------------------------------ prefill
+
+    def forward(
+        self,
+        hidden: torch.Tensor,  # [B, S, H] bf16
+        kv_cache: Optional[Tuple[torch.Tensor, torch.Tensor]],
+        prefix_length: int,
+    ) -> torch.Tensor:
+        B, S, H = hidden.shape
+        hidden = hidden.to(torch.bfloat16)
+        end = prefix_length + S
+        self._ensure_rope(end)
+
+        xn = self.hip.rms_norm(hidden, self.ln1_w, self.eps)
+        qkv = torch.matmul(xn, self.wqkv_t)  # [B, S, qkv] bf16 (rocBLAS)
+        q = qkv[..., : self.qh * self.hd].view(B, S, self.qh, self.hd).transpose(1, 2)
+        k = qkv[..., self.qh * self.hd : (self.qh + self.kh) * self.hd].view(B, S, self.kh, self.hd).transpose(1, 2)
+        v = qkv[..., (self.qh + self.kh) * self.hd :].view(B, S, self.kh, self.hd).transpose(1, 2)
+        pos = torch.arange(prefix_length, end, device=self.device).unsqueeze(0).expand(B, S).contiguous()
+        q, k = self.hip.apply_rope(q.contiguous(), k.contiguous(), self.rope_cos, self.rope_sin, pos)
+        if kv_cache is not None:
+            k_cache, v_cache = kv_cache
+            k_cache[:B, :, prefix_length:end].copy_(k)
+            v_cache[:B, :, prefix_length:end].copy_(v)
+            attn = reference.attention(
+                q, k_cache[:B, :, :end], v_cache[:B, :, :end], causal=S > 1, kv_offset=prefix_length
+            )
+        else:
+            attn = reference.attention(q, k, v, causal=True)
+        attn = attn.transpose(1, 2).reshape(B, S, self.qh * self.hd).to(torch.bfloat16)
+        h2 = hidden + torch.matmul(attn, self.wo_t)
+        xn2 = self.hip.rms_norm(h2, self.ln2_w, self.eps)
+        gateup = torch.matmul(xn2, self.wgateup_t)
+        inter = self.wgateup_t.shape[1] // 2
+        act = self.hip.swiglu(gateup[..., :inter].contiguous(), gateup[..., inter:].contiguous())
+        return h2 + torch.matmul(act, self.wdown_t)

@@ -82,7 +82,11 @@ def load_pretrained_block(
             logger.warning("block %d: missing keys %s", block_index, report.missing_keys)
     else:
         init_random_block_(block, config, block_index)
-    return block.to(device=device, dtype=torch_dtype).eval()
+    block = block.to(device=device, dtype=torch_dtype).eval()
+    device = torch.device(device)
+    if device.type == "cuda" and hasattr(block, "optimize_for_inference"):
+        block.optimize_for_inference()
+    return block
 
 
 def init_random_block_(block: torch.nn.Module, config: ModelConfig, block_index: int) -> None:

@@ -1,0 +1,199 @@
+"""Numerics tests for the CDNA4 HIP kernel suite vs plain-PyTorch fp32
+references (run on a real MI355X via `pytest -m gpu`)."""
+
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+requires_gpu = pytest.mark.skipif(not torch.cuda.is_available(), reason="needs MI355X")
+
+
+@pytest.fixture(scope="module")
+def hip():
+    from petals_amd import ops
+
+    mod = ops._load_hip_ops()
+    assert mod is not None, f"HIP extension must load on a GPU box: {ops._hip_import_error!r}"
+    return mod
+
+
+@requires_gpu
+def test_rms_norm(hip):
+    from petals_amd.ops import reference
+
+    torch.manual_seed(0)
+    x = torch.randn(5, 4096, device="cuda").to(torch.bfloat16)
+    w = torch.randn(4096, device="cuda").to(torch.bfloat16)
+    out = hip.rms_norm(x, w, 1e-5)
+    ref = reference.rms_norm(x.float().cpu(), w.float().cpu(), 1e-5)
+    assert torch.allclose(out.float().cpu(), ref, atol=2e-2, rtol=2e-2)
+
+    out32 = hip.rms_norm_f32out(x, w, 1e-5)
+    assert out32.dtype == torch.float32
+    assert torch.allclose(out32.cpu(), ref, atol=2e-2, rtol=2e-2)
+
+
+@requires_gpu
+def test_swiglu(hip):
+    torch.manual_seed(0)
+    g = torch.randn(3, 1000, device="cuda").to(torch.bfloat16)
+    u = torch.randn(3, 1000, device="cuda").to(torch.bfloat16)
+    out = hip.swiglu(g, u)
+    ref = torch.nn.functional.silu(g.float()) * u.float()
+    assert torch.allclose(out.float(), ref, atol=3e-2, rtol=3e-2)
+
+
+@requires_gpu
+def test_apply_rope(hip):
+    from petals_amd.ops import reference
+
+    torch.manual_seed(0)
+    b, qh, kh, s, hd = 2, 8, 2, 7, 128
+    q = torch.randn(b, qh, s, hd, device="cuda").to(torch.bfloat16)
+    k = torch.randn(b, kh, s, hd, device="cuda").to(torch.bfloat16)
+    cos, sin = reference.build_rope_cache(hd, 64)
+    pos = torch.arange(3, 3 + s).unsqueeze(0).expand(b, s).contiguous().cuda()
+    q2, k2 = hip.apply_rope(q, k, cos.cuda(), sin.cuda(), pos)
+    qr, kr = reference.apply_rope(q.float().cpu(), k.float().cpu(), cos, sin, pos.cpu())
+    assert torch.allclose(q2.float().cpu(), qr, atol=2e-2, rtol=2e-2)
+    assert torch.allclose(k2.float().cpu(), kr, atol=2e-2, rtol=2e-2)
+
+
+@requires_gpu
+@pytest.mark.parametrize("batch", [1, 2, 4])
+@pytest.mark.parametrize("shape", [(512, 1024), (4096, 4096), (1000, 1536)])
+def test_gemv_plain(hip, batch, shape):
+    torch.manual_seed(1)
+    in_dim, out_dim = shape
+    wt = (torch.randn(in_dim, out_dim, device="cuda") * 0.05).to(torch.bfloat16)
+    x = torch.randn(batch, in_dim, device="cuda")
+    ws = torch.empty(0, device="cuda")
+    y = hip.gemv_bf16(wt, x, ws, None, 0)  # plain f32
+    ref = x @ wt.float()
+    assert torch.allclose(y, ref, atol=1e-2, rtol=1e-2), (y - ref).abs().max()
+
+
+@requires_gpu
+def test_gemv_residual(hip):
+    torch.manual_seed(2)
+    wt = (torch.randn(2048, 1024, device="cuda") * 0.05).to(torch.bfloat16)
+    x = torch.randn(2, 2048, device="cuda")
+    res = torch.randn(2, 1024, device="cuda").to(torch.bfloat16)
+    ws = torch.empty(0, device="cuda")
+    y = hip.gemv_bf16(wt, x, ws, res, 2)  # residual bf16
+    ref = res.float() + x @ wt.float()
+    assert torch.allclose(y.float(), ref, atol=5e-2, rtol=5e-2)
+
+
+@requires_gpu
+def test_gemv_swiglu(hip):
+    torch.manual_seed(3)
+    wt = (torch.randn(1024, 2 * 768, device="cuda") * 0.05).to(torch.bfloat16)
+    x = torch.randn(1, 1024, device="cuda")
+    ws = torch.empty(0, device="cuda")
+    y = hip.gemv_bf16(wt, x, ws, None, 3)  # swiglu f32
+    full = x @ wt.float()
+    g, u = full[:, :768], full[:, 768:]
+    ref = torch.nn.functional.silu(g) * u
+    assert torch.allclose(y, ref, atol=2e-2, rtol=2e-2)
+
+
+@requires_gpu
+@pytest.mark.parametrize("gq,kv_heads,kv_len", [(8, 8, 500), (1, 32, 77), (4, 8, 1)])
+def test_attn_decode(hip, gq, kv_heads, kv_len):
+    from petals_amd.ops import reference
+
+    torch.manual_seed(4)
+    b, hd, lmax = 2, 128, 640
+    q = torch.randn(b, kv_heads * gq * hd, device="cuda")
+    k_cache = torch.zeros(b, kv_heads, lmax, hd, device="cuda", dtype=torch.bfloat16)
+    v_cache = torch.zeros_like(k_cache)
+    k_cache[:, :, :kv_len] = (torch.randn(b, kv_heads, kv_len, hd, device="cuda") * 0.5).to(torch.bfloat16)
+    v_cache[:, :, :kv_len] = (torch.randn(b, kv_heads, kv_len, hd, device="cuda") * 0.5).to(torch.bfloat16)
+    kv_len_t = torch.tensor([kv_len], dtype=torch.int32, device="cuda")
+    empty = torch.empty(0, device="cuda")
+    out = hip.attn_decode_fused(q, k_cache, v_cache, kv_len_t, gq, 0, empty, empty, 1.0 / math.sqrt(hd))
+
+    q_ref = q.view(b, kv_heads * gq, 1, hd).float().cpu()
+    ref = reference.attention(
+        q_ref,
+        k_cache[:, :, :kv_len].float().cpu(),
+        v_cache[:, :, :kv_len].float().cpu(),
+        causal=False,
+    )
+    ref = ref.view(b, -1)
+    assert torch.allclose(out.cpu(), ref, atol=2e-2, rtol=2e-2), (out.cpu() - ref).abs().max()
+
+
+@requires_gpu
+def test_llama_block_fast_decode_matches_cpu(hip):
+    """Full fused decode path vs the fp32 CPU block with the same weights."""
+    from petals_amd.models import get_model_block
+    from petals_amd.models.config_base import load_model_config
+    from petals_amd.server.from_pretrained import init_random_block_
+
+    cfg = load_model_config("test-llama")
+    cfg.hidden_size, cfg.num_attention_heads, cfg.num_key_value_heads, cfg.intermediate_size = 512, 4, 2, 1024
+    blk_cpu = get_model_block(cfg, 0)
+    init_random_block_(blk_cpu, cfg, 0)
+    blk_cpu = blk_cpu.float().eval()
+
+    blk_gpu = get_model_block(cfg, 0)
+    blk_gpu.load_state_dict(blk_cpu.state_dict())
+    blk_gpu = blk_gpu.to("cuda", torch.bfloat16).eval().optimize_for_inference()
+
+    torch.manual_seed(5)
+    B, S = 2, 9
+    x = torch.randn(B, S, 512) * 0.5
+    ks, vs = blk_cpu.kv_cache_shape(B, 32)
+    kc, vc = torch.zeros(ks), torch.zeros(vs)
+    kg = torch.zeros(ks, device="cuda", dtype=torch.bfloat16)
+    vg = torch.zeros(vs, device="cuda", dtype=torch.bfloat16)
+
+    # prefill 6 tokens (fast prefill path), then decode 3 (fused kernels)
+    y_cpu, y_gpu = [], []
+    y_cpu.append(blk_cpu(x[:, :6], kv_cache=(kc, vc), prefix_length=0))
+    y_gpu.append(blk_gpu(x[:, :6].cuda().bfloat16(), kv_cache=(kg, vg), prefix_length=0))
+    for t in range(6, S):
+        y_cpu.append(blk_cpu(x[:, t : t + 1], kv_cache=(kc, vc), prefix_length=t))
+        y_gpu.append(blk_gpu(x[:, t : t + 1].cuda().bfloat16(), kv_cache=(kg, vg), prefix_length=t))
+    ref = torch.cat(y_cpu, 1)
+    out = torch.cat([y.float().cpu() for y in y_gpu], 1)
+    assert torch.allclose(out, ref, atol=0.05, rtol=0.05), (out - ref).abs().max()
+
+
+@requires_gpu
+def test_gpu_server_e2e_generate():
+    """Tiny llama swarm: 1 GPU server (bf16, fused path) vs local CPU fp32."""
+    import tempfile
+
+    from petals_amd.dht.node import DHT
+    from petals_amd.server.server import Server
+    from petals_amd.utils.auto_config import AutoDistributedModelForCausalLM
+
+    boot = DHT(host="127.0.0.1")
+    server = Server(
+        "test-llama",
+        initial_peers=[boot.listen_addr],
+        host="127.0.0.1",
+        device="cuda",
+        torch_dtype="bfloat16",
+        block_indices="0:4",
+        dht_prefix="gpu-e2e",
+        throughput=1.0,
+    ).start()
+    try:
+        model = AutoDistributedModelForCausalLM.from_pretrained(
+            "test-llama", initial_peers=[boot.listen_addr], dht_prefix="gpu-e2e",
+            show_route=False, max_retries=1,
+        )
+        ids = torch.randint(0, 128, (1, 5))
+        out = model.generate(ids, max_new_tokens=8, do_sample=False)
+        assert out.shape == (1, 13)
+        model.transformer.h.sequence_manager.shutdown()
+    finally:
+        server.shutdown()
+        boot.shutdown()
