@@ -45,10 +45,9 @@ def main():
 
     import torch.distributed as dist
 
-    from petals_amd.models import get_model_block
     from petals_amd.models.config_base import load_model_config
     from petals_amd.parallel.pipeline import PipelineStage, init_process_group_from_env, split_blocks
-    from petals_amd.server.from_pretrained import init_random_block_
+    from petals_amd.server.from_pretrained import build_empty_block, init_random_block_
 
     rank, world = init_process_group_from_env()
     assert world == args.gpus or args.gpus == 1, f"WORLD_SIZE={world} but --gpus={args.gpus}"
@@ -77,9 +76,9 @@ def main():
     t0 = time.time()
     blocks = []
     for i in my_span:
-        blk = get_model_block(config, i)
+        blk = build_empty_block(config, i, device, dtype)
         init_random_block_(blk, config, i)
-        blk = blk.to(device=device, dtype=dtype).eval()
+        blk = blk.eval()
         if use_cuda and hasattr(blk, "optimize_for_inference"):
             blk.optimize_for_inference()
         blocks.append(blk)

@@ -25,7 +25,7 @@ class BloomAttention(nn.Module):
         self.hidden_size = config.hidden_size
         self.query_key_value = nn.Linear(self.hidden_size, 3 * self.hidden_size, bias=True)
         self.dense = nn.Linear(self.hidden_size, self.hidden_size, bias=True)
-        self.register_buffer("alibi_slopes", ops.build_alibi_slopes(self.num_heads), persistent=False)
+        self._alibi_slopes = None  # lazy: blocks may be built on the meta device
 
     def forward(self, hidden_states, kv_cache=None, prefix_length: int = 0):
         b, q_len, _ = hidden_states.shape
@@ -35,7 +35,9 @@ class BloomAttention(nn.Module):
         v = fused[..., 2, :].transpose(1, 2)
 
         end = prefix_length + q_len
-        slopes = self.alibi_slopes.to(hidden_states.device)
+        if self._alibi_slopes is None or self._alibi_slopes.device != hidden_states.device:
+            self._alibi_slopes = ops.build_alibi_slopes(self.num_heads).to(hidden_states.device)
+        slopes = self._alibi_slopes
         if kv_cache is not None:
             k_cache, v_cache = kv_cache
             k_cache[:b, :, prefix_length:end].copy_(k)

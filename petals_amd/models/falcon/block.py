@@ -36,9 +36,8 @@ class FalconAttention(nn.Module):
             qkv_out = 3 * h
         self.query_key_value = nn.Linear(h, qkv_out, bias=config.bias)
         self.dense = nn.Linear(self.num_heads * self.head_dim, h, bias=config.bias)
-        cos, sin = ops.build_rope_cache(self.head_dim, config.max_position_embeddings, theta=config.rope_theta)
-        self.register_buffer("rope_cos", cos, persistent=False)
-        self.register_buffer("rope_sin", sin, persistent=False)
+        self.rope_cos = None  # lazy (meta-device construction)
+        self.rope_sin = None
 
     def _split_qkv(self, fused: torch.Tensor, b: int, q_len: int):
         if self.config.new_decoder_architecture:
@@ -57,11 +56,12 @@ class FalconAttention(nn.Module):
         return (x.transpose(1, 2) for x in (q, k, v))  # [b, heads, len, hd]
 
     def _ensure_rope(self, needed: int, device):
-        if self.rope_cos.shape[0] < needed:
-            cos, sin = ops.build_rope_cache(self.head_dim, max(needed, 2 * self.rope_cos.shape[0]), theta=self.config.rope_theta)
+        if self.rope_cos is None or self.rope_cos.shape[0] < needed or self.rope_cos.device != torch.device(device):
+            prev = 0 if self.rope_cos is None else self.rope_cos.shape[0]
+            cos, sin = ops.build_rope_cache(
+                self.head_dim, max(needed, 2 * prev, self.config.max_position_embeddings), theta=self.config.rope_theta
+            )
             self.rope_cos, self.rope_sin = cos.to(device), sin.to(device)
-        elif self.rope_cos.device != device:
-            self.rope_cos, self.rope_sin = self.rope_cos.to(device), self.rope_sin.to(device)
 
     def forward(self, hidden_states, kv_cache=None, prefix_length: int = 0):
         b, q_len, _ = hidden_states.shape

@@ -44,28 +44,23 @@ class LlamaAttention(nn.Module):
         self.v_proj = nn.Linear(self.hidden_size, self.num_kv_heads * self.head_dim, bias=bias)
         self.o_proj = nn.Linear(self.num_heads * self.head_dim, self.hidden_size, bias=bias)
 
-        cos, sin = ops.build_rope_cache(
-            self.head_dim,
-            config.max_position_embeddings,
-            theta=config.rope_theta,
-            rope_scaling=config.rope_scaling,
-        )
-        self.register_buffer("rope_cos", cos, persistent=False)
-        self.register_buffer("rope_sin", sin, persistent=False)
+        # rope tables are plain (non-module) attributes built lazily on first
+        # use: blocks are constructed on the meta device (cheap) and
+        # materialized straight on the GPU
+        self.rope_cos = None
+        self.rope_sin = None
 
     def _ensure_rope(self, needed_len: int, device, ref: torch.Tensor):
-        if self.rope_cos.shape[0] < needed_len:
+        if self.rope_cos is None or self.rope_cos.shape[0] < needed_len or self.rope_cos.device != torch.device(device):
+            prev = 0 if self.rope_cos is None else self.rope_cos.shape[0]
             cos, sin = ops.build_rope_cache(
                 self.head_dim,
-                max(needed_len, 2 * self.rope_cos.shape[0]),
+                max(needed_len, 2 * prev, self.config.max_position_embeddings),
                 theta=self.config.rope_theta,
                 rope_scaling=self.config.rope_scaling,
             )
             self.rope_cos = cos.to(device)
             self.rope_sin = sin.to(device)
-        elif self.rope_cos.device != device:
-            self.rope_cos = self.rope_cos.to(device)
-            self.rope_sin = self.rope_sin.to(device)
 
     def forward(
         self,

@@ -25,8 +25,10 @@ def init_process_group_from_env() -> tuple[int, int]:
     if world > 1 and not dist.is_initialized():
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "29571")
-        dist.init_process_group(backend="nccl", rank=rank, world_size=world)
-        torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", rank)))
+        backend = "nccl" if torch.cuda.is_available() else "gloo"  # nccl == RCCL on ROCm
+        dist.init_process_group(backend=backend, rank=rank, world_size=world)
+        if torch.cuda.is_available():
+            torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", rank)))
     return rank, world
 
 

@@ -64,6 +64,14 @@ def load_block_state_dict(model_dir: str, config: ModelConfig, block_index: int)
     return state_dict
 
 
+def build_empty_block(config: ModelConfig, block_index: int, device, torch_dtype) -> torch.nn.Module:
+    """Construct a block on the meta device (no allocation / default-init cost)
+    and materialize it empty straight on the target device."""
+    with torch.device("meta"):
+        block = get_model_block(config, block_index)
+    return block.to_empty(device=device).to(torch_dtype)
+
+
 def load_pretrained_block(
     model_name_or_dir: str,
     config: ModelConfig,
@@ -74,28 +82,31 @@ def load_pretrained_block(
 ) -> torch.nn.Module:
     """Build one block and fill it with checkpoint weights (or deterministic
     random weights when no local checkpoint exists)."""
-    block = get_model_block(config, block_index)
+    device = torch.device(device)
+    block = build_empty_block(config, block_index, device, torch_dtype)
     if os.path.isdir(model_name_or_dir):
         sd = load_block_state_dict(model_name_or_dir, config, block_index)
-        report = block.load_state_dict(sd, strict=False)
+        sd = {k: v.to(device=device, dtype=torch_dtype if v.is_floating_point() else None) for k, v in sd.items()}
+        report = block.load_state_dict(sd, strict=False, assign=True)
         if report.missing_keys:
             logger.warning("block %d: missing keys %s", block_index, report.missing_keys)
     else:
         init_random_block_(block, config, block_index)
-    block = block.to(device=device, dtype=torch_dtype).eval()
-    device = torch.device(device)
+    block = block.eval()
     if device.type == "cuda" and hasattr(block, "optimize_for_inference"):
         block.optimize_for_inference()
     return block
 
 
 def init_random_block_(block: torch.nn.Module, config: ModelConfig, block_index: int) -> None:
-    """Deterministic random init: same (model, block) => same weights everywhere."""
+    """Deterministic random init: same (model, block) => same weights on every
+    server *of the same device type* (CPU and GPU RNG sequences differ)."""
     import zlib
 
     key = f"{config.name_or_path or config.model_type}:{block_index}"
     seed = (zlib.crc32(key.encode()) & 0x7FFFFFFF) or 1
-    gen = torch.Generator().manual_seed(seed)
+    device = next(block.parameters()).device
+    gen = torch.Generator(device=device).manual_seed(seed)
     with torch.no_grad():
         for name, p in sorted(block.named_parameters()):
             if p.dim() >= 2:

@@ -99,9 +99,9 @@ def measure_compute_rps(
     forward_tokens: int = 1024,
 ) -> Dict[str, float]:
     device = torch.device(device)
-    block = get_model_block(config, 0)
+    block = build_empty_block(config, 0, device, dtype)
     init_random_block_(block, config, 0)
-    block = block.to(device=device, dtype=dtype).eval()
+    block = block.eval()
 
     k_shape, v_shape = block.kv_cache_shape(1, 1024)
     k = torch.zeros(k_shape, device=device, dtype=dtype)
