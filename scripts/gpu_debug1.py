@@ -1,0 +1,104 @@
+"""Debug: attn_decode NaN at (gq=4, kv_len=1) + kernel-level microbenches."""
+
+import math
+import time
+
+import torch
+
+from petals_amd import ops
+
+hip = ops._load_hip_ops()
+assert hip is not None
+
+
+def attn_debug():
+    b, kv_heads, hd, lmax, kv_len = 2, 8, 128, 640, 1
+    kc = torch.zeros(b, kv_heads, lmax, hd, device="cuda", dtype=torch.bfloat16)
+    vc = torch.zeros_like(kc)
+    kc[:, :, :kv_len] = 0.5
+    vc[:, :, :kv_len] = 0.25
+    empty = torch.empty(0, device="cuda")
+    kv_len_t = torch.tensor([kv_len], dtype=torch.int32, device="cuda")
+    for gq in (1, 2, 4, 8):
+        q = torch.randn(b, kv_heads * gq * hd, device="cuda")
+        for splits in (1, 2, 3, 0):
+            out = hip.attn_decode_fused(q, kc, vc, kv_len_t, gq, splits, empty, empty, 1 / math.sqrt(hd))
+            print(f"gq={gq} splits={splits}: nan={torch.isnan(out).any().item()} "
+                  f"out[0,:3]={out[0,:3].tolist()} expect~0.25", flush=True)
+
+
+def bench_kernel(fn, n=50):
+    fn()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(n):
+        fn()
+    torch.cuda.synchronize()
+    return (time.perf_counter() - t0) / n
+
+
+def gemv_bench():
+    ws = torch.empty(64 * 57344, dtype=torch.float32, device="cuda")
+    for in_dim, out_dim, name in ((8192, 10240, "qkv"), (8192, 8192, "o"), (8192, 57344, "gateup"), (28672, 8192, "down"), (8192, 32000, "head")):
+        wt = (torch.randn(in_dim, out_dim, device="cuda") * 0.02).to(torch.bfloat16)
+        x = torch.randn(1, in_dim, device="cuda")
+        t = bench_kernel(lambda: hip.gemv_bf16(wt, x, ws, None, 0))
+        gb = in_dim * out_dim * 2 / 1e9
+        print(f"gemv {name} [{in_dim},{out_dim}]: {t*1e6:.1f} us, {gb/t:.0f} GB/s", flush=True)
+
+
+def rmsnorm_bench():
+    x = torch.randn(1, 8192, device="cuda").to(torch.bfloat16)
+    w = torch.ones(8192, device="cuda", dtype=torch.bfloat16)
+    t = bench_kernel(lambda: hip.rms_norm_f32out(x, w, 1e-5))
+    print(f"rms_norm_f32out [1,8192]: {t*1e6:.1f} us", flush=True)
+
+
+def attn_bench():
+    b, kv_heads, gq, hd, lmax = 1, 8, 8, 128, 4096
+    kc = (torch.randn(b, kv_heads, lmax, hd, device="cuda") * 0.3).to(torch.bfloat16)
+    vc = kc.clone()
+    q = torch.randn(b, kv_heads * gq * hd, device="cuda")
+    empty = torch.empty(0, device="cuda")
+    for kv_len in (128, 1024, 4000):
+        kv_len_t = torch.tensor([kv_len], dtype=torch.int32, device="cuda")
+        t = bench_kernel(lambda: hip.attn_decode_fused(q, kc, vc, kv_len_t, gq, 0, empty, empty, 1 / math.sqrt(hd)))
+        gb = 2 * kv_heads * kv_len * hd * 2 / 1e9
+        print(f"attn_decode kv_len={kv_len}: {t*1e6:.1f} us, {gb/t:.0f} GB/s", flush=True)
+
+
+def layer_decode_bench():
+    """One full fused 70B layer decode step, timed."""
+    from petals_amd.models.config_base import load_model_config
+    from petals_amd.server.from_pretrained import build_empty_block, init_random_block_
+
+    config = load_model_config("llama-2-70b")
+    blk = build_empty_block(config, 0, "cuda", torch.bfloat16)
+    init_random_block_(blk, config, 0)
+    blk = blk.eval().optimize_for_inference()
+    ks, vs = blk.kv_cache_shape(1, 256)
+    k = torch.zeros(ks, device="cuda", dtype=torch.bfloat16)
+    v = torch.zeros(vs, device="cuda", dtype=torch.bfloat16)
+    h = torch.randn(1, 1, config.hidden_size, device="cuda", dtype=torch.bfloat16)
+    with torch.inference_mode():
+        t = bench_kernel(lambda: blk(h, kv_cache=(k, v), prefix_length=100), n=100)
+    gb = 1.69
+    print(f"70B layer decode: {t*1e6:.1f} us ({gb/t:.0f} GB/s vs 1.69 GB weights; ideal ~280us)", flush=True)
+
+    # python-overhead check: measure without sync inside loop but calling python chain
+    with torch.inference_mode():
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(100):
+            blk(h, kv_cache=(k, v), prefix_length=100)
+        wall_nosync = (time.perf_counter() - t0) / 100
+        torch.cuda.synchronize()
+    print(f"70B layer decode (no per-call sync): {wall_nosync*1e6:.1f} us", flush=True)
+
+
+if __name__ == "__main__":
+    attn_debug()
+    rmsnorm_bench()
+    gemv_bench()
+    attn_bench()
+    layer_decode_bench()
