@@ -1,5 +1,7 @@
 """Debug: attn_decode NaN at (gq=4, kv_len=1) + kernel-level microbenches."""
 
+import os, sys
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import math
 import time
 
