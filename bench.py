@@ -130,7 +130,7 @@ def main():
             return h  # unused on non-zero ranks
 
     def one_token(token_ids, prefix: int):
-        """Generate the next token given current ids [B,1]."""
+        """Generate the next token given current ids [B,1] (eager path)."""
         if rank == 0:
             h = embed[token_ids.view(-1)].view(B, 1, H)
         else:
@@ -140,6 +140,67 @@ def main():
             logits = head_logits(out[:, -1, :])
             return logits.argmax(dim=-1, keepdim=True)
         return token_ids
+
+    # ---- hipGraph decode: capture each rank's whole span step once; replay
+    # per token with only the device position advancing (utils/graphs.py)
+    graph_state = {}
+
+    def build_graphs(cur_id):
+        from petals_amd.ops.fused_decode import DecodeContext
+        from petals_amd.utils.graphs import GraphedCallable
+
+        ctx = DecodeContext(device)
+        ctx.set_position(args.prompt_len)
+        gs = {"ctx": ctx}
+        if rank == 0:
+            gs["cur_id"] = cur_id.clone()
+
+            def span_fn():
+                h = embed.index_select(0, gs["cur_id"].view(-1)).view(B, 1, H)
+                for blk, (k, v) in zip(blocks, kv_caches):
+                    h = blk(h, kv_cache=(k, v), ctx=ctx)
+                return h
+
+            gs["g_span"] = GraphedCallable(span_fn, [])
+            if world == 1:
+                def head_fn():
+                    logits = head_logits(gs["g_span"].static_outputs[0][:, -1, :])
+                    gs["cur_id"].copy_(logits.argmax(dim=-1, keepdim=True))
+                    return logits
+                gs["g_head"] = GraphedCallable(head_fn, [])
+            else:
+                gs["h_back"] = torch.empty(B, 1, H, device=device, dtype=dtype)
+
+                def head_fn():
+                    logits = head_logits(gs["h_back"][:, -1, :])
+                    gs["cur_id"].copy_(logits.argmax(dim=-1, keepdim=True))
+                    return logits
+                gs["g_head"] = GraphedCallable(head_fn, [])
+        else:
+            gs["h_in"] = torch.empty(B, 1, H, device=device, dtype=dtype)
+
+            def span_fn():
+                h = gs["h_in"]
+                for blk, (k, v) in zip(blocks, kv_caches):
+                    h = blk(h, kv_cache=(k, v), ctx=ctx)
+                return h
+
+            gs["g_span"] = GraphedCallable(span_fn, [])
+        return gs
+
+    def one_token_graphed(prefix: int):
+        gs = graph_state
+        gs["ctx"].set_position(prefix)
+        if rank == 0:
+            h = gs["g_span"].replay()
+            if world > 1:
+                stage.send(h)
+                dist.recv(gs["h_back"], src=stage.prev_rank)
+            gs["g_head"].replay()
+        else:
+            dist.recv(gs["h_in"], src=stage.prev_rank)
+            gs["g_span"].replay()
+            stage.send(gs["g_span"].static_outputs[0])
 
     # --- prefill
     prefix = 0
@@ -154,9 +215,19 @@ def main():
     else:
         cur = torch.zeros(B, 1, dtype=torch.long, device=device)
 
+    # --- capture decode graphs (GPU) or stay eager (CPU)
+    use_graphs = use_cuda and not os.environ.get("PETALS_AMD_NO_GRAPHS")
+    if use_graphs:
+        graph_state.update(build_graphs(cur if rank == 0 else torch.zeros(B, 1, dtype=torch.long, device=device)))
+        step_fn = one_token_graphed
+    else:
+        def step_fn(p):
+            nonlocal cur
+            cur = one_token(cur, p)
+
     # --- warmup decode
     for _ in range(args.warmup):
-        cur = one_token(cur, prefix)
+        step_fn(prefix)
         prefix += 1
 
     # --- timed region
@@ -166,7 +237,7 @@ def main():
         torch.cuda.synchronize(device)
     t_start = time.perf_counter()
     for _ in range(args.steps):
-        cur = one_token(cur, prefix)
+        step_fn(prefix)
         prefix += 1
     if use_cuda:
         torch.cuda.synchronize(device)

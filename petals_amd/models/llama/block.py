@@ -148,13 +148,14 @@ class LlamaBlock(nn.Module):
         hidden_states: torch.Tensor,
         kv_cache: Optional[Tuple[torch.Tensor, torch.Tensor]] = None,
         prefix_length: int = 0,
+        ctx=None,  # ops.fused_decode.DecodeContext (device-resident position)
     ) -> torch.Tensor:
         if self._fast is not None:
             if torch.is_grad_enabled() and hidden_states.requires_grad:
                 assert kv_cache is None, "training forward does not use the KV cache"
                 return self._fast.forward_autograd(hidden_states, prefix_length)
             if kv_cache is not None and hidden_states.shape[1] == 1 and hidden_states.shape[0] <= 8:
-                return self._fast.decode_step(hidden_states, kv_cache[0], kv_cache[1], prefix_length)
+                return self._fast.decode_step(hidden_states, kv_cache[0], kv_cache[1], prefix_length, ctx=ctx)
             return self._fast.forward(hidden_states, kv_cache, prefix_length)
 
         residual = hidden_states

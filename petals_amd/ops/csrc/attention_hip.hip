@@ -54,7 +54,7 @@ __global__ __launch_bounds__(64) void attn_decode_kernel(
   float m[GQ], l[GQ], o[GQ][EPL];
 #pragma unroll
   for (int g = 0; g < GQ; ++g) {
-    m[g] = -INFINITY;
+    m[g] = NEG_SENTINEL;
     l[g] = 0.f;
 #pragma unroll
     for (int e = 0; e < EPL; ++e) o[g][e] = 0.f;
@@ -134,7 +134,7 @@ __global__ __launch_bounds__(64) void attn_decode_kernel(
     float osum = 0.f, lsum = 0.f;
 #pragma unroll
     for (int r = 0; r < GROUPS; ++r) {
-      const float w = (c_m[r][g] == -INFINITY) ? 0.f : __expf(c_m[r][g] - m_star);
+      const float w = (c_m[r][g] <= NEG_THRESHOLD) ? 0.f : __expf(c_m[r][g] - m_star);
       osum += w * c_o[r][g][d];
       lsum += w * c_l[r][g];
     }
@@ -153,7 +153,7 @@ __global__ void attn_decode_combine_kernel(
     int n_splits, int gq, int hd) {
   const int bkv = blockIdx.x;
   const int g = blockIdx.y;
-  float m_star = -INFINITY;
+  float m_star = NEG_SENTINEL;
   for (int s = 0; s < n_splits; ++s)
     m_star = fmaxf(m_star, part_ml[(((size_t)bkv * n_splits + s) * gq + g) * 2]);
   for (int d = threadIdx.x; d < hd; d += blockDim.x) {
@@ -161,7 +161,7 @@ __global__ void attn_decode_combine_kernel(
     for (int s = 0; s < n_splits; ++s) {
       const float ms = part_ml[(((size_t)bkv * n_splits + s) * gq + g) * 2];
       const float ls = part_ml[(((size_t)bkv * n_splits + s) * gq + g) * 2 + 1];
-      if (ms == -INFINITY) continue;
+      if (ms <= NEG_THRESHOLD) continue;
       const float w = __expf(ms - m_star);
       osum += w * part_o[(((size_t)bkv * n_splits + s) * gq + g) * hd + d];
       lsum += w * ls;

@@ -50,3 +50,8 @@ __device__ __forceinline__ float group16_reduce_sum(float x) {
       TORCH_CHECK(false, "HIP kernel launch failed: ", hipGetErrorString(e)); \
     }                                                                        \
   } while (0)
+
+// finite stand-ins for -inf: the extension compiles with -ffast-math, under
+// which IEEE inf comparisons/arithmetic are not reliable
+#define NEG_SENTINEL  (-1e30f)
+#define NEG_THRESHOLD (-1e29f)

@@ -48,14 +48,41 @@ __global__ void gemv_bf16_kernel(
 
   const bool full = (out0 + 8) <= out_dim;
   if (full) {
-    for (int i = i_begin; i < i_end; ++i) {
+    // UNROLL x 16B weight loads in flight per wave: HBM latency (~900 cyc)
+    // needs many outstanding loads; the scalar x loads are wave-uniform
+    constexpr int UNROLL = 8;
+    const unsigned short* wp = wt + (size_t)i_begin * out_dim + out0;
+    int i = i_begin;
+    for (; i + UNROLL <= i_end; i += UNROLL) {
+      short8 w8[UNROLL];
+#pragma unroll
+      for (int u = 0; u < UNROLL; ++u)
+        w8[u] = *reinterpret_cast<const short8*>(wp + (size_t)u * out_dim);
+      float xs[BATCH][UNROLL];
+#pragma unroll
+      for (int b = 0; b < BATCH; ++b)
+#pragma unroll
+        for (int u = 0; u < UNROLL; ++u) xs[b][u] = x[(size_t)b * in_dim + i + u];
+#pragma unroll
+      for (int u = 0; u < UNROLL; ++u) {
+        float wf[8];
+#pragma unroll
+        for (int v = 0; v < 8; ++v) wf[v] = bf16_to_f32((unsigned short)w8[u][v]);
+#pragma unroll
+        for (int b = 0; b < BATCH; ++b)
+#pragma unroll
+          for (int v = 0; v < 8; ++v) acc[b][v] = fmaf(wf[v], xs[b][u], acc[b][v]);
+      }
+      wp += (size_t)UNROLL * out_dim;
+    }
+    for (; i < i_end; ++i) {
       const short8 w8 = *reinterpret_cast<const short8*>(wt + (size_t)i * out_dim + out0);
       float wf[8];
 #pragma unroll
       for (int v = 0; v < 8; ++v) wf[v] = bf16_to_f32((unsigned short)w8[v]);
 #pragma unroll
       for (int b = 0; b < BATCH; ++b) {
-        const float xv = x[(size_t)b * in_dim + i];  // wave-uniform -> s_load
+        const float xv = x[(size_t)b * in_dim + i];
 #pragma unroll
         for (int v = 0; v < 8; ++v) acc[b][v] = fmaf(wf[v], xv, acc[b][v]);
       }
@@ -135,10 +162,11 @@ __global__ void gemv_reduce_kernel(
 
 static int pick_splits(int in_dim, int out_dim) {
   const long out_waves = (out_dim + GEMV_OUT_PER_WAVE - 1) / GEMV_OUT_PER_WAVE;
-  // target ~1024 workgroups so 256 CUs have >=4 each; round splits to a
-  // divisor-ish value of in_dim chunks of >=128 rows
-  long splits = (1024 + out_waves - 1) / out_waves;
-  long max_splits = (in_dim + 127) / 128;
+  // target ~4096 single-wave workgroups (16/CU) so HBM latency is hidden by
+  // wave-level parallelism; chunks of >=64 input rows keep the partial-reduce
+  // traffic negligible vs the weight read
+  long splits = (4096 + out_waves - 1) / out_waves;
+  long max_splits = (in_dim + 63) / 64;
   if (splits > max_splits) splits = max_splits;
   if (splits < 1) splits = 1;
   return (int)splits;

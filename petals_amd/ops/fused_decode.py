@@ -45,6 +45,27 @@ def _get_ws(device: torch.device, key: str, numel: int) -> torch.Tensor:
     return ws
 
 
+class DecodeContext:
+    """Device-resident decode position shared by all blocks of a step.
+
+    Makes the whole token step hipGraph-capturable: kernels read the position
+    from device memory, so a captured graph replays correctly as `advance()`
+    (or `set_position`) moves the position — no host-side re-capture."""
+
+    def __init__(self, device: torch.device):
+        self.pos = torch.zeros(1, dtype=torch.int32, device=device)  # prefix length
+        self.kv_len = torch.zeros(1, dtype=torch.int32, device=device)  # prefix + 1
+
+    def set_position(self, prefix_length: int) -> None:
+        self.pos.fill_(prefix_length)
+        self.kv_len.fill_(prefix_length + 1)
+
+    def advance(self) -> None:
+        """On-device increment (graph-capturable)."""
+        self.pos.add_(1)
+        self.kv_len.add_(1)
+
+
 class LlamaFastPath:
     def __init__(self, block, hip_ops):
         cfg = block.config
@@ -105,7 +126,8 @@ class LlamaFastPath:
         hidden: torch.Tensor,  # [B, 1, H] bf16
         k_cache: torch.Tensor,  # [Bc, KV, Lmax, hd] bf16
         v_cache: torch.Tensor,
-        prefix_length: int,
+        prefix_length: int = -1,
+        ctx: Optional[DecodeContext] = None,
     ) -> torch.Tensor:
         B = hidden.shape[0]
         H = hidden.shape[-1]
@@ -113,19 +135,27 @@ class LlamaFastPath:
         if h.dtype != torch.bfloat16:
             h = h.to(torch.bfloat16)
         h = h.contiguous()
-        self._ensure_rope(prefix_length + 1)
-        self._pos.fill_(prefix_length)
-        self._kv_len.fill_(prefix_length + 1)
+        if ctx is None:
+            assert prefix_length >= 0
+            self._ensure_rope(prefix_length + 1)
+            self._pos.fill_(prefix_length)
+            self._kv_len.fill_(prefix_length + 1)
+            pos, kv_len = self._pos, self._kv_len
+        else:
+            # position lives on device (graph-capturable); rope table must
+            # already cover the cache (ensure_rope(lmax) at session start)
+            self._ensure_rope(k_cache.shape[2])
+            pos, kv_len = ctx.pos, ctx.kv_len
         ws = _get_ws(self.device, "gemv", 64 * B * max(self.wgateup_t.shape[1], self.wqkv_t.shape[1], H))
 
         xn = self.hip.rms_norm_f32out(h, self.ln1_w, self.eps)  # [B, H] f32
         qkv = self.hip.gemv_bf16(self.wqkv_t, xn, ws, None, _EPI_PLAIN_F32)  # [B, qkv] f32
         self.hip.rope_cache_write(
-            qkv, self.rope_cos, self.rope_sin, self._pos, k_cache[:B], v_cache[:B], self.qh, self.kh
+            qkv, self.rope_cos, self.rope_sin, pos, k_cache[:B], v_cache[:B], self.qh, self.kh
         )
         q = qkv[:, : self.qh * self.hd]
         attn = self.hip.attn_decode_fused(
-            q.contiguous(), k_cache[:B], v_cache[:B], self._kv_len, self.gq, 0,
+            q.contiguous(), k_cache[:B], v_cache[:B], kv_len, self.gq, 0,
             self._empty_f32, self._empty_f32, self.scale,
         )  # [B, H] f32
         h2 = self.hip.gemv_bf16(self.wo_t, attn, ws, h, _EPI_RESIDUAL_BF16)  # [B, H] bf16
