@@ -12,7 +12,7 @@ void rope_cache_write(
     torch::Tensor k_cache, torch::Tensor v_cache, int64_t qh, int64_t kh);
 torch::Tensor gemv_bf16(
     torch::Tensor wt, torch::Tensor x, torch::Tensor workspace,
-    c10::optional<torch::Tensor> residual, int64_t epilogue);
+    c10::optional<torch::Tensor> residual, int64_t epilogue, int64_t splits_override);
 torch::Tensor attn_decode_fused(
     torch::Tensor q, torch::Tensor k_cache, torch::Tensor v_cache, torch::Tensor kv_len,
     int64_t gq, int64_t n_splits, torch::Tensor part_o, torch::Tensor part_ml, double scale);
@@ -23,6 +23,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("swiglu", &swiglu, "silu(gate) * up (bf16)");
   m.def("apply_rope", &apply_rope, "rotate q,k by positions (bf16)");
   m.def("rope_cache_write", &rope_cache_write, "fused decode rope + kv cache write");
-  m.def("gemv_bf16", &gemv_bf16, "split-K bf16 gemv with fused epilogue");
+  m.def("gemv_bf16", &gemv_bf16, "split-K bf16 gemv with fused epilogue",
+        py::arg("wt"), py::arg("x"), py::arg("workspace"), py::arg("residual"), py::arg("epilogue"), py::arg("splits") = 0);
   m.def("attn_decode_fused", &attn_decode_fused, "GQA decode attention (flash-decoding)");
 }

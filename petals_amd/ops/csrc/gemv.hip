@@ -176,7 +176,8 @@ torch::Tensor gemv_bf16(
     torch::Tensor x,         // [batch, in] f32
     torch::Tensor workspace, // [n_splits_max, batch, out] f32 (preallocated, may be empty)
     c10::optional<torch::Tensor> residual,  // [batch, out] bf16
-    int64_t epilogue) {
+    int64_t epilogue,
+    int64_t splits_override) {
   TORCH_CHECK(wt.is_cuda() && wt.dtype() == torch::kBFloat16 && wt.dim() == 2);
   TORCH_CHECK(x.is_cuda() && x.dtype() == torch::kFloat32 && x.dim() == 2);
   const int in_dim = wt.size(0), out_dim = wt.size(1);
@@ -184,7 +185,7 @@ torch::Tensor gemv_bf16(
   const int batch = x.size(0);
   TORCH_CHECK(batch <= 8, "decode gemv supports batch <= 8");
 
-  const int splits = pick_splits(in_dim, out_dim);
+  const int splits = splits_override > 0 ? (int)splits_override : pick_splits(in_dim, out_dim);
   const int i_per_split = (in_dim + splits - 1) / splits;
 
   torch::Tensor partials;

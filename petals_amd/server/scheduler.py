@@ -39,7 +39,11 @@ class TaskPrioritizer:
 
 class PriorityRuntime:
     def __init__(self, device: Optional[torch.device] = None, sync_every: bool = True):
-        self.device = torch.device(device) if device is not None else None
+        if device is not None:
+            device = torch.device(device)
+            if device.type == "cuda" and device.index is None:
+                device = torch.device("cuda", torch.cuda.current_device() if torch.cuda.is_available() else 0)
+        self.device = device
         self._queue: list = []
         self._cv = threading.Condition()
         self._counter = itertools.count()
