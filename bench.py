@@ -41,6 +41,7 @@ def main():
     p.add_argument("--prompt-len", type=int, default=128)
     p.add_argument("--batch", type=int, default=1)
     p.add_argument("--device", default="cuda")
+    p.add_argument("--quant", default="none", choices=["none", "nf4"])
     args = p.parse_args()
 
     import torch.distributed as dist
@@ -80,7 +81,7 @@ def main():
         init_random_block_(blk, config, i)
         blk = blk.eval()
         if use_cuda and hasattr(blk, "optimize_for_inference"):
-            blk.optimize_for_inference()
+            blk.optimize_for_inference(quant=args.quant)
         blocks.append(blk)
     kv_caches = []
     for blk in blocks:
@@ -266,7 +267,7 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": tokens_per_s / baseline if args.model == "llama-2-70b" else None,
-            "dtype": "bf16" if use_cuda else "fp32",
+            "dtype": ("bf16" if args.quant == "none" else "nf4-weights/bf16-compute") if use_cuda else "fp32",
             "data": "synthetic prompt, random-init weights (no network)",
             "config": {
                 "model": args.model,
@@ -274,6 +275,7 @@ def main():
                 "seq_len": args.prompt_len + args.warmup + args.steps,
                 "prompt_len": args.prompt_len,
                 "parallelism": f"pp{world}" if world > 1 else "single",
+                "quant": args.quant,
             },
         }
         print(json.dumps(result), flush=True)

@@ -127,9 +127,10 @@ class LlamaBlock(nn.Module):
 
     _fast = None  # LlamaFastPath after optimize_for_inference()
 
-    def optimize_for_inference(self) -> "LlamaBlock":
-        """Repack weights into the MI355X kernel layout and enable the fused
-        decode path. Requires the HIP extension; frees nn.Linear weights."""
+    def optimize_for_inference(self, quant: str = "none") -> "LlamaBlock":
+        """Repack weights into the MI355X kernel layout (optionally NF4
+        quantize-on-load) and enable the fused decode path. Requires the HIP
+        extension; frees nn.Linear weights."""
         from petals_amd import ops as _ops
         from petals_amd.ops.fused_decode import LlamaFastPath
 
@@ -139,8 +140,21 @@ class LlamaBlock(nn.Module):
                 f"cannot optimize block for MI355X: HIP extension missing ({_ops._hip_import_error!r})"
             )
         assert next(self.parameters()).device.type == "cuda", "optimize_for_inference needs a GPU block"
-        assert not self.config.attention_bias and not self.config.mlp_bias, "fast path assumes no biases"
-        self._fast = LlamaFastPath(self, hip)
+        gq = self.config.num_attention_heads // self.config.n_kv_heads
+        if (
+            self.config.head_dim not in (64, 128)
+            or gq not in (1, 2, 4, 6, 8, 16)
+            or self.config.attention_bias
+            or getattr(self.config, "mlp_bias", False)
+        ):
+            import logging
+
+            logging.getLogger(__name__).warning(
+                "block geometry (head_dim=%s, gq=%s, bias=%s) outside the fused fast path; "
+                "serving via generic HIP ops", self.config.head_dim, gq, self.config.attention_bias,
+            )
+            return self
+        self._fast = LlamaFastPath(self, hip, quant=quant)
         return self
 
     def forward(
@@ -154,7 +168,8 @@ class LlamaBlock(nn.Module):
             if torch.is_grad_enabled() and hidden_states.requires_grad:
                 assert kv_cache is None, "training forward does not use the KV cache"
                 return self._fast.forward_autograd(hidden_states, prefix_length)
-            if kv_cache is not None and hidden_states.shape[1] == 1 and hidden_states.shape[0] <= 8:
+            max_b = 4 if self._fast.quant == "nf4" else 8
+            if kv_cache is not None and hidden_states.shape[1] == 1 and hidden_states.shape[0] <= max_b:
                 return self._fast.decode_step(hidden_states, kv_cache[0], kv_cache[1], prefix_length, ctx=ctx)
             return self._fast.forward(hidden_states, kv_cache, prefix_length)
 

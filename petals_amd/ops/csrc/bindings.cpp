@@ -16,6 +16,11 @@ torch::Tensor gemv_bf16(
 torch::Tensor attn_decode_fused(
     torch::Tensor q, torch::Tensor k_cache, torch::Tensor v_cache, torch::Tensor kv_len,
     int64_t gq, int64_t n_splits, torch::Tensor part_o, torch::Tensor part_ml, double scale);
+std::vector<torch::Tensor> nf4_quantize(torch::Tensor w);
+torch::Tensor nf4_dequantize(torch::Tensor packed, torch::Tensor absmax);
+torch::Tensor gemv_nf4(
+    torch::Tensor packed, torch::Tensor absmax, torch::Tensor x, torch::Tensor workspace,
+    c10::optional<torch::Tensor> residual, int64_t epilogue, int64_t splits_override);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rms_norm", &rms_norm, "RMSNorm (bf16 -> bf16)");
@@ -26,4 +31,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemv_bf16", &gemv_bf16, "split-K bf16 gemv with fused epilogue",
         py::arg("wt"), py::arg("x"), py::arg("workspace"), py::arg("residual"), py::arg("epilogue"), py::arg("splits") = 0);
   m.def("attn_decode_fused", &attn_decode_fused, "GQA decode attention (flash-decoding)");
+  m.def("nf4_quantize", &nf4_quantize, "blockwise NF4 quantize [in,out] bf16 -> (packed u8, absmax bf16)");
+  m.def("nf4_dequantize", &nf4_dequantize, "NF4 -> bf16");
+  m.def("gemv_nf4", &gemv_nf4, "split-K NF4 gemv (dequant fused) with epilogue",
+        py::arg("packed"), py::arg("absmax"), py::arg("x"), py::arg("workspace"),
+        py::arg("residual"), py::arg("epilogue"), py::arg("splits") = 0);
 }

@@ -15,6 +15,7 @@
 // ("Element-wise"/"GEMM" vectorization rules; G13).
 
 #include "common.h"
+#include "gemv_reduce.h"
 #include <torch/extension.h>
 #include <ATen/cuda/CUDAContext.h>
 
@@ -110,62 +111,15 @@ __global__ void gemv_bf16_kernel(
   }
 }
 
-// ------------------------------------------------------------- reduce + epi
-
-enum GemvEpilogue : int {
-  EPI_PLAIN_F32 = 0,     // y_f32[b, out] = sum
-  EPI_PLAIN_BF16 = 1,    // y_bf16[b, out] = sum
-  EPI_RESIDUAL_BF16 = 2, // y_bf16 = residual_bf16 + sum
-  EPI_SWIGLU_F32 = 3,    // out = silu(sum[:half]) * sum[half:], y_f32[b, half]
-};
-
-__global__ void gemv_reduce_kernel(
-    const float* __restrict__ partials,  // [n_splits, batch, out]
-    const unsigned short* __restrict__ residual,  // [batch, out] or null
-    void* __restrict__ y,
-    int n_splits,
-    int batch,
-    int out_dim,
-    int epilogue) {
-  const int half = out_dim >> 1;
-  const int n_out = (epilogue == EPI_SWIGLU_F32) ? half : out_dim;
-  const int total = batch * n_out;
-  for (int idx = blockIdx.x * blockDim.x + threadIdx.x; idx < total; idx += gridDim.x * blockDim.x) {
-    const int b = idx / n_out;
-    const int o = idx - b * n_out;
-    if (epilogue == EPI_SWIGLU_F32) {
-      float g = 0.f, u = 0.f;
-      for (int s = 0; s < n_splits; ++s) {
-        const float* base = partials + ((size_t)s * batch + b) * out_dim;
-        g += base[o];
-        u += base[o + half];
-      }
-      const float act = g / (1.f + __expf(-g)) * u;
-      reinterpret_cast<float*>(y)[(size_t)b * half + o] = act;
-    } else {
-      float sum = 0.f;
-      for (int s = 0; s < n_splits; ++s) sum += partials[((size_t)s * batch + b) * out_dim + o];
-      if (epilogue == EPI_PLAIN_F32) {
-        reinterpret_cast<float*>(y)[(size_t)b * out_dim + o] = sum;
-      } else if (epilogue == EPI_PLAIN_BF16) {
-        reinterpret_cast<unsigned short*>(y)[(size_t)b * out_dim + o] = f32_to_bf16(sum);
-      } else {  // EPI_RESIDUAL_BF16
-        const float r = bf16_to_f32(residual[(size_t)b * out_dim + o]);
-        reinterpret_cast<unsigned short*>(y)[(size_t)b * out_dim + o] = f32_to_bf16(r + sum);
-      }
-    }
-  }
-}
-
 // ------------------------------------------------------------------- host
 
 static int pick_splits(int in_dim, int out_dim) {
   const long out_waves = (out_dim + GEMV_OUT_PER_WAVE - 1) / GEMV_OUT_PER_WAVE;
-  // target ~4096 single-wave workgroups (16/CU) so HBM latency is hidden by
-  // wave-level parallelism; chunks of >=64 input rows keep the partial-reduce
-  // traffic negligible vs the weight read
-  long splits = (4096 + out_waves - 1) / out_waves;
-  long max_splits = (in_dim + 63) / 64;
+  // measured knee (profiles/gemv_sweep): ~768 single-wave workgroups (3/CU)
+  // with input chunks of >=256 rows; more splits shrink chunks below the
+  // latency-amortization point, fewer leave CUs idle
+  long splits = (768 + out_waves - 1) / out_waves;
+  long max_splits = (in_dim + 255) / 256;
   if (splits > max_splits) splits = max_splits;
   if (splits < 1) splits = 1;
   return (int)splits;
@@ -220,27 +174,8 @@ torch::Tensor gemv_bf16(
 #undef LAUNCH_GEMV
   HIP_CHECK_LAST();
 
-  // epilogue
-  const int half = out_dim / 2;
-  const int n_out = (epilogue == EPI_SWIGLU_F32) ? half : out_dim;
-  torch::Tensor y;
-  if (epilogue == EPI_PLAIN_F32) {
-    y = torch::empty({batch, out_dim}, x.options());
-  } else if (epilogue == EPI_SWIGLU_F32) {
-    y = torch::empty({batch, half}, x.options());
-  } else {
-    y = torch::empty({batch, out_dim}, wt.options());
-  }
-  const unsigned short* res_p = nullptr;
-  if (epilogue == EPI_RESIDUAL_BF16) {
-    TORCH_CHECK(residual.has_value(), "residual required");
-    TORCH_CHECK(residual->dtype() == torch::kBFloat16 && residual->is_contiguous());
-    res_p = reinterpret_cast<const unsigned short*>(residual->data_ptr());
-  }
-  const int total = batch * n_out;
-  int rblocks = std::min((total + 255) / 256, 2048);
-  gemv_reduce_kernel<<<rblocks, 256, 0, stream>>>(
-      part_p, res_p, y.data_ptr(), splits, batch, out_dim, (int)epilogue);
-  HIP_CHECK_LAST();
+  // epilogue (shared with the NF4 path)
+  torch::Tensor y = launch_gemv_reduce(
+      partials, residual, splits, batch, out_dim, (int)epilogue, x.options(), wt.options());
   return y;
 }

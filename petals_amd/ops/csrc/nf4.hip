@@ -1,0 +1,276 @@
+// NF4 4-bit blockwise quantization for CDNA4 (gfx950).
+//
+// Format (our own serving format; same 4.25 bits/param footprint as the
+// reference's bitsandbytes LinearNF4 — blocksize 64, reference
+// utils/convert_block.py:76-111, block_utils.py:46 — but with bf16 absmax
+// instead of double-quantized int8 absmax: same size, better accuracy):
+//   packed  uint8 [in, out/2]   two nibbles per byte (even elem = low nibble)
+//   absmax  bf16  [in, out/64]  per-64-element scale along the OUT dim
+//
+// Weights stay TRANSPOSED [in, out] like the bf16 path, so the same
+// x-broadcast GEMV structure applies; each lane owns 16 consecutive outputs
+// (8 B packed per input row) and dequantizes in-register through an
+// LDS-resident 16-entry NF4 LUT.
+
+#include "common.h"
+#include "gemv_reduce.h"
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+
+__device__ __constant__ float NF4_LUT_C[16] = {
+    -1.0f, -0.6961928009986877f, -0.5250730514526367f, -0.39491748809814453f,
+    -0.28444138169288635f, -0.18477343022823334f, -0.09105003625154495f, 0.0f,
+    0.07958029955625534f, 0.16093020141124725f, 0.24611230194568634f,
+    0.33791524171829224f, 0.4407098293304443f, 0.5626170039176941f,
+    0.7229568362236023f, 1.0f};
+
+// ------------------------------------------------------------- quantization
+
+// one thread per 64-element block along OUT
+__global__ void nf4_quantize_kernel(
+    const unsigned short* __restrict__ w,  // [in, out] bf16
+    unsigned char* __restrict__ packed,    // [in, out/2]
+    unsigned short* __restrict__ absmax,   // [in, out/64] bf16
+    long n_blocks,
+    int out_dim) {
+  const long blk = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (blk >= n_blocks) return;
+  const long base = blk * 64;  // element offset (rows are multiples of 64 wide)
+  float vals[64];
+  float amax = 0.f;
+#pragma unroll
+  for (int j = 0; j < 64; ++j) {
+    vals[j] = bf16_to_f32(w[base + j]);
+    amax = fmaxf(amax, fabsf(vals[j]));
+  }
+  const float amax_bf = bf16_to_f32(f32_to_bf16(amax));  // store-rounded scale
+  absmax[blk] = f32_to_bf16(amax);
+  const float inv = amax_bf > 0.f ? 1.0f / amax_bf : 0.f;
+#pragma unroll
+  for (int j = 0; j < 32; ++j) {
+    unsigned char lo, hi;
+#pragma unroll
+    for (int half = 0; half < 2; ++half) {
+      const float v = vals[2 * j + half] * inv;
+      // nearest NF4 level via linear scan (16 levels)
+      int best = 0;
+      float bd = fabsf(v - NF4_LUT_C[0]);
+#pragma unroll
+      for (int l = 1; l < 16; ++l) {
+        const float d = fabsf(v - NF4_LUT_C[l]);
+        if (d < bd) { bd = d; best = l; }
+      }
+      if (half == 0) lo = (unsigned char)best; else hi = (unsigned char)best;
+    }
+    packed[blk * 32 + j] = (unsigned char)(lo | (hi << 4));
+  }
+}
+
+// --------------------------------------------------------------- dequantize
+
+__global__ void nf4_dequant_kernel(
+    const unsigned char* __restrict__ packed,
+    const unsigned short* __restrict__ absmax,
+    unsigned short* __restrict__ out,  // bf16
+    long n_elems) {
+  const long i = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 2;
+  if (i >= n_elems) return;
+  const unsigned char byte = packed[i >> 1];
+  const float scale = bf16_to_f32(absmax[i >> 6]);
+  out[i] = f32_to_bf16(NF4_LUT_C[byte & 0xF] * scale);
+  if (i + 1 < n_elems) out[i + 1] = f32_to_bf16(NF4_LUT_C[byte >> 4] * scale);
+}
+
+// -------------------------------------------------------------- NF4 gemv
+
+#define NF4_OUT_PER_WAVE 1024  // 64 lanes x 16 outputs (8 B packed / row)
+
+template <int BATCH>
+__global__ void gemv_nf4_kernel(
+    const unsigned char* __restrict__ packed,   // [in, out/2]
+    const unsigned short* __restrict__ absmax,  // [in, out/64]
+    const float* __restrict__ x,                // [BATCH, in]
+    float* __restrict__ partials,               // [n_splits, BATCH, out]
+    int in_dim,
+    int out_dim,
+    int i_per_split) {
+  __shared__ float lut[16];
+  if (threadIdx.x < 16) lut[threadIdx.x] = NF4_LUT_C[threadIdx.x];
+  __syncthreads();
+
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int out0 = blockIdx.x * NF4_OUT_PER_WAVE + lane * 16;
+  if (out0 >= out_dim) return;
+  const int split = blockIdx.y;
+  const int i_begin = split * i_per_split;
+  const int i_end = min(i_begin + i_per_split, in_dim);
+  const bool full = (out0 + 16) <= out_dim;
+
+  float acc[BATCH][16];
+#pragma unroll
+  for (int b = 0; b < BATCH; ++b)
+#pragma unroll
+    for (int v = 0; v < 16; ++v) acc[b][v] = 0.f;
+
+  if (full) {
+    constexpr int UNROLL = 8;
+    const int half_out = out_dim >> 1;
+    const unsigned char* pp = packed + (size_t)i_begin * half_out + (out0 >> 1);
+    int i = i_begin;
+    for (; i + UNROLL <= i_end; i += UNROLL) {
+      uint2 pk[UNROLL];  // 8 bytes = 16 nibbles
+#pragma unroll
+      for (int u = 0; u < UNROLL; ++u)
+        pk[u] = *reinterpret_cast<const uint2*>(pp + (size_t)u * half_out);
+      float am[UNROLL];
+#pragma unroll
+      for (int u = 0; u < UNROLL; ++u)
+        am[u] = bf16_to_f32(absmax[(size_t)(i + u) * (out_dim >> 6) + (out0 >> 6)]);
+      float xs[BATCH][UNROLL];
+#pragma unroll
+      for (int b = 0; b < BATCH; ++b)
+#pragma unroll
+        for (int u = 0; u < UNROLL; ++u) xs[b][u] = x[(size_t)b * in_dim + i + u];
+#pragma unroll
+      for (int u = 0; u < UNROLL; ++u) {
+        float wf[16];
+        const unsigned int w0 = pk[u].x, w1 = pk[u].y;
+#pragma unroll
+        for (int v = 0; v < 8; ++v) wf[v] = lut[(w0 >> (4 * v)) & 0xF];
+#pragma unroll
+        for (int v = 0; v < 8; ++v) wf[8 + v] = lut[(w1 >> (4 * v)) & 0xF];
+#pragma unroll
+        for (int b = 0; b < BATCH; ++b) {
+          const float xa = xs[b][u] * am[u];
+#pragma unroll
+          for (int v = 0; v < 16; ++v) acc[b][v] = fmaf(wf[v], xa, acc[b][v]);
+        }
+      }
+      pp += (size_t)UNROLL * half_out;
+    }
+    for (; i < i_end; ++i) {
+      const uint2 pk = *reinterpret_cast<const uint2*>(packed + (size_t)i * half_out + (out0 >> 1));
+      const float am = bf16_to_f32(absmax[(size_t)i * (out_dim >> 6) + (out0 >> 6)]);
+#pragma unroll
+      for (int b = 0; b < BATCH; ++b) {
+        const float xa = x[(size_t)b * in_dim + i] * am;
+#pragma unroll
+        for (int v = 0; v < 8; ++v) acc[b][v] = fmaf(lut[(pk.x >> (4 * v)) & 0xF], xa, acc[b][v]);
+#pragma unroll
+        for (int v = 0; v < 8; ++v) acc[b][8 + v] = fmaf(lut[(pk.y >> (4 * v)) & 0xF], xa, acc[b][8 + v]);
+      }
+    }
+  } else {
+    for (int i = i_begin; i < i_end; ++i) {
+      const float am = bf16_to_f32(absmax[(size_t)i * (out_dim >> 6) + (out0 >> 6)]);
+      for (int b = 0; b < BATCH; ++b) {
+        const float xa = x[(size_t)b * in_dim + i] * am;
+        for (int v = 0; v < out_dim - out0; ++v) {
+          const unsigned char byte = packed[(size_t)i * (out_dim >> 1) + ((out0 + v) >> 1)];
+          const int nib = ((out0 + v) & 1) ? (byte >> 4) : (byte & 0xF);
+          acc[b][v] = fmaf(lut[nib], xa, acc[b][v]);
+        }
+      }
+    }
+  }
+
+#pragma unroll
+  for (int b = 0; b < BATCH; ++b) {
+    float* dst = partials + ((size_t)split * BATCH + b) * out_dim + out0;
+    if (full) {
+#pragma unroll
+      for (int q = 0; q < 4; ++q)
+        reinterpret_cast<float4v*>(dst)[q] =
+            float4v{acc[b][4 * q], acc[b][4 * q + 1], acc[b][4 * q + 2], acc[b][4 * q + 3]};
+    } else {
+      for (int v = 0; v < out_dim - out0; ++v) dst[v] = acc[b][v];
+    }
+  }
+}
+
+// ------------------------------------------------------------------- host
+
+std::vector<torch::Tensor> nf4_quantize(torch::Tensor w) {
+  TORCH_CHECK(w.is_cuda() && w.dtype() == torch::kBFloat16 && w.dim() == 2);
+  TORCH_CHECK(w.size(1) % 64 == 0, "out dim must be a multiple of 64");
+  auto wc = w.contiguous();
+  const long in_dim = wc.size(0), out_dim = wc.size(1);
+  const long n_blocks = in_dim * out_dim / 64;
+  auto packed = torch::empty({in_dim, out_dim / 2}, w.options().dtype(torch::kUInt8));
+  auto absmax = torch::empty({in_dim, out_dim / 64}, w.options());
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const long blocks = (n_blocks + 255) / 256;
+  nf4_quantize_kernel<<<blocks, 256, 0, stream>>>(
+      reinterpret_cast<const unsigned short*>(wc.data_ptr()),
+      packed.data_ptr<unsigned char>(),
+      reinterpret_cast<unsigned short*>(absmax.data_ptr()),
+      n_blocks, (int)out_dim);
+  HIP_CHECK_LAST();
+  return {packed, absmax};
+}
+
+torch::Tensor nf4_dequantize(torch::Tensor packed, torch::Tensor absmax) {
+  TORCH_CHECK(packed.is_cuda() && packed.dtype() == torch::kUInt8 && packed.dim() == 2);
+  const long in_dim = packed.size(0), half_out = packed.size(1);
+  const long n = in_dim * half_out * 2;
+  auto out = torch::empty({in_dim, half_out * 2}, absmax.options());
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const long blocks = (n / 2 + 255) / 256;
+  nf4_dequant_kernel<<<blocks, 256, 0, stream>>>(
+      packed.data_ptr<unsigned char>(),
+      reinterpret_cast<const unsigned short*>(absmax.data_ptr()),
+      reinterpret_cast<unsigned short*>(out.data_ptr()), n);
+  HIP_CHECK_LAST();
+  return out;
+}
+
+torch::Tensor gemv_nf4(
+    torch::Tensor packed,    // [in, out/2] u8
+    torch::Tensor absmax,    // [in, out/64] bf16
+    torch::Tensor x,         // [batch, in] f32
+    torch::Tensor workspace,
+    c10::optional<torch::Tensor> residual,
+    int64_t epilogue,
+    int64_t splits_override) {
+  TORCH_CHECK(packed.is_cuda() && packed.dtype() == torch::kUInt8);
+  TORCH_CHECK(x.dtype() == torch::kFloat32 && x.dim() == 2);
+  const int in_dim = packed.size(0);
+  const int out_dim = packed.size(1) * 2;
+  const int batch = x.size(0);
+  TORCH_CHECK(x.size(1) == in_dim && batch <= 4);
+
+  const long out_waves = (out_dim + NF4_OUT_PER_WAVE - 1) / NF4_OUT_PER_WAVE;
+  long splits = splits_override > 0 ? splits_override : (768 + out_waves - 1) / out_waves;
+  long max_splits = (in_dim + 255) / 256;
+  if (splits > max_splits) splits = max_splits;
+  if (splits < 1) splits = 1;
+
+  torch::Tensor partials;
+  auto f32opts = x.options();
+  if (workspace.numel() >= (int64_t)splits * batch * out_dim) {
+    partials = workspace;
+  } else {
+    partials = torch::empty({(int64_t)splits, batch, out_dim}, f32opts);
+  }
+  const int i_per_split = (in_dim + splits - 1) / splits;
+  dim3 grid(out_waves, splits);
+  auto stream = at::cuda::getCurrentCUDAStream();
+
+#define LAUNCH_NF4(B)                                                         \
+  gemv_nf4_kernel<B><<<grid, WAVE, 0, stream>>>(                              \
+      packed.data_ptr<unsigned char>(),                                       \
+      reinterpret_cast<const unsigned short*>(absmax.data_ptr()),             \
+      x.data_ptr<float>(), partials.data_ptr<float>(), in_dim, out_dim, i_per_split)
+  switch (batch) {
+    case 1: LAUNCH_NF4(1); break;
+    case 2: LAUNCH_NF4(2); break;
+    case 3: LAUNCH_NF4(3); break;
+    case 4: LAUNCH_NF4(4); break;
+  }
+#undef LAUNCH_NF4
+  HIP_CHECK_LAST();
+
+  torch::Tensor y = launch_gemv_reduce(
+      partials, residual, (int)splits, batch, out_dim, (int)epilogue, f32opts, absmax.options());
+  return y;
+}

@@ -66,11 +66,42 @@ class DecodeContext:
         self.kv_len.add_(1)
 
 
+class _FastWeight:
+    """A transposed [in, out] weight in bf16 or NF4 (quantize-on-load)."""
+
+    def __init__(self, t_bf16: torch.Tensor, hip, quant: str):
+        self.hip = hip
+        self.quant = quant
+        self.in_dim, self.out_dim = t_bf16.shape
+        if quant == "nf4":
+            self.packed, self.absmax = hip.nf4_quantize(t_bf16.contiguous())
+            self.t = None
+        else:
+            self.t = t_bf16.contiguous()
+            self.packed = self.absmax = None
+
+    @property
+    def shape(self):
+        return (self.in_dim, self.out_dim)
+
+    def gemv(self, x, ws, residual, epilogue):
+        if self.quant == "nf4":
+            return self.hip.gemv_nf4(self.packed, self.absmax, x, ws, residual, epilogue)
+        return self.hip.gemv_bf16(self.t, x, ws, residual, epilogue)
+
+    def dense(self) -> torch.Tensor:
+        """bf16 [in, out] view for prefill GEMMs (dequantized on the fly for NF4)."""
+        if self.quant == "nf4":
+            return self.hip.nf4_dequantize(self.packed, self.absmax)
+        return self.t
+
+
 class LlamaFastPath:
-    def __init__(self, block, hip_ops):
+    def __init__(self, block, hip_ops, quant: str = "none"):
         cfg = block.config
         self.hip = hip_ops
         self.cfg = cfg
+        self.quant = quant
         self.hd = cfg.head_dim
         self.qh = cfg.num_attention_heads
         self.kh = cfg.n_kv_heads
@@ -84,13 +115,16 @@ class LlamaFastPath:
         def t(w):
             return w.detach().to(torch.bfloat16).t().contiguous()
 
-        self.wqkv_t = torch.cat(
-            [t(attn.q_proj.weight), t(attn.k_proj.weight), t(attn.v_proj.weight)], dim=1
+        self.wqkv_t = _FastWeight(
+            torch.cat([t(attn.q_proj.weight), t(attn.k_proj.weight), t(attn.v_proj.weight)], dim=1),
+            hip_ops, quant,
         )  # [H, qh*hd + 2*kh*hd]
-        self.wo_t = t(attn.o_proj.weight)
+        self.wo_t = _FastWeight(t(attn.o_proj.weight), hip_ops, quant)
         mlp = block.mlp
-        self.wgateup_t = torch.cat([t(mlp.gate_proj.weight), t(mlp.up_proj.weight)], dim=1)
-        self.wdown_t = t(mlp.down_proj.weight)
+        self.wgateup_t = _FastWeight(
+            torch.cat([t(mlp.gate_proj.weight), t(mlp.up_proj.weight)], dim=1), hip_ops, quant
+        )
+        self.wdown_t = _FastWeight(t(mlp.down_proj.weight), hip_ops, quant)
         self.ln1_w = block.input_layernorm.weight.detach().to(torch.bfloat16).contiguous()
         self.ln2_w = block.post_attention_layernorm.weight.detach().to(torch.bfloat16).contiguous()
 
@@ -149,7 +183,7 @@ class LlamaFastPath:
         ws = _get_ws(self.device, "gemv", 64 * B * max(self.wgateup_t.shape[1], self.wqkv_t.shape[1], H))
 
         xn = self.hip.rms_norm_f32out(h, self.ln1_w, self.eps)  # [B, H] f32
-        qkv = self.hip.gemv_bf16(self.wqkv_t, xn, ws, None, _EPI_PLAIN_F32)  # [B, qkv] f32
+        qkv = self.wqkv_t.gemv(xn, ws, None, _EPI_PLAIN_F32)  # [B, qkv] f32
         self.hip.rope_cache_write(
             qkv, self.rope_cos, self.rope_sin, pos, k_cache[:B], v_cache[:B], self.qh, self.kh
         )
@@ -158,10 +192,10 @@ class LlamaFastPath:
             q.contiguous(), k_cache[:B], v_cache[:B], kv_len, self.gq, 0,
             self._empty_f32, self._empty_f32, self.scale,
         )  # [B, H] f32
-        h2 = self.hip.gemv_bf16(self.wo_t, attn, ws, h, _EPI_RESIDUAL_BF16)  # [B, H] bf16
+        h2 = self.wo_t.gemv(attn, ws, h, _EPI_RESIDUAL_BF16)  # [B, H] bf16
         xn2 = self.hip.rms_norm_f32out(h2, self.ln2_w, self.eps)
-        act = self.hip.gemv_bf16(self.wgateup_t, xn2, ws, None, _EPI_SWIGLU_F32)  # [B, I] f32
-        h3 = self.hip.gemv_bf16(self.wdown_t, act, ws, h2, _EPI_RESIDUAL_BF16)
+        act = self.wgateup_t.gemv(xn2, ws, None, _EPI_SWIGLU_F32)  # [B, I] f32
+        h3 = self.wdown_t.gemv(act, ws, h2, _EPI_RESIDUAL_BF16)
         return h3.view(B, 1, H)
 
     # --------------------------------------------------- training (autograd)
@@ -174,7 +208,7 @@ class LlamaFastPath:
         end = prefix_length + S
         self._ensure_rope(end)
         xn = reference.rms_norm(hidden, self.ln1_w, self.eps)
-        qkv = torch.matmul(xn, self.wqkv_t)
+        qkv = torch.matmul(xn, self.wqkv_t.dense())
         q = qkv[..., : self.qh * self.hd].view(B, S, self.qh, self.hd).transpose(1, 2)
         k = qkv[..., self.qh * self.hd : (self.qh + self.kh) * self.hd].view(B, S, self.kh, self.hd).transpose(1, 2)
         v = qkv[..., (self.qh + self.kh) * self.hd :].view(B, S, self.kh, self.hd).transpose(1, 2)
@@ -182,12 +216,12 @@ class LlamaFastPath:
         q, k = reference.apply_rope(q, k, self.rope_cos, self.rope_sin, pos)
         attn = reference.attention(q, k, v, causal=True)
         attn = attn.transpose(1, 2).reshape(B, S, self.qh * self.hd).to(hidden.dtype)
-        h2 = hidden + torch.matmul(attn, self.wo_t)
+        h2 = hidden + torch.matmul(attn, self.wo_t.dense())
         xn2 = reference.rms_norm(h2, self.ln2_w, self.eps)
-        gateup = torch.matmul(xn2, self.wgateup_t)
+        gateup = torch.matmul(xn2, self.wgateup_t.dense())
         inter = self.wgateup_t.shape[1] // 2
         act = reference.swiglu(gateup[..., :inter], gateup[..., inter:]).to(hidden.dtype)
-        return h2 + torch.matmul(act, self.wdown_t)
+        return h2 + torch.matmul(act, self.wdown_t.dense())
 
     # ------------------------------------------------------------ prefill
 
@@ -203,7 +237,7 @@ class LlamaFastPath:
         self._ensure_rope(end)
 
         xn = self.hip.rms_norm(hidden, self.ln1_w, self.eps)
-        qkv = torch.matmul(xn, self.wqkv_t)  # [B, S, qkv] bf16 (rocBLAS)
+        qkv = torch.matmul(xn, self.wqkv_t.dense())  # [B, S, qkv] bf16 (rocBLAS)
         q = qkv[..., : self.qh * self.hd].view(B, S, self.qh, self.hd).transpose(1, 2)
         k = qkv[..., self.qh * self.hd : (self.qh + self.kh) * self.hd].view(B, S, self.kh, self.hd).transpose(1, 2)
         v = qkv[..., (self.qh + self.kh) * self.hd :].view(B, S, self.kh, self.hd).transpose(1, 2)
@@ -219,9 +253,9 @@ class LlamaFastPath:
         else:
             attn = reference.attention(q, k, v, causal=True)
         attn = attn.transpose(1, 2).reshape(B, S, self.qh * self.hd).to(torch.bfloat16)
-        h2 = hidden + torch.matmul(attn, self.wo_t)
+        h2 = hidden + torch.matmul(attn, self.wo_t.dense())
         xn2 = self.hip.rms_norm(h2, self.ln2_w, self.eps)
-        gateup = torch.matmul(xn2, self.wgateup_t)
+        gateup = torch.matmul(xn2, self.wgateup_t.dense())
         inter = self.wgateup_t.shape[1] // 2
         act = self.hip.swiglu(gateup[..., :inter].contiguous(), gateup[..., inter:].contiguous())
-        return h2 + torch.matmul(act, self.wdown_t)
+        return h2 + torch.matmul(act, self.wdown_t.dense())

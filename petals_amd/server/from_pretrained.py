@@ -79,6 +79,7 @@ def load_pretrained_block(
     *,
     torch_dtype: torch.dtype = torch.float32,
     device: torch.device = torch.device("cpu"),
+    quant_type: str = "none",
 ) -> torch.nn.Module:
     """Build one block and fill it with checkpoint weights (or deterministic
     random weights when no local checkpoint exists)."""
@@ -94,7 +95,7 @@ def load_pretrained_block(
         init_random_block_(block, config, block_index)
     block = block.eval()
     if device.type == "cuda" and hasattr(block, "optimize_for_inference"):
-        block.optimize_for_inference()
+        block.optimize_for_inference(quant=quant_type)
     return block
 
 

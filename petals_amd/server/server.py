@@ -242,7 +242,8 @@ class Server:
             block = await asyncio.get_event_loop().run_in_executor(
                 None,
                 lambda idx=i: load_pretrained_block(
-                    self.model_name_or_dir, self.config, idx, torch_dtype=self.torch_dtype, device=self.device
+                    self.model_name_or_dir, self.config, idx, torch_dtype=self.torch_dtype,
+                    device=self.device, quant_type=self.quant_type,
                 ),
             )
             self.backends[uid] = TransformerBackend(
