@@ -289,6 +289,7 @@ class TransformerConnectionHandler:
                         raise RpcError(
                             f"max_length exceeded: prefix {session.prefix_length} + {length_increment} > {max_length}"
                         )
+                    step_start_position = session.prefix_length
                     if length_increment > 0:
                         priority = self.prioritizer.prioritize(hidden_states, type="inference")
                         output = await self.runtime.submit(
@@ -310,7 +311,7 @@ class TransformerConnectionHandler:
                     next_servers = step_meta.get("next_servers")
                     pushed = False
                     if next_servers and not has_prompts and length_increment > 0:
-                        pushed = await self._push_outputs(output, step_meta, next_servers)
+                        pushed = await self._push_outputs(output, step_meta, next_servers, step_start_position)
                     if not pushed:
                         # the last server of a push chain (or any server when push
                         # is off/failed) returns outputs on its client stream
@@ -357,16 +358,20 @@ class TransformerConnectionHandler:
         session.pushed_inputs.put_nowait((dict(request.meta), list(request.tensors)))
         await stream.close(RpcMessage(meta={"ok": True}))
 
-    async def _push_outputs(self, output: torch.Tensor, step_meta: Dict[str, Any], next_servers) -> bool:
+    async def _push_outputs(
+        self, output: torch.Tensor, step_meta: Dict[str, Any], next_servers, step_start_position: int
+    ) -> bool:
         """Push this step's output into the next server's session. next_servers:
         [[host, port, session_id, start_block, end_block], ...]; we contact the
-        first entry."""
+        first entry. start_from_position propagates down the chain so rollbacks
+        (speculative decoding) rewind every span's cache."""
         try:
             host, port, next_session_id = next_servers[0][0], next_servers[0][1], next_servers[0][2]
             meta = {
                 "session_id": next_session_id,
                 "step_id": step_meta.get("step_id"),
                 "next_servers": next_servers[1:],
+                "start_from_position": step_start_position,
             }
             await asyncio.wait_for(
                 self.p2p.call_unary((host, port), "petals.rpc_push", RpcMessage(meta=meta, tensors=[output])),
