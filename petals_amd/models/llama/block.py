@@ -67,11 +67,21 @@ class LlamaAttention(nn.Module):
         hidden_states: torch.Tensor,
         kv_cache: Optional[Tuple[torch.Tensor, torch.Tensor]] = None,
         prefix_length: int = 0,
+        adapter=None,  # utils.peft.BlockAdapter
     ) -> torch.Tensor:
         b, q_len, _ = hidden_states.shape
-        q = self.q_proj(hidden_states).view(b, q_len, self.num_heads, self.head_dim).transpose(1, 2)
-        k = self.k_proj(hidden_states).view(b, q_len, self.num_kv_heads, self.head_dim).transpose(1, 2)
-        v = self.v_proj(hidden_states).view(b, q_len, self.num_kv_heads, self.head_dim).transpose(1, 2)
+
+        def proj(lin, key, heads):
+            y = lin(hidden_states)
+            if adapter is not None:
+                d = adapter.delta(key, hidden_states)
+                if d is not None:
+                    y = y + d
+            return y.view(b, q_len, heads, self.head_dim).transpose(1, 2)
+
+        q = proj(self.q_proj, "q", self.num_heads)
+        k = proj(self.k_proj, "k", self.num_kv_heads)
+        v = proj(self.v_proj, "v", self.num_kv_heads)
 
         end = prefix_length + q_len
         self._ensure_rope(end, hidden_states.device, hidden_states)
@@ -88,7 +98,12 @@ class LlamaAttention(nn.Module):
             attn = ops.attention(q, k, v, causal=True)
 
         attn = attn.transpose(1, 2).reshape(b, q_len, self.num_heads * self.head_dim)
-        return self.o_proj(attn)
+        out = self.o_proj(attn)
+        if adapter is not None:
+            d = adapter.delta("o", attn)
+            if d is not None:
+                out = out + d
+        return out
 
 
 class LlamaMLP(nn.Module):
@@ -99,8 +114,21 @@ class LlamaMLP(nn.Module):
         self.up_proj = nn.Linear(config.hidden_size, config.intermediate_size, bias=bias)
         self.down_proj = nn.Linear(config.intermediate_size, config.hidden_size, bias=bias)
 
-    def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return self.down_proj(ops.swiglu(self.gate_proj(x), self.up_proj(x)))
+    def forward(self, x: torch.Tensor, adapter=None) -> torch.Tensor:
+        gate, up = self.gate_proj(x), self.up_proj(x)
+        if adapter is not None:
+            dg, du = adapter.delta("gate", x), adapter.delta("up", x)
+            if dg is not None:
+                gate = gate + dg
+            if du is not None:
+                up = up + du
+        act = ops.swiglu(gate, up)
+        out = self.down_proj(act)
+        if adapter is not None:
+            dd = adapter.delta("down", act)
+            if dd is not None:
+                out = out + dd
+        return out
 
 
 class RMSNorm(nn.Module):
@@ -165,22 +193,32 @@ class LlamaBlock(nn.Module):
         ctx=None,  # ops.fused_decode.DecodeContext (device-resident position)
     ) -> torch.Tensor:
         if self._fast is not None:
+            from petals_amd.utils.peft import active_block_adapter
+
+            adapter = active_block_adapter(self)
             if torch.is_grad_enabled() and hidden_states.requires_grad:
                 assert kv_cache is None, "training forward does not use the KV cache"
-                return self._fast.forward_autograd(hidden_states, prefix_length)
+                return self._fast.forward_autograd(hidden_states, prefix_length, adapter=adapter)
             max_b = 4 if self._fast.quant == "nf4" else 8
             if kv_cache is not None and hidden_states.shape[1] == 1 and hidden_states.shape[0] <= max_b:
-                return self._fast.decode_step(hidden_states, kv_cache[0], kv_cache[1], prefix_length, ctx=ctx)
-            return self._fast.forward(hidden_states, kv_cache, prefix_length)
+                return self._fast.decode_step(
+                    hidden_states, kv_cache[0], kv_cache[1], prefix_length, ctx=ctx, adapter=adapter
+                )
+            return self._fast.forward(hidden_states, kv_cache, prefix_length, adapter=adapter)
 
+        from petals_amd.utils.peft import active_block_adapter
+
+        adapter = active_block_adapter(self)
         residual = hidden_states
         hidden_states = self.input_layernorm(hidden_states)
-        hidden_states = self.self_attn(hidden_states, kv_cache=kv_cache, prefix_length=prefix_length)
+        hidden_states = self.self_attn(
+            hidden_states, kv_cache=kv_cache, prefix_length=prefix_length, adapter=adapter
+        )
         hidden_states = residual + hidden_states
 
         residual = hidden_states
         hidden_states = self.post_attention_layernorm(hidden_states)
-        hidden_states = self.mlp(hidden_states)
+        hidden_states = self.mlp(hidden_states, adapter=adapter)
         return residual + hidden_states
 
     # --- cache geometry used by the server's MemoryCache ---

@@ -190,7 +190,9 @@ torch::Tensor attn_decode_fused(
   const int GQi = (int)gq;
   int n_splits = (int)n_splits_i;
   if (n_splits <= 0) {
-    n_splits = std::max(1, std::min(1024 / std::max(1, b * kv_heads), (lmax + 255) / 256));
+    // target ~1024 single-wave workgroups; each split should keep >=64 cache
+    // rows so the streaming loop amortizes setup
+    n_splits = std::max(1, std::min(1024 / std::max(1, b * kv_heads), (lmax + 63) / 64));
   }
   auto opts = q.options();
   if (part_o.numel() < (int64_t)b * kv_heads * n_splits * GQi * hd)

@@ -95,11 +95,16 @@ __global__ void gemv_nf4_kernel(
     int in_dim,
     int out_dim,
     int i_per_split) {
-  __shared__ float lut[16];
-  if (threadIdx.x < 16) lut[threadIdx.x] = NF4_LUT_C[threadIdx.x];
+  // LUT replicated across all 32 LDS banks: lane l always reads bank (l&31)
+  // -> conflict-free random-index lookups (naive 16-float LUT was ~4-way
+  // conflicted and made LDS the bottleneck)
+  __shared__ float lut[16 * 32];
+  for (int i = threadIdx.x; i < 16 * 32; i += blockDim.x) lut[i] = NF4_LUT_C[i >> 5];
   __syncthreads();
 
   const int lane = threadIdx.x & (WAVE - 1);
+  const float* lutb = lut + (lane & 31);
+#define NF4_L(nib) lutb[(nib) << 5]
   const int out0 = blockIdx.x * NF4_OUT_PER_WAVE + lane * 16;
   if (out0 >= out_dim) return;
   const int split = blockIdx.y;
@@ -137,9 +142,9 @@ __global__ void gemv_nf4_kernel(
         float wf[16];
         const unsigned int w0 = pk[u].x, w1 = pk[u].y;
 #pragma unroll
-        for (int v = 0; v < 8; ++v) wf[v] = lut[(w0 >> (4 * v)) & 0xF];
+        for (int v = 0; v < 8; ++v) wf[v] = NF4_L((w0 >> (4 * v)) & 0xF);
 #pragma unroll
-        for (int v = 0; v < 8; ++v) wf[8 + v] = lut[(w1 >> (4 * v)) & 0xF];
+        for (int v = 0; v < 8; ++v) wf[8 + v] = NF4_L((w1 >> (4 * v)) & 0xF);
 #pragma unroll
         for (int b = 0; b < BATCH; ++b) {
           const float xa = xs[b][u] * am[u];
@@ -156,9 +161,9 @@ __global__ void gemv_nf4_kernel(
       for (int b = 0; b < BATCH; ++b) {
         const float xa = x[(size_t)b * in_dim + i] * am;
 #pragma unroll
-        for (int v = 0; v < 8; ++v) acc[b][v] = fmaf(lut[(pk.x >> (4 * v)) & 0xF], xa, acc[b][v]);
+        for (int v = 0; v < 8; ++v) acc[b][v] = fmaf(NF4_L((pk.x >> (4 * v)) & 0xF), xa, acc[b][v]);
 #pragma unroll
-        for (int v = 0; v < 8; ++v) acc[b][8 + v] = fmaf(lut[(pk.y >> (4 * v)) & 0xF], xa, acc[b][8 + v]);
+        for (int v = 0; v < 8; ++v) acc[b][8 + v] = fmaf(NF4_L((pk.y >> (4 * v)) & 0xF), xa, acc[b][8 + v]);
       }
     }
   } else {
@@ -169,7 +174,7 @@ __global__ void gemv_nf4_kernel(
         for (int v = 0; v < out_dim - out0; ++v) {
           const unsigned char byte = packed[(size_t)i * (out_dim >> 1) + ((out0 + v) >> 1)];
           const int nib = ((out0 + v) & 1) ? (byte >> 4) : (byte & 0xF);
-          acc[b][v] = fmaf(lut[nib], xa, acc[b][v]);
+          acc[b][v] = fmaf(NF4_L(nib), xa, acc[b][v]);
         }
       }
     }

@@ -162,6 +162,7 @@ class LlamaFastPath:
         v_cache: torch.Tensor,
         prefix_length: int = -1,
         ctx: Optional[DecodeContext] = None,
+        adapter=None,
     ) -> torch.Tensor:
         B = hidden.shape[0]
         H = hidden.shape[-1]
@@ -184,6 +185,8 @@ class LlamaFastPath:
 
         xn = self.hip.rms_norm_f32out(h, self.ln1_w, self.eps)  # [B, H] f32
         qkv = self.wqkv_t.gemv(xn, ws, None, _EPI_PLAIN_F32)  # [B, qkv] f32
+        if adapter is not None:
+            self._apply_qkv_adapter(qkv, xn, adapter)
         self.hip.rope_cache_write(
             qkv, self.rope_cos, self.rope_sin, pos, k_cache[:B], v_cache[:B], self.qh, self.kh
         )
@@ -192,15 +195,54 @@ class LlamaFastPath:
             q.contiguous(), k_cache[:B], v_cache[:B], kv_len, self.gq, 0,
             self._empty_f32, self._empty_f32, self.scale,
         )  # [B, H] f32
-        h2 = self.wo_t.gemv(attn, ws, h, _EPI_RESIDUAL_BF16)  # [B, H] bf16
+        if adapter is not None:
+            d = adapter.delta("o", attn)
+            h2 = self.wo_t.gemv(attn, ws, h, _EPI_RESIDUAL_BF16)  # [B, H] bf16
+            if d is not None:
+                h2 = h2 + d.to(h2.dtype)
+        else:
+            h2 = self.wo_t.gemv(attn, ws, h, _EPI_RESIDUAL_BF16)  # [B, H] bf16
         xn2 = self.hip.rms_norm_f32out(h2, self.ln2_w, self.eps)
-        act = self.wgateup_t.gemv(xn2, ws, None, _EPI_SWIGLU_F32)  # [B, I] f32
-        h3 = self.wdown_t.gemv(act, ws, h2, _EPI_RESIDUAL_BF16)
+        if adapter is None:
+            act = self.wgateup_t.gemv(xn2, ws, None, _EPI_SWIGLU_F32)  # [B, I] f32
+            h3 = self.wdown_t.gemv(act, ws, h2, _EPI_RESIDUAL_BF16)
+        else:
+            gateup = self.wgateup_t.gemv(xn2, ws, None, _EPI_PLAIN_F32)  # [B, 2I] f32
+            inter = self.wgateup_t.shape[1] // 2
+            dg, du = adapter.delta("gate", xn2), adapter.delta("up", xn2)
+            gate, up = gateup[:, :inter], gateup[:, inter:]
+            if dg is not None:
+                gate = gate + dg.float()
+            if du is not None:
+                up = up + du.float()
+            act = (torch.nn.functional.silu(gate) * up).contiguous()
+            h3 = self.wdown_t.gemv(act, ws, h2, _EPI_RESIDUAL_BF16)
+            dd = adapter.delta("down", act)
+            if dd is not None:
+                h3 = h3 + dd.to(h3.dtype)
         return h3.view(B, 1, H)
 
     # --------------------------------------------------- training (autograd)
 
-    def forward_autograd(self, hidden: torch.Tensor, prefix_length: int = 0) -> torch.Tensor:
+    def _qkv_adapter_delta(self, x, adapter):
+        """Returns the concatenated qkv LoRA delta for inputs [..., H]."""
+        qd, kd = self.qh * self.hd, self.kh * self.hd
+        parts = []
+        for key, width in (("q", qd), ("k", kd), ("v", kd)):
+            d = adapter.delta(key, x)
+            parts.append(d if d is not None else x.new_zeros(*x.shape[:-1], width))
+        return torch.cat(parts, dim=-1)
+
+    def _apply_qkv_adapter(self, qkv, x, adapter):
+        """Adds LoRA deltas into the fused qkv buffer (decode fast path)."""
+        qd = self.qh * self.hd
+        kd = self.kh * self.hd
+        for key, lo, hi in (("q", 0, qd), ("k", qd, qd + kd), ("v", qd + kd, qd + 2 * kd)):
+            d = adapter.delta(key, x)
+            if d is not None:
+                qkv[:, lo:hi] += d.float()
+
+    def forward_autograd(self, hidden: torch.Tensor, prefix_length: int = 0, adapter=None) -> torch.Tensor:
         """Differentiable forward on the transposed weights (torch primitives
         only — the HIP kernels are inference-only). Used by rpc_backward on GPU
         servers; weights are frozen, grads flow to inputs/prompts."""
@@ -209,6 +251,8 @@ class LlamaFastPath:
         self._ensure_rope(end)
         xn = reference.rms_norm(hidden, self.ln1_w, self.eps)
         qkv = torch.matmul(xn, self.wqkv_t.dense())
+        if adapter is not None:
+            qkv = qkv + self._qkv_adapter_delta(xn, adapter)
         q = qkv[..., : self.qh * self.hd].view(B, S, self.qh, self.hd).transpose(1, 2)
         k = qkv[..., self.qh * self.hd : (self.qh + self.kh) * self.hd].view(B, S, self.kh, self.hd).transpose(1, 2)
         v = qkv[..., (self.qh + self.kh) * self.hd :].view(B, S, self.kh, self.hd).transpose(1, 2)
@@ -216,12 +260,29 @@ class LlamaFastPath:
         q, k = reference.apply_rope(q, k, self.rope_cos, self.rope_sin, pos)
         attn = reference.attention(q, k, v, causal=True)
         attn = attn.transpose(1, 2).reshape(B, S, self.qh * self.hd).to(hidden.dtype)
-        h2 = hidden + torch.matmul(attn, self.wo_t.dense())
+        o = torch.matmul(attn, self.wo_t.dense())
+        if adapter is not None:
+            d = adapter.delta("o", attn)
+            if d is not None:
+                o = o + d
+        h2 = hidden + o
         xn2 = reference.rms_norm(h2, self.ln2_w, self.eps)
         gateup = torch.matmul(xn2, self.wgateup_t.dense())
         inter = self.wgateup_t.shape[1] // 2
-        act = reference.swiglu(gateup[..., :inter], gateup[..., inter:]).to(hidden.dtype)
-        return h2 + torch.matmul(act, self.wdown_t.dense())
+        gate, up = gateup[..., :inter], gateup[..., inter:]
+        if adapter is not None:
+            dg, du = adapter.delta("gate", xn2), adapter.delta("up", xn2)
+            if dg is not None:
+                gate = gate + dg
+            if du is not None:
+                up = up + du
+        act = reference.swiglu(gate, up).to(hidden.dtype)
+        down = torch.matmul(act, self.wdown_t.dense())
+        if adapter is not None:
+            dd = adapter.delta("down", act)
+            if dd is not None:
+                down = down + dd
+        return h2 + down
 
     # ------------------------------------------------------------ prefill
 
@@ -230,6 +291,7 @@ class LlamaFastPath:
         hidden: torch.Tensor,  # [B, S, H] bf16
         kv_cache: Optional[Tuple[torch.Tensor, torch.Tensor]],
         prefix_length: int,
+        adapter=None,
     ) -> torch.Tensor:
         B, S, H = hidden.shape
         hidden = hidden.to(torch.bfloat16)
@@ -238,6 +300,8 @@ class LlamaFastPath:
 
         xn = self.hip.rms_norm(hidden, self.ln1_w, self.eps)
         qkv = torch.matmul(xn, self.wqkv_t.dense())  # [B, S, qkv] bf16 (rocBLAS)
+        if adapter is not None:
+            qkv = qkv + self._qkv_adapter_delta(xn, adapter).to(qkv.dtype)
         q = qkv[..., : self.qh * self.hd].view(B, S, self.qh, self.hd).transpose(1, 2)
         k = qkv[..., self.qh * self.hd : (self.qh + self.kh) * self.hd].view(B, S, self.kh, self.hd).transpose(1, 2)
         v = qkv[..., (self.qh + self.kh) * self.hd :].view(B, S, self.kh, self.hd).transpose(1, 2)
@@ -253,9 +317,26 @@ class LlamaFastPath:
         else:
             attn = reference.attention(q, k, v, causal=True)
         attn = attn.transpose(1, 2).reshape(B, S, self.qh * self.hd).to(torch.bfloat16)
-        h2 = hidden + torch.matmul(attn, self.wo_t.dense())
+        o = torch.matmul(attn, self.wo_t.dense())
+        if adapter is not None:
+            d = adapter.delta("o", attn)
+            if d is not None:
+                o = o + d
+        h2 = hidden + o
         xn2 = self.hip.rms_norm(h2, self.ln2_w, self.eps)
         gateup = torch.matmul(xn2, self.wgateup_t.dense())
         inter = self.wgateup_t.shape[1] // 2
-        act = self.hip.swiglu(gateup[..., :inter].contiguous(), gateup[..., inter:].contiguous())
-        return h2 + torch.matmul(act, self.wdown_t.dense())
+        gate, up = gateup[..., :inter].contiguous(), gateup[..., inter:].contiguous()
+        if adapter is not None:
+            dg, du = adapter.delta("gate", xn2), adapter.delta("up", xn2)
+            if dg is not None:
+                gate = gate + dg
+            if du is not None:
+                up = up + du
+        act = self.hip.swiglu(gate.contiguous(), up.contiguous())
+        down = torch.matmul(act, self.wdown_t.dense())
+        if adapter is not None:
+            dd = adapter.delta("down", act)
+            if dd is not None:
+                down = down + dd
+        return h2 + down

@@ -122,18 +122,25 @@ class TransformerConnectionHandler:
     # -------------------------------------------------------- rpc_forward
 
     def _forward_chain(
-        self, uids: List[ModuleUID], hidden_states: torch.Tensor, prompts: Optional[torch.Tensor]
+        self,
+        uids: List[ModuleUID],
+        hidden_states: torch.Tensor,
+        prompts: Optional[torch.Tensor],
+        active_adapter: Optional[str] = None,
     ) -> torch.Tensor:
         """Runs IN the runtime thread: chain of stateless block forwards."""
+        from petals_amd.utils.peft import using_adapter
+
         backend0 = self.backends[uids[0]]
         device, dtype = backend0.device, backend0.dtype
         hidden_states = hidden_states.to(device=device, dtype=dtype)
         prompt_list = self._split_prompts(prompts, len(uids), dtype, device)
-        for uid, prompt in zip(uids, prompt_list):
-            if prompt is not None:
-                hidden_states = hidden_states.clone()
-                hidden_states[:, : prompt.shape[1]] += prompt
-            hidden_states = self.backends[uid].forward(hidden_states)
+        with using_adapter(active_adapter):
+            for uid, prompt in zip(uids, prompt_list):
+                if prompt is not None:
+                    hidden_states = hidden_states.clone()
+                    hidden_states[:, : prompt.shape[1]] += prompt
+                hidden_states = self.backends[uid].forward(hidden_states)
         return hidden_states.cpu()
 
     async def _handle_forward(self, meta: Dict[str, Any], tensors: List[torch.Tensor]) -> List[torch.Tensor]:
@@ -142,7 +149,9 @@ class TransformerConnectionHandler:
         prompts = tensors[1] if len(tensors) > 1 else None
         assert hidden_states.ndim == 3
         priority = self.prioritizer.prioritize(hidden_states, points=meta.get("points", 0), type="forward")
-        out = await self.runtime.submit(priority, self._forward_chain, uids, hidden_states, prompts)
+        out = await self.runtime.submit(
+            priority, self._forward_chain, uids, hidden_states, prompts, meta.get("active_adapter")
+        )
         return [out]
 
     async def rpc_forward(self, request: RpcMessage, stream: RpcStream) -> None:
@@ -158,15 +167,24 @@ class TransformerConnectionHandler:
     # ------------------------------------------------------- rpc_backward
 
     def _backward_chain(
-        self, uids: List[ModuleUID], inputs: torch.Tensor, grad_outputs: torch.Tensor, prompts: Optional[torch.Tensor]
+        self,
+        uids: List[ModuleUID],
+        inputs: torch.Tensor,
+        grad_outputs: torch.Tensor,
+        prompts: Optional[torch.Tensor],
+        active_adapter: Optional[str] = None,
     ) -> Tuple[torch.Tensor, torch.Tensor]:
         """Runs IN the runtime thread. Re-runs forward to recover intermediate
         activations, then backward in reverse (parity: block_functions.py:84-141)."""
+        from petals_amd.utils.peft import using_adapter
+
         backend0 = self.backends[uids[0]]
         device, dtype = backend0.device, backend0.dtype
         inputs = inputs.to(device=device, dtype=dtype)
         grad_outputs = grad_outputs.to(device=device, dtype=dtype)
         prompt_list = self._split_prompts(prompts, len(uids), dtype, device)
+        adapter_scope = using_adapter(active_adapter)
+        adapter_scope.__enter__()
 
         # re-run forward (plain no_grad, NOT inference_mode: these activations
         # feed autograd below) to recover intermediate inputs
@@ -194,6 +212,7 @@ class TransformerConnectionHandler:
                 grad_prompts.append(grad[:, : prompt.shape[1]].clone())
             else:
                 grad_prompts.append(None)
+        adapter_scope.__exit__(None, None, None)
         grad_prompts.reverse()
         if any(gp is not None for gp in grad_prompts):
             ref = next(gp for gp in grad_prompts if gp is not None)
@@ -208,7 +227,7 @@ class TransformerConnectionHandler:
         prompts = tensors[2] if len(tensors) > 2 else None
         priority = self.prioritizer.prioritize(inputs, points=meta.get("points", 0), type="backward")
         grad_inputs, grad_prompts = await self.runtime.submit(
-            priority, self._backward_chain, uids, inputs, grad_outputs, prompts
+            priority, self._backward_chain, uids, inputs, grad_outputs, prompts, meta.get("active_adapter")
         )
         return [grad_inputs, grad_prompts]
 
@@ -241,13 +260,16 @@ class TransformerConnectionHandler:
         hidden_states = hidden_states.to(device=device, dtype=dtype)
         if hypo_ids is not None and not is_dummy(hypo_ids):
             hypo_ids = hypo_ids.to(device)
+        from petals_amd.utils.peft import using_adapter
+
         prompt_list = self._split_prompts(prompts, len(uids), dtype, device)
-        for uid, prompt, handle_pair in zip(uids, prompt_list, handles):
-            if prompt is not None:
-                hidden_states = hidden_states.clone()
-                hidden_states[:, : prompt.shape[1]] += prompt
-            info = InferenceMetadata(uid, prefix_length, tuple(handle_pair), active_adapter)
-            (hidden_states,) = self.backends[uid].inference_step(hidden_states, hypo_ids, info)
+        with using_adapter(active_adapter):
+            for uid, prompt, handle_pair in zip(uids, prompt_list, handles):
+                if prompt is not None:
+                    hidden_states = hidden_states.clone()
+                    hidden_states[:, : prompt.shape[1]] += prompt
+                info = InferenceMetadata(uid, prefix_length, tuple(handle_pair), active_adapter)
+                (hidden_states,) = self.backends[uid].inference_step(hidden_states, hypo_ids, info)
         return hidden_states.cpu()
 
     async def rpc_inference(self, request: RpcMessage, stream: RpcStream) -> None:

@@ -14,6 +14,7 @@ from __future__ import annotations
 
 import asyncio
 import logging
+import os
 import random
 import threading
 import time
@@ -64,6 +65,7 @@ class Server:
         balance_quality: float = 0.75,
         mean_balance_check_period: float = 120.0,
         quant_type: str = "none",
+        adapters: Sequence[str] = (),
         public_name: Optional[str] = None,
     ):
         self.config = load_model_config(model_name_or_dir)
@@ -97,6 +99,7 @@ class Server:
         self.balance_quality = balance_quality
         self.mean_balance_check_period = mean_balance_check_period
         self.public_name = public_name
+        self.adapters = tuple(adapters)
         self._throughput_setting = throughput
 
         self.module_uids = [make_uid(self.config.dht_prefix, i) for i in range(self.config.num_blocks)]
@@ -232,6 +235,7 @@ class Server:
             inference_rps=inference_rps,
             torch_dtype=str(self.torch_dtype).replace("torch.", ""),
             quant_type=self.quant_type,
+            adapters=tuple(os.path.basename(os.path.normpath(a)) for a in self.adapters),
         )
         await self._announce()
 
@@ -246,6 +250,12 @@ class Server:
                     device=self.device, quant_type=self.quant_type,
                 ),
             )
+            for adapter_dir in self.adapters:
+                from petals_amd.utils.peft import add_adapter_to_block, load_block_adapter
+
+                ad = load_block_adapter(adapter_dir, i, self.config.block_prefix)
+                if ad is not None:
+                    add_adapter_to_block(block, ad)
             self.backends[uid] = TransformerBackend(
                 uid, block, config=self.config, memory_cache=self.memory_cache, dtype=self.torch_dtype
             )
@@ -256,6 +266,7 @@ class Server:
             runtime=self.runtime,
             inference_max_length=self.inference_max_length,
             p2p=self.p2p,
+            adapters=tuple(os.path.basename(os.path.normpath(a)) for a in self.adapters),
         )
         self.handler.register(self.p2p)
 
