@@ -67,6 +67,8 @@ class Server:
         quant_type: str = "none",
         adapters: Sequence[str] = (),
         public_name: Optional[str] = None,
+        announce_host: Optional[str] = None,
+        skip_reachability_check: bool = False,
     ):
         self.config = load_model_config(model_name_or_dir)
         self.model_name_or_dir = model_name_or_dir
@@ -74,6 +76,8 @@ class Server:
             self.config.dht_prefix = dht_prefix
         self.initial_peers = [tuple(p) for p in initial_peers]
         self.host, self.port = host, port
+        self.announce_host = announce_host
+        self.skip_reachability_check = skip_reachability_check
 
         if device is None:
             device = "cuda" if torch.cuda.is_available() else "cpu"
@@ -178,8 +182,19 @@ class Server:
     async def _amain(self):
         self.p2p = P2PNode()
         await self.p2p.listen(host=self.host, port=self.port)
-        self.listen_addr = self.p2p.listen_addr
+        # the address other peers should dial (never announce 0.0.0.0)
+        host = self.announce_host or self.host
+        if host in ("0.0.0.0", "::"):
+            import socket
+
+            host = socket.gethostbyname(socket.gethostname())
+        self.listen_addr = (host, self.p2p.listen_addr[1])
         self.dht_node = await DHTNode.create(initial_peers=self.initial_peers, p2p=self.p2p)
+        from petals_amd.server.reachability import ReachabilityProtocol, validate_reachability
+
+        ReachabilityProtocol(self.p2p)
+        if not self.skip_reachability_check and self.initial_peers:
+            await validate_reachability(self.p2p, self.listen_addr, self.initial_peers, wait_time=30.0)
 
         self.runtime = PriorityRuntime(self.device).start()
         while not self._stop.is_set():
