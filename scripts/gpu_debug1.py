@@ -104,3 +104,17 @@ if __name__ == "__main__":
     gemv_bench()
     attn_bench()
     layer_decode_bench()
+    nf4_bench()
+
+
+def nf4_bench():
+    ws = torch.empty(64 * 57344, dtype=torch.float32, device="cuda")
+    for in_dim, out_dim, name in ((8192, 10240, "qkv"), (8192, 8192, "o"), (8192, 57344, "gateup"), (28672, 8192, "down")):
+        wt = (torch.randn(in_dim, out_dim, device="cuda") * 0.02).to(torch.bfloat16)
+        packed, absmax = hip.nf4_quantize(wt)
+        del wt
+        x = torch.randn(1, in_dim, device="cuda")
+        t = bench_kernel(lambda: hip.gemv_nf4(packed, absmax, x, ws, None, 0))
+        gb = (packed.numel() + absmax.numel() * 2) / 1e9
+        print(f"gemv_nf4 {name} [{in_dim},{out_dim}]: {t*1e6:.1f} us, {gb/t:.0f} GB/s packed ({gb*1000:.0f}MB)", flush=True)
+        torch.cuda.empty_cache()
