@@ -84,7 +84,7 @@ __global__ void nf4_dequant_kernel(
 
 // -------------------------------------------------------------- NF4 gemv
 
-#define NF4_OUT_PER_WAVE 2048  // 64 lanes x 32 outputs (16 B packed / row)
+#define NF4_OUT_PER_WAVE 1024  // 64 lanes x 16 outputs (8 B packed / row)
 
 template <int BATCH>
 __global__ void gemv_nf4_kernel(
@@ -105,29 +105,29 @@ __global__ void gemv_nf4_kernel(
   const int lane = threadIdx.x & (WAVE - 1);
   const float* lutb = lut + (lane & 31);
 #define NF4_L(nib) lutb[(nib) << 5]
-  const int out0 = blockIdx.x * NF4_OUT_PER_WAVE + lane * 32;
+  const int out0 = blockIdx.x * NF4_OUT_PER_WAVE + lane * 16;
   if (out0 >= out_dim) return;
   const int split = blockIdx.y;
   const int i_begin = split * i_per_split;
   const int i_end = min(i_begin + i_per_split, in_dim);
-  const bool full = (out0 + 32) <= out_dim;
+  const bool full = (out0 + 16) <= out_dim;
 
-  float acc[BATCH][32];
+  float acc[BATCH][16];
 #pragma unroll
   for (int b = 0; b < BATCH; ++b)
 #pragma unroll
-    for (int v = 0; v < 32; ++v) acc[b][v] = 0.f;
+    for (int v = 0; v < 16; ++v) acc[b][v] = 0.f;
 
   if (full) {
-    constexpr int UNROLL = 4;
+    constexpr int UNROLL = 8;
     const int half_out = out_dim >> 1;
     const unsigned char* pp = packed + (size_t)i_begin * half_out + (out0 >> 1);
     int i = i_begin;
     for (; i + UNROLL <= i_end; i += UNROLL) {
-      uint4 pk[UNROLL];  // 16 bytes = 32 nibbles (one dwordx4 per input row)
+      uint2 pk[UNROLL];  // 8 bytes = 16 nibbles
 #pragma unroll
       for (int u = 0; u < UNROLL; ++u)
-        pk[u] = *reinterpret_cast<const uint4*>(pp + (size_t)u * half_out);
+        pk[u] = *reinterpret_cast<const uint2*>(pp + (size_t)u * half_out);
       float am[UNROLL];
 #pragma unroll
       for (int u = 0; u < UNROLL; ++u)
@@ -139,32 +139,31 @@ __global__ void gemv_nf4_kernel(
         for (int u = 0; u < UNROLL; ++u) xs[b][u] = x[(size_t)b * in_dim + i + u];
 #pragma unroll
       for (int u = 0; u < UNROLL; ++u) {
-        const unsigned int wds[4] = {pk[u].x, pk[u].y, pk[u].z, pk[u].w};
+        float wf[16];
+        const unsigned int w0 = pk[u].x, w1 = pk[u].y;
+#pragma unroll
+        for (int v = 0; v < 8; ++v) wf[v] = NF4_L((w0 >> (4 * v)) & 0xF);
+#pragma unroll
+        for (int v = 0; v < 8; ++v) wf[8 + v] = NF4_L((w1 >> (4 * v)) & 0xF);
 #pragma unroll
         for (int b = 0; b < BATCH; ++b) {
           const float xa = xs[b][u] * am[u];
 #pragma unroll
-          for (int d = 0; d < 4; ++d) {
-#pragma unroll
-            for (int v = 0; v < 8; ++v)
-              acc[b][8 * d + v] = fmaf(NF4_L((wds[d] >> (4 * v)) & 0xFu), xa, acc[b][8 * d + v]);
-          }
+          for (int v = 0; v < 16; ++v) acc[b][v] = fmaf(wf[v], xa, acc[b][v]);
         }
       }
       pp += (size_t)UNROLL * half_out;
     }
     for (; i < i_end; ++i) {
-      const uint4 pk = *reinterpret_cast<const uint4*>(packed + (size_t)i * half_out + (out0 >> 1));
-      const unsigned int wds[4] = {pk.x, pk.y, pk.z, pk.w};
+      const uint2 pk = *reinterpret_cast<const uint2*>(packed + (size_t)i * half_out + (out0 >> 1));
       const float am = bf16_to_f32(absmax[(size_t)i * (out_dim >> 6) + (out0 >> 6)]);
 #pragma unroll
       for (int b = 0; b < BATCH; ++b) {
         const float xa = x[(size_t)b * in_dim + i] * am;
 #pragma unroll
-        for (int d = 0; d < 4; ++d)
+        for (int v = 0; v < 8; ++v) acc[b][v] = fmaf(NF4_L((pk.x >> (4 * v)) & 0xF), xa, acc[b][v]);
 #pragma unroll
-          for (int v = 0; v < 8; ++v)
-            acc[b][8 * d + v] = fmaf(NF4_L((wds[d] >> (4 * v)) & 0xF), xa, acc[b][8 * d + v]);
+        for (int v = 0; v < 8; ++v) acc[b][8 + v] = fmaf(NF4_L((pk.y >> (4 * v)) & 0xF), xa, acc[b][8 + v]);
       }
     }
   } else {
@@ -186,7 +185,7 @@ __global__ void gemv_nf4_kernel(
     float* dst = partials + ((size_t)split * BATCH + b) * out_dim + out0;
     if (full) {
 #pragma unroll
-      for (int q = 0; q < 8; ++q)
+      for (int q = 0; q < 4; ++q)
         reinterpret_cast<float4v*>(dst)[q] =
             float4v{acc[b][4 * q], acc[b][4 * q + 1], acc[b][4 * q + 2], acc[b][4 * q + 3]};
     } else {
@@ -244,11 +243,13 @@ torch::Tensor gemv_nf4(
   const int in_dim = packed.size(0);
   const int out_dim = packed.size(1) * 2;
   const int batch = x.size(0);
-  TORCH_CHECK(x.size(1) == in_dim && batch <= 2, "NF4 decode gemv supports batch <= 2");
+  TORCH_CHECK(x.size(1) == in_dim && batch <= 4, "NF4 decode gemv supports batch <= 4");
 
   const long out_waves = (out_dim + NF4_OUT_PER_WAVE - 1) / NF4_OUT_PER_WAVE;
-  long splits = splits_override > 0 ? splits_override : (768 + out_waves - 1) / out_waves;
-  long max_splits = (in_dim + 255) / 256;
+  // NF4 matrices are 4x smaller than bf16: allow chunks down to 64 input rows
+  // so small projections still spread over the 256 CUs
+  long splits = splits_override > 0 ? splits_override : (1536 + out_waves - 1) / out_waves;
+  long max_splits = (in_dim + 63) / 64;
   if (splits > max_splits) splits = max_splits;
   if (splits < 1) splits = 1;
 
@@ -271,6 +272,8 @@ torch::Tensor gemv_nf4(
   switch (batch) {
     case 1: LAUNCH_NF4(1); break;
     case 2: LAUNCH_NF4(2); break;
+    case 3: LAUNCH_NF4(3); break;
+    case 4: LAUNCH_NF4(4); break;
   }
 #undef LAUNCH_NF4
   HIP_CHECK_LAST();

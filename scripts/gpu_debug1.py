@@ -98,13 +98,6 @@ def layer_decode_bench():
     print(f"70B layer decode (no per-call sync): {wall_nosync*1e6:.1f} us", flush=True)
 
 
-if __name__ == "__main__":
-    attn_debug()
-    rmsnorm_bench()
-    gemv_bench()
-    attn_bench()
-    layer_decode_bench()
-    nf4_bench()
 
 
 def nf4_bench():
@@ -118,3 +111,12 @@ def nf4_bench():
         gb = (packed.numel() + absmax.numel() * 2) / 1e9
         print(f"gemv_nf4 {name} [{in_dim},{out_dim}]: {t*1e6:.1f} us, {gb/t:.0f} GB/s packed ({gb*1000:.0f}MB)", flush=True)
         torch.cuda.empty_cache()
+
+
+if __name__ == "__main__":
+    attn_debug()
+    rmsnorm_bench()
+    gemv_bench()
+    attn_bench()
+    layer_decode_bench()
+    nf4_bench()
