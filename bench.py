@@ -41,7 +41,7 @@ def main():
     p.add_argument("--prompt-len", type=int, default=128)
     p.add_argument("--batch", type=int, default=1)
     p.add_argument("--device", default="cuda")
-    p.add_argument("--quant", default="none", choices=["none", "nf4"])
+    p.add_argument("--quant", default="nf4", choices=["none", "nf4"], help="BASELINE config #3 names NF4 for the 70B pipeline; --quant none measures pure bf16")
     args = p.parse_args()
 
     import torch.distributed as dist
@@ -55,6 +55,8 @@ def main():
     world = max(world, 1)
 
     use_cuda = args.device == "cuda" and torch.cuda.is_available()
+    if not use_cuda:
+        args.quant = "none"
     if use_cuda:
         device = torch.device("cuda", int(os.environ.get("LOCAL_RANK", 0)))
         torch.cuda.set_device(device)
