@@ -39,6 +39,42 @@ class _Session:
         self.max_length = max_length
         self.prefix_length = 0
         self.pushed_inputs: asyncio.Queue = asyncio.Queue()
+        self.span_graph = None  # _SpanDecodeGraph once captured (GPU decode)
+
+
+class _SpanDecodeGraph:
+    """Per-session hipGraph of one whole-span decode step (all blocks): the
+    session's KV tensors have stable addresses for its lifetime, so a single
+    capture replays for every generated token with only the device-resident
+    position advancing (ops/fused_decode.DecodeContext). This gives the real
+    serving path the same launch-overhead-free decode as bench.py."""
+
+    def __init__(self, blocks, cache_pairs, batch: int, hidden_size: int, device, dtype, adapter_name,
+                 initial_position: int):
+        from petals_amd.ops.fused_decode import DecodeContext
+        from petals_amd.utils.graphs import GraphedCallable
+        from petals_amd.utils.peft import using_adapter
+
+        self.ctx = DecodeContext(device)
+        self.h_in = torch.empty(batch, 1, hidden_size, device=device, dtype=dtype)
+        # warmup/capture writes garbage K/V at this position — it MUST be the
+        # position the first real replay will overwrite (never the prefilled
+        # prefix)
+        self.ctx.set_position(initial_position)
+
+        def span_fn():
+            h = self.h_in
+            with using_adapter(adapter_name):
+                for block, (k, v) in zip(blocks, cache_pairs):
+                    h = block(h, kv_cache=(k, v), ctx=self.ctx)
+            return h
+
+        self.graph = GraphedCallable(span_fn, [])
+
+    def step(self, hidden_states: torch.Tensor, prefix_length: int) -> torch.Tensor:
+        self.ctx.set_position(prefix_length)
+        self.h_in.copy_(hidden_states.view(self.h_in.shape))
+        return self.graph.replay()
 
 
 class TransformerConnectionHandler:
@@ -252,13 +288,51 @@ class TransformerConnectionHandler:
         handles: List[Tuple[int, int]],
         prefix_length: int,
         active_adapter: Optional[str],
+        session: Optional[_Session] = None,
     ) -> torch.Tensor:
         """Runs IN the runtime thread: one inference step through the whole span
         (the single-process analog of reference _MergedInferenceStep)."""
         backend0 = self.backends[uids[0]]
         device, dtype = backend0.device, backend0.dtype
         hidden_states = hidden_states.to(device=device, dtype=dtype)
-        if hypo_ids is not None and not is_dummy(hypo_ids):
+        has_hypo = hypo_ids is not None and not is_dummy(hypo_ids)
+        has_prompts = prompts is not None and not is_dummy(prompts)
+
+        # fast serving path: whole-span hipGraph decode (1 new token, GPU,
+        # every block on the fused path). Beam reorder runs eagerly before the
+        # replay; prompts force the eager path (they change per step).
+        all_fast = all(getattr(self.backends[uid].block, "_fast", None) is not None for uid in uids)
+        max_graph_batch = 1
+        if all_fast:
+            quants = {self.backends[uid].block._fast.quant for uid in uids}
+            max_graph_batch = 2 if "nf4" in quants else 8
+        if (
+            session is not None
+            and device.type == "cuda"
+            and all_fast
+            and hidden_states.shape[1] == 1
+            and hidden_states.shape[0] <= max_graph_batch
+            and not has_prompts
+        ):
+            if has_hypo:
+                hypo = hypo_ids.to(device)
+                with self.memory_cache.use_cache(*(h for pair in handles for h in pair)) as tensors:
+                    for t in tensors:
+                        t[...] = t[hypo]
+            if session.span_graph is None:
+                cache_pairs = []
+                with self.memory_cache.use_cache(*(h for pair in handles for h in pair)) as tensors:
+                    for i in range(len(uids)):
+                        cache_pairs.append((tensors[2 * i], tensors[2 * i + 1]))
+                blocks = [self.backends[uid].block for uid in uids]
+                session.span_graph = _SpanDecodeGraph(
+                    blocks, cache_pairs, hidden_states.shape[0], hidden_states.shape[-1],
+                    device, dtype, active_adapter, initial_position=prefix_length,
+                )
+            out = session.span_graph.step(hidden_states, prefix_length)
+            return out.cpu()
+
+        if has_hypo:
             hypo_ids = hypo_ids.to(device)
         from petals_amd.utils.peft import using_adapter
 
@@ -324,6 +398,7 @@ class TransformerConnectionHandler:
                             handles,
                             session.prefix_length,
                             active_adapter,
+                            session,
                         )
                     else:
                         output = hidden_states
