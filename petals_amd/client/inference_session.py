@@ -171,8 +171,10 @@ class InferenceSession:
             assert hypo_ids.dtype == torch.int64
 
         inputs_device, inputs_dtype = inputs.device, inputs.dtype
-        inputs = inputs.cpu().float()
-        prompts = prompts.cpu().float() if not is_dummy(prompts) else prompts
+        inputs = inputs.cpu()
+        if inputs.dtype not in (torch.bfloat16, torch.float16, torch.float32):
+            inputs = inputs.float()
+        prompts = prompts.cpu() if not is_dummy(prompts) else prompts
         n = inputs.shape[1]
         if self._position + n > self._max_length:
             raise ValueError(f"max_length exceeded: {self._position} + {n} > {self._max_length}")

@@ -105,6 +105,7 @@ class DistributedModelBase(nn.Module, PTuneMixin):
     ) -> ModelOutput:
         assert (input_ids is None) != (inputs_embeds is None), "provide input_ids xor inputs_embeds"
         if inputs_embeds is None:
+            input_ids = input_ids.to(self.embed_tokens.weight.device)
             inputs_embeds = self._embed(input_ids)
 
         batch = inputs_embeds.shape[0]

@@ -40,6 +40,8 @@ def main():
         args.model, initial_peers=[boot.listen_addr], dht_prefix="serve-bench",
         show_route=False, max_retries=2,
     )
+    # thin client computes embeddings + LM head on the GPU too
+    model = model.to(device="cuda", dtype=torch.bfloat16)
     torch.manual_seed(0)
     ids = torch.randint(0, model.config.vocab_size, (1, args.prompt_len))
 
@@ -52,7 +54,7 @@ def main():
     elapsed = time.perf_counter() - t0
     tps = args.new_tokens / elapsed
     print(f"FULL-STACK serving: {tps:.2f} tokens/s ({elapsed/args.new_tokens*1000:.1f} ms/token) "
-          f"model={args.model} quant={args.quant} [includes TCP + fp32 wire + client head]", flush=True)
+          f"model={args.model} quant={args.quant} [includes TCP wire + client embeds/head on GPU]", flush=True)
 
     model.transformer.h.sequence_manager.shutdown()
     server.shutdown()
