@@ -70,7 +70,7 @@ class RemoteGenerationMixin:
 
         with ctx as sess, self.transformer.h.use_session(sess):
             generated = input_ids
-            unfinished = torch.ones(batch, dtype=torch.bool)
+            unfinished = torch.ones(batch, dtype=torch.bool, device=input_ids.device)
             step_input = input_ids
             for _ in range(max_new_tokens):
                 logits = self(input_ids=step_input).logits[:, -1, :]
@@ -80,6 +80,7 @@ class RemoteGenerationMixin:
                     next_token = self._sample(logits, temperature, top_k, top_p)
                 else:
                     next_token = logits.argmax(dim=-1)
+                next_token = next_token.to(generated.device)
                 if eos_token_id is not None:
                     if pad_token_id is not None:
                         next_token = torch.where(
