@@ -45,6 +45,16 @@ def main():
     torch.manual_seed(0)
     ids = torch.randint(0, model.config.vocab_size, (1, args.prompt_len))
 
+    try:
+        _run(model, ids, args)
+    finally:
+        model.transformer.h.sequence_manager.shutdown()
+        server.shutdown()
+        boot.shutdown()
+        os._exit(0)  # skip interpreter-teardown races between HIP and daemon threads
+
+
+def _run(model, ids, args):
     # warmup generation (includes session setup + graph capture)
     out = model.generate(ids, max_new_tokens=8, do_sample=False)
     assert out.shape[1] == args.prompt_len + 8
@@ -55,10 +65,6 @@ def main():
     tps = args.new_tokens / elapsed
     print(f"FULL-STACK serving: {tps:.2f} tokens/s ({elapsed/args.new_tokens*1000:.1f} ms/token) "
           f"model={args.model} quant={args.quant} [includes TCP wire + client embeds/head on GPU]", flush=True)
-
-    model.transformer.h.sequence_manager.shutdown()
-    server.shutdown()
-    boot.shutdown()
 
 
 if __name__ == "__main__":
