@@ -30,3 +30,9 @@ class ClientConfig:
     max_pinged: int = 3
 
     active_adapter: Optional[str] = None
+
+    # wire compression (utils/serialization): "none" | "float16" | "bfloat16" |
+    # "blockwise_8bit" — applied to activations we SEND; output_compression is
+    # requested from servers for what they send back
+    wire_compression: str = "none"
+    output_compression: str = "none"

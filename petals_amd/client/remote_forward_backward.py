@@ -29,16 +29,19 @@ async def run_remote_forward(
     *,
     metadata: Optional[Dict] = None,
     timeout: float = 180.0,
+    compression: str = "none",
 ) -> torch.Tensor:
     tensors = [hidden_states.cpu(), prompts.cpu()]
     meta = {"uids": CHAIN_DELIMITER.join(uids), **(metadata or {})}
+    compressions = [compression] * len(tensors) if compression and compression != "none" else None
     if _payload_bytes(tensors) > MAX_UNARY_PAYLOAD_SIZE // 2:
         stream = await p2p.open_stream(addr, "petals.rpc_forward_stream", RpcMessage(meta=meta), timeout=timeout)
-        await send_tensors_streamed(stream, tensors, close=True)
+        await send_tensors_streamed(stream, tensors, compressions=compressions, close=True)
         _, outs = await receive_tensors_streamed(stream, timeout=timeout)
     else:
         resp = await p2p.call_unary(
-            addr, "petals.rpc_forward", RpcMessage(meta=meta, tensors=tensors), timeout=timeout
+            addr, "petals.rpc_forward", RpcMessage(meta=meta, tensors=tensors), timeout=timeout,
+            compressions=compressions,
         )
         outs = resp.tensors
     return outs[0]
@@ -54,16 +57,19 @@ async def run_remote_backward(
     *,
     metadata: Optional[Dict] = None,
     timeout: float = 180.0,
+    compression: str = "none",
 ) -> Tuple[torch.Tensor, torch.Tensor]:
     tensors = [inputs.cpu(), grad_outputs.cpu(), prompts.cpu()]
     meta = {"uids": CHAIN_DELIMITER.join(uids), **(metadata or {})}
+    compressions = [compression] * len(tensors) if compression and compression != "none" else None
     if _payload_bytes(tensors) > MAX_UNARY_PAYLOAD_SIZE // 2:
         stream = await p2p.open_stream(addr, "petals.rpc_backward_stream", RpcMessage(meta=meta), timeout=timeout)
-        await send_tensors_streamed(stream, tensors, close=True)
+        await send_tensors_streamed(stream, tensors, compressions=compressions, close=True)
         _, outs = await receive_tensors_streamed(stream, timeout=timeout)
     else:
         resp = await p2p.call_unary(
-            addr, "petals.rpc_backward", RpcMessage(meta=meta, tensors=tensors), timeout=timeout
+            addr, "petals.rpc_backward", RpcMessage(meta=meta, tensors=tensors), timeout=timeout,
+            compressions=compressions,
         )
         outs = resp.tensors
     return outs[0], outs[1]

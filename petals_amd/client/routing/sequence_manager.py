@@ -233,10 +233,13 @@ class RemoteSequenceManager:
         return min(self.config.min_backoff * 2 ** (attempt_no - 1), self.config.max_backoff)
 
     def get_request_metadata(self, protocol: str, *args, **kwargs) -> Dict[str, Any]:
-        return {
+        meta = {
             "active_adapter": self.config.active_adapter,
             "points": 0,
         }
+        if self.config.output_compression != "none":
+            meta["output_compression"] = self.config.output_compression
+        return meta
 
     def address_of(self, peer_id: str) -> Tuple[str, int]:
         addr = self.addrs.get(peer_id)

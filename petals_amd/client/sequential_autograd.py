@@ -59,6 +59,7 @@ async def sequential_forward(
                 span_prompts,
                 metadata=metadata,
                 timeout=sequence_manager.config.request_timeout,
+                compression=sequence_manager.config.wire_compression,
             )
             assert out.shape == outputs.shape, f"bad output shape {out.shape} vs {outputs.shape}"
             intermediate_inputs.append(outputs)
@@ -111,6 +112,7 @@ async def sequential_backward(
                     span_prompts,
                     metadata=metadata,
                     timeout=sequence_manager.config.request_timeout,
+                    compression=sequence_manager.config.wire_compression,
                 )
                 if not is_dummy(grad_prompts):
                     grad_prompts_reversed.append(grad_prompts)

@@ -96,6 +96,19 @@ class DistributedModelBase(nn.Module, PTuneMixin):
     def word_embeddings(self):  # bloom-style alias
         return self.embed_tokens
 
+    @classmethod
+    def from_pretrained(cls, model_name_or_path: str, config=None, torch_dtype=torch.float32, **kwargs):
+        if config is None:
+            from petals_amd.utils.auto_config import AutoDistributedConfig
+
+            config = AutoDistributedConfig.from_pretrained(model_name_or_path, **kwargs)
+        model = cls(config)
+        if os.path.isdir(model_name_or_path):
+            _load_client_side_weights(model, model_name_or_path, config)
+        else:
+            _init_client_side_weights(model, config)
+        return model.to(torch_dtype)
+
     def forward(
         self,
         input_ids: Optional[torch.Tensor] = None,
@@ -240,7 +253,7 @@ def _load_client_side_weights(model, model_dir: str, config):
             for name in wanted:
                 if name in f.keys():
                     found[name] = f.get_tensor(name)
-    tfm = model.transformer
+    tfm = getattr(model, "transformer", None) or model
     if emb_name in found:
         tfm.embed_tokens.weight.data = found[emb_name].to(tfm.embed_tokens.weight.dtype)
     if norm_name in found:
@@ -268,7 +281,7 @@ def _init_client_side_weights(model, config):
     key = f"{config.name_or_path or config.model_type}:client"
     seed = (zlib.crc32(key.encode()) & 0x7FFFFFFF) or 1
     gen = torch.Generator().manual_seed(seed)
-    tfm = model.transformer
+    tfm = getattr(model, "transformer", None) or model
     with torch.no_grad():
         tfm.embed_tokens.weight.normal_(0, 0.02, generator=gen)
         if hasattr(model, "lm_head"):

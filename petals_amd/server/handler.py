@@ -190,9 +190,14 @@ class TransformerConnectionHandler:
         )
         return [out]
 
+    @staticmethod
+    def _out_compressions(meta, n):
+        c = meta.get("output_compression")
+        return [c] * n if c and c != "none" else None
+
     async def rpc_forward(self, request: RpcMessage, stream: RpcStream) -> None:
         outs = await asyncio.wait_for(self._handle_forward(request.meta, request.tensors), self.request_timeout)
-        await stream.close(RpcMessage(tensors=outs))
+        await stream.send(RpcMessage(tensors=outs), kind="end", compressions=self._out_compressions(request.meta, len(outs)))
 
     async def rpc_forward_stream(self, request: RpcMessage, stream: RpcStream) -> None:
         meta, tensors = await receive_tensors_streamed(stream, timeout=self.request_timeout)
@@ -269,7 +274,7 @@ class TransformerConnectionHandler:
 
     async def rpc_backward(self, request: RpcMessage, stream: RpcStream) -> None:
         outs = await asyncio.wait_for(self._handle_backward(request.meta, request.tensors), self.request_timeout)
-        await stream.close(RpcMessage(tensors=outs))
+        await stream.send(RpcMessage(tensors=outs), kind="end", compressions=self._out_compressions(request.meta, len(outs)))
 
     async def rpc_backward_stream(self, request: RpcMessage, stream: RpcStream) -> None:
         meta, tensors = await receive_tensors_streamed(stream, timeout=self.request_timeout)
