@@ -55,6 +55,21 @@ def main():
 
 
 def _run(model, ids, args):
+    # --- isolate raw session step latency (client + wire + server compute),
+    # no embeddings/head/generate logic
+    H = model.config.hidden_size
+    dev = model.transformer.embed_tokens.weight.device
+    h = torch.randn(1, 1, H, device=dev, dtype=model.transformer.embed_tokens.weight.dtype) * 0.02
+    with model.transformer.h.inference_session(max_length=64) as sess:
+        sess.step(h)  # session open + first step (graph capture server-side)
+        import time as _t
+
+        t0 = _t.perf_counter()
+        for _ in range(20):
+            sess.step(h)
+        dt = (_t.perf_counter() - t0) / 20
+        print(f"raw session step: {dt*1000:.2f} ms/token (client+wire+server)", flush=True)
+
     # warmup generation (includes session setup + graph capture)
     out = model.generate(ids, max_new_tokens=8, do_sample=False)
     assert out.shape[1] == args.prompt_len + 8
