@@ -61,6 +61,29 @@ class MixtralSparseMoeBlock(nn.Module):
 
 
 class MixtralBlock(nn.Module):
+    _fast = None
+
+    def optimize_for_inference(self, quant: str = "none") -> "MixtralBlock":
+        """MI355X fast path: fused attention decode + routed expert GEMVs."""
+        from petals_amd import ops as _ops
+        from petals_amd.ops.fused_moe import MixtralFastPath
+
+        hip = _ops._load_hip_ops()
+        if hip is None:
+            raise RuntimeError(
+                f"cannot optimize block for MI355X: HIP extension missing ({_ops._hip_import_error!r})"
+            )
+        gq = self.config.num_attention_heads // self.config.n_kv_heads
+        if self.config.head_dim not in (64, 128) or gq not in (1, 2, 4, 6, 8, 16):
+            import logging
+
+            logging.getLogger(__name__).warning(
+                "mixtral block geometry outside the fused fast path; serving via generic ops"
+            )
+            return self
+        self._fast = MixtralFastPath(self, hip, quant=quant)
+        return self
+
     def __init__(self, config: MixtralConfig, layer_idx: int = 0):
         super().__init__()
         self.config = config
@@ -70,7 +93,16 @@ class MixtralBlock(nn.Module):
         self.input_layernorm = RMSNorm(config.hidden_size, eps=config.layer_norm_eps)
         self.post_attention_layernorm = RMSNorm(config.hidden_size, eps=config.layer_norm_eps)
 
-    def forward(self, hidden_states, kv_cache=None, prefix_length: int = 0):
+    def forward(self, hidden_states, kv_cache=None, prefix_length: int = 0, ctx=None):
+        if self._fast is not None:
+            if torch.is_grad_enabled() and hidden_states.requires_grad:
+                assert kv_cache is None
+                return self._fast.forward_autograd(hidden_states, prefix_length)
+            max_b = 2 if self._fast.quant == "nf4" else 8
+            if kv_cache is not None and hidden_states.shape[1] == 1 and hidden_states.shape[0] <= max_b:
+                return self._fast.decode_step(hidden_states, kv_cache[0], kv_cache[1], prefix_length, ctx=ctx)
+            return self._fast.forward(hidden_states, kv_cache, prefix_length)
+
         residual = hidden_states
         hidden_states = self.input_layernorm(hidden_states)
         hidden_states = self.self_attn(hidden_states, kv_cache=kv_cache, prefix_length=prefix_length)

@@ -97,6 +97,8 @@ class _FastWeight:
 
 
 class LlamaFastPath:
+    graph_safe = True  # whole-span hipGraph capture is valid for this block
+
     def __init__(self, block, hip_ops, quant: str = "none"):
         cfg = block.config
         self.hip = hip_ops
@@ -120,19 +122,13 @@ class LlamaFastPath:
             hip_ops, quant,
         )  # [H, qh*hd + 2*kh*hd]
         self.wo_t = _FastWeight(t(attn.o_proj.weight), hip_ops, quant)
-        mlp = block.mlp
-        self.wgateup_t = _FastWeight(
-            torch.cat([t(mlp.gate_proj.weight), t(mlp.up_proj.weight)], dim=1), hip_ops, quant
-        )
-        self.wdown_t = _FastWeight(t(mlp.down_proj.weight), hip_ops, quant)
         self.ln1_w = block.input_layernorm.weight.detach().to(torch.bfloat16).contiguous()
         self.ln2_w = block.post_attention_layernorm.weight.detach().to(torch.bfloat16).contiguous()
-
-        # free the originals (they'd double the span's memory)
-        empty = torch.empty(0, device=device, dtype=torch.bfloat16)
-        for lin in (attn.q_proj, attn.k_proj, attn.v_proj, attn.o_proj, mlp.gate_proj, mlp.up_proj, mlp.down_proj):
-            lin.weight.data = empty
+        self._empty_bf16 = torch.empty(0, device=device, dtype=torch.bfloat16)
+        for lin in (attn.q_proj, attn.k_proj, attn.v_proj, attn.o_proj):
+            lin.weight.data = self._empty_bf16
         self.has_bias = attn.q_proj.bias is not None
+        self._init_mlp_weights(block, hip_ops, quant)
 
         self.rope_cos: Optional[torch.Tensor] = None
         self.rope_sin: Optional[torch.Tensor] = None
@@ -140,6 +136,41 @@ class LlamaFastPath:
         self._kv_len = torch.zeros(1, dtype=torch.int32, device=device)
         self.device = device
         self._empty_f32 = torch.empty(0, dtype=torch.float32, device=device)
+
+    def _init_mlp_weights(self, block, hip_ops, quant):
+        def t(w):
+            return w.detach().to(torch.bfloat16).t().contiguous()
+
+        mlp = block.mlp
+        self.wgateup_t = _FastWeight(
+            torch.cat([t(mlp.gate_proj.weight), t(mlp.up_proj.weight)], dim=1), hip_ops, quant
+        )
+        self.wdown_t = _FastWeight(t(mlp.down_proj.weight), hip_ops, quant)
+        for lin in (mlp.gate_proj, mlp.up_proj, mlp.down_proj):
+            lin.weight.data = self._empty_bf16
+
+    def _mlp_dense(self, xn2, adapter, autograd: bool):
+        """MLP on [B, S, H] inputs via rocBLAS matmuls on the transposed
+        (possibly NF4-dequantized) weights; differentiable when autograd."""
+        gateup = torch.matmul(xn2, self.wgateup_t.dense())
+        inter = self.wgateup_t.shape[1] // 2
+        gate, up = gateup[..., :inter], gateup[..., inter:]
+        if adapter is not None:
+            dg, du = adapter.delta("gate", xn2), adapter.delta("up", xn2)
+            if dg is not None:
+                gate = gate + dg
+            if du is not None:
+                up = up + du
+        if autograd:
+            act = reference.swiglu(gate, up).to(xn2.dtype)
+        else:
+            act = self.hip.swiglu(gate.contiguous(), up.contiguous())
+        down = torch.matmul(act, self.wdown_t.dense())
+        if adapter is not None:
+            dd = adapter.delta("down", act)
+            if dd is not None:
+                down = down + dd
+        return down
 
     def _ensure_rope(self, needed: int):
         if self.rope_cos is None or self.rope_cos.shape[0] < needed:
@@ -181,7 +212,7 @@ class LlamaFastPath:
             # already cover the cache (ensure_rope(lmax) at session start)
             self._ensure_rope(k_cache.shape[2])
             pos, kv_len = ctx.pos, ctx.kv_len
-        ws = _get_ws(self.device, "gemv", 64 * B * max(self.wgateup_t.shape[1], self.wqkv_t.shape[1], H))
+        ws = _get_ws(self.device, "gemv", 64 * B * self._max_gemv_out())
 
         xn = self.hip.rms_norm_f32out(h, self.ln1_w, self.eps)  # [B, H] f32
         qkv = self.wqkv_t.gemv(xn, ws, None, _EPI_PLAIN_F32)  # [B, qkv] f32
@@ -203,24 +234,30 @@ class LlamaFastPath:
         else:
             h2 = self.wo_t.gemv(attn, ws, h, _EPI_RESIDUAL_BF16)  # [B, H] bf16
         xn2 = self.hip.rms_norm_f32out(h2, self.ln2_w, self.eps)
+        h3 = self._mlp_decode(xn2, h2, ws, adapter)
+        return h3.view(B, 1, H)
+
+    def _max_gemv_out(self) -> int:
+        return max(self.wgateup_t.shape[1], self.wqkv_t.shape[1], self.wo_t.shape[1])
+
+    def _mlp_decode(self, xn2, h2, ws, adapter):
         if adapter is None:
             act = self.wgateup_t.gemv(xn2, ws, None, _EPI_SWIGLU_F32)  # [B, I] f32
-            h3 = self.wdown_t.gemv(act, ws, h2, _EPI_RESIDUAL_BF16)
-        else:
-            gateup = self.wgateup_t.gemv(xn2, ws, None, _EPI_PLAIN_F32)  # [B, 2I] f32
-            inter = self.wgateup_t.shape[1] // 2
-            dg, du = adapter.delta("gate", xn2), adapter.delta("up", xn2)
-            gate, up = gateup[:, :inter], gateup[:, inter:]
-            if dg is not None:
-                gate = gate + dg.float()
-            if du is not None:
-                up = up + du.float()
-            act = (torch.nn.functional.silu(gate) * up).contiguous()
-            h3 = self.wdown_t.gemv(act, ws, h2, _EPI_RESIDUAL_BF16)
-            dd = adapter.delta("down", act)
-            if dd is not None:
-                h3 = h3 + dd.to(h3.dtype)
-        return h3.view(B, 1, H)
+            return self.wdown_t.gemv(act, ws, h2, _EPI_RESIDUAL_BF16)
+        gateup = self.wgateup_t.gemv(xn2, ws, None, _EPI_PLAIN_F32)  # [B, 2I] f32
+        inter = self.wgateup_t.shape[1] // 2
+        dg, du = adapter.delta("gate", xn2), adapter.delta("up", xn2)
+        gate, up = gateup[:, :inter], gateup[:, inter:]
+        if dg is not None:
+            gate = gate + dg.float()
+        if du is not None:
+            up = up + du.float()
+        act = (torch.nn.functional.silu(gate) * up).contiguous()
+        h3 = self.wdown_t.gemv(act, ws, h2, _EPI_RESIDUAL_BF16)
+        dd = adapter.delta("down", act)
+        if dd is not None:
+            h3 = h3 + dd.to(h3.dtype)
+        return h3
 
     # --------------------------------------------------- training (autograd)
 
@@ -267,22 +304,7 @@ class LlamaFastPath:
                 o = o + d
         h2 = hidden + o
         xn2 = reference.rms_norm(h2, self.ln2_w, self.eps)
-        gateup = torch.matmul(xn2, self.wgateup_t.dense())
-        inter = self.wgateup_t.shape[1] // 2
-        gate, up = gateup[..., :inter], gateup[..., inter:]
-        if adapter is not None:
-            dg, du = adapter.delta("gate", xn2), adapter.delta("up", xn2)
-            if dg is not None:
-                gate = gate + dg
-            if du is not None:
-                up = up + du
-        act = reference.swiglu(gate, up).to(hidden.dtype)
-        down = torch.matmul(act, self.wdown_t.dense())
-        if adapter is not None:
-            dd = adapter.delta("down", act)
-            if dd is not None:
-                down = down + dd
-        return h2 + down
+        return h2 + self._mlp_dense(xn2, adapter, autograd=True).to(h2.dtype)
 
     # ------------------------------------------------------------ prefill
 
@@ -324,19 +346,4 @@ class LlamaFastPath:
                 o = o + d
         h2 = hidden + o
         xn2 = self.hip.rms_norm(h2, self.ln2_w, self.eps)
-        gateup = torch.matmul(xn2, self.wgateup_t.dense())
-        inter = self.wgateup_t.shape[1] // 2
-        gate, up = gateup[..., :inter].contiguous(), gateup[..., inter:].contiguous()
-        if adapter is not None:
-            dg, du = adapter.delta("gate", xn2), adapter.delta("up", xn2)
-            if dg is not None:
-                gate = gate + dg
-            if du is not None:
-                up = up + du
-        act = self.hip.swiglu(gate.contiguous(), up.contiguous())
-        down = torch.matmul(act, self.wdown_t.dense())
-        if adapter is not None:
-            dd = adapter.delta("down", act)
-            if dd is not None:
-                down = down + dd
-        return h2 + down
+        return h2 + self._mlp_dense(xn2, adapter, autograd=False).to(h2.dtype)

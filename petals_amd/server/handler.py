@@ -306,7 +306,11 @@ class TransformerConnectionHandler:
         # fast serving path: whole-span hipGraph decode (1 new token, GPU,
         # every block on the fused path). Beam reorder runs eagerly before the
         # replay; prompts force the eager path (they change per step).
-        all_fast = all(getattr(self.backends[uid].block, "_fast", None) is not None for uid in uids)
+        all_fast = all(
+            getattr(self.backends[uid].block, "_fast", None) is not None
+            and getattr(self.backends[uid].block._fast, "graph_safe", False)
+            for uid in uids
+        )
         max_graph_batch = 1
         if all_fast:
             quants = {self.backends[uid].block._fast.quant for uid in uids}
