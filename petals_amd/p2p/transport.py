@@ -280,6 +280,11 @@ class P2PNode:
         return self.listen_addr
 
     async def _on_inbound(self, reader: asyncio.StreamReader, writer: asyncio.StreamWriter):
+        sock = writer.get_extra_info("socket")
+        if sock is not None:
+            # without NODELAY on the accepted socket, small RPC responses sit in
+            # Nagle's buffer against the peer's delayed ACK (~40 ms per step)
+            sock.setsockopt(socket.IPPROTO_TCP, socket.TCP_NODELAY, 1)
         conn = Connection(self, reader, writer)
         self._inbound.append(conn)
         conn.start()
