@@ -40,6 +40,9 @@ def main(argv=None):
     parser.add_argument("--balance_quality", type=float, default=0.75)
     parser.add_argument("--mean_balance_check_period", type=float, default=120.0)
     parser.add_argument("--public_name", type=str, default=None)
+    parser.add_argument("--adapters", nargs="*", default=[], help="local PEFT adapter dirs to serve")
+    parser.add_argument("--announce_host", type=str, default=None)
+    parser.add_argument("--skip_reachability_check", action="store_true")
     args = parser.parse_args(argv)
 
     if args.config:
@@ -74,6 +77,9 @@ def main(argv=None):
         mean_balance_check_period=args.mean_balance_check_period,
         quant_type=args.quant_type,
         public_name=args.public_name,
+        adapters=args.adapters,
+        announce_host=args.announce_host,
+        skip_reachability_check=args.skip_reachability_check,
     )
     server.start()
     print(f"petals_amd server listening on {server.listen_addr} peer_id={server.peer_id}", flush=True)
