@@ -1,12 +1,13 @@
-"""Full-stack serving measurement: a REAL server (DHT + TCP RPC + scheduler +
-per-session span hipGraph) + a thin client on the same box — the end-to-end
-number users of the framework get, including wire serialization.
+"""Full-stack serving measurement: a REAL server process (DHT + TCP RPC +
+scheduler + per-session span hipGraph) + a thin client process — the
+end-to-end number users of the framework get, including wire serialization.
 
 Usage: python scripts/serve_bench.py [--model llama-2-7b] [--new-tokens 64]
 """
 
 import argparse
 import os
+import subprocess
 import sys
 import time
 
@@ -21,20 +22,39 @@ def main():
     p.add_argument("--new-tokens", type=int, default=64)
     p.add_argument("--prompt-len", type=int, default=32)
     p.add_argument("--quant", default="none")
+    p.add_argument("--in-process", action="store_true", help="run the server in this process instead")
     args = p.parse_args()
 
     from petals_amd.dht.node import DHT
-    from petals_amd.server.server import Server
     from petals_amd.utils.auto_config import AutoDistributedModelForCausalLM
 
     boot = DHT(host="127.0.0.1")
     t0 = time.time()
-    server = Server(
-        args.model, initial_peers=[boot.listen_addr], host="127.0.0.1",
-        device="cuda", torch_dtype="bfloat16", dht_prefix="serve-bench",
-        throughput=1000.0, quant_type=args.quant,
-    ).start()
-    print(f"server up in {time.time()-t0:.1f}s, blocks={server.num_blocks}", flush=True)
+    server = proc = None
+    if args.in_process:
+        from petals_amd.server.server import Server
+
+        server = Server(
+            args.model, initial_peers=[boot.listen_addr], host="127.0.0.1",
+            device="cuda", torch_dtype="bfloat16", dht_prefix="serve-bench",
+            throughput=1000.0, quant_type=args.quant,
+        ).start()
+    else:
+        proc = subprocess.Popen(
+            [sys.executable, "-m", "petals_amd.cli.run_server", args.model,
+             "--host", "127.0.0.1", "--initial_peers", f"127.0.0.1:{boot.listen_addr[1]}",
+             "--torch_dtype", "bfloat16", "--dht_prefix", "serve-bench",
+             "--throughput", "1000", "--quant_type", args.quant],
+            cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+            stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True,
+        )
+        while True:
+            line = proc.stdout.readline()
+            if "listening on" in line:
+                break
+            if proc.poll() is not None:
+                raise RuntimeError("server died during startup")
+    print(f"server up in {time.time()-t0:.1f}s", flush=True)
 
     model = AutoDistributedModelForCausalLM.from_pretrained(
         args.model, initial_peers=[boot.listen_addr], dht_prefix="serve-bench",
@@ -49,25 +69,25 @@ def main():
         _run(model, ids, args)
     finally:
         model.transformer.h.sequence_manager.shutdown()
-        server.shutdown()
+        if server is not None:
+            server.shutdown()
+        if proc is not None:
+            proc.terminate()
         boot.shutdown()
         os._exit(0)  # skip interpreter-teardown races between HIP and daemon threads
 
 
 def _run(model, ids, args):
-    # --- isolate raw session step latency (client + wire + server compute),
-    # no embeddings/head/generate logic
+    # --- isolate raw session step latency (client + wire + server compute)
     H = model.config.hidden_size
     dev = model.transformer.embed_tokens.weight.device
     h = torch.randn(1, 1, H, device=dev, dtype=model.transformer.embed_tokens.weight.dtype) * 0.02
     with model.transformer.h.inference_session(max_length=64) as sess:
         sess.step(h)  # session open + first step (graph capture server-side)
-        import time as _t
-
-        t0 = _t.perf_counter()
+        t0 = time.perf_counter()
         for _ in range(20):
             sess.step(h)
-        dt = (_t.perf_counter() - t0) / 20
+        dt = (time.perf_counter() - t0) / 20
         print(f"raw session step: {dt*1000:.2f} ms/token (client+wire+server)", flush=True)
 
     # warmup generation (includes session setup + graph capture)
@@ -79,7 +99,8 @@ def _run(model, ids, args):
     elapsed = time.perf_counter() - t0
     tps = args.new_tokens / elapsed
     print(f"FULL-STACK serving: {tps:.2f} tokens/s ({elapsed/args.new_tokens*1000:.1f} ms/token) "
-          f"model={args.model} quant={args.quant} [includes TCP wire + client embeds/head on GPU]", flush=True)
+          f"model={args.model} quant={args.quant} "
+          f"[{'in-process' if args.in_process else 'separate server process'}]", flush=True)
 
 
 if __name__ == "__main__":
