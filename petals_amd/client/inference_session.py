@@ -278,14 +278,26 @@ class InferenceSession:
         }
         tensors = [span_inputs, span_prompts, hypo_ids]
 
+        import os as _os
+        import time as _time
+
+        _trace = _os.environ.get("PETALS_AMD_STEP_TRACE")
+        _t0 = _time.perf_counter()
+
         async def roundtrip():
+            _ts = _time.perf_counter()
             await session._asend_step(meta, tensors)
+            _tsent = _time.perf_counter()
             while True:
                 msg = await session._arecv_step(self._manager.config.request_timeout)
                 if msg.meta.get("step_id") == step_id and msg.tensors:
+                    if _trace:
+                        print(f"[cli] in-loop send {( _tsent-_ts)*1e3:.2f} recv {(_time.perf_counter()-_tsent)*1e3:.2f} ms", flush=True)
                     return msg.tensors[0]
 
         out = self._manager.run_coroutine(roundtrip(), timeout=self._manager.config.request_timeout + 10)
+        if _trace:
+            print(f"[cli] step total {(_time.perf_counter()-_t0)*1e3:.2f} ms (incl thread hop)", flush=True)
         assert out.shape == span_inputs.shape, f"{out.shape} vs {span_inputs.shape}"
         # bookkeeping
         base = session.history[:, : session.position] if session.history is not None else span_inputs[:, :0]

@@ -14,6 +14,7 @@ from __future__ import annotations
 import asyncio
 import contextlib
 import logging
+import os
 import time
 from typing import Any, Dict, List, Optional, Sequence, Tuple
 
@@ -395,6 +396,8 @@ class TransformerConnectionHandler:
                             f"max_length exceeded: prefix {session.prefix_length} + {length_increment} > {max_length}"
                         )
                     step_start_position = session.prefix_length
+                    _trace = os.environ.get("PETALS_AMD_STEP_TRACE")
+                    _t0 = time.perf_counter()
                     if length_increment > 0:
                         priority = self.prioritizer.prioritize(hidden_states, type="inference")
                         output = await self.runtime.submit(
@@ -411,6 +414,8 @@ class TransformerConnectionHandler:
                         )
                     else:
                         output = hidden_states
+                    if _trace:
+                        print(f"[srv] compute {(time.perf_counter()-_t0)*1e3:.2f} ms", flush=True)
                     session.prefix_length += length_increment
 
                     has_prompts = prompts is not None and not is_dummy(prompts)
