@@ -54,6 +54,11 @@ def main():
                 break
             if proc.poll() is not None:
                 raise RuntimeError("server died during startup")
+        # drain further server output in a thread: a full pipe buffer would
+        # BLOCK the server on its own log writes
+        import threading
+
+        threading.Thread(target=lambda: [None for _ in proc.stdout], daemon=True).start()
     print(f"server up in {time.time()-t0:.1f}s", flush=True)
 
     model = AutoDistributedModelForCausalLM.from_pretrained(
