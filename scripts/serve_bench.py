@@ -47,6 +47,7 @@ def main():
              "--throughput", "1000", "--quant_type", args.quant],
             cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
             stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True,
+            env={**os.environ},
         )
         while True:
             line = proc.stdout.readline()
@@ -58,7 +59,13 @@ def main():
         # BLOCK the server on its own log writes
         import threading
 
-        threading.Thread(target=lambda: [None for _ in proc.stdout], daemon=True).start()
+        def _drain():
+            with open("gpurun_out/server_sub.log", "w") as f:
+                for line in proc.stdout:
+                    f.write(line)
+
+        os.makedirs("gpurun_out", exist_ok=True)
+        threading.Thread(target=_drain, daemon=True).start()
     print(f"server up in {time.time()-t0:.1f}s", flush=True)
 
     model = AutoDistributedModelForCausalLM.from_pretrained(
