@@ -22,7 +22,9 @@ def main():
     p.add_argument("--new-tokens", type=int, default=64)
     p.add_argument("--prompt-len", type=int, default=32)
     p.add_argument("--quant", default="none")
-    p.add_argument("--in-process", action="store_true", help="run the server in this process instead")
+    p.add_argument("--subprocess", action="store_true",
+                   help="server in a separate process (NB: two processes sharing one GPU "
+                        "context-switch on every hop; use for isolation tests only)")
     args = p.parse_args()
 
     from petals_amd.dht.node import DHT
@@ -31,7 +33,7 @@ def main():
     boot = DHT(host="127.0.0.1")
     t0 = time.time()
     server = proc = None
-    if args.in_process:
+    if not args.subprocess:
         from petals_amd.server.server import Server
 
         server = Server(
@@ -112,7 +114,7 @@ def _run(model, ids, args):
     tps = args.new_tokens / elapsed
     print(f"FULL-STACK serving: {tps:.2f} tokens/s ({elapsed/args.new_tokens*1000:.1f} ms/token) "
           f"model={args.model} quant={args.quant} "
-          f"[{'in-process' if args.in_process else 'separate server process'}]", flush=True)
+          f"[{'separate server process' if args.subprocess else 'co-located (one GPU context)'}]", flush=True)
 
 
 if __name__ == "__main__":
