@@ -77,9 +77,26 @@ def apply_rope(q, k, cos, sin, position_ids):
 
 
 def attention(q, k, v, *, causal: bool, kv_offset: int = 0, attn_bias=None, scale=None):
-    # prefill attention currently runs the fp32-softmax torch composition
-    # (chunked by the backend); a flash-style HIP prefill kernel is the
-    # planned replacement. Decode attention uses attn_decode_fused.
+    """Prefill attention. On GPU (bf16, head_dim 64/128, no additive bias) this
+    runs the MFMA flash kernel; otherwise the fp32-softmax torch composition."""
+    if (
+        q.is_cuda
+        and not _grad_mode(q, k, v)
+        and attn_bias is None
+        and q.dtype == torch.bfloat16
+        and k.dtype == torch.bfloat16
+        and q.shape[-1] in (64, 128)
+        and q.shape[1] % k.shape[1] == 0
+    ):
+        hops = _require_hip("attention_prefill")
+        if hops is not None:
+            import math
+
+            kv_len = k.shape[2]
+            return hops.attn_prefill_fused(
+                q.contiguous(), k.contiguous(), v.contiguous(), kv_len, int(kv_offset),
+                float(scale) if scale is not None else 1.0 / math.sqrt(q.shape[-1]), bool(causal),
+            )
     return reference.attention(q, k, v, causal=causal, kv_offset=kv_offset, attn_bias=attn_bias, scale=scale)
 
 
@@ -112,9 +129,29 @@ def attention_decode(q, k_cache, v_cache, kv_len: int, *, attn_bias=None, scale=
                     float(scale) if scale is not None else 1.0 / math.sqrt(hd),
                 )
                 return out.view(b, 1, n_heads, hd).permute(0, 2, 1, 3).to(q.dtype)
+    q_len = q.shape[2]
+    if (
+        q.is_cuda
+        and q_len > 1
+        and not _grad_mode(q)
+        and attn_bias is None
+        and q.dtype == torch.bfloat16
+        and k_cache.dtype == torch.bfloat16
+        and q.shape[-1] in (64, 128)
+        and q.shape[1] % k_cache.shape[1] == 0
+    ):
+        # multi-token (prefill/chunk) step over the cache: MFMA flash kernel
+        hops = _require_hip("attention_prefill")
+        if hops is not None:
+            import math
+
+            return hops.attn_prefill_fused(
+                q.contiguous(), k_cache.contiguous(), v_cache.contiguous(), int(kv_len),
+                int(kv_len) - q_len,
+                float(scale) if scale is not None else 1.0 / math.sqrt(q.shape[-1]), True,
+            )
     k = k_cache[:, :, :kv_len]
     v = v_cache[:, :, :kv_len]
-    q_len = q.shape[2]
     return reference.attention(
         q, k, v, causal=q_len > 1, kv_offset=kv_len - q_len, attn_bias=attn_bias, scale=scale
     )

@@ -16,6 +16,9 @@ torch::Tensor gemv_bf16(
 torch::Tensor attn_decode_fused(
     torch::Tensor q, torch::Tensor k_cache, torch::Tensor v_cache, torch::Tensor kv_len,
     int64_t gq, int64_t n_splits, torch::Tensor part_o, torch::Tensor part_ml, double scale);
+torch::Tensor attn_prefill_fused(
+    torch::Tensor q, torch::Tensor k, torch::Tensor v,
+    int64_t kv_len, int64_t kv_offset, double scale, bool causal);
 std::vector<torch::Tensor> nf4_quantize(torch::Tensor w);
 torch::Tensor nf4_dequantize(torch::Tensor packed, torch::Tensor absmax);
 torch::Tensor gemv_nf4(
@@ -31,6 +34,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemv_bf16", &gemv_bf16, "split-K bf16 gemv with fused epilogue",
         py::arg("wt"), py::arg("x"), py::arg("workspace"), py::arg("residual"), py::arg("epilogue"), py::arg("splits") = 0);
   m.def("attn_decode_fused", &attn_decode_fused, "GQA decode attention (flash-decoding)");
+  m.def("attn_prefill_fused", &attn_prefill_fused, "MFMA flash prefill attention (bf16, causal, GQA)");
   m.def("nf4_quantize", &nf4_quantize, "blockwise NF4 quantize [in,out] bf16 -> (packed u8, absmax bf16)");
   m.def("nf4_dequantize", &nf4_dequantize, "NF4 -> bf16");
   m.def("gemv_nf4", &gemv_nf4, "split-K NF4 gemv (dequant fused) with epilogue",

@@ -333,11 +333,14 @@ class LlamaFastPath:
             k_cache, v_cache = kv_cache
             k_cache[:B, :, prefix_length:end].copy_(k)
             v_cache[:B, :, prefix_length:end].copy_(v)
-            attn = reference.attention(
-                q, k_cache[:B, :, :end], v_cache[:B, :, :end], causal=S > 1, kv_offset=prefix_length
+            attn = self.hip.attn_prefill_fused(
+                q.contiguous(), k_cache[:B].contiguous(), v_cache[:B].contiguous(),
+                end, prefix_length, self.scale, True,
             )
         else:
-            attn = reference.attention(q, k, v, causal=True)
+            attn = self.hip.attn_prefill_fused(
+                q.contiguous(), k.contiguous(), v.contiguous(), S, 0, self.scale, True
+            )
         attn = attn.transpose(1, 2).reshape(B, S, self.qh * self.hd).to(torch.bfloat16)
         o = torch.matmul(attn, self.wo_t.dense())
         if adapter is not None:

@@ -234,3 +234,33 @@ def test_mixtral_block_fast_decode_matches_cpu(hip):
     out = torch.cat([y.float().cpu() for y in y_gpu], 1)
     # bf16 routing can pick different experts on near-ties; require close overall
     assert torch.allclose(out, ref, atol=0.08, rtol=0.08), (out - ref).abs().max()
+
+
+@requires_gpu
+@pytest.mark.parametrize("case", [
+    dict(b=2, qh=8, kvh=2, s=67, hd=128, off=0, causal=True),
+    dict(b=1, qh=4, kvh=4, s=200, hd=128, off=0, causal=True),
+    dict(b=1, qh=8, kvh=8, s=33, hd=64, off=50, causal=True),   # chunk over a cache prefix
+    dict(b=2, qh=4, kvh=1, s=64, hd=128, off=0, causal=False),
+])
+def test_attn_prefill_mfma(hip, case):
+    """MFMA flash prefill vs fp32-softmax reference."""
+    from petals_amd.ops import reference
+
+    torch.manual_seed(9)
+    b, qh, kvh, s, hd, off = case["b"], case["qh"], case["kvh"], case["s"], case["hd"], case["off"]
+    kv_len = off + s
+    lmax = kv_len + 16
+    q = (torch.randn(b, qh, s, hd, device="cuda") * 0.5).to(torch.bfloat16)
+    k = torch.zeros(b, kvh, lmax, hd, device="cuda", dtype=torch.bfloat16)
+    v = torch.zeros_like(k)
+    k[:, :, :kv_len] = (torch.randn(b, kvh, kv_len, hd, device="cuda") * 0.5).to(torch.bfloat16)
+    v[:, :, :kv_len] = (torch.randn(b, kvh, kv_len, hd, device="cuda") * 0.5).to(torch.bfloat16)
+    import math
+    out = hip.attn_prefill_fused(q, k, v, kv_len, off, 1.0 / math.sqrt(hd), case["causal"])
+    ref = reference.attention(
+        q.float().cpu(), k[:, :, :kv_len].float().cpu(), v[:, :, :kv_len].float().cpu(),
+        causal=case["causal"], kv_offset=off,
+    )
+    err = (out.float().cpu() - ref).abs().max()
+    assert torch.allclose(out.float().cpu(), ref, atol=3e-2, rtol=3e-2), err
