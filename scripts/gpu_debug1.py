@@ -120,3 +120,21 @@ if __name__ == "__main__":
     attn_bench()
     layer_decode_bench()
     nf4_bench()
+    prefill_attn_bench()
+
+
+def prefill_attn_bench():
+    import math
+    from petals_amd.ops import reference
+
+    for s in (512, 2048, 4096):
+        b, qh, kvh, hd = 1, 64, 8, 128  # llama-2-70b shape
+        q = (torch.randn(b, qh, s, hd, device="cuda") * 0.3).to(torch.bfloat16)
+        k = (torch.randn(b, kvh, s, hd, device="cuda") * 0.3).to(torch.bfloat16)
+        v = k.clone()
+        sc = 1.0 / math.sqrt(hd)
+        t_mfma = bench_kernel(lambda: hip.attn_prefill_fused(q, k, v, s, 0, sc, True), n=20)
+        t_ref = bench_kernel(lambda: reference.attention(q, k, v, causal=True), n=5)
+        flops = 4 * b * qh * s * s * hd / 2  # causal half
+        print(f"prefill attn S={s}: mfma {t_mfma*1e3:.2f} ms ({flops/t_mfma/1e12:.0f} TF) "
+              f"vs matmul+softmax {t_ref*1e3:.2f} ms", flush=True)
