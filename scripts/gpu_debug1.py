@@ -113,14 +113,6 @@ def nf4_bench():
         torch.cuda.empty_cache()
 
 
-if __name__ == "__main__":
-    attn_debug()
-    rmsnorm_bench()
-    gemv_bench()
-    attn_bench()
-    layer_decode_bench()
-    nf4_bench()
-    prefill_attn_bench()
 
 
 def prefill_attn_bench():
@@ -138,3 +130,13 @@ def prefill_attn_bench():
         flops = 4 * b * qh * s * s * hd / 2  # causal half
         print(f"prefill attn S={s}: mfma {t_mfma*1e3:.2f} ms ({flops/t_mfma/1e12:.0f} TF) "
               f"vs matmul+softmax {t_ref*1e3:.2f} ms", flush=True)
+
+
+if __name__ == "__main__":
+    attn_debug()
+    rmsnorm_bench()
+    gemv_bench()
+    attn_bench()
+    layer_decode_bench()
+    nf4_bench()
+    prefill_attn_bench()
