@@ -16,6 +16,23 @@ enum GemvEpilogue : int {
   EPI_SWIGLU_F32 = 3,    // out = silu(sum[:half]) * sum[half:], y_f32[b, half]
 };
 
+// deterministic split sum with 4 independent accumulators so the loads
+// pipeline (the naive single-chain version was latency-bound and cost as much
+// as the gemv itself on small projections)
+static __device__ __forceinline__ float reduce_splits(
+    const float* __restrict__ partials, int n_splits, size_t stride, size_t off) {
+  float s0 = 0.f, s1 = 0.f, s2 = 0.f, s3 = 0.f;
+  int s = 0;
+  for (; s + 4 <= n_splits; s += 4) {
+    s0 += partials[(size_t)(s + 0) * stride + off];
+    s1 += partials[(size_t)(s + 1) * stride + off];
+    s2 += partials[(size_t)(s + 2) * stride + off];
+    s3 += partials[(size_t)(s + 3) * stride + off];
+  }
+  for (; s < n_splits; ++s) s0 += partials[(size_t)s * stride + off];
+  return (s0 + s1) + (s2 + s3);
+}
+
 static __global__ void gemv_reduce_kernel_impl(
     const float* __restrict__ partials,  // [n_splits, batch, out]
     const unsigned short* __restrict__ residual,  // [batch, out] or null
@@ -27,29 +44,26 @@ static __global__ void gemv_reduce_kernel_impl(
   const int half = out_dim >> 1;
   const int n_out = (epilogue == EPI_SWIGLU_F32) ? half : out_dim;
   const int total = batch * n_out;
-  for (int idx = blockIdx.x * blockDim.x + threadIdx.x; idx < total; idx += gridDim.x * blockDim.x) {
-    const int b = idx / n_out;
-    const int o = idx - b * n_out;
-    if (epilogue == EPI_SWIGLU_F32) {
-      float g = 0.f, u = 0.f;
-      for (int s = 0; s < n_splits; ++s) {
-        const float* base = partials + ((size_t)s * batch + b) * out_dim;
-        g += base[o];
-        u += base[o + half];
-      }
-      const float act = g / (1.f + __expf(-g)) * u;
-      reinterpret_cast<float*>(y)[(size_t)b * half + o] = act;
-    } else {
-      float sum = 0.f;
-      for (int s = 0; s < n_splits; ++s) sum += partials[((size_t)s * batch + b) * out_dim + o];
-      if (epilogue == EPI_PLAIN_F32) {
-        reinterpret_cast<float*>(y)[(size_t)b * out_dim + o] = sum;
-      } else if (epilogue == EPI_PLAIN_BF16) {
-        reinterpret_cast<unsigned short*>(y)[(size_t)b * out_dim + o] = f32_to_bf16(sum);
-      } else {  // EPI_RESIDUAL_BF16
-        const float r = bf16_to_f32(residual[(size_t)b * out_dim + o]);
-        reinterpret_cast<unsigned short*>(y)[(size_t)b * out_dim + o] = f32_to_bf16(r + sum);
-      }
+  const size_t stride = (size_t)batch * out_dim;
+  const int idx = blockIdx.x * blockDim.x + threadIdx.x;
+  if (idx >= total) return;
+  const int b = idx / n_out;
+  const int o = idx - b * n_out;
+  const size_t row = (size_t)b * out_dim;
+  if (epilogue == EPI_SWIGLU_F32) {
+    const float g = reduce_splits(partials, n_splits, stride, row + o);
+    const float u = reduce_splits(partials, n_splits, stride, row + o + half);
+    const float act = g / (1.f + __expf(-g)) * u;
+    reinterpret_cast<float*>(y)[(size_t)b * half + o] = act;
+  } else {
+    const float sum = reduce_splits(partials, n_splits, stride, row + o);
+    if (epilogue == EPI_PLAIN_F32) {
+      reinterpret_cast<float*>(y)[row + o] = sum;
+    } else if (epilogue == EPI_PLAIN_BF16) {
+      reinterpret_cast<unsigned short*>(y)[row + o] = f32_to_bf16(sum);
+    } else {  // EPI_RESIDUAL_BF16
+      const float r = bf16_to_f32(residual[row + o]);
+      reinterpret_cast<unsigned short*>(y)[row + o] = f32_to_bf16(r + sum);
     }
   }
 }
@@ -77,9 +91,10 @@ static inline torch::Tensor launch_gemv_reduce(
     res_p = reinterpret_cast<const unsigned short*>(residual->data_ptr());
   }
   const int total = batch * n_out;
-  int rblocks = std::min((total + 255) / 256, 2048);
+  const int rthreads = 64;  // small blocks -> enough workgroups to spread CUs
+  int rblocks = (total + rthreads - 1) / rthreads;
   auto stream = c10::hip::getCurrentHIPStream().stream();
-  gemv_reduce_kernel_impl<<<rblocks, 256, 0, stream>>>(
+  gemv_reduce_kernel_impl<<<rblocks, rthreads, 0, stream>>>(
       partials.data_ptr<float>(), res_p, y.data_ptr(), n_splits, batch, out_dim, epilogue);
   HIP_CHECK_LAST();
   return y;
