@@ -249,7 +249,7 @@ torch::Tensor gemv_nf4(
   // NF4 matrices are 4x smaller than bf16: allow chunks down to 64 input rows
   // so small projections still spread over the 256 CUs
   long splits = splits_override > 0 ? splits_override : (1536 + out_waves - 1) / out_waves;
-  long max_splits = (in_dim + 63) / 64;
+  long max_splits = (in_dim + 31) / 32;
   if (splits > max_splits) splits = max_splits;
   if (splits < 1) splits = 1;
 

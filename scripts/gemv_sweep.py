@@ -47,5 +47,28 @@ def main():
         torch.cuda.empty_cache()
 
 
+def nf4_sweep():
+    ws = torch.empty(512 * 57344, dtype=torch.float32, device="cuda")
+    shapes = [(8192, 10240, "qkv"), (8192, 8192, "o"), (8192, 57344, "gateup"), (28672, 8192, "down")]
+    for in_dim, out_dim, name in shapes:
+        wt = (torch.randn(in_dim, out_dim, device="cuda") * 0.02).to(torch.bfloat16)
+        packed, absmax = hip.nf4_quantize(wt)
+        del wt
+        x = torch.randn(1, in_dim, device="cuda")
+        gb = (packed.numel() + absmax.numel() * 2) / 1e9
+        row, best = [], (0, 0)
+        for splits in (0, 16, 32, 64, 128, 192, 256, 448):
+            if splits * 16 > in_dim and splits:
+                continue
+            t = bench(lambda: hip.gemv_nf4(packed, absmax, x, ws, None, 0, splits))
+            bw = gb / t
+            row.append(f"s{splits}:{bw:.0f}")
+            if bw > best[0]:
+                best = (bw, splits)
+        print(f"gemv_nf4 {name} [{in_dim}x{out_dim}] {gb*1000:.0f}MB: {' '.join(row)}  BEST s{best[1]} {best[0]:.0f} GB/s", flush=True)
+        torch.cuda.empty_cache()
+
+
 if __name__ == "__main__":
     main()
+    nf4_sweep()
