@@ -66,8 +66,15 @@ class DecodeContext:
         self.kv_len.add_(1)
 
 
+_split_autotune_cache: Dict[Tuple[str, int, int], int] = {}
+
+
 class _FastWeight:
-    """A transposed [in, out] weight in bf16 or NF4 (quantize-on-load)."""
+    """A transposed [in, out] weight in bf16 or NF4 (quantize-on-load).
+
+    Split-K counts are autotuned once per (kind, in, out) shape at load time
+    (profiles/gemv_split_sweep.log shows the optimum varies ~2x by shape);
+    the result is cached process-wide so only the first block pays."""
 
     def __init__(self, t_bf16: torch.Tensor, hip, quant: str):
         self.hip = hip
@@ -79,15 +86,53 @@ class _FastWeight:
         else:
             self.t = t_bf16.contiguous()
             self.packed = self.absmax = None
+        self.splits = self._autotune(t_bf16.device)
+
+    def _autotune(self, device) -> int:
+        key = (self.quant, self.in_dim, self.out_dim)
+        if key in _split_autotune_cache:
+            return _split_autotune_cache[key]
+        if device.type != "cuda":
+            return 0
+        import time
+
+        if self.quant == "nf4":
+            candidates = [0, 64, 128, 192, 256]
+            max_chunk = 32
+        else:
+            candidates = [0, 8, 16, 32, 64]
+            max_chunk = 64
+        candidates = [s for s in candidates if s == 0 or s * max_chunk <= self.in_dim]
+        x = torch.randn(1, self.in_dim, device=device)
+        ws = _get_ws(device, "gemv", 64 * max(self.out_dim, 1))
+        best, best_t = 0, float("inf")
+        for s in candidates:
+            try:
+                self._gemv_raw(x, ws, None, 0, s)  # warm
+                torch.cuda.synchronize()
+                t0 = time.perf_counter()
+                for _ in range(8):
+                    self._gemv_raw(x, ws, None, 0, s)
+                torch.cuda.synchronize()
+                dt = time.perf_counter() - t0
+            except RuntimeError:
+                continue
+            if dt < best_t:
+                best, best_t = s, dt
+        _split_autotune_cache[key] = best
+        return best
 
     @property
     def shape(self):
         return (self.in_dim, self.out_dim)
 
-    def gemv(self, x, ws, residual, epilogue):
+    def _gemv_raw(self, x, ws, residual, epilogue, splits):
         if self.quant == "nf4":
-            return self.hip.gemv_nf4(self.packed, self.absmax, x, ws, residual, epilogue)
-        return self.hip.gemv_bf16(self.t, x, ws, residual, epilogue)
+            return self.hip.gemv_nf4(self.packed, self.absmax, x, ws, residual, epilogue, splits)
+        return self.hip.gemv_bf16(self.t, x, ws, residual, epilogue, splits)
+
+    def gemv(self, x, ws, residual, epilogue):
+        return self._gemv_raw(x, ws, residual, epilogue, self.splits)
 
     def dense(self) -> torch.Tensor:
         """bf16 [in, out] view for prefill GEMMs (dequantized on the fly for NF4)."""
