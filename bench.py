@@ -218,10 +218,18 @@ def main():
     else:
         cur = torch.zeros(B, 1, dtype=torch.long, device=device)
 
-    # --- capture decode graphs (GPU) or stay eager (CPU)
-    use_graphs = use_cuda and not os.environ.get("PETALS_AMD_NO_GRAPHS")
+    # --- capture decode graphs (GPU) or stay eager (CPU / MoE / capture failure)
+    graph_safe = all(getattr(getattr(blk, "_fast", None), "graph_safe", False) for blk in blocks)
+    use_graphs = use_cuda and graph_safe and not os.environ.get("PETALS_AMD_NO_GRAPHS")
     if use_graphs:
-        graph_state.update(build_graphs(cur if rank == 0 else torch.zeros(B, 1, dtype=torch.long, device=device)))
+        try:
+            graph_state.update(
+                build_graphs(cur if rank == 0 else torch.zeros(B, 1, dtype=torch.long, device=device))
+            )
+        except Exception as e:  # noqa: BLE001
+            log(f"[bench] graph capture failed ({e!r}); falling back to eager decode")
+            use_graphs = False
+    if use_graphs:
         step_fn = one_token_graphed
     else:
         def step_fn(p):
