@@ -50,7 +50,7 @@ __global__ void gemv_bf16_kernel(
   if (full) {
     // UNROLL x 16B weight loads in flight per wave: HBM latency (~900 cyc)
     // needs many outstanding loads; the scalar x loads are wave-uniform
-    constexpr int UNROLL = 8;
+    constexpr int UNROLL = 16;  // 256 B of weight loads in flight per wave
     const unsigned short* wp = wt + (size_t)i_begin * out_dim + out0;
     int i = i_begin;
     for (; i + UNROLL <= i_end; i += UNROLL) {

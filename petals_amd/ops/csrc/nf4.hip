@@ -118,7 +118,7 @@ __global__ void gemv_nf4_kernel(
     for (int v = 0; v < 16; ++v) acc[b][v] = 0.f;
 
   if (full) {
-    constexpr int UNROLL = 8;
+    constexpr int UNROLL = 16;  // 128 B of packed loads in flight per wave
     const int half_out = out_dim >> 1;
     const unsigned char* pp = packed + (size_t)i_begin * half_out + (out0 >> 1);
     int i = i_begin;
