@@ -100,7 +100,7 @@ class _FastWeight:
             candidates = [0, 64, 128, 192, 256]
             max_chunk = 32
         else:
-            candidates = [0, 8, 16, 32, 64]
+            candidates = [0, 8, 16, 32, 64, 128]
             max_chunk = 64
         candidates = [s for s in candidates if s == 0 or s * max_chunk <= self.in_dim]
         x = torch.randn(1, self.in_dim, device=device)
