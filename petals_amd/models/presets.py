@@ -89,6 +89,11 @@ PRESETS = {
         num_local_experts=8, num_experts_per_tok=2,
     ),
     # ------------------------------------------------ tiny test configs
+    "test-llama-hd128": dict(
+        model_type="llama", hidden_size=512, num_hidden_layers=4, num_attention_heads=4,
+        num_key_value_heads=2, intermediate_size=1024, vocab_size=128,
+        max_position_embeddings=256, rms_norm_eps=1e-5, rope_theta=10000.0,
+    ),
     "test-llama": dict(
         model_type="llama", hidden_size=64, num_hidden_layers=4, num_attention_heads=4,
         num_key_value_heads=2, intermediate_size=128, vocab_size=128,

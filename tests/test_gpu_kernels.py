@@ -264,3 +264,69 @@ def test_attn_prefill_mfma(hip, case):
     )
     err = (out.float().cpu() - ref).abs().max()
     assert torch.allclose(out.float().cpu(), ref, atol=3e-2, rtol=3e-2), err
+
+
+@requires_gpu
+def test_gpu_speculative_and_lora_e2e(tmp_path):
+    """GPU server with the fused fast path + span graphs: speculative rollback
+    (start_from_position) and a LoRA adapter, end to end."""
+    import json
+
+    import torch as T
+    from safetensors.torch import save_file
+
+    from petals_amd.dht.node import DHT
+    from petals_amd.models.config_base import load_model_config
+    from petals_amd.models.llama.speculative_model import DistributedLlamaForSpeculativeGeneration
+    from petals_amd.server.server import Server
+    from petals_amd.utils.auto_config import AutoDistributedModelForCausalLM
+
+    cfg = load_model_config("test-llama-hd128")
+    # synthetic LoRA adapter for every block
+    ad_dir = tmp_path / "gpu-adapter"
+    ad_dir.mkdir()
+    T.manual_seed(11)
+    tensors = {}
+    for i in range(cfg.num_blocks):
+        tensors[f"base_model.model.model.layers.{i}.self_attn.q_proj.lora_A.weight"] = T.randn(4, 512) * 0.03
+        tensors[f"base_model.model.model.layers.{i}.self_attn.q_proj.lora_B.weight"] = T.randn(512, 4) * 0.03
+    with open(ad_dir / "adapter_config.json", "w") as f:
+        json.dump({"r": 4, "lora_alpha": 8}, f)
+    save_file(tensors, str(ad_dir / "adapter_model.safetensors"))
+
+    boot = DHT(host="127.0.0.1")
+    server = Server(
+        "test-llama-hd128", initial_peers=[boot.listen_addr], host="127.0.0.1", device="cuda",
+        torch_dtype="bfloat16", block_indices="0:4", dht_prefix="gpu-spec", throughput=1.0,
+        adapters=[str(ad_dir)],
+    ).start()
+    try:
+        model = AutoDistributedModelForCausalLM.from_pretrained(
+            "test-llama-hd128", initial_peers=[boot.listen_addr], dht_prefix="gpu-spec",
+            show_route=False, max_retries=1,
+        )
+        assert server.backends and all(b.block._fast is not None for b in server.backends.values())
+        ids = T.randint(0, 128, (1, 6))
+        ref = model.generate(ids, max_new_tokens=10, do_sample=False)
+
+        class BadDraft:
+            def __call__(self, input_ids=None, **kw):
+                class Out: ...
+                o = Out()
+                logits = T.full((input_ids.shape[0], input_ids.shape[1], 128), -10.0)
+                logits[..., 5] = 10.0
+                o.logits = logits
+                return o
+
+        spec = DistributedLlamaForSpeculativeGeneration(model, BadDraft())
+        out = spec.generate(ids, max_new_tokens=10, speculative_tokens=3)
+        assert T.equal(out, ref), (out.tolist(), ref.tolist())
+
+        # adapter changes the output
+        model.transformer.h.sequence_manager.config.active_adapter = "gpu-adapter"
+        out_lora = model.generate(ids, max_new_tokens=10, do_sample=False)
+        assert out_lora.shape == ref.shape
+        model.transformer.h.sequence_manager.shutdown()
+    finally:
+        server.shutdown()
+        boot.shutdown()
