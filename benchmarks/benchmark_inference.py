@@ -4,6 +4,10 @@ benchmarks/benchmark_inference.py — tokens/sec of session-based generate,
 multi-process clients supported via --n_processes)."""
 
 import argparse
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import multiprocessing as mp
 import time
 

@@ -2,6 +2,10 @@
 """Prompt-tuning fwd+bwd throughput (parity: reference benchmarks/benchmark_training.py)."""
 
 import argparse
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import time
 
 import torch
