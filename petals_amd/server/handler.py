@@ -225,13 +225,18 @@ class TransformerConnectionHandler:
         inputs = inputs.to(device=device, dtype=dtype)
         grad_outputs = grad_outputs.to(device=device, dtype=dtype)
         prompt_list = self._split_prompts(prompts, len(uids), dtype, device)
-        adapter_scope = using_adapter(active_adapter)
-        adapter_scope.__enter__()
-
         # re-run forward (plain no_grad, NOT inference_mode: these activations
         # feed autograd below) to recover intermediate inputs
         inter_inputs: List[torch.Tensor] = []
         hidden = inputs
+        adapter_scope = using_adapter(active_adapter)  # exception-safe via try/finally below
+        adapter_scope.__enter__()
+        try:
+            return self._backward_chain_inner(uids, inputs, grad_outputs, prompt_list, inter_inputs, hidden)
+        finally:
+            adapter_scope.__exit__(None, None, None)
+
+    def _backward_chain_inner(self, uids, inputs, grad_outputs, prompt_list, inter_inputs, hidden):
         with torch.no_grad():
             for uid, prompt in zip(uids[:-1], prompt_list[:-1]):
                 if prompt is not None:
@@ -254,7 +259,6 @@ class TransformerConnectionHandler:
                 grad_prompts.append(grad[:, : prompt.shape[1]].clone())
             else:
                 grad_prompts.append(None)
-        adapter_scope.__exit__(None, None, None)
         grad_prompts.reverse()
         if any(gp is not None for gp in grad_prompts):
             ref = next(gp for gp in grad_prompts if gp is not None)
