@@ -153,3 +153,29 @@ def test_training_forward_backward(client_model, hf_checkpoint):
     assert torch.allclose(embeds.grad, embeds_ref.grad, atol=1e-4, rtol=1e-3), (
         (embeds.grad - embeds_ref.grad).abs().max()
     )
+
+
+def test_auto_span_selection(hf_checkpoint):
+    """Servers without pinned blocks pick the least-covered span (block_selection)."""
+    from petals_amd.dht.node import DHT
+    from petals_amd.server.server import Server
+
+    path, _ = hf_checkpoint
+    boot = DHT(host="127.0.0.1")
+    s1 = Server(path, initial_peers=[boot.listen_addr], host="127.0.0.1", device="cpu",
+                torch_dtype="float32", num_blocks=2, dht_prefix="auto-span", throughput=1.0).start()
+    import time as _t
+
+    _t.sleep(0.5)
+    s2 = Server(path, initial_peers=[boot.listen_addr], host="127.0.0.1", device="cpu",
+                torch_dtype="float32", num_blocks=2, dht_prefix="auto-span", throughput=1.0).start()
+    try:
+        covered = sorted(
+            set(range(s1.server_info.start_block, s1.server_info.end_block))
+            | set(range(s2.server_info.start_block, s2.server_info.end_block))
+        )
+        assert covered == [0, 1, 2, 3], (s1.server_info, s2.server_info)
+    finally:
+        s1.shutdown()
+        s2.shutdown()
+        boot.shutdown()
