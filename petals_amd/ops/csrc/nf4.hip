@@ -94,16 +94,16 @@ __global__ void gemv_nf4_kernel(
     int in_dim,
     int out_dim,
     int i_per_split) {
-  // LUT replicated across all 32 LDS banks: lane l always reads bank (l&31)
-  // -> conflict-free random-index lookups (naive 16-float LUT was ~4-way
-  // conflicted and made LDS the bottleneck)
-  __shared__ float lut[16 * 32];
-  for (int i = threadIdx.x; i < 16 * 32; i += blockDim.x) lut[i] = NF4_LUT_C[i >> 5];
+  // PAIR LUT: one 8-byte read dequantizes a whole packed byte (two elements),
+  // halving LDS traffic vs per-nibble lookups; 256 float2 entries = 2 KiB.
+  // (the earlier 16-float bank-replicated LUT was LDS-issue bound)
+  __shared__ float2 lut2[256];
+  for (int i = threadIdx.x; i < 256; i += blockDim.x)
+    lut2[i] = make_float2(NF4_LUT_C[i & 0xF], NF4_LUT_C[i >> 4]);
   __syncthreads();
 
   const int lane = threadIdx.x & (WAVE - 1);
-  const float* lutb = lut + (lane & 31);
-#define NF4_L(nib) lutb[(nib) << 5]
+#define NF4_L2(byte) lut2[(byte)]
   const int out0 = blockIdx.x * NF4_OUT_PER_WAVE + lane * 16;
   if (out0 >= out_dim) return;
   const int split = blockIdx.y;
@@ -139,11 +139,15 @@ __global__ void gemv_nf4_kernel(
 #pragma unroll
       for (int u = 0; u < UNROLL; ++u) {
         float wf[16];
-        const unsigned int w0 = pk[u].x, w1 = pk[u].y;
+        const unsigned int wds2[2] = {pk[u].x, pk[u].y};
 #pragma unroll
-        for (int v = 0; v < 8; ++v) wf[v] = NF4_L((w0 >> (4 * v)) & 0xF);
+        for (int d = 0; d < 2; ++d)
 #pragma unroll
-        for (int v = 0; v < 8; ++v) wf[8 + v] = NF4_L((w1 >> (4 * v)) & 0xF);
+          for (int p2 = 0; p2 < 4; ++p2) {
+            const float2 w2 = NF4_L2((wds2[d] >> (8 * p2)) & 0xFFu);
+            wf[8 * d + 2 * p2] = w2.x;
+            wf[8 * d + 2 * p2 + 1] = w2.y;
+          }
 #pragma unroll
         for (int b = 0; b < BATCH; ++b) {
           const float xa = xs[b][u] * am[u];
@@ -156,13 +160,18 @@ __global__ void gemv_nf4_kernel(
     for (; i < i_end; ++i) {
       const uint2 pk = *reinterpret_cast<const uint2*>(packed + (size_t)i * half_out + (out0 >> 1));
       const float am = bf16_to_f32(absmax[(size_t)i * (out_dim >> 6) + (out0 >> 6)]);
+      const unsigned int twds[2] = {pk.x, pk.y};
 #pragma unroll
       for (int b = 0; b < BATCH; ++b) {
         const float xa = x[(size_t)b * in_dim + i] * am;
 #pragma unroll
-        for (int v = 0; v < 8; ++v) acc[b][v] = fmaf(NF4_L((pk.x >> (4 * v)) & 0xF), xa, acc[b][v]);
+        for (int d = 0; d < 2; ++d)
 #pragma unroll
-        for (int v = 0; v < 8; ++v) acc[b][8 + v] = fmaf(NF4_L((pk.y >> (4 * v)) & 0xF), xa, acc[b][8 + v]);
+          for (int p2 = 0; p2 < 4; ++p2) {
+            const float2 w2 = NF4_L2((twds[d] >> (8 * p2)) & 0xFFu);
+            acc[b][8 * d + 2 * p2] = fmaf(w2.x, xa, acc[b][8 * d + 2 * p2]);
+            acc[b][8 * d + 2 * p2 + 1] = fmaf(w2.y, xa, acc[b][8 * d + 2 * p2 + 1]);
+          }
       }
     }
   } else {
@@ -172,8 +181,8 @@ __global__ void gemv_nf4_kernel(
         const float xa = x[(size_t)b * in_dim + i] * am;
         for (int v = 0; v < out_dim - out0; ++v) {
           const unsigned char byte = packed[(size_t)i * (out_dim >> 1) + ((out0 + v) >> 1)];
-          const int nib = ((out0 + v) & 1) ? (byte >> 4) : (byte & 0xF);
-          acc[b][v] = fmaf(NF4_L(nib), xa, acc[b][v]);
+          const float2 w2 = NF4_L2(byte);
+          acc[b][v] = fmaf(((out0 + v) & 1) ? w2.y : w2.x, xa, acc[b][v]);
         }
       }
     }
