@@ -97,7 +97,7 @@ class _FastWeight:
         import time
 
         if self.quant == "nf4":
-            candidates = [0, 64, 128, 192, 256]
+            candidates = [0, 32, 64, 128, 192, 256]
             max_chunk = 32
         else:
             candidates = [0, 8, 16, 32, 64, 128]
