@@ -98,18 +98,13 @@ __global__ void gemv_nf4_kernel(
   // PAIR LUT: one 8-byte read dequantizes a whole packed byte (two elements),
   // halving LDS traffic vs per-nibble lookups; 256 float2 entries = 2 KiB.
   // (the earlier 16-float bank-replicated LUT was LDS-issue bound)
-  // two copies split the wave's random lookups across disjoint bank ranges
-  __shared__ float2 lut2[2][256];
-  for (int i = threadIdx.x; i < 256; i += blockDim.x) {
-    const float2 v = make_float2(NF4_LUT_C[i & 0xF], NF4_LUT_C[i >> 4]);
-    lut2[0][i] = v;
-    lut2[1][i] = v;
-  }
+  __shared__ float2 lut2[256];
+  for (int i = threadIdx.x; i < 256; i += blockDim.x)
+    lut2[i] = make_float2(NF4_LUT_C[i & 0xF], NF4_LUT_C[i >> 4]);
   __syncthreads();
 
   const int lane = threadIdx.x & (WAVE - 1);
-  const float2* lutb = lut2[lane & 1];
-#define NF4_L2(byte) lutb[(byte)]
+#define NF4_L2(byte) lut2[(byte)]
   const int out0 = blockIdx.x * NF4_OUT_PER_WAVE + lane * 16;
   if (out0 >= out_dim) return;
   const int split = blockIdx.y;
