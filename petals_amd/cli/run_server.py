@@ -9,7 +9,6 @@ from __future__ import annotations
 import argparse
 import logging
 import signal
-import sys
 import time
 
 

@@ -15,19 +15,17 @@ multi-token step with start_from_position, rebuilding its cache exactly.
 
 from __future__ import annotations
 
-import asyncio
 import itertools
 import logging
 import time
 import uuid
-from typing import AsyncIterator, List, Optional, Tuple
+from typing import List, Optional
 
 import torch
 
-from petals_amd.client.config import ClientConfig
 from petals_amd.client.routing.sequence_manager import RemoteSequenceManager
 from petals_amd.data_structures import CHAIN_DELIMITER, RemoteSpanInfo
-from petals_amd.p2p.transport import RpcError, RpcMessage, RpcStream
+from petals_amd.p2p.transport import RpcMessage, RpcStream
 from petals_amd.utils.misc import DUMMY, DUMMY_INT64, is_dummy
 
 logger = logging.getLogger(__name__)

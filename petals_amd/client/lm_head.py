@@ -4,7 +4,6 @@ chunked matmul fallback for low-memory CPUs)."""
 from __future__ import annotations
 
 import dataclasses
-from typing import Optional
 
 import torch
 import torch.nn.functional as F

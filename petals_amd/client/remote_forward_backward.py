@@ -6,7 +6,7 @@ MAX_UNARY_PAYLOAD_SIZE/2, :107).
 
 from __future__ import annotations
 
-from typing import Dict, List, Optional, Sequence, Tuple
+from typing import Dict, Optional, Sequence, Tuple
 
 import torch
 

@@ -4,12 +4,11 @@ from __future__ import annotations
 
 import contextvars
 from contextlib import contextmanager
-from typing import Optional, Sequence
+from typing import Optional
 
 import torch
 from torch import nn
 
-from petals_amd.client.config import ClientConfig
 from petals_amd.client.inference_session import InferenceSession
 from petals_amd.client.routing.sequence_manager import RemoteSequenceManager
 from petals_amd.client.sequential_autograd import _RemoteSequentialAutogradFunction

@@ -19,7 +19,7 @@ from typing import Any, Dict, List, Optional, Sequence, Tuple
 
 from petals_amd.client.config import ClientConfig
 from petals_amd.client.routing.sequence_info import RemoteSequenceInfo
-from petals_amd.data_structures import ModuleUID, RemoteSpanInfo, ServerState
+from petals_amd.data_structures import ModuleUID, RemoteSpanInfo
 from petals_amd.dht.node import DHT
 from petals_amd.utils.dht import get_remote_module_infos
 from petals_amd.utils.ping import PingAggregator

@@ -8,7 +8,7 @@ HF fused-QKV layout: query_key_value output reshapes to
 
 from __future__ import annotations
 
-from typing import Optional, Tuple
+from typing import Tuple
 
 import torch
 from torch import nn

@@ -11,7 +11,7 @@ path is the fused HIP kernel.)
 
 from __future__ import annotations
 
-from typing import Optional, Tuple
+from typing import Tuple
 
 import torch
 from torch import nn

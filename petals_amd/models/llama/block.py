@@ -20,7 +20,6 @@ Block forward contract (used by the server backend and tests):
 
 from __future__ import annotations
 
-import math
 from typing import Optional, Tuple
 
 import torch

@@ -10,7 +10,7 @@ grouped-GEMM HIP kernel can slot in on GPU.)
 
 from __future__ import annotations
 
-from typing import Optional, Tuple
+from typing import Tuple
 
 import torch
 import torch.nn.functional as F

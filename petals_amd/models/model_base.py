@@ -16,10 +16,9 @@ import dataclasses
 import json
 import logging
 import os
-from typing import Optional, Sequence, Tuple
+from typing import Optional, Tuple
 
 import torch
-import torch.nn.functional as F
 from torch import nn
 
 from petals_amd.client.config import ClientConfig
@@ -27,7 +26,7 @@ from petals_amd.client.lm_head import LMHead, LMHeadConfig
 from petals_amd.client.ptune import PTuneConfig, PTuneMixin
 from petals_amd.client.remote_generation import RemoteGenerationMixin
 from petals_amd.client.remote_sequential import RemoteSequential
-from petals_amd.utils.misc import DUMMY, is_dummy
+from petals_amd.utils.misc import DUMMY
 
 logger = logging.getLogger(__name__)
 

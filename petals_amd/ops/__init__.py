@@ -8,7 +8,7 @@ fallback; set PETALS_AMD_ALLOW_TORCH_FALLBACK=1 to override for debugging).
 from __future__ import annotations
 
 import os
-from typing import Optional, Tuple
+from typing import Optional
 
 import torch
 

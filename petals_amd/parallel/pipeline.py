@@ -12,7 +12,7 @@ Used by bench.py and by co-located server deployments; the TCP path in
 from __future__ import annotations
 
 import os
-from typing import List, Optional, Sequence
+from typing import List, Sequence
 
 import torch
 import torch.distributed as dist

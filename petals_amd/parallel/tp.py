@@ -17,7 +17,6 @@ reference's `--tensor_parallel_devices cpu cpu` CI servers.
 
 from __future__ import annotations
 
-import math
 from typing import Optional, Tuple
 
 import torch

@@ -9,7 +9,7 @@ re-placement of every server (eps guards against oscillation).
 
 from __future__ import annotations
 
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, List, Optional
 
 import numpy as np
 

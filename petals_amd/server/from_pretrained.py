@@ -13,7 +13,7 @@ from __future__ import annotations
 import json
 import logging
 import os
-from typing import Dict, Optional
+from typing import Dict
 
 import torch
 

@@ -18,7 +18,7 @@ import itertools
 import logging
 import threading
 import time
-from typing import AsyncContextManager, Dict, List, Optional, Sequence, Tuple
+from typing import Dict, List, Optional, Tuple
 
 import torch
 

@@ -24,7 +24,7 @@ import torch
 
 from petals_amd.data_structures import ServerInfo, ServerState, get_dht_time, make_uid
 from petals_amd.dht.node import DHTNode
-from petals_amd.models.config_base import ModelConfig, load_model_config
+from petals_amd.models.config_base import load_model_config
 from petals_amd.p2p.transport import P2PNode
 from petals_amd.server import block_selection
 from petals_amd.server.backend import TransformerBackend
@@ -33,7 +33,6 @@ from petals_amd.server.handler import TransformerConnectionHandler
 from petals_amd.server.memory_cache import MemoryCache
 from petals_amd.server.scheduler import PriorityRuntime
 from petals_amd.server.throughput import get_server_throughput
-from petals_amd.utils.dht import declare_active_modules, declare_model, get_remote_module_infos
 from petals_amd.utils.misc import get_size_in_bytes
 
 logger = logging.getLogger(__name__)

@@ -21,7 +21,6 @@ from typing import Dict, Optional
 
 import torch
 
-from petals_amd.models import get_model_block
 from petals_amd.models.config_base import ModelConfig
 from petals_amd.server.from_pretrained import init_random_block_
 

@@ -6,10 +6,9 @@ by each model family subpackage.
 
 from __future__ import annotations
 
-import dataclasses
 import json
 import os
-from typing import Dict, Optional, Type
+from typing import Dict
 
 _REGISTRY: Dict[str, Dict[str, type]] = {}
 

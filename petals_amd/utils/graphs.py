@@ -8,7 +8,7 @@ single captured graph replays for every token of a session.
 
 from __future__ import annotations
 
-from typing import Callable, List, Optional, Sequence
+from typing import Callable, Sequence
 
 import torch
 

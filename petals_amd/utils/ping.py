@@ -5,7 +5,7 @@ from __future__ import annotations
 import asyncio
 import math
 import time
-from typing import Dict, Sequence, Tuple
+from typing import Dict, Tuple
 
 from petals_amd.p2p.transport import P2PNode, RpcMessage
 
