@@ -52,11 +52,15 @@ __global__ __launch_bounds__(WAVES * 64) void attn_prefill_kernel(
   const int hi = lane >> 4;       // fragment k-group (and C row group)
   const int q0 = q0_wg + wave * QTILE;  // this wave's first q row
 
-  // LDS: K tile row-major [KVTILE][HD+KPAD]; V tile transposed [HD][KVTILE+KPAD];
-  // per-wave P scratch [QTILE][KVTILE+KPAD]
+  // LDS: K tile row-major [KVTILE][HD+KPAD]; V tile transposed [HD][KVTILE+KPAD]
+  // with an XOR swizzle on the dim-group (rows 8 apart land 0 mod 128 B with a
+  // 16 B-aligned stride, so the scalar transpose writes were 16-way
+  // bank-conflicted — the dominant cost in the v1 PMC profile); per-wave P
+  // scratch [QTILE][KVTILE+KPAD]
   __shared__ unsigned short k_lds[KVTILE][HD + KPAD];
-  __shared__ unsigned short vt_lds[HD][KVTILE + KPAD];
+  __shared__ unsigned char vt_raw[HD * (KVTILE + KPAD) * 2];
   __shared__ unsigned short p_lds[WAVES][QTILE][KVTILE + KPAD];
+#define VT_BYTE(dim, key_byte)   (((unsigned)(dim) * ((KVTILE + KPAD) * 2) + (unsigned)(key_byte)) ^ ((((unsigned)(dim) >> 3) & 7u) << 4))
 
   constexpr int KCH = HD / 32;  // 32-wide k-dim chunks per head dim
 
@@ -106,7 +110,8 @@ __global__ __launch_bounds__(WAVES * 64) void attn_prefill_kernel(
       }
       *reinterpret_cast<bf16x8*>(&k_lds[row][c8]) = kv8;
 #pragma unroll
-      for (int e = 0; e < 8; ++e) vt_lds[c8 + e][row] = (unsigned short)vv8[e];
+      for (int e = 0; e < 8; ++e)
+        *reinterpret_cast<unsigned short*>(&vt_raw[VT_BYTE(c8 + e, row * 2)]) = (unsigned short)vv8[e];
     }
     __syncthreads();
 
@@ -175,7 +180,8 @@ __global__ __launch_bounds__(WAVES * 64) void attn_prefill_kernel(
 #pragma unroll
     for (int d = 0; d < HD / 16; ++d) {
       // B[k = key][n = dim] = VT[dim = d*16 + col][key = hi*8 + reg]
-      const bf16x8 vfrag = *reinterpret_cast<const bf16x8*>(&vt_lds[d * 16 + col][hi * 8]);
+      const bf16x8 vfrag =
+          *reinterpret_cast<const bf16x8*>(&vt_raw[VT_BYTE(d * 16 + col, hi * 16)]);
       acc_o[d] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(p_frag, vfrag, acc_o[d], 0, 0, 0);
     }
   }
