@@ -24,6 +24,14 @@ def build(verbose: bool = False) -> str:
     from torch.utils.cpp_extension import load
 
     os.makedirs(BUILD_DIR, exist_ok=True)
+    # torch's hipify writes `<name>_hip.<ext>` next to each source and skips
+    # regeneration based on mtimes; a repo snapshot (gpurun) can scramble those
+    # and leave a stale artifact compiling instead of the real source. Purge
+    # them so every build regenerates from the true sources.
+    import glob
+
+    for stale in glob.glob(os.path.join(SRC_DIR, "*_hip.*")):
+        os.unlink(stale)
     sources = [os.path.join(SRC_DIR, s) for s in SOURCES if os.path.exists(os.path.join(SRC_DIR, s))]
     module = load(
         name="petals_amd_hip_ops",

@@ -162,15 +162,15 @@ __global__ __launch_bounds__(WAVES * 64) void attn_prefill_kernel(
     // ---- write P (bf16) into per-wave LDS in [q_row][key] layout
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      *reinterpret_cast<unsigned short*>(&p_raw[P_BYTE(wave, hi * 4 + r, col * 2)]) = f32_to_bf16(p[0][r]);
-      *reinterpret_cast<unsigned short*>(&p_raw[P_BYTE(wave, hi * 4 + r, 32 + col * 2)]) = f32_to_bf16(p[1][r]);
+      p_lds[wave][hi * 4 + r][col] = f32_to_bf16(p[0][r]);
+      p_lds[wave][hi * 4 + r][16 + col] = f32_to_bf16(p[1][r]);
     }
     // the P round-trip is per-wave, but a block barrier is the simple safe
     // ordering (the compiler may not prove the write/read regions disjoint)
     __syncthreads();
 
     // P A-fragment: A[row = lane&15][k = hi*8 + reg] over the 32 keys
-    const bf16x8 p_frag = *reinterpret_cast<const bf16x8*>(&p_raw[P_BYTE(wave, col, hi * 16)]);
+    const bf16x8 p_frag = *reinterpret_cast<const bf16x8*>(&p_lds[wave][col][hi * 8]);
 
     // ---- rescale O, then PV
 #pragma unroll
