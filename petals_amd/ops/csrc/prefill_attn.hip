@@ -2,12 +2,15 @@
 //
 // Geometry: one workgroup = 4 waves = 64 query rows of ONE (batch, q_head);
 // each wave owns a 16-row q tile held in registers as mfma_f32_16x16x32_bf16
-// A-fragments. K/V tiles of 32 keys are staged in LDS by the whole workgroup
+// A-fragments. K/V tiles of KVT keys are staged in LDS by the whole workgroup
 // (K row-major padded; V transposed so the PV B-fragments are contiguous
-// ds_read_b128 rows). Per tile: 8 QK^T MFMAs -> causal mask -> online softmax
-// (per-lane row stats: the C-layout keeps each q row's 4 accumulator rows in
-// the same lane) -> P through a per-wave LDS round-trip into A-fragment
-// layout -> 8 PV MFMAs. Fragment layouts HW-verified by scripts/mfma_verify.hip.
+// ds_read_b128 rows). Per tile: KVT/16*KCH QK^T MFMAs -> causal mask -> online
+// softmax (per-lane row stats: the C-layout keeps each q row's 4 accumulator
+// rows in the same lane) -> P through a per-wave LDS round-trip into
+// A-fragment layout -> KVT/32*HD/16 PV MFMAs. Fragment layouts HW-verified by
+// scripts/mfma_verify.hip. KVT=64 default (amortizes softmax shuffles and
+// barriers over 2x the MFMA work vs KVT=32); PETALS_PREFILL_KVT=32 selects
+// the narrow tile for A/B.
 //
 // Replaces the chunked rocBLAS-matmul + fp32-softmax prefill composition
 // (ops/reference.py attention) on the GPU path; the reference framework used
@@ -17,15 +20,24 @@
 #include <torch/extension.h>
 #include <ATen/cuda/CUDAContext.h>
 
+#include <cstdlib>
+
 using bf16x8 = __attribute__((ext_vector_type(8))) short;
 using f32x4 = __attribute__((ext_vector_type(4))) float;
 
-#define QTILE 16   // q rows per wave
-#define WAVES 4    // waves per workgroup
-#define KVTILE 32  // keys per LDS tile
-#define KPAD 8     // LDS row padding (elements) against bank conflicts
+#define QTILE 16  // q rows per wave
+#define WAVES 4   // waves per workgroup
+#define KPAD 8    // LDS row padding (elements) against bank conflicts
 
-template <int HD>
+// V-transpose LDS addressing: XOR swizzle on the dim-group (rows 8 apart land
+// 0 mod 128 B with any 16 B-aligned pitch, so the scalar transpose writes were
+// 16-way bank-conflicted — the dominant cost in the v1 PMC profile). Verified
+// injective with stage-write multiplicity 2 (b16 ideal) and PV-read
+// multiplicity 8 (b128 ideal) for both KVT=32 and KVT=64.
+#define VT_BYTE(dim, key_byte) \
+  ((((unsigned)(dim)) * ((KVT + KPAD) * 2) + (unsigned)(key_byte)) ^ ((((unsigned)(dim) >> 3) & 7u) << 4))
+
+template <int HD, int KVT>
 __global__ __launch_bounds__(WAVES * 64) void attn_prefill_kernel(
     const unsigned short* __restrict__ q,   // [B, QH, S, HD]
     const unsigned short* __restrict__ k,   // [B, KVH, Lmax, HD]
@@ -52,17 +64,15 @@ __global__ __launch_bounds__(WAVES * 64) void attn_prefill_kernel(
   const int hi = lane >> 4;       // fragment k-group (and C row group)
   const int q0 = q0_wg + wave * QTILE;  // this wave's first q row
 
-  // LDS: K tile row-major [KVTILE][HD+KPAD]; V tile transposed [HD][KVTILE+KPAD]
-  // with an XOR swizzle on the dim-group (rows 8 apart land 0 mod 128 B with a
-  // 16 B-aligned stride, so the scalar transpose writes were 16-way
-  // bank-conflicted — the dominant cost in the v1 PMC profile); per-wave P
-  // scratch [QTILE][KVTILE+KPAD]
-  __shared__ unsigned short k_lds[KVTILE][HD + KPAD];
-  __shared__ unsigned char vt_raw[HD * (KVTILE + KPAD) * 2];
-  __shared__ unsigned short p_lds[WAVES][QTILE][KVTILE + KPAD];
-#define VT_BYTE(dim, key_byte)   (((unsigned)(dim) * ((KVTILE + KPAD) * 2) + (unsigned)(key_byte)) ^ ((((unsigned)(dim) >> 3) & 7u) << 4))
+  // LDS: K tile row-major [KVT][HD+KPAD]; V tile transposed+swizzled
+  // [HD][KVT+KPAD]; per-wave P scratch [QTILE][KVT+KPAD]
+  __shared__ unsigned short k_lds[KVT][HD + KPAD];
+  __shared__ unsigned char vt_raw[HD * (KVT + KPAD) * 2];
+  __shared__ unsigned short p_lds[WAVES][QTILE][KVT + KPAD];
 
-  constexpr int KCH = HD / 32;  // 32-wide k-dim chunks per head dim
+  constexpr int KCH = HD / 32;   // 32-wide k-dim chunks per head dim
+  constexpr int NB = KVT / 16;   // 16-key S column blocks per tile
+  constexpr int PKC = KVT / 32;  // 32-key PV k-chunks per tile
 
   // ---- load this wave's q tile into A-fragments (zero-padded past s_q)
   bf16x8 q_frag[KCH];
@@ -78,7 +88,7 @@ __global__ __launch_bounds__(WAVES * 64) void attn_prefill_kernel(
     }
   }
 
-  // ---- accumulators: O in C-layout (8 dim-blocks x f32x4), softmax stats
+  // ---- accumulators: O in C-layout (HD/16 dim-blocks x f32x4), softmax stats
   f32x4 acc_o[HD / 16];
 #pragma unroll
   for (int d = 0; d < HD / 16; ++d) acc_o[d] = f32x4{0.f, 0.f, 0.f, 0.f};
@@ -95,11 +105,11 @@ __global__ __launch_bounds__(WAVES * 64) void attn_prefill_kernel(
 
   const size_t kv_base = (((size_t)b * kv_heads + kvh) * lmax) * HD;
 
-  for (int j0 = 0; j0 < kv_end; j0 += KVTILE) {
-    const int tile_n = min(KVTILE, kv_end - j0);
+  for (int j0 = 0; j0 < kv_end; j0 += KVT) {
+    const int tile_n = min(KVT, kv_end - j0);
     // ---- stage K tile (row-major) and V tile (transposed), 256 threads
     __syncthreads();
-    for (int idx = tid; idx < KVTILE * (HD / 8); idx += WAVES * WAVE) {
+    for (int idx = tid; idx < KVT * (HD / 8); idx += WAVES * WAVE) {
       const int row = idx / (HD / 8);
       const int c8 = (idx - row * (HD / 8)) * 8;
       bf16x8 kv8 = {0, 0, 0, 0, 0, 0, 0, 0};
@@ -115,10 +125,10 @@ __global__ __launch_bounds__(WAVES * 64) void attn_prefill_kernel(
     }
     __syncthreads();
 
-    // ---- S = Q K^T : two 16-key column blocks
-    f32x4 s_acc[2];
+    // ---- S = Q K^T : NB 16-key column blocks
+    f32x4 s_acc[NB];
 #pragma unroll
-    for (int nb = 0; nb < 2; ++nb) {
+    for (int nb = 0; nb < NB; ++nb) {
       s_acc[nb] = f32x4{0.f, 0.f, 0.f, 0.f};
 #pragma unroll
       for (int kc = 0; kc < KCH; ++kc) {
@@ -129,48 +139,60 @@ __global__ __launch_bounds__(WAVES * 64) void attn_prefill_kernel(
     }
 
     // ---- causal mask + online softmax (per-lane rows hi*4+r, col = key)
-    float p[2][4];  // [nb][r] probabilities for this lane's slots
+    float p[NB][4];  // [nb][r] probabilities for this lane's slots
     float corr[4];
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int qrow = q0 + hi * 4 + r;
       const int q_abs = kv_offset + qrow;
-      float s0 = s_acc[0][r] * scale;
-      float s1 = s_acc[1][r] * scale;
-      const int key0 = j0 + col, key1 = j0 + 16 + col;
-      const bool dead0 = key0 >= tile_n + j0 || (causal && key0 > q_abs) || qrow >= s_q;
-      const bool dead1 = key1 >= tile_n + j0 || (causal && key1 > q_abs) || qrow >= s_q;
-      if (dead0) s0 = NEG_SENTINEL;
-      if (dead1) s1 = NEG_SENTINEL;
+      // NOTE: masking and the running max are deliberately SEPARATE loops.
+      // Fusing them (mask + `mx = fmaxf(mx, s[nb])` in one loop body) makes
+      // amdclang (ROCm 7.2, gfx950, -O3, with or without -ffast-math)
+      // miscompile the conditional sentinel store: dead keys keep their raw
+      // scores for the r==0 slot of every accumulator row group, leaking
+      // masked keys into the softmax (bisected in scripts/prefill_bisect.hip).
+      float s[NB];
+#pragma unroll
+      for (int nb = 0; nb < NB; ++nb) {
+        s[nb] = s_acc[nb][r] * scale;
+        const int key = j0 + nb * 16 + col;
+        if (key >= tile_n + j0 || (causal && key > q_abs) || qrow >= s_q) s[nb] = NEG_SENTINEL;
+      }
+      float mx = NEG_SENTINEL;
+#pragma unroll
+      for (int nb = 0; nb < NB; ++nb) mx = fmaxf(mx, s[nb]);
       // row max across the 16 lanes holding this row (xor within low 4 bits)
-      float mx = fmaxf(s0, s1);
 #pragma unroll
       for (int off = 8; off > 0; off >>= 1) mx = fmaxf(mx, __shfl_xor(mx, off, WAVE));
       const float m_new = fmaxf(m_row[r], mx);
       corr[r] = (m_row[r] <= NEG_THRESHOLD) ? 0.f : __expf(m_row[r] - m_new);
-      const float p0 = (s0 <= NEG_THRESHOLD) ? 0.f : __expf(s0 - m_new);
-      const float p1 = (s1 <= NEG_THRESHOLD) ? 0.f : __expf(s1 - m_new);
-      float lsum = p0 + p1;
+      float lsum = 0.f;
+#pragma unroll
+      for (int nb = 0; nb < NB; ++nb) {
+        p[nb][r] = (s[nb] <= NEG_THRESHOLD) ? 0.f : __expf(s[nb] - m_new);
+        lsum += p[nb][r];
+      }
 #pragma unroll
       for (int off = 8; off > 0; off >>= 1) lsum += __shfl_xor(lsum, off, WAVE);
       l_row[r] = l_row[r] * corr[r] + lsum;
       m_row[r] = m_new;
-      p[0][r] = p0;
-      p[1][r] = p1;
     }
 
     // ---- write P (bf16) into per-wave LDS in [q_row][key] layout
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      p_lds[wave][hi * 4 + r][col] = f32_to_bf16(p[0][r]);
-      p_lds[wave][hi * 4 + r][16 + col] = f32_to_bf16(p[1][r]);
-    }
+    for (int r = 0; r < 4; ++r)
+#pragma unroll
+      for (int nb = 0; nb < NB; ++nb)
+        p_lds[wave][hi * 4 + r][nb * 16 + col] = f32_to_bf16(p[nb][r]);
     // the P round-trip is per-wave, but a block barrier is the simple safe
     // ordering (the compiler may not prove the write/read regions disjoint)
     __syncthreads();
 
-    // P A-fragment: A[row = lane&15][k = hi*8 + reg] over the 32 keys
-    const bf16x8 p_frag = *reinterpret_cast<const bf16x8*>(&p_lds[wave][col][hi * 8]);
+    // P A-fragments: A[row = lane&15][k = pk*32 + hi*8 + reg] over KVT keys
+    bf16x8 p_frag[PKC];
+#pragma unroll
+    for (int pk = 0; pk < PKC; ++pk)
+      p_frag[pk] = *reinterpret_cast<const bf16x8*>(&p_lds[wave][col][pk * 32 + hi * 8]);
 
     // ---- rescale O, then PV
 #pragma unroll
@@ -179,10 +201,13 @@ __global__ __launch_bounds__(WAVES * 64) void attn_prefill_kernel(
       for (int r = 0; r < 4; ++r) acc_o[d][r] *= corr[r];
 #pragma unroll
     for (int d = 0; d < HD / 16; ++d) {
-      // B[k = key][n = dim] = VT[dim = d*16 + col][key = hi*8 + reg]
-      const bf16x8 vfrag =
-          *reinterpret_cast<const bf16x8*>(&vt_raw[VT_BYTE(d * 16 + col, hi * 16)]);
-      acc_o[d] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(p_frag, vfrag, acc_o[d], 0, 0, 0);
+#pragma unroll
+      for (int pk = 0; pk < PKC; ++pk) {
+        // B[k = key][n = dim] = VT[dim = d*16 + col][key = pk*32 + hi*8 + reg]
+        const bf16x8 vfrag =
+            *reinterpret_cast<const bf16x8*>(&vt_raw[VT_BYTE(d * 16 + col, pk * 64 + hi * 16)]);
+        acc_o[d] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(p_frag[pk], vfrag, acc_o[d], 0, 0, 0);
+      }
     }
   }
 
@@ -218,20 +243,30 @@ torch::Tensor attn_prefill_fused(
   auto stream = at::cuda::getCurrentCUDAStream();
   const float sc = (float)scale;
 
+  static const int kvt_env = [] {
+    const char* s = std::getenv("PETALS_PREFILL_KVT");
+    return s ? std::atoi(s) : 64;
+  }();
+
+  const auto* qp = reinterpret_cast<const unsigned short*>(q.data_ptr());
+  const auto* kp = reinterpret_cast<const unsigned short*>(k.data_ptr());
+  const auto* vp = reinterpret_cast<const unsigned short*>(v.data_ptr());
+  auto* op = reinterpret_cast<unsigned short*>(out.data_ptr());
+
   if (HD == 128) {
-    attn_prefill_kernel<128><<<grid, WAVES * WAVE, 0, stream>>>(
-        reinterpret_cast<const unsigned short*>(q.data_ptr()),
-        reinterpret_cast<const unsigned short*>(k.data_ptr()),
-        reinterpret_cast<const unsigned short*>(v.data_ptr()),
-        reinterpret_cast<unsigned short*>(out.data_ptr()),
-        QH, KVH, S, LMAX, (int)kv_len, (int)kv_offset, sc, causal ? 1 : 0);
+    if (kvt_env == 32)
+      attn_prefill_kernel<128, 32><<<grid, WAVES * WAVE, 0, stream>>>(
+          qp, kp, vp, op, QH, KVH, S, LMAX, (int)kv_len, (int)kv_offset, sc, causal ? 1 : 0);
+    else
+      attn_prefill_kernel<128, 64><<<grid, WAVES * WAVE, 0, stream>>>(
+          qp, kp, vp, op, QH, KVH, S, LMAX, (int)kv_len, (int)kv_offset, sc, causal ? 1 : 0);
   } else {
-    attn_prefill_kernel<64><<<grid, WAVES * WAVE, 0, stream>>>(
-        reinterpret_cast<const unsigned short*>(q.data_ptr()),
-        reinterpret_cast<const unsigned short*>(k.data_ptr()),
-        reinterpret_cast<const unsigned short*>(v.data_ptr()),
-        reinterpret_cast<unsigned short*>(out.data_ptr()),
-        QH, KVH, S, LMAX, (int)kv_len, (int)kv_offset, sc, causal ? 1 : 0);
+    if (kvt_env == 32)
+      attn_prefill_kernel<64, 32><<<grid, WAVES * WAVE, 0, stream>>>(
+          qp, kp, vp, op, QH, KVH, S, LMAX, (int)kv_len, (int)kv_offset, sc, causal ? 1 : 0);
+    else
+      attn_prefill_kernel<64, 64><<<grid, WAVES * WAVE, 0, stream>>>(
+          qp, kp, vp, op, QH, KVH, S, LMAX, (int)kv_len, (int)kv_offset, sc, causal ? 1 : 0);
   }
   HIP_CHECK_LAST();
   return out;

@@ -133,10 +133,16 @@ def prefill_attn_bench():
 
 
 if __name__ == "__main__":
-    attn_debug()
-    rmsnorm_bench()
-    gemv_bench()
-    attn_bench()
-    layer_decode_bench()
-    nf4_bench()
-    prefill_attn_bench()
+    _all = {
+        "attn_debug": attn_debug,
+        "rmsnorm": rmsnorm_bench,
+        "gemv": gemv_bench,
+        "attn": attn_bench,
+        "layer": layer_decode_bench,
+        "nf4": nf4_bench,
+        "prefill": prefill_attn_bench,
+    }
+    picked = [a for a in sys.argv[1:] if a in _all]
+    for name, fn in _all.items():
+        if not picked or name in picked:
+            fn()
