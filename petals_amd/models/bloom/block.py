@@ -75,7 +75,43 @@ class BloomBlock(nn.Module):
         self.mlp = BloomMLP(config)
         self.apply_residual_post_ln = config.apply_residual_connection_post_layernorm
 
-    def forward(self, hidden_states, kv_cache=None, prefix_length: int = 0):
+    _fast = None  # BloomFastPath after optimize_for_inference()
+
+    def optimize_for_inference(self, quant: str = "none") -> "BloomBlock":
+        """Repack weights into the MI355X kernel layout (LayerNorm + ALiBi +
+        GELU fused decode chain); frees nn.Linear weights."""
+        from petals_amd import ops as _ops
+        from petals_amd.ops.fused_decode import BloomFastPath
+
+        hip = _ops._load_hip_ops()
+        if hip is None:
+            raise RuntimeError(
+                f"cannot optimize block for MI355X: HIP extension missing ({_ops._hip_import_error!r})"
+            )
+        assert next(self.parameters()).device.type == "cuda", "optimize_for_inference needs a GPU block"
+        if self.config.head_dim not in (64, 128):
+            import logging
+
+            logging.getLogger(__name__).warning(
+                "BLOOM head_dim=%s outside the fused fast path; serving via generic HIP ops",
+                self.config.head_dim,
+            )
+            return self
+        self._fast = BloomFastPath(self, hip, quant=quant)
+        return self
+
+    def forward(self, hidden_states, kv_cache=None, prefix_length: int = 0, ctx=None):
+        if self._fast is not None:
+            if torch.is_grad_enabled() and hidden_states.requires_grad:
+                assert kv_cache is None, "training forward does not use the KV cache"
+                return self._fast.forward_autograd(hidden_states, prefix_length)
+            max_b = 4 if self._fast.quant == "nf4" else 8
+            if kv_cache is not None and hidden_states.shape[1] == 1 and hidden_states.shape[0] <= max_b:
+                return self._fast.decode_step(
+                    hidden_states, kv_cache[0], kv_cache[1], prefix_length, ctx=ctx
+                )
+            return self._fast.forward(hidden_states, kv_cache, prefix_length)
+
         ln_out = self.input_layernorm(hidden_states)
         residual = ln_out if self.apply_residual_post_ln else hidden_states
         attn = self.self_attention(ln_out, kv_cache=kv_cache, prefix_length=prefix_length)

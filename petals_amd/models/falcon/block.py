@@ -108,7 +108,53 @@ class FalconBlock(nn.Module):
             if not config.parallel_attn:
                 self.post_attention_layernorm = nn.LayerNorm(config.hidden_size, eps=eps)
 
-    def forward(self, hidden_states, kv_cache=None, prefix_length: int = 0):
+    _fast = None  # FalconFastPath after optimize_for_inference()
+
+    def optimize_for_inference(self, quant: str = "none") -> "FalconBlock":
+        """Repack weights into the MI355X kernel layout (parallel attn+MLP,
+        rope GQA fused decode chain); frees nn.Linear weights. Only the
+        new-decoder architecture (40B/180B) with a supported (head_dim, gqa)
+        geometry takes the fused path; other variants serve via generic ops."""
+        from petals_amd import ops as _ops
+        from petals_amd.ops.fused_decode import FalconFastPath
+
+        hip = _ops._load_hip_ops()
+        if hip is None:
+            raise RuntimeError(
+                f"cannot optimize block for MI355X: HIP extension missing ({_ops._hip_import_error!r})"
+            )
+        assert next(self.parameters()).device.type == "cuda", "optimize_for_inference needs a GPU block"
+        cfg = self.config
+        gq = cfg.num_attention_heads // cfg.n_kv_heads
+        if (
+            not cfg.new_decoder_architecture
+            or cfg.head_dim not in (64, 128)
+            or gq not in (1, 2, 4, 6, 8, 16)
+            or cfg.bias
+        ):
+            import logging
+
+            logging.getLogger(__name__).warning(
+                "Falcon geometry (new_decoder=%s, head_dim=%s, gq=%s, bias=%s) outside the fused "
+                "fast path; serving via generic HIP ops",
+                cfg.new_decoder_architecture, cfg.head_dim, gq, cfg.bias,
+            )
+            return self
+        self._fast = FalconFastPath(self, hip, quant=quant)
+        return self
+
+    def forward(self, hidden_states, kv_cache=None, prefix_length: int = 0, ctx=None):
+        if self._fast is not None:
+            if torch.is_grad_enabled() and hidden_states.requires_grad:
+                assert kv_cache is None, "training forward does not use the KV cache"
+                return self._fast.forward_autograd(hidden_states, prefix_length)
+            max_b = 4 if self._fast.quant == "nf4" else 8
+            if kv_cache is not None and hidden_states.shape[1] == 1 and hidden_states.shape[0] <= max_b:
+                return self._fast.decode_step(
+                    hidden_states, kv_cache[0], kv_cache[1], prefix_length, ctx=ctx
+                )
+            return self._fast.forward(hidden_states, kv_cache, prefix_length)
+
         residual = hidden_states
         if self.config.new_decoder_architecture:
             attn_in = self.ln_attn(hidden_states)

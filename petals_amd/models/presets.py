@@ -103,8 +103,17 @@ PRESETS = {
         model_type="bloom", hidden_size=64, num_hidden_layers=4, num_attention_heads=4,
         vocab_size=128, max_position_embeddings=256, layer_norm_epsilon=1e-5,
     ),
+    "test-bloom-hd64": dict(  # fused-path geometry (head_dim 64) for GPU tests
+        model_type="bloom", hidden_size=256, num_hidden_layers=4, num_attention_heads=4,
+        vocab_size=128, max_position_embeddings=256, layer_norm_epsilon=1e-5,
+    ),
     "test-falcon": dict(
         model_type="falcon", hidden_size=64, num_hidden_layers=4, num_attention_heads=4,
+        num_kv_heads=2, vocab_size=128, max_position_embeddings=256,
+        layer_norm_epsilon=1e-5, parallel_attn=True, new_decoder_architecture=True, bias=False,
+    ),
+    "test-falcon-hd64": dict(  # fused-path geometry (head_dim 64, gq 2) for GPU tests
+        model_type="falcon", hidden_size=256, num_hidden_layers=4, num_attention_heads=4,
         num_kv_heads=2, vocab_size=128, max_position_embeddings=256,
         layer_norm_epsilon=1e-5, parallel_attn=True, new_decoder_architecture=True, bias=False,
     ),

@@ -21,6 +21,8 @@ __global__ __launch_bounds__(64) void attn_decode_kernel(
     float* __restrict__ part_o,            // [B*KV, splits, GQ, HD]
     float* __restrict__ part_ml,           // [B*KV, splits, GQ, 2]
     const int* __restrict__ kv_len_ptr,
+    const float* __restrict__ alibi,  // [kv_heads*GQ] slopes or null (ALiBi adds slope*j; the
+                                      // row-constant part of the bias cancels in softmax)
     int kv_heads,
     int lmax,
     int n_splits,
@@ -62,6 +64,10 @@ __global__ __launch_bounds__(64) void attn_decode_kernel(
   const unsigned short* k_base = k_cache + ((size_t)b * kv_heads + kv) * lmax * HD;
   const unsigned short* v_base = v_cache + ((size_t)b * kv_heads + kv) * lmax * HD;
 
+  float sl[GQ];
+#pragma unroll
+  for (int g = 0; g < GQ; ++g) sl[g] = alibi ? alibi[kv * GQ + g] : 0.f;
+
   for (int j = j_begin + grp; j < j_end; j += GROUPS) {
     // load K row slice: 16 lanes x EPL elems, coalesced
     const unsigned short* krow = k_base + (size_t)j * HD + gl * EPL;
@@ -81,7 +87,7 @@ __global__ __launch_bounds__(64) void attn_decode_kernel(
       float partial = 0.f;
 #pragma unroll
       for (int e = 0; e < EPL; ++e) partial = fmaf(kf[e], q_lds[g][gl * EPL + e], partial);
-      s[g] = group16_reduce_sum(partial);  // all 16 lanes get the row score
+      s[g] = group16_reduce_sum(partial) + sl[g] * j;  // all lanes get the row score
     }
     // V row slice
     const unsigned short* vrow = v_base + (size_t)j * HD + gl * EPL;
@@ -182,7 +188,8 @@ torch::Tensor attn_decode_fused(
     int64_t n_splits_i,
     torch::Tensor part_o,   // workspace [B*KV, splits, GQ, HD] f32 (or empty)
     torch::Tensor part_ml,  // workspace [B*KV, splits, GQ, 2] f32 (or empty)
-    double scale) {
+    double scale,
+    c10::optional<torch::Tensor> alibi_slopes) {  // [kv_heads*GQ] f32
   TORCH_CHECK(q.is_cuda() && q.dtype() == torch::kFloat32);
   TORCH_CHECK(k_cache.dtype() == torch::kBFloat16 && k_cache.dim() == 4);
   const int b = q.size(0);
@@ -210,12 +217,18 @@ torch::Tensor attn_decode_fused(
   float* pmlp = part_ml.data_ptr<float>();
   const int* lenp = kv_len.data_ptr<int>();
   const float sc = (float)scale;
+  const float* alibi_p = nullptr;
+  if (alibi_slopes.has_value() && alibi_slopes->defined() && alibi_slopes->numel() > 0) {
+    TORCH_CHECK(alibi_slopes->is_contiguous() && alibi_slopes->dtype() == torch::kFloat32);
+    TORCH_CHECK(alibi_slopes->numel() == (int64_t)kv_heads * GQi, "alibi slopes must be [q_heads]");
+    alibi_p = alibi_slopes->data_ptr<float>();
+  }
 
   bool launched = false;
 #define ATTN_CASE(HDV, GQV)                                                   \
   if (hd == HDV && GQi == GQV) {                                              \
     attn_decode_kernel<HDV, GQV><<<grid, 64, 0, stream>>>(                    \
-        qp, kp, vp, pop, pmlp, lenp, kv_heads, lmax, n_splits, sc);           \
+        qp, kp, vp, pop, pmlp, lenp, alibi_p, kv_heads, lmax, n_splits, sc);  \
     launched = true;                                                          \
   }
   ATTN_CASE(128, 1) ATTN_CASE(128, 2) ATTN_CASE(128, 4) ATTN_CASE(128, 6)

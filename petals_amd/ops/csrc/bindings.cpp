@@ -4,40 +4,58 @@
 
 torch::Tensor rms_norm(torch::Tensor x, torch::Tensor w, double eps);
 torch::Tensor rms_norm_f32out(torch::Tensor x, torch::Tensor w, double eps);
+torch::Tensor layer_norm(torch::Tensor x, torch::Tensor w, torch::Tensor b, double eps);
+torch::Tensor layer_norm_f32out(torch::Tensor x, torch::Tensor w, torch::Tensor b, double eps);
 torch::Tensor swiglu(torch::Tensor gate, torch::Tensor up);
 std::vector<torch::Tensor> apply_rope(
     torch::Tensor q, torch::Tensor k, torch::Tensor cos_t, torch::Tensor sin_t, torch::Tensor pos);
 void rope_cache_write(
     torch::Tensor qkv, torch::Tensor cos_t, torch::Tensor sin_t, torch::Tensor pos,
     torch::Tensor k_cache, torch::Tensor v_cache, int64_t qh, int64_t kh);
+void kv_cache_write(
+    torch::Tensor qkv, torch::Tensor pos,
+    torch::Tensor k_cache, torch::Tensor v_cache, int64_t qh, int64_t kh);
 torch::Tensor gemv_bf16(
     torch::Tensor wt, torch::Tensor x, torch::Tensor workspace,
-    c10::optional<torch::Tensor> residual, int64_t epilogue, int64_t splits_override);
+    c10::optional<torch::Tensor> residual, int64_t epilogue, int64_t splits_override,
+    c10::optional<torch::Tensor> bias);
 torch::Tensor attn_decode_fused(
     torch::Tensor q, torch::Tensor k_cache, torch::Tensor v_cache, torch::Tensor kv_len,
-    int64_t gq, int64_t n_splits, torch::Tensor part_o, torch::Tensor part_ml, double scale);
+    int64_t gq, int64_t n_splits, torch::Tensor part_o, torch::Tensor part_ml, double scale,
+    c10::optional<torch::Tensor> alibi_slopes);
 torch::Tensor attn_prefill_fused(
     torch::Tensor q, torch::Tensor k, torch::Tensor v,
-    int64_t kv_len, int64_t kv_offset, double scale, bool causal);
+    int64_t kv_len, int64_t kv_offset, double scale, bool causal,
+    c10::optional<torch::Tensor> alibi_slopes);
 std::vector<torch::Tensor> nf4_quantize(torch::Tensor w);
 torch::Tensor nf4_dequantize(torch::Tensor packed, torch::Tensor absmax);
 torch::Tensor gemv_nf4(
     torch::Tensor packed, torch::Tensor absmax, torch::Tensor x, torch::Tensor workspace,
-    c10::optional<torch::Tensor> residual, int64_t epilogue, int64_t splits_override);
+    c10::optional<torch::Tensor> residual, int64_t epilogue, int64_t splits_override,
+    c10::optional<torch::Tensor> bias);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rms_norm", &rms_norm, "RMSNorm (bf16 -> bf16)");
   m.def("rms_norm_f32out", &rms_norm_f32out, "RMSNorm (bf16 -> f32)");
+  m.def("layer_norm", &layer_norm, "LayerNorm with weight+bias (bf16 -> bf16)");
+  m.def("layer_norm_f32out", &layer_norm_f32out, "LayerNorm with weight+bias (bf16 -> f32)");
   m.def("swiglu", &swiglu, "silu(gate) * up (bf16)");
   m.def("apply_rope", &apply_rope, "rotate q,k by positions (bf16)");
   m.def("rope_cache_write", &rope_cache_write, "fused decode rope + kv cache write");
+  m.def("kv_cache_write", &kv_cache_write, "decode kv cache write without rope (ALiBi families)");
   m.def("gemv_bf16", &gemv_bf16, "split-K bf16 gemv with fused epilogue",
-        py::arg("wt"), py::arg("x"), py::arg("workspace"), py::arg("residual"), py::arg("epilogue"), py::arg("splits") = 0);
-  m.def("attn_decode_fused", &attn_decode_fused, "GQA decode attention (flash-decoding)");
-  m.def("attn_prefill_fused", &attn_prefill_fused, "MFMA flash prefill attention (bf16, causal, GQA)");
+        py::arg("wt"), py::arg("x"), py::arg("workspace"), py::arg("residual"), py::arg("epilogue"),
+        py::arg("splits") = 0, py::arg("bias") = py::none());
+  m.def("attn_decode_fused", &attn_decode_fused, "GQA decode attention (flash-decoding)",
+        py::arg("q"), py::arg("k_cache"), py::arg("v_cache"), py::arg("kv_len"), py::arg("gq"),
+        py::arg("n_splits"), py::arg("part_o"), py::arg("part_ml"), py::arg("scale"),
+        py::arg("alibi_slopes") = py::none());
+  m.def("attn_prefill_fused", &attn_prefill_fused, "MFMA flash prefill attention (bf16, causal, GQA)",
+        py::arg("q"), py::arg("k"), py::arg("v"), py::arg("kv_len"), py::arg("kv_offset"),
+        py::arg("scale"), py::arg("causal"), py::arg("alibi_slopes") = py::none());
   m.def("nf4_quantize", &nf4_quantize, "blockwise NF4 quantize [in,out] bf16 -> (packed u8, absmax bf16)");
   m.def("nf4_dequantize", &nf4_dequantize, "NF4 -> bf16");
   m.def("gemv_nf4", &gemv_nf4, "split-K NF4 gemv (dequant fused) with epilogue",
         py::arg("packed"), py::arg("absmax"), py::arg("x"), py::arg("workspace"),
-        py::arg("residual"), py::arg("epilogue"), py::arg("splits") = 0);
+        py::arg("residual"), py::arg("epilogue"), py::arg("splits") = 0, py::arg("bias") = py::none());
 }

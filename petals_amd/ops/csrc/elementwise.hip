@@ -109,6 +109,122 @@ torch::Tensor rms_norm_f32out(torch::Tensor x, torch::Tensor w, double eps) {
   return y.view(shape);
 }
 
+// ------------------------------------------------------------- layer_norm
+
+// one workgroup (256 threads) per row, two-pass mean/variance in registers;
+// BLOOM/Falcon blocks use LayerNorm with bias where Llama uses RMSNorm
+// (reference models/bloom/block.py wraps HF BloomBlock whose norms run in
+// torch; here the norm is one fused kernel feeding the gemv in f32).
+template <bool OUT_F32>
+__global__ void layer_norm_kernel(
+    const unsigned short* __restrict__ x,  // [rows, dim] bf16
+    const unsigned short* __restrict__ w,  // [dim] bf16
+    const unsigned short* __restrict__ bias,  // [dim] bf16
+    void* __restrict__ y,                  // [rows, dim] bf16 or f32
+    int dim,
+    float eps) {
+  constexpr int T = 256;
+  const int row = blockIdx.x;
+  const int tid = threadIdx.x;
+  const unsigned short* xr = x + (size_t)row * dim;
+
+  float sum = 0.f, sumsq = 0.f;
+  const int chunk = T * 8;
+  for (int base = 0; base < dim; base += chunk) {
+    const int idx = base + tid * 8;
+    if (idx + 8 <= dim) {
+      const short8 v = *reinterpret_cast<const short8*>(xr + idx);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const float f = bf16_to_f32((unsigned short)v[j]);
+        sum += f;
+        sumsq = fmaf(f, f, sumsq);
+      }
+    } else {
+      for (int j = idx; j < dim; ++j) {
+        const float f = bf16_to_f32(xr[j]);
+        sum += f;
+        sumsq = fmaf(f, f, sumsq);
+      }
+    }
+  }
+  __shared__ float red_s[T / WAVE];
+  __shared__ float red_q[T / WAVE];
+  sum = wave_reduce_sum(sum);
+  sumsq = wave_reduce_sum(sumsq);
+  if ((tid & (WAVE - 1)) == 0) {
+    red_s[tid / WAVE] = sum;
+    red_q[tid / WAVE] = sumsq;
+  }
+  __syncthreads();
+  float tot_s = 0.f, tot_q = 0.f;
+#pragma unroll
+  for (int i = 0; i < T / WAVE; ++i) {
+    tot_s += red_s[i];
+    tot_q += red_q[i];
+  }
+  const float mean = tot_s / dim;
+  const float var = tot_q / dim - mean * mean;
+  const float inv = rsqrtf(var + eps);
+
+  for (int base = 0; base < dim; base += chunk) {
+    const int idx = base + tid * 8;
+    if (idx + 8 <= dim) {
+      const short8 v = *reinterpret_cast<const short8*>(xr + idx);
+      const short8 wv = *reinterpret_cast<const short8*>(w + idx);
+      const short8 bv = *reinterpret_cast<const short8*>(bias + idx);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const float f = (bf16_to_f32((unsigned short)v[j]) - mean) * inv *
+                            bf16_to_f32((unsigned short)wv[j]) +
+                        bf16_to_f32((unsigned short)bv[j]);
+        if (OUT_F32)
+          reinterpret_cast<float*>(y)[(size_t)row * dim + idx + j] = f;
+        else
+          reinterpret_cast<unsigned short*>(y)[(size_t)row * dim + idx + j] = f32_to_bf16(f);
+      }
+    } else {
+      for (int j = idx; j < dim; ++j) {
+        const float f = (bf16_to_f32(xr[j]) - mean) * inv * bf16_to_f32(w[j]) + bf16_to_f32(bias[j]);
+        if (OUT_F32)
+          reinterpret_cast<float*>(y)[(size_t)row * dim + j] = f;
+        else
+          reinterpret_cast<unsigned short*>(y)[(size_t)row * dim + j] = f32_to_bf16(f);
+      }
+    }
+  }
+}
+
+static torch::Tensor layer_norm_impl(torch::Tensor x, torch::Tensor w, torch::Tensor b, double eps, bool f32out) {
+  TORCH_CHECK(x.is_cuda() && x.dtype() == torch::kBFloat16, "layer_norm expects bf16 CUDA tensor");
+  TORCH_CHECK(w.dtype() == torch::kBFloat16 && b.dtype() == torch::kBFloat16);
+  auto shape = x.sizes().vec();
+  const int dim = shape.back();
+  auto x2 = x.contiguous().view({-1, dim});
+  const int rows = x2.size(0);
+  auto y = f32out ? torch::empty({rows, dim}, x.options().dtype(torch::kFloat32)) : torch::empty_like(x2);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  auto* xp = reinterpret_cast<const unsigned short*>(x2.data_ptr());
+  auto* wp = reinterpret_cast<const unsigned short*>(w.contiguous().data_ptr());
+  auto* bp = reinterpret_cast<const unsigned short*>(b.contiguous().data_ptr());
+  if (f32out)
+    layer_norm_kernel<true><<<rows, 256, 0, stream>>>(xp, wp, bp, y.data_ptr(), dim, (float)eps);
+  else
+    layer_norm_kernel<false><<<rows, 256, 0, stream>>>(xp, wp, bp, y.data_ptr(), dim, (float)eps);
+  HIP_CHECK_LAST();
+  return f32out ? y : y.view(shape);
+}
+
+torch::Tensor layer_norm(torch::Tensor x, torch::Tensor w, torch::Tensor b, double eps) {
+  auto shape = x.sizes().vec();
+  auto y = layer_norm_impl(x, w, b, eps, false);
+  return y.view(shape);
+}
+
+torch::Tensor layer_norm_f32out(torch::Tensor x, torch::Tensor w, torch::Tensor b, double eps) {
+  return layer_norm_impl(x, w, b, eps, true);
+}
+
 // ------------------------------------------------------------------ swiglu
 
 __global__ void swiglu_kernel(
@@ -259,6 +375,48 @@ __global__ void rope_cache_write_kernel(
       v_cache[(((size_t)bi * kh + kh_i) * lmax + p) * hd + d] = f32_to_bf16(v);
     }
   }
+}
+
+// kv cache write WITHOUT rope for ALiBi families (BLOOM): same [q|k|v] fused
+// layout as rope_cache_write, k and v copied bf16 into the caches at *pos.
+__global__ void kv_cache_write_kernel(
+    const float* __restrict__ qkv,
+    const int* __restrict__ pos_ptr,
+    unsigned short* __restrict__ k_cache,  // [bcap, kh, lmax, hd]
+    unsigned short* __restrict__ v_cache,
+    int b, int qh, int kh, int lmax, int hd) {
+  const int p = *pos_ptr;
+  const int row_elems = qh * hd + 2 * kh * hd;
+  const long total = (long)b * kh * hd;
+  for (long idx = (long)blockIdx.x * blockDim.x + threadIdx.x; idx < 2 * total;
+       idx += (long)gridDim.x * blockDim.x) {
+    const bool is_v = idx >= total;
+    long t = is_v ? idx - total : idx;
+    const int d = t % hd; t /= hd;
+    const int kh_i = t % kh; t /= kh;
+    const int bi = t;
+    const float val = qkv[(size_t)bi * row_elems + qh * hd + (is_v ? kh * hd : 0) + kh_i * hd + d];
+    unsigned short* cache = is_v ? v_cache : k_cache;
+    cache[(((size_t)bi * kh + kh_i) * lmax + p) * hd + d] = f32_to_bf16(val);
+  }
+}
+
+void kv_cache_write(
+    torch::Tensor qkv, torch::Tensor pos,
+    torch::Tensor k_cache, torch::Tensor v_cache, int64_t qh, int64_t kh) {
+  TORCH_CHECK(qkv.is_cuda() && qkv.dtype() == torch::kFloat32 && qkv.dim() == 2);
+  TORCH_CHECK(k_cache.dtype() == torch::kBFloat16 && k_cache.dim() == 4);
+  const int b = qkv.size(0);
+  const int hd = k_cache.size(3), lmax = k_cache.size(2);
+  const long total = 2L * b * kh * hd;
+  const long blocks = std::min((total + 255) / 256 + 1, (long)1024);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  kv_cache_write_kernel<<<blocks, 256, 0, stream>>>(
+      qkv.data_ptr<float>(), pos.data_ptr<int>(),
+      reinterpret_cast<unsigned short*>(k_cache.data_ptr()),
+      reinterpret_cast<unsigned short*>(v_cache.data_ptr()),
+      b, (int)qh, (int)kh, lmax, hd);
+  HIP_CHECK_LAST();
 }
 
 void rope_cache_write(

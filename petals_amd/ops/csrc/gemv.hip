@@ -131,7 +131,8 @@ torch::Tensor gemv_bf16(
     torch::Tensor workspace, // [n_splits_max, batch, out] f32 (preallocated, may be empty)
     c10::optional<torch::Tensor> residual,  // [batch, out] bf16
     int64_t epilogue,
-    int64_t splits_override) {
+    int64_t splits_override,
+    c10::optional<torch::Tensor> bias) {  // [out] bf16, added pre-activation
   TORCH_CHECK(wt.is_cuda() && wt.dtype() == torch::kBFloat16 && wt.dim() == 2);
   TORCH_CHECK(x.is_cuda() && x.dtype() == torch::kFloat32 && x.dim() == 2);
   const int in_dim = wt.size(0), out_dim = wt.size(1);
@@ -176,6 +177,6 @@ torch::Tensor gemv_bf16(
 
   // epilogue (shared with the NF4 path)
   torch::Tensor y = launch_gemv_reduce(
-      partials, residual, splits, batch, out_dim, (int)epilogue, x.options(), wt.options());
+      partials, residual, bias, splits, batch, out_dim, (int)epilogue, x.options(), wt.options());
   return y;
 }

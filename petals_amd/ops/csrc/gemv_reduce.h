@@ -14,6 +14,7 @@ enum GemvEpilogue : int {
   EPI_PLAIN_BF16 = 1,    // y_bf16[b, out] = sum
   EPI_RESIDUAL_BF16 = 2, // y_bf16 = residual_bf16 + sum
   EPI_SWIGLU_F32 = 3,    // out = silu(sum[:half]) * sum[half:], y_f32[b, half]
+  EPI_GELU_F32 = 4,      // y_f32[b, out] = gelu_tanh(sum) (BLOOM/Falcon MLPs)
 };
 
 // deterministic split sum with 4 independent accumulators so the loads
@@ -33,9 +34,17 @@ static __device__ __forceinline__ float reduce_splits(
   return (s0 + s1) + (s2 + s3);
 }
 
+static __device__ __forceinline__ float gelu_tanh_f32(float x) {
+  // matches torch F.gelu(approximate="tanh") (= HF BLOOM/Falcon GELU)
+  const float c = 0.79788456080286535588f;  // sqrt(2/pi)
+  const float t = tanhf(c * (x + 0.044715f * x * x * x));
+  return 0.5f * x * (1.f + t);
+}
+
 static __global__ void gemv_reduce_kernel_impl(
     const float* __restrict__ partials,  // [n_splits, batch, out]
     const unsigned short* __restrict__ residual,  // [batch, out] or null
+    const unsigned short* __restrict__ bias,      // [out] bf16 or null
     void* __restrict__ y,
     int n_splits,
     int batch,
@@ -51,16 +60,23 @@ static __global__ void gemv_reduce_kernel_impl(
   const int o = idx - b * n_out;
   const size_t row = (size_t)b * out_dim;
   if (epilogue == EPI_SWIGLU_F32) {
-    const float g = reduce_splits(partials, n_splits, stride, row + o);
-    const float u = reduce_splits(partials, n_splits, stride, row + o + half);
+    float g = reduce_splits(partials, n_splits, stride, row + o);
+    float u = reduce_splits(partials, n_splits, stride, row + o + half);
+    if (bias) {
+      g += bf16_to_f32(bias[o]);
+      u += bf16_to_f32(bias[o + half]);
+    }
     const float act = g / (1.f + __expf(-g)) * u;
     reinterpret_cast<float*>(y)[(size_t)b * half + o] = act;
   } else {
-    const float sum = reduce_splits(partials, n_splits, stride, row + o);
+    float sum = reduce_splits(partials, n_splits, stride, row + o);
+    if (bias) sum += bf16_to_f32(bias[o]);
     if (epilogue == EPI_PLAIN_F32) {
       reinterpret_cast<float*>(y)[row + o] = sum;
     } else if (epilogue == EPI_PLAIN_BF16) {
       reinterpret_cast<unsigned short*>(y)[row + o] = f32_to_bf16(sum);
+    } else if (epilogue == EPI_GELU_F32) {
+      reinterpret_cast<float*>(y)[row + o] = gelu_tanh_f32(sum);
     } else {  // EPI_RESIDUAL_BF16
       const float r = bf16_to_f32(residual[row + o]);
       reinterpret_cast<unsigned short*>(y)[row + o] = f32_to_bf16(r + sum);
@@ -72,6 +88,7 @@ static __global__ void gemv_reduce_kernel_impl(
 static inline torch::Tensor launch_gemv_reduce(
     torch::Tensor partials,
     c10::optional<torch::Tensor> residual,
+    c10::optional<torch::Tensor> bias,
     int n_splits,
     int batch,
     int out_dim,
@@ -81,21 +98,30 @@ static inline torch::Tensor launch_gemv_reduce(
   const int half = out_dim / 2;
   const int n_out = (epilogue == EPI_SWIGLU_F32) ? half : out_dim;
   torch::Tensor y;
-  if (epilogue == EPI_PLAIN_F32) y = torch::empty({batch, out_dim}, f32_opts);
-  else if (epilogue == EPI_SWIGLU_F32) y = torch::empty({batch, half}, f32_opts);
-  else y = torch::empty({batch, out_dim}, bf16_opts);
+  if (epilogue == EPI_PLAIN_F32 || epilogue == EPI_GELU_F32)
+    y = torch::empty({batch, out_dim}, f32_opts);
+  else if (epilogue == EPI_SWIGLU_F32)
+    y = torch::empty({batch, half}, f32_opts);
+  else
+    y = torch::empty({batch, out_dim}, bf16_opts);
   const unsigned short* res_p = nullptr;
   if (epilogue == EPI_RESIDUAL_BF16) {
     TORCH_CHECK(residual.has_value(), "residual required for EPI_RESIDUAL_BF16");
     TORCH_CHECK(residual->is_contiguous());
     res_p = reinterpret_cast<const unsigned short*>(residual->data_ptr());
   }
+  const unsigned short* bias_p = nullptr;
+  if (bias.has_value() && bias->defined() && bias->numel() > 0) {
+    TORCH_CHECK(bias->is_contiguous() && bias->dtype() == torch::kBFloat16);
+    TORCH_CHECK(bias->numel() == out_dim, "bias must be [out_dim] (pre-activation)");
+    bias_p = reinterpret_cast<const unsigned short*>(bias->data_ptr());
+  }
   const int total = batch * n_out;
   const int rthreads = 64;  // small blocks -> enough workgroups to spread CUs
   int rblocks = (total + rthreads - 1) / rthreads;
   auto stream = c10::hip::getCurrentHIPStream().stream();
   gemv_reduce_kernel_impl<<<rblocks, rthreads, 0, stream>>>(
-      partials.data_ptr<float>(), res_p, y.data_ptr(), n_splits, batch, out_dim, epilogue);
+      partials.data_ptr<float>(), res_p, bias_p, y.data_ptr(), n_splits, batch, out_dim, epilogue);
   HIP_CHECK_LAST();
   return y;
 }

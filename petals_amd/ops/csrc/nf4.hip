@@ -245,7 +245,8 @@ torch::Tensor gemv_nf4(
     torch::Tensor workspace,
     c10::optional<torch::Tensor> residual,
     int64_t epilogue,
-    int64_t splits_override) {
+    int64_t splits_override,
+    c10::optional<torch::Tensor> bias) {  // [out] bf16, added pre-activation
   TORCH_CHECK(packed.is_cuda() && packed.dtype() == torch::kUInt8);
   TORCH_CHECK(x.dtype() == torch::kFloat32 && x.dim() == 2);
   const int in_dim = packed.size(0);
@@ -287,6 +288,6 @@ torch::Tensor gemv_nf4(
   HIP_CHECK_LAST();
 
   torch::Tensor y = launch_gemv_reduce(
-      partials, residual, (int)splits, batch, out_dim, (int)epilogue, f32opts, absmax.options());
+      partials, residual, bias, (int)splits, batch, out_dim, (int)epilogue, f32opts, absmax.options());
   return y;
 }

@@ -43,6 +43,8 @@ __global__ __launch_bounds__(WAVES * 64) void attn_prefill_kernel(
     const unsigned short* __restrict__ k,   // [B, KVH, Lmax, HD]
     const unsigned short* __restrict__ v,   // [B, KVH, Lmax, HD]
     unsigned short* __restrict__ out,       // [B, QH, S, HD]
+    const float* __restrict__ alibi,        // [QH] ALiBi slopes or null (adds slope*key_abs;
+                                            // the per-row constant cancels in softmax)
     int q_heads,
     int kv_heads,
     int s_q,
@@ -63,6 +65,7 @@ __global__ __launch_bounds__(WAVES * 64) void attn_prefill_kernel(
   const int col = lane & 15;      // fragment column (and C col)
   const int hi = lane >> 4;       // fragment k-group (and C row group)
   const int q0 = q0_wg + wave * QTILE;  // this wave's first q row
+  const float slope = alibi ? alibi[qh] : 0.f;  // wave-uniform (one head per wg)
 
   // LDS: K tile row-major [KVT][HD+KPAD]; V tile transposed+swizzled
   // [HD][KVT+KPAD]; per-wave P scratch [QTILE][KVT+KPAD]
@@ -154,8 +157,8 @@ __global__ __launch_bounds__(WAVES * 64) void attn_prefill_kernel(
       float s[NB];
 #pragma unroll
       for (int nb = 0; nb < NB; ++nb) {
-        s[nb] = s_acc[nb][r] * scale;
         const int key = j0 + nb * 16 + col;
+        s[nb] = s_acc[nb][r] * scale + slope * key;
         if (key >= tile_n + j0 || (causal && key > q_abs) || qrow >= s_q) s[nb] = NEG_SENTINEL;
       }
       float mx = NEG_SENTINEL;
@@ -230,7 +233,8 @@ torch::Tensor attn_prefill_fused(
     int64_t kv_len,
     int64_t kv_offset,
     double scale,
-    bool causal) {
+    bool causal,
+    c10::optional<torch::Tensor> alibi_slopes) {  // [QH] f32
   TORCH_CHECK(q.is_cuda() && q.dtype() == torch::kBFloat16 && q.dim() == 4);
   TORCH_CHECK(k.dtype() == torch::kBFloat16 && k.dim() == 4);
   const int B = q.size(0), QH = q.size(1), S = q.size(2), HD = q.size(3);
@@ -248,6 +252,12 @@ torch::Tensor attn_prefill_fused(
     return s ? std::atoi(s) : 64;
   }();
 
+  const float* alibi_p = nullptr;
+  if (alibi_slopes.has_value() && alibi_slopes->defined() && alibi_slopes->numel() > 0) {
+    TORCH_CHECK(alibi_slopes->is_contiguous() && alibi_slopes->dtype() == torch::kFloat32);
+    TORCH_CHECK(alibi_slopes->numel() == QH, "alibi slopes must be [q_heads]");
+    alibi_p = alibi_slopes->data_ptr<float>();
+  }
   const auto* qp = reinterpret_cast<const unsigned short*>(q.data_ptr());
   const auto* kp = reinterpret_cast<const unsigned short*>(k.data_ptr());
   const auto* vp = reinterpret_cast<const unsigned short*>(v.data_ptr());
@@ -256,17 +266,17 @@ torch::Tensor attn_prefill_fused(
   if (HD == 128) {
     if (kvt_env == 32)
       attn_prefill_kernel<128, 32><<<grid, WAVES * WAVE, 0, stream>>>(
-          qp, kp, vp, op, QH, KVH, S, LMAX, (int)kv_len, (int)kv_offset, sc, causal ? 1 : 0);
+          qp, kp, vp, op, alibi_p, QH, KVH, S, LMAX, (int)kv_len, (int)kv_offset, sc, causal ? 1 : 0);
     else
       attn_prefill_kernel<128, 64><<<grid, WAVES * WAVE, 0, stream>>>(
-          qp, kp, vp, op, QH, KVH, S, LMAX, (int)kv_len, (int)kv_offset, sc, causal ? 1 : 0);
+          qp, kp, vp, op, alibi_p, QH, KVH, S, LMAX, (int)kv_len, (int)kv_offset, sc, causal ? 1 : 0);
   } else {
     if (kvt_env == 32)
       attn_prefill_kernel<64, 32><<<grid, WAVES * WAVE, 0, stream>>>(
-          qp, kp, vp, op, QH, KVH, S, LMAX, (int)kv_len, (int)kv_offset, sc, causal ? 1 : 0);
+          qp, kp, vp, op, alibi_p, QH, KVH, S, LMAX, (int)kv_len, (int)kv_offset, sc, causal ? 1 : 0);
     else
       attn_prefill_kernel<64, 64><<<grid, WAVES * WAVE, 0, stream>>>(
-          qp, kp, vp, op, QH, KVH, S, LMAX, (int)kv_len, (int)kv_offset, sc, causal ? 1 : 0);
+          qp, kp, vp, op, alibi_p, QH, KVH, S, LMAX, (int)kv_len, (int)kv_offset, sc, causal ? 1 : 0);
   }
   HIP_CHECK_LAST();
   return out;

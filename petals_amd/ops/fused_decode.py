@@ -32,6 +32,7 @@ _EPI_PLAIN_F32 = 0
 _EPI_PLAIN_BF16 = 1
 _EPI_RESIDUAL_BF16 = 2
 _EPI_SWIGLU_F32 = 3
+_EPI_GELU_F32 = 4
 
 _workspaces: Dict[Tuple[torch.device, str], torch.Tensor] = {}
 
@@ -126,13 +127,13 @@ class _FastWeight:
     def shape(self):
         return (self.in_dim, self.out_dim)
 
-    def _gemv_raw(self, x, ws, residual, epilogue, splits):
+    def _gemv_raw(self, x, ws, residual, epilogue, splits, bias=None):
         if self.quant == "nf4":
-            return self.hip.gemv_nf4(self.packed, self.absmax, x, ws, residual, epilogue, splits)
-        return self.hip.gemv_bf16(self.t, x, ws, residual, epilogue, splits)
+            return self.hip.gemv_nf4(self.packed, self.absmax, x, ws, residual, epilogue, splits, bias)
+        return self.hip.gemv_bf16(self.t, x, ws, residual, epilogue, splits, bias)
 
-    def gemv(self, x, ws, residual, epilogue):
-        return self._gemv_raw(x, ws, residual, epilogue, self.splits)
+    def gemv(self, x, ws, residual, epilogue, bias=None):
+        return self._gemv_raw(x, ws, residual, epilogue, self.splits, bias)
 
     def dense(self) -> torch.Tensor:
         """bf16 [in, out] view for prefill GEMMs (dequantized on the fly for NF4)."""
@@ -395,3 +396,341 @@ class LlamaFastPath:
         h2 = hidden + o
         xn2 = self.hip.rms_norm(h2, self.ln2_w, self.eps)
         return h2 + self._mlp_dense(xn2, adapter, autograd=False).to(h2.dtype)
+
+
+# ---------------------------------------------------------------------------
+# BLOOM: LayerNorm -> fused-QKV (per-head interleaved in the checkpoint,
+# re-permuted to [q|k|v] at load) -> ALiBi MHA -> LayerNorm -> GELU MLP,
+# biases on every projection. Decode runs the same 8-kernel chain as Llama
+# with layer_norm/kv_cache_write/alibi variants (all graph-capturable).
+# Parity: reference models/bloom/block.py wraps HF BloomBlock in torch.
+# ---------------------------------------------------------------------------
+
+
+def _bloom_qkv_perm(qh: int, hd: int) -> torch.Tensor:
+    """new [q|k|v] column index -> old per-head-interleaved [h,3,hd] index."""
+    idx = torch.arange(qh * hd)
+    heads, d = idx // hd, idx % hd
+    perm = torch.empty(3 * qh * hd, dtype=torch.long)
+    for which in range(3):
+        perm[which * qh * hd + idx] = heads * 3 * hd + which * hd + d
+    return perm
+
+
+class BloomFastPath:
+    graph_safe = True
+
+    def __init__(self, block, hip_ops, quant: str = "none"):
+        cfg = block.config
+        self.hip = hip_ops
+        self.cfg = cfg
+        self.quant = quant
+        self.hd = cfg.head_dim
+        self.qh = cfg.num_attention_heads
+        self.kh = self.qh  # MHA
+        self.gq = 1
+        self.scale = 1.0 / math.sqrt(self.hd)
+        self.eps = cfg.layer_norm_eps
+        self.post_ln_residual = bool(getattr(cfg, "apply_residual_connection_post_layernorm", False))
+
+        attn = block.self_attention
+        device = attn.query_key_value.weight.device
+
+        def t(w):
+            return w.detach().to(torch.bfloat16).t().contiguous()
+
+        def bias(lin):
+            return lin.bias.detach().to(torch.bfloat16).contiguous()
+
+        perm = _bloom_qkv_perm(self.qh, self.hd).to(device)
+        self.wqkv_t = _FastWeight(t(attn.query_key_value.weight)[:, perm].contiguous(), hip_ops, quant)
+        self.qkv_bias = attn.query_key_value.bias.detach().to(torch.bfloat16)[perm].contiguous()
+        self.wo_t = _FastWeight(t(attn.dense.weight), hip_ops, quant)
+        self.o_bias = bias(attn.dense)
+        mlp = block.mlp
+        self.w_h4h = _FastWeight(t(mlp.dense_h_to_4h.weight), hip_ops, quant)
+        self.b_h4h = bias(mlp.dense_h_to_4h)
+        self.w_4hh = _FastWeight(t(mlp.dense_4h_to_h.weight), hip_ops, quant)
+        self.b_4hh = bias(mlp.dense_4h_to_h)
+        self.ln1_w = block.input_layernorm.weight.detach().to(torch.bfloat16).contiguous()
+        self.ln1_b = block.input_layernorm.bias.detach().to(torch.bfloat16).contiguous()
+        self.ln2_w = block.post_attention_layernorm.weight.detach().to(torch.bfloat16).contiguous()
+        self.ln2_b = block.post_attention_layernorm.bias.detach().to(torch.bfloat16).contiguous()
+        self._empty_bf16 = torch.empty(0, device=device, dtype=torch.bfloat16)
+        for lin in (attn.query_key_value, attn.dense, mlp.dense_h_to_4h, mlp.dense_4h_to_h):
+            lin.weight.data = self._empty_bf16
+
+        self.slopes = reference.build_alibi_slopes(self.qh).to(device=device, dtype=torch.float32).contiguous()
+        self._pos = torch.zeros(1, dtype=torch.int32, device=device)
+        self._kv_len = torch.zeros(1, dtype=torch.int32, device=device)
+        self.device = device
+        self._empty_f32 = torch.empty(0, dtype=torch.float32, device=device)
+
+    def _max_gemv_out(self) -> int:
+        return max(self.wqkv_t.shape[1], self.w_h4h.shape[1])
+
+    # ------------------------------------------------------------- decode
+
+    @torch.inference_mode()
+    def decode_step(self, hidden, k_cache, v_cache, prefix_length: int = -1, ctx=None, adapter=None):
+        assert adapter is None, "LoRA adapters are served on the generic path for BLOOM"
+        B, H = hidden.shape[0], hidden.shape[-1]
+        h = hidden.view(B, H)
+        if h.dtype != torch.bfloat16:
+            h = h.to(torch.bfloat16)
+        h = h.contiguous()
+        if ctx is None:
+            assert prefix_length >= 0
+            self._pos.fill_(prefix_length)
+            self._kv_len.fill_(prefix_length + 1)
+            pos, kv_len = self._pos, self._kv_len
+        else:
+            pos, kv_len = ctx.pos, ctx.kv_len
+        ws = _get_ws(self.device, "gemv", 64 * B * self._max_gemv_out())
+
+        xn = self.hip.layer_norm_f32out(h, self.ln1_w, self.ln1_b, self.eps)
+        res1 = self.hip.layer_norm(h, self.ln1_w, self.ln1_b, self.eps) if self.post_ln_residual else h
+        qkv = self.wqkv_t.gemv(xn, ws, None, _EPI_PLAIN_F32, bias=self.qkv_bias)
+        self.hip.kv_cache_write(qkv, pos, k_cache[:B], v_cache[:B], self.qh, self.kh)
+        q = qkv[:, : self.qh * self.hd]
+        attn = self.hip.attn_decode_fused(
+            q.contiguous(), k_cache[:B], v_cache[:B], kv_len, self.gq, 0,
+            self._empty_f32, self._empty_f32, self.scale, self.slopes,
+        )
+        h2 = self.wo_t.gemv(attn, ws, res1, _EPI_RESIDUAL_BF16, bias=self.o_bias)
+        xn2 = self.hip.layer_norm_f32out(h2, self.ln2_w, self.ln2_b, self.eps)
+        res2 = self.hip.layer_norm(h2, self.ln2_w, self.ln2_b, self.eps) if self.post_ln_residual else h2
+        act = self.w_h4h.gemv(xn2, ws, None, _EPI_GELU_F32, bias=self.b_h4h)
+        h3 = self.w_4hh.gemv(act, ws, res2, _EPI_RESIDUAL_BF16, bias=self.b_4hh)
+        return h3.view(B, 1, H)
+
+    # -------------------------------------------------- prefill / training
+
+    def _split_heads(self, qkv, B, S):
+        qd = self.qh * self.hd
+        q = qkv[..., :qd].view(B, S, self.qh, self.hd).transpose(1, 2)
+        k = qkv[..., qd : 2 * qd].view(B, S, self.kh, self.hd).transpose(1, 2)
+        v = qkv[..., 2 * qd :].view(B, S, self.kh, self.hd).transpose(1, 2)
+        return q, k, v
+
+    def _body(self, hidden, kv_cache, prefix_length, autograd: bool):
+        B, S, H = hidden.shape
+        hidden = hidden.to(torch.bfloat16)
+        end = prefix_length + S
+        fln = torch.nn.functional.layer_norm
+
+        def ln(x, w, b):
+            if autograd or x.device.type != "cuda":
+                return fln(x.float(), (H,), w.float(), b.float(), self.eps).to(torch.bfloat16)
+            return self.hip.layer_norm(x, w, b, self.eps)
+
+        xn = ln(hidden, self.ln1_w, self.ln1_b)
+        res1 = xn if self.post_ln_residual else hidden
+        qkv = torch.matmul(xn, self.wqkv_t.dense()) + self.qkv_bias
+        q, k, v = self._split_heads(qkv, B, S)
+        if kv_cache is not None and not autograd:
+            k_cache, v_cache = kv_cache
+            k_cache[:B, :, prefix_length:end].copy_(k)
+            v_cache[:B, :, prefix_length:end].copy_(v)
+            attn = self.hip.attn_prefill_fused(
+                q.contiguous(), k_cache[:B].contiguous(), v_cache[:B].contiguous(),
+                end, prefix_length, self.scale, True, self.slopes,
+            )
+        elif not autograd and hidden.device.type == "cuda":
+            attn = self.hip.attn_prefill_fused(
+                q.contiguous(), k.contiguous(), v.contiguous(), S, 0, self.scale, True, self.slopes
+            )
+        else:
+            k_pos = torch.arange(end, device=hidden.device, dtype=torch.float32)
+            bias = (self.slopes.to(hidden.device)[:, None, None] * k_pos[None, None, :]).unsqueeze(0)
+            attn = reference.attention(q.float(), k.float(), v.float(), causal=True,
+                                       kv_offset=prefix_length, attn_bias=bias.float())
+        attn = attn.transpose(1, 2).reshape(B, S, H).to(torch.bfloat16)
+        o = torch.matmul(attn, self.wo_t.dense()) + self.o_bias
+        h2 = res1 + o
+        xn2 = ln(h2, self.ln2_w, self.ln2_b)
+        res2 = xn2 if self.post_ln_residual else h2
+        inter = torch.matmul(xn2, self.w_h4h.dense()) + self.b_h4h
+        act = reference.gelu(inter.float()).to(torch.bfloat16)
+        out = torch.matmul(act, self.w_4hh.dense()) + self.b_4hh
+        return res2 + out
+
+    def forward(self, hidden, kv_cache, prefix_length, adapter=None):
+        assert adapter is None
+        return self._body(hidden, kv_cache, prefix_length, autograd=False)
+
+    def forward_autograd(self, hidden, prefix_length: int = 0, adapter=None):
+        assert adapter is None
+        return self._body(hidden, None, prefix_length, autograd=True)
+
+
+# ---------------------------------------------------------------------------
+# Falcon (new_decoder_architecture: 40B/180B): parallel attention + MLP with
+# two LayerNorms on the same input, rope GQA, no biases, GELU MLP.
+# Decode: ln_attn/ln_mlp -> qkv gemv -> rope+cache write -> GQA flash decode
+# -> dense gemv (+resid) -> gelu gemv -> down gemv (+resid) = 9 kernels.
+# Parity: reference models/falcon/block.py wraps HF FalconDecoderLayer and
+# CUDA-graphs only the QKV split.
+# ---------------------------------------------------------------------------
+
+
+def _falcon_qkv_perm(qh: int, kh: int, hd: int) -> torch.Tensor:
+    """new [q|k|v] column -> old [(kv group)(gq q heads, k, v)(hd)] column."""
+    gq = qh // kh
+    group = (gq + 2) * hd
+    perm = torch.empty((qh + 2 * kh) * hd, dtype=torch.long)
+    iq = torch.arange(qh * hd)
+    kvg, rem = (iq // hd) // gq, iq % (hd)
+    slot = (iq // hd) % gq
+    perm[iq] = kvg * group + slot * hd + rem
+    ik = torch.arange(kh * hd)
+    perm[qh * hd + ik] = (ik // hd) * group + gq * hd + (ik % hd)
+    perm[(qh + kh) * hd + ik] = (ik // hd) * group + (gq + 1) * hd + (ik % hd)
+    return perm
+
+
+class FalconFastPath:
+    graph_safe = True
+
+    def __init__(self, block, hip_ops, quant: str = "none"):
+        cfg = block.config
+        assert cfg.new_decoder_architecture, "fused Falcon path covers the new-decoder architecture"
+        self.hip = hip_ops
+        self.cfg = cfg
+        self.quant = quant
+        self.hd = cfg.head_dim
+        self.qh = cfg.num_attention_heads
+        self.kh = cfg.n_kv_heads
+        self.gq = self.qh // self.kh
+        self.scale = 1.0 / math.sqrt(self.hd)
+        self.eps = cfg.layer_norm_eps
+
+        attn = block.self_attention
+        device = attn.query_key_value.weight.device
+
+        def t(w):
+            return w.detach().to(torch.bfloat16).t().contiguous()
+
+        perm = _falcon_qkv_perm(self.qh, self.kh, self.hd).to(device)
+        self.wqkv_t = _FastWeight(t(attn.query_key_value.weight)[:, perm].contiguous(), hip_ops, quant)
+        self.wo_t = _FastWeight(t(attn.dense.weight), hip_ops, quant)
+        mlp = block.mlp
+        self.w_h4h = _FastWeight(t(mlp.dense_h_to_4h.weight), hip_ops, quant)
+        self.w_4hh = _FastWeight(t(mlp.dense_4h_to_h.weight), hip_ops, quant)
+        self.ln_attn_w = block.ln_attn.weight.detach().to(torch.bfloat16).contiguous()
+        self.ln_attn_b = block.ln_attn.bias.detach().to(torch.bfloat16).contiguous()
+        self.ln_mlp_w = block.ln_mlp.weight.detach().to(torch.bfloat16).contiguous()
+        self.ln_mlp_b = block.ln_mlp.bias.detach().to(torch.bfloat16).contiguous()
+        self._empty_bf16 = torch.empty(0, device=device, dtype=torch.bfloat16)
+        for lin in (attn.query_key_value, attn.dense, mlp.dense_h_to_4h, mlp.dense_4h_to_h):
+            lin.weight.data = self._empty_bf16
+
+        self.rope_cos: Optional[torch.Tensor] = None
+        self.rope_sin: Optional[torch.Tensor] = None
+        self._pos = torch.zeros(1, dtype=torch.int32, device=device)
+        self._kv_len = torch.zeros(1, dtype=torch.int32, device=device)
+        self.device = device
+        self._empty_f32 = torch.empty(0, dtype=torch.float32, device=device)
+
+    def _ensure_rope(self, needed: int):
+        if self.rope_cos is None or self.rope_cos.shape[0] < needed:
+            cos, sin = reference.build_rope_cache(
+                self.hd, max(needed, self.cfg.max_position_embeddings), theta=self.cfg.rope_theta
+            )
+            self.rope_cos = cos.to(self.device, torch.float32).contiguous()
+            self.rope_sin = sin.to(self.device, torch.float32).contiguous()
+
+    def _max_gemv_out(self) -> int:
+        return max(self.wqkv_t.shape[1], self.w_h4h.shape[1])
+
+    @torch.inference_mode()
+    def decode_step(self, hidden, k_cache, v_cache, prefix_length: int = -1, ctx=None, adapter=None):
+        assert adapter is None, "LoRA adapters are served on the generic path for Falcon"
+        B, H = hidden.shape[0], hidden.shape[-1]
+        h = hidden.view(B, H)
+        if h.dtype != torch.bfloat16:
+            h = h.to(torch.bfloat16)
+        h = h.contiguous()
+        if ctx is None:
+            assert prefix_length >= 0
+            self._ensure_rope(prefix_length + 1)
+            self._pos.fill_(prefix_length)
+            self._kv_len.fill_(prefix_length + 1)
+            pos, kv_len = self._pos, self._kv_len
+        else:
+            self._ensure_rope(k_cache.shape[2])
+            pos, kv_len = ctx.pos, ctx.kv_len
+        ws = _get_ws(self.device, "gemv", 64 * B * self._max_gemv_out())
+
+        xn_attn = self.hip.layer_norm_f32out(h, self.ln_attn_w, self.ln_attn_b, self.eps)
+        xn_mlp = self.hip.layer_norm_f32out(h, self.ln_mlp_w, self.ln_mlp_b, self.eps)
+        qkv = self.wqkv_t.gemv(xn_attn, ws, None, _EPI_PLAIN_F32)
+        self.hip.rope_cache_write(
+            qkv, self.rope_cos, self.rope_sin, pos, k_cache[:B], v_cache[:B], self.qh, self.kh
+        )
+        q = qkv[:, : self.qh * self.hd]
+        attn = self.hip.attn_decode_fused(
+            q.contiguous(), k_cache[:B], v_cache[:B], kv_len, self.gq, 0,
+            self._empty_f32, self._empty_f32, self.scale,
+        )
+        h2 = self.wo_t.gemv(attn, ws, h, _EPI_RESIDUAL_BF16)  # resid + attn
+        act = self.w_h4h.gemv(xn_mlp, ws, None, _EPI_GELU_F32)
+        h3 = self.w_4hh.gemv(act, ws, h2, _EPI_RESIDUAL_BF16)  # ... + mlp
+        return h3.view(B, 1, H)
+
+    def _split_heads(self, qkv, B, S):
+        qd, kd = self.qh * self.hd, self.kh * self.hd
+        q = qkv[..., :qd].view(B, S, self.qh, self.hd).transpose(1, 2)
+        k = qkv[..., qd : qd + kd].view(B, S, self.kh, self.hd).transpose(1, 2)
+        v = qkv[..., qd + kd :].view(B, S, self.kh, self.hd).transpose(1, 2)
+        return q, k, v
+
+    def _body(self, hidden, kv_cache, prefix_length, autograd: bool):
+        B, S, H = hidden.shape
+        hidden = hidden.to(torch.bfloat16)
+        end = prefix_length + S
+        self._ensure_rope(end)
+        fln = torch.nn.functional.layer_norm
+
+        def ln(x, w, b):
+            if autograd or x.device.type != "cuda":
+                return fln(x.float(), (H,), w.float(), b.float(), self.eps).to(torch.bfloat16)
+            return self.hip.layer_norm(x, w, b, self.eps)
+
+        xn_attn = ln(hidden, self.ln_attn_w, self.ln_attn_b)
+        xn_mlp = ln(hidden, self.ln_mlp_w, self.ln_mlp_b)
+        qkv = torch.matmul(xn_attn, self.wqkv_t.dense())
+        q, k, v = self._split_heads(qkv, B, S)
+        pos = torch.arange(prefix_length, end, device=hidden.device)
+        if autograd:
+            q, k = reference.apply_rope(q, k, self.rope_cos, self.rope_sin, pos)
+            attn = reference.attention(q.float(), k.float(), v.float(), causal=True, kv_offset=prefix_length)
+        else:
+            posb = pos.unsqueeze(0).expand(B, S).contiguous()
+            q, k = self.hip.apply_rope(q.contiguous(), k.contiguous(), self.rope_cos, self.rope_sin, posb)
+            if kv_cache is not None:
+                k_cache, v_cache = kv_cache
+                k_cache[:B, :, prefix_length:end].copy_(k)
+                v_cache[:B, :, prefix_length:end].copy_(v)
+                attn = self.hip.attn_prefill_fused(
+                    q.contiguous(), k_cache[:B].contiguous(), v_cache[:B].contiguous(),
+                    end, prefix_length, self.scale, True,
+                )
+            else:
+                attn = self.hip.attn_prefill_fused(
+                    q.contiguous(), k.contiguous(), v.contiguous(), S, 0, self.scale, True
+                )
+        attn = attn.transpose(1, 2).reshape(B, S, self.qh * self.hd).to(torch.bfloat16)
+        o = torch.matmul(attn, self.wo_t.dense())
+        inter = torch.matmul(xn_mlp, self.w_h4h.dense())
+        act = reference.gelu(inter.float()).to(torch.bfloat16)
+        mlp_out = torch.matmul(act, self.w_4hh.dense())
+        return hidden + o + mlp_out
+
+    def forward(self, hidden, kv_cache, prefix_length, adapter=None):
+        assert adapter is None
+        return self._body(hidden, kv_cache, prefix_length, autograd=False)
+
+    def forward_autograd(self, hidden, prefix_length: int = 0, adapter=None):
+        assert adapter is None
+        return self._body(hidden, None, prefix_length, autograd=True)

@@ -330,3 +330,130 @@ def test_gpu_speculative_and_lora_e2e(tmp_path):
     finally:
         server.shutdown()
         boot.shutdown()
+
+
+@requires_gpu
+def test_layer_norm(hip):
+    torch.manual_seed(3)
+    x = (torch.randn(5, 1024, device="cuda") * 2 + 0.3).to(torch.bfloat16)
+    w = torch.randn(1024, device="cuda").to(torch.bfloat16)
+    b = torch.randn(1024, device="cuda").to(torch.bfloat16)
+    ref = torch.nn.functional.layer_norm(x.float().cpu(), (1024,), w.float().cpu(), b.float().cpu(), 1e-5)
+    out = hip.layer_norm(x, w, b, 1e-5)
+    assert torch.allclose(out.float().cpu(), ref, atol=3e-2, rtol=3e-2), (out.float().cpu() - ref).abs().max()
+    out32 = hip.layer_norm_f32out(x, w, b, 1e-5)
+    assert out32.dtype == torch.float32
+    assert torch.allclose(out32.cpu(), ref, atol=3e-2, rtol=3e-2)
+
+
+@requires_gpu
+def test_gemv_bias_and_gelu(hip):
+    torch.manual_seed(4)
+    in_dim, out_dim, B = 512, 768, 2
+    wt = (torch.randn(in_dim, out_dim, device="cuda") * 0.05).to(torch.bfloat16)
+    x = torch.randn(B, in_dim, device="cuda")
+    bias = (torch.randn(out_dim, device="cuda") * 0.5).to(torch.bfloat16)
+    ws = torch.empty(0, device="cuda")
+    ref_lin = x.cpu() @ wt.float().cpu() + bias.float().cpu()
+    # plain f32 + bias
+    out = hip.gemv_bf16(wt, x, ws, None, 0, 0, bias)
+    assert torch.allclose(out.cpu(), ref_lin, atol=2e-2, rtol=2e-2)
+    # gelu(tanh) epilogue + bias
+    out_g = hip.gemv_bf16(wt, x, ws, None, 4, 0, bias)
+    ref_g = torch.nn.functional.gelu(ref_lin, approximate="tanh")
+    assert torch.allclose(out_g.cpu(), ref_g, atol=2e-2, rtol=2e-2), (out_g.cpu() - ref_g).abs().max()
+    # nf4 path with bias
+    packed, absmax = hip.nf4_quantize(wt)
+    wt_dq = hip.nf4_dequantize(packed, absmax)
+    ref_nf4 = torch.nn.functional.gelu(x.cpu() @ wt_dq.float().cpu() + bias.float().cpu(), approximate="tanh")
+    out_n = hip.gemv_nf4(packed, absmax, x, ws, None, 4, 0, bias)
+    assert torch.allclose(out_n.cpu(), ref_nf4, atol=2e-2, rtol=2e-2)
+
+
+@requires_gpu
+def test_attn_decode_alibi(hip):
+    """ALiBi decode attention vs the reference with an explicit bias matrix."""
+    from petals_amd.ops import reference
+
+    torch.manual_seed(6)
+    B, H, hd, kv_len, lmax = 2, 8, 64, 73, 96
+    q = torch.randn(B, H, 1, hd, device="cuda").float()
+    k_cache = torch.zeros(B, H, lmax, hd, device="cuda", dtype=torch.bfloat16)
+    v_cache = torch.zeros_like(k_cache)
+    k_cache[:, :, :kv_len] = (torch.randn(B, H, kv_len, hd, device="cuda") * 0.5).to(torch.bfloat16)
+    v_cache[:, :, :kv_len] = (torch.randn(B, H, kv_len, hd, device="cuda") * 0.5).to(torch.bfloat16)
+    slopes = reference.build_alibi_slopes(H).to("cuda", torch.float32)
+    kvl = torch.tensor([kv_len], dtype=torch.int32, device="cuda")
+    empty = torch.empty(0, dtype=torch.float32, device="cuda")
+    scale = 1.0 / math.sqrt(hd)
+    out = hip.attn_decode_fused(
+        q.view(B, H * hd).contiguous(), k_cache, v_cache, kvl, 1, 0, empty, empty, scale, slopes
+    )
+    k_pos = torch.arange(kv_len, dtype=torch.float32)
+    bias = (slopes.cpu()[:, None, None] * k_pos[None, None, :]).unsqueeze(0)
+    ref = reference.attention(
+        q.float().cpu(), k_cache[:, :, :kv_len].float().cpu(), v_cache[:, :, :kv_len].float().cpu(),
+        causal=False, attn_bias=bias,
+    ).view(B, H * hd)
+    assert torch.allclose(out.float().cpu(), ref, atol=3e-2, rtol=3e-2), (out.float().cpu() - ref).abs().max()
+
+
+@requires_gpu
+def test_kv_cache_write(hip):
+    torch.manual_seed(7)
+    B, qh, kh, hd, lmax, pos = 2, 4, 4, 64, 32, 11
+    qkv = torch.randn(B, (qh + 2 * kh) * hd, device="cuda")
+    k_cache = torch.zeros(B, kh, lmax, hd, device="cuda", dtype=torch.bfloat16)
+    v_cache = torch.zeros_like(k_cache)
+    p = torch.tensor([pos], dtype=torch.int32, device="cuda")
+    hip.kv_cache_write(qkv, p, k_cache, v_cache, qh, kh)
+    k_ref = qkv[:, qh * hd : (qh + kh) * hd].view(B, kh, hd)
+    v_ref = qkv[:, (qh + kh) * hd :].view(B, kh, hd)
+    assert torch.allclose(k_cache[:, :, pos].float(), k_ref, atol=1e-2, rtol=1e-2)
+    assert torch.allclose(v_cache[:, :, pos].float(), v_ref, atol=1e-2, rtol=1e-2)
+    assert k_cache[:, :, pos + 1].abs().max() == 0 and k_cache[:, :, pos - 1].abs().max() == 0
+
+
+def _block_fused_vs_cpu(preset: str, hidden: int):
+    from petals_amd.models import get_model_block
+    from petals_amd.models.config_base import load_model_config
+    from petals_amd.server.from_pretrained import init_random_block_
+
+    cfg = load_model_config(preset)
+    blk_cpu = get_model_block(cfg, 0)
+    init_random_block_(blk_cpu, cfg, 0)
+    blk_cpu = blk_cpu.float().eval()
+
+    blk_gpu = get_model_block(cfg, 0)
+    blk_gpu.load_state_dict(blk_cpu.state_dict())
+    blk_gpu = blk_gpu.to("cuda", torch.bfloat16).eval().optimize_for_inference()
+    assert blk_gpu._fast is not None, f"{preset} must take the fused path"
+
+    torch.manual_seed(5)
+    B, S = 2, 9
+    x = torch.randn(B, S, hidden) * 0.5
+    ks, vs = blk_cpu.kv_cache_shape(B, 32)
+    kc, vc = torch.zeros(ks), torch.zeros(vs)
+    kg = torch.zeros(ks, device="cuda", dtype=torch.bfloat16)
+    vg = torch.zeros(vs, device="cuda", dtype=torch.bfloat16)
+
+    y_cpu = [blk_cpu(x[:, :6], kv_cache=(kc, vc), prefix_length=0)]
+    y_gpu = [blk_gpu(x[:, :6].cuda().bfloat16(), kv_cache=(kg, vg), prefix_length=0)]
+    for t in range(6, S):
+        y_cpu.append(blk_cpu(x[:, t : t + 1], kv_cache=(kc, vc), prefix_length=t))
+        y_gpu.append(blk_gpu(x[:, t : t + 1].cuda().bfloat16(), kv_cache=(kg, vg), prefix_length=t))
+    ref = torch.cat(y_cpu, 1)
+    out = torch.cat([y.float().cpu() for y in y_gpu], 1)
+    assert torch.allclose(out, ref, atol=0.05, rtol=0.05), (out - ref).abs().max()
+
+
+@requires_gpu
+def test_bloom_block_fast_decode_matches_cpu(hip):
+    """BLOOM fused path (LayerNorm + ALiBi + GELU biases) vs fp32 CPU block."""
+    _block_fused_vs_cpu("test-bloom-hd64", 256)
+
+
+@requires_gpu
+def test_falcon_block_fast_decode_matches_cpu(hip):
+    """Falcon new-decoder fused path (parallel attn+MLP, rope GQA) vs CPU."""
+    _block_fused_vs_cpu("test-falcon-hd64", 256)
