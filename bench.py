@@ -66,6 +66,8 @@ def main():
         dtype = torch.float32
 
     config = load_model_config(args.model)
+    if os.environ.get("PETALS_AMD_BENCH_BLOCKS"):  # debug: cap the span size
+        config.num_hidden_layers = int(os.environ["PETALS_AMD_BENCH_BLOCKS"])
     H = config.hidden_size
     B = args.batch
     max_len = args.prompt_len + args.warmup + args.steps + 8

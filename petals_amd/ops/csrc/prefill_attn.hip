@@ -37,7 +37,7 @@ using f32x4 = __attribute__((ext_vector_type(4))) float;
 #define VT_BYTE(dim, key_byte) \
   ((((unsigned)(dim)) * ((KVT + KPAD) * 2) + (unsigned)(key_byte)) ^ ((((unsigned)(dim) >> 3) & 7u) << 4))
 
-template <int HD, int KVT>
+template <int HD, int KVT, bool ALIBI>
 __global__ __launch_bounds__(WAVES * 64) void attn_prefill_kernel(
     const unsigned short* __restrict__ q,   // [B, QH, S, HD]
     const unsigned short* __restrict__ k,   // [B, KVH, Lmax, HD]
@@ -65,7 +65,7 @@ __global__ __launch_bounds__(WAVES * 64) void attn_prefill_kernel(
   const int col = lane & 15;      // fragment column (and C col)
   const int hi = lane >> 4;       // fragment k-group (and C row group)
   const int q0 = q0_wg + wave * QTILE;  // this wave's first q row
-  const float slope = alibi ? alibi[qh] : 0.f;  // wave-uniform (one head per wg)
+  const float slope = ALIBI ? alibi[qh] : 0.f;  // wave-uniform (one head per wg)
 
   // LDS: K tile row-major [KVT][HD+KPAD]; V tile transposed+swizzled
   // [HD][KVT+KPAD]; per-wave P scratch [QTILE][KVT+KPAD]
@@ -144,6 +144,11 @@ __global__ __launch_bounds__(WAVES * 64) void attn_prefill_kernel(
     // ---- causal mask + online softmax (per-lane rows hi*4+r, col = key)
     float p[NB][4];  // [nb][r] probabilities for this lane's slots
     float corr[4];
+    // interior tiles (no tail, no causal edge, no s_q edge for ANY row of
+    // this wave) skip all masking VALU: +10% at S=4096 (PMC showed the
+    // kernel issue-bound on VALU at 17:1 VALU:MFMA)
+    const bool interior =
+        (tile_n == KVT) && (q0 + QTILE <= s_q) && (!causal || j0 + KVT <= kv_offset + q0 + 1);
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int qrow = q0 + hi * 4 + r;
@@ -155,11 +160,20 @@ __global__ __launch_bounds__(WAVES * 64) void attn_prefill_kernel(
       // scores for the r==0 slot of every accumulator row group, leaking
       // masked keys into the softmax (bisected in scripts/prefill_bisect.hip).
       float s[NB];
+      if (interior) {
 #pragma unroll
-      for (int nb = 0; nb < NB; ++nb) {
-        const int key = j0 + nb * 16 + col;
-        s[nb] = s_acc[nb][r] * scale + slope * key;
-        if (key >= tile_n + j0 || (causal && key > q_abs) || qrow >= s_q) s[nb] = NEG_SENTINEL;
+        for (int nb = 0; nb < NB; ++nb) {
+          s[nb] = s_acc[nb][r] * scale;
+          if (ALIBI) s[nb] += slope * (j0 + nb * 16 + col);
+        }
+      } else {
+#pragma unroll
+        for (int nb = 0; nb < NB; ++nb) {
+          const int key = j0 + nb * 16 + col;
+          s[nb] = s_acc[nb][r] * scale;
+          if (ALIBI) s[nb] += slope * key;
+          if (key >= tile_n + j0 || (causal && key > q_abs) || qrow >= s_q) s[nb] = NEG_SENTINEL;
+        }
       }
       float mx = NEG_SENTINEL;
 #pragma unroll
@@ -263,21 +277,24 @@ torch::Tensor attn_prefill_fused(
   const auto* vp = reinterpret_cast<const unsigned short*>(v.data_ptr());
   auto* op = reinterpret_cast<unsigned short*>(out.data_ptr());
 
+#define PF_LAUNCH(HDV, KVTV, AB)                                                            \
+  attn_prefill_kernel<HDV, KVTV, AB><<<grid, WAVES * WAVE, 0, stream>>>(                    \
+      qp, kp, vp, op, alibi_p, QH, KVH, S, LMAX, (int)kv_len, (int)kv_offset, sc, causal ? 1 : 0)
+  const bool ab = alibi_p != nullptr;
   if (HD == 128) {
-    if (kvt_env == 32)
-      attn_prefill_kernel<128, 32><<<grid, WAVES * WAVE, 0, stream>>>(
-          qp, kp, vp, op, alibi_p, QH, KVH, S, LMAX, (int)kv_len, (int)kv_offset, sc, causal ? 1 : 0);
-    else
-      attn_prefill_kernel<128, 64><<<grid, WAVES * WAVE, 0, stream>>>(
-          qp, kp, vp, op, alibi_p, QH, KVH, S, LMAX, (int)kv_len, (int)kv_offset, sc, causal ? 1 : 0);
+    if (kvt_env == 32) {
+      if (ab) PF_LAUNCH(128, 32, true); else PF_LAUNCH(128, 32, false);
+    } else {
+      if (ab) PF_LAUNCH(128, 64, true); else PF_LAUNCH(128, 64, false);
+    }
   } else {
-    if (kvt_env == 32)
-      attn_prefill_kernel<64, 32><<<grid, WAVES * WAVE, 0, stream>>>(
-          qp, kp, vp, op, alibi_p, QH, KVH, S, LMAX, (int)kv_len, (int)kv_offset, sc, causal ? 1 : 0);
-    else
-      attn_prefill_kernel<64, 64><<<grid, WAVES * WAVE, 0, stream>>>(
-          qp, kp, vp, op, alibi_p, QH, KVH, S, LMAX, (int)kv_len, (int)kv_offset, sc, causal ? 1 : 0);
+    if (kvt_env == 32) {
+      if (ab) PF_LAUNCH(64, 32, true); else PF_LAUNCH(64, 32, false);
+    } else {
+      if (ab) PF_LAUNCH(64, 64, true); else PF_LAUNCH(64, 64, false);
+    }
   }
+#undef PF_LAUNCH
   HIP_CHECK_LAST();
   return out;
 }

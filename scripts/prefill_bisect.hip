@@ -118,6 +118,8 @@ __global__ __launch_bounds__(WAVES * 64) void kern(
 
     float p[NB][4];
     float corr[4];
+    const bool interior =
+        (tile_n == KVT) && (q0 + QTILE <= s_q) && (!causal || j0 + KVT <= kv_offset + q0 + 1);
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int qrow = q0 + hi * 4 + r;
@@ -147,6 +149,38 @@ __global__ __launch_bounds__(WAVES * 64) void kern(
         m_row[r] = m_new;
         p[0][r] = p0;
         p[1][r] = p1;
+      } else if constexpr (VAR == 5) {
+        // split-mask form + mask-free interior tiles (all rows of this wave
+        // live for the whole tile: no tail, no causal edge, no s_q edge)
+        float s[NB];
+        if (interior) {
+#pragma unroll
+          for (int nb = 0; nb < NB; ++nb) s[nb] = s_acc[nb][r] * scale;
+        } else {
+#pragma unroll
+          for (int nb = 0; nb < NB; ++nb) {
+            s[nb] = s_acc[nb][r] * scale;
+            const int key = j0 + nb * 16 + col;
+            if (key >= tile_n + j0 || (causal && key > q_abs) || qrow >= s_q) s[nb] = NEG_SENTINEL;
+          }
+        }
+        float mx = NEG_SENTINEL;
+#pragma unroll
+        for (int nb = 0; nb < NB; ++nb) mx = fmaxf(mx, s[nb]);
+#pragma unroll
+        for (int off = 8; off > 0; off >>= 1) mx = fmaxf(mx, __shfl_xor(mx, off, WAVE));
+        const float m_new = fmaxf(m_row[r], mx);
+        corr[r] = (m_row[r] <= NEG_THRESHOLD) ? 0.f : __expf(m_row[r] - m_new);
+        float lsum = 0.f;
+#pragma unroll
+        for (int nb = 0; nb < NB; ++nb) {
+          p[nb][r] = (s[nb] <= NEG_THRESHOLD) ? 0.f : __expf(s[nb] - m_new);
+          lsum += p[nb][r];
+        }
+#pragma unroll
+        for (int off = 8; off > 0; off >>= 1) lsum += __shfl_xor(lsum, off, WAVE);
+        l_row[r] = l_row[r] * corr[r] + lsum;
+        m_row[r] = m_new;
       } else if constexpr (VAR == 2 || VAR == 4) {
         // split: mask loop, then mx loop
         float s[NB];
@@ -365,6 +399,7 @@ int main() {
   RUN(128, 64, 0, "KVT64 VAR0 (new exact)")
   RUN(128, 64, 2, "KVT64 VAR2 (split mask/mx)")
   RUN(128, 64, 4, "KVT64 VAR4 (split + P swz)")
+  RUN(128, 64, 5, "KVT64 VAR5 (interior fast)")
 
   // ---- timing at the llama-2-70b prefill shape ----
   {
@@ -402,6 +437,7 @@ int main() {
     TIME(32, 4, "time KVT32 split+Pswz")
     TIME(64, 2, "time KVT64 split")
     TIME(64, 4, "time KVT64 split+Pswz")
+    TIME(64, 5, "time KVT64 interior-fast")
   }
   return 0;
 }
