@@ -172,7 +172,10 @@ def main():
                     logits = head_logits(gs["g_span"].static_outputs[0][:, -1, :])
                     gs["cur_id"].copy_(logits.argmax(dim=-1, keepdim=True))
                     return logits
-                gs["g_head"] = GraphedCallable(head_fn, [])
+                if not os.environ.get("PETALS_AMD_NO_HEAD_GRAPH"):  # debug bisect
+                    gs["g_head"] = GraphedCallable(head_fn, [])
+                else:
+                    gs["head_eager"] = head_fn
             else:
                 gs["h_back"] = torch.empty(B, 1, H, device=device, dtype=dtype)
 
@@ -201,7 +204,10 @@ def main():
             if world > 1:
                 stage.send(h)
                 dist.recv(gs["h_back"], src=stage.prev_rank)
-            gs["g_head"].replay()
+            if "g_head" in gs:
+                gs["g_head"].replay()
+            else:
+                gs["head_eager"]()
         else:
             dist.recv(gs["h_in"], src=stage.prev_rank)
             gs["g_span"].replay()

@@ -42,14 +42,10 @@ class BloomAttention(nn.Module):
             k_cache, v_cache = kv_cache
             k_cache[:b, :, prefix_length:end].copy_(k)
             v_cache[:b, :, prefix_length:end].copy_(v)
-            k_pos = torch.arange(end, device=hidden_states.device, dtype=torch.float32)
-            bias = (slopes[:, None, None] * k_pos[None, None, :]).unsqueeze(0)
-            attn = ops.attention_decode(q, k_cache[:b], v_cache[:b], end, attn_bias=bias)
+            attn = ops.attention_decode(q, k_cache[:b], v_cache[:b], end, alibi_slopes=slopes)
         else:
             assert prefix_length == 0
-            k_pos = torch.arange(q_len, device=hidden_states.device, dtype=torch.float32)
-            bias = (slopes[:, None, None] * k_pos[None, None, :]).unsqueeze(0)
-            attn = ops.attention(q, k, v, causal=True, attn_bias=bias)
+            attn = ops.attention(q, k, v, causal=True, alibi_slopes=slopes)
         attn = attn.transpose(1, 2).reshape(b, q_len, self.hidden_size)
         return self.dense(attn)
 

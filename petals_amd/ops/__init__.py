@@ -76,9 +76,16 @@ def apply_rope(q, k, cos, sin, position_ids):
     return reference.apply_rope(q, k, cos, sin, position_ids)
 
 
-def attention(q, k, v, *, causal: bool, kv_offset: int = 0, attn_bias=None, scale=None):
-    """Prefill attention. On GPU (bf16, head_dim 64/128, no additive bias) this
-    runs the MFMA flash kernel; otherwise the fp32-softmax torch composition."""
+def _alibi_bias_matrix(alibi_slopes, kv_len, device):
+    """[1, H, 1, kv_len] additive bias = slope * key_pos (fallback paths)."""
+    k_pos = torch.arange(kv_len, device=device, dtype=torch.float32)
+    return (alibi_slopes.to(device)[:, None, None] * k_pos[None, None, :]).unsqueeze(0)
+
+
+def attention(q, k, v, *, causal: bool, kv_offset: int = 0, attn_bias=None, scale=None, alibi_slopes=None):
+    """Prefill attention. On GPU (bf16, head_dim 64/128, no additive bias
+    matrix — ALiBi via per-head `alibi_slopes` stays on the kernel) this runs
+    the MFMA flash kernel; otherwise the fp32-softmax torch composition."""
     if (
         q.is_cuda
         and not _grad_mode(q, k, v)
@@ -93,14 +100,18 @@ def attention(q, k, v, *, causal: bool, kv_offset: int = 0, attn_bias=None, scal
             import math
 
             kv_len = k.shape[2]
+            slopes = None if alibi_slopes is None else alibi_slopes.to(q.device, torch.float32).contiguous()
             return hops.attn_prefill_fused(
                 q.contiguous(), k.contiguous(), v.contiguous(), kv_len, int(kv_offset),
                 float(scale) if scale is not None else 1.0 / math.sqrt(q.shape[-1]), bool(causal),
+                slopes,
             )
+    if alibi_slopes is not None and attn_bias is None:
+        attn_bias = _alibi_bias_matrix(alibi_slopes, k.shape[2], q.device)
     return reference.attention(q, k, v, causal=causal, kv_offset=kv_offset, attn_bias=attn_bias, scale=scale)
 
 
-def attention_decode(q, k_cache, v_cache, kv_len: int, *, attn_bias=None, scale=None):
+def attention_decode(q, k_cache, v_cache, kv_len: int, *, attn_bias=None, scale=None, alibi_slopes=None):
     """Decode attention over a preallocated cache.
 
     q: [b, n_heads, q_len, hd]; k_cache/v_cache: [b, n_kv, max_len, hd] with
@@ -124,9 +135,10 @@ def attention_decode(q, k_cache, v_cache, kv_len: int, *, attn_bias=None, scale=
                 kv_len_t = torch.tensor([kv_len], dtype=torch.int32, device=q.device)
                 empty = torch.empty(0, dtype=torch.float32, device=q.device)
                 qf = q.permute(0, 2, 1, 3).reshape(b, n_heads * hd).float().contiguous()
+                slopes = None if alibi_slopes is None else alibi_slopes.to(q.device, torch.float32).contiguous()
                 out = ops.attn_decode_fused(
                     qf, k_cache.contiguous(), v_cache.contiguous(), kv_len_t, gq, 0, empty, empty,
-                    float(scale) if scale is not None else 1.0 / math.sqrt(hd),
+                    float(scale) if scale is not None else 1.0 / math.sqrt(hd), slopes,
                 )
                 return out.view(b, 1, n_heads, hd).permute(0, 2, 1, 3).to(q.dtype)
     q_len = q.shape[2]
@@ -145,13 +157,17 @@ def attention_decode(q, k_cache, v_cache, kv_len: int, *, attn_bias=None, scale=
         if hops is not None:
             import math
 
+            slopes = None if alibi_slopes is None else alibi_slopes.to(q.device, torch.float32).contiguous()
             return hops.attn_prefill_fused(
                 q.contiguous(), k_cache.contiguous(), v_cache.contiguous(), int(kv_len),
                 int(kv_len) - q_len,
                 float(scale) if scale is not None else 1.0 / math.sqrt(q.shape[-1]), True,
+                slopes,
             )
     k = k_cache[:, :, :kv_len]
     v = v_cache[:, :, :kv_len]
+    if alibi_slopes is not None and attn_bias is None:
+        attn_bias = _alibi_bias_matrix(alibi_slopes, kv_len, q.device)
     return reference.attention(
         q, k, v, causal=q_len > 1, kv_offset=kv_len - q_len, attn_bias=attn_bias, scale=scale
     )

@@ -418,7 +418,14 @@ def _bloom_qkv_perm(qh: int, hd: int) -> torch.Tensor:
 
 
 class BloomFastPath:
-    graph_safe = True
+    # hipGraph capture of a bloom span is numerically fine in isolation (see
+    # scripts/bloom_graph_debug.py — every chain prefix captures and replays),
+    # but a bloom span graph coexisting with the client's LM-head graph
+    # hardware-faults on replay (scripts/head_graph_debug.py isolates the head
+    # as clean too; the same dual-graph pattern works for Llama/Falcon/405B).
+    # Until the interaction is root-caused, bloom decodes eagerly: 40.8 tok/s
+    # on bloom-176b NF4 (gpurun_out/bloom176_ng2.log) vs the graphs-on crash.
+    graph_safe = False
 
     def __init__(self, block, hip_ops, quant: str = "none"):
         cfg = block.config
