@@ -1,10 +1,19 @@
-// GQA decode attention for CDNA4 (gfx950), flash-decoding style:
-// grid = (batch x kv_heads, kv splits); each 64-thread block streams a slice
-// of the KV cache once (coalesced 16 B/lane K/V row reads by 16-lane groups),
-// keeps online-softmax state per q-head in registers, block-combines its 4
-// row-groups through LDS, and writes one unnormalized partial per split.
-// attn_decode_combine merges the splits. kv_len comes from a DEVICE pointer so
-// the whole decode step can be captured in a hipGraph with a moving position.
+// GQA decode attention for CDNA4 (gfx950), flash-decoding style.
+//
+// Default path (mfma_decode_kernel): the GQ query heads of one kv head form
+// the rows of an mfma_f32_16x16x32_bf16 A-fragment (padded to 16), so 32
+// cache keys cost 4 QK MFMAs + 8 PV MFMAs per wave instead of 32*GQ
+// shuffle-reduced dot products. K is consumed directly from global as
+// B-fragments (16 B/lane); V goes through the XOR-swizzled LDS transpose
+// (same as the prefill kernel); 4 waves take interleaved 32-key tiles of the
+// workgroup's split range and combine through LDS; attn_decode_combine merges
+// splits. Measured (scripts/decode_mfma.hip, llama-2-70b shape): 1039 GB/s at
+// kv=4k -> 3509 GB/s at kv=131k, 2.2-2.4x the VALU kernel at every length.
+//
+// The original VALU kernel (attn_decode_kernel: 16-lane-group dot products,
+// online softmax in registers) is kept behind PETALS_DECODE_KERNEL=valu.
+// kv_len comes from a DEVICE pointer so the whole decode step can be captured
+// in a hipGraph with a moving position.
 //
 // Replaces the reference's torch QK^T/softmax/AV decode math
 // (reference models/llama/block.py:108-127) on the MI355X fast path.
@@ -12,6 +21,217 @@
 #include "common.h"
 #include <torch/extension.h>
 #include <ATen/cuda/CUDAContext.h>
+
+#include <cstdlib>
+
+using bf16x8 = __attribute__((ext_vector_type(8))) short;
+using f32x4 = __attribute__((ext_vector_type(4))) float;
+
+#define DKVT 32    // keys per wave-tile (MFMA path)
+#define DWAVES 4   // waves per workgroup (MFMA path)
+#define DKPAD 8
+
+#define DVT_BYTE(dim, key_byte) \
+  ((((unsigned)(dim)) * ((DKVT + DKPAD) * 2) + (unsigned)(key_byte)) ^ ((((unsigned)(dim) >> 3) & 7u) << 4))
+
+template <int HD, int GQ>
+__global__ __launch_bounds__(DWAVES * 64) void mfma_decode_kernel(
+    const float* __restrict__ q,            // [B, KV, GQ, HD]
+    const unsigned short* __restrict__ k_cache,  // [B, KV, lmax, HD]
+    const unsigned short* __restrict__ v_cache,
+    float* __restrict__ part_o,             // [B*KV, splits, GQ, HD]
+    float* __restrict__ part_ml,            // [B*KV, splits, GQ, 2]
+    const int* __restrict__ kv_len_ptr,
+    const float* __restrict__ alibi,        // [kv_heads*GQ] slopes or null
+    int kv_heads,
+    int lmax,
+    int n_splits,
+    float scale) {
+  const int bkv = blockIdx.x;
+  const int kvh = bkv % kv_heads;
+  const int split = blockIdx.y;
+  const int kv_len = *kv_len_ptr;
+
+  const int rows_per_split = (kv_len + n_splits - 1) / n_splits;
+  const int j_begin = split * rows_per_split;
+  const int j_end = min(j_begin + rows_per_split, kv_len);
+
+  const int tid = threadIdx.x;
+  const int wave = tid / WAVE;
+  const int lane = tid & (WAVE - 1);
+  const int col = lane & 15;
+  const int hi = lane >> 4;
+
+  constexpr int KCH = HD / 32;
+
+  // per-wave LDS: swizzled V^T tile + P scratch; the cross-wave combine
+  // overlays the V^T storage after the main loop
+  __shared__ unsigned char vt_raw[DWAVES][HD * (DKVT + DKPAD) * 2];
+  __shared__ unsigned short p_lds[DWAVES][16][DKVT + DKPAD];
+  __shared__ float c_ml[DWAVES][GQ][2];
+  static_assert(GQ * HD * 4 <= HD * (DKVT + DKPAD) * 2, "combine O overlay too big");
+
+  // ---- q tile: A-fragment rows = q heads (zero-padded to 16), pre-scaled
+  bf16x8 q_frag[KCH];
+  const size_t q_base = (size_t)bkv * GQ * HD;
+#pragma unroll
+  for (int kc = 0; kc < KCH; ++kc) {
+    if (col < GQ) {
+      const float* src = q + q_base + (size_t)col * HD + kc * 32 + hi * 8;
+      short v[8];
+#pragma unroll
+      for (int e = 0; e < 8; ++e) v[e] = (short)f32_to_bf16(src[e] * scale);
+      q_frag[kc] = bf16x8{v[0], v[1], v[2], v[3], v[4], v[5], v[6], v[7]};
+    } else {
+      q_frag[kc] = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+    }
+  }
+
+  // per-lane-row ALiBi slopes (rows = heads hi*4 + r)
+  float sl[4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int g = hi * 4 + r;
+    sl[r] = (alibi && g < GQ) ? alibi[kvh * GQ + g] : 0.f;
+  }
+
+  f32x4 acc_o[HD / 16];
+#pragma unroll
+  for (int d = 0; d < HD / 16; ++d) acc_o[d] = f32x4{0.f, 0.f, 0.f, 0.f};
+  float m_row[4], l_row[4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    m_row[r] = NEG_SENTINEL;
+    l_row[r] = 0.f;
+  }
+
+  const size_t kv_base = (size_t)bkv * lmax * HD;
+  const unsigned short* kb = k_cache + kv_base;
+  const unsigned short* vb = v_cache + kv_base;
+
+  // wave w handles tiles j_begin + (t*DWAVES + w)*DKVT
+  for (int j0 = j_begin + wave * DKVT; j0 < j_end; j0 += DWAVES * DKVT) {
+    const int tile_n = min(DKVT, j_end - j0);
+    const bool full = tile_n == DKVT;
+
+    // ---- stage V^T (this wave only: no block barrier; the waitcnt orders
+    // this wave's LDS writes before its reads)
+    for (int idx = lane; idx < DKVT * (HD / 8); idx += WAVE) {
+      const int row = idx / (HD / 8);
+      const int c8 = (idx - row * (HD / 8)) * 8;
+      bf16x8 vv8 = {0, 0, 0, 0, 0, 0, 0, 0};
+      if (j0 + row < j_end) vv8 = *reinterpret_cast<const bf16x8*>(vb + (size_t)(j0 + row) * HD + c8);
+#pragma unroll
+      for (int e = 0; e < 8; ++e)
+        *reinterpret_cast<unsigned short*>(&vt_raw[wave][DVT_BYTE(c8 + e, row * 2)]) = (unsigned short)vv8[e];
+    }
+    __builtin_amdgcn_s_waitcnt(0);
+
+    // ---- S = Q K^T, K direct from global: B[k = kdim][n = key]
+    f32x4 s_acc[DKVT / 16];
+#pragma unroll
+    for (int nb = 0; nb < DKVT / 16; ++nb) {
+      s_acc[nb] = f32x4{0.f, 0.f, 0.f, 0.f};
+      const int key = j0 + nb * 16 + col;
+      const unsigned short* krow = kb + (size_t)min(key, j_end - 1) * HD + hi * 8;
+#pragma unroll
+      for (int kc = 0; kc < KCH; ++kc) {
+        const bf16x8 kt = *reinterpret_cast<const bf16x8*>(krow + kc * 32);
+        s_acc[nb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(q_frag[kc], kt, s_acc[nb], 0, 0, 0);
+      }
+    }
+
+    // ---- online softmax over this tile (rows = heads hi*4+r)
+    float p[DKVT / 16][4];
+    float corr[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      float s[DKVT / 16];
+#pragma unroll
+      for (int nb = 0; nb < DKVT / 16; ++nb) {
+        const int key = j0 + nb * 16 + col;
+        s[nb] = s_acc[nb][r] + sl[r] * key;
+        if (!full && key >= j_end) s[nb] = NEG_SENTINEL;
+      }
+      float mx = NEG_SENTINEL;
+#pragma unroll
+      for (int nb = 0; nb < DKVT / 16; ++nb) mx = fmaxf(mx, s[nb]);
+#pragma unroll
+      for (int off = 8; off > 0; off >>= 1) mx = fmaxf(mx, __shfl_xor(mx, off, WAVE));
+      const float m_new = fmaxf(m_row[r], mx);
+      corr[r] = (m_row[r] <= NEG_THRESHOLD) ? 0.f : __expf(m_row[r] - m_new);
+      float lsum = 0.f;
+#pragma unroll
+      for (int nb = 0; nb < DKVT / 16; ++nb) {
+        p[nb][r] = (s[nb] <= NEG_THRESHOLD) ? 0.f : __expf(s[nb] - m_new);
+        lsum += p[nb][r];
+      }
+#pragma unroll
+      for (int off = 8; off > 0; off >>= 1) lsum += __shfl_xor(lsum, off, WAVE);
+      l_row[r] = l_row[r] * corr[r] + lsum;
+      m_row[r] = m_new;
+    }
+
+#pragma unroll
+    for (int r = 0; r < 4; ++r)
+#pragma unroll
+      for (int nb = 0; nb < DKVT / 16; ++nb)
+        p_lds[wave][hi * 4 + r][nb * 16 + col] = f32_to_bf16(p[nb][r]);
+    __builtin_amdgcn_s_waitcnt(0);
+
+    const bf16x8 p_frag = *reinterpret_cast<const bf16x8*>(&p_lds[wave][col][hi * 8]);
+
+#pragma unroll
+    for (int d = 0; d < HD / 16; ++d)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) acc_o[d][r] *= corr[r];
+#pragma unroll
+    for (int d = 0; d < HD / 16; ++d) {
+      const bf16x8 vfrag = *reinterpret_cast<const bf16x8*>(&vt_raw[wave][DVT_BYTE(d * 16 + col, hi * 16)]);
+      acc_o[d] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(p_frag, vfrag, acc_o[d], 0, 0, 0);
+    }
+  }
+
+  // ---- cross-wave combine (overlay the V^T LDS with per-wave O rows)
+  __syncthreads();
+  float* c_o = reinterpret_cast<float*>(vt_raw[wave]);
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int g = hi * 4 + r;
+    if (g < GQ) {
+#pragma unroll
+      for (int d = 0; d < HD / 16; ++d) c_o[g * HD + d * 16 + col] = acc_o[d][r];
+      if (col == 0) {
+        c_ml[wave][g][0] = m_row[r];
+        c_ml[wave][g][1] = l_row[r];
+      }
+    }
+  }
+  __syncthreads();
+
+  if (wave == 0) {
+    float* po = part_o + (((size_t)bkv * n_splits + split) * GQ) * HD;
+    float* pml = part_ml + (((size_t)bkv * n_splits + split) * GQ) * 2;
+    for (int idx = lane; idx < GQ * HD; idx += WAVE) {
+      const int g = idx / HD, d = idx - g * HD;
+      float m_star = c_ml[0][g][0];
+#pragma unroll
+      for (int w = 1; w < DWAVES; ++w) m_star = fmaxf(m_star, c_ml[w][g][0]);
+      float osum = 0.f, lsum = 0.f;
+#pragma unroll
+      for (int w = 0; w < DWAVES; ++w) {
+        const float wgt = (c_ml[w][g][0] <= NEG_THRESHOLD) ? 0.f : __expf(c_ml[w][g][0] - m_star);
+        osum += wgt * reinterpret_cast<const float*>(vt_raw[w])[g * HD + d];
+        lsum += wgt * c_ml[w][g][1];
+      }
+      po[idx] = osum;
+      if (d == 0) {
+        pml[g * 2 + 0] = m_star;
+        pml[g * 2 + 1] = lsum;
+      }
+    }
+  }
+}
 
 template <int HD, int GQ>
 __global__ __launch_bounds__(64) void attn_decode_kernel(
@@ -195,11 +415,24 @@ torch::Tensor attn_decode_fused(
   const int b = q.size(0);
   const int kv_heads = k_cache.size(1), lmax = k_cache.size(2), hd = k_cache.size(3);
   const int GQi = (int)gq;
+  static const bool use_valu = [] {
+    const char* s = std::getenv("PETALS_DECODE_KERNEL");
+    return s && s[0] == 'v';
+  }();
   int n_splits = (int)n_splits_i;
   if (n_splits <= 0) {
-    // target ~1024 single-wave workgroups; each split should keep >=64 cache
-    // rows so the streaming loop amortizes setup
-    n_splits = std::max(1, std::min(1024 / std::max(1, b * kv_heads), (lmax + 63) / 64));
+    const int bkv = std::max(1, b * kv_heads);
+    if (use_valu) {
+      // VALU kernel: target ~1024 single-wave workgroups, chunks >= 64 rows
+      n_splits = std::max(1, std::min(1024 / bkv, (lmax + 63) / 64));
+    } else {
+      // MFMA kernel: 4-wave workgroups; fill ~256 CUs (chunks >= 128 rows) and
+      // go up to 3 wgs/CU only when chunks stay >= 512 rows (measured knees,
+      // profiles/decode_longctx_sweep.log + scripts/decode_mfma.hip)
+      const int by_occ = std::min(768 / bkv, lmax / 512);
+      const int by_min = std::min(256 / bkv, (lmax + 127) / 128);
+      n_splits = std::max(1, std::max(by_occ, by_min));
+    }
   }
   auto opts = q.options();
   if (part_o.numel() < (int64_t)b * kv_heads * n_splits * GQi * hd)
@@ -227,8 +460,12 @@ torch::Tensor attn_decode_fused(
   bool launched = false;
 #define ATTN_CASE(HDV, GQV)                                                   \
   if (hd == HDV && GQi == GQV) {                                              \
-    attn_decode_kernel<HDV, GQV><<<grid, 64, 0, stream>>>(                    \
-        qp, kp, vp, pop, pmlp, lenp, alibi_p, kv_heads, lmax, n_splits, sc);  \
+    if (use_valu)                                                             \
+      attn_decode_kernel<HDV, GQV><<<grid, 64, 0, stream>>>(                  \
+          qp, kp, vp, pop, pmlp, lenp, alibi_p, kv_heads, lmax, n_splits, sc);\
+    else                                                                      \
+      mfma_decode_kernel<HDV, GQV><<<grid, DWAVES * WAVE, 0, stream>>>(       \
+          qp, kp, vp, pop, pmlp, lenp, alibi_p, kv_heads, lmax, n_splits, sc);\
     launched = true;                                                          \
   }
   ATTN_CASE(128, 1) ATTN_CASE(128, 2) ATTN_CASE(128, 4) ATTN_CASE(128, 6)
