@@ -37,7 +37,7 @@ from petals_amd.utils.misc import get_size_in_bytes
 
 logger = logging.getLogger(__name__)
 
-DTYPE_MAP = {"bfloat16": torch.bfloat16, "float16": torch.float16, "float32": torch.float32, "auto": None}
+from petals_amd.constants import DTYPE_MAP  # noqa: E402  (parity: reference constants.py:18)
 
 
 class Server:

@@ -33,3 +33,13 @@ def docstring_from(source):
         return fn
 
     return wrapper
+
+
+def sample_up_to(population, k: int):
+    """Up to k random items (parity: reference utils/random.py:7)."""
+    import random
+
+    population = list(population)
+    if len(population) > k:
+        population = random.sample(population, k)
+    return population
