@@ -343,6 +343,14 @@ class TransformerConnectionHandler:
                     blocks, cache_pairs, hidden_states.shape[0], hidden_states.shape[-1],
                     device, dtype, active_adapter, initial_position=prefix_length,
                 )
+            if os.environ.get("PETALS_AMD_STEP_TRACE"):
+                _g0 = time.perf_counter()
+                out = session.span_graph.step(hidden_states, prefix_length)
+                torch.cuda.synchronize()
+                _g1 = time.perf_counter()
+                res = out.cpu()
+                print(f"[rt] graph {( _g1-_g0)*1e3:.2f} d2h {(time.perf_counter()-_g1)*1e3:.2f} ms", flush=True)
+                return res
             out = session.span_graph.step(hidden_states, prefix_length)
             return out.cpu()
 

@@ -27,6 +27,10 @@ import torch
 
 logger = logging.getLogger(__name__)
 
+import os as _os
+
+_TRACE = bool(_os.environ.get("PETALS_AMD_STEP_TRACE"))
+
 
 class TaskPrioritizer:
     """Policy hook (parity: server/task_prioritizer.py:15). Lower = sooner."""
@@ -80,6 +84,10 @@ class PriorityRuntime:
                     continue
                 priority, _seq, fn, args, kwargs, loop, future = heapq.heappop(self._queue)
             t0 = time.perf_counter()
+            if _TRACE and isinstance(kwargs.get("_enq"), float):
+                print(f"[rt] queue-wait {(t0 - kwargs.pop('_enq'))*1e3:.2f} ms", flush=True)
+            else:
+                kwargs.pop("_enq", None)
             try:
                 stream = self._get_stream(priority)
                 if stream is not None:
@@ -93,10 +101,14 @@ class PriorityRuntime:
                 loop.call_soon_threadsafe(_set_exception_safe, future, e)
             finally:
                 self.stats["tasks"] += 1
+                if _TRACE:
+                    print(f"[rt] task {(time.perf_counter()-t0)*1e3:.2f} ms", flush=True)
                 self.stats["busy_time"] += time.perf_counter() - t0
 
     async def submit(self, priority: float, fn: Callable, *args, **kwargs) -> Any:
         """Schedule fn on the runtime thread; await its result."""
+        if _TRACE:
+            kwargs["_enq"] = time.perf_counter()
         loop = asyncio.get_event_loop()
         future: asyncio.Future = loop.create_future()
         with self._cv:

@@ -27,6 +27,13 @@ def main():
                         "context-switch on every hop; use for isolation tests only)")
     args = p.parse_args()
 
+    if os.environ.get("PETALS_AMD_GC_TUNE"):
+        import gc
+
+        gc.collect()
+        gc.freeze()
+        gc.set_threshold(100000, 50, 50)
+
     from petals_amd.dht.node import DHT
     from petals_amd.utils.auto_config import AutoDistributedModelForCausalLM
 
