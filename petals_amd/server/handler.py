@@ -327,6 +327,7 @@ class TransformerConnectionHandler:
             and hidden_states.shape[1] == 1
             and hidden_states.shape[0] <= max_graph_batch
             and not has_prompts
+            and not os.environ.get("PETALS_AMD_NO_GRAPHS")
         ):
             if has_hypo:
                 hypo = hypo_ids.to(device)
