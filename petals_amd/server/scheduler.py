@@ -55,6 +55,16 @@ class PriorityRuntime:
         self._thread: Optional[threading.Thread] = None
         self._streams = {}
         self.stats = {"tasks": 0, "busy_time": 0.0}
+        # keep-warm (opt-in, PETALS_AMD_KEEP_WARM=<seconds>): for a short
+        # window after each task, keep the device queue non-empty while idle.
+        # Per-token serving leaves ~10 ms host gaps in which this pool's host
+        # power management parks sclk, making ~every 3rd step run ~5x slower
+        # (PARITY.md "serving cadence"). 1-element ticks every 1 ms were NOT
+        # enough to hold clocks on the SR-IOV pool host; the knob stays for
+        # bare-metal deployments where guest clocks are controllable.
+        self._keep_warm_s = float(_os.environ.get("PETALS_AMD_KEEP_WARM", "0"))
+        self._warm_until = 0.0
+        self._warm_buf: Optional[torch.Tensor] = None
 
     def start(self):
         self._thread = threading.Thread(target=self._run, name="PriorityRuntime", daemon=True)
@@ -77,7 +87,16 @@ class PriorityRuntime:
         while True:
             with self._cv:
                 while not self._queue and not self._shutdown:
-                    self._cv.wait(timeout=1.0)
+                    if (
+                        self.device is not None
+                        and self.device.type == "cuda"
+                        and self._keep_warm_s > 0
+                        and time.monotonic() < self._warm_until
+                    ):
+                        self._warm_tick()
+                        self._cv.wait(timeout=0.001)
+                    else:
+                        self._cv.wait(timeout=1.0)
                 if self._shutdown and not self._queue:
                     return
                 if not self._queue:
@@ -100,10 +119,20 @@ class PriorityRuntime:
             except BaseException as e:  # noqa: BLE001
                 loop.call_soon_threadsafe(_set_exception_safe, future, e)
             finally:
+                self._warm_until = time.monotonic() + self._keep_warm_s
                 self.stats["tasks"] += 1
                 if _TRACE:
                     print(f"[rt] task {(time.perf_counter()-t0)*1e3:.2f} ms", flush=True)
                 self.stats["busy_time"] += time.perf_counter() - t0
+
+    def _warm_tick(self):
+        """Enqueue a trivial kernel (no sync) so the device queue never drains
+        during the keep-warm window."""
+        if self._warm_buf is None:
+            self._warm_buf = torch.zeros(1, device=self.device)
+        stream = self._get_stream(1.0)
+        with torch.cuda.stream(stream):
+            self._warm_buf.add_(0.0)
 
     async def submit(self, priority: float, fn: Callable, *args, **kwargs) -> Any:
         """Schedule fn on the runtime thread; await its result."""
