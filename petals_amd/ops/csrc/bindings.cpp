@@ -7,6 +7,7 @@ torch::Tensor rms_norm_f32out(torch::Tensor x, torch::Tensor w, double eps);
 torch::Tensor layer_norm(torch::Tensor x, torch::Tensor w, torch::Tensor b, double eps);
 torch::Tensor layer_norm_f32out(torch::Tensor x, torch::Tensor w, torch::Tensor b, double eps);
 torch::Tensor swiglu(torch::Tensor gate, torch::Tensor up);
+void warm_spin(int64_t microseconds);
 std::vector<torch::Tensor> apply_rope(
     torch::Tensor q, torch::Tensor k, torch::Tensor cos_t, torch::Tensor sin_t, torch::Tensor pos);
 void rope_cache_write(
@@ -40,6 +41,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layer_norm", &layer_norm, "LayerNorm with weight+bias (bf16 -> bf16)");
   m.def("layer_norm_f32out", &layer_norm_f32out, "LayerNorm with weight+bias (bf16 -> f32)");
   m.def("swiglu", &swiglu, "silu(gate) * up (bf16)");
+  m.def("warm_spin", &warm_spin, "occupy the device queue for ~N us (keep-warm)");
   m.def("apply_rope", &apply_rope, "rotate q,k by positions (bf16)");
   m.def("rope_cache_write", &rope_cache_write, "fused decode rope + kv cache write");
   m.def("kv_cache_write", &kv_cache_write, "decode kv cache write without rope (ALiBi families)");

@@ -225,6 +225,26 @@ torch::Tensor layer_norm_f32out(torch::Tensor x, torch::Tensor w, torch::Tensor 
   return layer_norm_impl(x, w, b, eps, true);
 }
 
+// --------------------------------------------------------------- warm_spin
+
+// one tiny workgroup that occupies the device queue for ~us microseconds
+// (wall_clock64 ticks at a fixed ~100 MHz, clock-independent). Used by the
+// scheduler's opt-in keep-warm path: per-token serving leaves ~10 ms host
+// gaps in which aggressive power management parks the clocks, making the
+// next step run several times slower; a queue that never drains avoids the
+// idle state at negligible power (s_sleep in the loop).
+__global__ void warm_spin_kernel(long long ticks) {
+  const long long start = wall_clock64();
+  while (wall_clock64() - start < ticks) __builtin_amdgcn_s_sleep(32);
+}
+
+void warm_spin(int64_t microseconds) {
+  const long long us = std::min<int64_t>(std::max<int64_t>(microseconds, 1), 2000);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  warm_spin_kernel<<<1, 64, 0, stream>>>(us * 100);  // 100 MHz wall clock
+  HIP_CHECK_LAST();
+}
+
 // ------------------------------------------------------------------ swiglu
 
 __global__ void swiglu_kernel(

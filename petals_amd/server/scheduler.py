@@ -126,13 +126,20 @@ class PriorityRuntime:
                 self.stats["busy_time"] += time.perf_counter() - t0
 
     def _warm_tick(self):
-        """Enqueue a trivial kernel (no sync) so the device queue never drains
-        during the keep-warm window."""
+        """Keep the device queue non-empty during the keep-warm window: a
+        ~1 ms bounded spin kernel (ops warm_spin, 1 workgroup, s_sleep loop)
+        if the HIP extension is loaded, else a trivial add."""
         if self._warm_buf is None:
             self._warm_buf = torch.zeros(1, device=self.device)
+            from petals_amd import ops as _ops
+
+            self._warm_spin = getattr(_ops._load_hip_ops(), "warm_spin", None)
         stream = self._get_stream(1.0)
         with torch.cuda.stream(stream):
-            self._warm_buf.add_(0.0)
+            if self._warm_spin is not None:
+                self._warm_spin(1000)
+            else:
+                self._warm_buf.add_(0.0)
 
     async def submit(self, priority: float, fn: Callable, *args, **kwargs) -> Any:
         """Schedule fn on the runtime thread; await its result."""
