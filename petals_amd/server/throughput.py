@@ -98,6 +98,8 @@ def measure_compute_rps(
     forward_tokens: int = 1024,
 ) -> Dict[str, float]:
     device = torch.device(device)
+    from petals_amd.server.from_pretrained import build_empty_block, init_random_block_
+
     block = build_empty_block(config, 0, device, dtype)
     init_random_block_(block, config, 0)
     block = block.eval()
