@@ -41,7 +41,7 @@ def main():
     p.add_argument("--prompt-len", type=int, default=128)
     p.add_argument("--batch", type=int, default=1)
     p.add_argument("--device", default="cuda")
-    p.add_argument("--quant", default="nf4", choices=["none", "nf4"], help="BASELINE config #3 names NF4 for the 70B pipeline; --quant none measures pure bf16")
+    p.add_argument("--quant", default="nf4", choices=["none", "nf4", "int8"], help="BASELINE config #3 names NF4 for the 70B pipeline; --quant none measures pure bf16")
     args = p.parse_args()
 
     import torch.distributed as dist
@@ -285,7 +285,7 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": tokens_per_s / baseline if args.model == "llama-2-70b" else None,
-            "dtype": ("bf16" if args.quant == "none" else "nf4-weights/bf16-compute") if use_cuda else "fp32",
+            "dtype": ("bf16" if args.quant == "none" else f"{args.quant}-weights/bf16-compute") if use_cuda else "fp32",
             "data": "synthetic prompt, random-init weights (no network)",
             "config": {
                 "model": args.model,

@@ -15,7 +15,7 @@ SRC_DIR = os.path.join(PKG_DIR, "csrc")
 BUILD_DIR = os.path.join(PKG_DIR, "_build")
 SO_PATH = os.path.join(PKG_DIR, "_hip_ops.so")
 
-SOURCES = ["bindings.cpp", "elementwise.hip", "gemv.hip", "attention.hip", "nf4.hip", "prefill_attn.hip", "moe.hip"]
+SOURCES = ["bindings.cpp", "elementwise.hip", "gemv.hip", "attention.hip", "nf4.hip", "int8.hip", "prefill_attn.hip", "moe.hip"]
 
 
 def build(verbose: bool = False) -> str:

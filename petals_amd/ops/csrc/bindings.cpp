@@ -29,6 +29,10 @@ torch::Tensor attn_prefill_fused(
     int64_t kv_len, int64_t kv_offset, double scale, bool causal,
     c10::optional<torch::Tensor> alibi_slopes);
 std::vector<torch::Tensor> nf4_quantize(torch::Tensor w);
+torch::Tensor gemv_int8(
+    torch::Tensor q, torch::Tensor scale, torch::Tensor x, torch::Tensor workspace,
+    c10::optional<torch::Tensor> residual, int64_t epilogue, int64_t splits_override,
+    c10::optional<torch::Tensor> bias);
 torch::Tensor nf4_dequantize(torch::Tensor packed, torch::Tensor absmax);
 torch::Tensor gemv_nf4(
     torch::Tensor packed, torch::Tensor absmax, torch::Tensor x, torch::Tensor workspace,
@@ -55,6 +59,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_prefill_fused", &attn_prefill_fused, "MFMA flash prefill attention (bf16, causal, GQA)",
         py::arg("q"), py::arg("k"), py::arg("v"), py::arg("kv_len"), py::arg("kv_offset"),
         py::arg("scale"), py::arg("causal"), py::arg("alibi_slopes") = py::none());
+  m.def("gemv_int8", &gemv_int8, "split-K weight-only int8 gemv (per-column scales) with epilogue",
+        py::arg("q"), py::arg("scale"), py::arg("x"), py::arg("workspace"),
+        py::arg("residual"), py::arg("epilogue"), py::arg("splits") = 0, py::arg("bias") = py::none());
   m.def("nf4_quantize", &nf4_quantize, "blockwise NF4 quantize [in,out] bf16 -> (packed u8, absmax bf16)");
   m.def("nf4_dequantize", &nf4_dequantize, "NF4 -> bf16");
   m.def("gemv_nf4", &gemv_nf4, "split-K NF4 gemv (dequant fused) with epilogue",

@@ -71,7 +71,9 @@ _split_autotune_cache: Dict[Tuple[str, int, int], int] = {}
 
 
 class _FastWeight:
-    """A transposed [in, out] weight in bf16 or NF4 (quantize-on-load).
+    """A transposed [in, out] weight in bf16, NF4 or weight-only int8
+    (quantize-on-load; int8 = per-out-column symmetric absmax, the MI355X
+    stand-in for the reference's optional LLM.int8 path).
 
     Split-K counts are autotuned once per (kind, in, out) shape at load time
     (profiles/gemv_split_sweep.log shows the optimum varies ~2x by shape);
@@ -81,12 +83,16 @@ class _FastWeight:
         self.hip = hip
         self.quant = quant
         self.in_dim, self.out_dim = t_bf16.shape
+        self.t = self.packed = self.absmax = self.q8 = self.scale8 = None
         if quant == "nf4":
             self.packed, self.absmax = hip.nf4_quantize(t_bf16.contiguous())
-            self.t = None
+        elif quant == "int8":
+            w = t_bf16.float()
+            scale = w.abs().amax(dim=0).clamp_min(1e-8) / 127.0
+            self.q8 = torch.round(w / scale).clamp(-127, 127).to(torch.int8).contiguous()
+            self.scale8 = scale.to(torch.bfloat16).contiguous()
         else:
             self.t = t_bf16.contiguous()
-            self.packed = self.absmax = None
         self.splits = self._autotune(t_bf16.device)
 
     def _autotune(self, device) -> int:
@@ -100,6 +106,9 @@ class _FastWeight:
         if self.quant == "nf4":
             candidates = [0, 32, 64, 128, 192, 256]
             max_chunk = 32
+        elif self.quant == "int8":
+            candidates = [0, 16, 32, 64, 128, 192]
+            max_chunk = 128
         else:
             candidates = [0, 8, 16, 32, 64, 128]
             max_chunk = 64
@@ -130,15 +139,19 @@ class _FastWeight:
     def _gemv_raw(self, x, ws, residual, epilogue, splits, bias=None):
         if self.quant == "nf4":
             return self.hip.gemv_nf4(self.packed, self.absmax, x, ws, residual, epilogue, splits, bias)
+        if self.quant == "int8":
+            return self.hip.gemv_int8(self.q8, self.scale8, x, ws, residual, epilogue, splits, bias)
         return self.hip.gemv_bf16(self.t, x, ws, residual, epilogue, splits, bias)
 
     def gemv(self, x, ws, residual, epilogue, bias=None):
         return self._gemv_raw(x, ws, residual, epilogue, self.splits, bias)
 
     def dense(self) -> torch.Tensor:
-        """bf16 [in, out] view for prefill GEMMs (dequantized on the fly for NF4)."""
+        """bf16 [in, out] view for prefill GEMMs (dequantized on the fly when quantized)."""
         if self.quant == "nf4":
             return self.hip.nf4_dequantize(self.packed, self.absmax)
+        if self.quant == "int8":
+            return (self.q8.to(torch.float32) * self.scale8.float()).to(torch.bfloat16)
         return self.t
 
 

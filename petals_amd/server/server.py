@@ -129,6 +129,8 @@ class Server:
         dtype_bytes = get_size_in_bytes(self.torch_dtype)
         if self.quant_type == "nf4":
             dtype_bytes = 0.53125  # 4.25 bits/param (parity: block_utils.py:46)
+        elif self.quant_type == "int8":
+            dtype_bytes = 1.002  # 8 bits/param + per-column bf16 scales
         h, inter = cfg.hidden_size, cfg.intermediate_size
         kv = cfg.n_kv_heads * cfg.head_dim
         attn = h * h + 2 * h * kv + h * h

@@ -457,3 +457,59 @@ def test_bloom_block_fast_decode_matches_cpu(hip):
 def test_falcon_block_fast_decode_matches_cpu(hip):
     """Falcon new-decoder fused path (parallel attn+MLP, rope GQA) vs CPU."""
     _block_fused_vs_cpu("test-falcon-hd64", 256)
+
+
+@requires_gpu
+def test_gemv_int8(hip):
+    """Weight-only int8 gemv (per-column scales) vs the same math in torch."""
+    torch.manual_seed(11)
+    in_dim, out_dim, B = 1024, 1536, 2
+    w = (torch.randn(in_dim, out_dim, device="cuda") * 0.04).float()
+    scale = w.abs().amax(dim=0).clamp_min(1e-8) / 127.0
+    q = torch.round(w / scale).clamp(-127, 127).to(torch.int8).contiguous()
+    sc_bf16 = scale.to(torch.bfloat16).contiguous()
+    x = torch.randn(B, in_dim, device="cuda")
+    ws = torch.empty(0, device="cuda")
+    ref = x.cpu() @ (q.float().cpu() * sc_bf16.float().cpu())
+    out = hip.gemv_int8(q, sc_bf16, x, ws, None, 0)
+    assert torch.allclose(out.cpu(), ref, atol=2e-2, rtol=2e-2), (out.cpu() - ref).abs().max()
+    # residual epilogue
+    res = torch.randn(B, out_dim, device="cuda").to(torch.bfloat16)
+    out_r = hip.gemv_int8(q, sc_bf16, x, ws, res, 2)
+    assert torch.allclose(out_r.float().cpu(), ref + res.float().cpu(), atol=4e-2, rtol=4e-2)
+
+
+@requires_gpu
+def test_llama_block_int8_decode_matches_cpu(hip):
+    """Full fused decode on weight-only int8 vs the fp32 CPU block."""
+    from petals_amd.models import get_model_block
+    from petals_amd.models.config_base import load_model_config
+    from petals_amd.server.from_pretrained import init_random_block_
+
+    cfg = load_model_config("test-llama-hd128")
+    blk_cpu = get_model_block(cfg, 0)
+    init_random_block_(blk_cpu, cfg, 0)
+    blk_cpu = blk_cpu.float().eval()
+
+    blk_gpu = get_model_block(cfg, 0)
+    blk_gpu.load_state_dict(blk_cpu.state_dict())
+    blk_gpu = blk_gpu.to("cuda", torch.bfloat16).eval().optimize_for_inference(quant="int8")
+    assert blk_gpu._fast is not None and blk_gpu._fast.quant == "int8"
+
+    torch.manual_seed(5)
+    B, S, H = 2, 9, cfg.hidden_size
+    x = torch.randn(B, S, H) * 0.5
+    ks, vs = blk_cpu.kv_cache_shape(B, 32)
+    kc, vc = torch.zeros(ks), torch.zeros(vs)
+    kg = torch.zeros(ks, device="cuda", dtype=torch.bfloat16)
+    vg = torch.zeros(vs, device="cuda", dtype=torch.bfloat16)
+
+    y_cpu = [blk_cpu(x[:, :6], kv_cache=(kc, vc), prefix_length=0)]
+    y_gpu = [blk_gpu(x[:, :6].cuda().bfloat16(), kv_cache=(kg, vg), prefix_length=0)]
+    for t in range(6, S):
+        y_cpu.append(blk_cpu(x[:, t : t + 1], kv_cache=(kc, vc), prefix_length=t))
+        y_gpu.append(blk_gpu(x[:, t : t + 1].cuda().bfloat16(), kv_cache=(kg, vg), prefix_length=t))
+    ref = torch.cat(y_cpu, 1)
+    out = torch.cat([y.float().cpu() for y in y_gpu], 1)
+    # int8 weights: looser bound than bf16/nf4 block tests
+    assert torch.allclose(out, ref, atol=0.08, rtol=0.08), (out - ref).abs().max()

@@ -95,3 +95,18 @@ class TestNF4GPU:
             outs[quant] = h.float().cpu()
         diff = (outs["none"] - outs["nf4"]).abs().max()
         assert diff < 0.2, diff  # quantization error, not a wiring bug
+
+
+def test_int8_weight_roundtrip_error():
+    """Per-out-column symmetric int8 (the _FastWeight int8 scheme) keeps
+    relative weight error within ~1% on gaussian weights (CPU math)."""
+    import torch
+
+    torch.manual_seed(2)
+    w = torch.randn(512, 768) * 0.05
+    scale = w.abs().amax(dim=0).clamp_min(1e-8) / 127.0
+    q = torch.round(w / scale).clamp(-127, 127).to(torch.int8)
+    deq = q.float() * scale
+    rel = (deq - w).norm() / w.norm()
+    assert rel < 0.012, rel.item()
+    assert (deq - w).abs().max() <= scale.max() * 0.5 + 1e-6
