@@ -179,3 +179,54 @@ def test_auto_span_selection(hf_checkpoint):
         s1.shutdown()
         s2.shutdown()
         boot.shutdown()
+
+
+def test_inference_failover_mid_session(hf_checkpoint):
+    """Kill the server an open session is using; the session must recover by
+    replaying its input history into a redundant server, with logits matching
+    the uninterrupted reference (the petals failover semantics,
+    reference client/inference_session.py history replay)."""
+    from petals_amd.dht.node import DHT
+    from petals_amd.server.server import Server
+    from petals_amd.utils.auto_config import AutoDistributedModelForCausalLM
+
+    path, hf_model = hf_checkpoint
+    boot = DHT(host="127.0.0.1")
+    servers = []
+    for _ in range(2):  # two full-coverage servers
+        s = Server(
+            path, initial_peers=[boot.listen_addr], host="127.0.0.1", device="cpu",
+            torch_dtype="float32", block_indices="0:4", dht_prefix="test-failover",
+            throughput=1.0, update_period=1.0,
+        )
+        servers.append(s.start())
+    model = AutoDistributedModelForCausalLM.from_pretrained(
+        path, initial_peers=[boot.listen_addr], dht_prefix="test-failover",
+        show_route=False, max_retries=3, min_backoff=0.2, request_timeout=20.0,
+    )
+    try:
+        torch.manual_seed(7)
+        ids = torch.randint(0, 128, (1, 8))
+        with torch.no_grad():
+            ref = model(input_ids=ids).logits  # uninterrupted reference
+
+        outs = []
+        with model.transformer.h.inference_session(max_length=16) as sess:
+            with model.transformer.h.use_session(sess):
+                with torch.no_grad():
+                    outs.append(model(input_ids=ids[:, :4]).logits)
+                    # kill the server this session is attached to
+                    used_peer = sess._sessions[0].span.peer_id
+                    victim = next(s for s in servers if s.peer_id == used_peer)
+                    victim.shutdown()
+                    for t in range(4, 8):
+                        outs.append(model(input_ids=ids[:, t : t + 1]).logits)
+                survivor = next(s for s in servers if s.peer_id != used_peer)
+                assert sess._sessions[0].span.peer_id == survivor.peer_id, "session must have failed over"
+        step_logits = torch.cat(outs, dim=1)
+        assert torch.allclose(step_logits, ref, atol=1e-4, rtol=1e-3), (step_logits - ref).abs().max()
+    finally:
+        model.transformer.h.sequence_manager.shutdown()
+        for s in servers:
+            s.shutdown()
+        boot.shutdown()
