@@ -230,3 +230,33 @@ def test_inference_failover_mid_session(hf_checkpoint):
         for s in servers:
             s.shutdown()
         boot.shutdown()
+
+
+def test_server_to_server_push_used(client_model, monkeypatch):
+    """Steady-state multi-span decode must take the rpc_push fast path (server
+    hands activations to the next server; the client only talks to the first
+    and last spans) — and still match the one-shot forward logits."""
+    from petals_amd.client import inference_session as isess
+
+    calls = {"push": 0}
+    orig = isess.InferenceSession._step_pushed
+
+    def spy(self, *a, **k):
+        calls["push"] += 1
+        return orig(self, *a, **k)
+
+    monkeypatch.setattr(isess.InferenceSession, "_step_pushed", spy)
+    torch.manual_seed(9)
+    ids = torch.randint(0, 128, (1, 6))
+    with torch.no_grad():
+        ref = client_model(input_ids=ids).logits
+    outs = []
+    with client_model.transformer.h.inference_session(max_length=12) as sess:
+        with client_model.transformer.h.use_session(sess):
+            with torch.no_grad():
+                outs.append(client_model(input_ids=ids[:, :3]).logits)
+                for t in range(3, 6):
+                    outs.append(client_model(input_ids=ids[:, t : t + 1]).logits)
+    assert calls["push"] >= 1, "server-to-server push was never used"
+    step_logits = torch.cat(outs, dim=1)
+    assert torch.allclose(step_logits, ref, atol=1e-4, rtol=1e-3), (step_logits - ref).abs().max()
