@@ -260,3 +260,36 @@ def test_server_to_server_push_used(client_model, monkeypatch):
     assert calls["push"] >= 1, "server-to-server push was never used"
     step_logits = torch.cat(outs, dim=1)
     assert torch.allclose(step_logits, ref, atol=1e-4, rtol=1e-3), (step_logits - ref).abs().max()
+
+
+def test_beam_search_matches_local_mirror(client_model, hf_checkpoint):
+    """Remote beam search (beams = server-side batch rows, KV caches reordered
+    via hypo_ids each step) vs a local mirror of the same algorithm that
+    recomputes from scratch every step: identical sequences prove the
+    server-side cache reorder is correct."""
+    import torch.nn.functional as F
+
+    _, hf_model = hf_checkpoint
+    torch.manual_seed(3)
+    ids = torch.randint(0, 128, (1, 5))
+    num_beams, new_tokens = 2, 5
+    out = client_model.generate(ids, max_new_tokens=new_tokens, num_beams=num_beams)
+
+    expanded = ids.expand(num_beams, -1)
+    with torch.no_grad():
+        logits = hf_model(expanded).logits[:, -1, :].float()
+    logprobs = F.log_softmax(logits[0:1], dim=-1)
+    scores, next_tokens = logprobs.topk(num_beams, dim=-1)
+    beam_scores = scores[0]
+    sequences = torch.cat([expanded, next_tokens[0][:, None]], dim=1)
+    for _ in range(new_tokens - 1):
+        with torch.no_grad():
+            logits = hf_model(sequences).logits[:, -1, :].float()  # full recompute, no KV cache
+        logprobs = F.log_softmax(logits, dim=-1)
+        total = beam_scores[:, None] + logprobs
+        vocab = total.shape[-1]
+        beam_scores, flat_idx = total.reshape(-1).topk(num_beams)
+        beam_idx, token_idx = flat_idx // vocab, flat_idx % vocab
+        sequences = torch.cat([sequences[beam_idx], token_idx[:, None]], dim=1)
+    ref = sequences[beam_scores.argmax()][None]
+    assert torch.equal(out, ref), (out, ref)
