@@ -178,6 +178,18 @@ class Server:
                 loop.run_until_complete(self._ashutdown())
             except Exception:  # noqa: BLE001
                 pass
+            # drain leftovers (e.g. rpc_inference generators awaiting a pushed
+            # step when a client session dies with the server) so closing the
+            # loop doesn't finalize their coroutines on a dead loop
+            try:
+                pending = [t for t in asyncio.all_tasks(loop) if not t.done()]
+                for t in pending:
+                    t.cancel()
+                if pending:
+                    loop.run_until_complete(asyncio.gather(*pending, return_exceptions=True))
+                loop.run_until_complete(loop.shutdown_asyncgens())
+            except Exception:  # noqa: BLE001
+                pass
             loop.close()
 
     async def _amain(self):
