@@ -115,6 +115,21 @@ def nf4_bench():
 
 
 
+def int8_bench():
+    for name, in_dim, out_dim in (("qkv", 8192, 10240), ("gateup", 8192, 57344), ("down", 28672, 8192)):
+        wt = (torch.randn(in_dim, out_dim, device="cuda") * 0.02).float()
+        scale = (wt.abs().amax(dim=0).clamp_min(1e-8) / 127.0)
+        q = torch.round(wt / scale).clamp(-127, 127).to(torch.int8).contiguous()
+        sc = scale.to(torch.bfloat16).contiguous()
+        del wt
+        x = torch.randn(1, in_dim, device="cuda")
+        ws = torch.empty(0, device="cuda")
+        t = bench_kernel(lambda: hip.gemv_int8(q, sc, x, ws, None, 0))
+        gb = q.numel() / 1e9
+        print(f"gemv_int8 {name} [{in_dim},{out_dim}]: {t*1e6:.1f} us, {gb/t:.0f} GB/s ({gb*1000:.0f}MB)", flush=True)
+        torch.cuda.empty_cache()
+
+
 def prefill_attn_bench():
     import math
     from petals_amd.ops import reference
@@ -147,18 +162,3 @@ if __name__ == "__main__":
     for name, fn in _all.items():
         if not picked or name in picked:
             fn()
-
-
-def int8_bench():
-    for name, in_dim, out_dim in (("qkv", 8192, 10240), ("gateup", 8192, 57344), ("down", 28672, 8192)):
-        wt = (torch.randn(in_dim, out_dim, device="cuda") * 0.02).float()
-        scale = (wt.abs().amax(dim=0).clamp_min(1e-8) / 127.0)
-        q = torch.round(wt / scale).clamp(-127, 127).to(torch.int8).contiguous()
-        sc = scale.to(torch.bfloat16).contiguous()
-        del wt
-        x = torch.randn(1, in_dim, device="cuda")
-        ws = torch.empty(0, device="cuda")
-        t = bench_kernel(lambda: hip.gemv_int8(q, sc, x, ws, None, 0))
-        gb = q.numel() / 1e9
-        print(f"gemv_int8 {name} [{in_dim},{out_dim}]: {t*1e6:.1f} us, {gb/t:.0f} GB/s ({gb*1000:.0f}MB)", flush=True)
-        torch.cuda.empty_cache()
