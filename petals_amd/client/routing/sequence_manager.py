@@ -64,22 +64,26 @@ class RemoteSequenceManager:
 
     # ------------------------------------------------------------- updates
 
+    def _filter_servers(self, infos) -> None:
+        """Drop banned / not-allowed / blocked servers in place (client policy:
+        reference sequence_manager.py bans + ClientConfig allow/block lists)."""
+        now = time.monotonic()
+        for info in infos:
+            if info is None:
+                continue
+            for peer_id in list(info.servers.keys()):
+                ban = self._bans.get(peer_id)
+                if ban is not None and ban.banned_until > now:
+                    del info.servers[peer_id]
+                elif self.config.allowed_servers is not None and peer_id not in self.config.allowed_servers:
+                    del info.servers[peer_id]
+                elif self.config.blocked_servers is not None and peer_id in self.config.blocked_servers:
+                    del info.servers[peer_id]
+
     def update(self, wait: bool = True) -> None:
         infos, addrs = get_remote_module_infos(self.dht, self.block_uids)
         with self._lock:
-            # filter allowed/blocked servers and active bans
-            now = time.monotonic()
-            for info in infos:
-                if info is None:
-                    continue
-                for peer_id in list(info.servers.keys()):
-                    ban = self._bans.get(peer_id)
-                    if ban is not None and ban.banned_until > now:
-                        del info.servers[peer_id]
-                    elif self.config.allowed_servers is not None and peer_id not in self.config.allowed_servers:
-                        del info.servers[peer_id]
-                    elif self.config.blocked_servers is not None and peer_id in self.config.blocked_servers:
-                        del info.servers[peer_id]
+            self._filter_servers(infos)
             self.state.update_([i if (i and i.servers) else None for i in infos])
             self.addrs.update(addrs)
             self._last_update = time.monotonic()

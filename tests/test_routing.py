@@ -46,16 +46,9 @@ class _FakeManager(RemoteSequenceManager):
 
     def update(self, wait=True):
         infos = make_infos(len(self.block_uids), self._spans_src)
-        now = time.monotonic()
-        for info in infos:
-            if info is None:
-                continue
-            for pid in list(info.servers):
-                ban = self._bans.get(pid)
-                if ban and ban.banned_until > now:
-                    del info.servers[pid]
+        self._filter_servers(infos)  # the REAL ban/allow/block policy
         self.state.update_([i if (i and i.servers) else None for i in infos])
-        self._last_update = now
+        self._last_update = time.monotonic()
 
 
 def test_max_throughput_covers_all_blocks():
@@ -108,3 +101,16 @@ def test_should_choose_other_blocks():
     spans = {"A": (0, 4, 1.0), "B": (4, 8, 1.0)}
     infos = make_infos(8, spans)
     assert not block_selection.should_choose_other_blocks("A", infos, balance_quality=0.75)
+
+
+def test_allowed_and_blocked_servers():
+    """ClientConfig allow/deny lists filter the routing table."""
+    spans = {"A": (0, 8, 1.0), "B": (0, 8, 1.0)}
+    mgr = _FakeManager(8, spans, allowed_servers=["A"])
+    mgr.update()
+    for _ in range(4):
+        assert [s.peer_id for s in mgr.make_sequence(mode="max_throughput")] == ["A"]
+    mgr = _FakeManager(8, spans, blocked_servers=["A"])
+    mgr.update()
+    for _ in range(4):
+        assert [s.peer_id for s in mgr.make_sequence(mode="max_throughput")] == ["B"]
