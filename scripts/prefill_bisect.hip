@@ -86,24 +86,73 @@ __global__ __launch_bounds__(WAVES * 64) void kern(
   const int kv_end = causal ? min(kv_len, wg_last_q_abs + 1) : kv_len;
   const size_t kv_base = (((size_t)b * kv_heads + kvh) * lmax) * HD;
 
+  constexpr int SIT = (KVT * (HD / 8) + WAVES * WAVE - 1) / (WAVES * WAVE);  // staging iters/thread
+  bf16x8 k_reg[SIT], v_reg[SIT];
+  if constexpr (VAR == 6) {
+    // preload tile 0 into registers
+#pragma unroll
+    for (int it = 0; it < SIT; ++it) {
+      const int idx = tid + it * WAVES * WAVE;
+      const int row = idx / (HD / 8);
+      const int c8 = (idx - row * (HD / 8)) * 8;
+      k_reg[it] = v_reg[it] = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+      if (idx < KVT * (HD / 8) && row < kv_end) {
+        k_reg[it] = *reinterpret_cast<const bf16x8*>(k + kv_base + (size_t)row * HD + c8);
+        v_reg[it] = *reinterpret_cast<const bf16x8*>(v + kv_base + (size_t)row * HD + c8);
+      }
+    }
+  }
+
   for (int j0 = 0; j0 < kv_end; j0 += KVT) {
     const int tile_n = min(KVT, kv_end - j0);
     __syncthreads();
-    for (int idx = tid; idx < KVT * (HD / 8); idx += WAVES * WAVE) {
-      const int row = idx / (HD / 8);
-      const int c8 = (idx - row * (HD / 8)) * 8;
-      bf16x8 kv8 = {0, 0, 0, 0, 0, 0, 0, 0};
-      bf16x8 vv8 = {0, 0, 0, 0, 0, 0, 0, 0};
-      if (j0 + row < kv_end) {
-        kv8 = *reinterpret_cast<const bf16x8*>(k + kv_base + (size_t)(j0 + row) * HD + c8);
-        vv8 = *reinterpret_cast<const bf16x8*>(v + kv_base + (size_t)(j0 + row) * HD + c8);
-      }
-      *reinterpret_cast<bf16x8*>(&k_lds[row][c8]) = kv8;
+    if constexpr (VAR == 6) {
+      // write the prefetched tile, then start the NEXT tile's global loads so
+      // they overlap the MFMA/softmax compute below
 #pragma unroll
-      for (int e = 0; e < 8; ++e)
-        *reinterpret_cast<unsigned short*>(&vt_raw[VT_BYTE(c8 + e, row * 2)]) = (unsigned short)vv8[e];
+      for (int it = 0; it < SIT; ++it) {
+        const int idx = tid + it * WAVES * WAVE;
+        if (idx < KVT * (HD / 8)) {
+          const int row = idx / (HD / 8);
+          const int c8 = (idx - row * (HD / 8)) * 8;
+          *reinterpret_cast<bf16x8*>(&k_lds[row][c8]) = k_reg[it];
+#pragma unroll
+          for (int e = 0; e < 8; ++e)
+            *reinterpret_cast<unsigned short*>(&vt_raw[VT_BYTE(c8 + e, row * 2)]) = (unsigned short)v_reg[it][e];
+        }
+      }
+      __syncthreads();
+      if (j0 + KVT < kv_end) {
+        const int jn = j0 + KVT;
+#pragma unroll
+        for (int it = 0; it < SIT; ++it) {
+          const int idx = tid + it * WAVES * WAVE;
+          const int row = idx / (HD / 8);
+          const int c8 = (idx - row * (HD / 8)) * 8;
+          k_reg[it] = v_reg[it] = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+          if (idx < KVT * (HD / 8) && jn + row < kv_end) {
+            k_reg[it] = *reinterpret_cast<const bf16x8*>(k + kv_base + (size_t)(jn + row) * HD + c8);
+            v_reg[it] = *reinterpret_cast<const bf16x8*>(v + kv_base + (size_t)(jn + row) * HD + c8);
+          }
+        }
+      }
+    } else {
+      for (int idx = tid; idx < KVT * (HD / 8); idx += WAVES * WAVE) {
+        const int row = idx / (HD / 8);
+        const int c8 = (idx - row * (HD / 8)) * 8;
+        bf16x8 kv8 = {0, 0, 0, 0, 0, 0, 0, 0};
+        bf16x8 vv8 = {0, 0, 0, 0, 0, 0, 0, 0};
+        if (j0 + row < kv_end) {
+          kv8 = *reinterpret_cast<const bf16x8*>(k + kv_base + (size_t)(j0 + row) * HD + c8);
+          vv8 = *reinterpret_cast<const bf16x8*>(v + kv_base + (size_t)(j0 + row) * HD + c8);
+        }
+        *reinterpret_cast<bf16x8*>(&k_lds[row][c8]) = kv8;
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          *reinterpret_cast<unsigned short*>(&vt_raw[VT_BYTE(c8 + e, row * 2)]) = (unsigned short)vv8[e];
+      }
+      __syncthreads();
     }
-    __syncthreads();
 
     f32x4 s_acc[NB];
 #pragma unroll
@@ -149,7 +198,7 @@ __global__ __launch_bounds__(WAVES * 64) void kern(
         m_row[r] = m_new;
         p[0][r] = p0;
         p[1][r] = p1;
-      } else if constexpr (VAR == 5) {
+      } else if constexpr (VAR == 5 || VAR == 6) {
         // split-mask form + mask-free interior tiles (all rows of this wave
         // live for the whole tile: no tail, no causal edge, no s_q edge)
         float s[NB];
@@ -400,6 +449,8 @@ int main() {
   RUN(128, 64, 2, "KVT64 VAR2 (split mask/mx)")
   RUN(128, 64, 4, "KVT64 VAR4 (split + P swz)")
   RUN(128, 64, 5, "KVT64 VAR5 (interior fast)")
+  RUN(128, 64, 6, "KVT64 VAR6 (prefetch)")
+  RUN(128, 32, 6, "KVT32 VAR6 (prefetch)")
 
   // ---- timing at the llama-2-70b prefill shape ----
   {
@@ -438,6 +489,7 @@ int main() {
     TIME(64, 2, "time KVT64 split")
     TIME(64, 4, "time KVT64 split+Pswz")
     TIME(64, 5, "time KVT64 interior-fast")
+    TIME(64, 6, "time KVT64 prefetch")
   }
   return 0;
 }
