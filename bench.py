@@ -183,7 +183,10 @@ def main():
                     logits = head_logits(gs["h_back"][:, -1, :])
                     gs["cur_id"].copy_(logits.argmax(dim=-1, keepdim=True))
                     return logits
-                gs["g_head"] = GraphedCallable(head_fn, [])
+                if not os.environ.get("PETALS_AMD_NO_HEAD_GRAPH"):  # debug bisect
+                    gs["g_head"] = GraphedCallable(head_fn, [])
+                else:
+                    gs["head_eager"] = head_fn
         else:
             gs["h_in"] = torch.empty(B, 1, H, device=device, dtype=dtype)
 
