@@ -135,11 +135,14 @@ async def sequential_backward(
                     sequence_manager, inputs, prompts, start_index=span.start, end_index=span.end
                 )
                 assert len(new_inters) > 0 and len(new_spans) > 0
-                # substitute the failed span with the new sub-chain
-                intermediate_inputs.extend(new_inters[1:])
-                forward_sequences.extend(new_spans[1:])
-                inputs = new_inters[0]
-                span = new_spans[0]
+                # substitute the failed span with the new sub-chain; grad_outputs
+                # corresponds to the LAST sub-span's output, so backward must
+                # proceed deepest-sub-span first: push all recomputed entries and
+                # pop the last one
+                intermediate_inputs.extend(new_inters)
+                forward_sequences.extend(new_spans)
+                inputs = intermediate_inputs.pop()
+                span = forward_sequences.pop()
 
     grad_prompts = (
         torch.cat(list(reversed(grad_prompts_reversed)), dim=0) if grad_prompts_reversed else DUMMY
@@ -196,7 +199,9 @@ class _RemoteSequentialAutogradFunction(torch.autograd.Function):
         forward_sequences: List[Sequence[Any]] = ctx.sequences_for_batches
         sequence_manager = ctx.sequence_manager
 
-        batch_size = max(grad_outputs.shape[0] // len(intermediate_input_batches), 1)
+        # must mirror forward's split boundaries exactly (same formula), else
+        # grad splits diverge from input splits when the batch doesn't divide evenly
+        batch_size = max(MAX_TOKENS_IN_BATCH // max(grad_outputs.shape[1], 1), 1)
         grad_output_batches: Sequence[torch.Tensor] = grad_outputs.split(batch_size)
         assert len(grad_output_batches) == len(intermediate_input_batches)
 

@@ -17,8 +17,12 @@ from petals_amd.data_structures import RemoteModuleInfo, RemoteSpanInfo, ServerS
 
 
 def compute_throughputs(spans: Dict[str, RemoteSpanInfo], *, total_blocks: int) -> np.ndarray:
+    # deterministic accumulation order: FP addition is not associative, so an
+    # undefined order yields slightly different sums for the same server set and
+    # can cause excess block replacements (swarm churn)
     throughputs = np.zeros(total_blocks)
-    for span in spans.values():
+    for peer_id in sorted(spans):
+        span = spans[peer_id]
         if span.state != ServerState.OFFLINE:
             throughputs[span.start : span.end] += span.throughput
     return throughputs

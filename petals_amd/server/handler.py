@@ -460,19 +460,21 @@ class TransformerConnectionHandler:
                     await task
             if not done:
                 raise RpcError("inference step timed out")
-            task = done.pop()
-            try:
-                item = task.result()
-            except RpcError as e:
-                if "closed" in str(e):
-                    return  # client closed the stream: session over
-                raise
-            if isinstance(item, RpcMessage):
-                if item.kind == "end" and not item.tensors and not item.meta.get("step_id"):
-                    return  # graceful close
-                yield item.meta, item.tensors
-            else:  # pushed step: (meta, tensors)
-                yield item
+            # both tasks may complete in the same tick; each completed task has
+            # already consumed its message, so process EVERY one or a step is lost
+            for task in done:
+                try:
+                    item = task.result()
+                except RpcError as e:
+                    if "closed" in str(e):
+                        return  # client closed the stream: session over
+                    raise
+                if isinstance(item, RpcMessage):
+                    if item.kind == "end" and not item.tensors and not item.meta.get("step_id"):
+                        return  # graceful close
+                    yield item.meta, item.tensors
+                else:  # pushed step: (meta, tensors)
+                    yield item
 
     async def rpc_push(self, request: RpcMessage, stream: RpcStream) -> None:
         session_id = request.meta.get("session_id")
