@@ -146,6 +146,121 @@ class RpcStream:
 
 RpcHandler = Callable[[RpcMessage, RpcStream], Awaitable[None]]
 
+# ---------------------------------------------------------------- in-process
+# Registry of P2PNodes listening in THIS process. When a client dials an
+# address served by the same process (co-located client+server — the common
+# deployment for a GPU node that both serves blocks and runs the thin client),
+# the RPC bypasses sockets AND serialization entirely: RpcMessages cross
+# loops by reference, so GPU tensors never leave the device. Disable with
+# PETALS_AMD_NO_INPROC=1 (used by wire-level tests).
+
+_INPROC_NODES: Dict[Tuple[str, int], "P2PNode"] = {}
+
+
+def _inproc_enabled() -> bool:
+    return not os.environ.get("PETALS_AMD_NO_INPROC")
+
+
+class InProcStream:
+    """One endpoint of an in-process RPC stream. Mirrors RpcStream's interface
+    and EOF/error semantics; each endpoint is owned by one asyncio loop and
+    fed thread-safely from the peer's loop."""
+
+    _EOF = StopAsyncIteration
+
+    def __init__(self, owner_loop: asyncio.AbstractEventLoop, rpc: str, rid: int):
+        self._loop = owner_loop
+        self.rpc = rpc
+        self.rid = rid
+        self._items = __import__("collections").deque()
+        self._lock = __import__("threading").Lock()
+        self._event: Optional[asyncio.Event] = None  # created lazily ON the owner loop
+        self.peer: Optional["InProcStream"] = None
+        self._closed_outbound = False
+        self.is_inproc = True
+
+    # -- feeding (called from any thread)
+
+    def _deliver(self, item) -> None:
+        with self._lock:
+            self._items.append(item)
+        try:
+            self._loop.call_soon_threadsafe(self._notify)
+        except RuntimeError:
+            pass  # owner loop already closed
+
+    def _notify(self) -> None:
+        if self._event is not None:
+            self._event.set()
+
+    def _feed_eof(self, exc_text: Optional[str] = None) -> None:
+        self._deliver(exc_text if exc_text is not None else self._EOF)
+
+    # -- receiving (called on the owner loop)
+
+    def _pop(self):
+        with self._lock:
+            return self._items.popleft() if self._items else None
+
+    async def _next_item(self, timeout: Optional[float]):
+        if self._event is None:
+            self._event = asyncio.Event()
+        deadline = None if timeout is None else asyncio.get_event_loop().time() + timeout
+        while True:
+            self._event.clear()
+            item = self._pop()
+            if item is not None:
+                return item
+            remaining = None if deadline is None else deadline - asyncio.get_event_loop().time()
+            if remaining is not None and remaining <= 0:
+                raise asyncio.TimeoutError()
+            await asyncio.wait_for(self._event.wait(), remaining)
+
+    async def receive(self, timeout: Optional[float] = None) -> RpcMessage:
+        item = await self._next_item(timeout)
+        if item is self._EOF:
+            self._deliver(self._EOF)  # sticky
+            raise RpcError(f"stream {self.rpc}#{self.rid}: closed")
+        if isinstance(item, str):
+            self._deliver(item)
+            raise RpcError(f"stream {self.rpc}#{self.rid}: remote error: {item}")
+        return item
+
+    def __aiter__(self):
+        return self
+
+    async def __anext__(self) -> RpcMessage:
+        item = await self._next_item(None)
+        if item is self._EOF:
+            self._deliver(self._EOF)
+            raise StopAsyncIteration
+        if isinstance(item, str):
+            self._deliver(item)
+            raise RpcError(f"stream {self.rpc}#{self.rid}: remote error: {item}")
+        if item.kind == "end":
+            self._deliver(self._EOF)
+        return item
+
+    # -- sending
+
+    async def send(self, msg: RpcMessage, kind: str = "item", compressions=None) -> None:
+        if self._closed_outbound:
+            raise RpcError(f"stream {self.rpc}#{self.rid} already closed for sending")
+        if kind == "end":
+            self._closed_outbound = True
+        if self.peer is not None:
+            self.peer._deliver(RpcMessage(meta=dict(msg.meta), tensors=list(msg.tensors), kind=kind))
+
+    async def close(self, msg: Optional[RpcMessage] = None) -> None:
+        if not self._closed_outbound:
+            await self.send(msg or RpcMessage(), kind="end")
+
+    async def error(self, text: str) -> None:
+        if not self._closed_outbound:
+            self._closed_outbound = True
+            if self.peer is not None:
+                self.peer._feed_eof(text)
+
 
 class Connection:
     """One TCP connection; multiplexes many RPC streams."""
@@ -268,6 +383,8 @@ class P2PNode:
         self._conns: Dict[Tuple[str, int], Connection] = {}
         self._inbound: List[Connection] = []
         self._conn_locks: Dict[Tuple[str, int], asyncio.Lock] = {}
+        self._loop: Optional[asyncio.AbstractEventLoop] = None
+        self._inproc_keys: List[Tuple[str, int]] = []
 
     def add_handler(self, rpc: str, handler: RpcHandler) -> None:
         self.handlers[rpc] = handler
@@ -277,6 +394,11 @@ class P2PNode:
         sockets = self._server.sockets
         addr = sockets[0].getsockname()
         self.listen_addr = (host, addr[1])
+        self._loop = asyncio.get_event_loop()
+        for h in {host, "127.0.0.1"}:
+            key = (h, addr[1])
+            _INPROC_NODES[key] = self
+            self._inproc_keys.append(key)
         return self.listen_addr
 
     async def _on_inbound(self, reader: asyncio.StreamReader, writer: asyncio.StreamWriter):
@@ -325,6 +447,10 @@ class P2PNode:
 
         With ``end=True`` the request also closes our outbound side (unary
         request)."""
+        addr = (addr[0], int(addr[1]))
+        target = _INPROC_NODES.get(addr) if _inproc_enabled() else None
+        if target is not None and target._loop is not None and not target._loop.is_closed():
+            return self._open_inproc(target, rpc, request, end)
         conn = await self.connect(addr, timeout)
         rid = conn.next_rid()
         stream = RpcStream(conn, rpc, rid)
@@ -333,6 +459,29 @@ class P2PNode:
             stream._closed_outbound = True
         await conn.send_frame(_encode_envelope(rpc, rid, "req_end" if end else "req", request, compressions))
         return stream
+
+    def _open_inproc(self, target: "P2PNode", rpc: str, request: RpcMessage, end: bool) -> InProcStream:
+        local = InProcStream(asyncio.get_event_loop(), rpc, id(request))
+        remote = InProcStream(target._loop, rpc, id(request))
+        local.peer = remote
+        remote.peer = local
+        if end:
+            local._closed_outbound = True
+        first = RpcMessage(
+            meta=dict(request.meta), tensors=list(request.tensors), kind="end" if end else "item"
+        )
+
+        def _start():
+            handler = target.handlers.get(rpc)
+            if handler is None:
+                local._feed_eof(f"unknown rpc {rpc!r}")
+                return
+            if first.kind == "end":
+                remote._feed_eof()
+            asyncio.ensure_future(_run_inproc_handler(handler, first, remote))
+
+        target._loop.call_soon_threadsafe(_start)
+        return local
 
     async def call_unary(
         self, addr: Tuple[str, int], rpc: str, request: RpcMessage, timeout: float = 30.0, compressions=None
@@ -343,6 +492,10 @@ class P2PNode:
         return await stream.receive(timeout=timeout)
 
     async def shutdown(self):
+        for key in self._inproc_keys:
+            if _INPROC_NODES.get(key) is self:
+                _INPROC_NODES.pop(key, None)
+        self._inproc_keys.clear()
         for conn in list(self._conns.values()) + list(self._inbound):
             await conn.close()
         if self._server is not None:
@@ -351,3 +504,17 @@ class P2PNode:
                 await self._server.wait_closed()
             except Exception:  # noqa: BLE001
                 pass
+
+
+async def _run_inproc_handler(handler: RpcHandler, first: RpcMessage, stream: InProcStream):
+    try:
+        await handler(first, stream)
+        await stream.close()
+    except asyncio.CancelledError:
+        raise
+    except Exception as e:  # noqa: BLE001
+        logger.debug("in-proc rpc handler %s failed: %r", stream.rpc, e, exc_info=True)
+        try:
+            await stream.error(f"{type(e).__name__}: {e}")
+        except Exception:  # noqa: BLE001
+            pass

@@ -23,7 +23,10 @@ def swarm():
 
 
 @pytest.mark.parametrize("compression", ["float16", "blockwise_8bit"])
-def test_compressed_forward_close(swarm, compression):
+def test_compressed_forward_close(swarm, compression, monkeypatch):
+    # this test exercises the WIRE (lossy compression); the in-process
+    # transport shortcut is lossless by design, so force sockets here
+    monkeypatch.setenv("PETALS_AMD_NO_INPROC", "1")
     from petals_amd.utils.auto_config import AutoDistributedModel
 
     boot = swarm
