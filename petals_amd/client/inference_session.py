@@ -15,6 +15,7 @@ multi-token step with start_from_position, rebuilding its cache exactly.
 
 from __future__ import annotations
 
+import asyncio
 import itertools
 import logging
 import time
@@ -169,10 +170,12 @@ class InferenceSession:
             assert hypo_ids.dtype == torch.int64
 
         inputs_device, inputs_dtype = inputs.device, inputs.dtype
-        inputs = inputs.cpu()
+        # inputs stay on their device: the in-proc/mesh paths move them without
+        # ever visiting the host, and socket serialization does .cpu() itself
+        inputs = inputs.detach()
         if inputs.dtype not in (torch.bfloat16, torch.float16, torch.float32):
             inputs = inputs.float()
-        prompts = prompts.cpu() if not is_dummy(prompts) else prompts
+        prompts = prompts.detach() if not is_dummy(prompts) else prompts
         n = inputs.shape[1]
         if self._position + n > self._max_length:
             raise ValueError(f"max_length exceeded: {self._position} + {n} > {self._max_length}")
@@ -306,24 +309,51 @@ class InferenceSession:
 
     def _step_pushed(self, inputs: torch.Tensor, hypo_ids: torch.Tensor, step_id: str) -> torch.Tensor:
         """Send to the first span; servers hand activations to each other over
-        rpc_push; we await the last span's output."""
+        rpc_push (RCCL for co-located spans); we await the last span's output.
+        When the last span's server shares a LocalMesh with THIS process, the
+        final activation also comes back over RCCL and the client stream
+        carries only metadata."""
+        from petals_amd.parallel.mesh import get_local_mesh
+
         first, last = self._sessions[0], self._sessions[-1]
         next_servers = []
         for s in self._sessions[1:]:
             host, port = self._manager.address_of(s.span.peer_id)
-            next_servers.append([host, port, s.session_id, s.span.start, s.span.end])
+            si = s.span.server_info
+            next_servers.append(
+                [host, port, s.session_id, s.span.start, s.span.end,
+                 getattr(si, "mesh_id", None), getattr(si, "mesh_rank", None)]
+            )
         meta = {
             "step_id": step_id,
             "start_from_position": self._position,
             "next_servers": next_servers,
         }
+        last_info = last.span.server_info
+        local_mesh = get_local_mesh(getattr(last_info, "mesh_id", None))
+        if (
+            local_mesh is not None
+            and getattr(last_info, "mesh_rank", None) is not None
+            and last_info.mesh_rank != local_mesh.rank
+        ):
+            meta["output_via_mesh"] = {"mesh_id": local_mesh.mesh_id, "rank": local_mesh.rank}
         tensors = [inputs, DUMMY, hypo_ids]
 
         async def roundtrip():
             await first._asend_step(meta, tensors)
             while True:
                 msg = await last._arecv_step(self._manager.config.request_timeout)
-                if msg.meta.get("step_id") == step_id and msg.tensors:
+                if msg.meta.get("step_id") != step_id:
+                    continue
+                tvm = msg.meta.get("tensors_via_mesh")
+                if tvm is not None:
+                    handle = local_mesh.post_recv(
+                        int(tvm["src_rank"]), int(tvm["ticket"]), tvm["shape"], tvm["dtype"]
+                    )
+                    return await asyncio.wait_for(
+                        asyncio.wrap_future(handle.future), self._manager.config.request_timeout
+                    )
+                if msg.tensors:
                     return msg.tensors[0]
 
         out = self._manager.run_coroutine(roundtrip(), timeout=self._manager.config.request_timeout + 10)

@@ -62,6 +62,10 @@ class ServerInfo:
     using_relay: bool = False
     cache_tokens_left: Optional[int] = None
     next_pings: Optional[Dict[str, float]] = None  # peer_id hex -> rtt seconds
+    # co-located RCCL/xGMI tier (parallel/mesh.py): servers sharing a mesh
+    # exchange activations over RCCL p2p instead of TCP
+    mesh_id: Optional[str] = None
+    mesh_rank: Optional[int] = None
 
     def to_dict(self) -> Dict[str, Any]:
         d = dataclasses.asdict(self)

@@ -130,25 +130,19 @@ class LocalMesh:
 
     # ------------------------------------------------------------ transfers
 
-    def next_send_ticket(self, dst: int) -> int:
-        """Reserve the ticket for an upcoming send to `dst`. Call ONLY from a
-        single thread per server (the asyncio loop) so that reservation order
-        equals enqueue order."""
-        with self._lock:
-            t = self._send_tickets[dst]
-            self._send_tickets[dst] = t + 1
-            return t
-
-    def send(self, tensor: torch.Tensor, dst: int, ticket: int, ready_event=None) -> Future:
-        """Enqueue a send; returns a Future resolved when the transfer is on
-        the wire (stream-synchronized)."""
+    def send(self, tensor: torch.Tensor, dst: int, ready_event=None) -> Tuple[Future, int]:
+        """Enqueue a send; the per-dst ticket is assigned atomically at enqueue
+        time (so ticket order == issue order, with no holes). Returns
+        (future resolved when the transfer is on the wire, ticket)."""
         fut: Future = Future()
         with self._cv:
             if self._broken:
                 raise MeshError(f"mesh broken: {self._broken}")
+            ticket = self._send_tickets[dst]
+            self._send_tickets[dst] = ticket + 1
             self._queue.append(("send", dst, ticket, tensor, ready_event, fut))
             self._cv.notify_all()
-        return fut
+        return fut, ticket
 
     def post_recv(self, src: int, ticket: int, shape, dtype_str: str) -> MeshRecvHandle:
         """Register an expected inbound transfer (called when the TCP meta for
@@ -257,6 +251,25 @@ class LocalMesh:
 
     def _any_ready_locked(self) -> bool:
         return any(self._recv_next[src] in pending for src, pending in self._recv_pending.items())
+
+
+# Process-local mesh registry: the thin client (and bench harnesses) in a
+# server process use the same mesh for final-output delivery.
+_LOCAL_MESH: Dict[str, LocalMesh] = {}
+
+
+def register_local_mesh(mesh: LocalMesh) -> None:
+    _LOCAL_MESH[mesh.mesh_id] = mesh
+
+
+def get_local_mesh(mesh_id: Optional[str] = None) -> Optional[LocalMesh]:
+    if mesh_id is not None:
+        m = _LOCAL_MESH.get(mesh_id)
+        return m if m is not None and m.is_usable else None
+    for m in _LOCAL_MESH.values():
+        if m.is_usable:
+            return m
+    return None
 
 
 def _wait_and_resolve(work, fut: Future, result, keepalive=()):

@@ -93,6 +93,7 @@ class TransformerConnectionHandler:
         server_info_extra=None,
         p2p: Optional[P2PNode] = None,
         adapters: Sequence[str] = (),
+        mesh=None,
     ):
         self.backends = backends
         self.memory_cache = memory_cache
@@ -105,6 +106,7 @@ class TransformerConnectionHandler:
         self.server_info_extra = server_info_extra or (lambda: {})
         self.adapters = tuple(adapters)
         self.p2p = p2p
+        self.mesh = mesh  # LocalMesh: RCCL/xGMI hand-off tier for co-located spans
         self._sessions: Dict[str, _Session] = {}
 
     def register(self, p2p: P2PNode) -> None:
@@ -299,6 +301,7 @@ class TransformerConnectionHandler:
         prefix_length: int,
         active_adapter: Optional[str],
         session: Optional[_Session] = None,
+        keep_on_device: bool = False,
     ) -> torch.Tensor:
         """Runs IN the runtime thread: one inference step through the whole span
         (the single-process analog of reference _MergedInferenceStep)."""
@@ -349,11 +352,13 @@ class TransformerConnectionHandler:
                 out = session.span_graph.step(hidden_states, prefix_length)
                 torch.cuda.synchronize()
                 _g1 = time.perf_counter()
-                res = out.cpu()
-                print(f"[rt] graph {( _g1-_g0)*1e3:.2f} d2h {(time.perf_counter()-_g1)*1e3:.2f} ms", flush=True)
+                # clone: the graph's static output is overwritten by the next
+                # replay; anything leaving the runtime must own its storage
+                res = out.clone() if keep_on_device else out.cpu()
+                print(f"[rt] graph {( _g1-_g0)*1e3:.2f} out {(time.perf_counter()-_g1)*1e3:.2f} ms", flush=True)
                 return res
             out = session.span_graph.step(hidden_states, prefix_length)
-            return out.cpu()
+            return out.clone() if keep_on_device else out.cpu()
 
         if has_hypo:
             hypo_ids = hypo_ids.to(device)
@@ -367,7 +372,7 @@ class TransformerConnectionHandler:
                     hidden_states[:, : prompt.shape[1]] += prompt
                 info = InferenceMetadata(uid, prefix_length, tuple(handle_pair), active_adapter)
                 (hidden_states,) = self.backends[uid].inference_step(hidden_states, hypo_ids, info)
-        return hidden_states.cpu()
+        return hidden_states if keep_on_device else hidden_states.cpu()
 
     async def rpc_inference(self, request: RpcMessage, stream: RpcStream) -> None:
         meta = request.meta
@@ -401,6 +406,10 @@ class TransformerConnectionHandler:
                             raise RpcError("start_from_position is ahead of the cache")
                         session.prefix_length = pos
                     hidden_states = tensors[0]
+                    if not torch.is_tensor(hidden_states):  # MeshRecvHandle: inputs arrive over RCCL
+                        hidden_states = await asyncio.wait_for(
+                            asyncio.wrap_future(hidden_states.future), self.step_timeout
+                        )
                     prompts = tensors[1] if len(tensors) > 1 else None
                     hypo_ids = tensors[2] if len(tensors) > 2 else None
                     length_increment = hidden_states.shape[1] if hidden_states.numel() > 0 else 0
@@ -411,6 +420,18 @@ class TransformerConnectionHandler:
                     step_start_position = session.prefix_length
                     _trace = os.environ.get("PETALS_AMD_STEP_TRACE")
                     _t0 = time.perf_counter()
+
+                    has_prompts = prompts is not None and not is_dummy(prompts)
+                    next_servers = step_meta.get("next_servers")
+                    will_push = bool(next_servers) and not has_prompts and length_increment > 0
+                    output_via_mesh = step_meta.get("output_via_mesh")
+                    # keep the output on the GPU when it leaves over RCCL (mesh
+                    # push / mesh output) or stays in-process (co-located client)
+                    keep_on_device = (
+                        (will_push and self._mesh_dst(next_servers[0]) is not None)
+                        or (not will_push and self._mesh_out_dst(output_via_mesh) is not None)
+                        or getattr(stream, "is_inproc", False)
+                    )
                     if length_increment > 0:
                         priority = self.prioritizer.prioritize(hidden_states, type="inference")
                         output = await self.runtime.submit(
@@ -424,6 +445,7 @@ class TransformerConnectionHandler:
                             session.prefix_length,
                             active_adapter,
                             session,
+                            keep_on_device,
                         )
                     else:
                         output = hidden_states
@@ -431,17 +453,25 @@ class TransformerConnectionHandler:
                         print(f"[srv] compute {(time.perf_counter()-_t0)*1e3:.2f} ms", flush=True)
                     session.prefix_length += length_increment
 
-                    has_prompts = prompts is not None and not is_dummy(prompts)
-                    next_servers = step_meta.get("next_servers")
                     pushed = False
-                    if next_servers and not has_prompts and length_increment > 0:
+                    if will_push:
                         pushed = await self._push_outputs(output, step_meta, next_servers, step_start_position)
                     if not pushed:
                         # the last server of a push chain (or any server when push
                         # is off/failed) returns outputs on its client stream
-                        await stream.send(
-                            RpcMessage(meta={"step_id": step_meta.get("step_id")}, tensors=[output])
-                        )
+                        out_meta = {"step_id": step_meta.get("step_id")}
+                        mesh_dst = self._mesh_out_dst(output_via_mesh)
+                        if mesh_dst is not None and torch.is_tensor(output) and output.device.type == self.mesh.device.type:
+                            try:
+                                fut, ticket = self.mesh.send(output, dst=mesh_dst)
+                                out_meta["tensors_via_mesh"] = self._mesh_desc(output, ticket)
+                                await stream.send(RpcMessage(meta=out_meta))
+                                continue
+                            except Exception as e:  # noqa: BLE001
+                                logger.warning("mesh output delivery failed (%r); using the stream", e)
+                        if output.device.type != "cpu" and not getattr(stream, "is_inproc", False):
+                            output = output.cpu()
+                        await stream.send(RpcMessage(meta=out_meta, tensors=[output]))
         finally:
             self._sessions.pop(session_id, None)
 
@@ -476,21 +506,68 @@ class TransformerConnectionHandler:
                 else:  # pushed step: (meta, tensors)
                     yield item
 
+    # ------------------------------------------------- mesh (RCCL) hand-off
+
+    def _mesh_dst(self, next_server_entry) -> Optional[int]:
+        """Next-server mesh rank iff it shares a usable mesh with this server.
+        Entries: [host, port, session_id, start, end, mesh_id, mesh_rank]."""
+        if self.mesh is None or not self.mesh.is_usable or self.mesh.device is None:
+            return None
+        if len(next_server_entry) < 7:
+            return None
+        mesh_id, mesh_rank = next_server_entry[5], next_server_entry[6]
+        if mesh_id != self.mesh.mesh_id or mesh_rank is None or mesh_rank == self.mesh.rank:
+            return None
+        return int(mesh_rank)
+
+    def _mesh_out_dst(self, output_via_mesh) -> Optional[int]:
+        """Client-requested final-output mesh delivery ({mesh_id, rank})."""
+        if not output_via_mesh or self.mesh is None or not self.mesh.is_usable or self.mesh.device is None:
+            return None
+        if output_via_mesh.get("mesh_id") != self.mesh.mesh_id:
+            return None
+        rank = output_via_mesh.get("rank")
+        if rank is None or rank == self.mesh.rank:
+            return None
+        return int(rank)
+
+    def _mesh_desc(self, tensor: torch.Tensor, ticket: int) -> Dict[str, Any]:
+        from petals_amd.utils.serialization import _DTYPE_TO_STR
+
+        return {
+            "mesh_id": self.mesh.mesh_id,
+            "src_rank": self.mesh.rank,
+            "ticket": ticket,
+            "shape": list(tensor.shape),
+            "dtype": _DTYPE_TO_STR[tensor.dtype],
+        }
+
     async def rpc_push(self, request: RpcMessage, stream: RpcStream) -> None:
         session_id = request.meta.get("session_id")
         session = self._sessions.get(session_id)
         if session is None:
             raise RpcError(f"no active inference session {session_id!r}")
-        session.pushed_inputs.put_nowait((dict(request.meta), list(request.tensors)))
+        tensors: List[Any] = list(request.tensors)
+        tvm = request.meta.get("tensors_via_mesh")
+        if tvm is not None:
+            if self.mesh is None or not self.mesh.is_usable or tvm.get("mesh_id") != self.mesh.mesh_id:
+                raise RpcError("mesh transfer addressed to a server without that mesh")
+            handle = self.mesh.post_recv(
+                int(tvm["src_rank"]), int(tvm["ticket"]), tvm["shape"], tvm["dtype"]
+            )
+            tensors = [handle]
+        session.pushed_inputs.put_nowait((dict(request.meta), tensors))
         await stream.close(RpcMessage(meta={"ok": True}))
 
     async def _push_outputs(
         self, output: torch.Tensor, step_meta: Dict[str, Any], next_servers, step_start_position: int
     ) -> bool:
         """Push this step's output into the next server's session. next_servers:
-        [[host, port, session_id, start_block, end_block], ...]; we contact the
-        first entry. start_from_position propagates down the chain so rollbacks
-        (speculative decoding) rewind every span's cache."""
+        [[host, port, session_id, start_block, end_block, mesh_id, mesh_rank],
+        ...]; we contact the first entry. start_from_position propagates down
+        the chain so rollbacks (speculative decoding) rewind every span's
+        cache. Co-located next servers get the activation over RCCL/xGMI (the
+        TCP message then carries only metadata)."""
         try:
             host, port, next_session_id = next_servers[0][0], next_servers[0][1], next_servers[0][2]
             meta = {
@@ -498,9 +575,19 @@ class TransformerConnectionHandler:
                 "step_id": step_meta.get("step_id"),
                 "next_servers": next_servers[1:],
                 "start_from_position": step_start_position,
+                "output_via_mesh": step_meta.get("output_via_mesh"),
             }
+            tensors = [output]
+            mesh_dst = self._mesh_dst(next_servers[0])
+            if mesh_dst is not None and output.device.type == self.mesh.device.type:
+                _fut, ticket = self.mesh.send(output, dst=mesh_dst)
+                meta["tensors_via_mesh"] = self._mesh_desc(output, ticket)
+                tensors = []
+            elif output.device.type != "cpu":
+                output = output.cpu()
+                tensors = [output]
             await asyncio.wait_for(
-                self.p2p.call_unary((host, port), "petals.rpc_push", RpcMessage(meta=meta, tensors=[output])),
+                self.p2p.call_unary((host, port), "petals.rpc_push", RpcMessage(meta=meta, tensors=tensors)),
                 timeout=10.0,
             )
             return True

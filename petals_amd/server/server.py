@@ -68,6 +68,7 @@ class Server:
         public_name: Optional[str] = None,
         announce_host: Optional[str] = None,
         skip_reachability_check: bool = False,
+        mesh=None,
     ):
         self.config = load_model_config(model_name_or_dir)
         self.model_name_or_dir = model_name_or_dir
@@ -104,6 +105,12 @@ class Server:
         self.public_name = public_name
         self.adapters = tuple(adapters)
         self._throughput_setting = throughput
+        # LocalMesh for the co-located RCCL/xGMI activation hand-off tier
+        self.mesh = mesh
+        if mesh is not None:
+            from petals_amd.parallel.mesh import register_local_mesh
+
+            register_local_mesh(mesh)
 
         self.module_uids = [make_uid(self.config.dht_prefix, i) for i in range(self.config.num_blocks)]
 
@@ -264,6 +271,8 @@ class Server:
             torch_dtype=str(self.torch_dtype).replace("torch.", ""),
             quant_type=self.quant_type,
             adapters=tuple(os.path.basename(os.path.normpath(a)) for a in self.adapters),
+            mesh_id=self.mesh.mesh_id if self.mesh is not None else None,
+            mesh_rank=self.mesh.rank if self.mesh is not None else None,
         )
         await self._announce()
 
@@ -295,6 +304,7 @@ class Server:
             inference_max_length=self.inference_max_length,
             p2p=self.p2p,
             adapters=tuple(os.path.basename(os.path.normpath(a)) for a in self.adapters),
+            mesh=self.mesh,
         )
         self.handler.register(self.p2p)
 

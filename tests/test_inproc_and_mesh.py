@@ -104,12 +104,12 @@ def _mesh_worker(rank, world, port, fail_q):
         mesh = LocalMesh("test-mesh", rank, world, device=torch.device("cpu"))
         n = 6
         if rank == 0:
-            # send n tensors to rank 1 in ticket order
+            # send n tensors to rank 1; tickets are auto-assigned in order
             futs = []
             for i in range(n):
-                ticket = mesh.next_send_ticket(1)
+                fut, ticket = mesh.send(torch.full((2, 3), float(i)), dst=1)
                 assert ticket == i
-                futs.append(mesh.send(torch.full((2, 3), float(i)), dst=1, ticket=ticket))
+                futs.append(fut)
             for f in futs:
                 f.result(timeout=30)
             # bidirectional: also receive one from rank 1 (no deadlock)
@@ -122,8 +122,8 @@ def _mesh_worker(rank, world, port, fail_q):
             handles = {}
             for i in reversed(range(n)):
                 handles[i] = mesh.post_recv(src=0, ticket=i, shape=(2, 3), dtype_str="f32")
-            t = mesh.next_send_ticket(0)
-            mesh.send(torch.arange(4.0), dst=0, ticket=t).result(timeout=30)
+            fut, _t = mesh.send(torch.arange(4.0), dst=0)
+            fut.result(timeout=30)
             for i in range(n):
                 got = handles[i].result(timeout=30)
                 assert torch.equal(got, torch.full((2, 3), float(i))), (i, got)
