@@ -32,6 +32,154 @@ def log(msg):
         print(msg, flush=True)
 
 
+def _emit_result(args, world, use_cuda, elapsed, B, stack):
+    ms_per_step = elapsed / args.steps * 1000
+    tokens_per_s = args.steps * B / elapsed
+    baseline = 6.0  # reference headline: Llama-2-70B, 6 tok/s on the public swarm
+    par = ("swarm-pp%d" % world if world > 1 else "swarm-single") if stack == "serve" else (
+        f"pp{world}" if world > 1 else "single")
+    result = {
+        "metric": "single-batch generate tokens/sec, Llama-2-70B across 1/2/4/8 MI355X servers",
+        "value": tokens_per_s,
+        "unit": "tokens/s",
+        "n_gpus": world,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": ms_per_step,
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": tokens_per_s / baseline if args.model == "llama-2-70b" else None,
+        "dtype": ("bf16" if args.quant == "none" else f"{args.quant}-weights/bf16-compute") if use_cuda else "fp32",
+        "data": "synthetic prompt, random-init weights (no network)",
+        "config": {
+            "model": args.model,
+            "global_batch": B,
+            "seq_len": args.prompt_len + args.warmup + args.steps,
+            "prompt_len": args.prompt_len,
+            "parallelism": par,
+            "quant": args.quant if use_cuda else "none",
+            "stack": stack,
+        },
+    }
+    print(json.dumps(result), flush=True)
+
+
+def run_serve_stack(args, rank, world, use_cuda, config):
+    """The REAL serving stack, rank-per-GPU: every rank runs a full Server
+    (DHT announce, handler RPCs, PriorityRuntime, MemoryCache, per-session
+    span hipGraphs) on its span; rank 0 additionally runs the thin client
+    (embeddings + LM head + sampling). Adjacent spans hand activations over
+    the RCCL/xGMI mesh (parallel/mesh.py); the co-located client<->rank0 hop
+    is in-process (p2p/transport.py InProcStream). This is what a user of
+    `petals_amd.cli.run_server` + AutoDistributedModelForCausalLM gets."""
+    import torch.distributed as dist
+
+    from petals_amd.dht.node import DHT
+    from petals_amd.parallel.pipeline import split_blocks
+    from petals_amd.server.server import Server
+
+    B = args.batch
+    device = torch.device("cuda", int(os.environ.get("LOCAL_RANK", 0))) if use_cuda else torch.device("cpu")
+    quant = args.quant if use_cuda else "none"
+
+    mesh = None
+    if world > 1:
+        from petals_amd.parallel.mesh import LocalMesh
+
+        # dedicated process group: mesh p2p must never share a communicator
+        # with the benchmark's own barriers/all-reduces
+        pg = dist.new_group(list(range(world)))
+        mesh = LocalMesh("bench-mesh", rank, world, device=device, group=pg)
+
+    boot = None
+    boot_addr = [None]
+    if rank == 0:
+        boot = DHT(host="127.0.0.1")
+        boot_addr = [list(boot.listen_addr)]
+    if world > 1:
+        dist.broadcast_object_list(boot_addr, src=0)
+    initial_peers = [tuple(boot_addr[0])]
+
+    spans = split_blocks(config.num_blocks, world)
+    my_span = spans[rank]
+    t0 = time.time()
+    server = Server(
+        args.model,
+        initial_peers=initial_peers,
+        host="127.0.0.1",
+        device=str(device),
+        torch_dtype="bfloat16" if use_cuda else "float32",
+        block_indices=f"{my_span.start}:{my_span.stop}",
+        dht_prefix="bench-serve",
+        throughput=1000.0,
+        quant_type=quant,
+        mesh=mesh,
+        update_period=30.0,
+    ).start()
+    log(f"[bench] serve stack: rank {rank} serving blocks {my_span.start}:{my_span.stop} "
+        f"in {time.time()-t0:.1f}s")
+
+    def barrier():
+        if world > 1:
+            dist.barrier()
+
+    barrier()  # every rank's server is up and announced before the client starts
+
+    elapsed = 0.0
+    if rank == 0:
+        from petals_amd.utils.auto_config import AutoDistributedModelForCausalLM
+
+        model = AutoDistributedModelForCausalLM.from_pretrained(
+            args.model, initial_peers=initial_peers, dht_prefix="bench-serve",
+            show_route=False, max_retries=2,
+        )
+        if use_cuda:
+            model = model.to(device=device, dtype=torch.bfloat16)
+        gen = torch.Generator().manual_seed(1234)
+        prompt = torch.randint(0, config.vocab_size, (B, args.prompt_len), generator=gen)
+        if use_cuda:
+            prompt = prompt.to(device)
+        max_len = args.prompt_len + args.warmup + args.steps + 8
+        with model.transformer.h.inference_session(max_length=max_len, batch_size=B) as sess, \
+                model.transformer.h.use_session(sess):
+            out = model.generate(prompt, max_new_tokens=max(args.warmup, 1), do_sample=False,
+                                 eos_token_id=-1)
+            barrier()
+            if use_cuda:
+                torch.cuda.synchronize(device)
+            t_start = time.perf_counter()
+            out = model.generate(out[:, -1:], max_new_tokens=args.steps, do_sample=False,
+                                 eos_token_id=-1)
+            if use_cuda:
+                torch.cuda.synchronize(device)
+            elapsed = time.perf_counter() - t_start
+            barrier()
+            assert out.shape[1] == args.steps + 1
+        model.transformer.h.sequence_manager.shutdown()
+    else:
+        barrier()  # start of timed region
+        barrier()  # end of timed region
+
+    # max over ranks (only rank 0 measured, but keep the collective contract)
+    if world > 1:
+        e = torch.tensor([elapsed], device=device if use_cuda else "cpu")
+        dist.all_reduce(e, op=dist.ReduceOp.MAX)
+        elapsed = float(e.item())
+
+    if rank == 0:
+        _emit_result(args, world, use_cuda, elapsed, B, "serve")
+
+    server.shutdown()
+    if mesh is not None:
+        mesh.shutdown()
+    if boot is not None:
+        boot.shutdown()
+    if world > 1:
+        dist.barrier()
+        dist.destroy_process_group()
+    os._exit(0)  # skip interpreter-teardown races between HIP and daemon threads
+
+
 def main():
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
@@ -42,6 +190,9 @@ def main():
     p.add_argument("--batch", type=int, default=1)
     p.add_argument("--device", default="cuda")
     p.add_argument("--quant", default="nf4", choices=["none", "nf4", "int8"], help="BASELINE config #3 names NF4 for the 70B pipeline; --quant none measures pure bf16")
+    p.add_argument("--stack", default="pipeline", choices=["pipeline", "serve"],
+                   help="pipeline: bare rank-per-GPU RCCL pipeline; serve: the REAL serving stack "
+                        "(DHT + Server + handler + client sessions) with the RCCL mesh hand-off")
     args = p.parse_args()
 
     import torch.distributed as dist
@@ -71,6 +222,10 @@ def main():
     H = config.hidden_size
     B = args.batch
     max_len = args.prompt_len + args.warmup + args.steps + 8
+
+    if args.stack == "serve":
+        run_serve_stack(args, rank, world, use_cuda, config)
+        return
 
     spans = split_blocks(config.num_blocks, world)
     my_span = spans[rank]
@@ -274,32 +429,7 @@ def main():
         elapsed = float(e.item())
 
     if rank == 0:
-        ms_per_step = elapsed / args.steps * 1000
-        tokens_per_s = args.steps * B / elapsed
-        baseline = 6.0  # reference headline: Llama-2-70B, 6 tok/s on the public swarm
-        result = {
-            "metric": "single-batch generate tokens/sec, Llama-2-70B across 1/2/4/8 MI355X servers",
-            "value": tokens_per_s,
-            "unit": "tokens/s",
-            "n_gpus": world,
-            "steps": args.steps,
-            "warmup": args.warmup,
-            "ms_per_step": ms_per_step,
-            "higher_is_better": True,
-            "scaling": "weak",
-            "vs_baseline": tokens_per_s / baseline if args.model == "llama-2-70b" else None,
-            "dtype": ("bf16" if args.quant == "none" else f"{args.quant}-weights/bf16-compute") if use_cuda else "fp32",
-            "data": "synthetic prompt, random-init weights (no network)",
-            "config": {
-                "model": args.model,
-                "global_batch": B,
-                "seq_len": args.prompt_len + args.warmup + args.steps,
-                "prompt_len": args.prompt_len,
-                "parallelism": f"pp{world}" if world > 1 else "single",
-                "quant": args.quant,
-            },
-        }
-        print(json.dumps(result), flush=True)
+        _emit_result(args, world, use_cuda, elapsed, B, "pipeline")
 
     if world > 1:
         dist.destroy_process_group()

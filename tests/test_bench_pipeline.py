@@ -40,6 +40,31 @@ def test_bench_single_process_cpu():
     _check_json_line(out.stdout, 1)
 
 
+def test_bench_serve_stack_single_cpu():
+    out = subprocess.run(
+        [sys.executable, "bench.py", "--gpus", "1", "--steps", "3", "--warmup", "1",
+         "--model", "test-llama", "--prompt-len", "8", "--device", "cpu", "--stack", "serve"],
+        cwd=REPO, capture_output=True, text=True, timeout=300,
+    )
+    assert out.returncode == 0, (out.stdout[-2000:], out.stderr[-2000:])
+    result = _check_json_line(out.stdout, 1)
+    assert result["config"]["stack"] == "serve"
+
+
+def test_bench_serve_stack_two_rank_cpu():
+    port = _free_port()
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1", "--nproc-per-node", "2",
+         "--master-addr", "127.0.0.1", "--master-port", str(port),
+         "bench.py", "--gpus", "2", "--steps", "3", "--warmup", "1",
+         "--model", "test-llama", "--prompt-len", "8", "--device", "cpu", "--stack", "serve"],
+        cwd=REPO, capture_output=True, text=True, timeout=420,
+    )
+    assert out.returncode == 0, (out.stdout[-2000:], out.stderr[-2000:])
+    result = _check_json_line(out.stdout, 2)
+    assert result["config"]["parallelism"] == "swarm-pp2"
+
+
 def test_bench_two_rank_pipeline_cpu():
     port = _free_port()
     out = subprocess.run(
