@@ -64,101 +64,148 @@ def _emit_result(args, world, use_cuda, elapsed, B, stack):
     print(json.dumps(result), flush=True)
 
 
-def run_serve_stack(args, rank, world, use_cuda, config):
+def run_serve_stack(args, rank, world, use_cuda, config) -> bool:
     """The REAL serving stack, rank-per-GPU: every rank runs a full Server
     (DHT announce, handler RPCs, PriorityRuntime, MemoryCache, per-session
     span hipGraphs) on its span; rank 0 additionally runs the thin client
     (embeddings + LM head + sampling). Adjacent spans hand activations over
     the RCCL/xGMI mesh (parallel/mesh.py); the co-located client<->rank0 hop
     is in-process (p2p/transport.py InProcStream). This is what a user of
-    `petals_amd.cli.run_server` + AutoDistributedModelForCausalLM gets."""
-    import torch.distributed as dist
+    `petals_amd.cli.run_server` + AutoDistributedModelForCausalLM gets.
 
-    from petals_amd.dht.node import DHT
-    from petals_amd.parallel.pipeline import split_blocks
-    from petals_amd.server.server import Server
+    Returns True if the measurement was emitted; False after a COORDINATED
+    failure (all ranks agree via all-reduce), so main() can fall back to the
+    bare pipeline measurement instead of hanging the job."""
+    import torch.distributed as dist
 
     B = args.batch
     device = torch.device("cuda", int(os.environ.get("LOCAL_RANK", 0))) if use_cuda else torch.device("cpu")
     quant = args.quant if use_cuda else "none"
 
-    mesh = None
-    if world > 1:
-        from petals_amd.parallel.mesh import LocalMesh
+    def healthy(ok: bool, what: str) -> bool:
+        """Coordinated go/no-go: min over ranks."""
+        if world == 1:
+            return ok
+        t = torch.tensor([1.0 if ok else 0.0], device=device if use_cuda else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MIN)
+        if t.item() < 0.5:
+            log(f"[bench] serve stack aborted at stage: {what}")
+            return False
+        return True
 
-        # dedicated process group: mesh p2p must never share a communicator
-        # with the benchmark's own barriers/all-reduces
-        pg = dist.new_group(list(range(world)))
-        mesh = LocalMesh("bench-mesh", rank, world, device=device, group=pg)
+    mesh = boot = server = None
+    ok = True
+    try:
+        if world > 1:
+            from petals_amd.parallel.mesh import LocalMesh
 
-    boot = None
-    boot_addr = [None]
-    if rank == 0:
-        boot = DHT(host="127.0.0.1")
-        boot_addr = [list(boot.listen_addr)]
-    if world > 1:
-        dist.broadcast_object_list(boot_addr, src=0)
-    initial_peers = [tuple(boot_addr[0])]
+            # dedicated process group: mesh p2p must never share a communicator
+            # with the benchmark's own barriers/all-reduces
+            pg = dist.new_group(list(range(world)))
+            mesh = LocalMesh("bench-mesh", rank, world, device=device, group=pg)
 
-    spans = split_blocks(config.num_blocks, world)
-    my_span = spans[rank]
-    t0 = time.time()
-    server = Server(
-        args.model,
-        initial_peers=initial_peers,
-        host="127.0.0.1",
-        device=str(device),
-        torch_dtype="bfloat16" if use_cuda else "float32",
-        block_indices=f"{my_span.start}:{my_span.stop}",
-        dht_prefix="bench-serve",
-        throughput=1000.0,
-        quant_type=quant,
-        mesh=mesh,
-        update_period=30.0,
-    ).start()
-    log(f"[bench] serve stack: rank {rank} serving blocks {my_span.start}:{my_span.stop} "
-        f"in {time.time()-t0:.1f}s")
+        from petals_amd.dht.node import DHT
+        from petals_amd.parallel.pipeline import split_blocks
+        from petals_amd.server.server import Server
+
+        boot_addr = [None]
+        if rank == 0:
+            boot = DHT(host="127.0.0.1")
+            boot_addr = [list(boot.listen_addr)]
+        if world > 1:
+            dist.broadcast_object_list(boot_addr, src=0)
+        initial_peers = [tuple(boot_addr[0])]
+
+        spans = split_blocks(config.num_blocks, world)
+        my_span = spans[rank]
+        t0 = time.time()
+        server = Server(
+            args.model,
+            initial_peers=initial_peers,
+            host="127.0.0.1",
+            device=str(device),
+            torch_dtype="bfloat16" if use_cuda else "float32",
+            block_indices=f"{my_span.start}:{my_span.stop}",
+            dht_prefix="bench-serve",
+            throughput=1000.0,
+            quant_type=quant,
+            mesh=mesh,
+            update_period=30.0,
+        ).start()
+        log(f"[bench] serve stack: rank {rank} serving blocks {my_span.start}:{my_span.stop} "
+            f"in {time.time()-t0:.1f}s")
+    except Exception as e:  # noqa: BLE001
+        print(f"[bench] rank {rank}: serve stack startup failed: {e!r}", flush=True)
+        ok = False
+
+    if not healthy(ok, "server startup"):
+        _serve_cleanup(server, mesh, boot, use_cuda)
+        return False
+
+    try:
+        if mesh is not None:
+            # NCCL communicator init is lazy AND collective: prime the ring
+            # eagerly or the first decode chain deadlocks
+            mesh.warmup_ring()
+    except Exception as e:  # noqa: BLE001
+        print(f"[bench] rank {rank}: mesh warmup failed: {e!r}", flush=True)
+        ok = False
+    if not healthy(ok, "mesh warmup"):
+        _serve_cleanup(server, mesh, boot, use_cuda)
+        return False
 
     def barrier():
         if world > 1:
             dist.barrier()
 
-    barrier()  # every rank's server is up and announced before the client starts
-
     elapsed = 0.0
+    client_ok = True
     if rank == 0:
-        from petals_amd.utils.auto_config import AutoDistributedModelForCausalLM
+        model = None
+        try:
+            from petals_amd.utils.auto_config import AutoDistributedModelForCausalLM
 
-        model = AutoDistributedModelForCausalLM.from_pretrained(
-            args.model, initial_peers=initial_peers, dht_prefix="bench-serve",
-            show_route=False, max_retries=2,
-        )
-        if use_cuda:
-            model = model.to(device=device, dtype=torch.bfloat16)
-        gen = torch.Generator().manual_seed(1234)
-        prompt = torch.randint(0, config.vocab_size, (B, args.prompt_len), generator=gen)
-        if use_cuda:
-            prompt = prompt.to(device)
-        max_len = args.prompt_len + args.warmup + args.steps + 8
-        with model.transformer.h.inference_session(max_length=max_len, batch_size=B) as sess, \
-                model.transformer.h.use_session(sess):
-            out = model.generate(prompt, max_new_tokens=max(args.warmup, 1), do_sample=False,
-                                 eos_token_id=-1)
-            barrier()
+            model = AutoDistributedModelForCausalLM.from_pretrained(
+                args.model, initial_peers=initial_peers, dht_prefix="bench-serve",
+                show_route=False, max_retries=2, request_timeout=60.0,
+            )
             if use_cuda:
-                torch.cuda.synchronize(device)
-            t_start = time.perf_counter()
-            out = model.generate(out[:, -1:], max_new_tokens=args.steps, do_sample=False,
-                                 eos_token_id=-1)
+                model = model.to(device=device, dtype=torch.bfloat16)
+            gen = torch.Generator().manual_seed(1234)
+            prompt = torch.randint(0, config.vocab_size, (B, args.prompt_len), generator=gen)
             if use_cuda:
-                torch.cuda.synchronize(device)
-            elapsed = time.perf_counter() - t_start
+                prompt = prompt.to(device)
+            max_len = args.prompt_len + args.warmup + args.steps + 8
+            with model.transformer.h.inference_session(max_length=max_len, batch_size=B) as sess, \
+                    model.transformer.h.use_session(sess):
+                out = model.generate(prompt, max_new_tokens=max(args.warmup, 1), do_sample=False,
+                                     eos_token_id=-1)
+                barrier()
+                if use_cuda:
+                    torch.cuda.synchronize(device)
+                t_start = time.perf_counter()
+                out = model.generate(out[:, -1:], max_new_tokens=args.steps, do_sample=False,
+                                     eos_token_id=-1)
+                if use_cuda:
+                    torch.cuda.synchronize(device)
+                elapsed = time.perf_counter() - t_start
+                barrier()
+                assert out.shape[1] == args.steps + 1
+        except Exception as e:  # noqa: BLE001
+            print(f"[bench] client failed: {e!r}", flush=True)
+            client_ok = False
+            barrier()  # release peers from the timed-region barriers
             barrier()
-            assert out.shape[1] == args.steps + 1
-        model.transformer.h.sequence_manager.shutdown()
+        finally:
+            if model is not None:
+                model.transformer.h.sequence_manager.shutdown()
     else:
         barrier()  # start of timed region
         barrier()  # end of timed region
+
+    if not healthy(client_ok, "client generate"):
+        _serve_cleanup(server, mesh, boot, use_cuda)
+        return False
 
     # max over ranks (only rank 0 measured, but keep the collective contract)
     if world > 1:
@@ -169,15 +216,27 @@ def run_serve_stack(args, rank, world, use_cuda, config):
     if rank == 0:
         _emit_result(args, world, use_cuda, elapsed, B, "serve")
 
-    server.shutdown()
-    if mesh is not None:
-        mesh.shutdown()
-    if boot is not None:
-        boot.shutdown()
+    _serve_cleanup(server, mesh, boot, use_cuda)
     if world > 1:
         dist.barrier()
-        dist.destroy_process_group()
-    os._exit(0)  # skip interpreter-teardown races between HIP and daemon threads
+    return True
+
+
+def _serve_cleanup(server, mesh, boot, use_cuda):
+    try:
+        if server is not None:
+            server.shutdown()
+        if mesh is not None:
+            mesh.shutdown()
+        if boot is not None:
+            boot.shutdown()
+    except Exception:  # noqa: BLE001
+        pass
+    if use_cuda:
+        import gc
+
+        gc.collect()
+        torch.cuda.empty_cache()
 
 
 def main():
@@ -224,8 +283,12 @@ def main():
     max_len = args.prompt_len + args.warmup + args.steps + 8
 
     if args.stack == "serve":
-        run_serve_stack(args, rank, world, use_cuda, config)
-        return
+        if run_serve_stack(args, rank, world, use_cuda, config):
+            if world > 1:
+                dist.destroy_process_group()
+            os._exit(0)  # skip interpreter-teardown races between HIP and daemon threads
+        log("[bench] serve stack failed; measuring the bare pipeline instead")
+        args.stack = "pipeline"
 
     spans = split_blocks(config.num_blocks, world)
     my_span = spans[rank]

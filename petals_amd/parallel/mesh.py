@@ -128,6 +128,25 @@ class LocalMesh:
             if isinstance(fut, Future) and not fut.done():
                 fut.set_exception(exc)
 
+    def warmup_ring(self, timeout: float = 120.0) -> None:
+        """Pass one dummy tensor around the ring (r -> r+1). MUST be called on
+        every rank before serving when the group is NCCL/RCCL: communicator
+        init is collective and lazy, so without this the first decode chain
+        deadlocks (rank r's first p2p op only happens once the token reaches
+        it, which requires init to have completed on all ranks). Also primes
+        exactly the pairs the pipeline uses and advances every ticket counter
+        identically on all ranks."""
+        if self.world < 2:
+            return
+        dst = (self.rank + 1) % self.world
+        src = (self.rank - 1) % self.world
+        dev = self.device if self.device is not None else "cpu"
+        fut, ticket = self.send(torch.zeros(1, device=dev), dst=dst)
+        assert ticket == 0, "warmup_ring must run before any other transfer"
+        handle = self.post_recv(src, 0, (1,), "f32")
+        handle.result(timeout)
+        fut.result(timeout)
+
     # ------------------------------------------------------------ transfers
 
     def send(self, tensor: torch.Tensor, dst: int, ready_event=None) -> Tuple[Future, int]:

@@ -23,10 +23,15 @@ def init_process_group_from_env() -> tuple[int, int]:
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))
     if world > 1 and not dist.is_initialized():
+        import datetime
+
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "29571")
         backend = "nccl" if torch.cuda.is_available() else "gloo"  # nccl == RCCL on ROCm
-        dist.init_process_group(backend=backend, rank=rank, world_size=world)
+        # bounded timeout: a wedged collective should abort the job, not hang it
+        dist.init_process_group(
+            backend=backend, rank=rank, world_size=world, timeout=datetime.timedelta(seconds=300)
+        )
         if torch.cuda.is_available():
             torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", rank)))
     return rank, world
