@@ -42,6 +42,19 @@ class MeshError(Exception):
     pass
 
 
+class _SendEntry:
+    """One (possibly pre-announced) outbound transfer."""
+
+    __slots__ = ("ticket", "tensor", "ready_event", "fut", "committed")
+
+    def __init__(self):
+        self.ticket = -1
+        self.tensor: Optional[torch.Tensor] = None
+        self.ready_event = None
+        self.fut: Future = Future()
+        self.committed = False
+
+
 class MeshRecvHandle:
     """A pending inbound transfer; resolves to a device tensor once the RCCL
     recv has completed (and its stream has been synchronized)."""
@@ -123,10 +136,10 @@ class LocalMesh:
         for h in pending:
             if not h.future.done():
                 h.future.set_exception(exc)
-        for item in items:
-            fut = item[-1]
-            if isinstance(fut, Future) and not fut.done():
-                fut.set_exception(exc)
+        for item in items:  # ("send", dst, _SendEntry)
+            entry = item[-1]
+            if isinstance(entry, _SendEntry) and not entry.fut.done():
+                entry.fut.set_exception(exc)
 
     def warmup_ring(self, timeout: float = 120.0) -> None:
         """Pass one dummy tensor around the ring (r -> r+1). MUST be called on
@@ -153,15 +166,44 @@ class LocalMesh:
         """Enqueue a send; the per-dst ticket is assigned atomically at enqueue
         time (so ticket order == issue order, with no holes). Returns
         (future resolved when the transfer is on the wire, ticket)."""
-        fut: Future = Future()
+        ticket, commit, _abort = self.send_deferred(dst)
+        fut = commit(tensor, ready_event)
+        return fut, ticket
+
+    def send_deferred(self, dst: int):
+        """Reserve the next ticket to `dst` NOW and supply the tensor later
+        ("pre-announce"): the caller can ship the ticket to the receiver over
+        TCP before the producing compute has even started, so the control
+        plane runs ahead of the data plane. The comm thread holds back all
+        later sends to the same dst until this entry is committed (ticket
+        order is preserved).
+
+        Returns (ticket, commit(tensor, ready_event=None) -> Future,
+        abort(reason)). An aborted entry breaks the mesh (the receiver is
+        already expecting this ticket; silently skipping it would desync the
+        pair) — the swarm falls back to TCP."""
+        entry = _SendEntry()
         with self._cv:
             if self._broken:
                 raise MeshError(f"mesh broken: {self._broken}")
             ticket = self._send_tickets[dst]
             self._send_tickets[dst] = ticket + 1
-            self._queue.append(("send", dst, ticket, tensor, ready_event, fut))
+            entry.ticket = ticket
+            self._queue.append(("send", dst, entry))
             self._cv.notify_all()
-        return fut, ticket
+
+        def commit(tensor: torch.Tensor, ready_event=None) -> Future:
+            with self._cv:
+                entry.tensor = tensor
+                entry.ready_event = ready_event
+                entry.committed = True
+                self._cv.notify_all()
+            return entry.fut
+
+        def abort(reason: str) -> None:
+            self.mark_broken(f"deferred send to {dst} (ticket {entry.ticket}) aborted: {reason}")
+
+        return ticket, commit, abort
 
     def post_recv(self, src: int, ticket: int, shape, dtype_str: str) -> MeshRecvHandle:
         """Register an expected inbound transfer (called when the TCP meta for
@@ -206,29 +248,42 @@ class LocalMesh:
 
             self._waiters = ThreadPoolExecutor(max_workers=32, thread_name_prefix=f"mesh-wait-{self.mesh_id}")
         outstanding: List[tuple] = []  # (kind, work, future, buf_or_None)
+        send_q: Dict[int, collections.deque] = collections.defaultdict(collections.deque)
         while True:
             with self._cv:
-                if not self._queue and not self._any_ready_locked() and not outstanding and not self._shutdown:
+                if (
+                    not self._queue
+                    and not self._any_ready_locked()
+                    and not outstanding
+                    and not self._any_committed_locked(send_q)
+                    and not self._shutdown
+                ):
                     self._cv.wait(timeout=0.5)
                 if self._shutdown:
                     return
+                while self._queue:  # entries arrive in ticket order per dst
+                    _, dst, entry = self._queue.popleft()
+                    send_q[dst].append(entry)
                 sends = []
-                while self._queue:
-                    sends.append(self._queue.popleft())
+                for dst, q in send_q.items():
+                    # strict ticket order per pair: stop at the first
+                    # uncommitted (pre-announced, compute in flight) entry
+                    while q and q[0].committed:
+                        sends.append((dst, q.popleft()))
                 recvs = self._ready_recvs()
             try:
                 use_stream = self._stream is not None
                 ctx = torch.cuda.stream(self._stream) if use_stream else _nullcontext()
                 with ctx:
-                    for _, dst, _ticket, tensor, ready_event, fut in sends:
-                        if ready_event is not None and use_stream:
-                            self._stream.wait_event(ready_event)
-                        t = tensor if tensor.is_contiguous() else tensor.contiguous()
+                    for dst, entry in sends:
+                        if entry.ready_event is not None and use_stream:
+                            self._stream.wait_event(entry.ready_event)
+                        t = entry.tensor if entry.tensor.is_contiguous() else entry.tensor.contiguous()
                         w = dist.isend(t, dst=dst, group=self.group)
                         if use_cuda:
-                            outstanding.append(("send", w, fut, t))  # keep t alive until done
+                            outstanding.append(("send", w, entry.fut, t))  # keep t alive until done
                         else:
-                            self._waiters.submit(_wait_and_resolve, w, fut, True, (t,))
+                            self._waiters.submit(_wait_and_resolve, w, entry.fut, True, (t,))
                     for src, _ticket, handle in recvs:
                         buf = torch.empty(
                             handle.shape,
@@ -255,10 +310,9 @@ class LocalMesh:
                     time.sleep(20e-6)
             except Exception as e:  # noqa: BLE001
                 logger.exception("mesh comm thread failed")
-                for item in sends:
-                    fut = item[-1]
-                    if not fut.done():
-                        fut.set_exception(e)
+                for _dst, entry in sends:
+                    if not entry.fut.done():
+                        entry.fut.set_exception(e)
                 for _, _, h in recvs:
                     if not h.future.done():
                         h.future.set_exception(e)
@@ -270,6 +324,10 @@ class LocalMesh:
 
     def _any_ready_locked(self) -> bool:
         return any(self._recv_next[src] in pending for src, pending in self._recv_pending.items())
+
+    @staticmethod
+    def _any_committed_locked(send_q) -> bool:
+        return any(q and q[0].committed for q in send_q.values())
 
 
 # Process-local mesh registry: the thin client (and bench harnesses) in a

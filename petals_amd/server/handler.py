@@ -78,6 +78,26 @@ class _SpanDecodeGraph:
         return self.graph.replay()
 
 
+def _dtype_str(dtype) -> str:
+    from petals_amd.utils.serialization import _DTYPE_TO_STR
+
+    return _DTYPE_TO_STR[dtype]
+
+
+def _commit_if_needed(commit_cb, out: torch.Tensor) -> None:
+    """Runs IN the runtime thread: hand the produced activation to a
+    pre-announced mesh transfer. The CUDA event lets the comm thread isend as
+    soon as the producing stream reaches this point — before the runtime's
+    end-of-task synchronize."""
+    if commit_cb is None:
+        return
+    evt = None
+    if out.is_cuda:
+        evt = torch.cuda.Event()
+        evt.record()
+    commit_cb(out, evt)
+
+
 class TransformerConnectionHandler:
     def __init__(
         self,
@@ -302,11 +322,17 @@ class TransformerConnectionHandler:
         active_adapter: Optional[str],
         session: Optional[_Session] = None,
         keep_on_device: bool = False,
+        commit_cb=None,
     ) -> torch.Tensor:
         """Runs IN the runtime thread: one inference step through the whole span
         (the single-process analog of reference _MergedInferenceStep)."""
         backend0 = self.backends[uids[0]]
         device, dtype = backend0.device, backend0.dtype
+        if not torch.is_tensor(hidden_states):
+            # MeshRecvHandle: the step was submitted BEFORE its input landed;
+            # block here (the actual data dependency) instead of in the
+            # asyncio loop so launch work overlaps the upstream hop's compute
+            hidden_states = hidden_states.result(self.step_timeout)
         hidden_states = hidden_states.to(device=device, dtype=dtype)
         has_hypo = hypo_ids is not None and not is_dummy(hypo_ids)
         has_prompts = prompts is not None and not is_dummy(prompts)
@@ -355,10 +381,13 @@ class TransformerConnectionHandler:
                 # clone: the graph's static output is overwritten by the next
                 # replay; anything leaving the runtime must own its storage
                 res = out.clone() if keep_on_device else out.cpu()
+                _commit_if_needed(commit_cb, res)
                 print(f"[rt] graph {( _g1-_g0)*1e3:.2f} out {(time.perf_counter()-_g1)*1e3:.2f} ms", flush=True)
                 return res
             out = session.span_graph.step(hidden_states, prefix_length)
-            return out.clone() if keep_on_device else out.cpu()
+            out = out.clone() if keep_on_device else out.cpu()
+            _commit_if_needed(commit_cb, out)
+            return out
 
         if has_hypo:
             hypo_ids = hypo_ids.to(device)
@@ -372,7 +401,9 @@ class TransformerConnectionHandler:
                     hidden_states[:, : prompt.shape[1]] += prompt
                 info = InferenceMetadata(uid, prefix_length, tuple(handle_pair), active_adapter)
                 (hidden_states,) = self.backends[uid].inference_step(hidden_states, hypo_ids, info)
-        return hidden_states if keep_on_device else hidden_states.cpu()
+        hidden_states = hidden_states if keep_on_device else hidden_states.cpu()
+        _commit_if_needed(commit_cb, hidden_states)
+        return hidden_states
 
     async def rpc_inference(self, request: RpcMessage, stream: RpcStream) -> None:
         meta = request.meta
@@ -406,13 +437,15 @@ class TransformerConnectionHandler:
                             raise RpcError("start_from_position is ahead of the cache")
                         session.prefix_length = pos
                     hidden_states = tensors[0]
-                    if not torch.is_tensor(hidden_states):  # MeshRecvHandle: inputs arrive over RCCL
-                        hidden_states = await asyncio.wait_for(
-                            asyncio.wrap_future(hidden_states.future), self.step_timeout
-                        )
                     prompts = tensors[1] if len(tensors) > 1 else None
                     hypo_ids = tensors[2] if len(tensors) > 2 else None
-                    length_increment = hidden_states.shape[1] if hidden_states.numel() > 0 else 0
+                    if torch.is_tensor(hidden_states):
+                        length_increment = hidden_states.shape[1] if hidden_states.numel() > 0 else 0
+                    else:  # MeshRecvHandle: inputs arrive over RCCL; the shape
+                        # is known from the pre-announce, so the step can be
+                        # queued before the data lands (the runtime thread
+                        # blocks on the actual dependency)
+                        length_increment = hidden_states.shape[1]
                     if session.prefix_length + length_increment > max_length:
                         raise RpcError(
                             f"max_length exceeded: prefix {session.prefix_length} + {length_increment} > {max_length}"
@@ -432,44 +465,119 @@ class TransformerConnectionHandler:
                         or (not will_push and self._mesh_out_dst(output_via_mesh) is not None)
                         or getattr(stream, "is_inproc", False)
                     )
+
+                    # --- pre-announce: for mesh hand-offs, reserve the ticket
+                    # and ship the (tiny) TCP metadata BEFORE computing, so the
+                    # control plane of hop r+1 runs while hop r's GPU works;
+                    # the activation is committed to the wire by the runtime
+                    # thread the moment it exists (ticket order keeps pairs in
+                    # sync even though metas may arrive out of order)
+                    commit_cb = abort_cb = None
+                    pre_kind = None
+                    ack_task = None
+                    if length_increment > 0:
+                        out_shape = list(hidden_states.shape)  # blocks preserve shape
+                        out_dtype = self.backends[uids[0]].dtype
+                        if will_push:
+                            mdst = self._mesh_dst(next_servers[0])
+                            if mdst is not None:
+                                try:
+                                    ticket, commit_cb, abort_cb = self.mesh.send_deferred(mdst)
+                                    push_meta = {
+                                        "session_id": next_servers[0][2],
+                                        "step_id": step_meta.get("step_id"),
+                                        "next_servers": next_servers[1:],
+                                        "start_from_position": step_start_position,
+                                        "output_via_mesh": output_via_mesh,
+                                        "tensors_via_mesh": {
+                                            "mesh_id": self.mesh.mesh_id,
+                                            "src_rank": self.mesh.rank,
+                                            "ticket": ticket,
+                                            "shape": out_shape,
+                                            "dtype": _dtype_str(out_dtype),
+                                        },
+                                    }
+                                    host, port = next_servers[0][0], next_servers[0][1]
+                                    ack_task = asyncio.ensure_future(
+                                        asyncio.wait_for(
+                                            self.p2p.call_unary(
+                                                (host, port), "petals.rpc_push", RpcMessage(meta=push_meta)
+                                            ),
+                                            timeout=10.0,
+                                        )
+                                    )
+                                    pre_kind = "push"
+                                except Exception as e:  # noqa: BLE001
+                                    logger.warning("mesh pre-announce failed (%r); TCP push", e)
+                                    commit_cb = abort_cb = None
+                        elif not has_prompts:
+                            modst = self._mesh_out_dst(output_via_mesh)
+                            if modst is not None:
+                                try:
+                                    ticket, commit_cb, abort_cb = self.mesh.send_deferred(modst)
+                                    await stream.send(RpcMessage(meta={
+                                        "step_id": step_meta.get("step_id"),
+                                        "tensors_via_mesh": self._mesh_desc_shaped(out_shape, out_dtype, ticket),
+                                    }))
+                                    pre_kind = "out"
+                                except Exception as e:  # noqa: BLE001
+                                    logger.warning("mesh out pre-announce failed (%r); stream", e)
+                                    commit_cb = abort_cb = None
+
                     if length_increment > 0:
                         priority = self.prioritizer.prioritize(hidden_states, type="inference")
-                        output = await self.runtime.submit(
-                            priority,
-                            self._inference_step_chain,
-                            uids,
-                            hidden_states,
-                            hypo_ids,
-                            prompts,
-                            handles,
-                            session.prefix_length,
-                            active_adapter,
-                            session,
-                            keep_on_device,
-                        )
+                        try:
+                            output = await self.runtime.submit(
+                                priority,
+                                self._inference_step_chain,
+                                uids,
+                                hidden_states,
+                                hypo_ids,
+                                prompts,
+                                handles,
+                                session.prefix_length,
+                                active_adapter,
+                                session,
+                                keep_on_device,
+                                commit_cb,
+                            )
+                        except BaseException as e:
+                            if abort_cb is not None:
+                                # the peer already expects this ticket: a
+                                # silent skip would desync the pair
+                                abort_cb(f"compute failed: {e!r}")
+                            if ack_task is not None:
+                                ack_task.cancel()
+                                with contextlib.suppress(Exception):
+                                    await ack_task
+                            raise
                     else:
                         output = hidden_states
                     if _trace:
                         print(f"[srv] compute {(time.perf_counter()-_t0)*1e3:.2f} ms", flush=True)
                     session.prefix_length += length_increment
 
+                    if pre_kind == "out":
+                        continue  # meta already on the stream; data on the mesh
+
                     pushed = False
-                    if will_push:
+                    if pre_kind == "push":
+                        try:
+                            await ack_task
+                            pushed = True
+                        except Exception as e:  # noqa: BLE001
+                            logger.warning("pre-announced push not acknowledged (%r)", e)
+                            if self.mesh is not None:
+                                # data already committed to the wire but the peer
+                                # never learned the ticket: the pair is desynced
+                                self.mesh.mark_broken(f"push ack failed: {e!r}")
+                    elif will_push:
                         pushed = await self._push_outputs(output, step_meta, next_servers, step_start_position)
                     if not pushed:
                         # the last server of a push chain (or any server when push
                         # is off/failed) returns outputs on its client stream
                         out_meta = {"step_id": step_meta.get("step_id")}
-                        mesh_dst = self._mesh_out_dst(output_via_mesh)
-                        if mesh_dst is not None and torch.is_tensor(output) and output.device.type == self.mesh.device.type:
-                            try:
-                                fut, ticket = self.mesh.send(output, dst=mesh_dst)
-                                out_meta["tensors_via_mesh"] = self._mesh_desc(output, ticket)
-                                await stream.send(RpcMessage(meta=out_meta))
-                                continue
-                            except Exception as e:  # noqa: BLE001
-                                logger.warning("mesh output delivery failed (%r); using the stream", e)
-                        if output.device.type != "cpu" and not getattr(stream, "is_inproc", False):
+                        if torch.is_tensor(output) and output.device.type != "cpu" and not getattr(stream, "is_inproc", False):
                             output = output.cpu()
                         await stream.send(RpcMessage(meta=out_meta, tensors=[output]))
         finally:
@@ -532,14 +640,15 @@ class TransformerConnectionHandler:
         return int(rank)
 
     def _mesh_desc(self, tensor: torch.Tensor, ticket: int) -> Dict[str, Any]:
-        from petals_amd.utils.serialization import _DTYPE_TO_STR
+        return self._mesh_desc_shaped(list(tensor.shape), tensor.dtype, ticket)
 
+    def _mesh_desc_shaped(self, shape, dtype, ticket: int) -> Dict[str, Any]:
         return {
             "mesh_id": self.mesh.mesh_id,
             "src_rank": self.mesh.rank,
             "ticket": ticket,
-            "shape": list(tensor.shape),
-            "dtype": _DTYPE_TO_STR[tensor.dtype],
+            "shape": list(shape),
+            "dtype": _dtype_str(dtype),
         }
 
     async def rpc_push(self, request: RpcMessage, stream: RpcStream) -> None:
