@@ -249,9 +249,10 @@ def main():
     p.add_argument("--batch", type=int, default=1)
     p.add_argument("--device", default="cuda")
     p.add_argument("--quant", default="nf4", choices=["none", "nf4", "int8"], help="BASELINE config #3 names NF4 for the 70B pipeline; --quant none measures pure bf16")
-    p.add_argument("--stack", default="pipeline", choices=["pipeline", "serve"],
-                   help="pipeline: bare rank-per-GPU RCCL pipeline; serve: the REAL serving stack "
-                        "(DHT + Server + handler + client sessions) with the RCCL mesh hand-off")
+    p.add_argument("--stack", default="serve", choices=["pipeline", "serve"],
+                   help="serve (default): the REAL serving stack (DHT + Server + handler + client "
+                        "sessions) with the RCCL mesh hand-off — what a user gets; pipeline: bare "
+                        "rank-per-GPU RCCL pipeline (kernel-harness upper bound)")
     args = p.parse_args()
 
     import torch.distributed as dist

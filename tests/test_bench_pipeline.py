@@ -71,7 +71,7 @@ def test_bench_two_rank_pipeline_cpu():
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1", "--nproc-per-node", "2",
          "--master-addr", "127.0.0.1", "--master-port", str(port),
          "bench.py", "--gpus", "2", "--steps", "3", "--warmup", "1",
-         "--model", "test-llama", "--prompt-len", "8", "--device", "cpu"],
+         "--model", "test-llama", "--prompt-len", "8", "--device", "cpu", "--stack", "pipeline"],
         cwd=REPO, capture_output=True, text=True, timeout=300,
     )
     assert out.returncode == 0, out.stderr[-2000:]
