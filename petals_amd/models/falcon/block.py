@@ -110,11 +110,20 @@ class FalconBlock(nn.Module):
 
     _fast = None  # FalconFastPath after optimize_for_inference()
 
+    # (head_dim, gq) combos instantiated in ops/csrc/attention.hip ATTN_CASE /
+    # ATTN_CASE_BIG — gq > 16 runs the multi-group MFMA decode (ceil(gq/16)
+    # A-fragment groups per block, K/V re-read per extra group)
+    _FUSED_DECODE_GEOMS = {
+        (128, 1), (128, 2), (128, 4), (128, 6), (128, 8), (128, 16), (128, 32),
+        (64, 1), (64, 2), (64, 4), (64, 8), (64, 16), (64, 29), (64, 32), (64, 71),
+    }
+
     def optimize_for_inference(self, quant: str = "none") -> "FalconBlock":
         """Repack weights into the MI355X kernel layout (parallel attn+MLP,
-        rope GQA fused decode chain); frees nn.Linear weights. Only the
-        new-decoder architecture (40B/180B) with a supported (head_dim, gqa)
-        geometry takes the fused path; other variants serve via generic ops."""
+        rope GQA fused decode chain); frees nn.Linear weights. Covers the
+        new-decoder architecture (40B/180B GQA) and the 7B-style old decoder
+        (MQA + parallel_attn, single layernorm); other variants (e.g. biased
+        falcon-rw) serve via generic ops."""
         from petals_amd import ops as _ops
         from petals_amd.ops.fused_decode import FalconFastPath
 
@@ -127,17 +136,16 @@ class FalconBlock(nn.Module):
         cfg = self.config
         gq = cfg.num_attention_heads // cfg.n_kv_heads
         if (
-            not cfg.new_decoder_architecture
-            or cfg.head_dim not in (64, 128)
-            or gq not in (1, 2, 4, 6, 8, 16)
+            (cfg.head_dim, gq) not in self._FUSED_DECODE_GEOMS
+            or (not cfg.new_decoder_architecture and not cfg.parallel_attn)
             or cfg.bias
         ):
             import logging
 
             logging.getLogger(__name__).warning(
-                "Falcon geometry (new_decoder=%s, head_dim=%s, gq=%s, bias=%s) outside the fused "
-                "fast path; serving via generic HIP ops",
-                cfg.new_decoder_architecture, cfg.head_dim, gq, cfg.bias,
+                "Falcon geometry (new_decoder=%s, parallel_attn=%s, head_dim=%s, gq=%s, bias=%s) "
+                "outside the fused fast path; serving via generic HIP ops",
+                cfg.new_decoder_architecture, cfg.parallel_attn, cfg.head_dim, gq, cfg.bias,
             )
             return self
         self._fast = FalconFastPath(self, hip, quant=quant)

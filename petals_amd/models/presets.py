@@ -117,6 +117,17 @@ PRESETS = {
         num_kv_heads=2, vocab_size=128, max_position_embeddings=256,
         layer_norm_epsilon=1e-5, parallel_attn=True, new_decoder_architecture=True, bias=False,
     ),
+    "test-falcon-gq29": dict(  # falcon-180b head geometry (gq=29, head_dim 64) for GPU tests
+        model_type="falcon", hidden_size=1856, num_hidden_layers=2, num_attention_heads=29,
+        num_kv_heads=1, vocab_size=128, max_position_embeddings=256,
+        layer_norm_epsilon=1e-5, parallel_attn=True, new_decoder_architecture=True, bias=False,
+    ),
+    "test-falcon-mqa71": dict(  # falcon-7b geometry (old decoder, MQA gq=71) for GPU tests
+        model_type="falcon", hidden_size=4544, num_hidden_layers=2, num_attention_heads=71,
+        vocab_size=128, max_position_embeddings=256,
+        layer_norm_epsilon=1e-5, parallel_attn=True, new_decoder_architecture=False,
+        multi_query=True, bias=False,
+    ),
     "test-mixtral": dict(
         model_type="mixtral", hidden_size=64, num_hidden_layers=4, num_attention_heads=4,
         num_key_value_heads=2, intermediate_size=128, vocab_size=128,

@@ -34,6 +34,12 @@ using f32x4 = __attribute__((ext_vector_type(4))) float;
 #define DVT_BYTE(dim, key_byte) \
   ((((unsigned)(dim)) * ((DKVT + DKPAD) * 2) + (unsigned)(key_byte)) ^ ((((unsigned)(dim) >> 3) & 7u) << 4))
 
+// Arbitrary GQ: heads are processed in ceil(GQ/16) groups of <=16 (one
+// A-fragment each). The group loop sits OUTSIDE the kv loop so live registers
+// stay at one group's footprint; K/V re-reads per extra group come from
+// L2/HBM and are negligible for the weight-bound decode shapes that need
+// GQ > 16 (falcon-7b MQA gq=71, falcon-180b gq=29 — reference
+// models/falcon/block.py:100-110).
 template <int HD, int GQ>
 __global__ __launch_bounds__(DWAVES * 64) void mfma_decode_kernel(
     const float* __restrict__ q,            // [B, KV, GQ, HD]
@@ -63,171 +69,179 @@ __global__ __launch_bounds__(DWAVES * 64) void mfma_decode_kernel(
   const int hi = lane >> 4;
 
   constexpr int KCH = HD / 32;
+  constexpr int GA = (GQ + 15) / 16;  // head groups of <=16
 
   // per-wave LDS: swizzled V^T tile + P scratch; the cross-wave combine
-  // overlays the V^T storage after the main loop
+  // overlays the V^T storage after the main loop (one <=16-head group at a
+  // time, so the overlay always fits)
   __shared__ unsigned char vt_raw[DWAVES][HD * (DKVT + DKPAD) * 2];
   __shared__ unsigned short p_lds[DWAVES][16][DKVT + DKPAD];
-  __shared__ float c_ml[DWAVES][GQ][2];
-  static_assert(GQ * HD * 4 <= HD * (DKVT + DKPAD) * 2, "combine O overlay too big");
-
-  // ---- q tile: A-fragment rows = q heads (zero-padded to 16), pre-scaled
-  bf16x8 q_frag[KCH];
-  const size_t q_base = (size_t)bkv * GQ * HD;
-#pragma unroll
-  for (int kc = 0; kc < KCH; ++kc) {
-    if (col < GQ) {
-      const float* src = q + q_base + (size_t)col * HD + kc * 32 + hi * 8;
-      short v[8];
-#pragma unroll
-      for (int e = 0; e < 8; ++e) v[e] = (short)f32_to_bf16(src[e] * scale);
-      q_frag[kc] = bf16x8{v[0], v[1], v[2], v[3], v[4], v[5], v[6], v[7]};
-    } else {
-      q_frag[kc] = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
-    }
-  }
-
-  // per-lane-row ALiBi slopes (rows = heads hi*4 + r)
-  float sl[4];
-#pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    const int g = hi * 4 + r;
-    sl[r] = (alibi && g < GQ) ? alibi[kvh * GQ + g] : 0.f;
-  }
-
-  f32x4 acc_o[HD / 16];
-#pragma unroll
-  for (int d = 0; d < HD / 16; ++d) acc_o[d] = f32x4{0.f, 0.f, 0.f, 0.f};
-  float m_row[4], l_row[4];
-#pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    m_row[r] = NEG_SENTINEL;
-    l_row[r] = 0.f;
-  }
+  __shared__ float c_ml[DWAVES][16][2];
+  static_assert(16 * HD * 4 <= HD * (DKVT + DKPAD) * 2, "combine O overlay too big");
 
   const size_t kv_base = (size_t)bkv * lmax * HD;
   const unsigned short* kb = k_cache + kv_base;
   const unsigned short* vb = v_cache + kv_base;
+  const size_t q_base = (size_t)bkv * GQ * HD;
 
-  // wave w handles tiles j_begin + (t*DWAVES + w)*DKVT
-  for (int j0 = j_begin + wave * DKVT; j0 < j_end; j0 += DWAVES * DKVT) {
-    const int tile_n = min(DKVT, j_end - j0);
-    const bool full = tile_n == DKVT;
+  for (int a = 0; a < GA; ++a) {
+    const int g0 = a * 16;  // first head of this group
+    if (a > 0) __syncthreads();  // wave0's combine reads of vt_raw are done
 
-    // ---- stage V^T (this wave only: no block barrier; the waitcnt orders
-    // this wave's LDS writes before its reads)
-    for (int idx = lane; idx < DKVT * (HD / 8); idx += WAVE) {
-      const int row = idx / (HD / 8);
-      const int c8 = (idx - row * (HD / 8)) * 8;
-      bf16x8 vv8 = {0, 0, 0, 0, 0, 0, 0, 0};
-      if (j0 + row < j_end) vv8 = *reinterpret_cast<const bf16x8*>(vb + (size_t)(j0 + row) * HD + c8);
+    // ---- q tile: A-fragment rows = q heads of this group (zero-padded), pre-scaled
+    bf16x8 q_frag[KCH];
 #pragma unroll
-      for (int e = 0; e < 8; ++e)
-        *reinterpret_cast<unsigned short*>(&vt_raw[wave][DVT_BYTE(c8 + e, row * 2)]) = (unsigned short)vv8[e];
-    }
-    __builtin_amdgcn_s_waitcnt(0);
-
-    // ---- S = Q K^T, K direct from global: B[k = kdim][n = key]
-    f32x4 s_acc[DKVT / 16];
+    for (int kc = 0; kc < KCH; ++kc) {
+      if (g0 + col < GQ) {
+        const float* src = q + q_base + (size_t)(g0 + col) * HD + kc * 32 + hi * 8;
+        short v[8];
 #pragma unroll
-    for (int nb = 0; nb < DKVT / 16; ++nb) {
-      s_acc[nb] = f32x4{0.f, 0.f, 0.f, 0.f};
-      const int key = j0 + nb * 16 + col;
-      const unsigned short* krow = kb + (size_t)min(key, j_end - 1) * HD + hi * 8;
-#pragma unroll
-      for (int kc = 0; kc < KCH; ++kc) {
-        const bf16x8 kt = *reinterpret_cast<const bf16x8*>(krow + kc * 32);
-        s_acc[nb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(q_frag[kc], kt, s_acc[nb], 0, 0, 0);
+        for (int e = 0; e < 8; ++e) v[e] = (short)f32_to_bf16(src[e] * scale);
+        q_frag[kc] = bf16x8{v[0], v[1], v[2], v[3], v[4], v[5], v[6], v[7]};
+      } else {
+        q_frag[kc] = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
       }
     }
 
-    // ---- online softmax over this tile (rows = heads hi*4+r)
-    float p[DKVT / 16][4];
-    float corr[4];
+    // per-lane-row ALiBi slopes (rows = heads g0 + hi*4 + r)
+    float sl[4];
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      float s[DKVT / 16];
+      const int g = g0 + hi * 4 + r;
+      sl[r] = (alibi && g < GQ) ? alibi[kvh * GQ + g] : 0.f;
+    }
+
+    f32x4 acc_o[HD / 16];
+#pragma unroll
+    for (int d = 0; d < HD / 16; ++d) acc_o[d] = f32x4{0.f, 0.f, 0.f, 0.f};
+    float m_row[4], l_row[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      m_row[r] = NEG_SENTINEL;
+      l_row[r] = 0.f;
+    }
+
+    // wave w handles tiles j_begin + (t*DWAVES + w)*DKVT
+    for (int j0 = j_begin + wave * DKVT; j0 < j_end; j0 += DWAVES * DKVT) {
+      const int tile_n = min(DKVT, j_end - j0);
+      const bool full = tile_n == DKVT;
+
+      // ---- stage V^T (this wave only: no block barrier; the waitcnt orders
+      // this wave's LDS writes before its reads)
+      for (int idx = lane; idx < DKVT * (HD / 8); idx += WAVE) {
+        const int row = idx / (HD / 8);
+        const int c8 = (idx - row * (HD / 8)) * 8;
+        bf16x8 vv8 = {0, 0, 0, 0, 0, 0, 0, 0};
+        if (j0 + row < j_end) vv8 = *reinterpret_cast<const bf16x8*>(vb + (size_t)(j0 + row) * HD + c8);
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          *reinterpret_cast<unsigned short*>(&vt_raw[wave][DVT_BYTE(c8 + e, row * 2)]) = (unsigned short)vv8[e];
+      }
+      __builtin_amdgcn_s_waitcnt(0);
+
+      // ---- S = Q K^T, K direct from global: B[k = kdim][n = key]
+      f32x4 s_acc[DKVT / 16];
 #pragma unroll
       for (int nb = 0; nb < DKVT / 16; ++nb) {
+        s_acc[nb] = f32x4{0.f, 0.f, 0.f, 0.f};
         const int key = j0 + nb * 16 + col;
-        s[nb] = s_acc[nb][r] + sl[r] * key;
-        if (!full && key >= j_end) s[nb] = NEG_SENTINEL;
+        const unsigned short* krow = kb + (size_t)min(key, j_end - 1) * HD + hi * 8;
+#pragma unroll
+        for (int kc = 0; kc < KCH; ++kc) {
+          const bf16x8 kt = *reinterpret_cast<const bf16x8*>(krow + kc * 32);
+          s_acc[nb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(q_frag[kc], kt, s_acc[nb], 0, 0, 0);
+        }
       }
-      float mx = NEG_SENTINEL;
+
+      // ---- online softmax over this tile (rows = heads g0 + hi*4+r)
+      float p[DKVT / 16][4];
+      float corr[4];
 #pragma unroll
-      for (int nb = 0; nb < DKVT / 16; ++nb) mx = fmaxf(mx, s[nb]);
+      for (int r = 0; r < 4; ++r) {
+        float s[DKVT / 16];
 #pragma unroll
-      for (int off = 8; off > 0; off >>= 1) mx = fmaxf(mx, __shfl_xor(mx, off, WAVE));
-      const float m_new = fmaxf(m_row[r], mx);
-      corr[r] = (m_row[r] <= NEG_THRESHOLD) ? 0.f : __expf(m_row[r] - m_new);
-      float lsum = 0.f;
+        for (int nb = 0; nb < DKVT / 16; ++nb) {
+          const int key = j0 + nb * 16 + col;
+          s[nb] = s_acc[nb][r] + sl[r] * key;
+          if (!full && key >= j_end) s[nb] = NEG_SENTINEL;
+        }
+        float mx = NEG_SENTINEL;
 #pragma unroll
-      for (int nb = 0; nb < DKVT / 16; ++nb) {
-        p[nb][r] = (s[nb] <= NEG_THRESHOLD) ? 0.f : __expf(s[nb] - m_new);
-        lsum += p[nb][r];
+        for (int nb = 0; nb < DKVT / 16; ++nb) mx = fmaxf(mx, s[nb]);
+#pragma unroll
+        for (int off = 8; off > 0; off >>= 1) mx = fmaxf(mx, __shfl_xor(mx, off, WAVE));
+        const float m_new = fmaxf(m_row[r], mx);
+        corr[r] = (m_row[r] <= NEG_THRESHOLD) ? 0.f : __expf(m_row[r] - m_new);
+        float lsum = 0.f;
+#pragma unroll
+        for (int nb = 0; nb < DKVT / 16; ++nb) {
+          p[nb][r] = (s[nb] <= NEG_THRESHOLD) ? 0.f : __expf(s[nb] - m_new);
+          lsum += p[nb][r];
+        }
+#pragma unroll
+        for (int off = 8; off > 0; off >>= 1) lsum += __shfl_xor(lsum, off, WAVE);
+        l_row[r] = l_row[r] * corr[r] + lsum;
+        m_row[r] = m_new;
       }
+
 #pragma unroll
-      for (int off = 8; off > 0; off >>= 1) lsum += __shfl_xor(lsum, off, WAVE);
-      l_row[r] = l_row[r] * corr[r] + lsum;
-      m_row[r] = m_new;
+      for (int r = 0; r < 4; ++r)
+#pragma unroll
+        for (int nb = 0; nb < DKVT / 16; ++nb)
+          p_lds[wave][hi * 4 + r][nb * 16 + col] = f32_to_bf16(p[nb][r]);
+      __builtin_amdgcn_s_waitcnt(0);
+
+      const bf16x8 p_frag = *reinterpret_cast<const bf16x8*>(&p_lds[wave][col][hi * 8]);
+
+#pragma unroll
+      for (int d = 0; d < HD / 16; ++d)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) acc_o[d][r] *= corr[r];
+#pragma unroll
+      for (int d = 0; d < HD / 16; ++d) {
+        const bf16x8 vfrag = *reinterpret_cast<const bf16x8*>(&vt_raw[wave][DVT_BYTE(d * 16 + col, hi * 16)]);
+        acc_o[d] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(p_frag, vfrag, acc_o[d], 0, 0, 0);
+      }
     }
 
+    // ---- cross-wave combine for this head group (overlay V^T LDS with O rows)
+    __syncthreads();
+    float* c_o = reinterpret_cast<float*>(vt_raw[wave]);
 #pragma unroll
-    for (int r = 0; r < 4; ++r)
+    for (int r = 0; r < 4; ++r) {
+      const int gl = hi * 4 + r;
+      if (g0 + gl < GQ) {
 #pragma unroll
-      for (int nb = 0; nb < DKVT / 16; ++nb)
-        p_lds[wave][hi * 4 + r][nb * 16 + col] = f32_to_bf16(p[nb][r]);
-    __builtin_amdgcn_s_waitcnt(0);
-
-    const bf16x8 p_frag = *reinterpret_cast<const bf16x8*>(&p_lds[wave][col][hi * 8]);
-
-#pragma unroll
-    for (int d = 0; d < HD / 16; ++d)
-#pragma unroll
-      for (int r = 0; r < 4; ++r) acc_o[d][r] *= corr[r];
-#pragma unroll
-    for (int d = 0; d < HD / 16; ++d) {
-      const bf16x8 vfrag = *reinterpret_cast<const bf16x8*>(&vt_raw[wave][DVT_BYTE(d * 16 + col, hi * 16)]);
-      acc_o[d] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(p_frag, vfrag, acc_o[d], 0, 0, 0);
-    }
-  }
-
-  // ---- cross-wave combine (overlay the V^T LDS with per-wave O rows)
-  __syncthreads();
-  float* c_o = reinterpret_cast<float*>(vt_raw[wave]);
-#pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    const int g = hi * 4 + r;
-    if (g < GQ) {
-#pragma unroll
-      for (int d = 0; d < HD / 16; ++d) c_o[g * HD + d * 16 + col] = acc_o[d][r];
-      if (col == 0) {
-        c_ml[wave][g][0] = m_row[r];
-        c_ml[wave][g][1] = l_row[r];
+        for (int d = 0; d < HD / 16; ++d) c_o[gl * HD + d * 16 + col] = acc_o[d][r];
+        if (col == 0) {
+          c_ml[wave][gl][0] = m_row[r];
+          c_ml[wave][gl][1] = l_row[r];
+        }
       }
     }
-  }
-  __syncthreads();
+    __syncthreads();
 
-  if (wave == 0) {
-    float* po = part_o + (((size_t)bkv * n_splits + split) * GQ) * HD;
-    float* pml = part_ml + (((size_t)bkv * n_splits + split) * GQ) * 2;
-    for (int idx = lane; idx < GQ * HD; idx += WAVE) {
-      const int g = idx / HD, d = idx - g * HD;
-      float m_star = c_ml[0][g][0];
+    if (wave == 0) {
+      const int gn = min(16, GQ - g0);  // heads in this group
+      float* po = part_o + (((size_t)bkv * n_splits + split) * GQ + g0) * HD;
+      float* pml = part_ml + (((size_t)bkv * n_splits + split) * GQ + g0) * 2;
+      for (int idx = lane; idx < gn * HD; idx += WAVE) {
+        const int gl = idx / HD, d = idx - gl * HD;
+        float m_star = c_ml[0][gl][0];
 #pragma unroll
-      for (int w = 1; w < DWAVES; ++w) m_star = fmaxf(m_star, c_ml[w][g][0]);
-      float osum = 0.f, lsum = 0.f;
+        for (int w = 1; w < DWAVES; ++w) m_star = fmaxf(m_star, c_ml[w][gl][0]);
+        float osum = 0.f, lsum = 0.f;
 #pragma unroll
-      for (int w = 0; w < DWAVES; ++w) {
-        const float wgt = (c_ml[w][g][0] <= NEG_THRESHOLD) ? 0.f : __expf(c_ml[w][g][0] - m_star);
-        osum += wgt * reinterpret_cast<const float*>(vt_raw[w])[g * HD + d];
-        lsum += wgt * c_ml[w][g][1];
-      }
-      po[idx] = osum;
-      if (d == 0) {
-        pml[g * 2 + 0] = m_star;
-        pml[g * 2 + 1] = lsum;
+        for (int w = 0; w < DWAVES; ++w) {
+          const float wgt = (c_ml[w][gl][0] <= NEG_THRESHOLD) ? 0.f : __expf(c_ml[w][gl][0] - m_star);
+          osum += wgt * reinterpret_cast<const float*>(vt_raw[w])[gl * HD + d];
+          lsum += wgt * c_ml[w][gl][1];
+        }
+        po[idx] = osum;
+        if (d == 0) {
+          pml[gl * 2 + 0] = m_star;
+          pml[gl * 2 + 1] = lsum;
+        }
       }
     }
   }
@@ -472,6 +486,18 @@ torch::Tensor attn_decode_fused(
   ATTN_CASE(128, 8) ATTN_CASE(128, 16)
   ATTN_CASE(64, 1) ATTN_CASE(64, 2) ATTN_CASE(64, 4) ATTN_CASE(64, 8) ATTN_CASE(64, 16)
 #undef ATTN_CASE
+  // GQ > 16 (multi-group MFMA path only; the VALU kernel would blow its
+  // per-lane register arrays): falcon-7b MQA (gq=71), falcon-180b (gq=29)
+#define ATTN_CASE_BIG(HDV, GQV)                                               \
+  if (hd == HDV && GQi == GQV) {                                              \
+    TORCH_CHECK(!use_valu, "VALU decode kernel does not support gq=", GQV);   \
+    mfma_decode_kernel<HDV, GQV><<<grid, DWAVES * WAVE, 0, stream>>>(         \
+        qp, kp, vp, pop, pmlp, lenp, alibi_p, kv_heads, lmax, n_splits, sc);  \
+    launched = true;                                                          \
+  }
+  ATTN_CASE_BIG(64, 29) ATTN_CASE_BIG(64, 71) ATTN_CASE_BIG(64, 32)
+  ATTN_CASE_BIG(128, 32)
+#undef ATTN_CASE_BIG
   TORCH_CHECK(launched, "unsupported (head_dim, gqa) = (", hd, ", ", GQi, ")");
   HIP_CHECK_LAST();
 

@@ -614,7 +614,13 @@ class FalconFastPath:
 
     def __init__(self, block, hip_ops, quant: str = "none"):
         cfg = block.config
-        assert cfg.new_decoder_architecture, "fused Falcon path covers the new-decoder architecture"
+        # two fused geometries: the new-decoder architecture (40B/180B:
+        # ln_attn + ln_mlp, GQA) and the 7B-style old decoder (single
+        # input_layernorm feeding BOTH attn and mlp, parallel residual, MQA —
+        # its fused-QKV layout equals the new-decoder layout with n_kv=1)
+        self.single_ln = not cfg.new_decoder_architecture
+        if self.single_ln:
+            assert cfg.parallel_attn, "fused Falcon old-decoder path needs parallel_attn"
         self.hip = hip_ops
         self.cfg = cfg
         self.quant = quant
@@ -637,10 +643,12 @@ class FalconFastPath:
         mlp = block.mlp
         self.w_h4h = _FastWeight(t(mlp.dense_h_to_4h.weight), hip_ops, quant)
         self.w_4hh = _FastWeight(t(mlp.dense_4h_to_h.weight), hip_ops, quant)
-        self.ln_attn_w = block.ln_attn.weight.detach().to(torch.bfloat16).contiguous()
-        self.ln_attn_b = block.ln_attn.bias.detach().to(torch.bfloat16).contiguous()
-        self.ln_mlp_w = block.ln_mlp.weight.detach().to(torch.bfloat16).contiguous()
-        self.ln_mlp_b = block.ln_mlp.bias.detach().to(torch.bfloat16).contiguous()
+        ln_attn = block.input_layernorm if self.single_ln else block.ln_attn
+        ln_mlp = block.input_layernorm if self.single_ln else block.ln_mlp
+        self.ln_attn_w = ln_attn.weight.detach().to(torch.bfloat16).contiguous()
+        self.ln_attn_b = ln_attn.bias.detach().to(torch.bfloat16).contiguous()
+        self.ln_mlp_w = ln_mlp.weight.detach().to(torch.bfloat16).contiguous()
+        self.ln_mlp_b = ln_mlp.bias.detach().to(torch.bfloat16).contiguous()
         self._empty_bf16 = torch.empty(0, device=device, dtype=torch.bfloat16)
         for lin in (attn.query_key_value, attn.dense, mlp.dense_h_to_4h, mlp.dense_4h_to_h):
             lin.weight.data = self._empty_bf16
@@ -683,7 +691,7 @@ class FalconFastPath:
         ws = _get_ws(self.device, "gemv", 64 * B * self._max_gemv_out())
 
         xn_attn = self.hip.layer_norm_f32out(h, self.ln_attn_w, self.ln_attn_b, self.eps)
-        xn_mlp = self.hip.layer_norm_f32out(h, self.ln_mlp_w, self.ln_mlp_b, self.eps)
+        xn_mlp = xn_attn if self.single_ln else self.hip.layer_norm_f32out(h, self.ln_mlp_w, self.ln_mlp_b, self.eps)
         qkv = self.wqkv_t.gemv(xn_attn, ws, None, _EPI_PLAIN_F32)
         self.hip.rope_cache_write(
             qkv, self.rope_cos, self.rope_sin, pos, k_cache[:B], v_cache[:B], self.qh, self.kh
@@ -718,7 +726,7 @@ class FalconFastPath:
             return self.hip.layer_norm(x, w, b, self.eps)
 
         xn_attn = ln(hidden, self.ln_attn_w, self.ln_attn_b)
-        xn_mlp = ln(hidden, self.ln_mlp_w, self.ln_mlp_b)
+        xn_mlp = xn_attn if self.single_ln else ln(hidden, self.ln_mlp_w, self.ln_mlp_b)
         qkv = torch.matmul(xn_attn, self.wqkv_t.dense())
         q, k, v = self._split_heads(qkv, B, S)
         pos = torch.arange(prefix_length, end, device=hidden.device)

@@ -460,6 +460,42 @@ def test_falcon_block_fast_decode_matches_cpu(hip):
 
 
 @requires_gpu
+def test_falcon_gq29_block_fast_decode_matches_cpu(hip):
+    """falcon-180b head geometry (gq=29): multi-group MFMA decode vs CPU."""
+    _block_fused_vs_cpu("test-falcon-gq29", 1856)
+
+
+@requires_gpu
+def test_falcon_mqa71_block_fast_decode_matches_cpu(hip):
+    """falcon-7b geometry (old decoder, MQA gq=71, single LN) vs CPU."""
+    _block_fused_vs_cpu("test-falcon-mqa71", 4544)
+
+
+@requires_gpu
+@pytest.mark.parametrize("gq,kv_heads,kv_len", [(29, 1, 333), (71, 1, 95), (32, 2, 200)])
+def test_attn_decode_big_gq(hip, gq, kv_heads, kv_len):
+    """gq > 16: the decode kernel loops ceil(gq/16) A-fragment head groups."""
+    from petals_amd.ops import reference
+
+    torch.manual_seed(6)
+    b, hd, lmax = 2, 64, 512
+    q = torch.randn(b, kv_heads * gq * hd, device="cuda")
+    k_cache = torch.zeros(b, kv_heads, lmax, hd, device="cuda", dtype=torch.bfloat16)
+    v_cache = torch.zeros_like(k_cache)
+    k_cache[:, :, :kv_len] = (torch.randn(b, kv_heads, kv_len, hd, device="cuda") * 0.5).to(torch.bfloat16)
+    v_cache[:, :, :kv_len] = (torch.randn(b, kv_heads, kv_len, hd, device="cuda") * 0.5).to(torch.bfloat16)
+    kv_len_t = torch.tensor([kv_len], dtype=torch.int32, device="cuda")
+    empty = torch.empty(0, device="cuda")
+    out = hip.attn_decode_fused(q, k_cache, v_cache, kv_len_t, gq, 0, empty, empty, 1.0 / math.sqrt(hd))
+
+    q_ref = q.view(b, kv_heads * gq, 1, hd).float().cpu()
+    kr = k_cache[:, :, :kv_len].float().cpu()
+    vr = v_cache[:, :, :kv_len].float().cpu()
+    ref = reference.attention(q_ref, kr, vr, causal=False).view(b, -1)
+    assert torch.allclose(out.cpu(), ref, atol=2e-2, rtol=2e-2), (out.cpu() - ref).abs().max()
+
+
+@requires_gpu
 def test_gemv_int8(hip):
     """Weight-only int8 gemv (per-column scales) vs the same math in torch."""
     torch.manual_seed(11)
