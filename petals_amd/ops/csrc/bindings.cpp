@@ -38,6 +38,12 @@ torch::Tensor gemv_nf4(
     torch::Tensor packed, torch::Tensor absmax, torch::Tensor x, torch::Tensor workspace,
     c10::optional<torch::Tensor> residual, int64_t epilogue, int64_t splits_override,
     c10::optional<torch::Tensor> bias);
+torch::Tensor gemv_bf16_moe(
+    torch::Tensor wt_all, torch::Tensor x, torch::Tensor sel, int64_t k_per_tok,
+    torch::Tensor workspace, int64_t epilogue, int64_t splits_override);
+torch::Tensor gemv_nf4_moe(
+    torch::Tensor packed_all, torch::Tensor absmax_all, torch::Tensor x, torch::Tensor sel,
+    int64_t k_per_tok, torch::Tensor workspace, int64_t epilogue, int64_t splits_override);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rms_norm", &rms_norm, "RMSNorm (bf16 -> bf16)");
@@ -67,4 +73,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemv_nf4", &gemv_nf4, "split-K NF4 gemv (dequant fused) with epilogue",
         py::arg("packed"), py::arg("absmax"), py::arg("x"), py::arg("workspace"),
         py::arg("residual"), py::arg("epilogue"), py::arg("splits") = 0, py::arg("bias") = py::none());
+  m.def("gemv_bf16_moe", &gemv_bf16_moe,
+        "device-routed MoE gemv: stacked bf16 expert weights, expert ids from a device tensor",
+        py::arg("wt_all"), py::arg("x"), py::arg("sel"), py::arg("k_per_tok"),
+        py::arg("workspace"), py::arg("epilogue"), py::arg("splits") = 0);
+  m.def("gemv_nf4_moe", &gemv_nf4_moe,
+        "device-routed MoE gemv: stacked NF4 expert weights, expert ids from a device tensor",
+        py::arg("packed_all"), py::arg("absmax_all"), py::arg("x"), py::arg("sel"),
+        py::arg("k_per_tok"), py::arg("workspace"), py::arg("epilogue"), py::arg("splits") = 0);
 }

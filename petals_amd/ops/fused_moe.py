@@ -1,16 +1,23 @@
 """MI355X fast path for Mixtral blocks: fused Llama-style attention decode +
-routed expert GEMVs (decode-time "grouped GEMM" degenerates to top-k indexed
-GEMVs on stacked transposed expert weights — each selected expert's weights
-are read exactly once per token).
+device-routed expert GEMVs.
 
-Not hipGraph-safe (expert selection is data-dependent), so the serving span
-graph is disabled for Mixtral spans (graph_safe = False); everything else
-(8-kernel attention decode, NF4 experts) applies.
+Decode routing runs entirely ON the GPU (router gemm -> softmax -> top-k ->
+int32 expert ids in a device tensor) and the expert GEMVs read their expert's
+stacked weights through that tensor (ops/csrc/moe.hip), so there is no host
+sync per step and the whole span is hipGraph-capturable (graph_safe) for the
+bf16 and NF4 quantizations. Each selected expert's weights are read exactly
+once per token — the decode-time degenerate case of a grouped GEMM.
+
+(The reference runs Mixtral experts densely inside one server's HF block with
+host-side routing: reference models/mixtral/block.py:73-81.)
+
+int8 expert weights keep the round-1 host-routed per-expert GEMV path
+(graph_safe = False).
 """
 
 from __future__ import annotations
 
-from typing import List
+from typing import List, Optional
 
 import torch
 import torch.nn.functional as F
@@ -23,9 +30,40 @@ from petals_amd.ops.fused_decode import (
 )
 
 
-class MixtralFastPath(LlamaFastPath):
-    graph_safe = False
+class _StackedExperts:
+    """All experts' transposed [in, out] weights stacked on a leading expert
+    axis, in the exact layout the moe.hip kernels stream."""
 
+    def __init__(self, t_list: List[torch.Tensor], hip, quant: str):
+        self.hip = hip
+        self.quant = quant
+        self.in_dim, self.out_dim = t_list[0].shape
+        if quant == "nf4":
+            packed, absmax = [], []
+            for t in t_list:
+                p, a = hip.nf4_quantize(t.contiguous())
+                packed.append(p)
+                absmax.append(a)
+            self.packed_all = torch.stack(packed).contiguous()
+            self.absmax_all = torch.stack(absmax).contiguous()
+        else:
+            assert quant == "none"
+            self.wt_all = torch.stack([t.contiguous() for t in t_list]).contiguous()
+
+    def moe_gemv(self, x: torch.Tensor, sel: torch.Tensor, k_per_tok: int, ws: torch.Tensor,
+                 epilogue: int) -> torch.Tensor:
+        if self.quant == "nf4":
+            return self.hip.gemv_nf4_moe(self.packed_all, self.absmax_all, x, sel, k_per_tok, ws, epilogue)
+        return self.hip.gemv_bf16_moe(self.wt_all, x, sel, k_per_tok, ws, epilogue)
+
+    def dense(self, e: int) -> torch.Tensor:
+        """Expert e as a dense bf16 [in, out] (prefill/training matmuls)."""
+        if self.quant == "nf4":
+            return self.hip.nf4_dequantize(self.packed_all[e], self.absmax_all[e])
+        return self.wt_all[e]
+
+
+class MixtralFastPath(LlamaFastPath):
     def _init_mlp_weights(self, block, hip_ops, quant):
         def t(w):
             return w.detach().to(torch.bfloat16).t().contiguous()
@@ -34,22 +72,35 @@ class MixtralFastPath(LlamaFastPath):
         self.top_k = moe.top_k
         self.num_experts = moe.num_experts
         self.w_router_t = t(moe.gate.weight)  # [H, E]
+        self.stacked_gu: Optional[_StackedExperts] = None
+        self.stacked_down: Optional[_StackedExperts] = None
         self.wgateup_e: List[_FastWeight] = []
         self.wdown_e: List[_FastWeight] = []
+        if quant in ("none", "nf4"):
+            gu_list = [torch.cat([t(e.w1.weight), t(e.w3.weight)], dim=1) for e in moe.experts]
+            down_list = [t(e.w2.weight) for e in moe.experts]
+            self.stacked_gu = _StackedExperts(gu_list, hip_ops, quant)
+            self.stacked_down = _StackedExperts(down_list, hip_ops, quant)
+            self.graph_safe = True  # device-routed decode: capturable
+        else:
+            for expert in moe.experts:
+                self.wgateup_e.append(
+                    _FastWeight(torch.cat([t(expert.w1.weight), t(expert.w3.weight)], dim=1), hip_ops, quant)
+                )
+                self.wdown_e.append(_FastWeight(t(expert.w2.weight), hip_ops, quant))
+            self.graph_safe = False  # host-routed fallback
         for expert in moe.experts:
-            self.wgateup_e.append(
-                _FastWeight(torch.cat([t(expert.w1.weight), t(expert.w3.weight)], dim=1), hip_ops, quant)
-            )
-            self.wdown_e.append(_FastWeight(t(expert.w2.weight), hip_ops, quant))
             for lin in (expert.w1, expert.w2, expert.w3):
                 lin.weight.data = self._empty_bf16
         moe.gate.weight.data = self._empty_bf16
 
     def _max_gemv_out(self) -> int:
-        return max(self.wgateup_e[0].shape[1], self.wqkv_t.shape[1], self.wo_t.shape[1])
+        gu_out = self.stacked_gu.out_dim if self.stacked_gu is not None else self.wgateup_e[0].shape[1]
+        return max(gu_out, self.wqkv_t.shape[1], self.wo_t.shape[1])
 
     def _route(self, xn2: torch.Tensor):
-        """Router on [tokens, H] f32/bf16 -> (weights [t, k], experts [t, k])."""
+        """Router on [tokens, H] f32/bf16 -> (weights [t, k], experts [t, k]).
+        All-device: safe inside hipGraph capture."""
         logits = xn2.to(self.w_router_t.dtype) @ self.w_router_t
         probs = F.softmax(logits.float(), dim=-1)
         weights, selected = torch.topk(probs, self.top_k, dim=-1)
@@ -57,21 +108,34 @@ class MixtralFastPath(LlamaFastPath):
         return weights, selected
 
     def _mlp_decode(self, xn2, h2, ws, adapter):
-        """xn2 f32 [B, H]; returns h3 bf16 [B, H]. Top-k expert GEMVs per row
-        (host-synced routing: ~10 us, amortized against ~MBs of expert reads)."""
+        """xn2 f32 [B, H]; returns h3 bf16 [B, H]."""
         B = xn2.shape[0]
         weights, selected = self._route(xn2)
-        sel = selected.tolist()
-        w = weights.tolist()
+        if self.stacked_gu is not None:
+            sel = selected.to(torch.int32).reshape(-1).contiguous()  # [B*k]
+            act = self.stacked_gu.moe_gemv(xn2.contiguous(), sel, self.top_k, ws, _EPI_SWIGLU_F32)
+            down = self.stacked_down.moe_gemv(act.contiguous(), sel, 1, ws, _EPI_PLAIN_F32)
+            moe_out = (down.view(B, self.top_k, -1) * weights.unsqueeze(-1)).sum(dim=1)
+            return (h2.float() + moe_out).to(torch.bfloat16)
+
+        # int8: host-synced routing (~10 us, amortized against MBs of expert reads)
+        sel_l = selected.tolist()
+        w_l = weights.tolist()
         moe_out = torch.zeros_like(xn2)  # f32 [B, H]
         for b in range(B):
             xb = xn2[b : b + 1].contiguous()
             for j in range(self.top_k):
-                e = sel[b][j]
+                e = sel_l[b][j]
                 act = self.wgateup_e[e].gemv(xb, ws, None, _EPI_SWIGLU_F32)  # [1, I] f32
                 down = self.wdown_e[e].gemv(act, ws, None, _EPI_PLAIN_F32)  # [1, H] f32
-                moe_out[b] += down[0] * w[b][j]
+                moe_out[b] += down[0] * w_l[b][j]
         return (h2.float() + moe_out).to(torch.bfloat16)
+
+    def _expert_dense(self, e: int) -> torch.Tensor:
+        return self.stacked_gu.dense(e) if self.stacked_gu is not None else self.wgateup_e[e].dense()
+
+    def _expert_down_dense(self, e: int) -> torch.Tensor:
+        return self.stacked_down.dense(e) if self.stacked_down is not None else self.wdown_e[e].dense()
 
     def _mlp_dense(self, xn2, adapter, autograd: bool):
         """Prefill / training MLP: token-grouped expert matmuls on the dense
@@ -79,16 +143,17 @@ class MixtralFastPath(LlamaFastPath):
         shape = xn2.shape
         x = xn2.reshape(-1, shape[-1])
         weights, selected = self._route(x)
-        out = torch.zeros(x.shape[0], self.wdown_e[0].shape[1], dtype=torch.float32, device=x.device)
+        out_dim = self.stacked_down.out_dim if self.stacked_down is not None else self.wdown_e[0].shape[1]
+        out = torch.zeros(x.shape[0], out_dim, dtype=torch.float32, device=x.device)
         expert_mask = F.one_hot(selected, num_classes=self.num_experts).permute(2, 1, 0)
         for e in range(self.num_experts):
             k_idx, tok_idx = torch.where(expert_mask[e])
             if tok_idx.numel() == 0:
                 continue
             xe = x[tok_idx].to(torch.bfloat16)
-            gateup = xe @ self.wgateup_e[e].dense()
+            gateup = xe @ self._expert_dense(e)
             inter = gateup.shape[-1] // 2
             act = F.silu(gateup[..., :inter].float()) * gateup[..., inter:].float()
-            down = act.to(torch.bfloat16) @ self.wdown_e[e].dense()
+            down = act.to(torch.bfloat16) @ self._expert_down_dense(e)
             out.index_add_(0, tok_idx, down.float() * weights[tok_idx, k_idx, None])
         return out.reshape(*shape[:-1], -1).to(xn2.dtype)

@@ -237,6 +237,53 @@ def test_mixtral_block_fast_decode_matches_cpu(hip):
 
 
 @requires_gpu
+@pytest.mark.parametrize("quant", ["none", "nf4"])
+def test_mixtral_decode_graph_capture(hip, quant):
+    """Device-routed MoE decode is hipGraph-safe: capture one decode step,
+    replay over new tokens, match the eager fused path exactly."""
+    from petals_amd.models import get_model_block
+    from petals_amd.models.config_base import load_model_config
+    from petals_amd.ops.fused_decode import DecodeContext
+    from petals_amd.server.from_pretrained import init_random_block_
+    from petals_amd.utils.graphs import GraphedCallable
+
+    cfg = load_model_config("test-mixtral")
+    cfg.hidden_size, cfg.num_attention_heads, cfg.num_key_value_heads, cfg.intermediate_size = 512, 4, 2, 1024
+    blk = get_model_block(cfg, 0)
+    init_random_block_(blk, cfg, 0)
+    blk = blk.to("cuda", torch.bfloat16).eval().optimize_for_inference(quant=quant)
+    assert blk._fast is not None and blk._fast.graph_safe, "device-routed MoE must be graph-safe"
+
+    blk2 = get_model_block(cfg, 0)
+    init_random_block_(blk2, cfg, 0)
+    blk2 = blk2.to("cuda", torch.bfloat16).eval().optimize_for_inference(quant=quant)
+
+    torch.manual_seed(7)
+    ks, vs = blk.kv_cache_shape(1, 16)
+    kg = torch.zeros(ks, device="cuda", dtype=torch.bfloat16)
+    vg = torch.zeros(vs, device="cuda", dtype=torch.bfloat16)
+    kg2, vg2 = kg.clone(), vg.clone()
+    xs = [torch.randn(1, 1, 512, device="cuda", dtype=torch.bfloat16) * 0.5 for _ in range(4)]
+
+    ctx = DecodeContext(torch.device("cuda"))
+    ctx.set_position(0)
+    h_in = torch.empty_like(xs[0])
+
+    def step():
+        return blk(h_in, kv_cache=(kg, vg), ctx=ctx)
+
+    g = GraphedCallable(step, [])
+    outs_graph, outs_eager = [], []
+    for t, x in enumerate(xs):
+        ctx.set_position(t)
+        h_in.copy_(x)
+        outs_graph.append(g.replay().clone())
+        outs_eager.append(blk2(x, kv_cache=(kg2, vg2), prefix_length=t))
+    for og, oe in zip(outs_graph, outs_eager):
+        assert torch.allclose(og.float(), oe.float(), atol=1e-3, rtol=1e-3), (og - oe).abs().max()
+
+
+@requires_gpu
 @pytest.mark.parametrize("case", [
     dict(b=2, qh=8, kvh=2, s=67, hd=128, off=0, causal=True),
     dict(b=1, qh=4, kvh=4, s=200, hd=128, off=0, causal=True),
