@@ -1,11 +1,17 @@
 """Autoregressive generation against remote sessions.
 
-Capability parity with the reference's RemoteGenerationMixin
-(client/remote_generation.py:84 — which delegates to HF GenerationMixin):
-greedy, temperature/top-k/top-p sampling, repetition penalty, simple beam
-search (server-side KV reorder via hypo_ids), resuming an open session across
-multiple generate() calls. Implemented natively to avoid coupling the client
-to transformers' generation internals.
+Two paths, matching the reference's RemoteGenerationMixin capability
+(reference client/remote_generation.py:20-163, which delegates to HF
+GenerationMixin):
+
+* the NATIVE loop (greedy / temperature / top-k / top-p / repetition penalty /
+  simple beam search with server-side KV reorder, session resume) — the fast
+  serving path, no per-token transformers overhead;
+* full `transformers.GenerationMixin` delegation via `_HFGenerateFacade`
+  whenever HF-specific machinery is requested (logits_processor,
+  stopping_criteria, streamer, generation_config, constrained decoding, ...):
+  a `RemotePastKeyValues` Cache shim tracks the server-side session position
+  and beam reorders ship to the servers as hypo_ids.
 """
 
 from __future__ import annotations
@@ -18,9 +24,184 @@ import torch.nn.functional as F
 
 from petals_amd.client.inference_session import InferenceSession
 
+# kwargs that route generate() through transformers.GenerationMixin
+_HF_ONLY_KWARGS = frozenset(
+    [
+        "logits_processor", "stopping_criteria", "streamer", "generation_config",
+        "prefix_allowed_tokens_fn", "synced_gpus", "assistant_model",
+        "negative_prompt_ids", "negative_prompt_attention_mask",
+        "bad_words_ids", "force_words_ids", "constraints", "num_beam_groups",
+        "diversity_penalty", "penalty_alpha", "top_a", "typical_p", "epsilon_cutoff",
+        "eta_cutoff", "no_repeat_ngram_size", "encoder_no_repeat_ngram_size",
+        "min_length", "min_new_tokens", "exponential_decay_length_penalty",
+        "suppress_tokens", "begin_suppress_tokens", "forced_bos_token_id",
+        "forced_eos_token_id", "guidance_scale", "low_memory", "length_penalty",
+        "early_stopping", "num_return_sequences", "output_scores", "output_logits",
+        "return_dict_in_generate", "use_hf_generate",
+    ]
+)
+
+
+class RemotePastKeyValues:
+    """Pretends to be a transformers Cache: only tracks the number of tokens
+    already consumed by the remote session (the real KV lives server-side).
+    Parity: reference client/remote_generation.py:20-42."""
+
+    is_compileable = False
+
+    def __init__(self, seen: int = 0):
+        self.layers = []  # transformers-5 Cache protocol
+        self._seen = seen
+        self.hypo_ids: Optional[torch.Tensor] = None
+
+    def get_seq_length(self, layer_idx: int = 0) -> int:
+        return self._seen
+
+    def get_max_cache_shape(self):
+        return None
+
+    def get_max_length(self):
+        return None
+
+    def update_seen(self, n: int) -> None:
+        self._seen += n
+
+    def __len__(self):
+        return 0
+
+    def __getitem__(self, _):
+        from petals_amd.utils.misc import DUMMY
+
+        return [DUMMY]
+
+
+_FACADE_CLS = None
+
+
+def _get_facade_cls():
+    """Lazily define the facade as a real transformers.GenerationMixin
+    SUBCLASS (generate() looks decoding methods up on type(self))."""
+    global _FACADE_CLS
+    if _FACADE_CLS is not None:
+        return _FACADE_CLS
+    import transformers
+    from transformers.modeling_outputs import CausalLMOutputWithPast
+
+    class _HFGenerateFacade(transformers.GenerationMixin):
+        """The object transformers.GenerationMixin.generate() runs against:
+        it exposes the HF model surface (config/generation_config/forward/
+        prepare_inputs_for_generation/_reorder_cache) over a Distributed*
+        model while keeping the real model's own config untouched."""
+
+        main_input_name = "input_ids"
+
+        def __init__(self, model):
+            self._model = model
+            cfg = model.config
+            self.config = transformers.PretrainedConfig(
+                vocab_size=cfg.vocab_size,
+                hidden_size=cfg.hidden_size,
+                is_decoder=True,
+                is_encoder_decoder=False,
+            )
+            self.generation_config = transformers.GenerationConfig(
+                eos_token_id=getattr(cfg, "eos_token_id", None),
+                bos_token_id=getattr(cfg, "bos_token_id", None),
+                pad_token_id=getattr(cfg, "pad_token_id", None),
+            )
+
+        @property
+        def device(self):
+            return self._model.lm_head.weight.device
+
+        def can_generate(self):
+            return True
+
+        def get_experts_implementation(self):
+            return {}
+
+        def set_experts_implementation(self, impl):
+            pass
+
+        def __call__(self, input_ids=None, past_key_values=None, **kwargs):
+            if past_key_values is not None and past_key_values.hypo_ids is not None:
+                self._model._next_hypo_ids = past_key_values.hypo_ids
+                past_key_values.hypo_ids = None
+            out = self._model(input_ids=input_ids)
+            if past_key_values is not None:
+                past_key_values.update_seen(input_ids.shape[1])
+            return CausalLMOutputWithPast(logits=out.logits, past_key_values=past_key_values)
+
+        forward = __call__
+
+        def prepare_inputs_for_generation(self, input_ids, past_key_values=None, **kwargs):
+            if past_key_values is not None and past_key_values.get_seq_length() > 0:
+                input_ids = input_ids[:, past_key_values.get_seq_length() :]
+            return {"input_ids": input_ids, "past_key_values": past_key_values}
+
+        def _reorder_cache(self, past_key_values, beam_idx):
+            # server-side KV reorder: ship beam_idx as hypo_ids with the next step
+            past_key_values.hypo_ids = beam_idx.to(torch.int64).cpu()
+            return past_key_values
+
+    _FACADE_CLS = _HFGenerateFacade
+    return _FACADE_CLS
+
 
 class RemoteGenerationMixin:
     """Mixin for Distributed*ForCausalLM models."""
+
+    @torch.inference_mode()
+    def generate_hf(
+        self,
+        input_ids: Optional[torch.Tensor] = None,
+        *,
+        session: Optional[InferenceSession] = None,
+        **kwargs,
+    ) -> torch.Tensor:
+        """Full `transformers.GenerationMixin.generate` against the swarm:
+        logits processors, stopping criteria, streamers, beam variants and
+        constrained decoding all work; the KV cache is the remote session.
+        Parity: reference client/remote_generation.py:84-163."""
+        kwargs.pop("use_hf_generate", None)
+        assert input_ids is not None and input_ids.ndim == 2, "input_ids must be [batch, seq]"
+        batch, prompt_len = input_ids.shape
+        num_beams = kwargs.get("num_beams", 1)
+        gc = kwargs.get("generation_config")
+        if num_beams == 1 and gc is not None:
+            num_beams = getattr(gc, "num_beams", 1) or 1
+        max_new_tokens = kwargs.get("max_new_tokens")
+        max_length = kwargs.get("max_length")
+        if max_new_tokens is None and max_length is None and gc is not None:
+            max_new_tokens = getattr(gc, "max_new_tokens", None)
+            max_length = getattr(gc, "max_length", None)
+        budget = (max_length - prompt_len) if (max_length is not None and max_new_tokens is None) else max_new_tokens
+        assert budget is not None and budget > 0, "provide max_new_tokens or max_length"
+
+        pre_seq_len = getattr(self, "pre_seq_len", 0) if getattr(self, "tuning_mode", None) else 0
+
+        ctx = contextlib.nullcontext(session)
+        if session is None:
+            resumed = self.transformer.h.active_session
+            if resumed is not None:
+                ctx = contextlib.nullcontext(resumed)
+            else:
+                ctx = self.transformer.h.inference_session(
+                    max_length=pre_seq_len + prompt_len + budget,
+                    batch_size=batch * max(num_beams, 1),
+                )
+
+        facade = _get_facade_cls()(self)
+        with ctx as sess, self.transformer.h.use_session(sess):
+            prev = sess.output_ids if sess.position > 0 else None
+            if prev is not None:
+                # resumed session: prepend the known context so HF criteria see
+                # the full text; the past-cache shim skips it server-side
+                input_ids = torch.cat([prev.to(input_ids.device), input_ids], dim=1)
+            past = RemotePastKeyValues(seen=sess.position)
+            out = facade.generate(input_ids, past_key_values=past, **kwargs)
+            sess.output_ids = out if torch.is_tensor(out) else out.sequences
+        return out
 
     @torch.inference_mode()
     def generate(
@@ -40,6 +221,21 @@ class RemoteGenerationMixin:
         session: Optional[InferenceSession] = None,
         **kwargs,
     ) -> torch.Tensor:
+        if any(k in _HF_ONLY_KWARGS and kwargs[k] is not None for k in kwargs):
+            # HF machinery requested: delegate to transformers.GenerationMixin
+            hf_kwargs = dict(kwargs)
+            for name, val in (
+                ("max_new_tokens", max_new_tokens), ("max_length", max_length),
+                ("do_sample", do_sample), ("temperature", temperature), ("top_k", top_k),
+                ("top_p", top_p), ("repetition_penalty", repetition_penalty),
+                ("num_beams", num_beams), ("eos_token_id", eos_token_id),
+                ("pad_token_id", pad_token_id),
+            ):
+                if val is not None and not (name == "temperature" and val == 1.0) and not (
+                    name in ("do_sample",) and val is False
+                ) and not (name == "num_beams" and val == 1):
+                    hf_kwargs[name] = val
+            return self.generate_hf(input_ids, session=session, **hf_kwargs)
         assert input_ids is not None and input_ids.ndim == 2, "input_ids must be [batch, seq]"
         if eos_token_id is None:
             eos_token_id = getattr(self.config, "eos_token_id", None)

@@ -115,6 +115,43 @@ def test_sampling_generate_deterministic(client_model):
     assert torch.equal(out1, out2)
 
 
+def test_hf_generate_logits_processor_and_stopping(client_model, hf_checkpoint):
+    """transformers.GenerationMixin delegation: logits processors and stopping
+    criteria work against the swarm and match the local HF model exactly."""
+    import transformers
+    from transformers import LogitsProcessorList, StoppingCriteriaList
+    from transformers.generation.logits_process import NoRepeatNGramLogitsProcessor
+    from transformers.generation.stopping_criteria import MaxLengthCriteria
+
+    _, hf_model = hf_checkpoint
+    torch.manual_seed(11)
+    ids = torch.randint(0, 128, (1, 5))
+    processors = LogitsProcessorList([NoRepeatNGramLogitsProcessor(2)])
+    criteria = StoppingCriteriaList([MaxLengthCriteria(max_length=11)])
+    ref = hf_model.generate(
+        ids, max_new_tokens=8, do_sample=False,
+        logits_processor=LogitsProcessorList([NoRepeatNGramLogitsProcessor(2)]),
+        stopping_criteria=StoppingCriteriaList([MaxLengthCriteria(max_length=11)]),
+    )
+    out = client_model.generate(
+        ids, max_new_tokens=8, do_sample=False,
+        logits_processor=processors, stopping_criteria=criteria,
+    )
+    assert torch.equal(out, ref), (out, ref)
+
+
+def test_hf_generate_beam_search_matches_local(client_model, hf_checkpoint):
+    """HF beam search via delegation: server-side KV reorder through hypo_ids."""
+    _, hf_model = hf_checkpoint
+    torch.manual_seed(12)
+    ids = torch.randint(0, 128, (1, 5))
+    ref = hf_model.generate(ids, max_new_tokens=6, num_beams=3, do_sample=False,
+                            early_stopping=True)
+    out = client_model.generate(ids, max_new_tokens=6, num_beams=3, do_sample=False,
+                                early_stopping=True)
+    assert torch.equal(out, ref), (out, ref)
+
+
 def test_inference_matches_forward(client_model):
     """Token-by-token session logits == one-shot forward logits."""
     torch.manual_seed(4)
