@@ -157,62 +157,120 @@ class RemoteSequenceManager:
             current = min(chosen.end, end)
         return span_sequence
 
-    def _make_sequence_min_latency(
-        self, start: int, end: int, cache_tokens_needed: Optional[int]
-    ) -> List[RemoteSpanInfo]:
-        """Dijkstra over block indices: for each span covering block i, an edge
-        i -> span.end with cost = client RTT (first hop) + blocks/inference_rps
-        + cache-allocation overhead (parity :177-289)."""
-        n = end - start
-        INF = math.inf
-        dist = [INF] * (n + 1)
-        prev: List[Optional[Tuple[int, RemoteSpanInfo]]] = [None] * (n + 1)
-        dist[0] = 0.0
-        rtts = self.ping_aggregator.to_dict()
+    @staticmethod
+    def _rtt_to_delay(rtt: Optional[float], *, default_delay: float = 0.15, max_delay: float = 5.0) -> float:
+        if rtt is None:
+            return default_delay
+        return min(rtt / 2, max_delay)
 
+    @staticmethod
+    def _has_cache_for(span: RemoteSpanInfo, cache_tokens_needed: Optional[int]) -> bool:
+        if cache_tokens_needed is None or span.server_info.cache_tokens_left is None:
+            return True
+        # pessimistic: assume the whole hosted span is used (false positives
+        # cost more than false negatives here)
+        return cache_tokens_needed * 2 * span.length <= span.server_info.cache_tokens_left
+
+    def _make_sequence_min_latency(
+        self,
+        start: int,
+        end: int,
+        cache_tokens_needed: Optional[int],
+        *,
+        overhead_delay: float = 0.018,  # serialization overhead per hop
+        default_inference_rps: float = 300.0,
+        alloc_delay: float = 10.0,  # penalty when the server must evict cache first
+    ) -> List[RemoteSpanInfo]:
+        """Dijkstra over (peer, block) nodes, mirroring the reference cost
+        model (reference sequence_manager.py:177-289): client<->server RTTs
+        price the first and last hops; chained server->server hops are priced
+        with the UPSTREAM server's gossiped `next_pings` RTT to the downstream
+        server (announced by ModuleAnnouncer pings, reference server.py:760);
+        compute edges cost 1/inference_rps per block; a chosen server is used
+        to its span end (keeps the graph O(servers * blocks))."""
         import heapq
 
-        heap = [(0.0, 0)]
-        visited = [False] * (n + 1)
+        missing = [i for i in range(start, end) if not self._spans_at(i)]
+        if missing:
+            raise MissingBlocksError(missing)
+
+        client_rtts = self.ping_aggregator.to_dict()
+        spans_at: Dict[int, List[RemoteSpanInfo]] = {i: self._spans_at(i) for i in range(start, end)}
+        span_of: Dict[str, RemoteSpanInfo] = {}
+        for i in range(start, end):
+            for s in spans_at[i]:
+                span_of[s.peer_id] = s
+
+        # adjacency: node -> list of (cost, node). Nodes: "start", "end",
+        # (peer_id, block_idx)
+        adj: Dict[Any, List[Tuple[float, Any]]] = {}
+
+        def add_edge(u, v, c):
+            adj.setdefault(u, []).append((c, v))
+
+        for span in spans_at[start]:
+            delay = self._rtt_to_delay(client_rtts.get(span.peer_id)) + overhead_delay
+            if not self._has_cache_for(span, cache_tokens_needed):
+                delay += alloc_delay
+            add_edge("start", (span.peer_id, start), delay)
+        for span in spans_at[end - 1]:
+            add_edge((span.peer_id, end), "end", self._rtt_to_delay(client_rtts.get(span.peer_id)))
+        for block_idx in range(start + 1, end):
+            for cur in spans_at[block_idx - 1]:
+                if cur.end != block_idx:
+                    continue  # a chosen server is used to its span end
+                for nxt in spans_at[block_idx]:
+                    rtt = None
+                    if cur.server_info.next_pings is not None:
+                        rtt = cur.server_info.next_pings.get(nxt.peer_id)
+                    delay = self._rtt_to_delay(rtt) + overhead_delay
+                    if not self._has_cache_for(nxt, cache_tokens_needed):
+                        delay += alloc_delay
+                    add_edge((cur.peer_id, block_idx), (nxt.peer_id, block_idx), delay)
+        for span in span_of.values():
+            rps = span.server_info.inference_rps or span.throughput or default_inference_rps
+            for block_idx in range(max(span.start, start), min(span.end, end)):
+                add_edge((span.peer_id, block_idx), (span.peer_id, block_idx + 1), 1.0 / max(rps, 1e-9))
+
+        dist: Dict[Any, float] = {"start": 0.0}
+        prev: Dict[Any, Any] = {}
+        heap = [(0.0, 0, "start")]
+        tie = 0
+        done = set()
         while heap:
-            d, u = heapq.heappop(heap)
-            if visited[u]:
+            d, _, u = heapq.heappop(heap)
+            if u in done:
                 continue
-            visited[u] = True
-            if u == n:
+            done.add(u)
+            if u == "end":
                 break
-            block_idx = start + u
-            for span in self._spans_at(block_idx):
-                if not (span.start <= block_idx < span.end):
-                    continue
-                v = min(span.end, end) - start
-                inference_rps = span.server_info.inference_rps or span.throughput or 1.0
-                compute_cost = (v - u) / max(inference_rps, 1e-9)
-                rtt = rtts.get(span.peer_id, 0.05)
-                cache_cost = 0.0
-                if cache_tokens_needed is not None:
-                    left = span.server_info.cache_tokens_left
-                    if left is not None and left < cache_tokens_needed * (v - u):
-                        cache_cost = 1.0  # will likely wait for cache eviction
-                cost = d + rtt + compute_cost + cache_cost
-                if cost < dist[v]:
-                    dist[v] = cost
-                    prev[v] = (u, span)
-                    heapq.heappush(heap, (cost, v))
-        if not math.isfinite(dist[n]):
-            missing = [start + u for u in range(n) if not any(True for _ in self._spans_at(start + u))]
-            raise MissingBlocksError(missing or list(range(start, end)))
-        # reconstruct
+            for c, v in adj.get(u, ()):
+                nd = d + c
+                if nd < dist.get(v, math.inf):
+                    dist[v] = nd
+                    prev[v] = u
+                    tie += 1
+                    heapq.heappush(heap, (nd, tie, v))
+        if "end" not in done:
+            raise MissingBlocksError(list(range(start, end)))
+
+        # walk back: nodes between "start" and "end" are (peer, block)
+        path = []
+        node = prev["end"]
+        while node != "start":
+            path.append(node)
+            node = prev[node]
+        path.reverse()
         seq: List[RemoteSpanInfo] = []
-        v = n
-        while v > 0:
-            u, span = prev[v]
-            span = dataclasses.replace(span)
-            span.start, span.end = start + u, start + v
-            seq.append(span)
-            v = u
-        seq.reverse()
-        return seq
+        for peer_id, block_idx in path:
+            if not seq or seq[-1].peer_id != peer_id:
+                span = dataclasses.replace(span_of[peer_id])
+                span.start = span.end = block_idx
+                seq.append(span)
+            else:
+                seq[-1].end = block_idx
+        # drop empty spans (measurement-noise triangle-inequality violations)
+        return [s for s in seq if s.length > 0]
 
     # ------------------------------------------------------------ failures
 

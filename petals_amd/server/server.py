@@ -128,6 +128,7 @@ class Server:
         self._restart_requested = False
         self.server_info: Optional[ServerInfo] = None
         self.listen_addr: Optional[Tuple[str, int]] = None
+        self._ping_agg = None  # PingAggregator for next-server RTT gossip
 
     # -------------------------------------------------------------- sizing
 
@@ -327,6 +328,10 @@ class Server:
         try:
             while not self._stop.is_set():
                 if time.monotonic() - last_announce >= self.update_period:
+                    try:
+                        await self._ping_next_servers()
+                    except Exception as e:  # noqa: BLE001
+                        logger.debug("next-server pings failed: %r", e)
                     await self._announce()
                     last_announce = time.monotonic()
                 if not self.runtime.alive:
@@ -362,6 +367,46 @@ class Server:
                 servers[peer_id] = info
             infos.append(RemoteModuleInfo(uid=uid, servers=servers) if servers else None)
         return infos, addrs
+
+    async def _ping_next_servers(self, max_pinged: int = 5):
+        """Ping servers most likely to be NEXT in an inference chain (those
+        hosting the block right after our span) and gossip the RTTs via
+        ServerInfo.next_pings — the client's min-latency router prices chained
+        server->server hops with them (parity: reference server/server.py:760,
+        client/routing/sequence_manager.py:253-268)."""
+        if self.server_info is None or self.server_info.end_block is None:
+            return
+        next_block = self.server_info.end_block
+        if next_block >= self.config.num_blocks:
+            self.server_info.next_pings = {}
+            return
+        infos, addrs = await self._get_infos()
+        info = infos[next_block]
+        if info is None:
+            return
+        candidates = {
+            pid: addrs[pid]
+            for pid in info.servers
+            if pid != self.p2p.peer_id and pid in addrs
+        }
+        if len(candidates) > max_pinged:
+            import random as _random
+
+            keys = _random.sample(sorted(candidates), max_pinged)
+            candidates = {k: candidates[k] for k in keys}
+        if not candidates:
+            self.server_info.next_pings = {}
+            return
+        if self._ping_agg is None:
+            from petals_amd.utils.ping import PingAggregator
+
+            self._ping_agg = PingAggregator(self.p2p)
+        await self._ping_agg.ping(candidates)
+        import math as _math
+
+        self.server_info.next_pings = {
+            p: rtt for p, rtt in self._ping_agg.to_dict().items() if _math.isfinite(rtt)
+        }
 
     async def _announce(self):
         if self.server_info is None:

@@ -68,6 +68,36 @@ def test_min_latency_prefers_fast_full_span():
     assert [s.peer_id for s in seq] == ["fast"]
 
 
+def test_min_latency_uses_server_to_server_next_pings():
+    """Chained hops are priced with the UPSTREAM server's gossiped next_pings
+    RTT, not the client's RTT: with asymmetric server->server latencies the
+    router must pick the chain whose HOP is fast, even when the client's RTT
+    to both candidates is identical."""
+    mgr = _FakeManager(
+        8,
+        {
+            "head": (0, 4, 50.0),
+            "near": (4, 8, 50.0),  # fast hop from head
+            "far": (4, 8, 50.0),   # slow hop from head (same compute/throughput)
+        },
+    )
+    # inject next_pings on the head server's gossip: near is 1 ms away, far 2 s
+    for info in mgr.state.block_infos:
+        if info and "head" in info.servers:
+            info.servers["head"].next_pings = {"near": 0.001, "far": 2.0}
+    mgr.state.update_(mgr.state.block_infos)
+    seq = mgr.make_sequence(mode="min_latency")
+    assert [s.peer_id for s in seq] == ["head", "near"], seq
+
+    # flip the RTTs: the router must follow
+    for info in mgr.state.block_infos:
+        if info and "head" in info.servers:
+            info.servers["head"].next_pings = {"near": 2.0, "far": 0.001}
+    mgr.state.update_(mgr.state.block_infos)
+    seq = mgr.make_sequence(mode="min_latency")
+    assert [s.peer_id for s in seq] == ["head", "far"], seq
+
+
 def test_missing_blocks_raises():
     mgr = _FakeManager(8, {"A": (0, 4, 1.0)})
     with pytest.raises(MissingBlocksError):
