@@ -31,6 +31,11 @@ class ClientConfig:
 
     active_adapter: Optional[str] = None
 
+    # transport security (p2p/transport.py STARTTLS): None = inherit the
+    # process default / PETALS_AMD_SECURE env; a swarm runs all-secure or
+    # all-plain
+    secure: Optional[bool] = None
+
     # wire compression (utils/serialization): "none" | "float16" | "bfloat16" |
     # "blockwise_8bit" — applied to activations we SEND; output_compression is
     # requested from servers for what they send back

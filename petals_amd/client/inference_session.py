@@ -318,11 +318,17 @@ class InferenceSession:
         first, last = self._sessions[0], self._sessions[-1]
         next_servers = []
         for s in self._sessions[1:]:
-            host, port = self._manager.address_of(s.span.peer_id)
+            addr = self._manager.address_of(s.span.peer_id)
             si = s.span.server_info
             next_servers.append(
-                [host, port, s.session_id, s.span.start, s.span.end,
-                 getattr(si, "mesh_id", None), getattr(si, "mesh_rank", None)]
+                {
+                    "addr": list(addr),
+                    "session_id": s.session_id,
+                    "start": s.span.start,
+                    "end": s.span.end,
+                    "mesh_id": getattr(si, "mesh_id", None),
+                    "mesh_rank": getattr(si, "mesh_rank", None),
+                }
             )
         meta = {
             "step_id": step_id,

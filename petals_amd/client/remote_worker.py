@@ -68,3 +68,13 @@ def get_worker() -> RemoteWorker:
         if _singleton is None:
             _singleton = RemoteWorker()
         return _singleton
+
+
+def reset_worker() -> None:
+    """Tear down the shared worker (its P2PNode captured the process's
+    security mode at first use — call this when flipping modes, e.g. tests)."""
+    global _singleton
+    with _singleton_lock:
+        if _singleton is not None:
+            _singleton.shutdown()
+            _singleton = None

@@ -49,6 +49,10 @@ class RemoteSequenceManager:
         dht: Optional[DHT] = None,
     ):
         self.config = config
+        if config.secure is not None:
+            from petals_amd.p2p import transport as _transport
+
+            _transport.DEFAULT_SECURE = config.secure
         self.block_uids = tuple(block_uids)
         self.dht = dht or DHT(initial_peers=config.initial_peers, client_mode=True)
         self._owns_dht = dht is None
@@ -87,6 +91,15 @@ class RemoteSequenceManager:
             self.state.update_([i if (i and i.servers) else None for i in infos])
             self.addrs.update(addrs)
             self._last_update = time.monotonic()
+        try:
+            # secure transports verify certificate fingerprints against the
+            # DHT-announced peer ids at connect time
+            p2p = self.p2p
+            for pid, addr in addrs.items():
+                if len(addr) == 2:
+                    p2p.expect_peer(addr, pid)
+        except Exception:  # noqa: BLE001
+            pass
 
     def _maybe_update(self):
         if time.monotonic() - self._last_update > self.config.update_period or not self.state.spans_by_priority:

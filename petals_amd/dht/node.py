@@ -357,6 +357,12 @@ class DHT:
                 port=self._port,
                 client_mode=self._client_mode,
             )
+            if not self._client_mode:
+                # bootstrap/visible DHT nodes double as circuit relays for
+                # NAT'd servers (p2p/relay.py)
+                from petals_amd.p2p.relay import RelayHub
+
+                RelayHub(self.node.p2p)
 
         loop.run_until_complete(boot())
         self._ready.set()

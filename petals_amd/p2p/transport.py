@@ -17,9 +17,18 @@ Every RPC is a message stream keyed by `rid`: a unary call is a stream with one
 "req" inbound and one "end" outbound. Handlers receive an `RpcStream` with an
 async-iterator of inbound messages and `send()`/`close()` for outbound ones.
 
-Identity: every node owns a random 16-byte peer id (hex). There is no
-encryption layer (the reference relied on libp2p's); swarm deployments should
-front this with their own network fabric.
+Identity & encryption: by default a node's peer id is a random 16-byte hex
+string and traffic is plaintext (fastest for trusted fabrics). With
+``secure=True`` the node owns an ed25519 keypair (p2p/identity.py), its peer
+id is the certificate fingerprint, and every connection starts with a magic
+line then upgrades via STARTTLS (TLS 1.3): clients verify that the server's
+certificate fingerprint equals the peer id announced in the DHT, so a relay
+or on-path attacker cannot impersonate a server. (The reference got identity
++ encryption + relays from go-libp2p; see also relay.py for the circuit-relay
+fallback for NAT'd servers.)
+
+Wire handshake (dialer speaks first): b"PAMD1\n" = plaintext frames follow;
+b"PAMDS\n" = server replies b"OK\n" and both sides wrap in TLS, then frames.
 """
 
 from __future__ import annotations
@@ -369,25 +378,72 @@ class Connection:
         await self._shutdown("closed locally")
 
 
+MAGIC_PLAIN = b"PAMD1\n"
+MAGIC_TLS = b"PAMDS\n"
+
+# process-wide default for P2PNode(secure=None): set from ClientConfig.secure
+# or the PETALS_AMD_SECURE env var (a swarm runs either all-secure or all-plain)
+DEFAULT_SECURE: Optional[bool] = None
+
+
+async def _starttls(reader: asyncio.StreamReader, writer: asyncio.StreamWriter,
+                    ctx, server_side: bool) -> None:
+    """Upgrade an asyncio stream pair to TLS in place (py3.10 recipe: swap the
+    transport under the existing StreamReaderProtocol/StreamWriter)."""
+    loop = asyncio.get_event_loop()
+    transport = writer.transport
+    protocol = transport.get_protocol()
+    new_transport = await loop.start_tls(transport, protocol, ctx, server_side=server_side)
+    writer._transport = new_transport  # noqa: SLF001
+
+
 class P2PNode:
     """A peer: can listen for inbound RPCs and open outbound RPC streams.
 
     Replaces the reference's per-process `p2pd` daemon + hivemind.P2P wrapper.
     """
 
-    def __init__(self, peer_id: Optional[str] = None):
-        self.peer_id = peer_id or os.urandom(16).hex()
+    def __init__(self, peer_id: Optional[str] = None, *, identity=None, secure: Optional[bool] = None):
+        if secure is None:
+            secure = DEFAULT_SECURE if DEFAULT_SECURE is not None else bool(os.environ.get("PETALS_AMD_SECURE"))
+        self.secure = secure
+        self.identity = identity
+        if secure and self.identity is None:
+            from petals_amd.p2p.identity import NodeIdentity
+
+            self.identity = NodeIdentity()
+        self.peer_id = (self.identity.peer_id if self.identity is not None else None) or peer_id or os.urandom(16).hex()
         self.handlers: Dict[str, RpcHandler] = {}
         self._server: Optional[asyncio.base_events.Server] = None
         self.listen_addr: Optional[Tuple[str, int]] = None
-        self._conns: Dict[Tuple[str, int], Connection] = {}
+        self._conns: Dict[tuple, Connection] = {}
         self._inbound: List[Connection] = []
-        self._conn_locks: Dict[Tuple[str, int], asyncio.Lock] = {}
+        self._conn_locks: Dict[tuple, asyncio.Lock] = {}
         self._loop: Optional[asyncio.AbstractEventLoop] = None
         self._inproc_keys: List[Tuple[str, int]] = []
+        self._expected_peers: Dict[Tuple[str, int], str] = {}  # addr -> announced peer id
+        self._ssl_server_ctx = None
+        self._ssl_client_ctx = None
 
     def add_handler(self, rpc: str, handler: RpcHandler) -> None:
         self.handlers[rpc] = handler
+
+    def expect_peer(self, addr, peer_id: str) -> None:
+        """Record the DHT-announced peer id for an address; secure connects
+        verify the certificate fingerprint against it."""
+        self._expected_peers[(addr[0], int(addr[1]))] = peer_id
+
+    def _server_ctx(self):
+        if self._ssl_server_ctx is None and self.identity is not None:
+            self._ssl_server_ctx = self.identity.server_ssl_context()
+        return self._ssl_server_ctx
+
+    def _client_ctx(self):
+        if self._ssl_client_ctx is None:
+            from petals_amd.p2p.identity import NodeIdentity
+
+            self._ssl_client_ctx = NodeIdentity.client_ssl_context()
+        return self._ssl_client_ctx
 
     async def listen(self, host: str = "127.0.0.1", port: int = 0) -> Tuple[str, int]:
         self._server = await asyncio.start_server(self._on_inbound, host=host, port=port)
@@ -407,6 +463,49 @@ class P2PNode:
             # without NODELAY on the accepted socket, small RPC responses sit in
             # Nagle's buffer against the peer's delayed ACK (~40 ms per step)
             sock.setsockopt(socket.IPPROTO_TCP, socket.TCP_NODELAY, 1)
+        await self.accept_stream_pair(reader, writer)
+
+    async def accept_stream_pair(self, reader: asyncio.StreamReader, writer: asyncio.StreamWriter):
+        """Accepting side of the transport handshake — used for direct inbound
+        TCP and for relay tunnels (p2p/relay.py) alike."""
+        try:
+            magic = await asyncio.wait_for(reader.readexactly(len(MAGIC_PLAIN)), 30.0)
+            if magic == MAGIC_TLS:
+                ctx = self._server_ctx()
+                if ctx is None:
+                    writer.close()
+                    return
+                writer.write(b"OK\n")
+                await writer.drain()
+                await _starttls(reader, writer, ctx, server_side=True)
+            elif magic == MAGIC_PLAIN:
+                if self.secure:
+                    logger.warning("refusing plaintext connection (secure mode)")
+                    writer.close()
+                    return
+            elif magic == b"PAMDR\n":  # relay: client leg asking for a tunnel
+                hub = getattr(self, "relay_hub", None)
+                target = (await asyncio.wait_for(reader.readline(), 30.0)).decode().strip()
+                if hub is None or not target:
+                    writer.close()
+                    return
+                await hub.open_tunnel(target, reader, writer)
+                return
+            elif magic == b"PAMDT\n":  # relay: target leg dialing back
+                hub = getattr(self, "relay_hub", None)
+                token = (await asyncio.wait_for(reader.readline(), 30.0)).decode().strip()
+                if hub is None or not hub.accept_tunnel(token, reader, writer):
+                    writer.close()
+                return
+            else:
+                writer.close()
+                return
+        except Exception:  # noqa: BLE001
+            try:
+                writer.close()
+            except Exception:  # noqa: BLE001
+                pass
+            return
         conn = Connection(self, reader, writer)
         self._inbound.append(conn)
         conn.start()
@@ -418,21 +517,67 @@ class P2PNode:
         if conn in self._inbound:
             self._inbound.remove(conn)
 
-    async def connect(self, addr: Tuple[str, int], timeout: float = 10.0) -> Connection:
-        addr = (addr[0], int(addr[1]))
+    async def connect(self, addr, timeout: float = 10.0) -> Connection:
+        addr = self._norm_addr(addr)
         lock = self._conn_locks.setdefault(addr, asyncio.Lock())
         async with lock:
             conn = self._conns.get(addr)
             if conn is not None and not conn.is_closed:
                 return conn
-            reader, writer = await asyncio.wait_for(asyncio.open_connection(*addr), timeout)
-            sock = writer.get_extra_info("socket")
-            if sock is not None:
-                sock.setsockopt(socket.IPPROTO_TCP, socket.TCP_NODELAY, 1)
-            conn = Connection(self, reader, writer)
-            conn.start()
+            if len(addr) >= 4 and addr[2] == "relay":
+                from petals_amd.p2p.relay import connect_via_relay
+
+                reader, writer = await connect_via_relay(self, (addr[0], addr[1]), addr[3], timeout)
+            else:
+                reader, writer = await asyncio.wait_for(asyncio.open_connection(addr[0], addr[1]), timeout)
+                sock = writer.get_extra_info("socket")
+                if sock is not None:
+                    sock.setsockopt(socket.IPPROTO_TCP, socket.TCP_NODELAY, 1)
+            conn = await self._handshake_outbound(addr, reader, writer)
             self._conns[addr] = conn
             return conn
+
+    @staticmethod
+    def _norm_addr(addr) -> tuple:
+        if len(addr) >= 4 and addr[2] == "relay":
+            return (addr[0], int(addr[1]), "relay", addr[3])
+        return (addr[0], int(addr[1]))
+
+    async def _handshake_outbound(self, addr, reader, writer) -> Connection:
+        """Dialer side of the magic/STARTTLS exchange + peer verification."""
+        expected = None
+        if len(addr) >= 4 and addr[2] == "relay":
+            expected = addr[3]  # the relay target IS the peer we must reach
+        else:
+            expected = self._expected_peers.get((addr[0], addr[1]))
+        if self.secure:
+            writer.write(MAGIC_TLS)
+            await writer.drain()
+            ok = await asyncio.wait_for(reader.readexactly(3), 30.0)
+            if ok != b"OK\n":
+                writer.close()
+                raise RpcError(f"peer at {addr} refused STARTTLS")
+            await _starttls(reader, writer, self._client_ctx(), server_side=False)
+            ssl_obj = writer.transport.get_extra_info("ssl_object")
+            der = ssl_obj.getpeercert(binary_form=True) if ssl_obj else None
+            if der is None:
+                writer.close()
+                raise RpcError("TLS peer presented no certificate")
+            from petals_amd.p2p.identity import cert_fingerprint
+
+            actual = cert_fingerprint(der)
+            if expected is not None and actual != expected:
+                writer.close()
+                raise RpcError(
+                    f"peer identity mismatch at {addr}: announced {expected[:8]}, "
+                    f"certificate fingerprint {actual[:8]} — possible impersonation"
+                )
+        else:
+            writer.write(MAGIC_PLAIN)
+            await writer.drain()
+        conn = Connection(self, reader, writer)
+        conn.start()
+        return conn
 
     async def open_stream(
         self,
@@ -447,8 +592,8 @@ class P2PNode:
 
         With ``end=True`` the request also closes our outbound side (unary
         request)."""
-        addr = (addr[0], int(addr[1]))
-        target = _INPROC_NODES.get(addr) if _inproc_enabled() else None
+        addr = self._norm_addr(addr)
+        target = _INPROC_NODES.get(addr) if _inproc_enabled() and len(addr) == 2 else None
         if target is not None and target._loop is not None and not target._loop.is_closed():
             return self._open_inproc(target, rpc, request, end)
         conn = await self.connect(addr, timeout)

@@ -484,7 +484,7 @@ class TransformerConnectionHandler:
                                 try:
                                     ticket, commit_cb, abort_cb = self.mesh.send_deferred(mdst)
                                     push_meta = {
-                                        "session_id": next_servers[0][2],
+                                        "session_id": next_servers[0]["session_id"],
                                         "step_id": step_meta.get("step_id"),
                                         "next_servers": next_servers[1:],
                                         "start_from_position": step_start_position,
@@ -497,11 +497,11 @@ class TransformerConnectionHandler:
                                             "dtype": _dtype_str(out_dtype),
                                         },
                                     }
-                                    host, port = next_servers[0][0], next_servers[0][1]
                                     ack_task = asyncio.ensure_future(
                                         asyncio.wait_for(
                                             self.p2p.call_unary(
-                                                (host, port), "petals.rpc_push", RpcMessage(meta=push_meta)
+                                                tuple(next_servers[0]["addr"]), "petals.rpc_push",
+                                                RpcMessage(meta=push_meta),
                                             ),
                                             timeout=10.0,
                                         )
@@ -618,12 +618,11 @@ class TransformerConnectionHandler:
 
     def _mesh_dst(self, next_server_entry) -> Optional[int]:
         """Next-server mesh rank iff it shares a usable mesh with this server.
-        Entries: [host, port, session_id, start, end, mesh_id, mesh_rank]."""
+        Entries: {"addr", "session_id", "start", "end", "mesh_id", "mesh_rank"}."""
         if self.mesh is None or not self.mesh.is_usable or self.mesh.device is None:
             return None
-        if len(next_server_entry) < 7:
-            return None
-        mesh_id, mesh_rank = next_server_entry[5], next_server_entry[6]
+        mesh_id = next_server_entry.get("mesh_id")
+        mesh_rank = next_server_entry.get("mesh_rank")
         if mesh_id != self.mesh.mesh_id or mesh_rank is None or mesh_rank == self.mesh.rank:
             return None
         return int(mesh_rank)
@@ -671,23 +670,24 @@ class TransformerConnectionHandler:
     async def _push_outputs(
         self, output: torch.Tensor, step_meta: Dict[str, Any], next_servers, step_start_position: int
     ) -> bool:
-        """Push this step's output into the next server's session. next_servers:
-        [[host, port, session_id, start_block, end_block, mesh_id, mesh_rank],
-        ...]; we contact the first entry. start_from_position propagates down
-        the chain so rollbacks (speculative decoding) rewind every span's
-        cache. Co-located next servers get the activation over RCCL/xGMI (the
-        TCP message then carries only metadata)."""
+        """Push this step's output into the next server's session. next_servers
+        entries: {"addr": [host, port(, "relay", peer)], "session_id", "start",
+        "end", "mesh_id", "mesh_rank"}; we contact the first entry.
+        start_from_position propagates down the chain so rollbacks
+        (speculative decoding) rewind every span's cache. Co-located next
+        servers get the activation over RCCL/xGMI (the TCP message then
+        carries only metadata)."""
         try:
-            host, port, next_session_id = next_servers[0][0], next_servers[0][1], next_servers[0][2]
+            entry = next_servers[0]
             meta = {
-                "session_id": next_session_id,
+                "session_id": entry["session_id"],
                 "step_id": step_meta.get("step_id"),
                 "next_servers": next_servers[1:],
                 "start_from_position": step_start_position,
                 "output_via_mesh": step_meta.get("output_via_mesh"),
             }
             tensors = [output]
-            mesh_dst = self._mesh_dst(next_servers[0])
+            mesh_dst = self._mesh_dst(entry)
             if mesh_dst is not None and output.device.type == self.mesh.device.type:
                 _fut, ticket = self.mesh.send(output, dst=mesh_dst)
                 meta["tensors_via_mesh"] = self._mesh_desc(output, ticket)
@@ -696,7 +696,7 @@ class TransformerConnectionHandler:
                 output = output.cpu()
                 tensors = [output]
             await asyncio.wait_for(
-                self.p2p.call_unary((host, port), "petals.rpc_push", RpcMessage(meta=meta, tensors=tensors)),
+                self.p2p.call_unary(tuple(entry["addr"]), "petals.rpc_push", RpcMessage(meta=meta, tensors=tensors)),
                 timeout=10.0,
             )
             return True

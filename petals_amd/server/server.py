@@ -69,6 +69,10 @@ class Server:
         announce_host: Optional[str] = None,
         skip_reachability_check: bool = False,
         mesh=None,
+        secure: Optional[bool] = None,
+        identity_path: Optional[str] = None,
+        use_relay: bool = True,
+        force_relay: bool = False,
     ):
         self.config = load_model_config(model_name_or_dir)
         self.model_name_or_dir = model_name_or_dir
@@ -129,6 +133,12 @@ class Server:
         self.server_info: Optional[ServerInfo] = None
         self.listen_addr: Optional[Tuple[str, int]] = None
         self._ping_agg = None  # PingAggregator for next-server RTT gossip
+        self.secure = secure
+        self.identity_path = identity_path
+        self.use_relay = use_relay
+        self.force_relay = force_relay
+        self._relayed = False
+        self._relay_addr: Optional[Tuple[str, int]] = None
 
     # -------------------------------------------------------------- sizing
 
@@ -201,7 +211,12 @@ class Server:
             loop.close()
 
     async def _amain(self):
-        self.p2p = P2PNode()
+        identity = None
+        if self.identity_path or self.secure:
+            from petals_amd.p2p.identity import NodeIdentity
+
+            identity = NodeIdentity(self.identity_path)
+        self.p2p = P2PNode(identity=identity, secure=self.secure)
         await self.p2p.listen(host=self.host, port=self.port)
         # the address other peers should dial (never announce 0.0.0.0)
         host = self.announce_host or self.host
@@ -211,11 +226,27 @@ class Server:
             host = socket.gethostbyname(socket.gethostname())
         self.listen_addr = (host, self.p2p.listen_addr[1])
         self.dht_node = await DHTNode.create(initial_peers=self.initial_peers, p2p=self.p2p)
-        from petals_amd.server.reachability import ReachabilityProtocol, validate_reachability
+        from petals_amd.p2p.relay import RelayClient, RelayHub
+        from petals_amd.server.reachability import ReachabilityProtocol, check_direct_reachability
 
         ReachabilityProtocol(self.p2p)
-        if not self.skip_reachability_check and self.initial_peers:
-            await validate_reachability(self.p2p, self.listen_addr, self.initial_peers, wait_time=30.0)
+        RelayHub(self.p2p)  # any reachable server can relay for NAT'd peers
+        unreachable = self.force_relay
+        if not unreachable and not self.skip_reachability_check and self.initial_peers:
+            ok = await check_direct_reachability(self.p2p, self.listen_addr, self.initial_peers)
+            unreachable = ok is False
+        if unreachable:
+            if not (self.use_relay and self.initial_peers):
+                raise RuntimeError(
+                    f"server at {self.listen_addr} is not reachable by the swarm and relays are disabled"
+                )
+            self._relay_addr = tuple(self.initial_peers[0])
+            await RelayClient(self.p2p, self._relay_addr).start()
+            self._relayed = True
+            logger.warning(
+                "server is not directly reachable; serving via relay %s (throughput penalized x0.2)",
+                self._relay_addr,
+            )
 
         self.runtime = PriorityRuntime(self.device).start()
         while not self._stop.is_set():
@@ -248,6 +279,10 @@ class Server:
         else:
             throughput = float(self._throughput_setting)
             inference_rps = forward_rps = network_rps = throughput
+        if self._relayed:
+            # relayed traffic crosses the relay's uplink twice (parity:
+            # reference server/throughput.py:96-107)
+            throughput *= 0.2
 
         cache_bytes = int(
             2
@@ -274,6 +309,7 @@ class Server:
             adapters=tuple(os.path.basename(os.path.normpath(a)) for a in self.adapters),
             mesh_id=self.mesh.mesh_id if self.mesh is not None else None,
             mesh_rank=self.mesh.rank if self.mesh is not None else None,
+            using_relay=self._relayed,
         )
         await self._announce()
 
@@ -421,7 +457,11 @@ class Server:
         ]
         if not uids:
             return
-        value = {"info": self.server_info.to_dict(), "addr": list(self.listen_addr)}
+        if self._relayed:
+            announced_addr = [self._relay_addr[0], self._relay_addr[1], "relay", self.p2p.peer_id]
+        else:
+            announced_addr = list(self.listen_addr)
+        value = {"info": self.server_info.to_dict(), "addr": announced_addr}
         entries = [(uid, self.p2p.peer_id, value, get_dht_time() + self.expiration) for uid in uids]
         try:
             await self.dht_node.store_many(entries)

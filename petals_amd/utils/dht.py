@@ -54,7 +54,9 @@ def get_remote_module_infos(
         for peer_id, (value, _expiration) in (found.get(uid) or {}).items():
             try:
                 info = ServerInfo.from_dict(value["info"])
-                addr = (value["addr"][0], int(value["addr"][1]))
+                # addr may be [host, port] or [relay_host, relay_port, "relay",
+                # target_peer_id] for NAT'd servers behind a circuit relay
+                addr = (value["addr"][0], int(value["addr"][1]), *value["addr"][2:])
             except (KeyError, TypeError, ValueError):
                 continue
             if active_only and info.state != 2:  # ServerState.ONLINE
