@@ -114,6 +114,7 @@ class TransformerConnectionHandler:
         p2p: Optional[P2PNode] = None,
         adapters: Sequence[str] = (),
         mesh=None,
+        default_compression: str = "none",
     ):
         self.backends = backends
         self.memory_cache = memory_cache
@@ -127,6 +128,7 @@ class TransformerConnectionHandler:
         self.adapters = tuple(adapters)
         self.p2p = p2p
         self.mesh = mesh  # LocalMesh: RCCL/xGMI hand-off tier for co-located spans
+        self.default_compression = default_compression
         self._sessions: Dict[str, _Session] = {}
 
     def register(self, p2p: P2PNode) -> None:
@@ -213,9 +215,8 @@ class TransformerConnectionHandler:
         )
         return [out]
 
-    @staticmethod
-    def _out_compressions(meta, n):
-        c = meta.get("output_compression")
+    def _out_compressions(self, meta, n):
+        c = meta.get("output_compression") or self.default_compression
         return [c] * n if c and c != "none" else None
 
     async def rpc_forward(self, request: RpcMessage, stream: RpcStream) -> None:

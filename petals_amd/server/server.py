@@ -73,6 +73,17 @@ class Server:
         identity_path: Optional[str] = None,
         use_relay: bool = True,
         force_relay: bool = False,
+        max_chunk_size_bytes: int = 256 * 1024 * 1024,
+        max_alloc_timeout: float = 600.0,
+        request_timeout: float = 3 * 60.0,
+        session_timeout: float = 30 * 60.0,
+        step_timeout: float = 5 * 60.0,
+        compression: str = "none",
+        stats_report_interval: Optional[float] = None,
+        cache_dir: Optional[str] = None,
+        max_disk_space: Optional[int] = None,
+        tensor_parallel_ranks: int = 1,
+        tp_group=None,
     ):
         self.config = load_model_config(model_name_or_dir)
         self.model_name_or_dir = model_name_or_dir
@@ -139,6 +150,19 @@ class Server:
         self.force_relay = force_relay
         self._relayed = False
         self._relay_addr: Optional[Tuple[str, int]] = None
+        self.max_chunk_size_bytes = max_chunk_size_bytes
+        self.max_alloc_timeout = max_alloc_timeout
+        self.request_timeout = request_timeout
+        self.session_timeout = session_timeout
+        self.step_timeout = step_timeout
+        self.compression = compression
+        self.stats_report_interval = stats_report_interval
+        self.cache_dir = cache_dir
+        self.max_disk_space = max_disk_space
+        # intra-server tensor parallelism (parallel/tp.py): this process is one
+        # of `tensor_parallel_ranks` ranks sharding every served block
+        self.tensor_parallel_ranks = int(tensor_parallel_ranks)
+        self.tp_group = tp_group
 
     # -------------------------------------------------------------- sizing
 
@@ -292,7 +316,7 @@ class Server:
             * get_size_in_bytes(self.torch_dtype)
             * len(block_indices)
         )
-        self.memory_cache = MemoryCache(cache_bytes, self.device)
+        self.memory_cache = MemoryCache(cache_bytes, self.device, alloc_timeout=self.max_alloc_timeout)
 
         self.server_info = ServerInfo(
             state=ServerState.JOINING,
@@ -331,7 +355,8 @@ class Server:
                 if ad is not None:
                     add_adapter_to_block(block, ad)
             self.backends[uid] = TransformerBackend(
-                uid, block, config=self.config, memory_cache=self.memory_cache, dtype=self.torch_dtype
+                uid, block, config=self.config, memory_cache=self.memory_cache, dtype=self.torch_dtype,
+                max_chunk_size_bytes=self.max_chunk_size_bytes,
             )
 
         self.handler = TransformerConnectionHandler(
@@ -342,6 +367,10 @@ class Server:
             p2p=self.p2p,
             adapters=tuple(os.path.basename(os.path.normpath(a)) for a in self.adapters),
             mesh=self.mesh,
+            request_timeout=self.request_timeout,
+            session_timeout=self.session_timeout,
+            step_timeout=self.step_timeout,
+            default_compression=self.compression,
         )
         self.handler.register(self.p2p)
 
@@ -361,8 +390,21 @@ class Server:
         # --- announce + health + rebalance loop
         next_balance_check = time.monotonic() + random.random() * 2 * self.mean_balance_check_period
         last_announce = time.monotonic()
+        last_stats = time.monotonic()
         try:
             while not self._stop.is_set():
+                if (
+                    self.stats_report_interval
+                    and time.monotonic() - last_stats >= self.stats_report_interval
+                ):
+                    st = self.runtime.stats
+                    logger.info(
+                        "runtime: %d tasks, %.1f s busy; cache: %.0f/%.0f MiB free",
+                        st["tasks"], st["busy_time"],
+                        self.memory_cache.bytes_left / (1 << 20),
+                        self.memory_cache.max_size_bytes / (1 << 20),
+                    )
+                    last_stats = time.monotonic()
                 if time.monotonic() - last_announce >= self.update_period:
                     try:
                         await self._ping_next_servers()
