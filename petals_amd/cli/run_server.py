@@ -139,6 +139,41 @@ def main(argv=None):
         import os as _os
 
         _os.environ["PETALS_AMD_CACHE"] = args.cache_dir
+
+    if args.tensor_parallel_ranks > 1:
+        # launched under torchrun, one process per GPU: rank 0 is the swarm
+        # Server; ranks 1.. execute their block shards in lockstep
+        import datetime
+        import os as _os
+
+        import torch
+        import torch.distributed as dist
+
+        if not dist.is_initialized():
+            backend = "nccl" if torch.cuda.is_available() else "gloo"
+            dist.init_process_group(backend, timeout=datetime.timedelta(seconds=600))
+        rank, world = dist.get_rank(), dist.get_world_size()
+        assert world == args.tensor_parallel_ranks, (
+            f"torchrun world size {world} != --tensor_parallel_ranks {args.tensor_parallel_ranks}"
+        )
+        if torch.cuda.is_available():
+            torch.cuda.set_device(int(_os.environ.get("LOCAL_RANK", rank)))
+        if rank != 0:
+            from petals_amd.constants import DTYPE_MAP
+            from petals_amd.models.config_base import load_model_config
+            from petals_amd.parallel.tp import TPShadowWorker
+
+            config = load_model_config(model)
+            device = torch.device("cuda", torch.cuda.current_device()) if torch.cuda.is_available() else torch.device("cpu")
+            dtype = DTYPE_MAP[args.torch_dtype]
+            if dtype is None:
+                dtype = torch.bfloat16 if device.type == "cuda" else torch.float32
+            TPShadowWorker(
+                model, config, device=device, torch_dtype=dtype, quant_type=args.quant_type,
+                group=None, rank=rank, world=world,
+            ).serve_forever()
+            return
+
     from petals_amd.server.server import Server
 
     server = Server(

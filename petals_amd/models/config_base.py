@@ -48,10 +48,13 @@ class ModelConfig:
     dht_prefix: str = ""
     # block_prefix is the state-dict path of the block list, e.g. "model.layers"
     block_prefix: str = "model.layers"
+    # explicit head_dim (tensor-parallel SHARD configs keep hidden_size while
+    # dividing the head count, so the derived value would be wrong)
+    head_dim_override: Optional[int] = None
 
     @property
     def head_dim(self) -> int:
-        return self.hidden_size // self.num_attention_heads
+        return self.head_dim_override or (self.hidden_size // self.num_attention_heads)
 
     @property
     def n_kv_heads(self) -> int:

@@ -158,11 +158,19 @@ class _FastWeight:
 class LlamaFastPath:
     graph_safe = True  # whole-span hipGraph capture is valid for this block
 
-    def __init__(self, block, hip_ops, quant: str = "none"):
+    def __init__(self, block, hip_ops, quant: str = "none", tp_world: int = 1, tp_group=None):
         cfg = block.config
         self.hip = hip_ops
         self.cfg = cfg
         self.quant = quant
+        # tensor parallelism (parallel/tp.py): cfg already IS the shard
+        # geometry; row-parallel outputs (o/down) skip the fused residual
+        # epilogue and are all-reduced before the residual add
+        self.tp_world = tp_world
+        self.tp_group = tp_group
+        if tp_world > 1:
+            # RCCL collectives inside hipGraph capture are not validated yet
+            self.graph_safe = False
         self.hd = cfg.head_dim
         self.qh = cfg.num_attention_heads
         self.kh = cfg.n_kv_heads
@@ -211,6 +219,7 @@ class LlamaFastPath:
     def _mlp_dense(self, xn2, adapter, autograd: bool):
         """MLP on [B, S, H] inputs via rocBLAS matmuls on the transposed
         (possibly NF4-dequantized) weights; differentiable when autograd."""
+        xn2 = self._tp_copy(xn2)
         gateup = torch.matmul(xn2, self.wgateup_t.dense())
         inter = self.wgateup_t.shape[1] // 2
         gate, up = gateup[..., :inter], gateup[..., inter:]
@@ -284,8 +293,12 @@ class LlamaFastPath:
         attn = self.hip.attn_decode_fused(
             q.contiguous(), k_cache[:B], v_cache[:B], kv_len, self.gq, 0,
             self._empty_f32, self._empty_f32, self.scale,
-        )  # [B, H] f32
-        if adapter is not None:
+        )  # [B, qh*hd] f32 (the shard's heads under TP)
+        if self.tp_world > 1:
+            part = self.wo_t.gemv(attn, ws, None, _EPI_PLAIN_F32)  # [B, H] f32 partial
+            self._tp_allreduce_(part)
+            h2 = (h.float() + part).to(torch.bfloat16)
+        elif adapter is not None:
             d = adapter.delta("o", attn)
             h2 = self.wo_t.gemv(attn, ws, h, _EPI_RESIDUAL_BF16)  # [B, H] bf16
             if d is not None:
@@ -296,10 +309,38 @@ class LlamaFastPath:
         h3 = self._mlp_decode(xn2, h2, ws, adapter)
         return h3.view(B, 1, H)
 
+    def _tp_copy(self, x: torch.Tensor) -> torch.Tensor:
+        """Column-parallel input boundary (identity fwd, grad all-reduce bwd)."""
+        if self.tp_world <= 1:
+            return x
+        from petals_amd.parallel.tp import copy_to_tp
+
+        return copy_to_tp(x, self.tp_group)
+
+    def _tp_allreduce_(self, t: torch.Tensor) -> None:
+        """In-place SUM across TP ranks (decode hot path; overridable in tests)."""
+        import torch.distributed as dist
+
+        dist.all_reduce(t, op=dist.ReduceOp.SUM, group=self.tp_group)
+
+    def _tp_reduce(self, x: torch.Tensor) -> torch.Tensor:
+        """Sum row-parallel partials across TP ranks (autograd-aware)."""
+        if self.tp_world <= 1:
+            return x
+        from petals_amd.parallel.tp import reduce_from_tp
+
+        return reduce_from_tp(x, self.tp_group)
+
     def _max_gemv_out(self) -> int:
         return max(self.wgateup_t.shape[1], self.wqkv_t.shape[1], self.wo_t.shape[1])
 
     def _mlp_decode(self, xn2, h2, ws, adapter):
+        if self.tp_world > 1:
+            assert adapter is None, "LoRA on TP shards is served via the generic path"
+            act = self.wgateup_t.gemv(xn2, ws, None, _EPI_SWIGLU_F32)  # [B, I/world] f32
+            part = self.wdown_t.gemv(act, ws, None, _EPI_PLAIN_F32)  # [B, H] f32 partial
+            self._tp_allreduce_(part)
+            return (h2.float() + part).to(torch.bfloat16)
         if adapter is None:
             act = self.wgateup_t.gemv(xn2, ws, None, _EPI_SWIGLU_F32)  # [B, I] f32
             return self.wdown_t.gemv(act, ws, h2, _EPI_RESIDUAL_BF16)
@@ -346,6 +387,7 @@ class LlamaFastPath:
         end = prefix_length + S
         self._ensure_rope(end)
         xn = reference.rms_norm(hidden, self.ln1_w, self.eps)
+        xn = self._tp_copy(xn)  # backward: all-reduce the shard-partial grads
         qkv = torch.matmul(xn, self.wqkv_t.dense())
         if adapter is not None:
             qkv = qkv + self._qkv_adapter_delta(xn, adapter)
@@ -356,14 +398,14 @@ class LlamaFastPath:
         q, k = reference.apply_rope(q, k, self.rope_cos, self.rope_sin, pos)
         attn = reference.attention(q, k, v, causal=True)
         attn = attn.transpose(1, 2).reshape(B, S, self.qh * self.hd).to(hidden.dtype)
-        o = torch.matmul(attn, self.wo_t.dense())
+        o = self._tp_reduce(torch.matmul(attn, self.wo_t.dense()))
         if adapter is not None:
             d = adapter.delta("o", attn)
             if d is not None:
                 o = o + d
         h2 = hidden + o
         xn2 = reference.rms_norm(h2, self.ln2_w, self.eps)
-        return h2 + self._mlp_dense(xn2, adapter, autograd=True).to(h2.dtype)
+        return h2 + self._tp_reduce(self._mlp_dense(xn2, adapter, autograd=True)).to(h2.dtype)
 
     # ------------------------------------------------------------ prefill
 
@@ -401,14 +443,14 @@ class LlamaFastPath:
                 q.contiguous(), k.contiguous(), v.contiguous(), S, 0, self.scale, True
             )
         attn = attn.transpose(1, 2).reshape(B, S, self.qh * self.hd).to(torch.bfloat16)
-        o = torch.matmul(attn, self.wo_t.dense())
+        o = self._tp_reduce(torch.matmul(attn, self.wo_t.dense()))
         if adapter is not None:
             d = adapter.delta("o", attn)
             if d is not None:
                 o = o + d
         h2 = hidden + o
         xn2 = self.hip.rms_norm(h2, self.ln2_w, self.eps)
-        return h2 + self._mlp_dense(xn2, adapter, autograd=False).to(h2.dtype)
+        return h2 + self._tp_reduce(self._mlp_dense(xn2, adapter, autograd=False)).to(h2.dtype)
 
 
 # ---------------------------------------------------------------------------

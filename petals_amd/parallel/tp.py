@@ -6,7 +6,17 @@ every GPU is its own process (one rank per GPU, torch.distributed over
 RCCL/xGMI), a block is sharded column-parallel for QKV/gate/up (split by
 heads / intermediate columns) and row-parallel for O/down, with one
 all-reduce(SUM) after attention and one after the MLP — reduce traffic is
-2 x hidden per token per block, bucketed by RCCL over the 7-link xGMI mesh.
+2 x hidden per token per block over the 7-link xGMI mesh.
+
+The fused MI355X decode path applies unchanged to a shard: a TP block exposes
+its SHARD geometry as `block.config`, so `optimize_for_inference()` builds the
+same LlamaFastPath (NF4/int8 gemv, MFMA flash decode, fused epilogues) over
+the shard weights, with the residual-fused epilogues split at the two reduce
+points (partial -> all_reduce -> +residual).
+
+Training: column-parallel layers consume the replicated input (grad wrt input
+needs an all-reduce in backward — `copy_to_tp`), row-parallel partial sums are
+reduced in forward with identity backward (`reduce_from_tp`).
 
 KV caches are per-rank shards ([batch, kv_heads/world, len, head_dim]) — the
 reference's PerDeviceTensors equivalent. Requires kv_heads % world == 0.
@@ -17,6 +27,7 @@ reference's `--tensor_parallel_devices cpu cpu` CI servers.
 
 from __future__ import annotations
 
+import dataclasses
 from typing import Optional, Tuple
 
 import torch
@@ -32,6 +43,58 @@ def _all_reduce(t: torch.Tensor, group=None) -> torch.Tensor:
     if dist.is_initialized() and dist.get_world_size(group) > 1:
         dist.all_reduce(t, op=dist.ReduceOp.SUM, group=group)
     return t
+
+
+class _CopyToTP(torch.autograd.Function):
+    """Identity forward; all-reduce(SUM) backward — wraps the replicated input
+    of column-parallel layers so input grads combine every rank's shard."""
+
+    @staticmethod
+    def forward(ctx, x, group):
+        ctx.group = group
+        return x
+
+    @staticmethod
+    def backward(ctx, grad):
+        return _all_reduce(grad.contiguous(), ctx.group), None
+
+
+class _ReduceFromTP(torch.autograd.Function):
+    """All-reduce(SUM) forward (combine row-parallel partials); identity
+    backward."""
+
+    @staticmethod
+    def forward(ctx, x, group):
+        return _all_reduce(x.contiguous(), group)
+
+    @staticmethod
+    def backward(ctx, grad):
+        return grad, None
+
+
+def copy_to_tp(x, group=None):
+    return _CopyToTP.apply(x, group) if torch.is_grad_enabled() and x.requires_grad else x
+
+
+def reduce_from_tp(x, group=None):
+    if torch.is_grad_enabled() and x.requires_grad:
+        return _ReduceFromTP.apply(x, group)
+    return _all_reduce(x, group)
+
+
+def shard_llama_config(config: LlamaConfig, world: int) -> LlamaConfig:
+    """This rank's shard geometry as a plain config (the fused fast path reads
+    its shapes from here)."""
+    assert config.num_attention_heads % world == 0, "q heads must divide tp world"
+    assert config.n_kv_heads % world == 0, "kv heads must divide tp world"
+    assert config.intermediate_size % world == 0
+    return dataclasses.replace(
+        config,
+        num_attention_heads=config.num_attention_heads // world,
+        num_key_value_heads=config.n_kv_heads // world,
+        intermediate_size=config.intermediate_size // world,
+        head_dim_override=config.head_dim,
+    )
 
 
 class TPLlamaAttention(nn.Module):
@@ -52,6 +115,7 @@ class TPLlamaAttention(nn.Module):
         self.o_proj = nn.Linear(self.num_heads * self.head_dim, h, bias=False)  # row-parallel: bias once
         self.rope_cos = None
         self.rope_sin = None
+        self.scale = None  # default 1/sqrt(head_dim) inside ops
 
     def _ensure_rope(self, needed: int, device):
         if self.rope_cos is None or self.rope_cos.shape[0] < needed or self.rope_cos.device != torch.device(device):
@@ -62,6 +126,10 @@ class TPLlamaAttention(nn.Module):
             self.rope_cos, self.rope_sin = cos.to(device), sin.to(device)
 
     def forward(self, x, kv_cache=None, prefix_length: int = 0):
+        # copy boundary: grads of the replicated input through THIS rank's
+        # column shard are partial -> all-reduced in backward. (Residual and
+        # layernorm paths are replicated and must NOT be reduced.)
+        x = copy_to_tp(x, self.group)
         b, q_len, _ = x.shape
         q = self.q_proj(x).view(b, q_len, self.num_heads, self.head_dim).transpose(1, 2)
         k = self.k_proj(x).view(b, q_len, self.num_kv_heads, self.head_dim).transpose(1, 2)
@@ -80,7 +148,7 @@ class TPLlamaAttention(nn.Module):
             attn = ops.attention(q, k, v, causal=True)
         attn = attn.transpose(1, 2).reshape(b, q_len, self.num_heads * self.head_dim)
         out = self.o_proj(attn)  # partial sum
-        return _all_reduce(out, self.group)
+        return reduce_from_tp(out, self.group)
 
 
 class TPLlamaMLP(nn.Module):
@@ -94,12 +162,18 @@ class TPLlamaMLP(nn.Module):
         self.down_proj = nn.Linear(inter, config.hidden_size, bias=False)
 
     def forward(self, x):
+        x = copy_to_tp(x, self.group)
         out = self.down_proj(ops.swiglu(self.gate_proj(x), self.up_proj(x)))
-        return _all_reduce(out, self.group)
+        return reduce_from_tp(out, self.group)
 
 
 class TPLlamaBlock(nn.Module):
-    """One rank's shard of a Llama decoder block."""
+    """One rank's shard of a Llama decoder block.
+
+    `self.config` reports the SHARD geometry, so `optimize_for_inference()`
+    can wrap the shard in the SAME fused MI355X path (LlamaFastPath: NF4/int8
+    gemv, MFMA flash decode) as an unsharded block — with the two
+    residual-fused epilogues split at the all-reduce points."""
 
     def __init__(self, config: LlamaConfig, layer_idx: int = 0, rank: Optional[int] = None,
                  world: Optional[int] = None, group=None):
@@ -108,13 +182,30 @@ class TPLlamaBlock(nn.Module):
             rank = dist.get_rank(group) if dist.is_initialized() else 0
         if world is None:
             world = dist.get_world_size(group) if dist.is_initialized() else 1
-        self.config = config
+        self.full_config = config
+        self.config = shard_llama_config(config, world)  # shard geometry
         self.layer_idx = layer_idx
         self.rank, self.world = rank, world
+        self.tp_group = group
         self.self_attn = TPLlamaAttention(config, rank, world, group)
         self.mlp = TPLlamaMLP(config, rank, world, group)
         self.input_layernorm = RMSNorm(config.hidden_size, eps=config.layer_norm_eps)
         self.post_attention_layernorm = RMSNorm(config.hidden_size, eps=config.layer_norm_eps)
+
+    _fast = None  # LlamaFastPath over the shard, after optimize_for_inference()
+
+    def optimize_for_inference(self, quant: str = "none") -> "TPLlamaBlock":
+        from petals_amd import ops as _ops
+        from petals_amd.ops.fused_decode import LlamaFastPath
+
+        hip = _ops._load_hip_ops()
+        if hip is None:
+            raise RuntimeError(
+                f"cannot optimize block for MI355X: HIP extension missing ({_ops._hip_import_error!r})"
+            )
+        assert next(self.parameters()).device.type == "cuda", "optimize_for_inference needs a GPU block"
+        self._fast = LlamaFastPath(self, hip, quant=quant, tp_world=self.world, tp_group=self.tp_group)
+        return self
 
     def load_from_full_state_dict(self, sd: dict) -> None:
         """Slice a FULL block state dict into this rank's shard (column split
@@ -142,6 +233,16 @@ class TPLlamaBlock(nn.Module):
         self.post_attention_layernorm.weight.data.copy_(sd["post_attention_layernorm.weight"])
 
     def forward(self, hidden_states, kv_cache=None, prefix_length: int = 0, ctx=None):
+        if self._fast is not None:
+            if torch.is_grad_enabled() and hidden_states.requires_grad:
+                assert kv_cache is None, "training forward does not use the KV cache"
+                return self._fast.forward_autograd(hidden_states, prefix_length)
+            max_b = 4 if self._fast.quant == "nf4" else 8
+            if kv_cache is not None and hidden_states.shape[1] == 1 and hidden_states.shape[0] <= max_b:
+                return self._fast.decode_step(
+                    hidden_states, kv_cache[0], kv_cache[1], prefix_length, ctx=ctx
+                )
+            return self._fast.forward(hidden_states, kv_cache, prefix_length)
         residual = hidden_states
         hidden_states = self.input_layernorm(hidden_states)
         hidden_states = self.self_attn(hidden_states, kv_cache=kv_cache, prefix_length=prefix_length)
@@ -154,3 +255,250 @@ class TPLlamaBlock(nn.Module):
     def kv_cache_shape(self, batch_size: int, max_length: int) -> Tuple[Tuple[int, ...], Tuple[int, ...]]:
         shape = (batch_size, self.self_attn.num_kv_heads, max_length, self.config.head_dim)
         return shape, shape
+
+
+def build_tp_block(config, layer_idx: int, *, rank: int, world: int, group=None) -> nn.Module:
+    """TP shard block factory. Llama-family geometry (GQA + SwiGLU) for now;
+    other families serve TP via the generic path or single-rank."""
+    if config.model_type in ("llama",):
+        return TPLlamaBlock(config, layer_idx, rank=rank, world=world, group=group)
+    raise NotImplementedError(f"tensor parallelism is not implemented for model_type={config.model_type!r}")
+
+
+# ---------------------------------------------------------------------------
+# TP serving coordination: rank 0 runs the FULL swarm Server (DHT, handler,
+# runtime); ranks 1..N-1 run TPShadowWorker loops that execute their block
+# shards in lockstep. Every coordinator op is two broadcasts on the tp group
+# (a small int64 control tensor, then payload tensors); rank0 serializes ops
+# under a lock so the collective order is identical on every rank.
+# ---------------------------------------------------------------------------
+
+OP_SHUTDOWN, OP_STEP, OP_OPEN, OP_CLOSE, OP_FWD, OP_BWD, OP_SPAN = 0, 1, 2, 3, 4, 5, 6
+_CTRL_LEN = 8
+
+
+class TPCoordinator:
+    """Rank-0 side: broadcast work to the shadow ranks, then run the same
+    compute locally (the collectives inside the blocks pair up)."""
+
+    def __init__(self, group, device, hidden_size: int, dtype):
+        import threading
+
+        self.group = group
+        self.device = torch.device(device)
+        self.hidden_size = hidden_size
+        self.dtype = dtype
+        self.lock = threading.Lock()
+
+    def _bcast_ctrl(self, *vals):
+        ctrl = torch.zeros(_CTRL_LEN, dtype=torch.int64, device=self.device)
+        for i, v in enumerate(vals):
+            ctrl[i] = int(v)
+        dist.broadcast(ctrl, src=0, group=self.group)
+
+    def _bcast_tensor(self, t: torch.Tensor):
+        dist.broadcast(t.to(self.device, self.dtype).contiguous(), src=0, group=self.group)
+
+    def span(self, start: int, end: int):
+        with self.lock:
+            self._bcast_ctrl(OP_SPAN, start, end)
+
+    def open_session(self, slot: int, batch: int, max_length: int):
+        with self.lock:
+            self._bcast_ctrl(OP_OPEN, slot, batch, max_length)
+
+    def close_session(self, slot: int):
+        with self.lock:
+            self._bcast_ctrl(OP_CLOSE, slot)
+
+    def _bcast_prompts(self, prompts):
+        """prompts: [n_blocks, b, pre_seq, H] or None — deep-ptune prompts are
+        added between blocks, so shadows need them to keep activations equal."""
+        if prompts is None:
+            return
+        dist.broadcast(prompts.to(self.device, self.dtype).contiguous(), src=0, group=self.group)
+
+    @staticmethod
+    def _prompt_dims(prompts):
+        return (1, prompts.shape[0], prompts.shape[2]) if prompts is not None else (0, 0, 0)
+
+    def step(self, slot: int, h: torch.Tensor, position: int, hypo_ids=None, max_chunk: int = 0,
+             prompts=None):
+        with self.lock:
+            b, n = h.shape[0], h.shape[1]
+            hp, _nb, pre = self._prompt_dims(prompts)
+            self._bcast_ctrl(OP_STEP, slot, b, n, position, 1 if hypo_ids is not None else 0,
+                             max_chunk, hp * pre)
+            self._bcast_tensor(h.view(b, n, self.hidden_size))
+            if hypo_ids is not None:
+                hy = hypo_ids.to(self.device, torch.int64).contiguous()
+                dist.broadcast(hy, src=0, group=self.group)
+            self._bcast_prompts(prompts)
+
+    def forward(self, h: torch.Tensor, prompts=None):
+        with self.lock:
+            b, s = h.shape[0], h.shape[1]
+            hp, _nb, pre = self._prompt_dims(prompts)
+            self._bcast_ctrl(OP_FWD, b, s, hp * pre)
+            self._bcast_tensor(h)
+            self._bcast_prompts(prompts)
+
+    def backward(self, inputs: torch.Tensor, grad_outputs: torch.Tensor, prompts=None):
+        with self.lock:
+            b, s = inputs.shape[0], inputs.shape[1]
+            hp, _nb, pre = self._prompt_dims(prompts)
+            self._bcast_ctrl(OP_BWD, b, s, hp * pre)
+            self._bcast_tensor(inputs)
+            self._bcast_tensor(grad_outputs)
+            self._bcast_prompts(prompts)
+
+    def shutdown(self):
+        with self.lock:
+            try:
+                self._bcast_ctrl(OP_SHUTDOWN)
+            except Exception:  # noqa: BLE001
+                pass
+
+
+class TPShadowWorker:
+    """Ranks 1..N-1: receive ops, run the local shard chain so its collectives
+    pair with rank 0's. Holds shard KV caches per (session slot, block)."""
+
+    def __init__(self, model_name_or_dir: str, config, *, device, torch_dtype, quant_type: str,
+                 group=None, rank: int, world: int):
+        self.model_name_or_dir = model_name_or_dir
+        self.config = config
+        self.device = torch.device(device)
+        self.dtype = torch_dtype
+        self.quant = quant_type
+        self.group = group
+        self.rank, self.world = rank, world
+        self.blocks = []
+        self.span = (0, 0)
+        self.sessions = {}  # slot -> {"caches": [(k, v)...], "position": int}
+
+    def _recv_ctrl(self):
+        ctrl = torch.zeros(_CTRL_LEN, dtype=torch.int64, device=self.device)
+        dist.broadcast(ctrl, src=0, group=self.group)
+        return [int(x) for x in ctrl.tolist()]
+
+    def _recv_tensor(self, *shape):
+        t = torch.empty(*shape, device=self.device, dtype=self.dtype)
+        dist.broadcast(t, src=0, group=self.group)
+        return t
+
+    def _load_span(self, start: int, end: int):
+        from petals_amd.server.from_pretrained import load_pretrained_block
+
+        self.blocks = [
+            load_pretrained_block(
+                self.model_name_or_dir, self.config, i, torch_dtype=self.dtype,
+                device=self.device, quant_type=self.quant,
+                tp_rank=self.rank, tp_world=self.world, tp_group=self.group,
+            )
+            for i in range(start, end)
+        ]
+        self.span = (start, end)
+        self.sessions.clear()
+
+    def serve_forever(self):
+        H = self.config.hidden_size
+        while True:
+            op, a, b, c, d, e, f, *_ = self._recv_ctrl()
+            if op == OP_SHUTDOWN:
+                return
+            if op == OP_SPAN:
+                self._load_span(a, b)
+            elif op == OP_OPEN:
+                caches = []
+                for blk in self.blocks:
+                    ks, vs = blk.kv_cache_shape(b, c)
+                    caches.append((
+                        torch.zeros(ks, device=self.device, dtype=self.dtype),
+                        torch.zeros(vs, device=self.device, dtype=self.dtype),
+                    ))
+                self.sessions[a] = {"caches": caches}
+            elif op == OP_CLOSE:
+                self.sessions.pop(a, None)
+            elif op == OP_STEP:
+                slot, batch, n, position, has_hypo, max_chunk, pre = a, b, c, d, e, f, _[0]
+                h = self._recv_tensor(batch, n, H)
+                hypo = None
+                if has_hypo:
+                    hypo = torch.zeros(batch, dtype=torch.int64, device=self.device)
+                    dist.broadcast(hypo, src=0, group=self.group)
+                prompts = self._recv_prompts(pre, batch, H)
+                sess = self.sessions.get(slot)
+                if sess is None:
+                    continue
+                mc = max_chunk if max_chunk > 0 else n
+                with torch.inference_mode():
+                    if hypo is not None:
+                        for k, v in sess["caches"]:
+                            k[...] = k[hypo]
+                            v[...] = v[hypo]
+                    for i, (blk, (k, v)) in enumerate(zip(self.blocks, sess["caches"])):
+                        if prompts is not None:
+                            h = h.clone()
+                            h[:, : prompts.shape[2]] += prompts[i]
+                        # chunk boundaries mirror rank0's backend.inference_step
+                        # so the collective sequences pair up exactly
+                        if n <= mc:
+                            h = blk(h, kv_cache=(k, v), prefix_length=position)
+                        else:
+                            out = torch.empty_like(h)
+                            for off in range(0, n, mc):
+                                chunk = h[:, off : off + mc]
+                                out[:, off : off + chunk.shape[1]] = blk(
+                                    chunk, kv_cache=(k, v), prefix_length=position + off
+                                )
+                            h = out
+            elif op == OP_FWD:
+                h = self._recv_tensor(a, b, H)
+                prompts = self._recv_prompts(c, a, H)
+                with torch.inference_mode():
+                    for i, blk in enumerate(self.blocks):
+                        if prompts is not None:
+                            h = h.clone()
+                            h[:, : prompts.shape[2]] += prompts[i]
+                        h = blk(h)
+            elif op == OP_BWD:
+                inputs = self._recv_tensor(a, b, H)
+                grad_outputs = self._recv_tensor(a, b, H)
+                prompts = self._recv_prompts(c, a, H)
+                self._run_backward(inputs, grad_outputs, prompts)
+
+    def _recv_prompts(self, pre: int, batch: int, H: int):
+        if pre <= 0:
+            return None
+        p = torch.empty(len(self.blocks), batch, pre, H, device=self.device, dtype=self.dtype)
+        dist.broadcast(p, src=0, group=self.group)
+        return p
+
+    def _run_backward(self, inputs, grad_outputs, prompts=None):
+        """Mirror handler._backward_chain: no-grad forward to recover
+        intermediate inputs, then per-block autograd backward in reverse —
+        same collective order as rank 0."""
+        inter = []
+        hidden = inputs
+
+        def add_prompt(h, i):
+            if prompts is None:
+                return h
+            h = h.clone()
+            h[:, : prompts.shape[2]] += prompts[i]
+            return h
+
+        with torch.no_grad():
+            for i, blk in enumerate(self.blocks[:-1]):
+                hidden = add_prompt(hidden, i)
+                inter.append(hidden)
+                hidden = blk(hidden)
+            inter.append(add_prompt(hidden, len(self.blocks) - 1))
+        grad = grad_outputs
+        for blk, hid in zip(reversed(self.blocks), reversed(inter)):
+            with torch.enable_grad():
+                x = hid.detach().requires_grad_(True)
+                out = blk(x)
+                out.backward(grad)
+                grad = x.grad

@@ -80,10 +80,34 @@ def load_pretrained_block(
     torch_dtype: torch.dtype = torch.float32,
     device: torch.device = torch.device("cpu"),
     quant_type: str = "none",
+    tp_rank: int = 0,
+    tp_world: int = 1,
+    tp_group=None,
 ) -> torch.nn.Module:
     """Build one block and fill it with checkpoint weights (or deterministic
-    random weights when no local checkpoint exists)."""
+    random weights when no local checkpoint exists). With tp_world > 1 the
+    result is this rank's TENSOR-PARALLEL SHARD of the block (parallel/tp.py):
+    the full-block weights are materialized on CPU, sliced, and only the shard
+    moves to the device."""
     device = torch.device(device)
+    if tp_world > 1:
+        from petals_amd.parallel.tp import build_tp_block
+
+        full = build_empty_block(config, block_index, torch.device("cpu"), torch_dtype)
+        if os.path.isdir(model_name_or_dir):
+            sd = load_block_state_dict(model_name_or_dir, config, block_index)
+            sd = {k: v.to(dtype=torch_dtype if v.is_floating_point() else None) for k, v in sd.items()}
+            full.load_state_dict(sd, strict=False, assign=True)
+        else:
+            init_random_block_(full, config, block_index)
+        shard = build_tp_block(config, block_index, rank=tp_rank, world=tp_world, group=tp_group)
+        shard = shard.to(torch_dtype)
+        shard.load_from_full_state_dict(full.state_dict())
+        del full
+        shard = shard.to(device).eval()
+        if device.type == "cuda":
+            shard.optimize_for_inference(quant=quant_type)
+        return shard
     block = build_empty_block(config, block_index, device, torch_dtype)
     if os.path.isdir(model_name_or_dir):
         sd = load_block_state_dict(model_name_or_dir, config, block_index)

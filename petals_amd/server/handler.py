@@ -41,6 +41,7 @@ class _Session:
         self.prefix_length = 0
         self.pushed_inputs: asyncio.Queue = asyncio.Queue()
         self.span_graph = None  # _SpanDecodeGraph once captured (GPU decode)
+        self.tp_slot = 0  # shadow-rank KV slot (tensor parallelism)
 
 
 class _SpanDecodeGraph:
@@ -115,6 +116,7 @@ class TransformerConnectionHandler:
         adapters: Sequence[str] = (),
         mesh=None,
         default_compression: str = "none",
+        tp_coord=None,
     ):
         self.backends = backends
         self.memory_cache = memory_cache
@@ -129,6 +131,8 @@ class TransformerConnectionHandler:
         self.p2p = p2p
         self.mesh = mesh  # LocalMesh: RCCL/xGMI hand-off tier for co-located spans
         self.default_compression = default_compression
+        self.tp_coord = tp_coord  # TPCoordinator: rank-0 side of intra-server TP
+        self._tp_slots = iter(range(1, 1 << 30))
         self._sessions: Dict[str, _Session] = {}
 
     def register(self, p2p: P2PNode) -> None:
@@ -195,6 +199,9 @@ class TransformerConnectionHandler:
         backend0 = self.backends[uids[0]]
         device, dtype = backend0.device, backend0.dtype
         hidden_states = hidden_states.to(device=device, dtype=dtype)
+        if self.tp_coord is not None:
+            p = prompts if (prompts is not None and not is_dummy(prompts)) else None
+            self.tp_coord.forward(hidden_states, p)
         prompt_list = self._split_prompts(prompts, len(uids), dtype, device)
         with using_adapter(active_adapter):
             for uid, prompt in zip(uids, prompt_list):
@@ -247,6 +254,9 @@ class TransformerConnectionHandler:
         device, dtype = backend0.device, backend0.dtype
         inputs = inputs.to(device=device, dtype=dtype)
         grad_outputs = grad_outputs.to(device=device, dtype=dtype)
+        if self.tp_coord is not None:
+            p = prompts if (prompts is not None and not is_dummy(prompts)) else None
+            self.tp_coord.backward(inputs, grad_outputs, p)
         prompt_list = self._split_prompts(prompts, len(uids), dtype, device)
         # re-run forward (plain no_grad, NOT inference_mode: these activations
         # feed autograd below) to recover intermediate inputs
@@ -335,6 +345,12 @@ class TransformerConnectionHandler:
             # asyncio loop so launch work overlaps the upstream hop's compute
             hidden_states = hidden_states.result(self.step_timeout)
         hidden_states = hidden_states.to(device=device, dtype=dtype)
+        if self.tp_coord is not None and session is not None:
+            hy = hypo_ids if (hypo_ids is not None and not is_dummy(hypo_ids)) else None
+            p = prompts if (prompts is not None and not is_dummy(prompts)) else None
+            h3d = hidden_states.view(hidden_states.shape[0], -1, hidden_states.shape[-1])
+            max_chunk = backend0._estimate_max_chunk_length(h3d, prefix_length)
+            self.tp_coord.step(session.tp_slot, h3d, prefix_length, hy, max_chunk, p)
         has_hypo = hypo_ids is not None and not is_dummy(hypo_ids)
         has_prompts = prompts is not None and not is_dummy(prompts)
 
@@ -427,6 +443,9 @@ class TransformerConnectionHandler:
         try:
             async with self.memory_cache.allocate_cache(*descriptors, timeout=alloc_timeout) as flat_handles:
                 handles = [tuple(flat_handles[2 * i : 2 * i + 2]) for i in range(len(uids))]
+                if self.tp_coord is not None:
+                    session.tp_slot = next(self._tp_slots)
+                    self.tp_coord.open_session(session.tp_slot, batch_size, max_length)
                 # confirm session is open (client waits for this before step 1)
                 await stream.send(RpcMessage(meta={"session_open": True, "session_id": session_id}))
                 async for step_meta, tensors in self._iterate_inference_steps(stream, session):
@@ -582,6 +601,8 @@ class TransformerConnectionHandler:
                             output = output.cpu()
                         await stream.send(RpcMessage(meta=out_meta, tensors=[output]))
         finally:
+            if self.tp_coord is not None and session.tp_slot:
+                self.tp_coord.close_session(session.tp_slot)
             self._sessions.pop(session_id, None)
 
     async def _iterate_inference_steps(self, stream: RpcStream, session: _Session):

@@ -272,6 +272,22 @@ class Server:
                 self._relay_addr,
             )
 
+        if self.tensor_parallel_ranks > 1:
+            import torch.distributed as dist
+
+            from petals_amd.parallel.tp import TPCoordinator
+
+            assert dist.is_initialized(), "TP serving needs torch.distributed (launch under torchrun)"
+            assert dist.get_rank(self.tp_group) == 0, (
+                "only TP rank 0 runs the Server; other ranks run TPShadowWorker (cli/run_server handles this)"
+            )
+            self._tp_coord = TPCoordinator(
+                self.tp_group, self.device, self.config.hidden_size,
+                self.torch_dtype if self.device.type == "cuda" else torch.float32,
+            )
+        else:
+            self._tp_coord = None
+
         self.runtime = PriorityRuntime(self.device).start()
         while not self._stop.is_set():
             await self._serve_once()
@@ -339,6 +355,10 @@ class Server:
 
         # --- load blocks (in a worker thread: file IO + H2D copies)
         served_uids = [make_uid(self.config.dht_prefix, i) for i in block_indices]
+        if self._tp_coord is not None:
+            # tell the shadow ranks which span to load (they build the same
+            # shard blocks and mirror every collective from here on)
+            self._tp_coord.span(block_indices[0], block_indices[-1] + 1)
         self.backends = {}
         for i, uid in zip(block_indices, served_uids):
             block = await asyncio.get_event_loop().run_in_executor(
@@ -346,6 +366,7 @@ class Server:
                 lambda idx=i: load_pretrained_block(
                     self.model_name_or_dir, self.config, idx, torch_dtype=self.torch_dtype,
                     device=self.device, quant_type=self.quant_type,
+                    tp_rank=0, tp_world=self.tensor_parallel_ranks, tp_group=self.tp_group,
                 ),
             )
             for adapter_dir in self.adapters:
@@ -371,6 +392,7 @@ class Server:
             session_timeout=self.session_timeout,
             step_timeout=self.step_timeout,
             default_compression=self.compression,
+            tp_coord=self._tp_coord,
         )
         self.handler.register(self.p2p)
 
@@ -511,6 +533,8 @@ class Server:
             logger.warning("announce failed: %r", e)
 
     async def _ashutdown(self):
+        if getattr(self, "_tp_coord", None) is not None:
+            self._tp_coord.shutdown()
         if self.runtime is not None:
             self.runtime.shutdown()
         if self.dht_node is not None:

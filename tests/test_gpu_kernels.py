@@ -518,6 +518,104 @@ def test_falcon_mqa71_block_fast_decode_matches_cpu(hip):
     _block_fused_vs_cpu("test-falcon-mqa71", 4544)
 
 
+class _FakeTPAllReduce:
+    """Simulates a 2-rank all-reduce on ONE GPU: both shard threads rendezvous
+    at a barrier, partials are summed, both continue with the total — the
+    exact dataflow of dist.all_reduce over RCCL, minus the wire."""
+
+    def __init__(self, world=2):
+        import threading
+
+        self.barrier = threading.Barrier(world)
+        self.world = world
+        self.slots = {}
+        self.lock = threading.Lock()
+        self.gen = 0
+
+    def __call__(self, t):
+        with self.lock:
+            self.slots[len(self.slots)] = t
+        idx = self.barrier.wait()
+        if idx == 0:
+            total = sum(v.float() for v in self.slots.values())
+            for v in self.slots.values():
+                v.copy_(total.to(v.dtype))
+            torch.cuda.synchronize()
+        self.barrier.wait()
+        if idx == 0:
+            self.slots.clear()
+        self.barrier.wait()
+
+
+@requires_gpu
+@pytest.mark.parametrize("quant", ["none", "nf4"])
+def test_tp_fused_shards_match_full_block(hip, quant):
+    """Two TP shards of a llama block (fused NF4/MFMA path, simulated
+    all-reduce) decode EXACTLY like the unsharded fused block."""
+    import threading
+
+    from petals_amd.models import get_model_block
+    from petals_amd.models.config_base import load_model_config
+    from petals_amd.parallel.tp import TPLlamaBlock
+    from petals_amd.server.from_pretrained import init_random_block_
+
+    cfg = load_model_config("test-llama-hd128")
+    full = get_model_block(cfg, 0)
+    init_random_block_(full, cfg, 0)
+    sd = {k: v.clone() for k, v in full.state_dict().items()}
+    full = full.to("cuda", torch.bfloat16).eval().optimize_for_inference(quant=quant)
+
+    reducer = _FakeTPAllReduce(2)
+    shards = []
+    for r in range(2):
+        blk = TPLlamaBlock(cfg, 0, rank=r, world=2)
+        blk.load_from_full_state_dict(sd)
+        blk = blk.to("cuda", torch.bfloat16).eval().optimize_for_inference(quant=quant)
+        blk._fast._tp_allreduce_ = reducer
+        shards.append(blk)
+
+    torch.manual_seed(9)
+    T = 4
+    xs = [torch.randn(1, 1, cfg.hidden_size, device="cuda", dtype=torch.bfloat16) * 0.5 for _ in range(T)]
+
+    ks, vs = full.kv_cache_shape(1, 16)
+    kf = torch.zeros(ks, device="cuda", dtype=torch.bfloat16)
+    vf = torch.zeros_like(kf)
+    ref = [full(x, kv_cache=(kf, vf), prefix_length=t) for t, x in enumerate(xs)]
+
+    outs = [None, None]
+    errs = []
+
+    def run_rank(r):
+        try:
+            torch.cuda.set_device(0)
+            blk = shards[r]
+            ks_, vs_ = blk.kv_cache_shape(1, 16)
+            k = torch.zeros(ks_, device="cuda", dtype=torch.bfloat16)
+            v = torch.zeros_like(k)
+            res = []
+            for t, x in enumerate(xs):
+                res.append(blk(x, kv_cache=(k, v), prefix_length=t))
+            outs[r] = res
+        except Exception as e:  # noqa: BLE001
+            errs.append((r, repr(e)))
+            try:
+                reducer.barrier.abort()
+            except Exception:  # noqa: BLE001
+                pass
+
+    threads = [threading.Thread(target=run_rank, args=(r,)) for r in range(2)]
+    for th in threads:
+        th.start()
+    for th in threads:
+        th.join(timeout=120)
+    assert not errs, errs
+    for t in range(T):
+        a, b = outs[0][t].float(), outs[1][t].float()
+        assert torch.allclose(a, b), "TP ranks must agree on the reduced output"
+        assert torch.allclose(a, ref[t].float(), atol=0.05, rtol=0.05), (a - ref[t].float()).abs().max()
+
+
 @requires_gpu
 @pytest.mark.parametrize("gq,kv_heads,kv_len", [(29, 1, 333), (71, 1, 95), (32, 2, 200)])
 def test_attn_decode_big_gq(hip, gq, kv_heads, kv_len):

@@ -36,8 +36,13 @@ def _tp_worker(rank, world, port, payload_path, result_path):
             parts.append(block(x[:, t : t + 1], kv_cache=(k, v), prefix_length=t))
         inc = torch.cat(parts, 1)
 
+        # training backward: grads wrt the replicated input must equal the
+        # full block's (copy_to_tp all-reduces input grads across shards)
+        xg = x.detach().clone().requires_grad_(True)
+        block(xg).square().mean().backward()
+
         if rank == 0:
-            torch.save({"out": out, "inc": inc}, result_path)
+            torch.save({"out": out, "inc": inc, "x_grad": xg.grad}, result_path)
     finally:
         dist.destroy_process_group()
 
@@ -53,6 +58,8 @@ def test_tp_block_matches_full(tmp_path):
     torch.manual_seed(0)
     x = torch.randn(2, 7, cfg.hidden_size)
     ref = full(x)
+    x_ref = x.detach().clone().requires_grad_(True)
+    full(x_ref).square().mean().backward()
 
     payload_path = str(tmp_path / "payload.pt")
     result_path = str(tmp_path / "result.pt")
@@ -68,3 +75,6 @@ def test_tp_block_matches_full(tmp_path):
     result = torch.load(result_path, weights_only=False)
     assert torch.allclose(result["out"], ref, atol=1e-5), (result["out"] - ref).abs().max()
     assert torch.allclose(result["inc"], ref, atol=1e-4), (result["inc"] - ref).abs().max()
+    assert torch.allclose(result["x_grad"], x_ref.grad, atol=1e-5), (
+        (result["x_grad"] - x_ref.grad).abs().max()
+    )
