@@ -65,7 +65,10 @@ __global__ __launch_bounds__(WAVES * 64) void attn_prefill_kernel(
   const int col = lane & 15;      // fragment column (and C col)
   const int hi = lane >> 4;       // fragment k-group (and C row group)
   const int q0 = q0_wg + wave * QTILE;  // this wave's first q row
-  const float slope = ALIBI ? alibi[qh] : 0.f;  // wave-uniform (one head per wg)
+  // softmax runs in the log2 domain (exp2 instead of exp): q is pre-scaled by
+  // scale*log2e at load, saving a VALU mul per score AND per exponential —
+  // the kernel was issue-bound on VALU at 17:1 VALU:MFMA (PMC, round 1)
+  const float slope = ALIBI ? alibi[qh] * 1.4426950408889634f : 0.f;  // wave-uniform
 
   // LDS: K tile row-major [KVT][HD+KPAD]; V tile transposed+swizzled
   // [HD][KVT+KPAD]; per-wave P scratch [QTILE][KVT+KPAD]
@@ -77,7 +80,9 @@ __global__ __launch_bounds__(WAVES * 64) void attn_prefill_kernel(
   constexpr int NB = KVT / 16;   // 16-key S column blocks per tile
   constexpr int PKC = KVT / 32;  // 32-key PV k-chunks per tile
 
-  // ---- load this wave's q tile into A-fragments (zero-padded past s_q)
+  // ---- load this wave's q tile into A-fragments (zero-padded past s_q),
+  // pre-scaled by scale*log2e (see the log2-domain softmax note above)
+  const float qscale = scale * 1.4426950408889634f;
   bf16x8 q_frag[KCH];
   const size_t q_base = (((size_t)b * q_heads + qh) * s_q) * HD;
   const int my_qrow = q0 + col;  // A: row = lane&15
@@ -85,7 +90,12 @@ __global__ __launch_bounds__(WAVES * 64) void attn_prefill_kernel(
   for (int kc = 0; kc < KCH; ++kc) {
     if (my_qrow < s_q) {
       const unsigned short* src = q + q_base + (size_t)my_qrow * HD + kc * 32 + hi * 8;
-      q_frag[kc] = *reinterpret_cast<const bf16x8*>(src);
+      const bf16x8 raw = *reinterpret_cast<const bf16x8*>(src);
+      short vs[8];
+#pragma unroll
+      for (int e = 0; e < 8; ++e)
+        vs[e] = (short)f32_to_bf16(bf16_to_f32((unsigned short)raw[e]) * qscale);
+      q_frag[kc] = bf16x8{vs[0], vs[1], vs[2], vs[3], vs[4], vs[5], vs[6], vs[7]};
     } else {
       q_frag[kc] = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
     }
@@ -163,14 +173,14 @@ __global__ __launch_bounds__(WAVES * 64) void attn_prefill_kernel(
       if (interior) {
 #pragma unroll
         for (int nb = 0; nb < NB; ++nb) {
-          s[nb] = s_acc[nb][r] * scale;
+          s[nb] = s_acc[nb][r];  // already scaled (q pre-scaled by scale*log2e)
           if (ALIBI) s[nb] += slope * (j0 + nb * 16 + col);
         }
       } else {
 #pragma unroll
         for (int nb = 0; nb < NB; ++nb) {
           const int key = j0 + nb * 16 + col;
-          s[nb] = s_acc[nb][r] * scale;
+          s[nb] = s_acc[nb][r];
           if (ALIBI) s[nb] += slope * key;
           if (key >= tile_n + j0 || (causal && key > q_abs) || qrow >= s_q) s[nb] = NEG_SENTINEL;
         }
@@ -182,11 +192,11 @@ __global__ __launch_bounds__(WAVES * 64) void attn_prefill_kernel(
 #pragma unroll
       for (int off = 8; off > 0; off >>= 1) mx = fmaxf(mx, __shfl_xor(mx, off, WAVE));
       const float m_new = fmaxf(m_row[r], mx);
-      corr[r] = (m_row[r] <= NEG_THRESHOLD) ? 0.f : __expf(m_row[r] - m_new);
+      corr[r] = (m_row[r] <= NEG_THRESHOLD) ? 0.f : exp2f(m_row[r] - m_new);
       float lsum = 0.f;
 #pragma unroll
       for (int nb = 0; nb < NB; ++nb) {
-        p[nb][r] = (s[nb] <= NEG_THRESHOLD) ? 0.f : __expf(s[nb] - m_new);
+        p[nb][r] = (s[nb] <= NEG_THRESHOLD) ? 0.f : exp2f(s[nb] - m_new);
         lsum += p[nb][r];
       }
 #pragma unroll

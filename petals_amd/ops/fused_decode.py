@@ -35,12 +35,21 @@ _EPI_SWIGLU_F32 = 3
 _EPI_GELU_F32 = 4
 
 _workspaces: Dict[Tuple[torch.device, str], torch.Tensor] = {}
+_retired_workspaces: list = []  # replaced but possibly graph-captured (see _get_ws)
 
 
 def _get_ws(device: torch.device, key: str, numel: int) -> torch.Tensor:
     k = (device, key)
     ws = _workspaces.get(k)
     if ws is None or ws.numel() < numel:
+        if ws is not None:
+            # NEVER free a replaced workspace: hipGraphs capture raw pointers,
+            # so freeing it lets the allocator hand the memory to someone else
+            # and a previously captured graph replays into foreign storage.
+            # (Root cause of the round-1 "bloom dual-graph hardware fault":
+            # the client LM-head graph for bloom's 250k vocab grew the shared
+            # gemv workspace AFTER the span graph had captured the old one.)
+            _retired_workspaces.append(ws)
         ws = torch.empty(numel, dtype=torch.float32, device=device)
         _workspaces[k] = ws
     return ws
@@ -147,12 +156,62 @@ class _FastWeight:
         return self._gemv_raw(x, ws, residual, epilogue, self.splits, bias)
 
     def dense(self) -> torch.Tensor:
-        """bf16 [in, out] view for prefill GEMMs (dequantized on the fly when quantized)."""
+        """bf16 [in, out] view for prefill GEMMs. Quantized weights dequantize
+        through a process-wide LRU (default 16 GiB of the 288 GB HBM,
+        PETALS_AMD_DEQUANT_CACHE_MB) so chunked prefills and back-to-back
+        sessions stop paying the per-matmul dequant (~337 us/weight measured
+        round 1, PARITY.md)."""
         if self.quant == "nf4":
-            return self.hip.nf4_dequantize(self.packed, self.absmax)
+            return _dequant_cached(self, lambda: self.hip.nf4_dequantize(self.packed, self.absmax))
         if self.quant == "int8":
-            return (self.q8.to(torch.float32) * self.scale8.float()).to(torch.bfloat16)
+            return _dequant_cached(
+                self, lambda: (self.q8.to(torch.float32) * self.scale8.float()).to(torch.bfloat16)
+            )
         return self.t
+
+
+# ---- dequantized-weight LRU (keyed by _FastWeight identity, byte-bounded)
+_dequant_cache: "OrderedDict[int, torch.Tensor]" = None  # type: ignore[assignment]
+_dequant_cache_bytes = 0
+
+
+def _dequant_cache_budget() -> int:
+    import os as _os
+
+    return int(_os.environ.get("PETALS_AMD_DEQUANT_CACHE_MB", "16384")) << 20
+
+
+def _dequant_cached(w, make) -> torch.Tensor:
+    global _dequant_cache, _dequant_cache_bytes
+    from collections import OrderedDict
+
+    if _dequant_cache is None:
+        _dequant_cache = OrderedDict()
+    key = id(w)
+    t = _dequant_cache.get(key)
+    if t is not None:
+        _dequant_cache.move_to_end(key)
+        return t
+    t = make()
+    budget = _dequant_cache_budget()
+    nbytes = t.numel() * t.element_size()
+    if nbytes > budget:
+        return t  # too big to cache at all
+    while _dequant_cache and _dequant_cache_bytes + nbytes > budget:
+        _k, old = _dequant_cache.popitem(last=False)
+        _dequant_cache_bytes -= old.numel() * old.element_size()
+    _dequant_cache[key] = t
+    _dequant_cache_bytes += nbytes
+    import weakref
+
+    def _evict(k=key):
+        global _dequant_cache_bytes
+        old = _dequant_cache.pop(k, None) if _dequant_cache else None
+        if old is not None:
+            _dequant_cache_bytes -= old.numel() * old.element_size()
+
+    weakref.finalize(w, _evict)  # id() reuse after GC must not alias entries
+    return t
 
 
 class LlamaFastPath:
@@ -473,14 +532,15 @@ def _bloom_qkv_perm(qh: int, hd: int) -> torch.Tensor:
 
 
 class BloomFastPath:
-    # hipGraph capture of a bloom span is numerically fine in isolation (see
-    # scripts/bloom_graph_debug.py — every chain prefix captures and replays),
-    # but a bloom span graph coexisting with the client's LM-head graph
-    # hardware-faults on replay (scripts/head_graph_debug.py isolates the head
-    # as clean too; the same dual-graph pattern works for Llama/Falcon/405B).
-    # Until the interaction is root-caused, bloom decodes eagerly: 40.8 tok/s
-    # on bloom-176b NF4 (gpurun_out/bloom176_ng2.log) vs the graphs-on crash.
-    graph_safe = False
+    # The round-1 "bloom dual-graph hardware fault" (a bloom span graph +
+    # the client's LM-head graph faulting on replay while each was clean in
+    # isolation) was ROOT-CAUSED to the shared gemv workspace: bloom's 250k
+    # vocab made the head's workspace request GROW the shared buffer, freeing
+    # the allocation the span graph had captured raw pointers into. _get_ws
+    # now retires (never frees) replaced workspaces, and the dual-graph
+    # pattern is exercised by tests/test_gpu_kernels.py::
+    # test_bloom_dual_graph_with_big_vocab_head.
+    graph_safe = True
 
     def __init__(self, block, hip_ops, quant: str = "none"):
         cfg = block.config

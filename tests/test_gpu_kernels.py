@@ -501,6 +501,66 @@ def test_bloom_block_fast_decode_matches_cpu(hip):
 
 
 @requires_gpu
+def test_bloom_dual_graph_with_big_vocab_head(hip):
+    """Regression for the round-1 dual-graph fault: a bloom-layout span graph
+    must survive the client's LM-head graph growing the shared gemv workspace
+    (bloom's 250k vocab). _get_ws retires replaced workspaces so captured raw
+    pointers stay valid."""
+    from petals_amd.models import get_model_block
+    from petals_amd.models.config_base import load_model_config
+    from petals_amd.ops.fused_decode import DecodeContext
+    from petals_amd.server.from_pretrained import init_random_block_
+    from petals_amd.utils.graphs import GraphedCallable
+
+    cfg = load_model_config("test-bloom-hd64")
+    blk = get_model_block(cfg, 0)
+    init_random_block_(blk, cfg, 0)
+    blk = blk.to("cuda", torch.bfloat16).eval().optimize_for_inference()
+    assert blk._fast is not None and blk._fast.graph_safe
+
+    H, vocab = cfg.hidden_size, 250880  # bloom-176b-sized head
+    ks, vs = blk.kv_cache_shape(1, 16)
+    kg = torch.zeros(ks, device="cuda", dtype=torch.bfloat16)
+    vg = torch.zeros_like(kg)
+    ctx = DecodeContext(torch.device("cuda"))
+    ctx.set_position(0)
+    h_in = torch.randn(1, 1, H, device="cuda", dtype=torch.bfloat16) * 0.5
+
+    g_span = GraphedCallable(lambda: blk(h_in, kv_cache=(kg, vg), ctx=ctx), [])
+
+    # capture the head AFTER the span graph — its gemv workspace request (64 *
+    # vocab floats) is far larger than the span's and used to free the span's
+    head_t = (torch.randn(H, vocab, device="cuda") * 0.02).to(torch.bfloat16)
+    norm_w = torch.ones(H, device="cuda", dtype=torch.bfloat16)
+    ws_empty = torch.empty(0, device="cuda")
+    h_last = torch.randn(1, H, device="cuda", dtype=torch.bfloat16)
+
+    def head_fn():
+        from petals_amd.ops.fused_decode import _get_ws
+
+        xn = hip.rms_norm_f32out(h_last, norm_w, 1e-5)
+        return hip.gemv_bf16(head_t, xn, _get_ws(torch.device("cuda", 0), "gemv", 64 * vocab), None, 0)
+
+    g_head = GraphedCallable(head_fn, [])
+
+    # interleaved replays: span graph must still write valid outputs
+    blk2 = get_model_block(cfg, 0)
+    init_random_block_(blk2, cfg, 0)
+    blk2 = blk2.to("cuda", torch.bfloat16).eval().optimize_for_inference()
+    kg2, vg2 = kg.clone(), vg.clone()
+    for t in range(4):
+        ctx.set_position(t)
+        x = torch.randn(1, 1, H, device="cuda", dtype=torch.bfloat16) * 0.5
+        h_in.copy_(x)
+        out = g_span.replay().clone()
+        g_head.replay()
+        ref = blk2(x, kv_cache=(kg2, vg2), prefix_length=t)
+        assert torch.allclose(out.float(), ref.float(), atol=1e-3, rtol=1e-3), (
+            (out - ref).abs().max()
+        )
+
+
+@requires_gpu
 def test_falcon_block_fast_decode_matches_cpu(hip):
     """Falcon new-decoder fused path (parallel attn+MLP, rope GQA) vs CPU."""
     _block_fused_vs_cpu("test-falcon-hd64", 256)
