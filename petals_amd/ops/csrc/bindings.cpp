@@ -37,7 +37,7 @@ torch::Tensor nf4_dequantize(torch::Tensor packed, torch::Tensor absmax);
 torch::Tensor gemv_nf4(
     torch::Tensor packed, torch::Tensor absmax, torch::Tensor x, torch::Tensor workspace,
     c10::optional<torch::Tensor> residual, int64_t epilogue, int64_t splits_override,
-    c10::optional<torch::Tensor> bias);
+    c10::optional<torch::Tensor> bias, c10::optional<torch::Tensor> absmax_t);
 torch::Tensor gemv_bf16_moe(
     torch::Tensor wt_all, torch::Tensor x, torch::Tensor sel, int64_t k_per_tok,
     torch::Tensor workspace, int64_t epilogue, int64_t splits_override);
@@ -72,7 +72,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("nf4_dequantize", &nf4_dequantize, "NF4 -> bf16");
   m.def("gemv_nf4", &gemv_nf4, "split-K NF4 gemv (dequant fused) with epilogue",
         py::arg("packed"), py::arg("absmax"), py::arg("x"), py::arg("workspace"),
-        py::arg("residual"), py::arg("epilogue"), py::arg("splits") = 0, py::arg("bias") = py::none());
+        py::arg("residual"), py::arg("epilogue"), py::arg("splits") = 0, py::arg("bias") = py::none(),
+        py::arg("absmax_t") = py::none());
   m.def("gemv_bf16_moe", &gemv_bf16_moe,
         "device-routed MoE gemv: stacked bf16 expert weights, expert ids from a device tensor",
         py::arg("wt_all"), py::arg("x"), py::arg("sel"), py::arg("k_per_tok"),

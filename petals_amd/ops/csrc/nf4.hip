@@ -89,6 +89,10 @@ template <int BATCH>
 __global__ void gemv_nf4_kernel(
     const unsigned char* __restrict__ packed,   // [in, out/2]
     const unsigned short* __restrict__ absmax,  // [in, out/64]
+    const unsigned short* __restrict__ absmax_t,  // [out/64, in] or null — the
+    // transposed copy turns the per-row 2-byte strided absmax gather into ONE
+    // contiguous 32 B load per 16-row unroll block (the strided stream was a
+    // latency-hiding bottleneck: 16 extra scattered loads in flight per block)
     const float* __restrict__ x,                // [BATCH, in]
     float* __restrict__ partials,               // [n_splits, BATCH, out]
     int in_dim,
@@ -121,6 +125,7 @@ __global__ void gemv_nf4_kernel(
     constexpr int UNROLL = 16;  // 128 B of packed loads in flight per wave
     const int half_out = out_dim >> 1;
     const unsigned char* pp = packed + (size_t)i_begin * half_out + (out0 >> 1);
+    const unsigned short* amt = absmax_t ? absmax_t + (size_t)(out0 >> 6) * in_dim : nullptr;
     int i = i_begin;
     for (; i + UNROLL <= i_end; i += UNROLL) {
       uint2 pk[UNROLL];  // 8 bytes = 16 nibbles
@@ -128,9 +133,19 @@ __global__ void gemv_nf4_kernel(
       for (int u = 0; u < UNROLL; ++u)
         pk[u] = *reinterpret_cast<const uint2*>(pp + (size_t)u * half_out);
       float am[UNROLL];
+      if (amt) {
+        const short8 a0 = *reinterpret_cast<const short8*>(amt + i);
+        const short8 a1 = *reinterpret_cast<const short8*>(amt + i + 8);
 #pragma unroll
-      for (int u = 0; u < UNROLL; ++u)
-        am[u] = bf16_to_f32(absmax[(size_t)(i + u) * (out_dim >> 6) + (out0 >> 6)]);
+        for (int e = 0; e < 8; ++e) {
+          am[e] = bf16_to_f32((unsigned short)a0[e]);
+          am[8 + e] = bf16_to_f32((unsigned short)a1[e]);
+        }
+      } else {
+#pragma unroll
+        for (int u = 0; u < UNROLL; ++u)
+          am[u] = bf16_to_f32(absmax[(size_t)(i + u) * (out_dim >> 6) + (out0 >> 6)]);
+      }
       float xs[BATCH][UNROLL];
 #pragma unroll
       for (int b = 0; b < BATCH; ++b)
@@ -246,7 +261,8 @@ torch::Tensor gemv_nf4(
     c10::optional<torch::Tensor> residual,
     int64_t epilogue,
     int64_t splits_override,
-    c10::optional<torch::Tensor> bias) {  // [out] bf16, added pre-activation
+    c10::optional<torch::Tensor> bias,      // [out] bf16, added pre-activation
+    c10::optional<torch::Tensor> absmax_t) {  // [out/64, in] bf16 (transposed copy)
   TORCH_CHECK(packed.is_cuda() && packed.dtype() == torch::kUInt8);
   TORCH_CHECK(x.dtype() == torch::kFloat32 && x.dim() == 2);
   const int in_dim = packed.size(0);
@@ -273,10 +289,16 @@ torch::Tensor gemv_nf4(
   dim3 grid(out_waves, splits);
   auto stream = at::cuda::getCurrentCUDAStream();
 
+  const unsigned short* amt_p = nullptr;
+  if (absmax_t.has_value() && absmax_t->defined() && absmax_t->numel() > 0) {
+    TORCH_CHECK(absmax_t->is_contiguous() && absmax_t->size(0) == out_dim / 64
+                && absmax_t->size(1) == in_dim);
+    amt_p = reinterpret_cast<const unsigned short*>(absmax_t->data_ptr());
+  }
 #define LAUNCH_NF4(B)                                                         \
   gemv_nf4_kernel<B><<<grid, WAVE, 0, stream>>>(                              \
       packed.data_ptr<unsigned char>(),                                       \
-      reinterpret_cast<const unsigned short*>(absmax.data_ptr()),             \
+      reinterpret_cast<const unsigned short*>(absmax.data_ptr()), amt_p,      \
       x.data_ptr<float>(), partials.data_ptr<float>(), in_dim, out_dim, i_per_split)
   switch (batch) {
     case 1: LAUNCH_NF4(1); break;

@@ -92,9 +92,12 @@ class _FastWeight:
         self.hip = hip
         self.quant = quant
         self.in_dim, self.out_dim = t_bf16.shape
-        self.t = self.packed = self.absmax = self.q8 = self.scale8 = None
+        self.t = self.packed = self.absmax = self.absmax_t = self.q8 = self.scale8 = None
         if quant == "nf4":
             self.packed, self.absmax = hip.nf4_quantize(t_bf16.contiguous())
+            # transposed absmax copy: one contiguous 32 B load per 16-row
+            # unroll block in the gemv instead of 16 strided 2 B gathers
+            self.absmax_t = self.absmax.t().contiguous()
         elif quant == "int8":
             w = t_bf16.float()
             scale = w.abs().amax(dim=0).clamp_min(1e-8) / 127.0
@@ -147,7 +150,8 @@ class _FastWeight:
 
     def _gemv_raw(self, x, ws, residual, epilogue, splits, bias=None):
         if self.quant == "nf4":
-            return self.hip.gemv_nf4(self.packed, self.absmax, x, ws, residual, epilogue, splits, bias)
+            return self.hip.gemv_nf4(self.packed, self.absmax, x, ws, residual, epilogue, splits, bias,
+                                     self.absmax_t)
         if self.quant == "int8":
             return self.hip.gemv_int8(self.q8, self.scale8, x, ws, residual, epilogue, splits, bias)
         return self.hip.gemv_bf16(self.t, x, ws, residual, epilogue, splits, bias)

@@ -56,16 +56,19 @@ def nf4_sweep():
         del wt
         x = torch.randn(1, in_dim, device="cuda")
         gb = (packed.numel() + absmax.numel() * 2) / 1e9
-        row, best = [], (0, 0)
-        for splits in (0, 16, 32, 64, 128, 192, 256, 448):
-            if splits * 16 > in_dim and splits:
-                continue
-            t = bench(lambda: hip.gemv_nf4(packed, absmax, x, ws, None, 0, splits))
-            bw = gb / t
-            row.append(f"s{splits}:{bw:.0f}")
-            if bw > best[0]:
-                best = (bw, splits)
-        print(f"gemv_nf4 {name} [{in_dim}x{out_dim}] {gb*1000:.0f}MB: {' '.join(row)}  BEST s{best[1]} {best[0]:.0f} GB/s", flush=True)
+        amt = absmax.t().contiguous()
+        for label, amt_arg in (("strided-am", None), ("am_t", amt)):
+            row, best = [], (0, 0)
+            for splits in (0, 16, 32, 64, 128, 192, 256, 448):
+                if splits * 16 > in_dim and splits:
+                    continue
+                t = bench(lambda: hip.gemv_nf4(packed, absmax, x, ws, None, 0, splits, None, amt_arg))
+                bw = gb / t
+                row.append(f"s{splits}:{bw:.0f}")
+                if bw > best[0]:
+                    best = (bw, splits)
+            print(f"gemv_nf4[{label}] {name} [{in_dim}x{out_dim}] {gb*1000:.0f}MB: {' '.join(row)}  "
+                  f"BEST s{best[1]} {best[0]:.0f} GB/s", flush=True)
         torch.cuda.empty_cache()
 
 
