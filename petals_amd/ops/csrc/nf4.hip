@@ -83,9 +83,12 @@ __global__ void nf4_dequant_kernel(
 
 // -------------------------------------------------------------- NF4 gemv
 
-#define NF4_OUT_PER_WAVE 1024  // 64 lanes x 16 outputs (8 B packed / row)
+#define NF4_OUT_PER_WAVE 1024  // default: 64 lanes x 16 outputs (8 B packed / row)
 
-template <int BATCH>
+// OPL = outputs per lane (16 or 8). OPL=8 doubles the workgroup count for the
+// same split depth: the NF4 inner loop's dependent LDS gathers need more
+// resident waves per SIMD to hide latency than the direct-load bf16 gemv.
+template <int BATCH, int OPL>
 __global__ void gemv_nf4_kernel(
     const unsigned char* __restrict__ packed,   // [in, out/2]
     const unsigned short* __restrict__ absmax,  // [in, out/64]
@@ -106,32 +109,40 @@ __global__ void gemv_nf4_kernel(
     lut2[i] = make_float2(NF4_LUT_C[i & 0xF], NF4_LUT_C[i >> 4]);
   __syncthreads();
 
+  constexpr int WORDS = OPL / 8;  // u32 packed words per lane per row
   const int lane = threadIdx.x & (WAVE - 1);
 #define NF4_L2(byte) lut2[(byte)]
-  const int out0 = blockIdx.x * NF4_OUT_PER_WAVE + lane * 16;
+  const int out0 = blockIdx.x * (WAVE * OPL) + lane * OPL;
   if (out0 >= out_dim) return;
   const int split = blockIdx.y;
   const int i_begin = split * i_per_split;
   const int i_end = min(i_begin + i_per_split, in_dim);
-  const bool full = (out0 + 16) <= out_dim;
+  const bool full = (out0 + OPL) <= out_dim;
 
-  float acc[BATCH][16];
+  float acc[BATCH][OPL];
 #pragma unroll
   for (int b = 0; b < BATCH; ++b)
 #pragma unroll
-    for (int v = 0; v < 16; ++v) acc[b][v] = 0.f;
+    for (int v = 0; v < OPL; ++v) acc[b][v] = 0.f;
 
   if (full) {
-    constexpr int UNROLL = 16;  // 128 B of packed loads in flight per wave
+    constexpr int UNROLL = 16;
     const int half_out = out_dim >> 1;
     const unsigned char* pp = packed + (size_t)i_begin * half_out + (out0 >> 1);
     const unsigned short* amt = absmax_t ? absmax_t + (size_t)(out0 >> 6) * in_dim : nullptr;
     int i = i_begin;
     for (; i + UNROLL <= i_end; i += UNROLL) {
-      uint2 pk[UNROLL];  // 8 bytes = 16 nibbles
+      unsigned int pk[UNROLL][WORDS];
 #pragma unroll
-      for (int u = 0; u < UNROLL; ++u)
-        pk[u] = *reinterpret_cast<const uint2*>(pp + (size_t)u * half_out);
+      for (int u = 0; u < UNROLL; ++u) {
+        if constexpr (WORDS == 2) {
+          const uint2 p2v = *reinterpret_cast<const uint2*>(pp + (size_t)u * half_out);
+          pk[u][0] = p2v.x;
+          pk[u][1] = p2v.y;
+        } else {
+          pk[u][0] = *reinterpret_cast<const unsigned int*>(pp + (size_t)u * half_out);
+        }
+      }
       float am[UNROLL];
       if (amt) {
         const short8 a0 = *reinterpret_cast<const short8*>(amt + i);
@@ -153,13 +164,12 @@ __global__ void gemv_nf4_kernel(
         for (int u = 0; u < UNROLL; ++u) xs[b][u] = x[(size_t)b * in_dim + i + u];
 #pragma unroll
       for (int u = 0; u < UNROLL; ++u) {
-        float wf[16];
-        const unsigned int wds2[2] = {pk[u].x, pk[u].y};
+        float wf[OPL];
 #pragma unroll
-        for (int d = 0; d < 2; ++d)
+        for (int d = 0; d < WORDS; ++d)
 #pragma unroll
           for (int p2 = 0; p2 < 4; ++p2) {
-            const float2 w2 = NF4_L2((wds2[d] >> (8 * p2)) & 0xFFu);
+            const float2 w2 = NF4_L2((pk[u][d] >> (8 * p2)) & 0xFFu);
             wf[8 * d + 2 * p2] = w2.x;
             wf[8 * d + 2 * p2 + 1] = w2.y;
           }
@@ -167,26 +177,27 @@ __global__ void gemv_nf4_kernel(
         for (int b = 0; b < BATCH; ++b) {
           const float xa = xs[b][u] * am[u];
 #pragma unroll
-          for (int v = 0; v < 16; ++v) acc[b][v] = fmaf(wf[v], xa, acc[b][v]);
+          for (int v = 0; v < OPL; ++v) acc[b][v] = fmaf(wf[v], xa, acc[b][v]);
         }
       }
       pp += (size_t)UNROLL * half_out;
     }
     for (; i < i_end; ++i) {
-      const uint2 pk = *reinterpret_cast<const uint2*>(packed + (size_t)i * half_out + (out0 >> 1));
       const float am = bf16_to_f32(absmax[(size_t)i * (out_dim >> 6) + (out0 >> 6)]);
-      const unsigned int twds[2] = {pk.x, pk.y};
 #pragma unroll
       for (int b = 0; b < BATCH; ++b) {
         const float xa = x[(size_t)b * in_dim + i] * am;
 #pragma unroll
-        for (int d = 0; d < 2; ++d)
+        for (int d = 0; d < WORDS; ++d) {
+          const unsigned int wd =
+              *reinterpret_cast<const unsigned int*>(packed + (size_t)i * half_out + (out0 >> 1) + 4 * d);
 #pragma unroll
           for (int p2 = 0; p2 < 4; ++p2) {
-            const float2 w2 = NF4_L2((twds[d] >> (8 * p2)) & 0xFFu);
+            const float2 w2 = NF4_L2((wd >> (8 * p2)) & 0xFFu);
             acc[b][8 * d + 2 * p2] = fmaf(w2.x, xa, acc[b][8 * d + 2 * p2]);
             acc[b][8 * d + 2 * p2 + 1] = fmaf(w2.y, xa, acc[b][8 * d + 2 * p2 + 1]);
           }
+        }
       }
     }
   } else {
@@ -208,7 +219,7 @@ __global__ void gemv_nf4_kernel(
     float* dst = partials + ((size_t)split * BATCH + b) * out_dim + out0;
     if (full) {
 #pragma unroll
-      for (int q = 0; q < 4; ++q)
+      for (int q = 0; q < OPL / 4; ++q)
         reinterpret_cast<float4v*>(dst)[q] =
             float4v{acc[b][4 * q], acc[b][4 * q + 1], acc[b][4 * q + 2], acc[b][4 * q + 3]};
     } else {
@@ -270,7 +281,16 @@ torch::Tensor gemv_nf4(
   const int batch = x.size(0);
   TORCH_CHECK(x.size(1) == in_dim && batch <= 4, "NF4 decode gemv supports batch <= 4");
 
-  const long out_waves = (out_dim + NF4_OUT_PER_WAVE - 1) / NF4_OUT_PER_WAVE;
+  // outputs per lane: 8 doubles the grid vs 16 (latency hiding for the
+  // LDS-gather dequant chain); PETALS_NF4_OPL overrides for A/B sweeps
+  static const int opl_env = [] {
+    const char* s = std::getenv("PETALS_NF4_OPL");
+    return s ? std::atoi(s) : 0;
+  }();
+  const int opl = opl_env ? opl_env : 8;
+  TORCH_CHECK(opl == 8 || opl == 16, "PETALS_NF4_OPL must be 8 or 16");
+
+  const long out_waves = (out_dim + (long)WAVE * opl - 1) / ((long)WAVE * opl);
   // NF4 matrices are 4x smaller than bf16: allow chunks down to 64 input rows
   // so small projections still spread over the 256 CUs
   long splits = splits_override > 0 ? splits_override : (1536 + out_waves - 1) / out_waves;
@@ -295,16 +315,25 @@ torch::Tensor gemv_nf4(
                 && absmax_t->size(1) == in_dim);
     amt_p = reinterpret_cast<const unsigned short*>(absmax_t->data_ptr());
   }
-#define LAUNCH_NF4(B)                                                         \
-  gemv_nf4_kernel<B><<<grid, WAVE, 0, stream>>>(                              \
+#define LAUNCH_NF4(B, OPL)                                                    \
+  gemv_nf4_kernel<B, OPL><<<grid, WAVE, 0, stream>>>(                         \
       packed.data_ptr<unsigned char>(),                                       \
       reinterpret_cast<const unsigned short*>(absmax.data_ptr()), amt_p,      \
       x.data_ptr<float>(), partials.data_ptr<float>(), in_dim, out_dim, i_per_split)
-  switch (batch) {
-    case 1: LAUNCH_NF4(1); break;
-    case 2: LAUNCH_NF4(2); break;
-    case 3: LAUNCH_NF4(3); break;
-    case 4: LAUNCH_NF4(4); break;
+  if (opl == 8) {
+    switch (batch) {
+      case 1: LAUNCH_NF4(1, 8); break;
+      case 2: LAUNCH_NF4(2, 8); break;
+      case 3: LAUNCH_NF4(3, 8); break;
+      case 4: LAUNCH_NF4(4, 8); break;
+    }
+  } else {
+    switch (batch) {
+      case 1: LAUNCH_NF4(1, 16); break;
+      case 2: LAUNCH_NF4(2, 16); break;
+      case 3: LAUNCH_NF4(3, 16); break;
+      case 4: LAUNCH_NF4(4, 16); break;
+    }
   }
 #undef LAUNCH_NF4
   HIP_CHECK_LAST();
