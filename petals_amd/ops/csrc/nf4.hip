@@ -88,7 +88,10 @@ __global__ void nf4_dequant_kernel(
 // OPL = outputs per lane (16 or 8). OPL=8 doubles the workgroup count for the
 // same split depth: the NF4 inner loop's dependent LDS gathers need more
 // resident waves per SIMD to hide latency than the direct-load bf16 gemv.
-template <int BATCH, int OPL>
+// LUTBF: 4-byte bf16-pair LUT entries instead of 8-byte float2 — the LUT
+// gathers read 8x the packed HBM bytes from LDS, so halving the entry width
+// halves the dominant LDS-bandwidth term (2 extra VALU bit-ops per pair).
+template <int BATCH, int OPL, bool LUTBF>
 __global__ void gemv_nf4_kernel(
     const unsigned char* __restrict__ packed,   // [in, out/2]
     const unsigned short* __restrict__ absmax,  // [in, out/64]
@@ -101,17 +104,36 @@ __global__ void gemv_nf4_kernel(
     int in_dim,
     int out_dim,
     int i_per_split) {
-  // PAIR LUT: one 8-byte read dequantizes a whole packed byte (two elements),
-  // halving LDS traffic vs per-nibble lookups; 256 float2 entries = 2 KiB.
+  // PAIR LUT: one read dequantizes a whole packed byte (two elements),
+  // halving LDS traffic vs per-nibble lookups; 256 entries.
   // (the earlier 16-float bank-replicated LUT was LDS-issue bound)
-  __shared__ float2 lut2[256];
-  for (int i = threadIdx.x; i < 256; i += blockDim.x)
-    lut2[i] = make_float2(NF4_LUT_C[i & 0xF], NF4_LUT_C[i >> 4]);
+  __shared__ float2 lut2[LUTBF ? 1 : 256];
+  __shared__ unsigned int lutp[LUTBF ? 256 : 1];
+  for (int i = threadIdx.x; i < 256; i += blockDim.x) {
+    if constexpr (LUTBF) {
+      lutp[i] = ((unsigned int)f32_to_bf16(NF4_LUT_C[i >> 4]) << 16) |
+                (unsigned int)f32_to_bf16(NF4_LUT_C[i & 0xF]);
+    } else {
+      lut2[i] = make_float2(NF4_LUT_C[i & 0xF], NF4_LUT_C[i >> 4]);
+    }
+  }
   __syncthreads();
+
+  auto lut_pair = [&](unsigned int byte) -> float2 {
+    if constexpr (LUTBF) {
+      const unsigned int wp = lutp[byte];
+      float2 r;
+      r.x = __uint_as_float(wp << 16);          // low nibble (bf16 -> f32 bits)
+      r.y = __uint_as_float(wp & 0xFFFF0000u);  // high nibble
+      return r;
+    } else {
+      return lut2[byte];
+    }
+  };
 
   constexpr int WORDS = OPL / 8;  // u32 packed words per lane per row
   const int lane = threadIdx.x & (WAVE - 1);
-#define NF4_L2(byte) lut2[(byte)]
+#define NF4_L2(byte) lut_pair(byte)
   const int out0 = blockIdx.x * (WAVE * OPL) + lane * OPL;
   if (out0 >= out_dim) return;
   const int split = blockIdx.y;
@@ -289,6 +311,10 @@ torch::Tensor gemv_nf4(
   }();
   const int opl = opl_env ? opl_env : 8;
   TORCH_CHECK(opl == 8 || opl == 16, "PETALS_NF4_OPL must be 8 or 16");
+  static const bool lut_bf16 = [] {
+    const char* s = std::getenv("PETALS_NF4_LUT");
+    return !(s && s[0] == 'f');  // default bf16-pair; PETALS_NF4_LUT=f32 for A/B
+  }();
 
   const long out_waves = (out_dim + (long)WAVE * opl - 1) / ((long)WAVE * opl);
   // NF4 matrices are 4x smaller than bf16: allow chunks down to 64 input rows
@@ -315,26 +341,24 @@ torch::Tensor gemv_nf4(
                 && absmax_t->size(1) == in_dim);
     amt_p = reinterpret_cast<const unsigned short*>(absmax_t->data_ptr());
   }
-#define LAUNCH_NF4(B, OPL)                                                    \
-  gemv_nf4_kernel<B, OPL><<<grid, WAVE, 0, stream>>>(                         \
+#define LAUNCH_NF4(B, OPL, LB)                                                \
+  gemv_nf4_kernel<B, OPL, LB><<<grid, WAVE, 0, stream>>>(                     \
       packed.data_ptr<unsigned char>(),                                       \
       reinterpret_cast<const unsigned short*>(absmax.data_ptr()), amt_p,      \
       x.data_ptr<float>(), partials.data_ptr<float>(), in_dim, out_dim, i_per_split)
-  if (opl == 8) {
-    switch (batch) {
-      case 1: LAUNCH_NF4(1, 8); break;
-      case 2: LAUNCH_NF4(2, 8); break;
-      case 3: LAUNCH_NF4(3, 8); break;
-      case 4: LAUNCH_NF4(4, 8); break;
-    }
-  } else {
-    switch (batch) {
-      case 1: LAUNCH_NF4(1, 16); break;
-      case 2: LAUNCH_NF4(2, 16); break;
-      case 3: LAUNCH_NF4(3, 16); break;
-      case 4: LAUNCH_NF4(4, 16); break;
-    }
+#define LAUNCH_NF4_B(OPL, LB)                                                 \
+  switch (batch) {                                                            \
+    case 1: LAUNCH_NF4(1, OPL, LB); break;                                    \
+    case 2: LAUNCH_NF4(2, OPL, LB); break;                                    \
+    case 3: LAUNCH_NF4(3, OPL, LB); break;                                    \
+    case 4: LAUNCH_NF4(4, OPL, LB); break;                                    \
   }
+  if (opl == 8) {
+    if (lut_bf16) { LAUNCH_NF4_B(8, true) } else { LAUNCH_NF4_B(8, false) }
+  } else {
+    if (lut_bf16) { LAUNCH_NF4_B(16, true) } else { LAUNCH_NF4_B(16, false) }
+  }
+#undef LAUNCH_NF4_B
 #undef LAUNCH_NF4
   HIP_CHECK_LAST();
 
