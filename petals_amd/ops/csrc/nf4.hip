@@ -179,11 +179,18 @@ __global__ void gemv_nf4_kernel(
         for (int u = 0; u < UNROLL; ++u)
           am[u] = bf16_to_f32(absmax[(size_t)(i + u) * (out_dim >> 6) + (out0 >> 6)]);
       }
+      // x loads MUST be vector-memory (vmcnt), not scalar (SMEM): SMEM shares
+      // the lgkm counter with the LDS LUT gathers and completes out of order,
+      // so any outstanding s_load forces lgkmcnt(0) before EVERY ds_read use
+      // (disassembly showed 2 ds_reads per full wait). volatile keeps the
+      // uniform-address loads out of the scalar unit.
       float xs[BATCH][UNROLL];
 #pragma unroll
-      for (int b = 0; b < BATCH; ++b)
+      for (int b = 0; b < BATCH; ++b) {
+        const volatile float* xvol = x + (size_t)b * in_dim + i;
 #pragma unroll
-        for (int u = 0; u < UNROLL; ++u) xs[b][u] = x[(size_t)b * in_dim + i + u];
+        for (int u = 0; u < UNROLL; ++u) xs[b][u] = xvol[u];
+      }
 #pragma unroll
       for (int u = 0; u < UNROLL; ++u) {
         float wf[OPL];
@@ -313,7 +320,9 @@ torch::Tensor gemv_nf4(
   TORCH_CHECK(opl == 8 || opl == 16, "PETALS_NF4_OPL must be 8 or 16");
   static const bool lut_bf16 = [] {
     const char* s = std::getenv("PETALS_NF4_LUT");
-    return !(s && s[0] == 'f');  // default bf16-pair; PETALS_NF4_LUT=f32 for A/B
+    return s && s[0] == 'b';  // default f32 pairs (bf16-pair measured SLOWER:
+    // the VALU unpack cost more than the halved LDS traffic bought,
+    // profiles/nf4_lut_sweep.log — LDS bandwidth is not the binding limit)
   }();
 
   const long out_waves = (out_dim + (long)WAVE * opl - 1) / ((long)WAVE * opl);
