@@ -179,18 +179,16 @@ __global__ void gemv_nf4_kernel(
         for (int u = 0; u < UNROLL; ++u)
           am[u] = bf16_to_f32(absmax[(size_t)(i + u) * (out_dim >> 6) + (out0 >> 6)]);
       }
-      // x loads MUST be vector-memory (vmcnt), not scalar (SMEM): SMEM shares
-      // the lgkm counter with the LDS LUT gathers and completes out of order,
-      // so any outstanding s_load forces lgkmcnt(0) before EVERY ds_read use
-      // (disassembly showed 2 ds_reads per full wait). volatile keeps the
-      // uniform-address loads out of the scalar unit.
+      // NB: scalar (SMEM) x loads share the lgkm counter with the LDS LUT
+      // gathers, limiting ds_read pipelining — but both measured alternatives
+      // (volatile VMEM: cache-bypassed broadcast re-reads; buffer-load
+      // intrinsics: worse scheduling) LOST to this form end to end
+      // (profiles/nf4_vmem_sweep.log), so the simple loads stay.
       float xs[BATCH][UNROLL];
 #pragma unroll
-      for (int b = 0; b < BATCH; ++b) {
-        const volatile float* xvol = x + (size_t)b * in_dim + i;
+      for (int b = 0; b < BATCH; ++b)
 #pragma unroll
-        for (int u = 0; u < UNROLL; ++u) xs[b][u] = xvol[u];
-      }
+        for (int u = 0; u < UNROLL; ++u) xs[b][u] = x[(size_t)b * in_dim + i + u];
 #pragma unroll
       for (int u = 0; u < UNROLL; ++u) {
         float wf[OPL];
