@@ -140,8 +140,11 @@ torch::Tensor gemv_bf16(
   const int batch = x.size(0);
   TORCH_CHECK(batch <= 8, "decode gemv supports batch <= 8");
 
-  const int splits = splits_override > 0 ? (int)splits_override : pick_splits(in_dim, out_dim);
-  const int i_per_split = (in_dim + splits - 1) / splits;
+  int splits = splits_override > 0 ? (int)splits_override : pick_splits(in_dim, out_dim);
+  // UNROLL(16)-aligned chunks: unaligned splits push rows onto the slow
+  // per-row tail path (same fix as the NF4 gemv)
+  const int i_per_split = (int)(((in_dim + splits - 1) / splits + 15) & ~15);
+  splits = (in_dim + i_per_split - 1) / i_per_split;
 
   torch::Tensor partials;
   if (workspace.numel() >= (int64_t)splits * batch * out_dim) {

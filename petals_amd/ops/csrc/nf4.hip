@@ -330,6 +330,10 @@ torch::Tensor gemv_nf4(
   long max_splits = (in_dim + 31) / 32;
   if (splits > max_splits) splits = max_splits;
   if (splits < 1) splits = 1;
+  // chunks must be UNROLL(16)-aligned or the per-row fallback tail eats the
+  // gain (an unaligned split count put up to a third of rows on the slow path)
+  int i_per_split_aligned = (int)(((in_dim + splits - 1) / splits + 15) & ~15L);
+  splits = (in_dim + i_per_split_aligned - 1) / i_per_split_aligned;
 
   torch::Tensor partials;
   auto f32opts = x.options();
@@ -338,7 +342,7 @@ torch::Tensor gemv_nf4(
   } else {
     partials = torch::empty({(int64_t)splits, batch, out_dim}, f32opts);
   }
-  const int i_per_split = (in_dim + splits - 1) / splits;
+  const int i_per_split = i_per_split_aligned;
   dim3 grid(out_waves, splits);
   auto stream = at::cuda::getCurrentCUDAStream();
 

@@ -116,7 +116,7 @@ class _FastWeight:
         import time
 
         if self.quant == "nf4":
-            candidates = [0, 32, 64, 128, 192, 256]
+            candidates = [0, 32, 64, 96, 128, 160, 192, 256, 320, 384]
             max_chunk = 32
         elif self.quant == "int8":
             candidates = [0, 16, 32, 64, 128, 192]

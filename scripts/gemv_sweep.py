@@ -57,9 +57,9 @@ def nf4_sweep():
         x = torch.randn(1, in_dim, device="cuda")
         gb = (packed.numel() + absmax.numel() * 2) / 1e9
         amt = absmax.t().contiguous()
-        for label, amt_arg in (("strided-am", None), ("am_t", amt)):
+        for label, amt_arg in (("am_t", amt),):
             row, best = [], (0, 0)
-            for splits in (0, 16, 32, 64, 128, 192, 256, 448):
+            for splits in (0, 16, 32, 64, 96, 128, 160, 192, 256, 320, 448):
                 if splits * 16 > in_dim and splits:
                     continue
                 t = bench(lambda: hip.gemv_nf4(packed, absmax, x, ws, None, 0, splits, None, amt_arg))
