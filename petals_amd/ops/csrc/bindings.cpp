@@ -16,6 +16,10 @@ void rope_cache_write(
 void kv_cache_write(
     torch::Tensor qkv, torch::Tensor pos,
     torch::Tensor k_cache, torch::Tensor v_cache, int64_t qh, int64_t kh);
+torch::Tensor qkv_rope_reduce(
+    torch::Tensor partials, c10::optional<torch::Tensor> cos_t, c10::optional<torch::Tensor> sin_t,
+    torch::Tensor pos, torch::Tensor k_cache, torch::Tensor v_cache,
+    int64_t qh, int64_t kh, bool rope, c10::optional<torch::Tensor> bias);
 torch::Tensor gemv_bf16(
     torch::Tensor wt, torch::Tensor x, torch::Tensor workspace,
     c10::optional<torch::Tensor> residual, int64_t epilogue, int64_t splits_override,
@@ -55,6 +59,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("apply_rope", &apply_rope, "rotate q,k by positions (bf16)");
   m.def("rope_cache_write", &rope_cache_write, "fused decode rope + kv cache write");
   m.def("kv_cache_write", &kv_cache_write, "decode kv cache write without rope (ALiBi families)");
+  m.def("qkv_rope_reduce", &qkv_rope_reduce,
+        "fused qkv split-K reduce + rope + kv-cache write (consumes RAW gemv partials)",
+        py::arg("partials"), py::arg("cos_t"), py::arg("sin_t"), py::arg("pos"),
+        py::arg("k_cache"), py::arg("v_cache"), py::arg("qh"), py::arg("kh"), py::arg("rope") = true,
+        py::arg("bias") = py::none());
   m.def("gemv_bf16", &gemv_bf16, "split-K bf16 gemv with fused epilogue",
         py::arg("wt"), py::arg("x"), py::arg("workspace"), py::arg("residual"), py::arg("epilogue"),
         py::arg("splits") = 0, py::arg("bias") = py::none());

@@ -8,15 +8,16 @@
 
 // --------------------------------------------------------------- rms_norm
 
-// one workgroup (256 threads) per row; row held in registers (short8 chunks)
-template <bool OUT_F32>
+// one workgroup per row; row held in registers (short8 chunks). T=1024 for
+// single-row decode calls (one wg is the whole launch — fewer serial chunk
+// phases shorten the latency chain), T=256 for many-row prefill.
+template <bool OUT_F32, int T>
 __global__ void rms_norm_kernel(
     const unsigned short* __restrict__ x,  // [rows, dim] bf16
     const unsigned short* __restrict__ w,  // [dim] bf16
     void* __restrict__ y,                  // [rows, dim] bf16 or f32
     int dim,
     float eps) {
-  constexpr int T = 256;
   const int row = blockIdx.x;
   const int tid = threadIdx.x;
   const unsigned short* xr = x + (size_t)row * dim;
@@ -85,10 +86,16 @@ torch::Tensor rms_norm(torch::Tensor x, torch::Tensor w, double eps) {
   const int rows = x2.size(0);
   auto y = torch::empty_like(x2);
   auto stream = at::cuda::getCurrentCUDAStream();
-  rms_norm_kernel<false><<<rows, 256, 0, stream>>>(
-      reinterpret_cast<const unsigned short*>(x2.data_ptr()),
-      reinterpret_cast<const unsigned short*>(w.contiguous().data_ptr()),
-      y.data_ptr(), dim, (float)eps);
+  if (rows <= 4)
+    rms_norm_kernel<false, 1024><<<rows, 1024, 0, stream>>>(
+        reinterpret_cast<const unsigned short*>(x2.data_ptr()),
+        reinterpret_cast<const unsigned short*>(w.contiguous().data_ptr()),
+        y.data_ptr(), dim, (float)eps);
+  else
+    rms_norm_kernel<false, 256><<<rows, 256, 0, stream>>>(
+        reinterpret_cast<const unsigned short*>(x2.data_ptr()),
+        reinterpret_cast<const unsigned short*>(w.contiguous().data_ptr()),
+        y.data_ptr(), dim, (float)eps);
   HIP_CHECK_LAST();
   return y.view(shape);
 }
@@ -101,10 +108,16 @@ torch::Tensor rms_norm_f32out(torch::Tensor x, torch::Tensor w, double eps) {
   const int rows = x2.size(0);
   auto y = torch::empty({rows, dim}, x.options().dtype(torch::kFloat32));
   auto stream = at::cuda::getCurrentCUDAStream();
-  rms_norm_kernel<true><<<rows, 256, 0, stream>>>(
-      reinterpret_cast<const unsigned short*>(x2.data_ptr()),
-      reinterpret_cast<const unsigned short*>(w.contiguous().data_ptr()),
-      y.data_ptr(), dim, (float)eps);
+  if (rows <= 4)
+    rms_norm_kernel<true, 1024><<<rows, 1024, 0, stream>>>(
+        reinterpret_cast<const unsigned short*>(x2.data_ptr()),
+        reinterpret_cast<const unsigned short*>(w.contiguous().data_ptr()),
+        y.data_ptr(), dim, (float)eps);
+  else
+    rms_norm_kernel<true, 256><<<rows, 256, 0, stream>>>(
+        reinterpret_cast<const unsigned short*>(x2.data_ptr()),
+        reinterpret_cast<const unsigned short*>(w.contiguous().data_ptr()),
+        y.data_ptr(), dim, (float)eps);
   HIP_CHECK_LAST();
   return y.view(shape);
 }
@@ -115,7 +128,7 @@ torch::Tensor rms_norm_f32out(torch::Tensor x, torch::Tensor w, double eps) {
 // BLOOM/Falcon blocks use LayerNorm with bias where Llama uses RMSNorm
 // (reference models/bloom/block.py wraps HF BloomBlock whose norms run in
 // torch; here the norm is one fused kernel feeding the gemv in f32).
-template <bool OUT_F32>
+template <bool OUT_F32, int T>
 __global__ void layer_norm_kernel(
     const unsigned short* __restrict__ x,  // [rows, dim] bf16
     const unsigned short* __restrict__ w,  // [dim] bf16
@@ -123,7 +136,6 @@ __global__ void layer_norm_kernel(
     void* __restrict__ y,                  // [rows, dim] bf16 or f32
     int dim,
     float eps) {
-  constexpr int T = 256;
   const int row = blockIdx.x;
   const int tid = threadIdx.x;
   const unsigned short* xr = x + (size_t)row * dim;
@@ -207,10 +219,17 @@ static torch::Tensor layer_norm_impl(torch::Tensor x, torch::Tensor w, torch::Te
   auto* xp = reinterpret_cast<const unsigned short*>(x2.data_ptr());
   auto* wp = reinterpret_cast<const unsigned short*>(w.contiguous().data_ptr());
   auto* bp = reinterpret_cast<const unsigned short*>(b.contiguous().data_ptr());
-  if (f32out)
-    layer_norm_kernel<true><<<rows, 256, 0, stream>>>(xp, wp, bp, y.data_ptr(), dim, (float)eps);
-  else
-    layer_norm_kernel<false><<<rows, 256, 0, stream>>>(xp, wp, bp, y.data_ptr(), dim, (float)eps);
+  if (rows <= 4) {
+    if (f32out)
+      layer_norm_kernel<true, 1024><<<rows, 1024, 0, stream>>>(xp, wp, bp, y.data_ptr(), dim, (float)eps);
+    else
+      layer_norm_kernel<false, 1024><<<rows, 1024, 0, stream>>>(xp, wp, bp, y.data_ptr(), dim, (float)eps);
+  } else {
+    if (f32out)
+      layer_norm_kernel<true, 256><<<rows, 256, 0, stream>>>(xp, wp, bp, y.data_ptr(), dim, (float)eps);
+    else
+      layer_norm_kernel<false, 256><<<rows, 256, 0, stream>>>(xp, wp, bp, y.data_ptr(), dim, (float)eps);
+  }
   HIP_CHECK_LAST();
   return f32out ? y : y.view(shape);
 }
@@ -395,6 +414,136 @@ __global__ void rope_cache_write_kernel(
       v_cache[(((size_t)bi * kh + kh_i) * lmax + p) * hd + d] = f32_to_bf16(v);
     }
   }
+}
+
+// ---------------- fused qkv split-K reduce + rope + kv-cache write ----------
+// Consumes the qkv gemv's RAW partials ([splits, b, qh*hd+2*kh*hd]) and in ONE
+// kernel: sums the splits (same 64-outputs x G split-group structure as
+// gemv_reduce), rotates q/k pairs by the device-resident position, writes
+// rotated q to a fresh [b, qh*hd] f32 tensor and k/v (bf16) into the caches.
+// Replaces reduce + rope_cache_write (or reduce + kv_cache_write for ALiBi
+// families) — one fewer kernel per block on the decode path.
+#define QKV_RED_GROUPS 8
+
+template <bool ROPE>
+__global__ __launch_bounds__(64 * QKV_RED_GROUPS) void qkv_rope_reduce_kernel(
+    const float* __restrict__ partials,  // [splits, b, row_elems]
+    const float* __restrict__ cos_t,     // [max_pos, hd]
+    const float* __restrict__ sin_t,
+    const int* __restrict__ pos_ptr,
+    const unsigned short* __restrict__ bias,  // [row_elems] bf16 or null
+    float* __restrict__ q_out,             // [b, qh*hd] f32 (rotated)
+    unsigned short* __restrict__ k_cache,  // [bcap, kh, lmax, hd]
+    unsigned short* __restrict__ v_cache,
+    int n_splits, int b, int qh, int kh, int lmax, int hd) {
+  constexpr int G = QKV_RED_GROUPS;
+  __shared__ float acc_a[G][64];
+  __shared__ float acc_b[G][64];
+  const int half = hd >> 1;
+  const int row_elems = qh * hd + 2 * kh * hd;
+  const int total = b * row_elems;
+  const int ol = threadIdx.x & 63;
+  const int sg = threadIdx.x >> 6;
+  const int idx = blockIdx.x * 64 + ol;
+  const bool live = idx < total;
+  const int bi = live ? idx / row_elems : 0;
+  const int o = live ? idx - bi * row_elems : 0;
+  const size_t stride = (size_t)b * row_elems;
+  const size_t row = (size_t)bi * row_elems;
+
+  // does this output need its rotation partner's sum too?
+  const bool in_q = o < qh * hd;
+  const bool in_k = !in_q && o < qh * hd + kh * hd;
+  const int d = in_q ? (o % hd) : (in_k ? ((o - qh * hd) % hd) : 0);
+  const bool rotate = ROPE && (in_q || in_k);
+  // partner offset: +half for d<half, -half otherwise
+  const int partner = rotate ? (d < half ? o + half : o - half) : o;
+
+  float a = 0.f, p2 = 0.f;
+  if (live) {
+    for (int s = sg; s < n_splits; s += G) a += partials[(size_t)s * stride + row + o];
+    if (rotate)
+      for (int s = sg; s < n_splits; s += G) p2 += partials[(size_t)s * stride + row + partner];
+  }
+  acc_a[sg][ol] = a;
+  acc_b[sg][ol] = p2;
+  __syncthreads();
+  if (sg != 0 || !live) return;
+  float sum = 0.f, psum = 0.f;
+#pragma unroll
+  for (int k = 0; k < G; ++k) {
+    sum += acc_a[k][ol];
+    psum += acc_b[k][ol];
+  }
+  if (bias) {
+    sum += bf16_to_f32(bias[o]);
+    if (rotate) psum += bf16_to_f32(bias[partner]);
+  }
+
+  const int p = *pos_ptr;
+  float val = sum;
+  if (rotate) {
+    const float c = cos_t[(size_t)p * hd + (d < half ? d : d - half)];
+    const float s = sin_t[(size_t)p * hd + (d < half ? d : d - half)];
+    // pair (x1, x2) at (d, d+half): out_d = x1*c - x2*s; out_{d+half} = x2*c + x1*s
+    val = (d < half) ? (sum * c - psum * s) : (sum * c + psum * s);
+  }
+  if (in_q) {
+    q_out[(size_t)bi * qh * hd + o] = val;
+  } else if (in_k) {
+    const int kh_i = (o - qh * hd) / hd;
+    k_cache[(((size_t)bi * kh + kh_i) * lmax + p) * hd + d] = f32_to_bf16(val);
+  } else {
+    const int t = o - qh * hd - kh * hd;
+    const int kh_i = t / hd;
+    const int dv = t - kh_i * hd;
+    v_cache[(((size_t)bi * kh + kh_i) * lmax + p) * hd + dv] = f32_to_bf16(val);
+  }
+}
+
+torch::Tensor qkv_rope_reduce(
+    torch::Tensor partials,  // [splits, b, row_elems] f32 (RAW gemv output)
+    c10::optional<torch::Tensor> cos_t,
+    c10::optional<torch::Tensor> sin_t,
+    torch::Tensor pos,       // device int32 [1]
+    torch::Tensor k_cache,   // [bcap, kh, lmax, hd] bf16
+    torch::Tensor v_cache,
+    int64_t qh, int64_t kh, bool rope,
+    c10::optional<torch::Tensor> bias) {
+  TORCH_CHECK(partials.is_cuda() && partials.dtype() == torch::kFloat32 && partials.dim() == 3);
+  const int n_splits = partials.size(0);
+  const int b = partials.size(1);
+  const int kh_i = k_cache.size(1), lmax = k_cache.size(2), hd = k_cache.size(3);
+  TORCH_CHECK(kh_i == kh);
+  TORCH_CHECK(partials.size(2) == qh * hd + 2 * kh * hd, "qkv width mismatch");
+  auto q_out = torch::empty({(int64_t)b, qh * hd}, partials.options());
+  const int total = b * (int)partials.size(2);
+  const int blocks = (total + 63) / 64;
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const float* cp = nullptr;
+  const float* sp = nullptr;
+  if (rope) {
+    TORCH_CHECK(cos_t.has_value() && sin_t.has_value(), "rope needs cos/sin tables");
+    cp = cos_t->data_ptr<float>();
+    sp = sin_t->data_ptr<float>();
+  }
+  const unsigned short* bp = nullptr;
+  if (bias.has_value() && bias->defined() && bias->numel() > 0) {
+    TORCH_CHECK(bias->is_contiguous() && bias->dtype() == torch::kBFloat16);
+    TORCH_CHECK(bias->numel() == partials.size(2));
+    bp = reinterpret_cast<const unsigned short*>(bias->data_ptr());
+  }
+#define LAUNCH_QRR(R)                                                          \
+  qkv_rope_reduce_kernel<R><<<blocks, 64 * QKV_RED_GROUPS, 0, stream>>>(       \
+      partials.data_ptr<float>(), cp, sp, pos.data_ptr<int>(), bp,             \
+      q_out.data_ptr<float>(),                                                 \
+      reinterpret_cast<unsigned short*>(k_cache.data_ptr()),                   \
+      reinterpret_cast<unsigned short*>(v_cache.data_ptr()),                   \
+      n_splits, b, (int)qh, (int)kh, lmax, hd)
+  if (rope) LAUNCH_QRR(true); else LAUNCH_QRR(false);
+#undef LAUNCH_QRR
+  HIP_CHECK_LAST();
+  return q_out;
 }
 
 // kv cache write WITHOUT rope for ALiBi families (BLOOM): same [q|k|v] fused

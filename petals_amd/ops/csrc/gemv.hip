@@ -179,6 +179,12 @@ torch::Tensor gemv_bf16(
   HIP_CHECK_LAST();
 
   // epilogue (shared with the NF4 path)
+  if (epilogue < 0) {
+    // RAW mode: the caller fuses its own reduce+epilogue (e.g. the qkv
+    // rope+cache-write reduce); the view's shape carries the split count
+    return partials.view(-1).narrow(0, 0, (int64_t)splits * batch * out_dim)
+        .view({(int64_t)splits, (int64_t)batch, (int64_t)out_dim});
+  }
   torch::Tensor y = launch_gemv_reduce(
       partials, residual, bias, splits, batch, out_dim, (int)epilogue, x.options(), wt.options());
   return y;

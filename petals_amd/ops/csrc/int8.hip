@@ -164,6 +164,12 @@ torch::Tensor gemv_int8(
 #undef LAUNCH_I8
   HIP_CHECK_LAST();
 
+  if (epilogue < 0) {
+    // RAW mode: the caller fuses its own reduce+epilogue (e.g. the qkv
+    // rope+cache-write reduce); the view's shape carries the split count
+    return partials.view(-1).narrow(0, 0, (int64_t)splits * batch * out_dim)
+        .view({(int64_t)splits, (int64_t)batch, (int64_t)out_dim});
+  }
   torch::Tensor y = launch_gemv_reduce(
       partials, residual, bias, (int)splits, batch, out_dim, (int)epilogue, f32opts, scale.options());
   return y;

@@ -28,6 +28,7 @@ import torch
 
 from petals_amd.ops import reference
 
+_EPI_RAW = -1  # return split partials; caller fuses its own reduce+epilogue
 _EPI_PLAIN_F32 = 0
 _EPI_PLAIN_BF16 = 1
 _EPI_RESIDUAL_BF16 = 2
@@ -346,15 +347,22 @@ class LlamaFastPath:
         ws = _get_ws(self.device, "gemv", 64 * B * self._max_gemv_out())
 
         xn = self.hip.rms_norm_f32out(h, self.ln1_w, self.eps)  # [B, H] f32
-        qkv = self.wqkv_t.gemv(xn, ws, None, _EPI_PLAIN_F32)  # [B, qkv] f32
-        if adapter is not None:
+        if adapter is None:
+            # fused qkv reduce + rope + cache write: one kernel fewer per block
+            parts = self.wqkv_t.gemv(xn, ws, None, _EPI_RAW)
+            q = self.hip.qkv_rope_reduce(
+                parts, self.rope_cos, self.rope_sin, pos, k_cache[:B], v_cache[:B],
+                self.qh, self.kh, True,
+            )  # [B, qh*hd] f32, rotated
+        else:
+            qkv = self.wqkv_t.gemv(xn, ws, None, _EPI_PLAIN_F32)  # [B, qkv] f32
             self._apply_qkv_adapter(qkv, xn, adapter)
-        self.hip.rope_cache_write(
-            qkv, self.rope_cos, self.rope_sin, pos, k_cache[:B], v_cache[:B], self.qh, self.kh
-        )
-        q = qkv[:, : self.qh * self.hd]
+            self.hip.rope_cache_write(
+                qkv, self.rope_cos, self.rope_sin, pos, k_cache[:B], v_cache[:B], self.qh, self.kh
+            )
+            q = qkv[:, : self.qh * self.hd].contiguous()
         attn = self.hip.attn_decode_fused(
-            q.contiguous(), k_cache[:B], v_cache[:B], kv_len, self.gq, 0,
+            q, k_cache[:B], v_cache[:B], kv_len, self.gq, 0,
             self._empty_f32, self._empty_f32, self.scale,
         )  # [B, qh*hd] f32 (the shard's heads under TP)
         if self.tp_world > 1:
@@ -616,11 +624,13 @@ class BloomFastPath:
 
         xn = self.hip.layer_norm_f32out(h, self.ln1_w, self.ln1_b, self.eps)
         res1 = self.hip.layer_norm(h, self.ln1_w, self.ln1_b, self.eps) if self.post_ln_residual else h
-        qkv = self.wqkv_t.gemv(xn, ws, None, _EPI_PLAIN_F32, bias=self.qkv_bias)
-        self.hip.kv_cache_write(qkv, pos, k_cache[:B], v_cache[:B], self.qh, self.kh)
-        q = qkv[:, : self.qh * self.hd]
+        parts = self.wqkv_t.gemv(xn, ws, None, _EPI_RAW)
+        q = self.hip.qkv_rope_reduce(
+            parts, None, None, pos, k_cache[:B], v_cache[:B], self.qh, self.kh, False,
+            bias=self.qkv_bias,
+        )  # ALiBi family: reduce + bias + cache write, no rotation
         attn = self.hip.attn_decode_fused(
-            q.contiguous(), k_cache[:B], v_cache[:B], kv_len, self.gq, 0,
+            q, k_cache[:B], v_cache[:B], kv_len, self.gq, 0,
             self._empty_f32, self._empty_f32, self.scale, self.slopes,
         )
         h2 = self.wo_t.gemv(attn, ws, res1, _EPI_RESIDUAL_BF16, bias=self.o_bias)
@@ -798,13 +808,13 @@ class FalconFastPath:
 
         xn_attn = self.hip.layer_norm_f32out(h, self.ln_attn_w, self.ln_attn_b, self.eps)
         xn_mlp = xn_attn if self.single_ln else self.hip.layer_norm_f32out(h, self.ln_mlp_w, self.ln_mlp_b, self.eps)
-        qkv = self.wqkv_t.gemv(xn_attn, ws, None, _EPI_PLAIN_F32)
-        self.hip.rope_cache_write(
-            qkv, self.rope_cos, self.rope_sin, pos, k_cache[:B], v_cache[:B], self.qh, self.kh
+        parts = self.wqkv_t.gemv(xn_attn, ws, None, _EPI_RAW)
+        q = self.hip.qkv_rope_reduce(
+            parts, self.rope_cos, self.rope_sin, pos, k_cache[:B], v_cache[:B],
+            self.qh, self.kh, True,
         )
-        q = qkv[:, : self.qh * self.hd]
         attn = self.hip.attn_decode_fused(
-            q.contiguous(), k_cache[:B], v_cache[:B], kv_len, self.gq, 0,
+            q, k_cache[:B], v_cache[:B], kv_len, self.gq, 0,
             self._empty_f32, self._empty_f32, self.scale,
         )
         h2 = self.wo_t.gemv(attn, ws, h, _EPI_RESIDUAL_BF16)  # resid + attn
