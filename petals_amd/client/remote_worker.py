@@ -1,15 +1,24 @@
-"""A dedicated asyncio worker thread for client-side RPC coroutines.
+"""A dedicated asyncio worker loop for client-side RPC coroutines.
 
 Parity with hivemind's RemoteExpertWorker (used at reference
 client/sequential_autograd.py:239): RPC work must NOT run on the DHT's event
 loop — routing code called from inside an RPC coroutine blocks on DHT lookups,
 which would deadlock a shared loop. The worker owns its own P2PNode (outbound
 connections only).
+
+Two execution modes:
+* INLINE (default): the CALLING thread pumps the worker loop via
+  run_until_complete, serialized by a lock. For the per-token decode path
+  this removes two thread wake-ups per step (submit + result) — worth
+  ~0.2-0.3 ms/token against a 13 ms step. Concurrent callers fall back to
+  blocking on the lock (client sessions are sequential by nature).
+* THREAD (PETALS_AMD_WORKER_THREAD=1): the round-1 background-thread design.
 """
 
 from __future__ import annotations
 
 import asyncio
+import os
 import threading
 from typing import Optional
 
@@ -23,8 +32,16 @@ class RemoteWorker:
         self._thread: Optional[threading.Thread] = None
         self._ready = threading.Event()
         self._lock = threading.Lock()
+        self._inline = not os.environ.get("PETALS_AMD_WORKER_THREAD")
+        self._inline_lock = threading.RLock()
 
     def _ensure_started(self):
+        if self._inline:
+            with self._lock:
+                if self.loop is None:
+                    self.loop = asyncio.new_event_loop()
+                    self.p2p = P2PNode()
+            return
         with self._lock:
             if self._thread is not None and self._thread.is_alive():
                 return
@@ -43,6 +60,12 @@ class RemoteWorker:
 
     def run_coroutine(self, coro, timeout: Optional[float] = None):
         self._ensure_started()
+        if self._inline:
+            with self._inline_lock:
+                if timeout is not None:
+                    coro = asyncio.wait_for(coro, timeout)
+                asyncio.set_event_loop(self.loop)
+                return self.loop.run_until_complete(coro)
         future = asyncio.run_coroutine_threadsafe(coro, self.loop)
         try:
             return future.result(timeout)
@@ -51,6 +74,16 @@ class RemoteWorker:
             raise
 
     def shutdown(self):
+        if self._inline:
+            with self._lock:
+                if self.loop is not None and not self.loop.is_closed() and not self.loop.is_running():
+                    try:
+                        self.loop.close()
+                    except Exception:  # noqa: BLE001
+                        pass
+                self.loop = None
+                self.p2p = None
+            return
         if self.loop is not None:
             self.loop.call_soon_threadsafe(self.loop.stop)
         if self._thread is not None:

@@ -334,6 +334,7 @@ class TransformerConnectionHandler:
         session: Optional[_Session] = None,
         keep_on_device: bool = False,
         commit_cb=None,
+        reply_cb=None,
     ) -> torch.Tensor:
         """Runs IN the runtime thread: one inference step through the whole span
         (the single-process analog of reference _MergedInferenceStep)."""
@@ -399,11 +400,22 @@ class TransformerConnectionHandler:
                 # replay; anything leaving the runtime must own its storage
                 res = out.clone() if keep_on_device else out.cpu()
                 _commit_if_needed(commit_cb, res)
+                if reply_cb is not None:
+                    if res.is_cuda:
+                        torch.cuda.current_stream().synchronize()
+                    reply_cb(res)
                 print(f"[rt] graph {( _g1-_g0)*1e3:.2f} out {(time.perf_counter()-_g1)*1e3:.2f} ms", flush=True)
                 return res
             out = session.span_graph.step(hidden_states, prefix_length)
             out = out.clone() if keep_on_device else out.cpu()
             _commit_if_needed(commit_cb, out)
+            if reply_cb is not None:
+                # the client reads this tensor on ITS stream: finish ours first
+                # (the scheduler syncs right after fn returns anyway, so this
+                # only moves the sync before the delivery, not adds one)
+                if out.is_cuda:
+                    torch.cuda.current_stream().synchronize()
+                reply_cb(out)
             return out
 
         if has_hypo:
@@ -420,6 +432,10 @@ class TransformerConnectionHandler:
                 (hidden_states,) = self.backends[uid].inference_step(hidden_states, hypo_ids, info)
         hidden_states = hidden_states if keep_on_device else hidden_states.cpu()
         _commit_if_needed(commit_cb, hidden_states)
+        if reply_cb is not None:
+            if hidden_states.is_cuda:
+                torch.cuda.current_stream().synchronize()
+            reply_cb(hidden_states)
         return hidden_states
 
     async def rpc_inference(self, request: RpcMessage, stream: RpcStream) -> None:
@@ -544,6 +560,24 @@ class TransformerConnectionHandler:
                                     logger.warning("mesh out pre-announce failed (%r); stream", e)
                                     commit_cb = abort_cb = None
 
+                    # direct reply from the runtime thread: a co-located
+                    # (in-process) client gets its output the moment compute
+                    # finishes, without waking the server loop first
+                    reply_cb = None
+                    replied_directly = False
+                    if (
+                        length_increment > 0
+                        and not will_push
+                        and self._mesh_out_dst(output_via_mesh) is None
+                        and getattr(stream, "is_inproc", False)
+                    ):
+                        _sid = step_meta.get("step_id")
+                        _peer = stream.peer
+
+                        def reply_cb(out, _sid=_sid, _peer=_peer):
+                            _peer._deliver(RpcMessage(meta={"step_id": _sid}, tensors=[out]))
+
+                        replied_directly = True
                     if length_increment > 0:
                         priority = self.prioritizer.prioritize(hidden_states, type="inference")
                         try:
@@ -560,6 +594,7 @@ class TransformerConnectionHandler:
                                 session,
                                 keep_on_device,
                                 commit_cb,
+                                reply_cb,
                             )
                         except BaseException as e:
                             if abort_cb is not None:
@@ -577,6 +612,8 @@ class TransformerConnectionHandler:
                         print(f"[srv] compute {(time.perf_counter()-_t0)*1e3:.2f} ms", flush=True)
                     session.prefix_length += length_increment
 
+                    if replied_directly:
+                        continue  # the runtime thread already delivered the reply
                     if pre_kind == "out":
                         continue  # meta already on the stream; data on the mesh
 
