@@ -23,8 +23,15 @@ class BloomAttention(nn.Module):
         self.num_heads = config.num_attention_heads
         self.head_dim = config.head_dim
         self.hidden_size = config.hidden_size
-        self.query_key_value = nn.Linear(self.hidden_size, 3 * self.hidden_size, bias=True)
-        self.dense = nn.Linear(self.hidden_size, self.hidden_size, bias=True)
+        # shard-safe sizes: for a full model num_heads*head_dim == hidden_size;
+        # a TP shard keeps hidden_size (replicated input) with fewer heads
+        qkv_out = 3 * self.num_heads * self.head_dim
+        self.query_key_value = nn.Linear(self.hidden_size, qkv_out, bias=True)
+        self.dense = nn.Linear(self.num_heads * self.head_dim, self.hidden_size, bias=True)
+        # ALiBi slopes depend on the GLOBAL head index: TP shard configs carry
+        # their slice window (alibi_start_head / alibi_total_heads)
+        self._alibi_total = getattr(config, "alibi_total_heads", None) or self.num_heads
+        self._alibi_start = getattr(config, "alibi_start_head", 0)
         self._alibi_slopes = None  # lazy: blocks may be built on the meta device
 
     def forward(self, hidden_states, kv_cache=None, prefix_length: int = 0):
@@ -36,7 +43,10 @@ class BloomAttention(nn.Module):
 
         end = prefix_length + q_len
         if self._alibi_slopes is None or self._alibi_slopes.device != hidden_states.device:
-            self._alibi_slopes = ops.build_alibi_slopes(self.num_heads).to(hidden_states.device)
+            full = ops.build_alibi_slopes(self._alibi_total)
+            self._alibi_slopes = full[self._alibi_start : self._alibi_start + self.num_heads].to(
+                hidden_states.device
+            )
         slopes = self._alibi_slopes
         if kv_cache is not None:
             k_cache, v_cache = kv_cache
@@ -46,7 +56,7 @@ class BloomAttention(nn.Module):
         else:
             assert prefix_length == 0
             attn = ops.attention(q, k, v, causal=True, alibi_slopes=slopes)
-        attn = attn.transpose(1, 2).reshape(b, q_len, self.hidden_size)
+        attn = attn.transpose(1, 2).reshape(b, q_len, self.num_heads * self.head_dim)
         return self.dense(attn)
 
 
@@ -72,6 +82,8 @@ class BloomBlock(nn.Module):
         self.apply_residual_post_ln = config.apply_residual_connection_post_layernorm
 
     _fast = None  # BloomFastPath after optimize_for_inference()
+    tp_world = 1  # set by parallel/tp.py TPBloomBlock shards
+    tp_group = None
 
     def optimize_for_inference(self, quant: str = "none") -> "BloomBlock":
         """Repack weights into the MI355X kernel layout (LayerNorm + ALiBi +
@@ -93,7 +105,7 @@ class BloomBlock(nn.Module):
                 self.config.head_dim,
             )
             return self
-        self._fast = BloomFastPath(self, hip, quant=quant)
+        self._fast = BloomFastPath(self, hip, quant=quant, tp_world=self.tp_world, tp_group=self.tp_group)
         return self
 
     def forward(self, hidden_states, kv_cache=None, prefix_length: int = 0, ctx=None):

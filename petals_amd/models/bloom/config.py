@@ -3,7 +3,7 @@
 from __future__ import annotations
 
 import dataclasses
-from typing import Any, Dict
+from typing import Any, Dict, Optional
 
 from petals_amd.models.config_base import ModelConfig, register_config
 
@@ -13,6 +13,10 @@ from petals_amd.models.config_base import ModelConfig, register_config
 class BloomConfig(ModelConfig):
     apply_residual_connection_post_layernorm: bool = False
     block_prefix: str = "h"  # state dict path: transformer.h.{i}
+    # TP shard configs: ALiBi slopes depend on the GLOBAL head index, so a
+    # shard carries its slice window over the full model's slope table
+    alibi_start_head: int = 0
+    alibi_total_heads: Optional[int] = None
 
     def __post_init__(self):
         if self.intermediate_size is None:

@@ -28,12 +28,14 @@ class FalconAttention(nn.Module):
         self.num_kv = config.n_kv_heads
         self.head_dim = config.head_dim
         h = config.hidden_size
+        # shard-safe sizes: a TP shard keeps hidden_size (replicated input)
+        # with fewer heads, so derive output widths from heads, not h
         if config.new_decoder_architecture:
             qkv_out = (self.num_heads + 2 * self.num_kv) * self.head_dim
         elif config.multi_query:
-            qkv_out = h + 2 * self.head_dim
+            qkv_out = (self.num_heads + 2) * self.head_dim
         else:
-            qkv_out = 3 * h
+            qkv_out = 3 * self.num_heads * self.head_dim
         self.query_key_value = nn.Linear(h, qkv_out, bias=config.bias)
         self.dense = nn.Linear(self.num_heads * self.head_dim, h, bias=config.bias)
         self.rope_cos = None  # lazy (meta-device construction)
@@ -109,6 +111,8 @@ class FalconBlock(nn.Module):
                 self.post_attention_layernorm = nn.LayerNorm(config.hidden_size, eps=eps)
 
     _fast = None  # FalconFastPath after optimize_for_inference()
+    tp_world = 1  # set by parallel/tp.py TPFalconBlock shards
+    tp_group = None
 
     # (head_dim, gq) combos instantiated in ops/csrc/attention.hip ATTN_CASE /
     # ATTN_CASE_BIG — gq > 16 runs the multi-group MFMA decode (ceil(gq/16)
@@ -148,7 +152,7 @@ class FalconBlock(nn.Module):
                 cfg.new_decoder_architecture, cfg.parallel_attn, cfg.head_dim, gq, cfg.bias,
             )
             return self
-        self._fast = FalconFastPath(self, hip, quant=quant)
+        self._fast = FalconFastPath(self, hip, quant=quant, tp_world=self.tp_world, tp_group=self.tp_group)
         return self
 
     def forward(self, hidden_states, kv_cache=None, prefix_length: int = 0, ctx=None):

@@ -128,6 +128,18 @@ PRESETS = {
         layer_norm_epsilon=1e-5, parallel_attn=True, new_decoder_architecture=False,
         multi_query=True, bias=False,
     ),
+    "test-falcon-mqa": dict(  # PRIME head count (5): exercises uneven TP head sharding
+        model_type="falcon", hidden_size=80, num_hidden_layers=2, num_attention_heads=5,
+        vocab_size=128, max_position_embeddings=256,
+        layer_norm_epsilon=1e-5, parallel_attn=True, new_decoder_architecture=False,
+        multi_query=True, bias=False,
+    ),
+    "test-falcon-classic": dict(  # rw-style: sequential residual, biased linears
+        model_type="falcon", hidden_size=64, num_hidden_layers=2, num_attention_heads=4,
+        vocab_size=128, max_position_embeddings=256,
+        layer_norm_epsilon=1e-5, parallel_attn=False, new_decoder_architecture=False,
+        multi_query=False, bias=True,
+    ),
     "test-mixtral": dict(
         model_type="mixtral", hidden_size=64, num_hidden_layers=4, num_attention_heads=4,
         num_key_value_heads=2, intermediate_size=128, vocab_size=128,

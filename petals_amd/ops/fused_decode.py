@@ -219,7 +219,38 @@ def _dequant_cached(w, make) -> torch.Tensor:
     return t
 
 
-class LlamaFastPath:
+class _TPFastPathMixin:
+    """Tensor-parallel hooks shared by the per-family fast paths. The shard's
+    row-parallel outputs (o/down/dense) skip the fused residual epilogue and
+    are all-reduced before the residual add (parallel/tp.py)."""
+
+    tp_world = 1
+    tp_group = None
+
+    def _tp_copy(self, x: torch.Tensor) -> torch.Tensor:
+        """Column-parallel input boundary (identity fwd, grad all-reduce bwd)."""
+        if self.tp_world <= 1:
+            return x
+        from petals_amd.parallel.tp import copy_to_tp
+
+        return copy_to_tp(x, self.tp_group)
+
+    def _tp_allreduce_(self, t: torch.Tensor) -> None:
+        """In-place SUM across TP ranks (decode hot path; overridable in tests)."""
+        import torch.distributed as dist
+
+        dist.all_reduce(t, op=dist.ReduceOp.SUM, group=self.tp_group)
+
+    def _tp_reduce(self, x: torch.Tensor) -> torch.Tensor:
+        """Sum row-parallel partials across TP ranks (autograd-aware)."""
+        if self.tp_world <= 1:
+            return x
+        from petals_amd.parallel.tp import reduce_from_tp
+
+        return reduce_from_tp(x, self.tp_group)
+
+
+class LlamaFastPath(_TPFastPathMixin):
     graph_safe = True  # whole-span hipGraph capture is valid for this block
 
     def __init__(self, block, hip_ops, quant: str = "none", tp_world: int = 1, tp_group=None):
@@ -380,28 +411,6 @@ class LlamaFastPath:
         h3 = self._mlp_decode(xn2, h2, ws, adapter)
         return h3.view(B, 1, H)
 
-    def _tp_copy(self, x: torch.Tensor) -> torch.Tensor:
-        """Column-parallel input boundary (identity fwd, grad all-reduce bwd)."""
-        if self.tp_world <= 1:
-            return x
-        from petals_amd.parallel.tp import copy_to_tp
-
-        return copy_to_tp(x, self.tp_group)
-
-    def _tp_allreduce_(self, t: torch.Tensor) -> None:
-        """In-place SUM across TP ranks (decode hot path; overridable in tests)."""
-        import torch.distributed as dist
-
-        dist.all_reduce(t, op=dist.ReduceOp.SUM, group=self.tp_group)
-
-    def _tp_reduce(self, x: torch.Tensor) -> torch.Tensor:
-        """Sum row-parallel partials across TP ranks (autograd-aware)."""
-        if self.tp_world <= 1:
-            return x
-        from petals_amd.parallel.tp import reduce_from_tp
-
-        return reduce_from_tp(x, self.tp_group)
-
     def _max_gemv_out(self) -> int:
         return max(self.wgateup_t.shape[1], self.wqkv_t.shape[1], self.wo_t.shape[1])
 
@@ -543,7 +552,7 @@ def _bloom_qkv_perm(qh: int, hd: int) -> torch.Tensor:
     return perm
 
 
-class BloomFastPath:
+class BloomFastPath(_TPFastPathMixin):
     # The round-1 "bloom dual-graph hardware fault" (a bloom span graph +
     # the client's LM-head graph faulting on replay while each was clean in
     # isolation) was ROOT-CAUSED to the shared gemv workspace: bloom's 250k
@@ -554,11 +563,15 @@ class BloomFastPath:
     # test_bloom_dual_graph_with_big_vocab_head.
     graph_safe = True
 
-    def __init__(self, block, hip_ops, quant: str = "none"):
+    def __init__(self, block, hip_ops, quant: str = "none", tp_world: int = 1, tp_group=None):
         cfg = block.config
         self.hip = hip_ops
         self.cfg = cfg
         self.quant = quant
+        self.tp_world = tp_world
+        self.tp_group = tp_group
+        if tp_world > 1:
+            self.graph_safe = False  # RCCL inside hipGraph capture unvalidated
         self.hd = cfg.head_dim
         self.qh = cfg.num_attention_heads
         self.kh = self.qh  # MHA
@@ -594,7 +607,15 @@ class BloomFastPath:
         for lin in (attn.query_key_value, attn.dense, mlp.dense_h_to_4h, mlp.dense_4h_to_h):
             lin.weight.data = self._empty_bf16
 
-        self.slopes = reference.build_alibi_slopes(self.qh).to(device=device, dtype=torch.float32).contiguous()
+        # ALiBi slopes depend on the GLOBAL head index: a TP shard takes its
+        # slice of the full model's slope table
+        total = getattr(cfg, "alibi_total_heads", None) or self.qh
+        start = getattr(cfg, "alibi_start_head", 0)
+        self.slopes = (
+            reference.build_alibi_slopes(total)[start : start + self.qh]
+            .to(device=device, dtype=torch.float32)
+            .contiguous()
+        )
         self._pos = torch.zeros(1, dtype=torch.int32, device=device)
         self._kv_len = torch.zeros(1, dtype=torch.int32, device=device)
         self.device = device
@@ -633,11 +654,23 @@ class BloomFastPath:
             q, k_cache[:B], v_cache[:B], kv_len, self.gq, 0,
             self._empty_f32, self._empty_f32, self.scale, self.slopes,
         )
-        h2 = self.wo_t.gemv(attn, ws, res1, _EPI_RESIDUAL_BF16, bias=self.o_bias)
+        if self.tp_world > 1:
+            # row-parallel dense: partial (rank-0-only bias baked in by the TP
+            # shard loader), all-reduce, then the replicated residual add
+            part = self.wo_t.gemv(attn, ws, None, _EPI_PLAIN_F32, bias=self.o_bias)
+            self._tp_allreduce_(part)
+            h2 = (res1.float() + part).to(torch.bfloat16)
+        else:
+            h2 = self.wo_t.gemv(attn, ws, res1, _EPI_RESIDUAL_BF16, bias=self.o_bias)
         xn2 = self.hip.layer_norm_f32out(h2, self.ln2_w, self.ln2_b, self.eps)
         res2 = self.hip.layer_norm(h2, self.ln2_w, self.ln2_b, self.eps) if self.post_ln_residual else h2
         act = self.w_h4h.gemv(xn2, ws, None, _EPI_GELU_F32, bias=self.b_h4h)
-        h3 = self.w_4hh.gemv(act, ws, res2, _EPI_RESIDUAL_BF16, bias=self.b_4hh)
+        if self.tp_world > 1:
+            part = self.w_4hh.gemv(act, ws, None, _EPI_PLAIN_F32, bias=self.b_4hh)
+            self._tp_allreduce_(part)
+            h3 = (res2.float() + part).to(torch.bfloat16)
+        else:
+            h3 = self.w_4hh.gemv(act, ws, res2, _EPI_RESIDUAL_BF16, bias=self.b_4hh)
         return h3.view(B, 1, H)
 
     # -------------------------------------------------- prefill / training
@@ -662,6 +695,7 @@ class BloomFastPath:
 
         xn = ln(hidden, self.ln1_w, self.ln1_b)
         res1 = xn if self.post_ln_residual else hidden
+        xn = self._tp_copy(xn)
         qkv = torch.matmul(xn, self.wqkv_t.dense()) + self.qkv_bias
         q, k, v = self._split_heads(qkv, B, S)
         if kv_cache is not None and not autograd:
@@ -681,15 +715,16 @@ class BloomFastPath:
             bias = (self.slopes.to(hidden.device)[:, None, None] * k_pos[None, None, :]).unsqueeze(0)
             attn = reference.attention(q.float(), k.float(), v.float(), causal=True,
                                        kv_offset=prefix_length, attn_bias=bias.float())
-        attn = attn.transpose(1, 2).reshape(B, S, H).to(torch.bfloat16)
+        attn = attn.transpose(1, 2).reshape(B, S, self.qh * self.hd).to(torch.bfloat16)
         o = torch.matmul(attn, self.wo_t.dense()) + self.o_bias
-        h2 = res1 + o
+        h2 = res1 + self._tp_reduce(o)
         xn2 = ln(h2, self.ln2_w, self.ln2_b)
         res2 = xn2 if self.post_ln_residual else h2
+        xn2 = self._tp_copy(xn2)
         inter = torch.matmul(xn2, self.w_h4h.dense()) + self.b_h4h
         act = reference.gelu(inter.float()).to(torch.bfloat16)
         out = torch.matmul(act, self.w_4hh.dense()) + self.b_4hh
-        return res2 + out
+        return res2 + self._tp_reduce(out)
 
     def forward(self, hidden, kv_cache, prefix_length, adapter=None):
         assert adapter is None
@@ -725,11 +760,15 @@ def _falcon_qkv_perm(qh: int, kh: int, hd: int) -> torch.Tensor:
     return perm
 
 
-class FalconFastPath:
+class FalconFastPath(_TPFastPathMixin):
     graph_safe = True
 
-    def __init__(self, block, hip_ops, quant: str = "none"):
+    def __init__(self, block, hip_ops, quant: str = "none", tp_world: int = 1, tp_group=None):
         cfg = block.config
+        self.tp_world = tp_world
+        self.tp_group = tp_group
+        if tp_world > 1:
+            self.graph_safe = False  # RCCL inside hipGraph capture unvalidated
         # two fused geometries: the new-decoder architecture (40B/180B:
         # ln_attn + ln_mlp, GQA) and the 7B-style old decoder (single
         # input_layernorm feeding BOTH attn and mlp, parallel residual, MQA —
@@ -817,6 +856,16 @@ class FalconFastPath:
             q, k_cache[:B], v_cache[:B], kv_len, self.gq, 0,
             self._empty_f32, self._empty_f32, self.scale,
         )
+        if self.tp_world > 1:
+            # parallel residual (attn + mlp both row-parallel partials):
+            # sum the two partials locally, then ONE all-reduce per block
+            ap = self.wo_t.gemv(attn, ws, None, _EPI_PLAIN_F32)
+            act = self.w_h4h.gemv(xn_mlp, ws, None, _EPI_GELU_F32)
+            part = self.w_4hh.gemv(act, ws, None, _EPI_PLAIN_F32)
+            part.add_(ap)
+            self._tp_allreduce_(part)
+            h3 = (h.float() + part).to(torch.bfloat16)
+            return h3.view(B, 1, H)
         h2 = self.wo_t.gemv(attn, ws, h, _EPI_RESIDUAL_BF16)  # resid + attn
         act = self.w_h4h.gemv(xn_mlp, ws, None, _EPI_GELU_F32)
         h3 = self.w_4hh.gemv(act, ws, h2, _EPI_RESIDUAL_BF16)  # ... + mlp
@@ -841,8 +890,8 @@ class FalconFastPath:
                 return fln(x.float(), (H,), w.float(), b.float(), self.eps).to(torch.bfloat16)
             return self.hip.layer_norm(x, w, b, self.eps)
 
-        xn_attn = ln(hidden, self.ln_attn_w, self.ln_attn_b)
-        xn_mlp = xn_attn if self.single_ln else ln(hidden, self.ln_mlp_w, self.ln_mlp_b)
+        xn_attn = self._tp_copy(ln(hidden, self.ln_attn_w, self.ln_attn_b))
+        xn_mlp = xn_attn if self.single_ln else self._tp_copy(ln(hidden, self.ln_mlp_w, self.ln_mlp_b))
         qkv = torch.matmul(xn_attn, self.wqkv_t.dense())
         q, k, v = self._split_heads(qkv, B, S)
         pos = torch.arange(prefix_length, end, device=hidden.device)
@@ -869,7 +918,7 @@ class FalconFastPath:
         inter = torch.matmul(xn_mlp, self.w_h4h.dense())
         act = reference.gelu(inter.float()).to(torch.bfloat16)
         mlp_out = torch.matmul(act, self.w_4hh.dense())
-        return hidden + o + mlp_out
+        return hidden + self._tp_reduce(o + mlp_out)
 
     def forward(self, hidden, kv_cache, prefix_length, adapter=None):
         assert adapter is None

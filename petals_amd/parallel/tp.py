@@ -257,11 +257,216 @@ class TPLlamaBlock(nn.Module):
         return shape, shape
 
 
+# ---------------------------------------------------------------------------
+# Falcon / BLOOM tensor parallelism. Unlike llama, these families shard
+# UNEVENLY when the head count does not divide the TP degree (falcon-7b has a
+# PRIME 71 query heads): rank r takes a contiguous near-even slice of heads
+# (or of kv GROUPS for the falcon new-decoder architecture, whose fused-QKV
+# weight is group-major), so per-rank shard configs differ in head count but
+# all produce [*, hidden] row-parallel partials for the same all-reduce.
+# Falcon MQA replicates the single k/v head on every rank (its weight rows are
+# loaded whole); parallel-attention falcons need only ONE all-reduce per block
+# (attn partial + mlp partial summed locally first).
+# ---------------------------------------------------------------------------
+
+
+def _split_count(total: int, world: int, rank: int) -> Tuple[int, int]:
+    """Contiguous near-even partition: (start, count) for this rank."""
+    base, rem = divmod(total, world)
+    start = rank * base + min(rank, rem)
+    return start, base + (1 if rank < rem else 0)
+
+
+def shard_falcon_config(config, world: int, rank: int):
+    from petals_amd.models.falcon.config import FalconConfig  # noqa: F401
+
+    i_start, i_n = _split_count(config.intermediate_size, world, rank)
+    if config.new_decoder_architecture:
+        gq = config.num_attention_heads // config.n_kv_heads
+        g_start, g_n = _split_count(config.n_kv_heads, world, rank)
+        return dataclasses.replace(
+            config,
+            num_attention_heads=g_n * gq,
+            num_kv_heads=g_n,
+            intermediate_size=i_n,
+            head_dim_override=config.head_dim,
+        )
+    q_start, q_n = _split_count(config.num_attention_heads, world, rank)
+    return dataclasses.replace(
+        config,
+        num_attention_heads=q_n,
+        intermediate_size=i_n,
+        head_dim_override=config.head_dim,
+    )  # __post_init__ re-derives num_key_value_heads (1 for MQA, =heads classic)
+
+
+def shard_bloom_config(config, world: int, rank: int):
+    h_start, h_n = _split_count(config.num_attention_heads, world, rank)
+    i_start, i_n = _split_count(config.intermediate_size, world, rank)
+    return dataclasses.replace(
+        config,
+        num_attention_heads=h_n,
+        num_key_value_heads=h_n,
+        intermediate_size=i_n,
+        head_dim_override=config.head_dim,
+        alibi_start_head=h_start,
+        alibi_total_heads=config.num_attention_heads,
+    )
+
+
+def _resolve_rank_world(rank, world, group):
+    if rank is None:
+        rank = dist.get_rank(group) if dist.is_initialized() else 0
+    if world is None:
+        world = dist.get_world_size(group) if dist.is_initialized() else 1
+    return rank, world
+
+
+from petals_amd.models.bloom.block import BloomBlock as _BloomBlockBase  # noqa: E402
+from petals_amd.models.falcon.block import FalconBlock as _FalconBlockBase  # noqa: E402
+
+
+class TPFalconBlock(_FalconBlockBase):
+    """One rank's shard of a Falcon decoder block (all three architectures:
+    new-decoder GQA 40B/180B, MQA 7B, classic MHA rw). `self.config` reports
+    the SHARD geometry so `optimize_for_inference()` builds the same fused
+    FalconFastPath over the shard (tp hooks split the residual epilogue at the
+    single per-block all-reduce)."""
+
+    def __init__(self, config, layer_idx: int = 0, rank: Optional[int] = None,
+                 world: Optional[int] = None, group=None):
+        rank, world = _resolve_rank_world(rank, world, group)
+        shard = shard_falcon_config(config, world, rank)
+        super().__init__(shard, layer_idx)
+        self.full_config = config
+        self.rank, self.world = rank, world
+        self.tp_world, self.tp_group = world, group
+
+    def load_from_full_state_dict(self, sd: dict) -> None:
+        cfg, full = self.config, self.full_config
+        r, w, hd = self.rank, self.world, cfg.head_dim
+        qkv_w = sd["self_attention.query_key_value.weight"]
+        qkv_b = sd.get("self_attention.query_key_value.bias")
+        if full.new_decoder_architecture:
+            gq = full.num_attention_heads // full.n_kv_heads
+            g_start, g_n = _split_count(full.n_kv_heads, w, r)
+            rows = slice(g_start * (gq + 2) * hd, (g_start + g_n) * (gq + 2) * hd)
+            cols = slice(g_start * gq * hd, (g_start + g_n) * gq * hd)
+            qkv_rows_w = qkv_w[rows]
+            qkv_rows_b = qkv_b[rows] if qkv_b is not None else None
+        elif full.multi_query:
+            q_start, q_n = _split_count(full.num_attention_heads, w, r)
+            qrows = slice(q_start * hd, (q_start + q_n) * hd)
+            kvrows = slice(full.num_attention_heads * hd, (full.num_attention_heads + 2) * hd)
+            qkv_rows_w = torch.cat([qkv_w[qrows], qkv_w[kvrows]], dim=0)  # q shard + replicated k,v
+            qkv_rows_b = torch.cat([qkv_b[qrows], qkv_b[kvrows]], dim=0) if qkv_b is not None else None
+            cols = qrows
+        else:  # classic MHA: per-head [q,k,v] interleave, 3*hd rows per head
+            h_start, h_n = _split_count(full.num_attention_heads, w, r)
+            rows = slice(h_start * 3 * hd, (h_start + h_n) * 3 * hd)
+            cols = slice(h_start * hd, (h_start + h_n) * hd)
+            qkv_rows_w = qkv_w[rows]
+            qkv_rows_b = qkv_b[rows] if qkv_b is not None else None
+        a = self.self_attention
+        a.query_key_value.weight.data.copy_(qkv_rows_w)
+        a.dense.weight.data.copy_(sd["self_attention.dense.weight"][:, cols])
+        if full.bias:
+            a.query_key_value.bias.data.copy_(qkv_rows_b)
+            # row-parallel biases apply ONCE: rank 0 carries them, others zero
+            a.dense.bias.data.copy_(sd["self_attention.dense.bias"]) if r == 0 else a.dense.bias.data.zero_()
+        i_start, i_n = _split_count(full.intermediate_size, w, r)
+        irows = slice(i_start, i_start + i_n)
+        m = self.mlp
+        m.dense_h_to_4h.weight.data.copy_(sd["mlp.dense_h_to_4h.weight"][irows])
+        m.dense_4h_to_h.weight.data.copy_(sd["mlp.dense_4h_to_h.weight"][:, irows])
+        if full.bias:
+            m.dense_h_to_4h.bias.data.copy_(sd["mlp.dense_h_to_4h.bias"][irows])
+            m.dense_4h_to_h.bias.data.copy_(sd["mlp.dense_4h_to_h.bias"]) if r == 0 else m.dense_4h_to_h.bias.data.zero_()
+        for name in ("ln_attn", "ln_mlp", "input_layernorm", "post_attention_layernorm"):
+            mod = getattr(self, name, None)
+            if mod is not None:
+                mod.weight.data.copy_(sd[f"{name}.weight"])
+                mod.bias.data.copy_(sd[f"{name}.bias"])
+
+    def forward(self, hidden_states, kv_cache=None, prefix_length: int = 0, ctx=None):
+        if self._fast is not None:  # fused path reduces internally (tp hooks)
+            return super().forward(hidden_states, kv_cache=kv_cache, prefix_length=prefix_length, ctx=ctx)
+        g = self.tp_group
+        residual = hidden_states
+        if self.config.new_decoder_architecture:
+            attn_in = copy_to_tp(self.ln_attn(hidden_states), g)
+            mlp_in = copy_to_tp(self.ln_mlp(hidden_states), g)
+            part = self.self_attention(attn_in, kv_cache=kv_cache, prefix_length=prefix_length) + self.mlp(mlp_in)
+            return residual + reduce_from_tp(part, g)  # ONE reduce (parallel residual)
+        attn_in = copy_to_tp(self.input_layernorm(hidden_states), g)
+        attn = self.self_attention(attn_in, kv_cache=kv_cache, prefix_length=prefix_length)
+        if self.config.parallel_attn:
+            return residual + reduce_from_tp(attn + self.mlp(attn_in), g)
+        hidden_states = residual + reduce_from_tp(attn, g)
+        mlp_in = copy_to_tp(self.post_attention_layernorm(hidden_states), g)
+        return hidden_states + reduce_from_tp(self.mlp(mlp_in), g)
+
+
+class TPBloomBlock(_BloomBlockBase):
+    """One rank's shard of a BLOOM decoder block: heads sharded (ALiBi slopes
+    sliced by GLOBAL head index via the shard config's alibi window)."""
+
+    def __init__(self, config, layer_idx: int = 0, rank: Optional[int] = None,
+                 world: Optional[int] = None, group=None):
+        rank, world = _resolve_rank_world(rank, world, group)
+        shard = shard_bloom_config(config, world, rank)
+        super().__init__(shard, layer_idx)
+        self.full_config = config
+        self.rank, self.world = rank, world
+        self.tp_world, self.tp_group = world, group
+
+    def load_from_full_state_dict(self, sd: dict) -> None:
+        full = self.full_config
+        r, w, hd = self.rank, self.world, self.config.head_dim
+        h_start, h_n = _split_count(full.num_attention_heads, w, r)
+        rows = slice(h_start * 3 * hd, (h_start + h_n) * 3 * hd)  # per-head [q,k,v] interleave
+        cols = slice(h_start * hd, (h_start + h_n) * hd)
+        a = self.self_attention
+        a.query_key_value.weight.data.copy_(sd["self_attention.query_key_value.weight"][rows])
+        a.query_key_value.bias.data.copy_(sd["self_attention.query_key_value.bias"][rows])
+        a.dense.weight.data.copy_(sd["self_attention.dense.weight"][:, cols])
+        # row-parallel biases apply ONCE: rank 0 carries them, others zero
+        a.dense.bias.data.copy_(sd["self_attention.dense.bias"]) if r == 0 else a.dense.bias.data.zero_()
+        i_start, i_n = _split_count(full.intermediate_size, w, r)
+        irows = slice(i_start, i_start + i_n)
+        m = self.mlp
+        m.dense_h_to_4h.weight.data.copy_(sd["mlp.dense_h_to_4h.weight"][irows])
+        m.dense_h_to_4h.bias.data.copy_(sd["mlp.dense_h_to_4h.bias"][irows])
+        m.dense_4h_to_h.weight.data.copy_(sd["mlp.dense_4h_to_h.weight"][:, irows])
+        m.dense_4h_to_h.bias.data.copy_(sd["mlp.dense_4h_to_h.bias"]) if r == 0 else m.dense_4h_to_h.bias.data.zero_()
+        for name in ("input_layernorm", "post_attention_layernorm"):
+            mod = getattr(self, name)
+            mod.weight.data.copy_(sd[f"{name}.weight"])
+            mod.bias.data.copy_(sd[f"{name}.bias"])
+
+    def forward(self, hidden_states, kv_cache=None, prefix_length: int = 0, ctx=None):
+        if self._fast is not None:  # fused path reduces internally (tp hooks)
+            return super().forward(hidden_states, kv_cache=kv_cache, prefix_length=prefix_length, ctx=ctx)
+        g = self.tp_group
+        ln_out = self.input_layernorm(hidden_states)
+        res = ln_out if self.apply_residual_post_ln else hidden_states
+        attn = self.self_attention(copy_to_tp(ln_out, g), kv_cache=kv_cache, prefix_length=prefix_length)
+        h2 = res + reduce_from_tp(attn, g)
+        ln2 = self.post_attention_layernorm(h2)
+        res2 = ln2 if self.apply_residual_post_ln else h2
+        return res2 + reduce_from_tp(self.mlp(copy_to_tp(ln2, g)), g)
+
+
 def build_tp_block(config, layer_idx: int, *, rank: int, world: int, group=None) -> nn.Module:
-    """TP shard block factory. Llama-family geometry (GQA + SwiGLU) for now;
-    other families serve TP via the generic path or single-rank."""
+    """TP shard block factory (llama GQA+SwiGLU, falcon all three
+    architectures, bloom MHA+ALiBi). Other families (e.g. Mixtral, whose
+    experts parallelize better by expert than by column) serve single-rank."""
     if config.model_type in ("llama",):
         return TPLlamaBlock(config, layer_idx, rank=rank, world=world, group=group)
+    if config.model_type == "falcon":
+        return TPFalconBlock(config, layer_idx, rank=rank, world=world, group=group)
+    if config.model_type == "bloom":
+        return TPBloomBlock(config, layer_idx, rank=rank, world=world, group=group)
     raise NotImplementedError(f"tensor parallelism is not implemented for model_type={config.model_type!r}")
 
 

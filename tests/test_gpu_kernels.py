@@ -608,18 +608,28 @@ class _FakeTPAllReduce:
 
 
 @requires_gpu
-@pytest.mark.parametrize("quant", ["none", "nf4"])
-def test_tp_fused_shards_match_full_block(hip, quant):
-    """Two TP shards of a llama block (fused NF4/MFMA path, simulated
-    all-reduce) decode EXACTLY like the unsharded fused block."""
+@pytest.mark.parametrize(
+    "model,quant",
+    [
+        ("test-llama-hd128", "none"),
+        ("test-llama-hd128", "nf4"),
+        ("test-falcon-hd64", "none"),  # single-reduce parallel residual
+        ("test-falcon-hd64", "nf4"),
+        ("test-bloom-hd64", "none"),  # ALiBi slope slicing + rank-0 biases
+        ("test-bloom-hd64", "nf4"),
+    ],
+)
+def test_tp_fused_shards_match_full_block(hip, model, quant):
+    """Two TP shards of a block (fused NF4/MFMA path, simulated all-reduce)
+    decode EXACTLY like the unsharded fused block — llama, falcon and bloom."""
     import threading
 
     from petals_amd.models import get_model_block
     from petals_amd.models.config_base import load_model_config
-    from petals_amd.parallel.tp import TPLlamaBlock
+    from petals_amd.parallel.tp import build_tp_block
     from petals_amd.server.from_pretrained import init_random_block_
 
-    cfg = load_model_config("test-llama-hd128")
+    cfg = load_model_config(model)
     full = get_model_block(cfg, 0)
     init_random_block_(full, cfg, 0)
     sd = {k: v.clone() for k, v in full.state_dict().items()}
@@ -628,9 +638,10 @@ def test_tp_fused_shards_match_full_block(hip, quant):
     reducer = _FakeTPAllReduce(2)
     shards = []
     for r in range(2):
-        blk = TPLlamaBlock(cfg, 0, rank=r, world=2)
+        blk = build_tp_block(cfg, 0, rank=r, world=2)
         blk.load_from_full_state_dict(sd)
         blk = blk.to("cuda", torch.bfloat16).eval().optimize_for_inference(quant=quant)
+        assert blk._fast is not None, "TP shard must take the fused path"
         blk._fast._tp_allreduce_ = reducer
         shards.append(blk)
 

@@ -18,11 +18,11 @@ def _tp_worker(rank, world, port, payload_path, result_path):
     dist.init_process_group("gloo", rank=rank, world_size=world)
     try:
         from petals_amd.models.config_base import load_model_config
-        from petals_amd.parallel.tp import TPLlamaBlock
+        from petals_amd.parallel.tp import build_tp_block
 
         payload = torch.load(payload_path, weights_only=False)
-        cfg = load_model_config("test-llama")
-        block = TPLlamaBlock(cfg, 0, rank=rank, world=world)
+        cfg = load_model_config(payload.get("model", "test-llama"))
+        block = build_tp_block(cfg, 0, rank=rank, world=world)
         block.load_from_full_state_dict(payload["state_dict"])
         x = payload["x"]
 
@@ -47,12 +47,22 @@ def _tp_worker(rank, world, port, payload_path, result_path):
         dist.destroy_process_group()
 
 
-def test_tp_block_matches_full(tmp_path):
+@pytest.mark.parametrize(
+    "model",
+    [
+        "test-llama",
+        "test-falcon",  # new-decoder GQA (single-reduce parallel residual)
+        "test-falcon-mqa",  # MQA with a PRIME head count (uneven shards, replicated kv)
+        "test-falcon-classic",  # sequential residual + biased linears (rank-0 bias)
+        "test-bloom",  # MHA + ALiBi (global-head slope slicing)
+    ],
+)
+def test_tp_block_matches_full(tmp_path, model):
     from petals_amd.models import get_model_block
     from petals_amd.models.config_base import load_model_config
     from petals_amd.server.from_pretrained import init_random_block_
 
-    cfg = load_model_config("test-llama")
+    cfg = load_model_config(model)
     full = get_model_block(cfg, 0)
     init_random_block_(full, cfg, 0)
     torch.manual_seed(0)
@@ -63,7 +73,7 @@ def test_tp_block_matches_full(tmp_path):
 
     payload_path = str(tmp_path / "payload.pt")
     result_path = str(tmp_path / "result.pt")
-    torch.save({"state_dict": full.state_dict(), "x": x}, payload_path)
+    torch.save({"state_dict": full.state_dict(), "x": x, "model": model}, payload_path)
 
     import socket
 
