@@ -34,9 +34,14 @@ class BloomAttention(nn.Module):
         self._alibi_start = getattr(config, "alibi_start_head", 0)
         self._alibi_slopes = None  # lazy: blocks may be built on the meta device
 
-    def forward(self, hidden_states, kv_cache=None, prefix_length: int = 0):
+    def forward(self, hidden_states, kv_cache=None, prefix_length: int = 0, adapter=None):
         b, q_len, _ = hidden_states.shape
-        fused = self.query_key_value(hidden_states).view(b, q_len, self.num_heads, 3, self.head_dim)
+        fused = self.query_key_value(hidden_states)
+        if adapter is not None:
+            d = adapter.delta("qkv", hidden_states)
+            if d is not None:
+                fused = fused + d
+        fused = fused.view(b, q_len, self.num_heads, 3, self.head_dim)
         q = fused[..., 0, :].transpose(1, 2)  # [b, heads, q_len, hd]
         k = fused[..., 1, :].transpose(1, 2)
         v = fused[..., 2, :].transpose(1, 2)
@@ -57,7 +62,12 @@ class BloomAttention(nn.Module):
             assert prefix_length == 0
             attn = ops.attention(q, k, v, causal=True, alibi_slopes=slopes)
         attn = attn.transpose(1, 2).reshape(b, q_len, self.num_heads * self.head_dim)
-        return self.dense(attn)
+        out = self.dense(attn)
+        if adapter is not None:
+            d = adapter.delta("dense", attn)
+            if d is not None:
+                out = out + d
+        return out
 
 
 class BloomMLP(nn.Module):
@@ -66,8 +76,19 @@ class BloomMLP(nn.Module):
         self.dense_h_to_4h = nn.Linear(config.hidden_size, config.intermediate_size, bias=True)
         self.dense_4h_to_h = nn.Linear(config.intermediate_size, config.hidden_size, bias=True)
 
-    def forward(self, x):
-        return self.dense_4h_to_h(ops.gelu(self.dense_h_to_4h(x)))
+    def forward(self, x, adapter=None):
+        inter = self.dense_h_to_4h(x)
+        if adapter is not None:
+            d = adapter.delta("h4h", x)
+            if d is not None:
+                inter = inter + d
+        act = ops.gelu(inter)
+        out = self.dense_4h_to_h(act)
+        if adapter is not None:
+            d = adapter.delta("4hh", act)
+            if d is not None:
+                out = out + d
+        return out
 
 
 class BloomBlock(nn.Module):
@@ -109,25 +130,28 @@ class BloomBlock(nn.Module):
         return self
 
     def forward(self, hidden_states, kv_cache=None, prefix_length: int = 0, ctx=None):
+        from petals_amd.utils.peft import active_block_adapter
+
+        adapter = active_block_adapter(self)
         if self._fast is not None:
             if torch.is_grad_enabled() and hidden_states.requires_grad:
                 assert kv_cache is None, "training forward does not use the KV cache"
-                return self._fast.forward_autograd(hidden_states, prefix_length)
+                return self._fast.forward_autograd(hidden_states, prefix_length, adapter=adapter)
             max_b = 4 if self._fast.quant == "nf4" else 8
             if kv_cache is not None and hidden_states.shape[1] == 1 and hidden_states.shape[0] <= max_b:
                 return self._fast.decode_step(
-                    hidden_states, kv_cache[0], kv_cache[1], prefix_length, ctx=ctx
+                    hidden_states, kv_cache[0], kv_cache[1], prefix_length, ctx=ctx, adapter=adapter
                 )
-            return self._fast.forward(hidden_states, kv_cache, prefix_length)
+            return self._fast.forward(hidden_states, kv_cache, prefix_length, adapter=adapter)
 
         ln_out = self.input_layernorm(hidden_states)
         residual = ln_out if self.apply_residual_post_ln else hidden_states
-        attn = self.self_attention(ln_out, kv_cache=kv_cache, prefix_length=prefix_length)
+        attn = self.self_attention(ln_out, kv_cache=kv_cache, prefix_length=prefix_length, adapter=adapter)
         hidden_states = residual + attn
 
         ln_out = self.post_attention_layernorm(hidden_states)
         residual = ln_out if self.apply_residual_post_ln else hidden_states
-        return residual + self.mlp(ln_out)
+        return residual + self.mlp(ln_out, adapter=adapter)
 
     def kv_cache_shape(self, batch_size: int, max_length: int) -> Tuple[Tuple[int, ...], Tuple[int, ...]]:
         shape = (batch_size, self.config.num_attention_heads, max_length, self.config.head_dim)

@@ -65,9 +65,14 @@ class FalconAttention(nn.Module):
             )
             self.rope_cos, self.rope_sin = cos.to(device), sin.to(device)
 
-    def forward(self, hidden_states, kv_cache=None, prefix_length: int = 0):
+    def forward(self, hidden_states, kv_cache=None, prefix_length: int = 0, adapter=None):
         b, q_len, _ = hidden_states.shape
-        q, k, v = self._split_qkv(self.query_key_value(hidden_states), b, q_len)
+        fused = self.query_key_value(hidden_states)
+        if adapter is not None:
+            d = adapter.delta("qkv", hidden_states)
+            if d is not None:
+                fused = fused + d
+        q, k, v = self._split_qkv(fused, b, q_len)
         end = prefix_length + q_len
         self._ensure_rope(end, hidden_states.device)
         pos = torch.arange(prefix_length, end, device=hidden_states.device)
@@ -81,7 +86,12 @@ class FalconAttention(nn.Module):
             assert prefix_length == 0
             attn = ops.attention(q, k, v, causal=True)
         attn = attn.transpose(1, 2).reshape(b, q_len, self.num_heads * self.head_dim)
-        return self.dense(attn)
+        out = self.dense(attn)
+        if adapter is not None:
+            d = adapter.delta("dense", attn)
+            if d is not None:
+                out = out + d
+        return out
 
 
 class FalconMLP(nn.Module):
@@ -90,8 +100,19 @@ class FalconMLP(nn.Module):
         self.dense_h_to_4h = nn.Linear(config.hidden_size, config.intermediate_size, bias=config.bias)
         self.dense_4h_to_h = nn.Linear(config.intermediate_size, config.hidden_size, bias=config.bias)
 
-    def forward(self, x):
-        return self.dense_4h_to_h(ops.gelu(self.dense_h_to_4h(x)))
+    def forward(self, x, adapter=None):
+        inter = self.dense_h_to_4h(x)
+        if adapter is not None:
+            d = adapter.delta("h4h", x)
+            if d is not None:
+                inter = inter + d
+        act = ops.gelu(inter)
+        out = self.dense_4h_to_h(act)
+        if adapter is not None:
+            d = adapter.delta("4hh", act)
+            if d is not None:
+                out = out + d
+        return out
 
 
 class FalconBlock(nn.Module):
@@ -156,30 +177,33 @@ class FalconBlock(nn.Module):
         return self
 
     def forward(self, hidden_states, kv_cache=None, prefix_length: int = 0, ctx=None):
+        from petals_amd.utils.peft import active_block_adapter
+
+        adapter = active_block_adapter(self)
         if self._fast is not None:
             if torch.is_grad_enabled() and hidden_states.requires_grad:
                 assert kv_cache is None, "training forward does not use the KV cache"
-                return self._fast.forward_autograd(hidden_states, prefix_length)
+                return self._fast.forward_autograd(hidden_states, prefix_length, adapter=adapter)
             max_b = 4 if self._fast.quant == "nf4" else 8
             if kv_cache is not None and hidden_states.shape[1] == 1 and hidden_states.shape[0] <= max_b:
                 return self._fast.decode_step(
-                    hidden_states, kv_cache[0], kv_cache[1], prefix_length, ctx=ctx
+                    hidden_states, kv_cache[0], kv_cache[1], prefix_length, ctx=ctx, adapter=adapter
                 )
-            return self._fast.forward(hidden_states, kv_cache, prefix_length)
+            return self._fast.forward(hidden_states, kv_cache, prefix_length, adapter=adapter)
 
         residual = hidden_states
         if self.config.new_decoder_architecture:
             attn_in = self.ln_attn(hidden_states)
             mlp_in = self.ln_mlp(hidden_states)
-            attn = self.self_attention(attn_in, kv_cache=kv_cache, prefix_length=prefix_length)
-            return residual + attn + self.mlp(mlp_in)
+            attn = self.self_attention(attn_in, kv_cache=kv_cache, prefix_length=prefix_length, adapter=adapter)
+            return residual + attn + self.mlp(mlp_in, adapter=adapter)
         attn_in = self.input_layernorm(hidden_states)
-        attn = self.self_attention(attn_in, kv_cache=kv_cache, prefix_length=prefix_length)
+        attn = self.self_attention(attn_in, kv_cache=kv_cache, prefix_length=prefix_length, adapter=adapter)
         if self.config.parallel_attn:
-            return residual + attn + self.mlp(attn_in)
+            return residual + attn + self.mlp(attn_in, adapter=adapter)
         hidden_states = residual + attn
         mlp_in = self.post_attention_layernorm(hidden_states)
-        return hidden_states + self.mlp(mlp_in)
+        return hidden_states + self.mlp(mlp_in, adapter=adapter)
 
     def kv_cache_shape(self, batch_size: int, max_length: int) -> Tuple[Tuple[int, ...], Tuple[int, ...]]:
         shape = (batch_size, self.config.n_kv_heads, max_length, self.config.head_dim)

@@ -590,6 +590,7 @@ class BloomFastPath(_TPFastPathMixin):
             return lin.bias.detach().to(torch.bfloat16).contiguous()
 
         perm = _bloom_qkv_perm(self.qh, self.hd).to(device)
+        self._qkv_perm = perm  # HF->flat [q|k|v] column map (LoRA deltas reuse it)
         self.wqkv_t = _FastWeight(t(attn.query_key_value.weight)[:, perm].contiguous(), hip_ops, quant)
         self.qkv_bias = attn.query_key_value.bias.detach().to(torch.bfloat16)[perm].contiguous()
         self.wo_t = _FastWeight(t(attn.dense.weight), hip_ops, quant)
@@ -628,7 +629,6 @@ class BloomFastPath(_TPFastPathMixin):
 
     @torch.inference_mode()
     def decode_step(self, hidden, k_cache, v_cache, prefix_length: int = -1, ctx=None, adapter=None):
-        assert adapter is None, "LoRA adapters are served on the generic path for BLOOM"
         B, H = hidden.shape[0], hidden.shape[-1]
         h = hidden.view(B, H)
         if h.dtype != torch.bfloat16:
@@ -645,11 +645,20 @@ class BloomFastPath(_TPFastPathMixin):
 
         xn = self.hip.layer_norm_f32out(h, self.ln1_w, self.ln1_b, self.eps)
         res1 = self.hip.layer_norm(h, self.ln1_w, self.ln1_b, self.eps) if self.post_ln_residual else h
-        parts = self.wqkv_t.gemv(xn, ws, None, _EPI_RAW)
-        q = self.hip.qkv_rope_reduce(
-            parts, None, None, pos, k_cache[:B], v_cache[:B], self.qh, self.kh, False,
-            bias=self.qkv_bias,
-        )  # ALiBi family: reduce + bias + cache write, no rotation
+        if adapter is None:
+            parts = self.wqkv_t.gemv(xn, ws, None, _EPI_RAW)
+            q = self.hip.qkv_rope_reduce(
+                parts, None, None, pos, k_cache[:B], v_cache[:B], self.qh, self.kh, False,
+                bias=self.qkv_bias,
+            )  # ALiBi family: reduce + bias + cache write, no rotation
+        else:
+            assert self.tp_world == 1, "LoRA on TP shards is served via the generic path"
+            qkv = self.wqkv_t.gemv(xn, ws, None, _EPI_PLAIN_F32, bias=self.qkv_bias)
+            d = adapter.delta("qkv", xn)
+            if d is not None:
+                qkv += d.float()[:, self._qkv_perm]  # deltas arrive in the HF column order
+            self.hip.kv_cache_write(qkv, pos, k_cache[:B], v_cache[:B], self.qh, self.kh)
+            q = qkv[:, : self.qh * self.hd].contiguous()
         attn = self.hip.attn_decode_fused(
             q, k_cache[:B], v_cache[:B], kv_len, self.gq, 0,
             self._empty_f32, self._empty_f32, self.scale, self.slopes,
@@ -662,15 +671,30 @@ class BloomFastPath(_TPFastPathMixin):
             h2 = (res1.float() + part).to(torch.bfloat16)
         else:
             h2 = self.wo_t.gemv(attn, ws, res1, _EPI_RESIDUAL_BF16, bias=self.o_bias)
+            if adapter is not None:
+                dd = adapter.delta("dense", attn)
+                if dd is not None:
+                    h2 = h2 + dd.to(h2.dtype)
         xn2 = self.hip.layer_norm_f32out(h2, self.ln2_w, self.ln2_b, self.eps)
         res2 = self.hip.layer_norm(h2, self.ln2_w, self.ln2_b, self.eps) if self.post_ln_residual else h2
-        act = self.w_h4h.gemv(xn2, ws, None, _EPI_GELU_F32, bias=self.b_h4h)
+        if adapter is not None:
+            inter = self.w_h4h.gemv(xn2, ws, None, _EPI_PLAIN_F32, bias=self.b_h4h)
+            dh = adapter.delta("h4h", xn2)
+            if dh is not None:
+                inter += dh.float()
+            act = reference.gelu(inter).contiguous()
+        else:
+            act = self.w_h4h.gemv(xn2, ws, None, _EPI_GELU_F32, bias=self.b_h4h)
         if self.tp_world > 1:
             part = self.w_4hh.gemv(act, ws, None, _EPI_PLAIN_F32, bias=self.b_4hh)
             self._tp_allreduce_(part)
             h3 = (res2.float() + part).to(torch.bfloat16)
         else:
             h3 = self.w_4hh.gemv(act, ws, res2, _EPI_RESIDUAL_BF16, bias=self.b_4hh)
+            if adapter is not None:
+                d4 = adapter.delta("4hh", act)
+                if d4 is not None:
+                    h3 = h3 + d4.to(h3.dtype)
         return h3.view(B, 1, H)
 
     # -------------------------------------------------- prefill / training
@@ -682,7 +706,7 @@ class BloomFastPath(_TPFastPathMixin):
         v = qkv[..., 2 * qd :].view(B, S, self.kh, self.hd).transpose(1, 2)
         return q, k, v
 
-    def _body(self, hidden, kv_cache, prefix_length, autograd: bool):
+    def _body(self, hidden, kv_cache, prefix_length, autograd: bool, adapter=None):
         B, S, H = hidden.shape
         hidden = hidden.to(torch.bfloat16)
         end = prefix_length + S
@@ -697,6 +721,10 @@ class BloomFastPath(_TPFastPathMixin):
         res1 = xn if self.post_ln_residual else hidden
         xn = self._tp_copy(xn)
         qkv = torch.matmul(xn, self.wqkv_t.dense()) + self.qkv_bias
+        if adapter is not None:
+            d = adapter.delta("qkv", xn)
+            if d is not None:
+                qkv = qkv + d.to(qkv.dtype)[..., self._qkv_perm]
         q, k, v = self._split_heads(qkv, B, S)
         if kv_cache is not None and not autograd:
             k_cache, v_cache = kv_cache
@@ -717,22 +745,34 @@ class BloomFastPath(_TPFastPathMixin):
                                        kv_offset=prefix_length, attn_bias=bias.float())
         attn = attn.transpose(1, 2).reshape(B, S, self.qh * self.hd).to(torch.bfloat16)
         o = torch.matmul(attn, self.wo_t.dense()) + self.o_bias
+        if adapter is not None:
+            d = adapter.delta("dense", attn)
+            if d is not None:
+                o = o + d.to(o.dtype)
         h2 = res1 + self._tp_reduce(o)
         xn2 = ln(h2, self.ln2_w, self.ln2_b)
         res2 = xn2 if self.post_ln_residual else h2
         xn2 = self._tp_copy(xn2)
         inter = torch.matmul(xn2, self.w_h4h.dense()) + self.b_h4h
+        if adapter is not None:
+            dh = adapter.delta("h4h", xn2)
+            if dh is not None:
+                inter = inter + dh.to(inter.dtype)
         act = reference.gelu(inter.float()).to(torch.bfloat16)
         out = torch.matmul(act, self.w_4hh.dense()) + self.b_4hh
+        if adapter is not None:
+            d4 = adapter.delta("4hh", act)
+            if d4 is not None:
+                out = out + d4.to(out.dtype)
         return res2 + self._tp_reduce(out)
 
     def forward(self, hidden, kv_cache, prefix_length, adapter=None):
-        assert adapter is None
-        return self._body(hidden, kv_cache, prefix_length, autograd=False)
+        assert adapter is None or self.tp_world == 1, "LoRA on TP shards is served via the generic path"
+        return self._body(hidden, kv_cache, prefix_length, autograd=False, adapter=adapter)
 
     def forward_autograd(self, hidden, prefix_length: int = 0, adapter=None):
-        assert adapter is None
-        return self._body(hidden, None, prefix_length, autograd=True)
+        assert adapter is None or self.tp_world == 1, "LoRA on TP shards is served via the generic path"
+        return self._body(hidden, None, prefix_length, autograd=True, adapter=adapter)
 
 
 # ---------------------------------------------------------------------------
@@ -793,6 +833,7 @@ class FalconFastPath(_TPFastPathMixin):
             return w.detach().to(torch.bfloat16).t().contiguous()
 
         perm = _falcon_qkv_perm(self.qh, self.kh, self.hd).to(device)
+        self._qkv_perm = perm  # HF->flat [q|k|v] column map (LoRA deltas reuse it)
         self.wqkv_t = _FastWeight(t(attn.query_key_value.weight)[:, perm].contiguous(), hip_ops, quant)
         self.wo_t = _FastWeight(t(attn.dense.weight), hip_ops, quant)
         mlp = block.mlp
@@ -828,7 +869,6 @@ class FalconFastPath(_TPFastPathMixin):
 
     @torch.inference_mode()
     def decode_step(self, hidden, k_cache, v_cache, prefix_length: int = -1, ctx=None, adapter=None):
-        assert adapter is None, "LoRA adapters are served on the generic path for Falcon"
         B, H = hidden.shape[0], hidden.shape[-1]
         h = hidden.view(B, H)
         if h.dtype != torch.bfloat16:
@@ -847,15 +887,41 @@ class FalconFastPath(_TPFastPathMixin):
 
         xn_attn = self.hip.layer_norm_f32out(h, self.ln_attn_w, self.ln_attn_b, self.eps)
         xn_mlp = xn_attn if self.single_ln else self.hip.layer_norm_f32out(h, self.ln_mlp_w, self.ln_mlp_b, self.eps)
-        parts = self.wqkv_t.gemv(xn_attn, ws, None, _EPI_RAW)
-        q = self.hip.qkv_rope_reduce(
-            parts, self.rope_cos, self.rope_sin, pos, k_cache[:B], v_cache[:B],
-            self.qh, self.kh, True,
-        )
+        if adapter is None:
+            parts = self.wqkv_t.gemv(xn_attn, ws, None, _EPI_RAW)
+            q = self.hip.qkv_rope_reduce(
+                parts, self.rope_cos, self.rope_sin, pos, k_cache[:B], v_cache[:B],
+                self.qh, self.kh, True,
+            )
+        else:
+            assert self.tp_world == 1, "LoRA on TP shards is served via the generic path"
+            qkv = self.wqkv_t.gemv(xn_attn, ws, None, _EPI_PLAIN_F32)  # flat [q|k|v] f32
+            d = adapter.delta("qkv", xn_attn)
+            if d is not None:
+                qkv += d.float()[:, self._qkv_perm]  # deltas arrive in the HF column order
+            self.hip.rope_cache_write(
+                qkv, self.rope_cos, self.rope_sin, pos, k_cache[:B], v_cache[:B], self.qh, self.kh
+            )
+            q = qkv[:, : self.qh * self.hd].contiguous()
         attn = self.hip.attn_decode_fused(
             q, k_cache[:B], v_cache[:B], kv_len, self.gq, 0,
             self._empty_f32, self._empty_f32, self.scale,
         )
+        if adapter is not None:
+            h2 = self.wo_t.gemv(attn, ws, h, _EPI_RESIDUAL_BF16)
+            dd = adapter.delta("dense", attn)
+            if dd is not None:
+                h2 = h2 + dd.to(h2.dtype)
+            inter = self.w_h4h.gemv(xn_mlp, ws, None, _EPI_PLAIN_F32)
+            dh = adapter.delta("h4h", xn_mlp)
+            if dh is not None:
+                inter += dh.float()
+            act = reference.gelu(inter).contiguous()
+            h3 = self.w_4hh.gemv(act, ws, h2, _EPI_RESIDUAL_BF16)
+            d4 = adapter.delta("4hh", act)
+            if d4 is not None:
+                h3 = h3 + d4.to(h3.dtype)
+            return h3.view(B, 1, H)
         if self.tp_world > 1:
             # parallel residual (attn + mlp both row-parallel partials):
             # sum the two partials locally, then ONE all-reduce per block
@@ -878,7 +944,7 @@ class FalconFastPath(_TPFastPathMixin):
         v = qkv[..., qd + kd :].view(B, S, self.kh, self.hd).transpose(1, 2)
         return q, k, v
 
-    def _body(self, hidden, kv_cache, prefix_length, autograd: bool):
+    def _body(self, hidden, kv_cache, prefix_length, autograd: bool, adapter=None):
         B, S, H = hidden.shape
         hidden = hidden.to(torch.bfloat16)
         end = prefix_length + S
@@ -893,6 +959,10 @@ class FalconFastPath(_TPFastPathMixin):
         xn_attn = self._tp_copy(ln(hidden, self.ln_attn_w, self.ln_attn_b))
         xn_mlp = xn_attn if self.single_ln else self._tp_copy(ln(hidden, self.ln_mlp_w, self.ln_mlp_b))
         qkv = torch.matmul(xn_attn, self.wqkv_t.dense())
+        if adapter is not None:
+            d = adapter.delta("qkv", xn_attn)
+            if d is not None:
+                qkv = qkv + d.to(qkv.dtype)[..., self._qkv_perm]
         q, k, v = self._split_heads(qkv, B, S)
         pos = torch.arange(prefix_length, end, device=hidden.device)
         if autograd:
@@ -916,14 +986,25 @@ class FalconFastPath(_TPFastPathMixin):
         attn = attn.transpose(1, 2).reshape(B, S, self.qh * self.hd).to(torch.bfloat16)
         o = torch.matmul(attn, self.wo_t.dense())
         inter = torch.matmul(xn_mlp, self.w_h4h.dense())
+        if adapter is not None:
+            d = adapter.delta("dense", attn)
+            if d is not None:
+                o = o + d.to(o.dtype)
+            dh = adapter.delta("h4h", xn_mlp)
+            if dh is not None:
+                inter = inter + dh.to(inter.dtype)
         act = reference.gelu(inter.float()).to(torch.bfloat16)
         mlp_out = torch.matmul(act, self.w_4hh.dense())
+        if adapter is not None:
+            d4 = adapter.delta("4hh", act)
+            if d4 is not None:
+                mlp_out = mlp_out + d4.to(mlp_out.dtype)
         return hidden + self._tp_reduce(o + mlp_out)
 
     def forward(self, hidden, kv_cache, prefix_length, adapter=None):
-        assert adapter is None
-        return self._body(hidden, kv_cache, prefix_length, autograd=False)
+        assert adapter is None or self.tp_world == 1, "LoRA on TP shards is served via the generic path"
+        return self._body(hidden, kv_cache, prefix_length, autograd=False, adapter=adapter)
 
     def forward_autograd(self, hidden, prefix_length: int = 0, adapter=None):
-        assert adapter is None
-        return self._body(hidden, None, prefix_length, autograd=True)
+        assert adapter is None or self.tp_world == 1, "LoRA on TP shards is served via the generic path"
+        return self._body(hidden, None, prefix_length, autograd=True, adapter=adapter)

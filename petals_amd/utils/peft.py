@@ -27,8 +27,9 @@ _active_adapter: contextvars.ContextVar[Optional[str]] = contextvars.ContextVar(
 )
 
 # projection keys supported per block (llama/mixtral naming)
-PROJ_KEYS = ("q", "k", "v", "o", "gate", "up", "down")
+PROJ_KEYS = ("q", "k", "v", "o", "gate", "up", "down", "qkv", "dense", "h4h", "4hh")
 _PROJ_PATHS = {
+    # llama family
     "q": "self_attn.q_proj",
     "k": "self_attn.k_proj",
     "v": "self_attn.v_proj",
@@ -36,6 +37,11 @@ _PROJ_PATHS = {
     "gate": "mlp.gate_proj",
     "up": "mlp.up_proj",
     "down": "mlp.down_proj",
+    # falcon / bloom family (fused-QKV checkpoints)
+    "qkv": "self_attention.query_key_value",
+    "dense": "self_attention.dense",
+    "h4h": "mlp.dense_h_to_4h",
+    "4hh": "mlp.dense_4h_to_h",
 }
 
 

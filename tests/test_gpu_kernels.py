@@ -461,10 +461,11 @@ def test_kv_cache_write(hip):
     assert k_cache[:, :, pos + 1].abs().max() == 0 and k_cache[:, :, pos - 1].abs().max() == 0
 
 
-def _block_fused_vs_cpu(preset: str, hidden: int):
+def _block_fused_vs_cpu(preset: str, hidden: int, adapter: bool = False):
     from petals_amd.models import get_model_block
     from petals_amd.models.config_base import load_model_config
     from petals_amd.server.from_pretrained import init_random_block_
+    from petals_amd.utils.peft import add_adapter_to_block, using_adapter
 
     cfg = load_model_config(preset)
     blk_cpu = get_model_block(cfg, 0)
@@ -476,19 +477,47 @@ def _block_fused_vs_cpu(preset: str, hidden: int):
     blk_gpu = blk_gpu.to("cuda", torch.bfloat16).eval().optimize_for_inference()
     assert blk_gpu._fast is not None, f"{preset} must take the fused path"
 
-    torch.manual_seed(5)
-    B, S = 2, 9
-    x = torch.randn(B, S, hidden) * 0.5
-    ks, vs = blk_cpu.kv_cache_shape(B, 32)
-    kc, vc = torch.zeros(ks), torch.zeros(vs)
-    kg = torch.zeros(ks, device="cuda", dtype=torch.bfloat16)
-    vg = torch.zeros(vs, device="cuda", dtype=torch.bfloat16)
+    ctx_mgr = None
+    if adapter:
+        # synthetic LoRA on qkv + the MLP down projection: the fused decode
+        # path must match the CPU eager adapter application
+        from petals_amd.utils.peft import BlockAdapter
 
-    y_cpu = [blk_cpu(x[:, :6], kv_cache=(kc, vc), prefix_length=0)]
-    y_gpu = [blk_gpu(x[:, :6].cuda().bfloat16(), kv_cache=(kg, vg), prefix_length=0)]
-    for t in range(6, S):
-        y_cpu.append(blk_cpu(x[:, t : t + 1], kv_cache=(kc, vc), prefix_length=t))
-        y_gpu.append(blk_gpu(x[:, t : t + 1].cuda().bfloat16(), kv_cache=(kg, vg), prefix_length=t))
+        torch.manual_seed(17)
+        qkv_lin = (blk_cpu.self_attention.query_key_value if hasattr(blk_cpu, "self_attention")
+                   else blk_cpu.self_attn.q_proj)
+        down_lin = (blk_cpu.mlp.dense_4h_to_h if hasattr(blk_cpu.mlp, "dense_4h_to_h")
+                    else blk_cpu.mlp.down_proj)
+        qkv_key = "qkv" if hasattr(blk_cpu, "self_attention") else "q"
+        down_key = "4hh" if hasattr(blk_cpu.mlp, "dense_4h_to_h") else "down"
+        projections = {
+            qkv_key: (torch.randn(4, qkv_lin.in_features) * 0.05,
+                      torch.randn(qkv_lin.out_features, 4) * 0.05, 2.0),
+            down_key: (torch.randn(4, down_lin.in_features) * 0.05,
+                       torch.randn(down_lin.out_features, 4) * 0.05, 2.0),
+        }
+        for blk in (blk_cpu, blk_gpu):
+            ad = BlockAdapter(name="t", projections={k: (a.clone(), b.clone(), s)
+                                                     for k, (a, b, s) in projections.items()})
+            add_adapter_to_block(blk, ad)
+        ctx_mgr = using_adapter("t")
+
+    import contextlib
+
+    with (ctx_mgr or contextlib.nullcontext()):
+        torch.manual_seed(5)
+        B, S = 2, 9
+        x = torch.randn(B, S, hidden) * 0.5
+        ks, vs = blk_cpu.kv_cache_shape(B, 32)
+        kc, vc = torch.zeros(ks), torch.zeros(vs)
+        kg = torch.zeros(ks, device="cuda", dtype=torch.bfloat16)
+        vg = torch.zeros(vs, device="cuda", dtype=torch.bfloat16)
+
+        y_cpu = [blk_cpu(x[:, :6], kv_cache=(kc, vc), prefix_length=0)]
+        y_gpu = [blk_gpu(x[:, :6].cuda().bfloat16(), kv_cache=(kg, vg), prefix_length=0)]
+        for t in range(6, S):
+            y_cpu.append(blk_cpu(x[:, t : t + 1], kv_cache=(kc, vc), prefix_length=t))
+            y_gpu.append(blk_gpu(x[:, t : t + 1].cuda().bfloat16(), kv_cache=(kg, vg), prefix_length=t))
     ref = torch.cat(y_cpu, 1)
     out = torch.cat([y.float().cpu() for y in y_gpu], 1)
     assert torch.allclose(out, ref, atol=0.05, rtol=0.05), (out - ref).abs().max()
@@ -498,6 +527,14 @@ def _block_fused_vs_cpu(preset: str, hidden: int):
 def test_bloom_block_fast_decode_matches_cpu(hip):
     """BLOOM fused path (LayerNorm + ALiBi + GELU biases) vs fp32 CPU block."""
     _block_fused_vs_cpu("test-bloom-hd64", 256)
+
+
+@requires_gpu
+@pytest.mark.parametrize("preset,hidden", [("test-falcon-hd64", 256), ("test-bloom-hd64", 256)])
+def test_block_fast_decode_with_adapter_matches_cpu(hip, preset, hidden):
+    """LoRA on the falcon/bloom FUSED decode paths (qkv delta permuted into
+    the fused [q|k|v] layout, kv_cache_write / rope_cache_write) vs CPU."""
+    _block_fused_vs_cpu(preset, hidden, adapter=True)
 
 
 @requires_gpu

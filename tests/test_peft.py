@@ -14,17 +14,36 @@ from petals_amd.utils.peft import BlockAdapter, add_adapter_to_block, load_block
 
 RANK, ALPHA = 4, 8
 
+# per-family adapter targets: (HF checkpoint path prefix, projections trained)
+_FAMILY = {
+    "test-llama": ("model.layers", ("self_attn.q_proj", "mlp.down_proj")),
+    "test-falcon": ("transformer.h", ("self_attention.query_key_value", "mlp.dense_4h_to_h")),
+    "test-bloom": ("h", ("self_attention.query_key_value", "mlp.dense_4h_to_h")),
+}
 
-def make_adapter_dir(tmp_path, config, n_blocks, seed=11):
+
+def _module_by_path(block, path):
+    mod = block
+    for part in path.split("."):
+        mod = getattr(mod, part)
+    return mod
+
+
+def make_adapter_dir(tmp_path, config, n_blocks, seed=11, model="test-llama"):
     from safetensors.torch import save_file
 
+    from petals_amd.models import get_model_block
+
     torch.manual_seed(seed)
+    prefix, projs = _FAMILY[model]
+    probe = get_model_block(config, 0)
     tensors = {}
-    h, inter = config.hidden_size, config.intermediate_size
     for i in range(n_blocks):
-        for proj, (in_d, out_d) in (("self_attn.q_proj", (h, h)), ("mlp.down_proj", (inter, h))):
-            tensors[f"base_model.model.model.layers.{i}.{proj}.lora_A.weight"] = torch.randn(RANK, in_d) * 0.05
-            tensors[f"base_model.model.model.layers.{i}.{proj}.lora_B.weight"] = torch.randn(out_d, RANK) * 0.05
+        for proj in projs:
+            lin = _module_by_path(probe, proj)
+            in_d, out_d = lin.in_features, lin.out_features
+            tensors[f"base_model.model.{prefix}.{i}.{proj}.lora_A.weight"] = torch.randn(RANK, in_d) * 0.05
+            tensors[f"base_model.model.{prefix}.{i}.{proj}.lora_B.weight"] = torch.randn(out_d, RANK) * 0.05
     d = tmp_path / "test-adapter"
     d.mkdir()
     with open(d / "adapter_config.json", "w") as f:
@@ -33,27 +52,35 @@ def make_adapter_dir(tmp_path, config, n_blocks, seed=11):
     return str(d)
 
 
+_KEY_TO_PATH = {
+    "q": "self_attn.q_proj", "k": "self_attn.k_proj", "v": "self_attn.v_proj",
+    "o": "self_attn.o_proj", "gate": "mlp.gate_proj", "up": "mlp.up_proj",
+    "down": "mlp.down_proj",
+    "qkv": "self_attention.query_key_value", "dense": "self_attention.dense",
+    "h4h": "mlp.dense_h_to_4h", "4hh": "mlp.dense_4h_to_h",
+}
+
+
 def merged_block(block, adapter: BlockAdapter):
     """Clone the block with LoRA merged into the dense weights."""
     import copy
 
     m = copy.deepcopy(block)
-    name_map = {"q": m.self_attn.q_proj, "k": m.self_attn.k_proj, "v": m.self_attn.v_proj,
-                "o": m.self_attn.o_proj, "gate": m.mlp.gate_proj, "up": m.mlp.up_proj,
-                "down": m.mlp.down_proj}
     with torch.no_grad():
         for key, (a, b, scale) in adapter.projections.items():
-            name_map[key].weight += (b.float() @ a.float()) * scale
+            _module_by_path(m, _KEY_TO_PATH[key]).weight += (b.float() @ a.float()) * scale
     return m
 
 
-def test_block_adapter_matches_merged(tmp_path):
-    cfg = load_model_config("test-llama")
+@pytest.mark.parametrize("model", ["test-llama", "test-falcon", "test-bloom"])
+def test_block_adapter_matches_merged(tmp_path, model):
+    cfg = load_model_config(model)
     block = get_model_block(cfg, 0)
     init_random_block_(block, cfg, 0)
-    adapter_dir = make_adapter_dir(tmp_path, cfg, 1)
-    ad = load_block_adapter(adapter_dir, 0, "model.layers")
-    assert ad is not None and set(ad.projections) == {"q", "down"}
+    adapter_dir = make_adapter_dir(tmp_path, cfg, 1, model=model)
+    ad = load_block_adapter(adapter_dir, 0, _FAMILY[model][0])
+    expected = {"q", "down"} if model == "test-llama" else {"qkv", "4hh"}
+    assert ad is not None and set(ad.projections) == expected
     add_adapter_to_block(block, ad)
 
     torch.manual_seed(1)
@@ -66,12 +93,13 @@ def test_block_adapter_matches_merged(tmp_path):
     assert torch.allclose(with_lora, ref, atol=1e-5), (with_lora - ref).abs().max()
 
 
-def test_block_adapter_with_cache(tmp_path):
-    cfg = load_model_config("test-llama")
+@pytest.mark.parametrize("model", ["test-llama", "test-falcon", "test-bloom"])
+def test_block_adapter_with_cache(tmp_path, model):
+    cfg = load_model_config(model)
     block = get_model_block(cfg, 0)
     init_random_block_(block, cfg, 0)
-    adapter_dir = make_adapter_dir(tmp_path, cfg, 1)
-    ad = load_block_adapter(adapter_dir, 0, "model.layers")
+    adapter_dir = make_adapter_dir(tmp_path, cfg, 1, model=model)
+    ad = load_block_adapter(adapter_dir, 0, _FAMILY[model][0])
     add_adapter_to_block(block, ad)
 
     torch.manual_seed(2)
