@@ -37,7 +37,10 @@ using f32x4 = __attribute__((ext_vector_type(4))) float;
 #define VT_BYTE(dim, key_byte) \
   ((((unsigned)(dim)) * ((KVT + KPAD) * 2) + (unsigned)(key_byte)) ^ ((((unsigned)(dim) >> 3) & 7u) << 4))
 
-template <int HD, int KVT, bool ALIBI>
+// QR = q rows per wave (16 or 32). QR=32 runs NSET=2 16-row MFMA row sets per
+// wave off ONE K/V staging pass: staging traffic and barriers per flop halve,
+// at the cost of ~2x accumulator VGPRs and a larger per-wave P scratch.
+template <int HD, int KVT, bool ALIBI, int QR>
 __global__ __launch_bounds__(WAVES * 64) void attn_prefill_kernel(
     const unsigned short* __restrict__ q,   // [B, QH, S, HD]
     const unsigned short* __restrict__ k,   // [B, KVH, Lmax, HD]
@@ -53,67 +56,74 @@ __global__ __launch_bounds__(WAVES * 64) void attn_prefill_kernel(
     int kv_offset,  // absolute position of q row 0 within the kv sequence
     float scale,
     int causal) {
+  constexpr int NSET = QR / 16;  // 16-row MFMA row sets per wave
   const int bh = blockIdx.x;          // b * q_heads + qh
   const int b = bh / q_heads;
   const int qh = bh - b * q_heads;
   const int kvh = qh / (q_heads / kv_heads);
-  const int q0_wg = blockIdx.y * (WAVES * QTILE);  // first q row of this wg
+  const int q0_wg = blockIdx.y * (WAVES * QR);  // first q row of this wg
 
   const int tid = threadIdx.x;
   const int wave = tid / WAVE;
   const int lane = tid & (WAVE - 1);
   const int col = lane & 15;      // fragment column (and C col)
   const int hi = lane >> 4;       // fragment k-group (and C row group)
-  const int q0 = q0_wg + wave * QTILE;  // this wave's first q row
+  const int q0 = q0_wg + wave * QR;  // this wave's first q row
   // softmax runs in the log2 domain (exp2 instead of exp): q is pre-scaled by
   // scale*log2e at load, saving a VALU mul per score AND per exponential —
   // the kernel was issue-bound on VALU at 17:1 VALU:MFMA (PMC, round 1)
   const float slope = ALIBI ? alibi[qh] * 1.4426950408889634f : 0.f;  // wave-uniform
 
   // LDS: K tile row-major [KVT][HD+KPAD]; V tile transposed+swizzled
-  // [HD][KVT+KPAD]; per-wave P scratch [QTILE][KVT+KPAD]
+  // [HD][KVT+KPAD]; per-wave P scratch [QR][KVT+KPAD]
   __shared__ unsigned short k_lds[KVT][HD + KPAD];
   __shared__ unsigned char vt_raw[HD * (KVT + KPAD) * 2];
-  __shared__ unsigned short p_lds[WAVES][QTILE][KVT + KPAD];
+  __shared__ unsigned short p_lds[WAVES][QR][KVT + KPAD];
 
   constexpr int KCH = HD / 32;   // 32-wide k-dim chunks per head dim
   constexpr int NB = KVT / 16;   // 16-key S column blocks per tile
   constexpr int PKC = KVT / 32;  // 32-key PV k-chunks per tile
 
-  // ---- load this wave's q tile into A-fragments (zero-padded past s_q),
+  // ---- load this wave's q tiles into A-fragments (zero-padded past s_q),
   // pre-scaled by scale*log2e (see the log2-domain softmax note above)
   const float qscale = scale * 1.4426950408889634f;
-  bf16x8 q_frag[KCH];
+  bf16x8 q_frag[NSET][KCH];
   const size_t q_base = (((size_t)b * q_heads + qh) * s_q) * HD;
-  const int my_qrow = q0 + col;  // A: row = lane&15
 #pragma unroll
-  for (int kc = 0; kc < KCH; ++kc) {
-    if (my_qrow < s_q) {
-      const unsigned short* src = q + q_base + (size_t)my_qrow * HD + kc * 32 + hi * 8;
-      const bf16x8 raw = *reinterpret_cast<const bf16x8*>(src);
-      short vs[8];
+  for (int qs = 0; qs < NSET; ++qs) {
+    const int my_qrow = q0 + qs * 16 + col;  // A: row = lane&15
 #pragma unroll
-      for (int e = 0; e < 8; ++e)
-        vs[e] = (short)f32_to_bf16(bf16_to_f32((unsigned short)raw[e]) * qscale);
-      q_frag[kc] = bf16x8{vs[0], vs[1], vs[2], vs[3], vs[4], vs[5], vs[6], vs[7]};
-    } else {
-      q_frag[kc] = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+    for (int kc = 0; kc < KCH; ++kc) {
+      if (my_qrow < s_q) {
+        const unsigned short* src = q + q_base + (size_t)my_qrow * HD + kc * 32 + hi * 8;
+        const bf16x8 raw = *reinterpret_cast<const bf16x8*>(src);
+        short vs[8];
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          vs[e] = (short)f32_to_bf16(bf16_to_f32((unsigned short)raw[e]) * qscale);
+        q_frag[qs][kc] = bf16x8{vs[0], vs[1], vs[2], vs[3], vs[4], vs[5], vs[6], vs[7]};
+      } else {
+        q_frag[qs][kc] = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+      }
     }
   }
 
   // ---- accumulators: O in C-layout (HD/16 dim-blocks x f32x4), softmax stats
-  f32x4 acc_o[HD / 16];
+  f32x4 acc_o[NSET][HD / 16];
+  float m_row[NSET][4], l_row[NSET][4];  // rows qs*16 + hi*4 + r (C layout)
 #pragma unroll
-  for (int d = 0; d < HD / 16; ++d) acc_o[d] = f32x4{0.f, 0.f, 0.f, 0.f};
-  float m_row[4], l_row[4];  // for rows hi*4 + r (C layout rows of this lane)
+  for (int qs = 0; qs < NSET; ++qs) {
 #pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    m_row[r] = NEG_SENTINEL;
-    l_row[r] = 0.f;
+    for (int d = 0; d < HD / 16; ++d) acc_o[qs][d] = f32x4{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      m_row[qs][r] = NEG_SENTINEL;
+      l_row[qs][r] = 0.f;
+    }
   }
 
   // causal bound for this WORKGROUP (max key any of its q rows may see)
-  const int wg_last_q_abs = kv_offset + min(q0_wg + WAVES * QTILE, s_q) - 1;
+  const int wg_last_q_abs = kv_offset + min(q0_wg + WAVES * QR, s_q) - 1;
   const int kv_end = causal ? min(kv_len, wg_last_q_abs + 1) : kv_len;
 
   const size_t kv_base = (((size_t)b * kv_heads + kvh) * lmax) * HD;
@@ -138,116 +148,126 @@ __global__ __launch_bounds__(WAVES * 64) void attn_prefill_kernel(
     }
     __syncthreads();
 
-    // ---- S = Q K^T : NB 16-key column blocks
-    f32x4 s_acc[NB];
+    float corr[NSET][4];
 #pragma unroll
-    for (int nb = 0; nb < NB; ++nb) {
-      s_acc[nb] = f32x4{0.f, 0.f, 0.f, 0.f};
-#pragma unroll
-      for (int kc = 0; kc < KCH; ++kc) {
-        // B[k][n] = K[key = nb*16 + col][kdim = kc*32 + hi*8 + reg]
-        const bf16x8 kt = *reinterpret_cast<const bf16x8*>(&k_lds[nb * 16 + col][kc * 32 + hi * 8]);
-        s_acc[nb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(q_frag[kc], kt, s_acc[nb], 0, 0, 0);
-      }
-    }
-
-    // ---- causal mask + online softmax (per-lane rows hi*4+r, col = key)
-    float p[NB][4];  // [nb][r] probabilities for this lane's slots
-    float corr[4];
-    // interior tiles (no tail, no causal edge, no s_q edge for ANY row of
-    // this wave) skip all masking VALU: +10% at S=4096 (PMC showed the
-    // kernel issue-bound on VALU at 17:1 VALU:MFMA)
-    const bool interior =
-        (tile_n == KVT) && (q0 + QTILE <= s_q) && (!causal || j0 + KVT <= kv_offset + q0 + 1);
-#pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int qrow = q0 + hi * 4 + r;
-      const int q_abs = kv_offset + qrow;
-      // NOTE: masking and the running max are deliberately SEPARATE loops.
-      // Fusing them (mask + `mx = fmaxf(mx, s[nb])` in one loop body) makes
-      // amdclang (ROCm 7.2, gfx950, -O3, with or without -ffast-math)
-      // miscompile the conditional sentinel store: dead keys keep their raw
-      // scores for the r==0 slot of every accumulator row group, leaking
-      // masked keys into the softmax (bisected in scripts/prefill_bisect.hip).
-      float s[NB];
-      if (interior) {
-#pragma unroll
-        for (int nb = 0; nb < NB; ++nb) {
-          s[nb] = s_acc[nb][r];  // already scaled (q pre-scaled by scale*log2e)
-          if (ALIBI) s[nb] += slope * (j0 + nb * 16 + col);
-        }
-      } else {
-#pragma unroll
-        for (int nb = 0; nb < NB; ++nb) {
-          const int key = j0 + nb * 16 + col;
-          s[nb] = s_acc[nb][r];
-          if (ALIBI) s[nb] += slope * key;
-          if (key >= tile_n + j0 || (causal && key > q_abs) || qrow >= s_q) s[nb] = NEG_SENTINEL;
-        }
-      }
-      float mx = NEG_SENTINEL;
-#pragma unroll
-      for (int nb = 0; nb < NB; ++nb) mx = fmaxf(mx, s[nb]);
-      // row max across the 16 lanes holding this row (xor within low 4 bits)
-#pragma unroll
-      for (int off = 8; off > 0; off >>= 1) mx = fmaxf(mx, __shfl_xor(mx, off, WAVE));
-      const float m_new = fmaxf(m_row[r], mx);
-      corr[r] = (m_row[r] <= NEG_THRESHOLD) ? 0.f : exp2f(m_row[r] - m_new);
-      float lsum = 0.f;
+    for (int qs = 0; qs < NSET; ++qs) {
+      const int q0s = q0 + qs * 16;  // first q row of this 16-row set
+      // ---- S = Q K^T : NB 16-key column blocks
+      f32x4 s_acc[NB];
 #pragma unroll
       for (int nb = 0; nb < NB; ++nb) {
-        p[nb][r] = (s[nb] <= NEG_THRESHOLD) ? 0.f : exp2f(s[nb] - m_new);
-        lsum += p[nb][r];
+        s_acc[nb] = f32x4{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+        for (int kc = 0; kc < KCH; ++kc) {
+          // B[k][n] = K[key = nb*16 + col][kdim = kc*32 + hi*8 + reg]
+          const bf16x8 kt = *reinterpret_cast<const bf16x8*>(&k_lds[nb * 16 + col][kc * 32 + hi * 8]);
+          s_acc[nb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(q_frag[qs][kc], kt, s_acc[nb], 0, 0, 0);
+        }
       }
-#pragma unroll
-      for (int off = 8; off > 0; off >>= 1) lsum += __shfl_xor(lsum, off, WAVE);
-      l_row[r] = l_row[r] * corr[r] + lsum;
-      m_row[r] = m_new;
-    }
 
-    // ---- write P (bf16) into per-wave LDS in [q_row][key] layout
+      // ---- causal mask + online softmax (per-lane rows hi*4+r, col = key)
+      float p[NB][4];  // [nb][r] probabilities for this lane's slots
+      // interior tiles (no tail, no causal edge, no s_q edge for ANY row of
+      // this set) skip all masking VALU: +10% at S=4096 (PMC showed the
+      // kernel issue-bound on VALU at 17:1 VALU:MFMA)
+      const bool interior =
+          (tile_n == KVT) && (q0s + QTILE <= s_q) && (!causal || j0 + KVT <= kv_offset + q0s + 1);
 #pragma unroll
-    for (int r = 0; r < 4; ++r)
+      for (int r = 0; r < 4; ++r) {
+        const int qrow = q0s + hi * 4 + r;
+        const int q_abs = kv_offset + qrow;
+        // NOTE: masking and the running max are deliberately SEPARATE loops.
+        // Fusing them (mask + `mx = fmaxf(mx, s[nb])` in one loop body) makes
+        // amdclang (ROCm 7.2, gfx950, -O3, with or without -ffast-math)
+        // miscompile the conditional sentinel store: dead keys keep their raw
+        // scores for the r==0 slot of every accumulator row group, leaking
+        // masked keys into the softmax (bisected in scripts/prefill_bisect.hip).
+        float s[NB];
+        if (interior) {
 #pragma unroll
-      for (int nb = 0; nb < NB; ++nb)
-        p_lds[wave][hi * 4 + r][nb * 16 + col] = f32_to_bf16(p[nb][r]);
+          for (int nb = 0; nb < NB; ++nb) {
+            s[nb] = s_acc[nb][r];  // already scaled (q pre-scaled by scale*log2e)
+            if (ALIBI) s[nb] += slope * (j0 + nb * 16 + col);
+          }
+        } else {
+#pragma unroll
+          for (int nb = 0; nb < NB; ++nb) {
+            const int key = j0 + nb * 16 + col;
+            s[nb] = s_acc[nb][r];
+            if (ALIBI) s[nb] += slope * key;
+            if (key >= tile_n + j0 || (causal && key > q_abs) || qrow >= s_q) s[nb] = NEG_SENTINEL;
+          }
+        }
+        float mx = NEG_SENTINEL;
+#pragma unroll
+        for (int nb = 0; nb < NB; ++nb) mx = fmaxf(mx, s[nb]);
+        // row max across the 16 lanes holding this row (xor within low 4 bits)
+#pragma unroll
+        for (int off = 8; off > 0; off >>= 1) mx = fmaxf(mx, __shfl_xor(mx, off, WAVE));
+        const float m_new = fmaxf(m_row[qs][r], mx);
+        corr[qs][r] = (m_row[qs][r] <= NEG_THRESHOLD) ? 0.f : exp2f(m_row[qs][r] - m_new);
+        float lsum = 0.f;
+#pragma unroll
+        for (int nb = 0; nb < NB; ++nb) {
+          p[nb][r] = (s[nb] <= NEG_THRESHOLD) ? 0.f : exp2f(s[nb] - m_new);
+          lsum += p[nb][r];
+        }
+#pragma unroll
+        for (int off = 8; off > 0; off >>= 1) lsum += __shfl_xor(lsum, off, WAVE);
+        l_row[qs][r] = l_row[qs][r] * corr[qs][r] + lsum;
+        m_row[qs][r] = m_new;
+      }
+
+      // ---- write P (bf16) into per-wave LDS in [q_row][key] layout
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+#pragma unroll
+        for (int nb = 0; nb < NB; ++nb)
+          p_lds[wave][qs * 16 + hi * 4 + r][nb * 16 + col] = f32_to_bf16(p[nb][r]);
+    }
     // the P round-trip is per-wave, but a block barrier is the simple safe
-    // ordering (the compiler may not prove the write/read regions disjoint)
+    // ordering (the compiler may not prove the write/read regions disjoint);
+    // ONE barrier covers all NSET row sets
     __syncthreads();
 
-    // P A-fragments: A[row = lane&15][k = pk*32 + hi*8 + reg] over KVT keys
-    bf16x8 p_frag[PKC];
 #pragma unroll
-    for (int pk = 0; pk < PKC; ++pk)
-      p_frag[pk] = *reinterpret_cast<const bf16x8*>(&p_lds[wave][col][pk * 32 + hi * 8]);
+    for (int qs = 0; qs < NSET; ++qs) {
+      // P A-fragments: A[row = lane&15][k = pk*32 + hi*8 + reg] over KVT keys
+      bf16x8 p_frag[PKC];
+#pragma unroll
+      for (int pk = 0; pk < PKC; ++pk)
+        p_frag[pk] = *reinterpret_cast<const bf16x8*>(&p_lds[wave][qs * 16 + col][pk * 32 + hi * 8]);
 
-    // ---- rescale O, then PV
+      // ---- rescale O, then PV
 #pragma unroll
-    for (int d = 0; d < HD / 16; ++d)
+      for (int d = 0; d < HD / 16; ++d)
 #pragma unroll
-      for (int r = 0; r < 4; ++r) acc_o[d][r] *= corr[r];
+        for (int r = 0; r < 4; ++r) acc_o[qs][d][r] *= corr[qs][r];
 #pragma unroll
-    for (int d = 0; d < HD / 16; ++d) {
+      for (int d = 0; d < HD / 16; ++d) {
 #pragma unroll
-      for (int pk = 0; pk < PKC; ++pk) {
-        // B[k = key][n = dim] = VT[dim = d*16 + col][key = pk*32 + hi*8 + reg]
-        const bf16x8 vfrag =
-            *reinterpret_cast<const bf16x8*>(&vt_raw[VT_BYTE(d * 16 + col, pk * 64 + hi * 16)]);
-        acc_o[d] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(p_frag[pk], vfrag, acc_o[d], 0, 0, 0);
+        for (int pk = 0; pk < PKC; ++pk) {
+          // B[k = key][n = dim] = VT[dim = d*16 + col][key = pk*32 + hi*8 + reg]
+          const bf16x8 vfrag =
+              *reinterpret_cast<const bf16x8*>(&vt_raw[VT_BYTE(d * 16 + col, pk * 64 + hi * 16)]);
+          acc_o[qs][d] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(p_frag[pk], vfrag, acc_o[qs][d], 0, 0, 0);
+        }
       }
     }
   }
 
-  // ---- write O: C layout row = hi*4 + r, col; scale by 1/l
+  // ---- write O: C layout row = qs*16 + hi*4 + r, col; scale by 1/l
 #pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    const int qrow = q0 + hi * 4 + r;
-    if (qrow >= s_q) continue;
-    const float inv_l = l_row[r] > 0.f ? 1.0f / l_row[r] : 0.f;
-    unsigned short* dst = out + q_base + (size_t)qrow * HD;
+  for (int qs = 0; qs < NSET; ++qs)
 #pragma unroll
-    for (int d = 0; d < HD / 16; ++d) dst[d * 16 + col] = f32_to_bf16(acc_o[d][r] * inv_l);
-  }
+    for (int r = 0; r < 4; ++r) {
+      const int qrow = q0 + qs * 16 + hi * 4 + r;
+      if (qrow >= s_q) continue;
+      const float inv_l = l_row[qs][r] > 0.f ? 1.0f / l_row[qs][r] : 0.f;
+      unsigned short* dst = out + q_base + (size_t)qrow * HD;
+#pragma unroll
+      for (int d = 0; d < HD / 16; ++d) dst[d * 16 + col] = f32_to_bf16(acc_o[qs][d][r] * inv_l);
+    }
 }
 
 torch::Tensor attn_prefill_fused(
@@ -267,7 +287,6 @@ torch::Tensor attn_prefill_fused(
   TORCH_CHECK(QH % KVH == 0);
   TORCH_CHECK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
   auto out = torch::empty_like(q);
-  dim3 grid(B * QH, (S + WAVES * QTILE - 1) / (WAVES * QTILE));
   auto stream = at::cuda::getCurrentCUDAStream();
   const float sc = (float)scale;
 
@@ -275,6 +294,14 @@ torch::Tensor attn_prefill_fused(
     const char* s = std::getenv("PETALS_PREFILL_KVT");
     return s ? std::atoi(s) : 64;
   }();
+  static const int qr_env = [] {
+    const char* s = std::getenv("PETALS_PREFILL_QROWS");
+    return s ? std::atoi(s) : 16;
+  }();
+  // QR=32 halves K/V staging + barriers per flop but needs >=8 row-tiles to
+  // fill the chip for short sequences; use it only when the grid stays large
+  const int qr = (qr_env == 32 && S >= 2 * WAVES * 32) ? 32 : 16;
+  dim3 grid(B * QH, (S + WAVES * qr - 1) / (WAVES * qr));
 
   const float* alibi_p = nullptr;
   if (alibi_slopes.has_value() && alibi_slopes->defined() && alibi_slopes->numel() > 0) {
@@ -287,23 +314,28 @@ torch::Tensor attn_prefill_fused(
   const auto* vp = reinterpret_cast<const unsigned short*>(v.data_ptr());
   auto* op = reinterpret_cast<unsigned short*>(out.data_ptr());
 
-#define PF_LAUNCH(HDV, KVTV, AB)                                                            \
-  attn_prefill_kernel<HDV, KVTV, AB><<<grid, WAVES * WAVE, 0, stream>>>(                    \
+#define PF_LAUNCH(HDV, KVTV, AB, QRV)                                                       \
+  attn_prefill_kernel<HDV, KVTV, AB, QRV><<<grid, WAVES * WAVE, 0, stream>>>(               \
       qp, kp, vp, op, alibi_p, QH, KVH, S, LMAX, (int)kv_len, (int)kv_offset, sc, causal ? 1 : 0)
+#define PF_QR(HDV, KVTV, AB)                                                                \
+  do {                                                                                      \
+    if (qr == 32) PF_LAUNCH(HDV, KVTV, AB, 32); else PF_LAUNCH(HDV, KVTV, AB, 16);          \
+  } while (0)
   const bool ab = alibi_p != nullptr;
   if (HD == 128) {
     if (kvt_env == 32) {
-      if (ab) PF_LAUNCH(128, 32, true); else PF_LAUNCH(128, 32, false);
+      if (ab) PF_QR(128, 32, true); else PF_QR(128, 32, false);
     } else {
-      if (ab) PF_LAUNCH(128, 64, true); else PF_LAUNCH(128, 64, false);
+      if (ab) PF_QR(128, 64, true); else PF_QR(128, 64, false);
     }
   } else {
     if (kvt_env == 32) {
-      if (ab) PF_LAUNCH(64, 32, true); else PF_LAUNCH(64, 32, false);
+      if (ab) PF_QR(64, 32, true); else PF_QR(64, 32, false);
     } else {
-      if (ab) PF_LAUNCH(64, 64, true); else PF_LAUNCH(64, 64, false);
+      if (ab) PF_QR(64, 64, true); else PF_QR(64, 64, false);
     }
   }
+#undef PF_QR
 #undef PF_LAUNCH
   HIP_CHECK_LAST();
   return out;
