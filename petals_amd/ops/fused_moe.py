@@ -27,7 +27,15 @@ from petals_amd.ops.fused_decode import (
     _EPI_SWIGLU_F32,
     LlamaFastPath,
     _FastWeight,
+    _dequant_cached,
 )
+
+
+class _CacheKey:
+    """Weakref-able identity key for the dequant LRU (plain object() cannot
+    carry weak references, which _dequant_cached uses for eviction)."""
+
+    __slots__ = ("__weakref__",)
 
 
 class _StackedExperts:
@@ -49,6 +57,9 @@ class _StackedExperts:
         else:
             assert quant == "none"
             self.wt_all = torch.stack([t.contiguous() for t in t_list]).contiguous()
+        # per-expert identity keys for the dequant LRU (fused_decode._dequant_cached
+        # keys by object id): multi-chunk prefills reuse each expert's dense form
+        self._cache_keys = [_CacheKey() for _ in t_list]
 
     def moe_gemv(self, x: torch.Tensor, sel: torch.Tensor, k_per_tok: int, ws: torch.Tensor,
                  epilogue: int) -> torch.Tensor:
@@ -57,9 +68,13 @@ class _StackedExperts:
         return self.hip.gemv_bf16_moe(self.wt_all, x, sel, k_per_tok, ws, epilogue)
 
     def dense(self, e: int) -> torch.Tensor:
-        """Expert e as a dense bf16 [in, out] (prefill/training matmuls)."""
+        """Expert e as a dense bf16 [in, out] (prefill/training matmuls),
+        LRU-cached so chunked prefills pay the dequant once per expert."""
         if self.quant == "nf4":
-            return self.hip.nf4_dequantize(self.packed_all[e], self.absmax_all[e])
+            return _dequant_cached(
+                self._cache_keys[e],
+                lambda: self.hip.nf4_dequantize(self.packed_all[e], self.absmax_all[e]),
+            )
         return self.wt_all[e]
 
 
