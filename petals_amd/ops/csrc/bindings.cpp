@@ -48,6 +48,10 @@ torch::Tensor gemv_bf16_moe(
 torch::Tensor gemv_nf4_moe(
     torch::Tensor packed_all, torch::Tensor absmax_all, torch::Tensor x, torch::Tensor sel,
     int64_t k_per_tok, torch::Tensor workspace, int64_t epilogue, int64_t splits_override);
+torch::Tensor moe_gemm(
+    c10::optional<torch::Tensor> wt_all, c10::optional<torch::Tensor> packed_all,
+    c10::optional<torch::Tensor> absmax_all, torch::Tensor x, torch::Tensor sorted_pairs,
+    torch::Tensor tile_expert, int64_t k_per_tok, int64_t n_rows);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rms_norm", &rms_norm, "RMSNorm (bf16 -> bf16)");
@@ -91,4 +95,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "device-routed MoE gemv: stacked NF4 expert weights, expert ids from a device tensor",
         py::arg("packed_all"), py::arg("absmax_all"), py::arg("x"), py::arg("sel"),
         py::arg("k_per_tok"), py::arg("workspace"), py::arg("epilogue"), py::arg("splits") = 0);
+  m.def("moe_gemm", &moe_gemm,
+        "grouped MFMA GEMM over expert-sorted (token, slot) pairs (prefill; NF4 dequant fused "
+        "into the LDS B-tile staging)",
+        py::arg("wt_all"), py::arg("packed_all"), py::arg("absmax_all"), py::arg("x"),
+        py::arg("sorted_pairs"), py::arg("tile_expert"), py::arg("k_per_tok"), py::arg("n_rows"));
 }

@@ -26,6 +26,9 @@
 #endif
 #define NF4_OUT_PER_WAVE 1024
 
+using bf16x8_moe = __attribute__((ext_vector_type(8))) short;
+using f32x4_moe = __attribute__((ext_vector_type(4))) float;
+
 static __device__ __constant__ float MOE_NF4_LUT[16] = {
     -1.0f, -0.6961928009986877f, -0.5250730514526367f, -0.39491748809814453f,
     -0.28444138169288635f, -0.18477343022823334f, -0.09105003625154495f, 0.0f,
@@ -210,7 +213,221 @@ __global__ void gemv_nf4_moe_kernel(
   }
 }
 
+// ------------------------------------------------- grouped GEMM (prefill)
+//
+// Prefill-time grouped GEMM over expert-sorted (token, slot) pairs: the host
+// (fused_moe.py) sorts the R = T*k pairs by expert with each expert's segment
+// padded to a multiple of MT rows, so every MT-row tile belongs to ONE expert
+// (tile_expert[tile], -1 = pure padding). Each workgroup computes an
+// MT x NT C tile with mfma_f32_16x16x32_bf16: the expert's B tile (KT x NT)
+// is staged per k-step into LDS TRANSPOSED with the attention kernel's XOR
+// swizzle (prefill_attn.hip VT_BYTE) so B-fragments are contiguous
+// ds_read_b128 rows; NF4 weights are dequantized on the way into LDS, so the
+// packed form (4.25 b/param) is what crosses HBM — no dense dequant pass.
+// Replaces the per-expert dense-matmul loop for prefill (each expert's
+// weights are read ceil(rows_e/MT) times instead of once per token).
+
+#define MOE_MT 32        // C tile rows (2 MFMA row blocks)
+#define MOE_NT 128       // C tile cols (4 waves x 2 MFMA col blocks)
+#define MOE_KT 64        // k (in-dim) step staged in LDS
+#define MOE_KPAD 8       // LDS k padding (elements)
+
+#define BT_BYTE(n, k_byte) \
+  ((((unsigned)(n)) * ((MOE_KT + MOE_KPAD) * 2) + (unsigned)(k_byte)) ^ ((((unsigned)(n) >> 3) & 7u) << 4))
+
+template <bool NF4>
+__global__ __launch_bounds__(256) void moe_gemm_kernel(
+    const unsigned short* __restrict__ wt_all,      // [E, in, out] bf16 (NF4: null)
+    const unsigned char* __restrict__ packed_all,   // [E, in, out/2] u8 (bf16: null)
+    const unsigned short* __restrict__ absmax_all,  // [E, in, out/64] bf16 (bf16: null)
+    const unsigned short* __restrict__ x,           // [T, in] bf16
+    const int* __restrict__ sorted_pairs,           // [Rp] pair id or -1 (padding)
+    const int* __restrict__ tile_expert,            // [Rp/MT] expert id or -1
+    unsigned short* __restrict__ c,                 // [R, out] bf16
+    int in_dim,
+    int out_dim,
+    int k_per_tok) {
+  const int tile = blockIdx.x;
+  const int e = tile_expert[tile];
+  if (e < 0) return;  // pure-padding tile
+  const int n0 = blockIdx.y * MOE_NT;
+
+  const int tid = threadIdx.x;
+  const int wave = tid / WAVE;
+  const int lane = tid & (WAVE - 1);
+  const int col = lane & 15;
+  const int hi = lane >> 4;
+
+  __shared__ float2 lut2[256];
+  if (NF4) {
+    for (int i = tid; i < 256; i += 256)
+      lut2[i] = make_float2(MOE_NF4_LUT[i & 0xF], MOE_NF4_LUT[i >> 4]);
+  }
+  __shared__ unsigned char bt_raw[MOE_NT * (MOE_KT + MOE_KPAD) * 2];
+
+  // this lane's A rows (2 row blocks): token index per MFMA A row
+  int tok[2];
+  int cpair[2][4];  // C rows: pair ids for rows rb*16 + hi*4 + r
+#pragma unroll
+  for (int rb = 0; rb < 2; ++rb) {
+    const int pr = sorted_pairs[tile * MOE_MT + rb * 16 + col];
+    tok[rb] = pr >= 0 ? pr / k_per_tok : -1;
+#pragma unroll
+    for (int r = 0; r < 4; ++r)
+      cpair[rb][r] = sorted_pairs[tile * MOE_MT + rb * 16 + hi * 4 + r];
+  }
+
+  f32x4_moe acc[2][2];
+#pragma unroll
+  for (int rb = 0; rb < 2; ++rb)
+#pragma unroll
+    for (int nb = 0; nb < 2; ++nb) acc[rb][nb] = f32x4_moe{0.f, 0.f, 0.f, 0.f};
+
+  const size_t w_stride = (size_t)in_dim * out_dim;
+
+  for (int kt = 0; kt < in_dim; kt += MOE_KT) {
+    // ---- stage B tile [MOE_KT k][MOE_NT n] transposed+swizzled into LDS
+    __syncthreads();
+    {
+      const int krow = kt + (tid >> 2);          // 64 rows, 4 threads per row
+      const int nloc = (tid & 3) * 32;           // 32 n values per thread
+      if (NF4) {
+        // 16 packed bytes -> 32 values; one absmax block per thread
+        unsigned int pk4[4] = {0u, 0u, 0u, 0u};
+        float am = 0.f;
+        if (krow < in_dim && n0 + nloc < out_dim) {
+          const unsigned char* prow =
+              packed_all + (size_t)e * (w_stride >> 1) + (size_t)krow * (out_dim >> 1) + ((n0 + nloc) >> 1);
+          const uint4 p = *reinterpret_cast<const uint4*>(prow);
+          pk4[0] = p.x; pk4[1] = p.y; pk4[2] = p.z; pk4[3] = p.w;
+          am = bf16_to_f32(absmax_all[(size_t)e * (w_stride >> 6) + (size_t)krow * (out_dim >> 6) + ((n0 + nloc) >> 6)]);
+        }
+        const int kb = (krow - kt) * 2;  // k byte offset in LDS rows
+#pragma unroll
+        for (int w4 = 0; w4 < 4; ++w4)
+#pragma unroll
+          for (int b = 0; b < 4; ++b) {
+            const float2 w2 = lut2[(pk4[w4] >> (8 * b)) & 0xFFu];
+            const int n = nloc + w4 * 8 + b * 2;
+            *reinterpret_cast<unsigned short*>(&bt_raw[BT_BYTE(n, kb)]) = f32_to_bf16(w2.x * am);
+            *reinterpret_cast<unsigned short*>(&bt_raw[BT_BYTE(n + 1, kb)]) = f32_to_bf16(w2.y * am);
+          }
+      } else {
+        const int kb = (krow - kt) * 2;
+#pragma unroll
+        for (int c8 = 0; c8 < 4; ++c8) {
+          short8 w8 = {0, 0, 0, 0, 0, 0, 0, 0};
+          if (krow < in_dim && n0 + nloc + c8 * 8 < out_dim)
+            w8 = *reinterpret_cast<const short8*>(
+                wt_all + (size_t)e * w_stride + (size_t)krow * out_dim + n0 + nloc + c8 * 8);
+#pragma unroll
+          for (int v = 0; v < 8; ++v)
+            *reinterpret_cast<unsigned short*>(&bt_raw[BT_BYTE(nloc + c8 * 8 + v, kb)]) =
+                (unsigned short)w8[v];
+        }
+      }
+    }
+    __syncthreads();
+
+    // ---- MFMA: 2 k-chunks x 2 row blocks x 2 col blocks
+#pragma unroll
+    for (int kc = 0; kc < MOE_KT / 32; ++kc) {
+      bf16x8_moe a_frag[2];
+#pragma unroll
+      for (int rb = 0; rb < 2; ++rb) {
+        if (tok[rb] >= 0 && kt + kc * 32 + hi * 8 + 8 <= in_dim) {
+          a_frag[rb] = *reinterpret_cast<const bf16x8_moe*>(
+              x + (size_t)tok[rb] * in_dim + kt + kc * 32 + hi * 8);
+        } else if (tok[rb] >= 0 && kt + kc * 32 + hi * 8 < in_dim) {
+          short vs[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+          for (int v = 0; v < 8 && kt + kc * 32 + hi * 8 + v < in_dim; ++v)
+            vs[v] = (short)x[(size_t)tok[rb] * in_dim + kt + kc * 32 + hi * 8 + v];
+          a_frag[rb] = bf16x8_moe{vs[0], vs[1], vs[2], vs[3], vs[4], vs[5], vs[6], vs[7]};
+        } else {
+          a_frag[rb] = bf16x8_moe{0, 0, 0, 0, 0, 0, 0, 0};
+        }
+      }
+#pragma unroll
+      for (int nb = 0; nb < 2; ++nb) {
+        const int n = wave * 32 + nb * 16 + col;
+        const bf16x8_moe b_frag =
+            *reinterpret_cast<const bf16x8_moe*>(&bt_raw[BT_BYTE(n, kc * 64 + hi * 16)]);
+#pragma unroll
+        for (int rb = 0; rb < 2; ++rb)
+          acc[rb][nb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag[rb], b_frag, acc[rb][nb], 0, 0, 0);
+      }
+    }
+  }
+
+  // ---- write C (bf16, original pair rows): C row = hi*4 + r within block
+#pragma unroll
+  for (int rb = 0; rb < 2; ++rb)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int pr = cpair[rb][r];
+      if (pr < 0) continue;
+#pragma unroll
+      for (int nb = 0; nb < 2; ++nb) {
+        const int n = n0 + wave * 32 + nb * 16 + col;
+        if (n < out_dim) c[(size_t)pr * out_dim + n] = f32_to_bf16(acc[rb][nb][r]);
+      }
+    }
+}
+
 // ------------------------------------------------------------------- host
+
+torch::Tensor moe_gemm(
+    c10::optional<torch::Tensor> wt_all,      // [E, in, out] bf16
+    c10::optional<torch::Tensor> packed_all,  // [E, in, out/2] u8
+    c10::optional<torch::Tensor> absmax_all,  // [E, in, out/64] bf16
+    torch::Tensor x,                          // [T, in] bf16
+    torch::Tensor sorted_pairs,               // [Rp] i32 (expert-sorted, MT-padded, -1 = pad)
+    torch::Tensor tile_expert,                // [Rp/MT] i32 (-1 = pure padding tile)
+    int64_t k_per_tok,
+    int64_t n_rows) {                         // R = T * k_per_tok (C rows)
+  TORCH_CHECK(x.is_cuda() && x.dtype() == torch::kBFloat16 && x.dim() == 2 && x.is_contiguous());
+  TORCH_CHECK(sorted_pairs.dtype() == torch::kInt32 && sorted_pairs.is_contiguous());
+  TORCH_CHECK(tile_expert.dtype() == torch::kInt32 && tile_expert.is_contiguous());
+  const bool nf4 = packed_all.has_value();
+  int in_dim, out_dim;
+  if (nf4) {
+    TORCH_CHECK(packed_all->dtype() == torch::kUInt8 && packed_all->dim() == 3 && packed_all->is_contiguous());
+    TORCH_CHECK(absmax_all.has_value() && absmax_all->dtype() == torch::kBFloat16);
+    in_dim = packed_all->size(1);
+    out_dim = packed_all->size(2) * 2;
+  } else {
+    TORCH_CHECK(wt_all.has_value() && wt_all->dtype() == torch::kBFloat16 && wt_all->dim() == 3);
+    in_dim = wt_all->size(1);
+    out_dim = wt_all->size(2);
+  }
+  TORCH_CHECK(x.size(1) == in_dim);
+  // staging loads are 16B/32-value vectors: callers fall back to the dense
+  // path for geometries outside these alignments (fused_moe.py)
+  TORCH_CHECK(out_dim % MOE_NT == 0, "moe_gemm needs out_dim % 128 == 0");
+  TORCH_CHECK(in_dim % MOE_KT == 0, "moe_gemm needs in_dim % 64 == 0");
+  const int Rp = sorted_pairs.size(0);
+  TORCH_CHECK(Rp % MOE_MT == 0 && tile_expert.size(0) == Rp / MOE_MT);
+
+  auto c = torch::empty({n_rows, (long)out_dim}, x.options());
+  dim3 grid(Rp / MOE_MT, (out_dim + MOE_NT - 1) / MOE_NT);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  if (nf4) {
+    moe_gemm_kernel<true><<<grid, 256, 0, stream>>>(
+        nullptr, packed_all->data_ptr<unsigned char>(),
+        reinterpret_cast<const unsigned short*>(absmax_all->data_ptr()),
+        reinterpret_cast<const unsigned short*>(x.data_ptr()),
+        sorted_pairs.data_ptr<int>(), tile_expert.data_ptr<int>(),
+        reinterpret_cast<unsigned short*>(c.data_ptr()), in_dim, out_dim, (int)k_per_tok);
+  } else {
+    moe_gemm_kernel<false><<<grid, 256, 0, stream>>>(
+        reinterpret_cast<const unsigned short*>(wt_all->data_ptr()), nullptr, nullptr,
+        reinterpret_cast<const unsigned short*>(x.data_ptr()),
+        sorted_pairs.data_ptr<int>(), tile_expert.data_ptr<int>(),
+        reinterpret_cast<unsigned short*>(c.data_ptr()), in_dim, out_dim, (int)k_per_tok);
+  }
+  HIP_CHECK_LAST();
+  return c;
+}
 
 torch::Tensor gemv_bf16_moe(
     torch::Tensor wt_all,    // [E, in, out] bf16

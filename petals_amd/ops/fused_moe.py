@@ -31,6 +31,33 @@ from petals_amd.ops.fused_decode import (
 )
 
 
+MOE_GEMM_MT = 32  # moe_gemm C-tile rows (ops/csrc/moe.hip MOE_MT)
+
+
+def sort_pairs_by_expert(selected: torch.Tensor, num_experts: int):
+    """Expert-sorted (token, slot) pair ids with each expert's segment padded
+    to MT rows, so every MT-row GEMM tile belongs to one expert (all device
+    ops — no host sync; the pad buffer is sized by the static upper bound).
+    Returns (sorted_pairs [Rp] i32 with -1 padding, tile_expert [Rp/MT] i32)."""
+    MT = MOE_GEMM_MT
+    dev = selected.device
+    sel = selected.reshape(-1)  # [R] expert per pair
+    R = sel.shape[0]
+    order = torch.argsort(sel, stable=True)
+    sel_sorted = sel[order]
+    counts = torch.bincount(sel, minlength=num_experts)
+    padded = (counts + MT - 1) // MT * MT
+    pad_off = torch.cumsum(padded, 0) - padded       # padded segment starts
+    unpad_off = torch.cumsum(counts, 0) - counts     # unpadded segment starts
+    slot = pad_off[sel_sorted] + (torch.arange(R, device=dev) - unpad_off[sel_sorted])
+    rp = (R + num_experts * (MT - 1) + MT - 1) // MT * MT  # static upper bound
+    sorted_pairs = torch.full((rp,), -1, dtype=torch.int32, device=dev)
+    sorted_pairs[slot] = order.to(torch.int32)
+    tile_expert = torch.full((rp // MT,), -1, dtype=torch.int32, device=dev)
+    tile_expert[slot // MT] = sel_sorted.to(torch.int32)
+    return sorted_pairs, tile_expert
+
+
 class _CacheKey:
     """Weakref-able identity key for the dequant LRU (plain object() cannot
     carry weak references, which _dequant_cached uses for eviction)."""
@@ -66,6 +93,19 @@ class _StackedExperts:
         if self.quant == "nf4":
             return self.hip.gemv_nf4_moe(self.packed_all, self.absmax_all, x, sel, k_per_tok, ws, epilogue)
         return self.hip.gemv_bf16_moe(self.wt_all, x, sel, k_per_tok, ws, epilogue)
+
+    def moe_gemm(self, x: torch.Tensor, sorted_pairs: torch.Tensor, tile_expert: torch.Tensor,
+                 k_per_tok: int, n_rows: int) -> torch.Tensor:
+        """Grouped MFMA GEMM (prefill): C[pair] = x[pair // k] @ W[expert(pair)]."""
+        if self.quant == "nf4":
+            return self.hip.moe_gemm(None, self.packed_all, self.absmax_all, x,
+                                     sorted_pairs, tile_expert, k_per_tok, n_rows)
+        return self.hip.moe_gemm(self.wt_all, None, None, x, sorted_pairs, tile_expert,
+                                 k_per_tok, n_rows)
+
+    @property
+    def gemm_ok(self) -> bool:
+        return self.out_dim % 128 == 0 and self.in_dim % 64 == 0
 
     def dense(self, e: int) -> torch.Tensor:
         """Expert e as a dense bf16 [in, out] (prefill/training matmuls),
@@ -153,11 +193,30 @@ class MixtralFastPath(LlamaFastPath):
         return self.stacked_down.dense(e) if self.stacked_down is not None else self.wdown_e[e].dense()
 
     def _mlp_dense(self, xn2, adapter, autograd: bool):
-        """Prefill / training MLP: token-grouped expert matmuls on the dense
-        (dequantized) transposed weights — the grouped-GEMM formulation."""
+        """Prefill / training MLP over the routed experts. Inference prefill
+        runs the grouped MFMA GEMM (ops/csrc/moe.hip moe_gemm: NF4 dequant
+        fused into the LDS B-tile staging — packed weights are what cross
+        HBM, read ceil(rows_e/32) times per expert instead of once per
+        token); training/CPU fall back to token-grouped dense matmuls."""
         shape = xn2.shape
         x = xn2.reshape(-1, shape[-1])
         weights, selected = self._route(x)
+        if (
+            not autograd
+            and x.device.type == "cuda"
+            and self.stacked_gu is not None
+            and self.stacked_gu.gemm_ok
+            and self.stacked_down.gemm_ok
+        ):
+            T, K = x.shape[0], self.top_k
+            sorted_pairs, tile_expert = sort_pairs_by_expert(selected, self.num_experts)
+            xb = x.to(torch.bfloat16).contiguous()
+            gateup = self.stacked_gu.moe_gemm(xb, sorted_pairs, tile_expert, K, T * K)
+            inter = gateup.shape[-1] // 2
+            act = self.hip.swiglu(gateup[:, :inter].contiguous(), gateup[:, inter:].contiguous())
+            down = self.stacked_down.moe_gemm(act.contiguous(), sorted_pairs, tile_expert, 1, T * K)
+            out = (down.view(T, K, -1).float() * weights.unsqueeze(-1)).sum(dim=1)
+            return out.reshape(*shape[:-1], -1).to(xn2.dtype)
         out_dim = self.stacked_down.out_dim if self.stacked_down is not None else self.wdown_e[0].shape[1]
         out = torch.zeros(x.shape[0], out_dim, dtype=torch.float32, device=x.device)
         expert_mask = F.one_hot(selected, num_classes=self.num_experts).permute(2, 1, 0)

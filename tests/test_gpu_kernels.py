@@ -238,6 +238,37 @@ def test_mixtral_block_fast_decode_matches_cpu(hip):
 
 @requires_gpu
 @pytest.mark.parametrize("quant", ["none", "nf4"])
+def test_moe_grouped_gemm_matches_dense(hip, quant):
+    """Prefill grouped MFMA GEMM (expert-sorted padded tiles, NF4 dequant
+    fused into the LDS B staging) vs per-expert dense matmuls."""
+    from petals_amd.ops.fused_moe import _StackedExperts, sort_pairs_by_expert
+
+    torch.manual_seed(11)
+    E, IN, OUT, T, K = 8, 512, 1152, 96, 2  # IN % 64 == 0, OUT % 128 == 0
+    wts = [(torch.randn(IN, OUT, device="cuda") * 0.05).to(torch.bfloat16) for _ in range(E)]
+    st = _StackedExperts(wts, hip, quant)
+    assert st.gemm_ok
+    x = (torch.randn(T, IN, device="cuda") * 0.5).to(torch.bfloat16)
+    sel = torch.randint(0, E, (T, K), device="cuda")
+    # make one expert empty and one over-full to exercise segment padding
+    sel[:40, 0] = 3
+    sel[sel == 7] = 1
+
+    sorted_pairs, tile_expert = sort_pairs_by_expert(sel, E)
+    out = st.moe_gemm(x, sorted_pairs, tile_expert, K, T * K)
+    assert out.shape == (T * K, OUT)
+
+    flat = sel.reshape(-1)
+    for pair in range(0, T * K, 7):
+        e = int(flat[pair])
+        ref = (x[pair // K].float() @ st.dense(e).float()).to(torch.bfloat16)
+        got = out[pair]
+        assert torch.allclose(got.float(), ref.float(), atol=0.02, rtol=0.02), (
+            pair, e, (got.float() - ref.float()).abs().max())
+
+
+@requires_gpu
+@pytest.mark.parametrize("quant", ["none", "nf4"])
 def test_mixtral_decode_graph_capture(hip, quant):
     """Device-routed MoE decode is hipGraph-safe: capture one decode step,
     replay over new tokens, match the eager fused path exactly."""
