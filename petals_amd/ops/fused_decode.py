@@ -219,6 +219,29 @@ def _dequant_cached(w, make) -> torch.Tensor:
     return t
 
 
+def decode_step_auto(fast, hidden, k_cache, v_cache, prefix_length: int = -1, ctx=None, adapter=None,
+                     max_b: int = None):
+    """Fused decode for any batch <= 8: batches past the kernel's register
+    budget (NF4 gemv accumulates [BATCH][OPL] in VGPRs, capped at 4) split
+    into sub-batches — the weights are re-read once per sub-batch, which
+    still beats falling back to the dense (dequantizing) prefill path by
+    ~5x at batch 8."""
+    if max_b is None:
+        max_b = 4 if fast.quant == "nf4" else 8
+    B = hidden.shape[0]
+    if B <= max_b:
+        return fast.decode_step(hidden, k_cache, v_cache, prefix_length, ctx=ctx, adapter=adapter)
+    outs = []
+    for i in range(0, B, max_b):
+        outs.append(
+            fast.decode_step(
+                hidden[i : i + max_b], k_cache[i : i + max_b], v_cache[i : i + max_b],
+                prefix_length, ctx=ctx, adapter=adapter,
+            )
+        )
+    return torch.cat(outs, dim=0)
+
+
 class _TPFastPathMixin:
     """Tensor-parallel hooks shared by the per-family fast paths. The shard's
     row-parallel outputs (o/down/dense) skip the fused residual epilogue and

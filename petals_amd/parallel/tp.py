@@ -237,10 +237,11 @@ class TPLlamaBlock(nn.Module):
             if torch.is_grad_enabled() and hidden_states.requires_grad:
                 assert kv_cache is None, "training forward does not use the KV cache"
                 return self._fast.forward_autograd(hidden_states, prefix_length)
-            max_b = 4 if self._fast.quant == "nf4" else 8
-            if kv_cache is not None and hidden_states.shape[1] == 1 and hidden_states.shape[0] <= max_b:
-                return self._fast.decode_step(
-                    hidden_states, kv_cache[0], kv_cache[1], prefix_length, ctx=ctx
+            if kv_cache is not None and hidden_states.shape[1] == 1 and hidden_states.shape[0] <= 8:
+                from petals_amd.ops.fused_decode import decode_step_auto
+
+                return decode_step_auto(
+                    self._fast, hidden_states, kv_cache[0], kv_cache[1], prefix_length, ctx=ctx
                 )
             return self._fast.forward(hidden_states, kv_cache, prefix_length)
         residual = hidden_states

@@ -561,6 +561,39 @@ def test_bloom_block_fast_decode_matches_cpu(hip):
 
 
 @requires_gpu
+def test_nf4_block_decode_batch_beyond_kernel_cap(hip):
+    """Batch 6 NF4 decode: the gemv kernel caps at BATCH=4, so the block
+    splits into sub-batches (decode_step_auto) instead of falling back to
+    the dense prefill path; numerics must match per-row batch-1 decode."""
+    from petals_amd.models import get_model_block
+    from petals_amd.models.config_base import load_model_config
+    from petals_amd.server.from_pretrained import init_random_block_
+
+    cfg = load_model_config("test-llama-hd128")
+    blk = get_model_block(cfg, 0)
+    init_random_block_(blk, cfg, 0)
+    blk = blk.to("cuda", torch.bfloat16).eval().optimize_for_inference(quant="nf4")
+
+    torch.manual_seed(21)
+    B, T = 6, 3
+    xs = [torch.randn(B, 1, cfg.hidden_size, device="cuda", dtype=torch.bfloat16) * 0.5 for _ in range(T)]
+    ks, vs = blk.kv_cache_shape(B, 16)
+    k6 = torch.zeros(ks, device="cuda", dtype=torch.bfloat16)
+    v6 = torch.zeros_like(k6)
+    out6 = [blk(x, kv_cache=(k6, v6), prefix_length=t) for t, x in enumerate(xs)]
+
+    ks1, vs1 = blk.kv_cache_shape(1, 16)
+    for b in range(B):
+        k1 = torch.zeros(ks1, device="cuda", dtype=torch.bfloat16)
+        v1 = torch.zeros_like(k1)
+        for t, x in enumerate(xs):
+            ref = blk(x[b : b + 1], kv_cache=(k1, v1), prefix_length=t)
+            got = out6[t][b : b + 1]
+            assert torch.allclose(got.float(), ref.float(), atol=0.02, rtol=0.02), (
+                b, t, (got.float() - ref.float()).abs().max())
+
+
+@requires_gpu
 @pytest.mark.parametrize("preset,hidden", [("test-falcon-hd64", 256), ("test-bloom-hd64", 256)])
 def test_block_fast_decode_with_adapter_matches_cpu(hip, preset, hidden):
     """LoRA on the falcon/bloom FUSED decode paths (qkv delta permuted into
