@@ -26,7 +26,7 @@
 // ---------------------------------------------------------------- gemv core
 
 template <int BATCH>
-__global__ void gemv_bf16_kernel(
+__global__ __launch_bounds__(WAVE) void gemv_bf16_kernel(
     const unsigned short* __restrict__ wt,  // [in, out]
     const float* __restrict__ x,            // [BATCH, in]
     float* __restrict__ partials,           // [n_splits, BATCH, out]
@@ -88,12 +88,17 @@ __global__ void gemv_bf16_kernel(
       }
     }
   } else {
-    const int nv = out_dim - out0;
     for (int i = i_begin; i < i_end; ++i) {
+#pragma unroll
       for (int b = 0; b < BATCH; ++b) {
         const float xv = x[(size_t)b * in_dim + i];
-        for (int v = 0; v < nv; ++v)
-          acc[b][v] = fmaf(bf16_to_f32(wt[(size_t)i * out_dim + out0 + v]), xv, acc[b][v]);
+        // compile-time trip count with a guard: a runtime-bounded loop here
+        // makes acc[][] dynamically indexed, forcing it onto the SCRATCH
+        // stack for the whole kernel (observed on the NF4 twin at BATCH>=5)
+#pragma unroll
+        for (int v = 0; v < 8; ++v)
+          if (out0 + v < out_dim)
+            acc[b][v] = fmaf(bf16_to_f32(wt[(size_t)i * out_dim + out0 + v]), xv, acc[b][v]);
       }
     }
   }
@@ -106,7 +111,9 @@ __global__ void gemv_bf16_kernel(
       d4[0] = float4v{acc[b][0], acc[b][1], acc[b][2], acc[b][3]};
       d4[1] = float4v{acc[b][4], acc[b][5], acc[b][6], acc[b][7]};
     } else {
-      for (int v = 0; v < out_dim - out0; ++v) dst[v] = acc[b][v];
+#pragma unroll
+      for (int v = 0; v < 8; ++v)
+        if (out0 + v < out_dim) dst[v] = acc[b][v];
     }
   }
 }

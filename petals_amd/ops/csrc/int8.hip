@@ -25,7 +25,7 @@
 using char8 = __attribute__((ext_vector_type(8))) char;
 
 template <int BATCH>
-__global__ void gemv_int8_kernel(
+__global__ __launch_bounds__(WAVE) void gemv_int8_kernel(
     const signed char* __restrict__ q,   // [in, out]
     const unsigned short* __restrict__ scale,  // [out] bf16 (absmax/127)
     const float* __restrict__ x,         // [BATCH, in]
@@ -83,20 +83,24 @@ __global__ void gemv_int8_kernel(
       }
     }
   } else {
-    const int nv = out_dim - out0;
     for (int i = i_begin; i < i_end; ++i) {
+#pragma unroll
       for (int b = 0; b < BATCH; ++b) {
         const float xv = x[(size_t)b * in_dim + i];
-        for (int v = 0; v < nv; ++v)
-          acc[b][v] = fmaf((float)q[(size_t)i * out_dim + out0 + v], xv, acc[b][v]);
+        // compile-time trip count + guard: runtime-bounded indexing of acc[][]
+        // would force it onto the scratch stack (see the NF4 twin)
+#pragma unroll
+        for (int v = 0; v < 8; ++v)
+          if (out0 + v < out_dim)
+            acc[b][v] = fmaf((float)q[(size_t)i * out_dim + out0 + v], xv, acc[b][v]);
       }
     }
   }
 
   // fold the per-column scale into the partial write (once per split)
   float sc[8];
-  const int nv = full ? 8 : out_dim - out0;
-  for (int v = 0; v < nv; ++v) sc[v] = bf16_to_f32(scale[out0 + v]);
+#pragma unroll
+  for (int v = 0; v < 8; ++v) sc[v] = (out0 + v < out_dim) ? bf16_to_f32(scale[out0 + v]) : 0.f;
 #pragma unroll
   for (int b = 0; b < BATCH; ++b) {
     float* dst = partials + ((size_t)split * BATCH + b) * out_dim + out0;
@@ -105,7 +109,9 @@ __global__ void gemv_int8_kernel(
       d4[0] = float4v{acc[b][0] * sc[0], acc[b][1] * sc[1], acc[b][2] * sc[2], acc[b][3] * sc[3]};
       d4[1] = float4v{acc[b][4] * sc[4], acc[b][5] * sc[5], acc[b][6] * sc[6], acc[b][7] * sc[7]};
     } else {
-      for (int v = 0; v < nv; ++v) dst[v] = acc[b][v] * sc[v];
+#pragma unroll
+      for (int v = 0; v < 8; ++v)
+        if (out0 + v < out_dim) dst[v] = acc[b][v] * sc[v];
     }
   }
 }

@@ -38,7 +38,7 @@ static __device__ __constant__ float MOE_NF4_LUT[16] = {
 
 // ------------------------------------------------------------ bf16 indexed
 
-__global__ void gemv_bf16_moe_kernel(
+__global__ __launch_bounds__(WAVE) void gemv_bf16_moe_kernel(
     const unsigned short* __restrict__ wt_all,  // [E, in, out]
     const float* __restrict__ x,                // [B, in]
     const int* __restrict__ sel,                // [R] expert per virtual row
@@ -92,11 +92,12 @@ __global__ void gemv_bf16_moe_kernel(
       for (int v = 0; v < 8; ++v) acc[v] = fmaf(bf16_to_f32((unsigned short)w8[v]), xv, acc[v]);
     }
   } else {
-    const int nv = out_dim - out0;
     for (int i = i_begin; i < i_end; ++i) {
       const float xv = xr[i];
-      for (int v = 0; v < nv; ++v)
-        acc[v] = fmaf(bf16_to_f32(wt[(size_t)i * out_dim + out0 + v]), xv, acc[v]);
+#pragma unroll
+      for (int v = 0; v < 8; ++v)
+        if (out0 + v < out_dim)
+          acc[v] = fmaf(bf16_to_f32(wt[(size_t)i * out_dim + out0 + v]), xv, acc[v]);
     }
   }
 
@@ -106,13 +107,15 @@ __global__ void gemv_bf16_moe_kernel(
     d4[0] = float4v{acc[0], acc[1], acc[2], acc[3]};
     d4[1] = float4v{acc[4], acc[5], acc[6], acc[7]};
   } else {
-    for (int v = 0; v < out_dim - out0; ++v) dst[v] = acc[v];
+#pragma unroll
+    for (int v = 0; v < 8; ++v)
+      if (out0 + v < out_dim) dst[v] = acc[v];
   }
 }
 
 // ------------------------------------------------------------- nf4 indexed
 
-__global__ void gemv_nf4_moe_kernel(
+__global__ __launch_bounds__(WAVE) void gemv_nf4_moe_kernel(
     const unsigned char* __restrict__ packed_all,   // [E, in, out/2]
     const unsigned short* __restrict__ absmax_all,  // [E, in, out/64]
     const float* __restrict__ x,                    // [B, in]
@@ -194,11 +197,13 @@ __global__ void gemv_nf4_moe_kernel(
   } else {
     for (int i = i_begin; i < i_end; ++i) {
       const float xa = xr[i] * bf16_to_f32(absmax[(size_t)i * (out_dim >> 6) + (out0 >> 6)]);
-      for (int v = 0; v < out_dim - out0; ++v) {
-        const unsigned char byte = packed[(size_t)i * (out_dim >> 1) + ((out0 + v) >> 1)];
-        const float2 w2 = lut2[byte];
-        acc[v] = fmaf(((out0 + v) & 1) ? w2.y : w2.x, xa, acc[v]);
-      }
+#pragma unroll
+      for (int v = 0; v < 16; ++v)
+        if (out0 + v < out_dim) {
+          const unsigned char byte = packed[(size_t)i * (out_dim >> 1) + ((out0 + v) >> 1)];
+          const float2 w2 = lut2[byte];
+          acc[v] = fmaf(((out0 + v) & 1) ? w2.y : w2.x, xa, acc[v]);
+        }
     }
   }
 
@@ -209,7 +214,9 @@ __global__ void gemv_nf4_moe_kernel(
       reinterpret_cast<float4v*>(dst)[q] =
           float4v{acc[4 * q], acc[4 * q + 1], acc[4 * q + 2], acc[4 * q + 3]};
   } else {
-    for (int v = 0; v < out_dim - out0; ++v) dst[v] = acc[v];
+#pragma unroll
+    for (int v = 0; v < 16; ++v)
+      if (out0 + v < out_dim) dst[v] = acc[v];
   }
 }
 

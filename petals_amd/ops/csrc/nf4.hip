@@ -92,7 +92,7 @@ __global__ void nf4_dequant_kernel(
 // gathers read 8x the packed HBM bytes from LDS, so halving the entry width
 // halves the dominant LDS-bandwidth term (2 extra VALU bit-ops per pair).
 template <int BATCH, int OPL, bool LUTBF>
-__global__ void gemv_nf4_kernel(
+__global__ __launch_bounds__(WAVE) void gemv_nf4_kernel(
     const unsigned char* __restrict__ packed,   // [in, out/2]
     const unsigned short* __restrict__ absmax,  // [in, out/64]
     const unsigned short* __restrict__ absmax_t,  // [out/64, in] or null — the
@@ -148,7 +148,9 @@ __global__ void gemv_nf4_kernel(
     for (int v = 0; v < OPL; ++v) acc[b][v] = 0.f;
 
   if (full) {
-    constexpr int UNROLL = 16;
+    // batch > 4: halve the unroll depth — xs[BATCH][UNROLL] + acc[BATCH][OPL]
+    // at full depth would blow the VGPR budget below 2 waves/SIMD
+    constexpr int UNROLL = (BATCH > 4) ? 8 : 16;
     const int half_out = out_dim >> 1;
     const unsigned char* pp = packed + (size_t)i_begin * half_out + (out0 >> 1);
     const unsigned short* amt = absmax_t ? absmax_t + (size_t)(out0 >> 6) * in_dim : nullptr;
@@ -168,11 +170,12 @@ __global__ void gemv_nf4_kernel(
       float am[UNROLL];
       if (amt) {
         const short8 a0 = *reinterpret_cast<const short8*>(amt + i);
-        const short8 a1 = *reinterpret_cast<const short8*>(amt + i + 8);
 #pragma unroll
-        for (int e = 0; e < 8; ++e) {
-          am[e] = bf16_to_f32((unsigned short)a0[e]);
-          am[8 + e] = bf16_to_f32((unsigned short)a1[e]);
+        for (int e = 0; e < 8; ++e) am[e] = bf16_to_f32((unsigned short)a0[e]);
+        if constexpr (UNROLL == 16) {
+          const short8 a1 = *reinterpret_cast<const short8*>(amt + i + 8);
+#pragma unroll
+          for (int e = 0; e < 8; ++e) am[8 + e] = bf16_to_f32((unsigned short)a1[e]);
         }
       } else {
 #pragma unroll
@@ -230,12 +233,19 @@ __global__ void gemv_nf4_kernel(
   } else {
     for (int i = i_begin; i < i_end; ++i) {
       const float am = bf16_to_f32(absmax[(size_t)i * (out_dim >> 6) + (out0 >> 6)]);
+#pragma unroll
       for (int b = 0; b < BATCH; ++b) {
         const float xa = x[(size_t)b * in_dim + i] * am;
-        for (int v = 0; v < out_dim - out0; ++v) {
-          const unsigned char byte = packed[(size_t)i * (out_dim >> 1) + ((out0 + v) >> 1)];
-          const float2 w2 = NF4_L2(byte);
-          acc[b][v] = fmaf(((out0 + v) & 1) ? w2.y : w2.x, xa, acc[b][v]);
+        // compile-time trip count with a guard: a runtime-bounded loop here
+        // makes acc[][] dynamically indexed, forcing it onto the SCRATCH
+        // stack for the whole kernel (observed at BATCH>=5: ~100x slowdown)
+#pragma unroll
+        for (int v = 0; v < OPL; ++v) {
+          if (out0 + v < out_dim) {
+            const unsigned char byte = packed[(size_t)i * (out_dim >> 1) + ((out0 + v) >> 1)];
+            const float2 w2 = NF4_L2(byte);
+            acc[b][v] = fmaf(((out0 + v) & 1) ? w2.y : w2.x, xa, acc[b][v]);
+          }
         }
       }
     }
@@ -250,7 +260,9 @@ __global__ void gemv_nf4_kernel(
         reinterpret_cast<float4v*>(dst)[q] =
             float4v{acc[b][4 * q], acc[b][4 * q + 1], acc[b][4 * q + 2], acc[b][4 * q + 3]};
     } else {
-      for (int v = 0; v < out_dim - out0; ++v) dst[v] = acc[b][v];
+#pragma unroll
+      for (int v = 0; v < OPL; ++v)
+        if (out0 + v < out_dim) dst[v] = acc[b][v];
     }
   }
 }
@@ -306,7 +318,7 @@ torch::Tensor gemv_nf4(
   const int in_dim = packed.size(0);
   const int out_dim = packed.size(1) * 2;
   const int batch = x.size(0);
-  TORCH_CHECK(x.size(1) == in_dim && batch <= 4, "NF4 decode gemv supports batch <= 4");
+  TORCH_CHECK(x.size(1) == in_dim && batch <= 8, "NF4 decode gemv supports batch <= 8");
 
   // outputs per lane: 8 doubles the grid vs 16 (latency hiding for the
   // LDS-gather dequant chain); PETALS_NF4_OPL overrides for A/B sweeps
@@ -363,6 +375,10 @@ torch::Tensor gemv_nf4(
     case 2: LAUNCH_NF4(2, OPL, LB); break;                                    \
     case 3: LAUNCH_NF4(3, OPL, LB); break;                                    \
     case 4: LAUNCH_NF4(4, OPL, LB); break;                                    \
+    case 5: LAUNCH_NF4(5, OPL, LB); break;                                    \
+    case 6: LAUNCH_NF4(6, OPL, LB); break;                                    \
+    case 7: LAUNCH_NF4(7, OPL, LB); break;                                    \
+    case 8: LAUNCH_NF4(8, OPL, LB); break;                                    \
   }
   if (opl == 8) {
     if (lut_bf16) { LAUNCH_NF4_B(8, true) } else { LAUNCH_NF4_B(8, false) }

@@ -22,6 +22,7 @@ hold two copies).
 from __future__ import annotations
 
 import math
+import os
 from typing import Dict, Optional, Tuple
 
 import torch
@@ -219,15 +220,21 @@ def _dequant_cached(w, make) -> torch.Tensor:
     return t
 
 
+def _nf4_decode_max_b() -> int:
+    # native BATCH 5-8 kernels exist (UNROLL drops 16->8 to fit the VGPR
+    # budget); sub-batching at 4 vs native-8 is an A/B — the measured winner
+    # is the default (profiles/bench_serve_batched.log)
+    return int(os.environ.get("PETALS_AMD_NF4_DECODE_MAXB", "8"))
+
+
 def decode_step_auto(fast, hidden, k_cache, v_cache, prefix_length: int = -1, ctx=None, adapter=None,
                      max_b: int = None):
-    """Fused decode for any batch <= 8: batches past the kernel's register
-    budget (NF4 gemv accumulates [BATCH][OPL] in VGPRs, capped at 4) split
+    """Fused decode for any batch <= 8: batches past the kernel cap split
     into sub-batches — the weights are re-read once per sub-batch, which
     still beats falling back to the dense (dequantizing) prefill path by
     ~5x at batch 8."""
     if max_b is None:
-        max_b = 4 if fast.quant == "nf4" else 8
+        max_b = _nf4_decode_max_b() if fast.quant == "nf4" else 8
     B = hidden.shape[0]
     if B <= max_b:
         return fast.decode_step(hidden, k_cache, v_cache, prefix_length, ctx=ctx, adapter=adapter)
