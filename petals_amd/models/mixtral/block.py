@@ -101,9 +101,11 @@ class MixtralBlock(nn.Module):
             if kv_cache is not None and hidden_states.shape[1] == 1 and hidden_states.shape[0] <= 8:
                 from petals_amd.ops.fused_decode import decode_step_auto
 
+                # moe gemv handles B*top_k <= 16 routed rows per launch; the
+                # dense-projection gemvs handle batch 8 natively
                 return decode_step_auto(
                     self._fast, hidden_states, kv_cache[0], kv_cache[1], prefix_length, ctx=ctx,
-                    max_b=2 if self._fast.quant == "nf4" else 8,
+                    max_b=min(8, 16 // max(self._fast.top_k, 1)),
                 )
             return self._fast.forward(hidden_states, kv_cache, prefix_length)
 
