@@ -366,7 +366,10 @@ class TransformerConnectionHandler:
         max_graph_batch = 1
         if all_fast:
             quants = {self.backends[uid].block._fast.quant for uid in uids}
-            max_graph_batch = 2 if "nf4" in quants else 8
+            # NF4 batch 3-8 runs the native BATCH kernels since the scratch
+            # fix; PETALS_AMD_GRAPH_MAX_BATCH caps span-graph capture for A/B
+            default = "8"
+            max_graph_batch = int(os.environ.get("PETALS_AMD_GRAPH_MAX_BATCH", default))
         if (
             session is not None
             and device.type == "cuda"
