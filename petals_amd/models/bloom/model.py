@@ -20,8 +20,11 @@ class DistributedBloomModel(DistributedModelBase):
     def _make_final_norm(self, config):
         return nn.LayerNorm(config.hidden_size, eps=config.layer_norm_eps)
 
-    def _embed(self, input_ids: torch.Tensor) -> torch.Tensor:
-        return self.word_embeddings_layernorm(self.embed_tokens(input_ids))
+    def _prepare_embeds(self, inputs_embeds: torch.Tensor) -> torch.Tensor:
+        # applied after prompt concat, to ids-derived AND caller-provided
+        # embeddings alike (parity: reference models/bloom/model.py:83; HF
+        # BloomModel layer-norms inputs_embeds in both paths)
+        return self.word_embeddings_layernorm(inputs_embeds)
 
 
 class DistributedBloomForCausalLM(DistributedForCausalLMBase):

@@ -91,6 +91,12 @@ class DistributedModelBase(nn.Module, PTuneMixin):
     def _embed(self, input_ids: torch.Tensor) -> torch.Tensor:
         return self.embed_tokens(input_ids)
 
+    def _prepare_embeds(self, inputs_embeds: torch.Tensor) -> torch.Tensor:
+        """Applied to the embedding stream regardless of whether it came from
+        input_ids or caller-provided inputs_embeds (BLOOM overrides with its
+        embedding LayerNorm)."""
+        return inputs_embeds
+
     @property
     def word_embeddings(self):  # bloom-style alias
         return self.embed_tokens
@@ -128,6 +134,10 @@ class DistributedModelBase(nn.Module, PTuneMixin):
         if use_prompts and at_start:
             prompts, intermediate_prompts = self.get_prompt(batch)
             inputs_embeds = torch.cat([prompts.to(inputs_embeds.dtype), inputs_embeds], dim=1)
+        # family hook applied to BOTH input_ids and caller-provided
+        # inputs_embeds, after prompt concat (BLOOM's embedding LayerNorm —
+        # reference models/bloom/model.py:83 applies it exactly here)
+        inputs_embeds = self._prepare_embeds(inputs_embeds)
 
         step_kwargs = {}
         if session is not None and hypo_ids is not None:

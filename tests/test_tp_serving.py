@@ -56,7 +56,7 @@ def _worker(rank, world, dist_port, ckpt_dir, fail_q):
         from petals_amd.server.server import Server
         from petals_amd.utils.auto_config import AutoDistributedModelForCausalLM
 
-        hf_model = transformers.LlamaForCausalLM.from_pretrained(path).eval()
+        hf_model = transformers.AutoModelForCausalLM.from_pretrained(path).eval()
         boot = DHT(host="127.0.0.1")
         server = Server(
             path,
@@ -99,11 +99,23 @@ def _worker(rank, world, dist_port, ckpt_dir, fail_q):
         raise
 
 
-def test_tp_serving_exact_match_world2(tmp_path):
+@pytest.mark.parametrize("family", ["llama", "falcon", "bloom"])
+def test_tp_serving_exact_match_world2(tmp_path, family):
     transformers = pytest.importorskip("transformers")
     torch.manual_seed(0)
-    cfg = transformers.LlamaConfig(**HF_CFG)
-    transformers.LlamaForCausalLM(cfg).eval().save_pretrained(
+    if family == "llama":
+        cfg = transformers.LlamaConfig(**HF_CFG)
+        hf_model = transformers.LlamaForCausalLM(cfg)
+    elif family == "falcon":
+        cfg = transformers.FalconConfig(
+            hidden_size=64, num_attention_heads=4, num_hidden_layers=4, vocab_size=128,
+            new_decoder_architecture=True, num_kv_heads=2, bias=False, parallel_attn=True,
+        )
+        hf_model = transformers.FalconForCausalLM(cfg)
+    else:
+        cfg = transformers.BloomConfig(hidden_size=64, n_head=4, n_layer=4, vocab_size=128)
+        hf_model = transformers.BloomForCausalLM(cfg)
+    hf_model.eval().save_pretrained(
         os.path.join(str(tmp_path), "ckpt"), safe_serialization=True
     )
     port = _free_port()
