@@ -330,3 +330,55 @@ def test_beam_search_matches_local_mirror(client_model, hf_checkpoint):
         sequences = torch.cat([sequences[beam_idx], token_idx[:, None]], dim=1)
     ref = sequences[beam_scores.argmax()][None]
     assert torch.equal(out, ref), (out, ref)
+
+
+@pytest.mark.parametrize("family", ["bloom", "mixtral"])
+def test_training_grads_via_inputs_embeds_nonllama(tmp_path, family):
+    """Training through caller-provided inputs_embeds (the reference's prompt
+    -tuning client mode) must match local HF for families whose model wrapper
+    transforms the embedding stream — regression for the BLOOM embedding
+    LayerNorm being skipped on the inputs_embeds path (applied only on the
+    input_ids path: generate() matched while training grads were ~3e-2 off)."""
+    transformers = pytest.importorskip("transformers")
+    import os as _os
+
+    torch.manual_seed(0)
+    if family == "bloom":
+        cfg = transformers.BloomConfig(hidden_size=64, n_head=4, n_layer=3, vocab_size=100)
+        hf = transformers.BloomForCausalLM(cfg).eval()
+    else:
+        cfg = transformers.MixtralConfig(
+            hidden_size=64, num_hidden_layers=3, num_attention_heads=4, num_key_value_heads=2,
+            intermediate_size=128, vocab_size=100, num_local_experts=4, num_experts_per_tok=2,
+        )
+        hf = transformers.MixtralForCausalLM(cfg).eval()
+    path = _os.path.join(str(tmp_path), "ckpt")
+    hf.save_pretrained(path, safe_serialization=True)
+
+    from petals_amd.dht.node import DHT
+    from petals_amd.server.server import Server
+    from petals_amd.utils.auto_config import AutoDistributedModelForCausalLM
+
+    boot = DHT(host="127.0.0.1")
+    server = Server(
+        path, initial_peers=[boot.listen_addr], host="127.0.0.1", device="cpu",
+        torch_dtype="float32", block_indices="0:3", dht_prefix=f"{family}-egrad",
+        throughput=1.0,
+    ).start()
+    try:
+        model = AutoDistributedModelForCausalLM.from_pretrained(
+            path, initial_peers=[boot.listen_addr], dht_prefix=f"{family}-egrad",
+            show_route=False, max_retries=1,
+        )
+        ids = torch.randint(0, 100, (1, 5))
+        er = hf.get_input_embeddings()(ids).detach().requires_grad_(True)
+        hf(inputs_embeds=er).logits.square().mean().backward()
+        e = hf.get_input_embeddings()(ids).detach().requires_grad_(True)
+        model(inputs_embeds=e).logits.square().mean().backward()
+        assert torch.allclose(e.grad, er.grad, atol=1e-4, rtol=1e-3), (
+            (e.grad - er.grad).abs().max()
+        )
+        model.transformer.h.sequence_manager.shutdown()
+    finally:
+        server.shutdown()
+        boot.shutdown()
