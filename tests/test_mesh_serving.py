@@ -86,6 +86,20 @@ def _worker(rank, world, dist_port, ckpt_dir, fail_q):
             assert mesh._send_tickets.get(1, 0) > 0, "no mesh sends from rank 0 to rank 1"
             # ...and rank1 sent the final activations back to this process
             assert mesh._recv_next.get(1, 0) > 0, "no mesh deliveries from rank 1"
+        dist.barrier()
+        # --- failure path: a broken mesh must fall back to TCP push with the
+        # exact same outputs (mark_broken on both ranks, as a real RCCL error
+        # would do on both ends of the pair)
+        mesh.mark_broken("test-induced failure")
+        dist.barrier()
+        if rank == 0:
+            sent_before = dict(mesh._send_tickets)
+            torch.manual_seed(7)
+            ids2 = torch.randint(0, 128, (1, 4))
+            ref2 = hf_model.generate(ids2, max_new_tokens=6, do_sample=False)
+            out2 = model.generate(ids2, max_new_tokens=6, do_sample=False)
+            assert torch.equal(out2, ref2), (out2, ref2)
+            assert dict(mesh._send_tickets) == sent_before, "broken mesh must not be used"
             model.transformer.h.sequence_manager.shutdown()
         dist.barrier()
         server.shutdown()
