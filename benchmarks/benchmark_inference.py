@@ -25,12 +25,15 @@ def benchmark_inference(process_idx, args, results):
     model = AutoDistributedModelForCausalLM.from_pretrained(
         args.model, initial_peers=[parse_addr(p) for p in args.initial_peers],
         dht_prefix=args.dht_prefix, show_route=False,
+        torch_dtype=getattr(torch, args.torch_dtype),
     )
-    ids = torch.randint(0, model.config.vocab_size, (1, 8))
+    if args.device != "cpu":
+        model = model.to(args.device)  # embeddings/norm/head; blocks stay remote
+    ids = torch.randint(0, model.config.vocab_size, (1, 8), device=args.device)
     with model.transformer.h.inference_session(max_length=args.seq_len) as session:
         with model.transformer.h.use_session(session):
             model(input_ids=ids)  # prefill
-            step = torch.randint(0, model.config.vocab_size, (1, 1))
+            step = torch.randint(0, model.config.vocab_size, (1, 1), device=args.device)
             t0 = None
             for i in range(args.seq_len - ids.shape[1] - 1):
                 if i == args.warmup_steps:
@@ -48,6 +51,7 @@ def main():
     parser.add_argument("--initial_peers", nargs="+", required=True, help="host:port")
     parser.add_argument("--dht_prefix", default=None)
     parser.add_argument("--torch_dtype", default="float32")
+    parser.add_argument("--device", default="cpu")
     parser.add_argument("--n_processes", default="1")
     parser.add_argument("--seq_len", type=int, default=128)
     parser.add_argument("--warmup_steps", type=int, default=4)
