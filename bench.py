@@ -381,6 +381,7 @@ def main():
 
             def span_fn():
                 h = embed.index_select(0, gs["cur_id"].view(-1)).view(B, 1, H)
+                ctx.norm_parts = None  # folded-norm hand-off starts fresh each pass
                 for blk, (k, v) in zip(blocks, kv_caches):
                     h = blk(h, kv_cache=(k, v), ctx=ctx)
                 return h
@@ -411,6 +412,7 @@ def main():
 
             def span_fn():
                 h = gs["h_in"]
+                ctx.norm_parts = None  # folded-norm hand-off starts fresh each pass
                 for blk, (k, v) in zip(blocks, kv_caches):
                     h = blk(h, kv_cache=(k, v), ctx=ctx)
                 return h

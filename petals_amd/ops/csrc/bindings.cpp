@@ -41,7 +41,10 @@ torch::Tensor nf4_dequantize(torch::Tensor packed, torch::Tensor absmax);
 torch::Tensor gemv_nf4(
     torch::Tensor packed, torch::Tensor absmax, torch::Tensor x, torch::Tensor workspace,
     c10::optional<torch::Tensor> residual, int64_t epilogue, int64_t splits_override,
-    c10::optional<torch::Tensor> bias, c10::optional<torch::Tensor> absmax_t);
+    c10::optional<torch::Tensor> bias, c10::optional<torch::Tensor> absmax_t,
+    c10::optional<torch::Tensor> x_parts, double fold_eps,
+    c10::optional<torch::Tensor> sumsq_out);
+torch::Tensor sumsq_rows(torch::Tensor x);
 torch::Tensor gemv_bf16_moe(
     torch::Tensor wt_all, torch::Tensor x, torch::Tensor sel, int64_t k_per_tok,
     torch::Tensor workspace, int64_t epilogue, int64_t splits_override);
@@ -83,10 +86,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("residual"), py::arg("epilogue"), py::arg("splits") = 0, py::arg("bias") = py::none());
   m.def("nf4_quantize", &nf4_quantize, "blockwise NF4 quantize [in,out] bf16 -> (packed u8, absmax bf16)");
   m.def("nf4_dequantize", &nf4_dequantize, "NF4 -> bf16");
-  m.def("gemv_nf4", &gemv_nf4, "split-K NF4 gemv (dequant fused) with epilogue",
+  m.def("gemv_nf4", &gemv_nf4, "split-K NF4 gemv (dequant fused) with epilogue; "
+        "x_parts switches to the folded-RMSNorm mode (bf16 x + sum(h^2) partials)",
         py::arg("packed"), py::arg("absmax"), py::arg("x"), py::arg("workspace"),
         py::arg("residual"), py::arg("epilogue"), py::arg("splits") = 0, py::arg("bias") = py::none(),
-        py::arg("absmax_t") = py::none());
+        py::arg("absmax_t") = py::none(), py::arg("x_parts") = py::none(),
+        py::arg("fold_eps") = 0.0, py::arg("sumsq_out") = py::none());
+  m.def("sumsq_rows", &sumsq_rows, "per-row sum(x^2) of a bf16 matrix -> [rows, 1] f32");
   m.def("gemv_bf16_moe", &gemv_bf16_moe,
         "device-routed MoE gemv: stacked bf16 expert weights, expert ids from a device tensor",
         py::arg("wt_all"), py::arg("x"), py::arg("sel"), py::arg("k_per_tok"),

@@ -606,3 +606,48 @@ void rope_cache_write(
       b, (int)qh, (int)kh, lmax, hd);
   HIP_CHECK_LAST();
 }
+
+// ---------------- row sum-of-squares (folded-RMSNorm first-block feeder) ----
+// One workgroup per row: parts layout matches gemv_reduce's sumsq_out
+// ([rows, 1] here — the consuming gemv sums n_parts entries either way).
+
+__global__ void sumsq_rows_kernel(
+    const unsigned short* __restrict__ x,  // [rows, dim] bf16
+    float* __restrict__ parts,             // [rows, 1]
+    int dim) {
+  const int row = blockIdx.x;
+  const unsigned short* xr = x + (size_t)row * dim;
+  float s = 0.f;
+  for (int i = threadIdx.x * 8; i + 8 <= dim; i += blockDim.x * 8) {
+    const short8 v = *reinterpret_cast<const short8*>(xr + i);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const float f = bf16_to_f32((unsigned short)v[j]);
+      s = fmaf(f, f, s);
+    }
+  }
+  for (int i = (dim & ~7) + threadIdx.x; i < dim; i += blockDim.x) {
+    const float f = bf16_to_f32(xr[i]);
+    s = fmaf(f, f, s);
+  }
+  __shared__ float red[1024 / WAVE];
+  s = wave_reduce_sum(s);
+  if ((threadIdx.x & (WAVE - 1)) == 0) red[threadIdx.x / WAVE] = s;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float t = 0.f;
+    for (int k = 0; k < (int)(blockDim.x / WAVE); ++k) t += red[k];
+    parts[row] = t;
+  }
+}
+
+torch::Tensor sumsq_rows(torch::Tensor x) {
+  TORCH_CHECK(x.is_cuda() && x.dtype() == torch::kBFloat16 && x.dim() == 2 && x.is_contiguous());
+  const int rows = x.size(0), dim = x.size(1);
+  auto parts = torch::empty({rows, 1}, x.options().dtype(torch::kFloat32));
+  auto stream = at::cuda::getCurrentCUDAStream();
+  sumsq_rows_kernel<<<rows, 1024, 0, stream>>>(
+      reinterpret_cast<const unsigned short*>(x.data_ptr()), parts.data_ptr<float>(), dim);
+  HIP_CHECK_LAST();
+  return parts;
+}

@@ -54,6 +54,10 @@ static __global__ __launch_bounds__(64 * GEMV_REDUCE_GROUPS) void gemv_reduce_ke
     const unsigned short* __restrict__ residual,  // [batch, out] or null
     const unsigned short* __restrict__ bias,      // [out] bf16 or null
     void* __restrict__ y,
+    float* __restrict__ sumsq_out,  // [batch, out/64] per-wg sum(y^2) or null —
+    // feeds the folded-RMSNorm decode path (the next gemv derives inv_rms from
+    // these instead of a separate norm kernel); computed over the bf16-ROUNDED
+    // stored values so it matches what a norm kernel reading y would see
     int n_splits,
     int batch,
     int out_dim,
@@ -106,7 +110,13 @@ static __global__ __launch_bounds__(64 * GEMV_REDUCE_GROUPS) void gemv_reduce_ke
       reinterpret_cast<float*>(y)[row + o] = gelu_tanh_f32(sum);
     } else {  // EPI_RESIDUAL_BF16
       const float r = bf16_to_f32(residual[row + o]);
-      reinterpret_cast<unsigned short*>(y)[row + o] = f32_to_bf16(r + sum);
+      const unsigned short packed = f32_to_bf16(r + sum);
+      reinterpret_cast<unsigned short*>(y)[row + o] = packed;
+      if (sumsq_out) {
+        const float v = bf16_to_f32(packed);
+        float sq = wave_reduce_sum(v * v);  // 64 lanes of sg==0 = one wave
+        if (ol == 0) sumsq_out[(size_t)b * (out_dim >> 6) + (o >> 6)] = sq;
+      }
     }
   }
 }
@@ -121,7 +131,8 @@ static inline torch::Tensor launch_gemv_reduce(
     int out_dim,
     int epilogue,
     const torch::TensorOptions& f32_opts,
-    const torch::TensorOptions& bf16_opts) {
+    const torch::TensorOptions& bf16_opts,
+    c10::optional<torch::Tensor> sumsq_out = c10::nullopt) {
   const int half = out_dim / 2;
   const int n_out = (epilogue == EPI_SWIGLU_F32) ? half : out_dim;
   torch::Tensor y;
@@ -143,12 +154,20 @@ static inline torch::Tensor launch_gemv_reduce(
     TORCH_CHECK(bias->numel() == out_dim, "bias must be [out_dim] (pre-activation)");
     bias_p = reinterpret_cast<const unsigned short*>(bias->data_ptr());
   }
+  float* ss_p = nullptr;
+  if (sumsq_out.has_value() && sumsq_out->defined() && sumsq_out->numel() > 0) {
+    TORCH_CHECK(epilogue == EPI_RESIDUAL_BF16, "sumsq_out only with the residual epilogue");
+    TORCH_CHECK(out_dim % 64 == 0, "sumsq_out needs out_dim % 64 == 0");
+    TORCH_CHECK(sumsq_out->dtype() == torch::kFloat32 && sumsq_out->is_contiguous());
+    TORCH_CHECK(sumsq_out->numel() >= (int64_t)batch * (out_dim / 64));
+    ss_p = sumsq_out->data_ptr<float>();
+  }
   const int total = batch * n_out;
   const int rthreads = 64 * GEMV_REDUCE_GROUPS;
   int rblocks = (total + 63) / 64;
   auto stream = c10::hip::getCurrentHIPStream().stream();
   gemv_reduce_kernel_impl<<<rblocks, rthreads, 0, stream>>>(
-      partials.data_ptr<float>(), res_p, bias_p, y.data_ptr(), n_splits, batch, out_dim, epilogue);
+      partials.data_ptr<float>(), res_p, bias_p, y.data_ptr(), ss_p, n_splits, batch, out_dim, epilogue);
   HIP_CHECK_LAST();
   return y;
 }

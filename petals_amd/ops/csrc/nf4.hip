@@ -91,7 +91,11 @@ __global__ void nf4_dequant_kernel(
 // LUTBF: 4-byte bf16-pair LUT entries instead of 8-byte float2 — the LUT
 // gathers read 8x the packed HBM bytes from LDS, so halving the entry width
 // halves the dominant LDS-bandwidth term (2 extra VALU bit-ops per pair).
-template <int BATCH, int OPL, bool LUTBF>
+// XS: folded-RMSNorm input mode — x arrives as the raw bf16 hidden state and
+// the kernel derives inv_rms from producer-side per-wg sum(h^2) partials
+// (gemv_reduce sumsq_out), replacing the standalone norm kernel entirely (the
+// norm WEIGHT is pre-folded into the quantized weight rows at load).
+template <int BATCH, int OPL, bool LUTBF, bool XS = false>
 __global__ __launch_bounds__(WAVE) void gemv_nf4_kernel(
     const unsigned char* __restrict__ packed,   // [in, out/2]
     const unsigned short* __restrict__ absmax,  // [in, out/64]
@@ -99,11 +103,15 @@ __global__ __launch_bounds__(WAVE) void gemv_nf4_kernel(
     // transposed copy turns the per-row 2-byte strided absmax gather into ONE
     // contiguous 32 B load per 16-row unroll block (the strided stream was a
     // latency-hiding bottleneck: 16 extra scattered loads in flight per block)
-    const float* __restrict__ x,                // [BATCH, in]
+    const float* __restrict__ x,                // [BATCH, in] f32 (XS: unused)
     float* __restrict__ partials,               // [n_splits, BATCH, out]
     int in_dim,
     int out_dim,
-    int i_per_split) {
+    int i_per_split,
+    const unsigned short* __restrict__ xb16 = nullptr,  // XS: [BATCH, in] bf16
+    const float* __restrict__ xs_parts = nullptr,       // XS: [BATCH, n_parts] sum(h^2) partials
+    int n_parts = 0,
+    float eps = 0.f) {
   // PAIR LUT: one read dequantizes a whole packed byte (two elements),
   // halving LDS traffic vs per-nibble lookups; 256 entries.
   // (the earlier 16-float bank-replicated LUT was LDS-issue bound)
@@ -133,6 +141,18 @@ __global__ __launch_bounds__(WAVE) void gemv_nf4_kernel(
 
   constexpr int WORDS = OPL / 8;  // u32 packed words per lane per row
   const int lane = threadIdx.x & (WAVE - 1);
+  // inv_rms is a per-row CONSTANT, so it scales the ACCUMULATOR once at the
+  // end instead of every x load: the sumsq-parts loads issue here (before the
+  // weight stream starts) and their reduction happens after the main loop —
+  // zero added latency on the critical path
+  float pfp[XS ? 2 * BATCH : 1];
+  if constexpr (XS) {
+#pragma unroll
+    for (int b = 0; b < BATCH; ++b) {
+      pfp[2 * b] = lane < n_parts ? xs_parts[(size_t)b * n_parts + lane] : 0.f;
+      pfp[2 * b + 1] = lane + WAVE < n_parts ? xs_parts[(size_t)b * n_parts + lane + WAVE] : 0.f;
+    }
+  }
 #define NF4_L2(byte) lut_pair(byte)
   const int out0 = blockIdx.x * (WAVE * OPL) + lane * OPL;
   if (out0 >= out_dim) return;
@@ -191,7 +211,9 @@ __global__ __launch_bounds__(WAVE) void gemv_nf4_kernel(
 #pragma unroll
       for (int b = 0; b < BATCH; ++b)
 #pragma unroll
-        for (int u = 0; u < UNROLL; ++u) xs[b][u] = x[(size_t)b * in_dim + i + u];
+        for (int u = 0; u < UNROLL; ++u)
+          xs[b][u] = XS ? bf16_to_f32(xb16[(size_t)b * in_dim + i + u])
+                        : x[(size_t)b * in_dim + i + u];
 #pragma unroll
       for (int u = 0; u < UNROLL; ++u) {
         float wf[OPL];
@@ -216,7 +238,9 @@ __global__ __launch_bounds__(WAVE) void gemv_nf4_kernel(
       const float am = bf16_to_f32(absmax[(size_t)i * (out_dim >> 6) + (out0 >> 6)]);
 #pragma unroll
       for (int b = 0; b < BATCH; ++b) {
-        const float xa = x[(size_t)b * in_dim + i] * am;
+        const float xv = XS ? bf16_to_f32(xb16[(size_t)b * in_dim + i])
+                            : x[(size_t)b * in_dim + i];
+        const float xa = xv * am;
 #pragma unroll
         for (int d = 0; d < WORDS; ++d) {
           const unsigned int wd =
@@ -235,7 +259,9 @@ __global__ __launch_bounds__(WAVE) void gemv_nf4_kernel(
       const float am = bf16_to_f32(absmax[(size_t)i * (out_dim >> 6) + (out0 >> 6)]);
 #pragma unroll
       for (int b = 0; b < BATCH; ++b) {
-        const float xa = x[(size_t)b * in_dim + i] * am;
+        const float xv = XS ? bf16_to_f32(xb16[(size_t)b * in_dim + i])
+                            : x[(size_t)b * in_dim + i];
+        const float xa = xv * am;
         // compile-time trip count with a guard: a runtime-bounded loop here
         // makes acc[][] dynamically indexed, forcing it onto the SCRATCH
         // stack for the whole kernel (observed at BATCH>=5: ~100x slowdown)
@@ -251,6 +277,15 @@ __global__ __launch_bounds__(WAVE) void gemv_nf4_kernel(
     }
   }
 
+  if constexpr (XS) {
+#pragma unroll
+    for (int b = 0; b < BATCH; ++b) {
+      const float total = wave_reduce_sum(pfp[2 * b] + pfp[2 * b + 1]);
+      const float inv = rsqrtf(total / in_dim + eps);
+#pragma unroll
+      for (int v = 0; v < OPL; ++v) acc[b][v] *= inv;
+    }
+  }
 #pragma unroll
   for (int b = 0; b < BATCH; ++b) {
     float* dst = partials + ((size_t)split * BATCH + b) * out_dim + out0;
@@ -306,15 +341,25 @@ torch::Tensor nf4_dequantize(torch::Tensor packed, torch::Tensor absmax) {
 torch::Tensor gemv_nf4(
     torch::Tensor packed,    // [in, out/2] u8
     torch::Tensor absmax,    // [in, out/64] bf16
-    torch::Tensor x,         // [batch, in] f32
+    torch::Tensor x,         // [batch, in] f32 (folded-norm mode: bf16 raw hidden)
     torch::Tensor workspace,
     c10::optional<torch::Tensor> residual,
     int64_t epilogue,
     int64_t splits_override,
     c10::optional<torch::Tensor> bias,      // [out] bf16, added pre-activation
-    c10::optional<torch::Tensor> absmax_t) {  // [out/64, in] bf16 (transposed copy)
+    c10::optional<torch::Tensor> absmax_t,  // [out/64, in] bf16 (transposed copy)
+    c10::optional<torch::Tensor> x_parts,   // folded-norm: [batch, n_parts] sum(h^2) partials
+    double fold_eps,
+    c10::optional<torch::Tensor> sumsq_out) {  // [batch, out/64] f32: emit sum(y^2) per wg
   TORCH_CHECK(packed.is_cuda() && packed.dtype() == torch::kUInt8);
-  TORCH_CHECK(x.dtype() == torch::kFloat32 && x.dim() == 2);
+  const bool xs_mode = x_parts.has_value() && x_parts->defined() && x_parts->numel() > 0;
+  if (xs_mode) {
+    TORCH_CHECK(x.dtype() == torch::kBFloat16 && x.dim() == 2 && x.is_contiguous(),
+                "folded-norm gemv takes the raw bf16 hidden state");
+    TORCH_CHECK(x_parts->dtype() == torch::kFloat32 && x_parts->is_contiguous());
+  } else {
+    TORCH_CHECK(x.dtype() == torch::kFloat32 && x.dim() == 2);
+  }
   const int in_dim = packed.size(0);
   const int out_dim = packed.size(1) * 2;
   const int batch = x.size(0);
@@ -328,6 +373,9 @@ torch::Tensor gemv_nf4(
   }();
   const int opl = opl_env ? opl_env : 8;
   TORCH_CHECK(opl == 8 || opl == 16, "PETALS_NF4_OPL must be 8 or 16");
+  TORCH_CHECK(!xs_mode || opl == 8, "folded-norm gemv is built for OPL=8");
+  TORCH_CHECK(!xs_mode || x_parts->size(1) <= 2 * WAVE,
+              "folded-norm gemv prefetches at most 128 sumsq partials");
   static const bool lut_bf16 = [] {
     const char* s = std::getenv("PETALS_NF4_LUT");
     return s && s[0] == 'b';  // default f32 pairs (bf16-pair measured SLOWER:
@@ -348,7 +396,7 @@ torch::Tensor gemv_nf4(
   splits = (in_dim + i_per_split_aligned - 1) / i_per_split_aligned;
 
   torch::Tensor partials;
-  auto f32opts = x.options();
+  auto f32opts = x.options().dtype(torch::kFloat32);
   if (workspace.numel() >= (int64_t)splits * batch * out_dim) {
     partials = workspace;
   } else {
@@ -369,6 +417,13 @@ torch::Tensor gemv_nf4(
       packed.data_ptr<unsigned char>(),                                       \
       reinterpret_cast<const unsigned short*>(absmax.data_ptr()), amt_p,      \
       x.data_ptr<float>(), partials.data_ptr<float>(), in_dim, out_dim, i_per_split)
+#define LAUNCH_NF4_XS(B)                                                      \
+  gemv_nf4_kernel<B, 8, false, true><<<grid, WAVE, 0, stream>>>(              \
+      packed.data_ptr<unsigned char>(),                                       \
+      reinterpret_cast<const unsigned short*>(absmax.data_ptr()), amt_p,      \
+      nullptr, partials.data_ptr<float>(), in_dim, out_dim, i_per_split,      \
+      reinterpret_cast<const unsigned short*>(x.data_ptr()),                  \
+      x_parts->data_ptr<float>(), (int)x_parts->size(1), (float)fold_eps)
 #define LAUNCH_NF4_B(OPL, LB)                                                 \
   switch (batch) {                                                            \
     case 1: LAUNCH_NF4(1, OPL, LB); break;                                    \
@@ -380,12 +435,24 @@ torch::Tensor gemv_nf4(
     case 7: LAUNCH_NF4(7, OPL, LB); break;                                    \
     case 8: LAUNCH_NF4(8, OPL, LB); break;                                    \
   }
-  if (opl == 8) {
+  if (xs_mode) {
+    switch (batch) {
+      case 1: LAUNCH_NF4_XS(1); break;
+      case 2: LAUNCH_NF4_XS(2); break;
+      case 3: LAUNCH_NF4_XS(3); break;
+      case 4: LAUNCH_NF4_XS(4); break;
+      case 5: LAUNCH_NF4_XS(5); break;
+      case 6: LAUNCH_NF4_XS(6); break;
+      case 7: LAUNCH_NF4_XS(7); break;
+      case 8: LAUNCH_NF4_XS(8); break;
+    }
+  } else if (opl == 8) {
     if (lut_bf16) { LAUNCH_NF4_B(8, true) } else { LAUNCH_NF4_B(8, false) }
   } else {
     if (lut_bf16) { LAUNCH_NF4_B(16, true) } else { LAUNCH_NF4_B(16, false) }
   }
 #undef LAUNCH_NF4_B
+#undef LAUNCH_NF4_XS
 #undef LAUNCH_NF4
   HIP_CHECK_LAST();
 
@@ -396,6 +463,7 @@ torch::Tensor gemv_nf4(
         .view({(int64_t)splits, (int64_t)batch, (int64_t)out_dim});
   }
   torch::Tensor y = launch_gemv_reduce(
-      partials, residual, bias, (int)splits, batch, out_dim, (int)epilogue, f32opts, absmax.options());
+      partials, residual, bias, (int)splits, batch, out_dim, (int)epilogue, f32opts,
+      absmax.options(), sumsq_out);
   return y;
 }

@@ -64,6 +64,10 @@ class DecodeContext:
     from device memory, so a captured graph replays correctly as `advance()`
     (or `set_position`) moves the position — no host-side re-capture."""
 
+    norm_parts = None  # folded-RMSNorm: producer-side sum(h^2) partials handed
+    # from block n-1's down-projection reduce to block n's qkv gemv; span
+    # loops MUST reset this to None before their first block each pass
+
     def __init__(self, device: torch.device):
         self.pos = torch.zeros(1, dtype=torch.int32, device=device)  # prefix length
         self.kv_len = torch.zeros(1, dtype=torch.int32, device=device)  # prefix + 1
@@ -150,16 +154,19 @@ class _FastWeight:
     def shape(self):
         return (self.in_dim, self.out_dim)
 
-    def _gemv_raw(self, x, ws, residual, epilogue, splits, bias=None):
+    def _gemv_raw(self, x, ws, residual, epilogue, splits, bias=None, x_parts=None,
+                  fold_eps=0.0, sumsq_out=None):
         if self.quant == "nf4":
             return self.hip.gemv_nf4(self.packed, self.absmax, x, ws, residual, epilogue, splits, bias,
-                                     self.absmax_t)
+                                     self.absmax_t, x_parts, fold_eps, sumsq_out)
+        assert x_parts is None and sumsq_out is None, "folded-norm gemv is NF4-only"
         if self.quant == "int8":
             return self.hip.gemv_int8(self.q8, self.scale8, x, ws, residual, epilogue, splits, bias)
         return self.hip.gemv_bf16(self.t, x, ws, residual, epilogue, splits, bias)
 
-    def gemv(self, x, ws, residual, epilogue, bias=None):
-        return self._gemv_raw(x, ws, residual, epilogue, self.splits, bias)
+    def gemv(self, x, ws, residual, epilogue, bias=None, x_parts=None, fold_eps=0.0, sumsq_out=None):
+        return self._gemv_raw(x, ws, residual, epilogue, self.splits, bias,
+                              x_parts=x_parts, fold_eps=fold_eps, sumsq_out=sumsq_out)
 
     def dense(self) -> torch.Tensor:
         """bf16 [in, out] view for prefill GEMMs. Quantized weights dequantize
@@ -309,13 +316,50 @@ class LlamaFastPath(_TPFastPathMixin):
         def t(w):
             return w.detach().to(torch.bfloat16).t().contiguous()
 
+        # folded RMSNorm (RMSNorm(x)·W == (x·inv_rms) @ diag(ln_w)·W): the norm
+        # WEIGHT is pre-multiplied into the quantized weight rows, and decode
+        # derives inv_rms from producer-side sum(h^2) partials — removing both
+        # norm kernels per block from the decode chain. MEASURED NEGATIVE end
+        # to end (74.2 vs 76.3 tok/s, profiles/fold_norm_ab.log): the sumsq
+        # emission + per-wave accumulator-scale tails across ~10k gemv waves
+        # cost more than the two removed 4.8 us norm kernels, even with the
+        # parts loads prefetched before the weight stream. Default OFF;
+        # PETALS_AMD_FOLD_NORM=1 keeps the path selectable (numerics covered
+        # by test_folded_norm_decode_matches_unfolded).
+        self.fold_norm = (
+            quant == "nf4"
+            and tp_world == 1
+            and getattr(type(self), "supports_fold_norm", True)
+            and cfg.hidden_size % 64 == 0
+            and cfg.hidden_size // 64 <= 128  # XS gemv prefetches <= 128 partials
+            and os.environ.get("PETALS_AMD_FOLD_NORM", "0") == "1"
+        )
+        self.ln1_w = block.input_layernorm.weight.detach().to(torch.bfloat16).contiguous()
+        self.ln2_w = block.post_attention_layernorm.weight.detach().to(torch.bfloat16).contiguous()
+        if self.fold_norm:
+            ln1f = block.input_layernorm.weight.detach().float().to(device)
+
+            def t1(w):
+                return (w.detach().float().t().to(device) * ln1f[:, None]).to(torch.bfloat16).contiguous()
+        else:
+            t1 = t
+
         self.wqkv_t = _FastWeight(
-            torch.cat([t(attn.q_proj.weight), t(attn.k_proj.weight), t(attn.v_proj.weight)], dim=1),
+            torch.cat([t1(attn.q_proj.weight), t1(attn.k_proj.weight), t1(attn.v_proj.weight)], dim=1),
             hip_ops, quant,
         )  # [H, qh*hd + 2*kh*hd]
         self.wo_t = _FastWeight(t(attn.o_proj.weight), hip_ops, quant)
-        self.ln1_w = block.input_layernorm.weight.detach().to(torch.bfloat16).contiguous()
-        self.ln2_w = block.post_attention_layernorm.weight.detach().to(torch.bfloat16).contiguous()
+        # norm weights as seen by the NON-fold code paths (prefill, autograd,
+        # adapters): identity when folded into the matmul weights
+        if self.fold_norm:
+            self.ln1_w_post = torch.ones_like(self.ln1_w)
+            self.ln2_w_post = torch.ones_like(self.ln2_w)
+            P = cfg.hidden_size // 64
+            self._ss1 = torch.empty(8, P, dtype=torch.float32, device=device)
+            self._ss2 = torch.empty(8, P, dtype=torch.float32, device=device)
+        else:
+            self.ln1_w_post = self.ln1_w
+            self.ln2_w_post = self.ln2_w
         self._empty_bf16 = torch.empty(0, device=device, dtype=torch.bfloat16)
         for lin in (attn.q_proj, attn.k_proj, attn.v_proj, attn.o_proj):
             lin.weight.data = self._empty_bf16
@@ -333,23 +377,34 @@ class LlamaFastPath(_TPFastPathMixin):
         def t(w):
             return w.detach().to(torch.bfloat16).t().contiguous()
 
+        if self.fold_norm:
+            ln2f = block.post_attention_layernorm.weight.detach().float().to(self.ln2_w.device)
+
+            def t2(w):
+                return (w.detach().float().t().to(ln2f.device) * ln2f[:, None]).to(torch.bfloat16).contiguous()
+        else:
+            t2 = t
+
         mlp = block.mlp
         self.wgateup_t = _FastWeight(
-            torch.cat([t(mlp.gate_proj.weight), t(mlp.up_proj.weight)], dim=1), hip_ops, quant
+            torch.cat([t2(mlp.gate_proj.weight), t2(mlp.up_proj.weight)], dim=1), hip_ops, quant
         )
         self.wdown_t = _FastWeight(t(mlp.down_proj.weight), hip_ops, quant)
         for lin in (mlp.gate_proj, mlp.up_proj, mlp.down_proj):
             lin.weight.data = self._empty_bf16
 
-    def _mlp_dense(self, xn2, adapter, autograd: bool):
+    def _mlp_dense(self, xn2, adapter, autograd: bool, x_delta=None):
         """MLP on [B, S, H] inputs via rocBLAS matmuls on the transposed
-        (possibly NF4-dequantized) weights; differentiable when autograd."""
+        (possibly NF4-dequantized) weights; differentiable when autograd.
+        `x_delta`: adapter-delta input when it differs from the matmul input
+        (folded-norm weights take the scale-only normed x)."""
         xn2 = self._tp_copy(xn2)
+        xd = xn2 if x_delta is None else x_delta
         gateup = torch.matmul(xn2, self.wgateup_t.dense())
         inter = self.wgateup_t.shape[1] // 2
         gate, up = gateup[..., :inter], gateup[..., inter:]
         if adapter is not None:
-            dg, du = adapter.delta("gate", xn2), adapter.delta("up", xn2)
+            dg, du = adapter.delta("gate", xd), adapter.delta("up", xd)
             if dg is not None:
                 gate = gate + dg
             if du is not None:
@@ -407,8 +462,22 @@ class LlamaFastPath(_TPFastPathMixin):
             pos, kv_len = ctx.pos, ctx.kv_len
         ws = _get_ws(self.device, "gemv", 64 * B * self._max_gemv_out())
 
-        xn = self.hip.rms_norm_f32out(h, self.ln1_w, self.eps)  # [B, H] f32
-        if adapter is None:
+        fold = self.fold_norm and adapter is None
+        if fold:
+            # folded RMSNorm: no norm kernels at all — the qkv gemv consumes the
+            # raw bf16 hidden plus producer-side sum(h^2) partials (from the
+            # previous block's down-reduce via ctx, or one sumsq kernel at the
+            # span head); the mlp gemv consumes the o-reduce's partials
+            parts_in = getattr(ctx, "norm_parts", None) if ctx is not None else None
+            if parts_in is None:
+                parts_in = self.hip.sumsq_rows(h)
+            parts = self.wqkv_t.gemv(h, ws, None, _EPI_RAW, x_parts=parts_in, fold_eps=self.eps)
+            q = self.hip.qkv_rope_reduce(
+                parts, self.rope_cos, self.rope_sin, pos, k_cache[:B], v_cache[:B],
+                self.qh, self.kh, True,
+            )
+        elif adapter is None:
+            xn = self.hip.rms_norm_f32out(h, self.ln1_w_post, self.eps)  # [B, H] f32
             # fused qkv reduce + rope + cache write: one kernel fewer per block
             parts = self.wqkv_t.gemv(xn, ws, None, _EPI_RAW)
             q = self.hip.qkv_rope_reduce(
@@ -416,8 +485,12 @@ class LlamaFastPath(_TPFastPathMixin):
                 self.qh, self.kh, True,
             )  # [B, qh*hd] f32, rotated
         else:
+            xn = self.hip.rms_norm_f32out(h, self.ln1_w_post, self.eps)
             qkv = self.wqkv_t.gemv(xn, ws, None, _EPI_PLAIN_F32)  # [B, qkv] f32
-            self._apply_qkv_adapter(qkv, xn, adapter)
+            # adapters were trained against the true normed activations: undo
+            # the fold for the delta input only
+            xn_a = xn * self.ln1_w.float() if self.fold_norm else xn
+            self._apply_qkv_adapter(qkv, xn_a, adapter)
             self.hip.rope_cache_write(
                 qkv, self.rope_cos, self.rope_sin, pos, k_cache[:B], v_cache[:B], self.qh, self.kh
             )
@@ -435,9 +508,17 @@ class LlamaFastPath(_TPFastPathMixin):
             h2 = self.wo_t.gemv(attn, ws, h, _EPI_RESIDUAL_BF16)  # [B, H] bf16
             if d is not None:
                 h2 = h2 + d.to(h2.dtype)
+        elif fold:
+            h2 = self.wo_t.gemv(attn, ws, h, _EPI_RESIDUAL_BF16, sumsq_out=self._ss1[:B])
+            act = self.wgateup_t.gemv(h2, ws, None, _EPI_SWIGLU_F32,
+                                      x_parts=self._ss1[:B], fold_eps=self.eps)
+            h3 = self.wdown_t.gemv(act, ws, h2, _EPI_RESIDUAL_BF16, sumsq_out=self._ss2[:B])
+            if ctx is not None:
+                ctx.norm_parts = self._ss2[:B]
+            return h3.view(B, 1, H)
         else:
             h2 = self.wo_t.gemv(attn, ws, h, _EPI_RESIDUAL_BF16)  # [B, H] bf16
-        xn2 = self.hip.rms_norm_f32out(h2, self.ln2_w, self.eps)
+        xn2 = self.hip.rms_norm_f32out(h2, self.ln2_w_post, self.eps)
         h3 = self._mlp_decode(xn2, h2, ws, adapter)
         return h3.view(B, 1, H)
 
@@ -456,7 +537,8 @@ class LlamaFastPath(_TPFastPathMixin):
             return self.wdown_t.gemv(act, ws, h2, _EPI_RESIDUAL_BF16)
         gateup = self.wgateup_t.gemv(xn2, ws, None, _EPI_PLAIN_F32)  # [B, 2I] f32
         inter = self.wgateup_t.shape[1] // 2
-        dg, du = adapter.delta("gate", xn2), adapter.delta("up", xn2)
+        xd = xn2 * self.ln2_w.float() if self.fold_norm else xn2  # true normed x for deltas
+        dg, du = adapter.delta("gate", xd), adapter.delta("up", xd)
         gate, up = gateup[:, :inter], gateup[:, inter:]
         if dg is not None:
             gate = gate + dg.float()
@@ -496,11 +578,12 @@ class LlamaFastPath(_TPFastPathMixin):
         B, S, H = hidden.shape
         end = prefix_length + S
         self._ensure_rope(end)
-        xn = reference.rms_norm(hidden, self.ln1_w, self.eps)
+        xn = reference.rms_norm(hidden, self.ln1_w_post, self.eps)
         xn = self._tp_copy(xn)  # backward: all-reduce the shard-partial grads
         qkv = torch.matmul(xn, self.wqkv_t.dense())
         if adapter is not None:
-            qkv = qkv + self._qkv_adapter_delta(xn, adapter)
+            xd = reference.rms_norm(hidden, self.ln1_w, self.eps) if self.fold_norm else xn
+            qkv = qkv + self._qkv_adapter_delta(xd, adapter)
         q = qkv[..., : self.qh * self.hd].view(B, S, self.qh, self.hd).transpose(1, 2)
         k = qkv[..., self.qh * self.hd : (self.qh + self.kh) * self.hd].view(B, S, self.kh, self.hd).transpose(1, 2)
         v = qkv[..., (self.qh + self.kh) * self.hd :].view(B, S, self.kh, self.hd).transpose(1, 2)
@@ -514,8 +597,9 @@ class LlamaFastPath(_TPFastPathMixin):
             if d is not None:
                 o = o + d
         h2 = hidden + o
-        xn2 = reference.rms_norm(h2, self.ln2_w, self.eps)
-        return h2 + self._tp_reduce(self._mlp_dense(xn2, adapter, autograd=True)).to(h2.dtype)
+        xn2 = reference.rms_norm(h2, self.ln2_w_post, self.eps)
+        xd2 = reference.rms_norm(h2, self.ln2_w, self.eps) if (self.fold_norm and adapter is not None) else None
+        return h2 + self._tp_reduce(self._mlp_dense(xn2, adapter, autograd=True, x_delta=xd2)).to(h2.dtype)
 
     # ------------------------------------------------------------ prefill
 
@@ -531,10 +615,11 @@ class LlamaFastPath(_TPFastPathMixin):
         end = prefix_length + S
         self._ensure_rope(end)
 
-        xn = self.hip.rms_norm(hidden, self.ln1_w, self.eps)
+        xn = self.hip.rms_norm(hidden, self.ln1_w_post, self.eps)
         qkv = torch.matmul(xn, self.wqkv_t.dense())  # [B, S, qkv] bf16 (rocBLAS)
         if adapter is not None:
-            qkv = qkv + self._qkv_adapter_delta(xn, adapter).to(qkv.dtype)
+            xd = self.hip.rms_norm(hidden, self.ln1_w, self.eps) if self.fold_norm else xn
+            qkv = qkv + self._qkv_adapter_delta(xd, adapter).to(qkv.dtype)
         q = qkv[..., : self.qh * self.hd].view(B, S, self.qh, self.hd).transpose(1, 2)
         k = qkv[..., self.qh * self.hd : (self.qh + self.kh) * self.hd].view(B, S, self.kh, self.hd).transpose(1, 2)
         v = qkv[..., (self.qh + self.kh) * self.hd :].view(B, S, self.kh, self.hd).transpose(1, 2)
@@ -559,8 +644,9 @@ class LlamaFastPath(_TPFastPathMixin):
             if d is not None:
                 o = o + d
         h2 = hidden + o
-        xn2 = self.hip.rms_norm(h2, self.ln2_w, self.eps)
-        return h2 + self._tp_reduce(self._mlp_dense(xn2, adapter, autograd=False)).to(h2.dtype)
+        xn2 = self.hip.rms_norm(h2, self.ln2_w_post, self.eps)
+        xd2 = self.hip.rms_norm(h2, self.ln2_w, self.eps) if (self.fold_norm and adapter is not None) else None
+        return h2 + self._tp_reduce(self._mlp_dense(xn2, adapter, autograd=False, x_delta=xd2)).to(h2.dtype)
 
 
 # ---------------------------------------------------------------------------

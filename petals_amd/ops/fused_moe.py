@@ -119,6 +119,10 @@ class _StackedExperts:
 
 
 class MixtralFastPath(LlamaFastPath):
+    # the router consumes the true ln2-weighted normed activations (softmax is
+    # not scale-invariant), so the MLP norm cannot fold into expert weights
+    supports_fold_norm = False
+
     def _init_mlp_weights(self, block, hip_ops, quant):
         def t(w):
             return w.detach().to(torch.bfloat16).t().contiguous()
@@ -192,7 +196,7 @@ class MixtralFastPath(LlamaFastPath):
     def _expert_down_dense(self, e: int) -> torch.Tensor:
         return self.stacked_down.dense(e) if self.stacked_down is not None else self.wdown_e[e].dense()
 
-    def _mlp_dense(self, xn2, adapter, autograd: bool):
+    def _mlp_dense(self, xn2, adapter, autograd: bool, x_delta=None):
         """Prefill / training MLP over the routed experts. Inference prefill
         runs the grouped MFMA GEMM (ops/csrc/moe.hip moe_gemm: NF4 dequant
         fused into the LDS B-tile staging — packed weights are what cross

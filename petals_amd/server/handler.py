@@ -66,6 +66,7 @@ class _SpanDecodeGraph:
 
         def span_fn():
             h = self.h_in
+            self.ctx.norm_parts = None  # folded-norm hand-off starts fresh each pass
             with using_adapter(adapter_name):
                 for block, (k, v) in zip(blocks, cache_pairs):
                     h = block(h, kv_cache=(k, v), ctx=self.ctx)

@@ -561,6 +561,71 @@ def test_bloom_block_fast_decode_matches_cpu(hip):
 
 
 @requires_gpu
+def test_folded_norm_decode_matches_unfolded(hip, monkeypatch):
+    """PETALS_AMD_FOLD_NORM: RMSNorm weights folded into the NF4 weights +
+    inv_rms from producer-side sumsq partials must match the unfolded fused
+    path, including the ctx (norm_parts) hand-off between chained blocks."""
+    from petals_amd.models import get_model_block
+    from petals_amd.models.config_base import load_model_config
+    from petals_amd.ops.fused_decode import DecodeContext
+    from petals_amd.server.from_pretrained import init_random_block_
+
+    cfg = load_model_config("test-llama-hd128")
+
+    def build(fold):
+        monkeypatch.setenv("PETALS_AMD_FOLD_NORM", "1" if fold else "0")
+        blks = []
+        for i in range(2):
+            blk = get_model_block(cfg, i)
+            init_random_block_(blk, cfg, i)
+            blk = blk.to("cuda", torch.bfloat16).eval().optimize_for_inference(quant="nf4")
+            assert blk._fast.fold_norm == fold
+            blks.append(blk)
+        return blks
+
+    ref_blks = build(False)
+    fold_blks = build(True)
+    torch.manual_seed(13)
+    B, T = 2, 4
+    xs = [torch.randn(B, 1, cfg.hidden_size, device="cuda", dtype=torch.bfloat16) * 0.5 for _ in range(T)]
+
+    def run(blks, use_ctx):
+        ks, vs = blks[0].kv_cache_shape(B, 16)
+        caches = [(torch.zeros(ks, device="cuda", dtype=torch.bfloat16),
+                   torch.zeros(vs, device="cuda", dtype=torch.bfloat16)) for _ in blks]
+        ctx = DecodeContext(torch.device("cuda")) if use_ctx else None
+        outs = []
+        for t, x in enumerate(xs):
+            if ctx is not None:
+                ctx.set_position(t)
+                ctx.norm_parts = None  # span loop resets each pass
+            h = x
+            for blk, (k, v) in zip(blks, caches):
+                h = blk(h, kv_cache=(k, v), prefix_length=t, ctx=ctx)
+            outs.append(h)
+        return outs
+
+    ref = run(ref_blks, use_ctx=False)
+    got_plain = run(fold_blks, use_ctx=False)   # per-block sumsq_rows
+    got_ctx = run(fold_blks, use_ctx=True)      # producer-side parts hand-off
+    for t in range(T):
+        # folded weights quantize on a slightly different NF4 grid -> small tol
+        assert torch.allclose(got_plain[t].float(), ref[t].float(), atol=0.06, rtol=0.06), (
+            t, (got_plain[t].float() - ref[t].float()).abs().max())
+        # parts-vs-sumsq_rows summation order differs by ULPs; bf16 rounding
+        # can flip on a few elements
+        assert torch.allclose(got_ctx[t].float(), got_plain[t].float(), atol=0.02, rtol=0.02), (
+            t, (got_ctx[t].float() - got_plain[t].float()).abs().max())
+
+    # prefill path with folded weights (scale-only norm) must also match
+    xp = torch.randn(1, 6, cfg.hidden_size, device="cuda", dtype=torch.bfloat16) * 0.5
+    pr = ref_blks[0](xp)
+    pf = fold_blks[0](xp)
+    assert torch.allclose(pf.float(), pr.float(), atol=0.06, rtol=0.06), (
+        (pf.float() - pr.float()).abs().max())
+
+
+@requires_gpu
 def test_nf4_block_decode_batch_beyond_kernel_cap(hip):
     """Batch 6 NF4 decode: the gemv kernel caps at BATCH=4, so the block
     splits into sub-batches (decode_step_auto) instead of falling back to
