@@ -52,7 +52,9 @@ __global__ __launch_bounds__(DWAVES * 64) void mfma_decode_kernel(
     int kv_heads,
     int lmax,
     int n_splits,
-    float scale) {
+    float scale,
+    float* __restrict__ out_direct) {  // n_splits==1: normalized out [B, KV*GQ*HD]
+                                       // written here; the combine kernel is skipped
   const int bkv = blockIdx.x;
   const int kvh = bkv % kv_heads;
   const int split = blockIdx.y;
@@ -237,10 +239,17 @@ __global__ __launch_bounds__(DWAVES * 64) void mfma_decode_kernel(
           osum += wgt * reinterpret_cast<const float*>(vt_raw[w])[gl * HD + d];
           lsum += wgt * c_ml[w][gl][1];
         }
-        po[idx] = osum;
-        if (d == 0) {
-          pml[gl * 2 + 0] = m_star;
-          pml[gl * 2 + 1] = lsum;
+        if (out_direct) {
+          // single split: normalize here (matches the combine kernel's
+          // n_splits==1 output exactly) and skip that kernel entirely —
+          // at short kv both kernels are pure launch latency
+          out_direct[((size_t)bkv * GQ + g0 + gl) * HD + d] = osum / fmaxf(lsum, 1e-20f);
+        } else {
+          po[idx] = osum;
+          if (d == 0) {
+            pml[gl * 2 + 0] = m_star;
+            pml[gl * 2 + 1] = lsum;
+          }
         }
       }
     }
@@ -446,12 +455,23 @@ torch::Tensor attn_decode_fused(
       const int by_occ = std::min(768 / bkv, lmax / 512);
       const int by_min = std::min(256 / bkv, (lmax + 127) / 128);
       n_splits = std::max(1, std::max(by_occ, by_min));
+      // measured dead end (profiles/attn_direct_ab.log): forcing one split
+      // + in-kernel normalization (combine kernel skipped) at short caches
+      // LOST 1.4 tok/s end to end — halving the split occupancy costs more
+      // than the removed launch. Off by default; the env knob keeps the
+      // single-split direct-out path selectable for re-evaluation.
+      static const int short_max = [] {
+        const char* e = std::getenv("PETALS_AMD_DECODE_SHORT_LMAX");
+        return e ? std::atoi(e) : 0;
+      }();
+      if (lmax <= short_max) n_splits = 1;
     }
   }
   auto opts = q.options();
-  if (part_o.numel() < (int64_t)b * kv_heads * n_splits * GQi * hd)
+  const bool will_direct = !use_valu && n_splits == 1;  // parts go unused
+  if (!will_direct && part_o.numel() < (int64_t)b * kv_heads * n_splits * GQi * hd)
     part_o = torch::empty({(int64_t)b * kv_heads, n_splits, GQi, hd}, opts);
-  if (part_ml.numel() < (int64_t)b * kv_heads * n_splits * GQi * 2)
+  if (!will_direct && part_ml.numel() < (int64_t)b * kv_heads * n_splits * GQi * 2)
     part_ml = torch::empty({(int64_t)b * kv_heads, n_splits, GQi, 2}, opts);
   auto out = torch::empty({(int64_t)b, (int64_t)kv_heads * GQi * hd}, opts);
 
@@ -471,6 +491,10 @@ torch::Tensor attn_decode_fused(
     alibi_p = alibi_slopes->data_ptr<float>();
   }
 
+  // single split on the MFMA path: the decode kernel normalizes and writes
+  // the output itself — the combine kernel (pure launch latency at short kv)
+  // is skipped entirely
+  float* direct_p = (!use_valu && n_splits == 1) ? out.data_ptr<float>() : nullptr;
   bool launched = false;
 #define ATTN_CASE(HDV, GQV)                                                   \
   if (hd == HDV && GQi == GQV) {                                              \
@@ -479,7 +503,8 @@ torch::Tensor attn_decode_fused(
           qp, kp, vp, pop, pmlp, lenp, alibi_p, kv_heads, lmax, n_splits, sc);\
     else                                                                      \
       mfma_decode_kernel<HDV, GQV><<<grid, DWAVES * WAVE, 0, stream>>>(       \
-          qp, kp, vp, pop, pmlp, lenp, alibi_p, kv_heads, lmax, n_splits, sc);\
+          qp, kp, vp, pop, pmlp, lenp, alibi_p, kv_heads, lmax, n_splits, sc, \
+          direct_p);                                                          \
     launched = true;                                                          \
   }
   ATTN_CASE(128, 1) ATTN_CASE(128, 2) ATTN_CASE(128, 4) ATTN_CASE(128, 6)
@@ -492,7 +517,8 @@ torch::Tensor attn_decode_fused(
   if (hd == HDV && GQi == GQV) {                                              \
     TORCH_CHECK(!use_valu, "VALU decode kernel does not support gq=", GQV);   \
     mfma_decode_kernel<HDV, GQV><<<grid, DWAVES * WAVE, 0, stream>>>(         \
-        qp, kp, vp, pop, pmlp, lenp, alibi_p, kv_heads, lmax, n_splits, sc);  \
+        qp, kp, vp, pop, pmlp, lenp, alibi_p, kv_heads, lmax, n_splits, sc,   \
+        direct_p);                                                            \
     launched = true;                                                          \
   }
   ATTN_CASE_BIG(64, 29) ATTN_CASE_BIG(64, 71) ATTN_CASE_BIG(64, 32)
@@ -501,9 +527,11 @@ torch::Tensor attn_decode_fused(
   TORCH_CHECK(launched, "unsupported (head_dim, gqa) = (", hd, ", ", GQi, ")");
   HIP_CHECK_LAST();
 
-  dim3 cgrid(b * kv_heads, GQi);
-  attn_decode_combine_kernel<<<cgrid, std::min(hd, 256), 0, stream>>>(
-      pop, pmlp, out.data_ptr<float>(), n_splits, GQi, hd);
-  HIP_CHECK_LAST();
+  if (direct_p == nullptr) {
+    dim3 cgrid(b * kv_heads, GQi);
+    attn_decode_combine_kernel<<<cgrid, std::min(hd, 256), 0, stream>>>(
+        pop, pmlp, out.data_ptr<float>(), n_splits, GQi, hd);
+    HIP_CHECK_LAST();
+  }
   return out;
 }
