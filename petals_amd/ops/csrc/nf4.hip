@@ -95,7 +95,12 @@ __global__ void nf4_dequant_kernel(
 // the kernel derives inv_rms from producer-side per-wg sum(h^2) partials
 // (gemv_reduce sumsq_out), replacing the standalone norm kernel entirely (the
 // norm WEIGHT is pre-folded into the quantized weight rows at load).
-template <int BATCH, int OPL, bool LUTBF, bool XS = false>
+// LREP: bank-shifted replicas of the pair LUT. PMC measured 4.5 bank
+// conflicts PER LDS INSTRUCTION on the single-copy LUT (64 random byte
+// indices into 256 float2 entries = 2 KB over 32 banks-pairs); each replica
+// is padded to 260 entries so copies land 8 banks apart, and lane l reads
+// copy l & (LREP-1).
+template <int BATCH, int OPL, bool LUTBF, bool XS = false, int LREP = 1>
 __global__ __launch_bounds__(WAVE) void gemv_nf4_kernel(
     const unsigned char* __restrict__ packed,   // [in, out/2]
     const unsigned short* __restrict__ absmax,  // [in, out/64]
@@ -115,17 +120,20 @@ __global__ __launch_bounds__(WAVE) void gemv_nf4_kernel(
   // PAIR LUT: one read dequantizes a whole packed byte (two elements),
   // halving LDS traffic vs per-nibble lookups; 256 entries.
   // (the earlier 16-float bank-replicated LUT was LDS-issue bound)
-  __shared__ float2 lut2[LUTBF ? 1 : 256];
+  __shared__ float2 lut2[LUTBF ? 1 : LREP][LUTBF ? 1 : 260];  // 260: replicas 8 banks apart
   __shared__ unsigned int lutp[LUTBF ? 256 : 1];
   for (int i = threadIdx.x; i < 256; i += blockDim.x) {
     if constexpr (LUTBF) {
       lutp[i] = ((unsigned int)f32_to_bf16(NF4_LUT_C[i >> 4]) << 16) |
                 (unsigned int)f32_to_bf16(NF4_LUT_C[i & 0xF]);
     } else {
-      lut2[i] = make_float2(NF4_LUT_C[i & 0xF], NF4_LUT_C[i >> 4]);
+#pragma unroll
+      for (int c = 0; c < LREP; ++c)
+        lut2[c][i] = make_float2(NF4_LUT_C[i & 0xF], NF4_LUT_C[i >> 4]);
     }
   }
   __syncthreads();
+  const int lrep = threadIdx.x & (LREP - 1);
 
   auto lut_pair = [&](unsigned int byte) -> float2 {
     if constexpr (LUTBF) {
@@ -135,7 +143,7 @@ __global__ __launch_bounds__(WAVE) void gemv_nf4_kernel(
       r.y = __uint_as_float(wp & 0xFFFF0000u);  // high nibble
       return r;
     } else {
-      return lut2[byte];
+      return lut2[lrep][byte];
     }
   };
 
@@ -376,6 +384,10 @@ torch::Tensor gemv_nf4(
   TORCH_CHECK(!xs_mode || opl == 8, "folded-norm gemv is built for OPL=8");
   TORCH_CHECK(!xs_mode || x_parts->size(1) <= 2 * WAVE,
               "folded-norm gemv prefetches at most 128 sumsq partials");
+  static const int lut_rep = [] {
+    const char* s = std::getenv("PETALS_NF4_LUT_REP");
+    return s ? std::atoi(s) : 1;
+  }();
   static const bool lut_bf16 = [] {
     const char* s = std::getenv("PETALS_NF4_LUT");
     return s && s[0] == 'b';  // default f32 pairs (bf16-pair measured SLOWER:
@@ -413,10 +425,16 @@ torch::Tensor gemv_nf4(
     amt_p = reinterpret_cast<const unsigned short*>(absmax_t->data_ptr());
   }
 #define LAUNCH_NF4(B, OPL, LB)                                                \
-  gemv_nf4_kernel<B, OPL, LB><<<grid, WAVE, 0, stream>>>(                     \
-      packed.data_ptr<unsigned char>(),                                       \
-      reinterpret_cast<const unsigned short*>(absmax.data_ptr()), amt_p,      \
-      x.data_ptr<float>(), partials.data_ptr<float>(), in_dim, out_dim, i_per_split)
+  if (lut_rep == 4 && OPL == 8 && !(LB))                                      \
+    gemv_nf4_kernel<B, OPL, LB, false, 4><<<grid, WAVE, 0, stream>>>(         \
+        packed.data_ptr<unsigned char>(),                                     \
+        reinterpret_cast<const unsigned short*>(absmax.data_ptr()), amt_p,    \
+        x.data_ptr<float>(), partials.data_ptr<float>(), in_dim, out_dim, i_per_split); \
+  else                                                                        \
+    gemv_nf4_kernel<B, OPL, LB><<<grid, WAVE, 0, stream>>>(                   \
+        packed.data_ptr<unsigned char>(),                                     \
+        reinterpret_cast<const unsigned short*>(absmax.data_ptr()), amt_p,    \
+        x.data_ptr<float>(), partials.data_ptr<float>(), in_dim, out_dim, i_per_split)
 #define LAUNCH_NF4_XS(B)                                                      \
   gemv_nf4_kernel<B, 8, false, true><<<grid, WAVE, 0, stream>>>(              \
       packed.data_ptr<unsigned char>(),                                       \
