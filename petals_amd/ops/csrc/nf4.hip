@@ -100,7 +100,11 @@ __global__ void nf4_dequant_kernel(
 // indices into 256 float2 entries = 2 KB over 32 banks-pairs); each replica
 // is padded to 260 entries so copies land 8 banks apart, and lane l reads
 // copy l & (LREP-1).
-template <int BATCH, int OPL, bool LUTBF, bool XS = false, int LREP = 1>
+// UN: override of the weight-load unroll depth (0 = default 16, 8 for
+// batch>4). The batch-1 kernel sits at 66 VGPRs with wait/busy ~12x
+// (profiles/pmc_gemv_shapes_r2.txt) — deeper unroll trades registers for
+// more HBM loads in flight.
+template <int BATCH, int OPL, bool LUTBF, bool XS = false, int LREP = 1, int UN = 0>
 __global__ __launch_bounds__(WAVE) void gemv_nf4_kernel(
     const unsigned char* __restrict__ packed,   // [in, out/2]
     const unsigned short* __restrict__ absmax,  // [in, out/64]
@@ -178,7 +182,7 @@ __global__ __launch_bounds__(WAVE) void gemv_nf4_kernel(
   if (full) {
     // batch > 4: halve the unroll depth — xs[BATCH][UNROLL] + acc[BATCH][OPL]
     // at full depth would blow the VGPR budget below 2 waves/SIMD
-    constexpr int UNROLL = (BATCH > 4) ? 8 : 16;
+    constexpr int UNROLL = UN ? UN : ((BATCH > 4) ? 8 : 16);
     const int half_out = out_dim >> 1;
     const unsigned char* pp = packed + (size_t)i_begin * half_out + (out0 >> 1);
     const unsigned short* amt = absmax_t ? absmax_t + (size_t)(out0 >> 6) * in_dim : nullptr;
@@ -197,13 +201,11 @@ __global__ __launch_bounds__(WAVE) void gemv_nf4_kernel(
       }
       float am[UNROLL];
       if (amt) {
-        const short8 a0 = *reinterpret_cast<const short8*>(amt + i);
 #pragma unroll
-        for (int e = 0; e < 8; ++e) am[e] = bf16_to_f32((unsigned short)a0[e]);
-        if constexpr (UNROLL == 16) {
-          const short8 a1 = *reinterpret_cast<const short8*>(amt + i + 8);
+        for (int c8 = 0; c8 < UNROLL / 8; ++c8) {
+          const short8 a8 = *reinterpret_cast<const short8*>(amt + i + c8 * 8);
 #pragma unroll
-          for (int e = 0; e < 8; ++e) am[8 + e] = bf16_to_f32((unsigned short)a1[e]);
+          for (int e = 0; e < 8; ++e) am[c8 * 8 + e] = bf16_to_f32((unsigned short)a8[e]);
         }
       } else {
 #pragma unroll
@@ -388,6 +390,10 @@ torch::Tensor gemv_nf4(
     const char* s = std::getenv("PETALS_NF4_LUT_REP");
     return s ? std::atoi(s) : 1;
   }();
+  static const int un_env = [] {
+    const char* s = std::getenv("PETALS_NF4_UNROLL");
+    return s ? std::atoi(s) : 0;
+  }();
   static const bool lut_bf16 = [] {
     const char* s = std::getenv("PETALS_NF4_LUT");
     return s && s[0] == 'b';  // default f32 pairs (bf16-pair measured SLOWER:
@@ -395,6 +401,7 @@ torch::Tensor gemv_nf4(
     // profiles/nf4_lut_sweep.log — LDS bandwidth is not the binding limit)
   }();
 
+  const int un32 = (un_env == 32 && batch <= 2 && opl == 8 && !lut_bf16) ? 32 : 0;
   const long out_waves = (out_dim + (long)WAVE * opl - 1) / ((long)WAVE * opl);
   // NF4 matrices are 4x smaller than bf16: allow chunks down to 64 input rows
   // so small projections still spread over the 256 CUs
@@ -402,9 +409,10 @@ torch::Tensor gemv_nf4(
   long max_splits = (in_dim + 31) / 32;
   if (splits > max_splits) splits = max_splits;
   if (splits < 1) splits = 1;
-  // chunks must be UNROLL(16)-aligned or the per-row fallback tail eats the
+  // chunks must be UNROLL-aligned or the per-row fallback tail eats the
   // gain (an unaligned split count put up to a third of rows on the slow path)
-  int i_per_split_aligned = (int)(((in_dim + splits - 1) / splits + 15) & ~15L);
+  const long un_mask = un32 ? 31L : 15L;
+  int i_per_split_aligned = (int)(((in_dim + splits - 1) / splits + un_mask) & ~un_mask);
   splits = (in_dim + i_per_split_aligned - 1) / i_per_split_aligned;
 
   torch::Tensor partials;
@@ -453,7 +461,18 @@ torch::Tensor gemv_nf4(
     case 7: LAUNCH_NF4(7, OPL, LB); break;                                    \
     case 8: LAUNCH_NF4(8, OPL, LB); break;                                    \
   }
-  if (xs_mode) {
+  if (un32) {
+    if (batch == 1)
+      gemv_nf4_kernel<1, 8, false, false, 1, 32><<<grid, WAVE, 0, stream>>>(
+          packed.data_ptr<unsigned char>(),
+          reinterpret_cast<const unsigned short*>(absmax.data_ptr()), amt_p,
+          x.data_ptr<float>(), partials.data_ptr<float>(), in_dim, out_dim, i_per_split);
+    else
+      gemv_nf4_kernel<2, 8, false, false, 1, 32><<<grid, WAVE, 0, stream>>>(
+          packed.data_ptr<unsigned char>(),
+          reinterpret_cast<const unsigned short*>(absmax.data_ptr()), amt_p,
+          x.data_ptr<float>(), partials.data_ptr<float>(), in_dim, out_dim, i_per_split);
+  } else if (xs_mode) {
     switch (batch) {
       case 1: LAUNCH_NF4_XS(1); break;
       case 2: LAUNCH_NF4_XS(2); break;
