@@ -166,28 +166,28 @@ def test_llama_block_fast_decode_matches_cpu(hip):
 
 
 @requires_gpu
-def test_gpu_server_e2e_generate():
-    """Tiny llama swarm: 1 GPU server (bf16, fused path) vs local CPU fp32."""
-    import tempfile
-
+@pytest.mark.parametrize("preset", ["test-llama", "test-mixtral", "test-bloom-hd64", "test-falcon-hd64"])
+def test_gpu_server_e2e_generate(preset):
+    """Tiny swarm per family: 1 GPU server (bf16, fused path + span graphs)
+    serving a real client generate()."""
     from petals_amd.dht.node import DHT
     from petals_amd.server.server import Server
     from petals_amd.utils.auto_config import AutoDistributedModelForCausalLM
 
     boot = DHT(host="127.0.0.1")
     server = Server(
-        "test-llama",
+        preset,
         initial_peers=[boot.listen_addr],
         host="127.0.0.1",
         device="cuda",
         torch_dtype="bfloat16",
         block_indices="0:4",
-        dht_prefix="gpu-e2e",
+        dht_prefix=f"gpu-e2e-{preset}",
         throughput=1.0,
     ).start()
     try:
         model = AutoDistributedModelForCausalLM.from_pretrained(
-            "test-llama", initial_peers=[boot.listen_addr], dht_prefix="gpu-e2e",
+            preset, initial_peers=[boot.listen_addr], dht_prefix=f"gpu-e2e-{preset}",
             show_route=False, max_retries=1,
         )
         ids = torch.randint(0, 128, (1, 5))
