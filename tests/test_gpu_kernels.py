@@ -166,6 +166,50 @@ def test_llama_block_fast_decode_matches_cpu(hip):
 
 
 @requires_gpu
+def test_gpu_session_rollback_start_from_position():
+    """Speculative-decoding rollback against a GPU server with span graphs:
+    rewinding the session position and re-stepping must give the same logits
+    as a fresh session fed the same prefix (KV overwritten in place)."""
+    from petals_amd.dht.node import DHT
+    from petals_amd.server.server import Server
+    from petals_amd.utils.auto_config import AutoDistributedModelForCausalLM
+
+    boot = DHT(host="127.0.0.1")
+    server = Server(
+        "test-llama", initial_peers=[boot.listen_addr], host="127.0.0.1",
+        device="cuda", torch_dtype="bfloat16", block_indices="0:4",
+        dht_prefix="gpu-rb", throughput=1.0,
+    ).start()
+    try:
+        model = AutoDistributedModelForCausalLM.from_pretrained(
+            "test-llama", initial_peers=[boot.listen_addr], dht_prefix="gpu-rb",
+            show_route=False, max_retries=1,
+        )
+        torch.manual_seed(4)
+        ids = torch.randint(0, 128, (1, 6))
+        extra = torch.randint(0, 128, (1, 3))
+        with model.transformer.h.inference_session(max_length=16) as sess:
+            with model.transformer.h.use_session(sess):
+                with torch.no_grad():
+                    model(input_ids=ids)           # position 6
+                    model(input_ids=extra)         # position 9 (to be discarded)
+                sess.position = 6                  # speculative rollback
+                with torch.no_grad():
+                    logits_rb = model(input_ids=extra[:, :1]).logits
+        with model.transformer.h.inference_session(max_length=16) as sess2:
+            with model.transformer.h.use_session(sess2):
+                with torch.no_grad():
+                    model(input_ids=ids)
+                    logits_fresh = model(input_ids=extra[:, :1]).logits
+        assert torch.allclose(logits_rb, logits_fresh, atol=2e-2, rtol=2e-2), (
+            (logits_rb - logits_fresh).abs().max())
+        model.transformer.h.sequence_manager.shutdown()
+    finally:
+        server.shutdown()
+        boot.shutdown()
+
+
+@requires_gpu
 @pytest.mark.parametrize("preset", ["test-llama", "test-mixtral", "test-bloom-hd64", "test-falcon-hd64"])
 def test_gpu_server_e2e_generate(preset):
     """Tiny swarm per family: 1 GPU server (bf16, fused path + span graphs)
