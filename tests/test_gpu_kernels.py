@@ -166,6 +166,50 @@ def test_llama_block_fast_decode_matches_cpu(hip):
 
 
 @requires_gpu
+def test_gpu_two_span_push_chain():
+    """Two GPU servers (spans 0:2 and 2:4) in one process: decode steps chain
+    server-to-server over rpc_push with per-span hipGraphs; generate must
+    produce the same ids as a single-server swarm."""
+    from petals_amd.dht.node import DHT
+    from petals_amd.server.server import Server
+    from petals_amd.utils.auto_config import AutoDistributedModelForCausalLM
+
+    boot = DHT(host="127.0.0.1")
+    servers = [
+        Server(
+            "test-llama", initial_peers=[boot.listen_addr], host="127.0.0.1",
+            device="cuda", torch_dtype="bfloat16", block_indices=spec,
+            dht_prefix="gpu-2span", throughput=1.0,
+        ).start()
+        for spec in ("0:2", "2:4")
+    ]
+    boot_single = DHT(host="127.0.0.1")
+    single = Server(
+        "test-llama", initial_peers=[boot_single.listen_addr], host="127.0.0.1",
+        device="cuda", torch_dtype="bfloat16", block_indices="0:4",
+        dht_prefix="gpu-1span", throughput=1.0,
+    ).start()
+    try:
+        torch.manual_seed(2)
+        ids = torch.randint(0, 128, (1, 5))
+        outs = {}
+        for prefix, peers in (("gpu-2span", boot), ("gpu-1span", boot_single)):
+            model = AutoDistributedModelForCausalLM.from_pretrained(
+                "test-llama", initial_peers=[peers.listen_addr], dht_prefix=prefix,
+                show_route=False, max_retries=1,
+            )
+            outs[prefix] = model.generate(ids, max_new_tokens=8, do_sample=False)
+            model.transformer.h.sequence_manager.shutdown()
+        assert torch.equal(outs["gpu-2span"], outs["gpu-1span"]), outs
+    finally:
+        for s in servers:
+            s.shutdown()
+        single.shutdown()
+        boot.shutdown()
+        boot_single.shutdown()
+
+
+@requires_gpu
 def test_gpu_session_rollback_start_from_position():
     """Speculative-decoding rollback against a GPU server with span graphs:
     rewinding the session position and re-stepping must give the same logits
