@@ -166,6 +166,58 @@ def test_llama_block_fast_decode_matches_cpu(hip):
 
 
 @requires_gpu
+def test_gpu_training_backward_grads(tmp_path):
+    """Client training fwd+bwd through a GPU server (bf16 fused autograd
+    path): input grads must track a local fp32 CPU swarm's within bf16
+    tolerance (cosine > 0.99). An HF checkpoint on disk pins identical
+    weights on both servers (the preset random init differs per device)."""
+    transformers = pytest.importorskip("transformers")
+    import os as _os
+
+    torch.manual_seed(0)
+    cfg = transformers.LlamaConfig(
+        hidden_size=64, num_hidden_layers=4, num_attention_heads=4, num_key_value_heads=2,
+        intermediate_size=128, vocab_size=128, max_position_embeddings=256,
+        tie_word_embeddings=False,
+    )
+    path = _os.path.join(str(tmp_path), "ckpt")
+    transformers.LlamaForCausalLM(cfg).eval().save_pretrained(path, safe_serialization=True)
+
+    from petals_amd.dht.node import DHT
+    from petals_amd.server.server import Server
+    from petals_amd.utils.auto_config import AutoDistributedModelForCausalLM
+
+    def run(device, dtype, prefix):
+        boot = DHT(host="127.0.0.1")
+        server = Server(
+            path, initial_peers=[boot.listen_addr], host="127.0.0.1",
+            device=device, torch_dtype=dtype, block_indices="0:4",
+            dht_prefix=prefix, throughput=1.0,
+        ).start()
+        try:
+            model = AutoDistributedModelForCausalLM.from_pretrained(
+                path, initial_peers=[boot.listen_addr], dht_prefix=prefix,
+                show_route=False, max_retries=1,
+            )
+            torch.manual_seed(8)
+            ids = torch.randint(0, 128, (1, 6))
+            e = model.transformer.embed_tokens(ids).detach().float().requires_grad_(True)
+            model(inputs_embeds=e).logits.float().square().mean().backward()
+            g = e.grad.clone()
+            model.transformer.h.sequence_manager.shutdown()
+            return g
+        finally:
+            server.shutdown()
+            boot.shutdown()
+
+    g_gpu = run("cuda", "bfloat16", "gpu-train")
+    g_cpu = run("cpu", "float32", "cpu-train")
+    assert torch.isfinite(g_gpu).all()
+    cos = torch.nn.functional.cosine_similarity(g_gpu.flatten(), g_cpu.flatten(), dim=0)
+    assert cos > 0.99, cos
+
+
+@requires_gpu
 def test_gpu_two_span_push_chain():
     """Two GPU servers (spans 0:2 and 2:4) in one process: decode steps chain
     server-to-server over rpc_push with per-span hipGraphs; generate must
