@@ -423,3 +423,19 @@ def test_chunked_prefill_matches_forward(tmp_path, hf_checkpoint):
     finally:
         server.shutdown()
         boot.shutdown()
+
+
+def test_worker_thread_mode(client_model, monkeypatch):
+    """PETALS_AMD_WORKER_THREAD=1 restores the background-thread client loop
+    (default is the inline pump); generate must still work."""
+    from petals_amd.client.remote_worker import reset_worker
+
+    monkeypatch.setenv("PETALS_AMD_WORKER_THREAD", "1")
+    reset_worker()
+    try:
+        ids = torch.randint(0, 128, (1, 4))
+        out = client_model.generate(ids, max_new_tokens=4, do_sample=False)
+        assert out.shape == (1, 8)
+    finally:
+        monkeypatch.delenv("PETALS_AMD_WORKER_THREAD")
+        reset_worker()
