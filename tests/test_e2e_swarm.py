@@ -425,17 +425,26 @@ def test_chunked_prefill_matches_forward(tmp_path, hf_checkpoint):
         boot.shutdown()
 
 
-def test_worker_thread_mode(client_model, monkeypatch):
+def test_worker_thread_mode(swarm, monkeypatch):
     """PETALS_AMD_WORKER_THREAD=1 restores the background-thread client loop
-    (default is the inline pump); generate must still work."""
+    (default is the inline pump); generate must still work. Uses a FRESH
+    client: the shared client_model may hold a resumable session from an
+    earlier test with a different batch size."""
     from petals_amd.client.remote_worker import reset_worker
+    from petals_amd.utils.auto_config import AutoDistributedModelForCausalLM
 
+    bootstrap, _, path = swarm
     monkeypatch.setenv("PETALS_AMD_WORKER_THREAD", "1")
     reset_worker()
     try:
+        model = AutoDistributedModelForCausalLM.from_pretrained(
+            path, initial_peers=[bootstrap.listen_addr], dht_prefix="test-llama-e2e",
+            show_route=False, max_retries=2, min_backoff=0.2,
+        )
         ids = torch.randint(0, 128, (1, 4))
-        out = client_model.generate(ids, max_new_tokens=4, do_sample=False)
+        out = model.generate(ids, max_new_tokens=4, do_sample=False)
         assert out.shape == (1, 8)
+        model.transformer.h.sequence_manager.shutdown()
     finally:
         monkeypatch.delenv("PETALS_AMD_WORKER_THREAD")
         reset_worker()
