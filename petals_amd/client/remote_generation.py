@@ -184,6 +184,12 @@ class RemoteGenerationMixin:
         if session is None:
             resumed = self.transformer.h.active_session
             if resumed is not None:
+                want = batch * max(num_beams, 1)
+                if getattr(resumed, "_batch_size", want) != want:
+                    raise ValueError(
+                        f"cannot resume the active inference session: it was opened with "
+                        f"batch_size={resumed._batch_size} but generate() needs {want}"
+                    )
                 ctx = contextlib.nullcontext(resumed)
             else:
                 ctx = self.transformer.h.inference_session(
