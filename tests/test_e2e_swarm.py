@@ -448,3 +448,58 @@ def test_worker_thread_mode(swarm, monkeypatch):
     finally:
         monkeypatch.delenv("PETALS_AMD_WORKER_THREAD")
         reset_worker()
+
+
+@pytest.mark.parametrize("family", ["llama", "bloom"])
+def test_sequence_classification_e2e(tmp_path, family):
+    """Distributed*ForSequenceClassification: forward through the swarm +
+    backward into the classifier head (reference model.py parity)."""
+    transformers = pytest.importorskip("transformers")
+    import os as _os
+    import time as _time
+
+    torch.manual_seed(0)
+    if family == "llama":
+        cfg = transformers.LlamaConfig(
+            hidden_size=64, num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+            intermediate_size=128, vocab_size=128, max_position_embeddings=256,
+            tie_word_embeddings=False,
+        )
+        hf = transformers.LlamaForCausalLM(cfg)
+    else:
+        cfg = transformers.BloomConfig(hidden_size=64, n_head=4, n_layer=2, vocab_size=128)
+        hf = transformers.BloomForCausalLM(cfg)
+    path = _os.path.join(str(tmp_path), "ckpt")
+    hf.eval().save_pretrained(path, safe_serialization=True)
+
+    from petals_amd.dht.node import DHT
+    from petals_amd.server.server import Server
+    from petals_amd.utils.auto_config import AutoDistributedConfig
+
+    boot = DHT(host="127.0.0.1")
+    server = Server(
+        path, initial_peers=[boot.listen_addr], host="127.0.0.1", device="cpu",
+        torch_dtype="float32", block_indices="0:2", dht_prefix=f"cls-{family}",
+        throughput=1.0,
+    ).start()
+    try:
+        _time.sleep(1)
+        if family == "llama":
+            from petals_amd.models.llama.model import DistributedLlamaForSequenceClassification as CLS
+        else:
+            from petals_amd.models.bloom.model import DistributedBloomForSequenceClassification as CLS
+        m = CLS.from_pretrained(
+            path, initial_peers=[boot.listen_addr], dht_prefix=f"cls-{family}", num_labels=3,
+            show_route=False, max_retries=3, min_backoff=0.3,
+        )
+        ids = torch.randint(0, 128, (2, 7))
+        out = m(input_ids=ids)
+        assert out.logits.shape == (2, 3)
+        loss = torch.nn.functional.cross_entropy(out.logits, torch.tensor([0, 2]))
+        loss.backward()
+        grads = [p.grad for p in m.score.parameters() if p.grad is not None]
+        assert grads and all(torch.isfinite(g).all() for g in grads)
+        m.transformer.h.sequence_manager.shutdown()
+    finally:
+        server.shutdown()
+        boot.shutdown()
