@@ -89,6 +89,8 @@ def main(argv=None):
     parser.add_argument("--announce_host", type=str, default=None)
     parser.add_argument("--public_ip", type=str, default=None, help="alias for --announce_host")
     parser.add_argument("--skip_reachability_check", action="store_true")
+    parser.add_argument("--force_relay", action="store_true",
+                        help="serve through a circuit relay even if directly reachable (NAT-path testing)")
     parser.add_argument("--no_auto_relay", action="store_true",
                         help="refuse to serve via a circuit relay when unreachable (fail instead)")
     parser.add_argument("--secure", action="store_true",
@@ -202,6 +204,7 @@ def main(argv=None):
         secure=True if args.secure else None,
         identity_path=args.identity_path,
         use_relay=not args.no_auto_relay,
+        force_relay=args.force_relay,
         max_chunk_size_bytes=args.max_chunk_size_bytes,
         max_alloc_timeout=args.max_alloc_timeout,
         request_timeout=args.request_timeout,
