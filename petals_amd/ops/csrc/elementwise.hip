@@ -423,10 +423,13 @@ __global__ void rope_cache_write_kernel(
 // rotated q to a fresh [b, qh*hd] f32 tensor and k/v (bf16) into the caches.
 // Replaces reduce + rope_cache_write (or reduce + kv_cache_write for ALiBi
 // families) — one fewer kernel per block on the decode path.
-#define QKV_RED_GROUPS 8
+#define QKV_RED_GROUPS 16  // 16 split groups x 64 outputs (1024-thread wgs):
+// halves each thread's serial split chain vs 8 — measured +0.9-1.0 tok/s on
+// every alternating pair (profiles/qrr_groups_ab.log); PETALS_QRR_GROUPS=8
+// selects the old width
 
-template <bool ROPE>
-__global__ __launch_bounds__(64 * QKV_RED_GROUPS) void qkv_rope_reduce_kernel(
+template <bool ROPE, int G_T = QKV_RED_GROUPS>
+__global__ __launch_bounds__(64 * G_T) void qkv_rope_reduce_kernel(
     const float* __restrict__ partials,  // [splits, b, row_elems]
     const float* __restrict__ cos_t,     // [max_pos, hd]
     const float* __restrict__ sin_t,
@@ -436,7 +439,7 @@ __global__ __launch_bounds__(64 * QKV_RED_GROUPS) void qkv_rope_reduce_kernel(
     unsigned short* __restrict__ k_cache,  // [bcap, kh, lmax, hd]
     unsigned short* __restrict__ v_cache,
     int n_splits, int b, int qh, int kh, int lmax, int hd) {
-  constexpr int G = QKV_RED_GROUPS;
+  constexpr int G = G_T;
   __shared__ float acc_a[G][64];
   __shared__ float acc_b[G][64];
   const int half = hd >> 1;
@@ -533,15 +536,23 @@ torch::Tensor qkv_rope_reduce(
     TORCH_CHECK(bias->numel() == partials.size(2));
     bp = reinterpret_cast<const unsigned short*>(bias->data_ptr());
   }
-#define LAUNCH_QRR(R)                                                          \
-  qkv_rope_reduce_kernel<R><<<blocks, 64 * QKV_RED_GROUPS, 0, stream>>>(       \
+  static const int qrr_g = [] {
+    const char* e = std::getenv("PETALS_QRR_GROUPS");
+    return e ? std::atoi(e) : QKV_RED_GROUPS;
+  }();
+#define LAUNCH_QRR_G(R, G)                                                     \
+  qkv_rope_reduce_kernel<R, G><<<blocks, 64 * G, 0, stream>>>(                 \
       partials.data_ptr<float>(), cp, sp, pos.data_ptr<int>(), bp,             \
       q_out.data_ptr<float>(),                                                 \
       reinterpret_cast<unsigned short*>(k_cache.data_ptr()),                   \
       reinterpret_cast<unsigned short*>(v_cache.data_ptr()),                   \
       n_splits, b, (int)qh, (int)kh, lmax, hd)
-  if (rope) LAUNCH_QRR(true); else LAUNCH_QRR(false);
-#undef LAUNCH_QRR
+  if (qrr_g == 8) {
+    if (rope) LAUNCH_QRR_G(true, 8); else LAUNCH_QRR_G(false, 8);
+  } else {
+    if (rope) LAUNCH_QRR_G(true, 16); else LAUNCH_QRR_G(false, 16);
+  }
+#undef LAUNCH_QRR_G
   HIP_CHECK_LAST();
   return q_out;
 }
