@@ -8,6 +8,7 @@
 // ROCm-native stream API: this header is consumed verbatim (torch's hipify
 // only rewrites the .hip TUs, not local headers)
 #include <c10/hip/HIPStream.h>
+#include <cstdlib>
 
 enum GemvEpilogue : int {
   EPI_PLAIN_F32 = 0,     // y_f32[b, out] = sum
@@ -47,9 +48,13 @@ static __device__ __forceinline__ float gelu_tanh_f32(float x) {
 // as the small gemvs themselves. Split groups multiply resident waves 8x and
 // cut each thread's serial chain 8x; a tiny LDS tree combines (deterministic
 // order preserved: fixed group partition, fixed add order).
-#define GEMV_REDUCE_GROUPS 8
+#define GEMV_REDUCE_GROUPS 16  // 16 split groups x 64 outputs (1024-thread
+// wgs): halves each thread's serial split chain vs the round-2 value of 8 —
+// +1.3 tok/s on every alternating A/B pair (profiles/qrr_groups_ab.log);
+// PETALS_REDUCE_GROUPS=8 selects the old width
 
-static __global__ __launch_bounds__(64 * GEMV_REDUCE_GROUPS) void gemv_reduce_kernel_impl(
+template <int G>
+static __global__ __launch_bounds__(64 * G) void gemv_reduce_kernel_impl(
     const float* __restrict__ partials,  // [n_splits, batch, out]
     const unsigned short* __restrict__ residual,  // [batch, out] or null
     const unsigned short* __restrict__ bias,      // [out] bf16 or null
@@ -62,7 +67,6 @@ static __global__ __launch_bounds__(64 * GEMV_REDUCE_GROUPS) void gemv_reduce_ke
     int batch,
     int out_dim,
     int epilogue) {
-  constexpr int G = GEMV_REDUCE_GROUPS;
   __shared__ float acc_g[G][64];
   __shared__ float acc_u[G][64];
   const int half = out_dim >> 1;
@@ -163,11 +167,18 @@ static inline torch::Tensor launch_gemv_reduce(
     ss_p = sumsq_out->data_ptr<float>();
   }
   const int total = batch * n_out;
-  const int rthreads = 64 * GEMV_REDUCE_GROUPS;
+  static const int red_g = [] {
+    const char* e = std::getenv("PETALS_REDUCE_GROUPS");
+    return e ? std::atoi(e) : GEMV_REDUCE_GROUPS;
+  }();
   int rblocks = (total + 63) / 64;
   auto stream = c10::hip::getCurrentHIPStream().stream();
-  gemv_reduce_kernel_impl<<<rblocks, rthreads, 0, stream>>>(
-      partials.data_ptr<float>(), res_p, bias_p, y.data_ptr(), ss_p, n_splits, batch, out_dim, epilogue);
+  if (red_g == 16)
+    gemv_reduce_kernel_impl<16><<<rblocks, 64 * 16, 0, stream>>>(
+        partials.data_ptr<float>(), res_p, bias_p, y.data_ptr(), ss_p, n_splits, batch, out_dim, epilogue);
+  else
+    gemv_reduce_kernel_impl<8><<<rblocks, 64 * 8, 0, stream>>>(
+        partials.data_ptr<float>(), res_p, bias_p, y.data_ptr(), ss_p, n_splits, batch, out_dim, epilogue);
   HIP_CHECK_LAST();
   return y;
 }
